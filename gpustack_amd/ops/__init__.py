@@ -1,0 +1,148 @@
+"""gpustack_amd.ops — dispatch layer for the CDNA4 HIP kernels.
+
+On a GPU box the in-tree `_hip_ops.so` (hand-written gfx950 kernels) is
+REQUIRED: there is no silent eager fallback — a missing extension raises at
+first use so a broken build can't masquerade as the native path. On CPU-only
+hosts the fp32 torch reference implementations run instead, which keeps the
+engine/control-plane logic testable without hardware.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import torch_ref
+
+_HIP = None
+_HIP_ERR: str | None = None
+
+
+def _load_hip():
+    global _HIP, _HIP_ERR
+    if _HIP is not None or _HIP_ERR is not None:
+        return _HIP
+    try:
+        import importlib.util
+        from pathlib import Path
+
+        so = Path(__file__).resolve().parent / "_hip_ops.so"
+        if not so.exists():
+            raise ImportError(
+                f"{so} not found — build it with `python -m gpustack_amd.ops.build` "
+                "(hipcc --offload-arch=gfx950)"
+            )
+        spec = importlib.util.spec_from_file_location("gpustack_amd.ops._hip_ops", so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _HIP = mod
+    except Exception as e:  # noqa: BLE001
+        _HIP_ERR = str(e)
+        raise
+    return _HIP
+
+
+def hip_available() -> bool:
+    if not torch.cuda.is_available():
+        return False
+    try:
+        return _load_hip() is not None
+    except Exception:  # noqa: BLE001
+        return False
+
+
+def _backend(t: torch.Tensor):
+    if t.is_cuda:
+        return _load_hip()  # raises loudly if the native build is missing
+    return None
+
+
+# --- op surface -----------------------------------------------------------
+
+def rms_norm(out, x, weight, eps: float) -> None:
+    hip = _backend(x)
+    if hip is not None:
+        hip.rms_norm(out, x, weight, eps)
+    else:
+        torch_ref.rms_norm(out, x, weight, eps)
+
+
+def fused_add_rms_norm(x, residual, weight, eps: float) -> None:
+    hip = _backend(x)
+    if hip is not None:
+        hip.fused_add_rms_norm(x, residual, weight, eps)
+    else:
+        torch_ref.fused_add_rms_norm(x, residual, weight, eps)
+
+
+def rotary_embedding(positions, q, k, cos_sin, head_dim: int, rot_dim: int) -> None:
+    hip = _backend(q)
+    if hip is not None:
+        hip.rotary_embedding(positions, q, k, cos_sin, head_dim, rot_dim)
+    else:
+        torch_ref.rotary_embedding(positions, q, k, cos_sin, head_dim, rot_dim)
+
+
+def silu_and_mul(out, x) -> None:
+    hip = _backend(x)
+    if hip is not None:
+        hip.silu_and_mul(out, x)
+    else:
+        torch_ref.silu_and_mul(out, x)
+
+
+def reshape_and_cache(k, v, k_cache, v_cache, slots) -> None:
+    hip = _backend(k)
+    if hip is not None:
+        hip.reshape_and_cache(k, v, k_cache, v_cache, slots)
+    else:
+        torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slots)
+
+
+def greedy_sample(logits) -> torch.Tensor:
+    out = torch.empty(logits.shape[0], dtype=torch.long, device=logits.device)
+    hip = _backend(logits)
+    if hip is not None:
+        hip.greedy_sample(out, logits)
+    else:
+        torch_ref.greedy_sample(out, logits)
+    return out
+
+
+def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale: float) -> None:
+    hip = _backend(q)
+    if hip is not None:
+        hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+    else:
+        torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+
+
+_PREFILL_BQ = 64
+
+
+def build_prefill_tiles(seq_lens: list[int], device) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Map varlen sequences to fixed 64-row q tiles for the MFMA kernel."""
+    starts, q0s, lens = [], [], []
+    tok = 0
+    for L in seq_lens:
+        for q0 in range(0, L, _PREFILL_BQ):
+            starts.append(tok)
+            q0s.append(q0)
+            lens.append(L)
+        tok += L
+    mk = lambda a: torch.tensor(a, dtype=torch.int32, device=device)
+    return mk(starts), mk(q0s), mk(lens)
+
+
+def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float) -> None:
+    hip = _backend(q)
+    if hip is not None:
+        ts, tq, tl = build_prefill_tiles(seq_lens, q.device)
+        hip.flash_prefill(out, q, k, v, ts, tq, tl, scale)
+    else:
+        torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale)
+
+
+def mfma_probe(a, b):
+    return _load_hip().mfma_probe(a, b)
+
+
+build_cos_sin_cache = torch_ref.build_cos_sin_cache

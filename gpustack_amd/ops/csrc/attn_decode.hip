@@ -1,0 +1,195 @@
+// Paged-attention decode for MI355X (gfx950) — the TPOT-dominant kernel.
+//
+// Single new token per sequence attends over the paged KV cache.
+// Replaces what the reference delegates to vLLM's paged_attention kernels
+// (SURVEY.md §2.9 #1); designed for CDNA4 rather than ported:
+//
+//  - one workgroup per (sequence, kv_head); GQA q-heads (GQ = Hq/Hkv) share
+//    the K/V stream so each KV byte is read exactly once per step.
+//  - 4 waves * 4 lane-groups of 16 = 16 independent online-softmax streams;
+//    a 16-lane group owns one token per iteration, lane = 8 head dims
+//    (16 B = u16x8 loads -> 1 KiB per wave per iteration, fully coalesced
+//    against the [block, kv_head, block_size, D] pool layout).
+//  - streams merge via 64-wide shuffles (cross-group) then LDS (cross-wave);
+//    accumulation entirely in f32.
+//
+// KV pool layout: [num_blocks, Hkv, BS, D] bf16, BS = 16 tokens.
+#include "common.h"
+
+namespace {
+
+constexpr int BS = 16;       // tokens per KV block (page)
+constexpr int NWAVES = 4;    // waves per workgroup
+constexpr int THREADS = NWAVES * WAVE_SIZE;
+
+template <int D, int GQ>
+__global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
+    unsigned short* __restrict__ out,        // [N, Hq, D]
+    const unsigned short* __restrict__ q,    // [N, Hq, D]
+    const unsigned short* __restrict__ kc,   // [B, Hkv, BS, D]
+    const unsigned short* __restrict__ vc,   // [B, Hkv, BS, D]
+    const int* __restrict__ block_tables,    // [N, max_blocks]
+    const int* __restrict__ seq_lens,        // [N]
+    int Hkv, int max_blocks, float scale) {
+  constexpr int LPG = 16;           // lanes per token-group
+  constexpr int DV = D / LPG;       // dims per lane (8 for D=128)
+  static_assert(DV == 8, "decode kernel assumes D = 128");
+
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;        // kv head
+  const int Hq = Hkv * GQ;
+  const int len = seq_lens[seq];
+  const int npages = (len + BS - 1) / BS;
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int g = lane >> 4;         // token-group within wave [0,4)
+  const int sub = lane & 15;       // dim-slice within group [0,16)
+
+  // Load q rows for this kv head's GQ query heads (f32 registers).
+  float qr[GQ][DV];
+#pragma unroll
+  for (int gq = 0; gq < GQ; ++gq) {
+    const unsigned short* qp =
+        q + ((long)seq * Hq + (long)h * GQ + gq) * D + sub * DV;
+    u16x8 u = *reinterpret_cast<const u16x8*>(qp);
+    bf8_to_f32(u, qr[gq]);
+  }
+
+  float m[GQ], s[GQ], acc[GQ][DV];
+#pragma unroll
+  for (int gq = 0; gq < GQ; ++gq) {
+    m[gq] = -INFINITY;
+    s[gq] = 0.f;
+#pragma unroll
+    for (int j = 0; j < DV; ++j) acc[gq][j] = 0.f;
+  }
+
+  const int* bt = block_tables + (long)seq * max_blocks;
+  for (int page = wave; page < npages; page += NWAVES) {
+    const long blk = bt[page];
+    const unsigned short* kbase = kc + ((blk * Hkv + h) * BS) * D;
+    const unsigned short* vbase = vc + ((blk * Hkv + h) * BS) * D;
+#pragma unroll
+    for (int it = 0; it < BS / 4; ++it) {
+      const int tok = it * 4 + g;
+      const bool valid = page * BS + tok < len;
+      // Coalesced: the wave's 64 lanes cover 4 tokens x 128 dims = 1 KiB.
+      u16x8 ku = *reinterpret_cast<const u16x8*>(kbase + tok * D + sub * DV);
+      u16x8 vu = *reinterpret_cast<const u16x8*>(vbase + tok * D + sub * DV);
+      float kf[DV], vf[DV];
+      bf8_to_f32(ku, kf);
+      bf8_to_f32(vu, vf);
+#pragma unroll
+      for (int gq = 0; gq < GQ; ++gq) {
+        float d = 0.f;
+#pragma unroll
+        for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[j];
+        d = group16_reduce_sum(d) * scale;   // all 16 lanes get the full dot
+        if (valid) {
+          const float nm = fmaxf(m[gq], d);
+          const float corr = __expf(m[gq] - nm);  // exp(-inf - finite) = 0
+          const float p = __expf(d - nm);
+          s[gq] = s[gq] * corr + p;
+#pragma unroll
+          for (int j = 0; j < DV; ++j)
+            acc[gq][j] = acc[gq][j] * corr + p * vf[j];
+          m[gq] = nm;
+        }
+      }
+    }
+  }
+
+  // Merge the 4 token-group streams within each wave (butterfly over
+  // lane masks 16 and 32). Lanes with equal `sub` end up identical.
+#pragma unroll
+  for (int mask = 16; mask <= 32; mask <<= 1) {
+#pragma unroll
+    for (int gq = 0; gq < GQ; ++gq) {
+      const float om = __shfl_xor(m[gq], mask, WAVE_SIZE);
+      const float os = __shfl_xor(s[gq], mask, WAVE_SIZE);
+      float oacc[DV];
+#pragma unroll
+      for (int j = 0; j < DV; ++j)
+        oacc[j] = __shfl_xor(acc[gq][j], mask, WAVE_SIZE);
+      const float nm = fmaxf(m[gq], om);
+      if (os > 0.f || s[gq] > 0.f) {
+        const float w1 = (s[gq] > 0.f) ? __expf(m[gq] - nm) : 0.f;
+        const float w2 = (os > 0.f) ? __expf(om - nm) : 0.f;
+        s[gq] = s[gq] * w1 + os * w2;
+#pragma unroll
+        for (int j = 0; j < DV; ++j)
+          acc[gq][j] = acc[gq][j] * w1 + oacc[j] * w2;
+        m[gq] = nm;
+      }
+    }
+  }
+
+  // Cross-wave merge through LDS.
+  __shared__ float lds_acc[NWAVES][GQ][D];
+  __shared__ float lds_m[NWAVES][GQ];
+  __shared__ float lds_s[NWAVES][GQ];
+  if (g == 0) {  // one representative lane-group per wave
+#pragma unroll
+    for (int gq = 0; gq < GQ; ++gq) {
+#pragma unroll
+      for (int j = 0; j < DV; ++j) lds_acc[wave][gq][sub * DV + j] = acc[gq][j];
+      if (sub == 0) {
+        lds_m[wave][gq] = m[gq];
+        lds_s[wave][gq] = s[gq];
+      }
+    }
+  }
+  __syncthreads();
+
+  // 256 threads cover (gq, d) output elements.
+  for (int u = threadIdx.x; u < GQ * D; u += THREADS) {
+    const int gq = u / D;
+    const int d = u % D;
+    float M = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < NWAVES; ++w)
+      if (lds_s[w][gq] > 0.f) M = fmaxf(M, lds_m[w][gq]);
+    float num = 0.f, den = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVES; ++w) {
+      if (lds_s[w][gq] > 0.f) {
+        const float wt = __expf(lds_m[w][gq] - M);
+        num += wt * lds_acc[w][gq][d];
+        den += wt * lds_s[w][gq];
+      }
+    }
+    out[((long)seq * Hq + (long)h * GQ + gq) * D + d] =
+        f2bf(den > 0.f ? num / den : 0.f);
+  }
+}
+
+}  // namespace
+
+void paged_attn_decode_launch(void* out, const void* q, const void* kc,
+                              const void* vc, const int* block_tables,
+                              const int* seq_lens, int N, int Hq, int Hkv,
+                              int D, int max_blocks, float scale,
+                              int* err_unsupported, hipStream_t s) {
+  const int GQ = Hq / Hkv;
+  dim3 grid(N, Hkv);
+  dim3 block(THREADS);
+  *err_unsupported = 0;
+  if (D != 128) { *err_unsupported = 1; return; }
+#define LAUNCH_GQ(G)                                                         \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<128, G>), grid, block, 0, s,  \
+                     (unsigned short*)out, (const unsigned short*)q,         \
+                     (const unsigned short*)kc, (const unsigned short*)vc,   \
+                     block_tables, seq_lens, Hkv, max_blocks, scale)
+  switch (GQ) {
+    case 1: LAUNCH_GQ(1); break;
+    case 2: LAUNCH_GQ(2); break;
+    case 4: LAUNCH_GQ(4); break;
+    case 5: LAUNCH_GQ(5); break;
+    case 6: LAUNCH_GQ(6); break;
+    case 7: LAUNCH_GQ(7); break;
+    case 8: LAUNCH_GQ(8); break;
+    default: *err_unsupported = 1; return;
+  }
+#undef LAUNCH_GQ
+}

@@ -1,0 +1,276 @@
+// Varlen causal flash-attention prefill for MI355X (gfx950) on MFMA.
+//
+// Replaces the prefill attention the reference gets from vLLM/SGLang
+// (SURVEY.md §2.9 #1). CDNA4-first structure (not a CUDA port):
+//
+//  - mfma_f32_16x16x32_bf16 tiles; wavefront-64 fragment layouts:
+//      A-frag: lane holds A[row = l&15][k = (l>>4)*8 + j]   (u16x8)
+//      B-frag: lane holds B[k = (l>>4)*8 + j][col = l&15]   (u16x8)
+//      C/D  : lane holds D[row = (l>>4)*4 + r][col = l&15]  (f32x4)
+//  - swapped QK^T (S^T = K · Q^T) so BOTH operands are row-contiguous
+//    u16x8 reads (K from LDS, Q from registers) — the HipKittens-style
+//    trick described in the CDNA4 guide (§B fused attention).
+//  - K tile LDS-staged with the ((row&7)<<4) byte-XOR swizzle (guide §6 G4:
+//    row-major [32][128] bf16 would be a 16-way bank conflict on b128 reads).
+//  - V tile staged TRANSPOSED (VT[128][32+4]) so PV B-fragments are
+//    contiguous b128 reads; P goes through a small per-wave LDS buffer to
+//    re-distribute S^T's lane layout into PV's A-fragment layout.
+//  - online softmax entirely in f32 registers; stats per q-row shared
+//    across lanes via 64-wide shuffles.
+//
+// Tiling: BQ = 64 q rows per workgroup (4 waves x 16 rows), BK = 32 kv.
+// v0 computes attention over the prefill chunk's own contiguous K/V
+// (full-prompt prefill); chunked prefill against the paged pool reuses the
+// decode path.
+#include "common.h"
+
+namespace {
+
+constexpr int BQ = 64;
+constexpr int BK = 32;
+constexpr int PF_D = 128;
+constexpr int PF_WAVES = 4;
+constexpr int PF_THREADS = PF_WAVES * WAVE_SIZE;
+constexpr int VT_PAD = 4;  // elements; breaks the 16-row bank cycle
+
+typedef __attribute__((ext_vector_type(8))) short s16x8;
+
+DEVICE_INLINE f32x4 mfma16x16x32_bf16(u16x8 a, u16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+      __builtin_bit_cast(s16x8, a), __builtin_bit_cast(s16x8, b), c, 0, 0, 0);
+}
+
+// byte offset of (kv, dbyte) in the swizzled K tile
+DEVICE_INLINE int kswz(int kv, int dbyte) {
+  return kv * (PF_D * 2) + (dbyte ^ ((kv & 7) << 4));
+}
+
+__global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
+    unsigned short* __restrict__ out,      // [T, Hq, D]
+    const unsigned short* __restrict__ q,  // [T, Hq, D]
+    const unsigned short* __restrict__ k,  // [T, Hkv, D]
+    const unsigned short* __restrict__ v,  // [T, Hkv, D]
+    const int* __restrict__ tile_start,    // [ntiles] seq start (global row)
+    const int* __restrict__ tile_q0,       // [ntiles] q-tile offset in seq
+    const int* __restrict__ tile_len,      // [ntiles] seq length
+    int Hq, int Hkv, float scale) {
+  const int tile = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (Hq / Hkv);
+  const int seq0 = tile_start[tile];
+  const int q0 = tile_q0[tile];
+  const int len = tile_len[tile];
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lc = lane & 15;        // column / row-id within 16
+  const int lg = lane >> 4;        // 4-lane group id
+
+  __shared__ unsigned short Kl[BK * PF_D];            // swizzled
+  __shared__ unsigned short VTl[PF_D][BK + VT_PAD];   // transposed V
+  __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
+
+  // Hoist this wave's 16 q rows into B-fragments (4 k-chunks of 32).
+  const int qrow_local = q0 + wave * 16 + lc;   // this lane's B-frag q row
+  const int qrow_clamped = (qrow_local < len) ? qrow_local : (len - 1);
+  u16x8 qfrag[4];
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk) {
+    const unsigned short* qp = q + ((long)(seq0 + qrow_clamped) * Hq + qh) * PF_D +
+                               kk * 32 + lg * 8;
+    qfrag[kk] = *reinterpret_cast<const u16x8*>(qp);
+  }
+
+  float mcol = -INFINITY;  // running max for q row `lc` (this wave)
+  float lcol = 0.f;        // running denom for q row `lc`
+  f32x4 o[PF_D / 16];      // O[q=(lg*4+r)][d=lc+nt*16]
+#pragma unroll
+  for (int nt = 0; nt < PF_D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int q_hi = q0 + BQ - 1;                    // last q row of this block
+  const int kv_end = min(len, q_hi + 1);           // causal bound
+  const int ntiles_kv = (kv_end + BK - 1) / BK;
+  const int wave_q_hi = q0 + wave * 16 + 15;       // this wave's causal bound
+
+  for (int kt = 0; kt < ntiles_kv; ++kt) {
+    const int kv0 = kt * BK;
+    // ---- cooperative staging: K (swizzled) and V^T ----
+    __syncthreads();
+    {
+      // K: 512 u16x8 chunks over 2 rounds; unit -> (kv, d0)
+      for (int u = threadIdx.x; u < BK * (PF_D / 8); u += PF_THREADS) {
+        const int kv = u / (PF_D / 8);
+        const int d0 = (u % (PF_D / 8)) * 8;
+        u16x8 val{0, 0, 0, 0, 0, 0, 0, 0};
+        if (kv0 + kv < len) {
+          val = *reinterpret_cast<const u16x8*>(
+              k + ((long)(seq0 + kv0 + kv) * Hkv + kvh) * PF_D + d0);
+        }
+        *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
+                                  kswz(kv, d0 * 2)) = val;
+      }
+      // V^T: unit -> (d, kv-chunk of 8); lane-contiguous d for coalescing
+      for (int u = threadIdx.x; u < PF_D * (BK / 8); u += PF_THREADS) {
+        const int d = u % PF_D;
+        const int kvc = (u / PF_D) * 8;
+        unsigned short tmp[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int kv = kv0 + kvc + j;
+          tmp[j] = (kv < len)
+                       ? v[((long)(seq0 + kv) * Hkv + kvh) * PF_D + d]
+                       : (unsigned short)0;
+        }
+        *reinterpret_cast<u16x8*>(&VTl[d][kvc]) =
+            *reinterpret_cast<u16x8*>(tmp);
+      }
+    }
+    __syncthreads();
+
+    if (kv0 > wave_q_hi) continue;  // fully masked for this wave
+
+    // ---- S^T = K · Q^T  (2 stiles x 4 k-chunks of MFMA) ----
+    f32x4 st[2];
+    st[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+    st[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int stile = 0; stile < 2; ++stile) {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const u16x8 a = *reinterpret_cast<const u16x8*>(
+            reinterpret_cast<char*>(Kl) +
+            kswz(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
+        st[stile] = mfma16x16x32_bf16(a, qfrag[kk], st[stile]);
+      }
+    }
+
+    // ---- mask + online softmax (stats per q row lc) ----
+    const int qpos = q0 + wave * 16 + lc;  // q row this lane's stats cover
+    float sv[8];
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int stile = 0; stile < 2; ++stile) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kvpos = kv0 + stile * 16 + lg * 4 + r;
+        float x = st[stile][r] * scale;
+        const bool ok = (kvpos <= qpos) && (kvpos < len) && (qpos < len);
+        x = ok ? x : -INFINITY;
+        sv[stile * 4 + r] = x;
+        tmax = fmaxf(tmax, x);
+      }
+    }
+#pragma unroll
+    for (int msk = 16; msk <= 32; msk <<= 1)
+      tmax = fmaxf(tmax, __shfl_xor(tmax, msk, WAVE_SIZE));
+
+    const float nm = fmaxf(mcol, tmax);
+    float corr = 1.f, tsum = 0.f;
+    float pv[8];
+    if (nm != -INFINITY) {
+      corr = __expf(mcol - nm);  // mcol = -inf -> 0
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        pv[i] = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - nm);
+        tsum += pv[i];
+      }
+      mcol = nm;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) pv[i] = 0.f;
+    }
+#pragma unroll
+    for (int msk = 16; msk <= 32; msk <<= 1)
+      tsum += __shfl_xor(tsum, msk, WAVE_SIZE);
+    lcol = lcol * corr + tsum;
+
+    // ---- stage P (bf16) into this wave's LDS buffer: P[q=lc][kv] ----
+#pragma unroll
+    for (int stile = 0; stile < 2; ++stile) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        Pl[wave][lc][stile * 16 + lg * 4 + r] = f2bf(pv[stile * 4 + r]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- rescale O by this tile's correction (row-matched via shuffle) ----
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int orow = lg * 4 + r;
+      const float c = __shfl(corr, orow, WAVE_SIZE);
+#pragma unroll
+      for (int nt = 0; nt < PF_D / 16; ++nt) o[nt][r] *= c;
+    }
+
+    // ---- O += P · V  (A = P from LDS, B = V^T rows from LDS) ----
+    const u16x8 pa =
+        *reinterpret_cast<const u16x8*>(&Pl[wave][lc][lg * 8]);
+#pragma unroll
+    for (int nt = 0; nt < PF_D / 16; ++nt) {
+      const u16x8 b =
+          *reinterpret_cast<const u16x8*>(&VTl[nt * 16 + lc][lg * 8]);
+      o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
+    }
+  }
+
+  // ---- epilogue: normalize rows and write ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int orow = lg * 4 + r;
+    const float denom = __shfl(lcol, orow, WAVE_SIZE);
+    const int qrow = q0 + wave * 16 + orow;
+    if (qrow >= len || denom <= 0.f) continue;
+    const float inv = 1.f / denom;
+#pragma unroll
+    for (int nt = 0; nt < PF_D / 16; ++nt) {
+      out[((long)(seq0 + qrow) * Hq + qh) * PF_D + nt * 16 + lc] =
+          f2bf(o[nt][r] * inv);
+    }
+  }
+}
+
+}  // namespace
+
+void flash_prefill_launch(void* out, const void* q, const void* k,
+                          const void* v, const int* tile_start,
+                          const int* tile_q0, const int* tile_len, int ntiles,
+                          int Hq, int Hkv, int D, float scale,
+                          int* err_unsupported, hipStream_t s) {
+  *err_unsupported = 0;
+  if (D != 128 || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  dim3 grid(ntiles, Hq);
+  hipLaunchKernelGGL(flash_prefill_kernel, grid, dim3(PF_THREADS), 0, s,
+                     (unsigned short*)out, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v,
+                     tile_start, tile_q0, tile_len, Hq, Hkv, scale);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe: D = A[16x32] * B[32x16] with the fragment layouts this
+// file assumes. The GPU unit test checks it against torch.matmul with random
+// asymmetric inputs (guide G9: transpose-detecting correctness checks).
+// ---------------------------------------------------------------------------
+namespace {
+__global__ void mfma_probe_kernel(float* __restrict__ d,
+                                  const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b) {
+  const int lane = threadIdx.x & 63;
+  const int lc = lane & 15, lg = lane >> 4;
+  u16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[lc * 32 + lg * 8 + j];   // A[row=lc][k=lg*8+j]
+    bf[j] = b[(lg * 8 + j) * 16 + lc]; // B[k=lg*8+j][col=lc]
+  }
+  f32x4 c{0.f, 0.f, 0.f, 0.f};
+  c = mfma16x16x32_bf16(af, bf, c);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(lg * 4 + r) * 16 + lc] = c[r];
+}
+}  // namespace
+
+void mfma_probe_launch(float* d, const void* a, const void* b, hipStream_t s) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, s, d,
+                     (const unsigned short*)a, (const unsigned short*)b);
+}

@@ -1,0 +1,166 @@
+// Torch bindings for the gpustack_amd HIP ops (native ROCm: no hipify,
+// streams come straight from c10::hip).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+// launchers (defined in the .hip translation units)
+void rms_norm_launch(void*, const void*, const void*, float, int, int, hipStream_t);
+void fused_add_rms_norm_launch(void*, void*, const void*, float, int, int, hipStream_t);
+void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, int, int, hipStream_t);
+void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
+void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, hipStream_t);
+void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
+void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, int*, hipStream_t);
+void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, int*, hipStream_t);
+void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
+
+#define HIP_CHECK_LAST()                                                     \
+  do {                                                                       \
+    hipError_t e_ = hipGetLastError();                                       \
+    TORCH_CHECK(e_ == hipSuccess, "HIP kernel launch failed: ",              \
+                hipGetErrorString(e_));                                      \
+  } while (0)
+
+namespace {
+
+hipStream_t cur_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
+  check_bf16(out, "out"); check_bf16(input, "input"); check_bf16(weight, "weight");
+  const int H = input.size(-1);
+  const long T = input.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  rms_norm_launch(out.data_ptr(), input.data_ptr(), weight.data_ptr(),
+                  (float)eps, (int)T, H, cur_stream(input));
+  HIP_CHECK_LAST();
+}
+
+void fused_add_rms_norm(at::Tensor x, at::Tensor residual, at::Tensor weight,
+                        double eps) {
+  check_bf16(x, "x"); check_bf16(residual, "residual"); check_bf16(weight, "weight");
+  const int H = x.size(-1);
+  const long T = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  fused_add_rms_norm_launch(x.data_ptr(), residual.data_ptr(),
+                            weight.data_ptr(), (float)eps, (int)T, H,
+                            cur_stream(x));
+  HIP_CHECK_LAST();
+}
+
+void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
+                      at::Tensor cos_sin, long head_dim, long rot_dim) {
+  check_bf16(q, "q"); check_bf16(k, "k");
+  TORCH_CHECK(positions.scalar_type() == at::kLong && positions.is_contiguous());
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
+  const int T = positions.size(0);
+  const int D = (int)head_dim, R = (int)rot_dim;
+  const int Hq = q.numel() / ((long)T * D);
+  const int Hk = k.numel() / ((long)T * D);
+  TORCH_CHECK((R / 2) % 8 == 0, "rot_dim/2 must be a multiple of 8");
+  rope_neox_launch(positions.data_ptr<long>(), q.data_ptr(), k.data_ptr(),
+                   cos_sin.data_ptr<float>(), T, Hq, Hk, D, R, cur_stream(q));
+  HIP_CHECK_LAST();
+}
+
+void silu_and_mul(at::Tensor out, at::Tensor x) {
+  check_bf16(out, "out"); check_bf16(x, "x");
+  const int I = out.size(-1);
+  const long T = out.numel() / I;
+  TORCH_CHECK(x.size(-1) == 2 * I && I % 8 == 0);
+  silu_and_mul_launch(out.data_ptr(), x.data_ptr(), T, I, cur_stream(x));
+  HIP_CHECK_LAST();
+}
+
+void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor slots) {
+  check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  TORCH_CHECK(slots.scalar_type() == at::kLong && slots.is_contiguous());
+  const int T = k.size(0), Hkv = k.size(1), D = k.size(2);
+  const int BS = k_cache.size(2);
+  TORCH_CHECK(k_cache.size(1) == Hkv && k_cache.size(3) == D && D % 8 == 0);
+  reshape_and_cache_launch(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+                           v_cache.data_ptr(), slots.data_ptr<long>(), T, Hkv,
+                           D, BS, cur_stream(k));
+  HIP_CHECK_LAST();
+}
+
+void greedy_sample(at::Tensor out, at::Tensor logits) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(out.scalar_type() == at::kLong && out.is_contiguous());
+  const int N = logits.size(0), V = logits.size(1);
+  greedy_sample_launch(out.data_ptr<long>(), logits.data_ptr(), N, V,
+                       cur_stream(logits));
+  HIP_CHECK_LAST();
+}
+
+void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor block_tables,
+                       at::Tensor seq_lens, double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q");
+  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt && block_tables.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int N = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hkv = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  TORCH_CHECK(BS == 16, "decode kernel assumes block_size 16");
+  const int max_blocks = block_tables.size(1);
+  int err = 0;
+  paged_attn_decode_launch(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                           v_cache.data_ptr(), block_tables.data_ptr<int>(),
+                           seq_lens.data_ptr<int>(), N, Hq, Hkv, D, max_blocks,
+                           (float)scale, &err, cur_stream(q));
+  TORCH_CHECK(!err, "paged_attn_decode: unsupported head_dim/GQ combination: D=",
+              D, " Hq=", Hq, " Hkv=", Hkv);
+  HIP_CHECK_LAST();
+}
+
+void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
+                   at::Tensor tile_start, at::Tensor tile_q0,
+                   at::Tensor tile_len, double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  TORCH_CHECK(tile_start.scalar_type() == at::kInt && tile_start.is_contiguous());
+  const int ntiles = tile_start.size(0);
+  const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
+  int err = 0;
+  flash_prefill_launch(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       tile_start.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                       tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D,
+                       (float)scale, &err, cur_stream(q));
+  TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D);
+  HIP_CHECK_LAST();
+}
+
+at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
+  check_bf16(a, "a"); check_bf16(b, "b");
+  auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  mfma_probe_launch(d.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                    cur_stream(a));
+  HIP_CHECK_LAST();
+  return d;
+}
+
+}  // namespace
+
+#undef HIP_CHECK_LAST
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm, "residual+=x; x=norm(residual)*w");
+  m.def("rotary_embedding", &rotary_embedding, "in-place neox RoPE on q,k");
+  m.def("silu_and_mul", &silu_and_mul, "silu(x[:d])*x[d:]");
+  m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged pool");
+  m.def("greedy_sample", &greedy_sample, "argmax over vocab");
+  m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention");
+  m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+}

@@ -1,0 +1,161 @@
+"""Plain-PyTorch fp32 reference implementations of every HIP op.
+
+These are the numerics oracle for the GPU kernel tests (tests compare the
+HIP kernels against these at fp32), and the CPU execution path that keeps
+the whole serving stack testable without a GPU (the reference
+tests everything host-only; we do the same for the control plane + engine
+logic and add real kernel tests on-device).
+"""
+from __future__ import annotations
+
+import torch
+
+
+def rms_norm(out: torch.Tensor, x: torch.Tensor, weight: torch.Tensor, eps: float) -> None:
+    xf = x.float()
+    inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    out.copy_((xf * inv * weight.float()).to(out.dtype))
+
+
+def fused_add_rms_norm(x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float) -> None:
+    summed = (x.float() + residual.float()).to(residual.dtype)
+    residual.copy_(summed)
+    sf = summed.float()
+    inv = torch.rsqrt(sf.pow(2).mean(-1, keepdim=True) + eps)
+    x.copy_((sf * inv * weight.float()).to(x.dtype))
+
+
+def build_cos_sin_cache(
+    head_dim: int,
+    rot_dim: int,
+    max_pos: int,
+    base: float = 10000.0,
+    scaling: dict | None = None,
+    device: torch.device | str = "cpu",
+) -> torch.Tensor:
+    """[max_pos, rot_dim] f32: first rot_dim/2 cos then rot_dim/2 sin.
+
+    `scaling` supports llama3-style rope scaling (keys: factor,
+    low_freq_factor, high_freq_factor, original_max_position_embeddings).
+    """
+    half = rot_dim // 2
+    inv_freq = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float64) * 2 / rot_dim))
+    if scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
+        factor = scaling["factor"]
+        lo = scaling["low_freq_factor"]
+        hi = scaling["high_freq_factor"]
+        orig = scaling["original_max_position_embeddings"]
+        wavelen = 2 * torch.pi / inv_freq
+        lo_wl = orig / lo
+        hi_wl = orig / hi
+        new = torch.where(wavelen > lo_wl, inv_freq / factor, inv_freq)
+        smooth = (orig / wavelen - lo) / (hi - lo)
+        mid = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+        is_mid = (wavelen <= lo_wl) & (wavelen >= hi_wl)
+        inv_freq = torch.where(is_mid, mid, new)
+    t = torch.arange(max_pos, dtype=torch.float64)
+    freqs = torch.outer(t, inv_freq)
+    cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1).float()
+    return cache.to(device)
+
+
+def rotary_embedding(
+    positions: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos_sin: torch.Tensor,
+    head_dim: int,
+    rot_dim: int,
+) -> None:
+    """In-place neox-style rotation. q: [T, Hq*D] or [T, Hq, D]; same for k."""
+    half = rot_dim // 2
+    cs = cos_sin[positions]  # [T, rot]
+    cos = cs[:, :half].unsqueeze(1)  # [T, 1, half]
+    sin = cs[:, half:].unsqueeze(1)
+    for x in (q, k):
+        T = x.shape[0]
+        xs = x.view(T, -1, head_dim)
+        x1 = xs[..., :half].float()
+        x2 = xs[..., half : 2 * half].float()
+        o1 = x1 * cos - x2 * sin
+        o2 = x2 * cos + x1 * sin
+        xs[..., :half] = o1.to(x.dtype)
+        xs[..., half : 2 * half] = o2.to(x.dtype)
+
+
+def silu_and_mul(out: torch.Tensor, x: torch.Tensor) -> None:
+    d = out.shape[-1]
+    xf = x.float()
+    out.copy_((torch.nn.functional.silu(xf[..., :d]) * xf[..., d:]).to(out.dtype))
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slots: torch.Tensor,
+) -> None:
+    """k/v: [T, Hkv, D]; caches: [B, Hkv, BS, D]; slots: [T] int64 (-1 skip)."""
+    bs = k_cache.shape[2]
+    mask = slots >= 0
+    idx = slots[mask]
+    blk = torch.div(idx, bs, rounding_mode="floor")
+    off = idx % bs
+    k_cache[blk, :, off] = k[mask]
+    v_cache[blk, :, off] = v[mask]
+
+
+def greedy_sample(out: torch.Tensor, logits: torch.Tensor) -> None:
+    out.copy_(logits.float().argmax(dim=-1))
+
+
+def paged_attn_decode(
+    out: torch.Tensor,
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+) -> None:
+    """q/out: [N, Hq, D]; caches [B, Hkv, BS, D]."""
+    N, Hq, D = q.shape
+    Hkv = k_cache.shape[1]
+    BS = k_cache.shape[2]
+    GQ = Hq // Hkv
+    for i in range(N):
+        L = int(seq_lens[i])
+        nblk = (L + BS - 1) // BS
+        blocks = block_tables[i, :nblk].long()
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        keys = keys.repeat_interleave(GQ, dim=0).float()  # [Hq, L, D]
+        vals = vals.repeat_interleave(GQ, dim=0).float()
+        qi = q[i].float().unsqueeze(1)  # [Hq, 1, D]
+        att = torch.softmax((qi @ keys.transpose(1, 2)) * scale, dim=-1)
+        out[i] = (att @ vals).squeeze(1).to(out.dtype)
+
+
+def varlen_prefill_attn(
+    out: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    seq_lens: list[int],
+    scale: float,
+) -> None:
+    """q: [T, Hq, D]; k/v: [T, Hkv, D]; causal within each sequence."""
+    Hq = q.shape[1]
+    Hkv = k.shape[1]
+    GQ = Hq // Hkv
+    start = 0
+    for L in seq_lens:
+        qs = q[start : start + L].float().permute(1, 0, 2)  # [Hq, L, D]
+        ks = k[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
+        vs = v[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
+        att = (qs @ ks.transpose(1, 2)) * scale
+        mask = torch.full((L, L), float("-inf"), device=q.device).triu(1)
+        att = torch.softmax(att + mask, dim=-1)
+        out[start : start + L] = (att @ vs).permute(1, 0, 2).to(out.dtype)
+        start += L

@@ -1,0 +1,149 @@
+"""On-device numerics: HIP/CDNA4 kernels vs the plain-torch fp32 reference.
+
+Every op the engine uses on the hot path is checked here against
+gpustack_amd.ops.torch_ref with random (asymmetric) inputs — including the
+MFMA fragment-layout probe (guide G9: transpose-detecting checks).
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gpustack_amd import ops
+from gpustack_amd.ops import torch_ref as R
+
+
+def _close(a, b, atol=2e-2, rtol=2e-2, frac=1.0):
+    a, b = a.float().cpu(), b.float().cpu()
+    ok = torch.isclose(a, b, atol=atol, rtol=rtol)
+    if frac >= 1.0:
+        assert bool(ok.all()), f"max abs err {(a-b).abs().max().item()}"
+    else:
+        assert ok.float().mean().item() >= frac
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+def test_hip_ext_loads():
+    assert ops.hip_available(), "native _hip_ops.so must load on a GPU box"
+
+
+def test_mfma_probe_layout():
+    # Random asymmetric A, B: catches any transposed fragment mapping.
+    a = torch.randn(16, 32, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(32, 16, dtype=torch.bfloat16, device="cuda")
+    d = ops.mfma_probe(a, b)
+    e = a.float() @ b.float()
+    _close(d, e, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("T,H", [(1, 4096), (17, 4096), (256, 8192), (3, 16384)])
+def test_rms_norm(T, H):
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.rand(H, dtype=torch.bfloat16, device="cuda") + 0.5
+    out = torch.empty_like(x)
+    ops.rms_norm(out, x, w, 1e-5)
+    ref = torch.empty_like(x)
+    R.rms_norm(ref, x, w, 1e-5)
+    _close(out, ref)
+
+
+@pytest.mark.parametrize("T,H", [(5, 4096), (128, 8192)])
+def test_fused_add_rms_norm(T, H):
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(T, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.rand(H, dtype=torch.bfloat16, device="cuda") + 0.5
+    x2, res2 = x.clone(), res.clone()
+    ops.fused_add_rms_norm(x, res, w, 1e-5)
+    R.fused_add_rms_norm(x2, res2, w, 1e-5)
+    _close(res, res2)
+    _close(x, x2)
+
+
+def test_rotary_embedding():
+    T, Hq, Hk, D = 33, 32, 8, 128
+    pos = torch.randint(0, 4096, (T,), device="cuda")
+    cache = ops.build_cos_sin_cache(D, D, 8192, device="cuda")
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
+    q2, k2 = q.clone(), k.clone()
+    ops.rotary_embedding(pos, q, k, cache, D, D)
+    R.rotary_embedding(pos, q2, k2, cache, D, D)
+    _close(q, q2)
+    _close(k, k2)
+
+
+def test_silu_and_mul():
+    x = torch.randn(77, 2 * 14336, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty(77, 14336, dtype=torch.bfloat16, device="cuda")
+    ref = torch.empty_like(out)
+    ops.silu_and_mul(out, x)
+    R.silu_and_mul(ref, x)
+    _close(out, ref)
+
+
+def test_reshape_and_cache():
+    T, Hkv, D, BS, B = 65, 8, 128, 16, 32
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(B, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    slots = torch.randperm(B * BS, device="cuda")[:T]
+    slots[5] = -1
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    R.reshape_and_cache(k, v, kc2, vc2, slots)
+    assert torch.equal(kc, kc2)
+    assert torch.equal(vc, vc2)
+
+
+@pytest.mark.parametrize("Hq,Hkv", [(32, 8), (8, 8), (64, 8), (40, 8)])
+@pytest.mark.parametrize("lens", [[1], [16], [17, 5, 160, 33], [2048]])
+def test_paged_attn_decode(Hq, Hkv, lens):
+    D, BS = 128, 16
+    N = len(lens)
+    maxb = (max(lens) + BS - 1) // BS
+    nblocks = sum((l + BS - 1) // BS for l in lens) + 2
+    kc = torch.randn(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    bt = torch.zeros(N, maxb, dtype=torch.int32, device="cuda")
+    nxt = 0
+    for i, l in enumerate(lens):
+        nb = (l + BS - 1) // BS
+        bt[i, :nb] = torch.arange(nxt, nxt + nb, dtype=torch.int32)
+        nxt += nb
+    q = torch.randn(N, Hq, D, dtype=torch.bfloat16, device="cuda")
+    sl = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    ref = torch.empty_like(q)
+    scale = 1 / math.sqrt(D)
+    ops.paged_attn_decode(out, q, kc, vc, bt, sl, scale)
+    R.paged_attn_decode(ref, q, kc, vc, bt, sl, scale)
+    _close(out, ref, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("Hq,Hkv", [(32, 8), (8, 8)])
+@pytest.mark.parametrize("lens", [[1], [64], [63, 70, 5], [300], [1024]])
+def test_flash_prefill(Hq, Hkv, lens):
+    D = 128
+    T = sum(lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    ref = torch.empty_like(q)
+    scale = 1 / math.sqrt(D)
+    ops.varlen_prefill_attn(out, q, k, v, lens, scale)
+    R.varlen_prefill_attn(ref, q, k, v, lens, scale)
+    _close(out, ref, atol=4e-2, rtol=4e-2)
+
+
+def test_greedy_sample():
+    logits = torch.randn(64, 128256, dtype=torch.bfloat16, device="cuda")
+    got = ops.greedy_sample(logits)
+    assert torch.equal(got.cpu(), logits.float().argmax(-1).cpu())
