@@ -1,0 +1,145 @@
+"""Engine + model configuration for the first-party CDNA4 inference engine.
+
+ModelSpec mirrors the fields the reference reads out of HF config.json via
+AutoConfig (reference: gpustack/scheduler/scheduler.py:216 evaluation path);
+we parse config.json directly so random-init serving needs no checkpoint.
+"""
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field
+from pathlib import Path
+
+
+@dataclass
+class ModelSpec:
+    architecture: str = "LlamaForCausalLM"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    rope_scaling: dict | None = None
+    rms_norm_eps: float = 1e-5
+    max_position_embeddings: int = 8192
+    tie_word_embeddings: bool = False
+    attention_bias: bool = False  # qwen2-style qkv bias
+    qk_norm: bool = False         # qwen3-style per-head q/k RMSNorm
+    eos_token_id: int = 128001
+
+    @property
+    def gqa_ratio(self) -> int:
+        return self.num_heads // self.num_kv_heads
+
+    def kv_bytes_per_token(self, dtype_size: int = 2) -> int:
+        return 2 * self.num_layers * self.num_kv_heads * self.head_dim * dtype_size
+
+    def weight_bytes(self, dtype_size: int = 2) -> int:
+        h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
+        qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
+        o = self.num_heads * self.head_dim * h
+        mlp = 3 * h * i
+        per_layer = qkv + o + mlp + 2 * h
+        emb = v * h * (1 if self.tie_word_embeddings else 2)
+        return (per_layer * self.num_layers + emb + h) * dtype_size
+
+    @classmethod
+    def from_hf_config(cls, cfg: dict) -> "ModelSpec":
+        arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        nh = cfg.get("num_attention_heads", 32)
+        hd = cfg.get("head_dim") or cfg.get("hidden_size", 4096) // nh
+        eos = cfg.get("eos_token_id", 2)
+        if isinstance(eos, list):
+            eos = eos[0]
+        return cls(
+            architecture=arch,
+            vocab_size=cfg.get("vocab_size", 32000),
+            hidden_size=cfg.get("hidden_size", 4096),
+            intermediate_size=cfg.get("intermediate_size", 11008),
+            num_layers=cfg.get("num_hidden_layers", 32),
+            num_heads=nh,
+            num_kv_heads=cfg.get("num_key_value_heads", nh),
+            head_dim=hd,
+            rope_theta=cfg.get("rope_theta", 10000.0),
+            rope_scaling=cfg.get("rope_scaling"),
+            rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
+            max_position_embeddings=cfg.get("max_position_embeddings", 4096),
+            tie_word_embeddings=cfg.get("tie_word_embeddings", False),
+            attention_bias=arch.startswith("Qwen2"),
+            qk_norm=arch.startswith("Qwen3"),
+            eos_token_id=eos,
+        )
+
+    @classmethod
+    def from_dir(cls, model_dir: str | Path) -> "ModelSpec":
+        with open(Path(model_dir) / "config.json") as f:
+            return cls.from_hf_config(json.load(f))
+
+
+# Named presets so the bench / tests / control plane can run with
+# random-init weights and no network (BASELINE.json: synthetic data).
+PRESETS: dict[str, ModelSpec] = {
+    "llama-3-8b": ModelSpec(),
+    "llama-3-70b": ModelSpec(
+        hidden_size=8192, intermediate_size=28672, num_layers=80,
+        num_heads=64, num_kv_heads=8,
+    ),
+    "qwen3-32b": ModelSpec(
+        architecture="Qwen3ForCausalLM", vocab_size=151936, hidden_size=5120,
+        intermediate_size=25600, num_layers=64, num_heads=64, num_kv_heads=8,
+        rope_theta=1000000.0, max_position_embeddings=40960, qk_norm=True,
+        eos_token_id=151645,
+    ),
+    "qwen3-14b": ModelSpec(
+        architecture="Qwen3ForCausalLM", vocab_size=151936, hidden_size=5120,
+        intermediate_size=17408, num_layers=40, num_heads=40, num_kv_heads=8,
+        rope_theta=1000000.0, max_position_embeddings=40960, qk_norm=True,
+        eos_token_id=151645,
+    ),
+    "qwen2.5-7b": ModelSpec(
+        architecture="Qwen2ForCausalLM", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_layers=28, num_heads=28, num_kv_heads=4,
+        rope_theta=1000000.0, attention_bias=True, eos_token_id=151645,
+    ),
+    # tiny CPU-testable model (OPT-125m-scale plumbing per BASELINE.json cfg 1)
+    "tiny": ModelSpec(
+        vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,
+        num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
+        rope_theta=10000.0, eos_token_id=1,
+    ),
+}
+
+
+@dataclass
+class EngineConfig:
+    model: str = "llama-3-8b"              # preset name or model dir
+    dtype: str = "bfloat16"
+    block_size: int = 16
+    gpu_memory_utilization: float = 0.90
+    max_num_seqs: int = 256
+    max_prefill_tokens: int = 8192         # per-step prefill token budget
+    max_model_len: int = 8192
+    kv_cache_blocks: int | None = None     # override (CPU tests)
+    tp_size: int = 1
+    tp_rank: int = 0
+    device: str = "cuda"
+    seed: int = 0
+    enforce_random_weights: bool = True    # no checkpoint: random init
+    model_dir: str | None = None
+    # host-DRAM KV offload tier (extended_kv_cache in the reference schema)
+    kv_offload_gb: float = 0.0
+
+    spec: ModelSpec = field(default_factory=ModelSpec)
+
+    def __post_init__(self):
+        if self.model in PRESETS:
+            self.spec = PRESETS[self.model]
+        elif self.model_dir:
+            self.spec = ModelSpec.from_dir(self.model_dir)
+        elif Path(self.model).is_dir():
+            self.model_dir = self.model
+            self.spec = ModelSpec.from_dir(self.model)
+        self.max_model_len = min(self.max_model_len, self.spec.max_position_embeddings)

@@ -1,0 +1,105 @@
+"""LLMEngine: continuous-batching serving engine for MI355X.
+
+The first-party replacement for the external engines the reference launches
+in containers. One engine instance owns one GPU (or one
+TP rank group); the worker's serve manager runs it in a subprocess behind
+an OpenAI-compatible HTTP endpoint (gpustack_amd/worker/engine_server.py).
+"""
+from __future__ import annotations
+
+import itertools
+import logging
+
+from ..parallel import Communicator
+from .config import EngineConfig
+from .model_runner import ModelRunner
+from .scheduler import Scheduler
+from .sequence import SamplingParams, Sequence, SeqStatus, StepOutput
+
+logger = logging.getLogger(__name__)
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, comm: Communicator | None = None):
+        self.cfg = cfg
+        self.runner = ModelRunner(cfg, comm)
+        kv = self.runner.init_kv_cache()
+        self.scheduler = Scheduler(cfg, kv)
+        self.seqs: dict[str, Sequence] = {}
+        self._counter = itertools.count()
+        logger.info(
+            "engine ready: model=%s kv_blocks=%d (%.1f GiB KV pool)",
+            cfg.model, kv.num_blocks,
+            kv.num_blocks * cfg.block_size * cfg.spec.kv_bytes_per_token() / 2**30,
+        )
+
+    # -- API ---------------------------------------------------------------
+    def add_request(
+        self,
+        prompt_token_ids: list[int],
+        params: SamplingParams | None = None,
+        request_id: str | None = None,
+    ) -> str:
+        rid = request_id or f"req-{next(self._counter)}"
+        params = params or SamplingParams()
+        if len(prompt_token_ids) > self.cfg.max_model_len - 1:
+            prompt_token_ids = prompt_token_ids[-(self.cfg.max_model_len - 1):]
+        seq = Sequence(rid, list(prompt_token_ids), params)
+        self.seqs[rid] = seq
+        self.scheduler.add(seq)
+        return rid
+
+    def abort_request(self, request_id: str) -> bool:
+        ok = self.scheduler.abort(request_id)
+        self.seqs.pop(request_id, None)
+        return ok
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_work()
+
+    @property
+    def num_running(self) -> int:
+        return len(self.scheduler.running)
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.scheduler.waiting)
+
+    def step(self) -> list[StepOutput]:
+        batch = self.scheduler.schedule()
+        if batch is None:
+            return []
+        token_ids = self.runner.execute(batch)
+        if batch.is_prefill:
+            self.scheduler.on_prefill_done(batch)
+        outputs: list[StepOutput] = []
+        for seq, tok in zip(batch.seqs, token_ids):
+            seq.record_first_token()
+            seq.output_token_ids.append(tok)
+            reason = self._finish_reason(seq, tok)
+            if reason:
+                self.scheduler.finish_seq(seq, reason)
+                self.seqs.pop(seq.request_id, None)
+            outputs.append(StepOutput(seq.request_id, tok, reason is not None, reason))
+        return outputs
+
+    def _finish_reason(self, seq: Sequence, tok: int) -> str | None:
+        p = seq.params
+        if not p.ignore_eos:
+            if tok == self.cfg.spec.eos_token_id or tok in p.stop_token_ids:
+                return "stop"
+        if len(seq.output_token_ids) >= p.max_tokens:
+            return "length"
+        if seq.num_tokens >= self.cfg.max_model_len:
+            return "length"
+        return None
+
+    # convenience for tests / offline use
+    def generate(self, prompts: list[list[int]], params: SamplingParams | None = None):
+        ids = [self.add_request(p, params) for p in prompts]
+        results = {i: [] for i in ids}
+        while self.has_unfinished():
+            for out in self.step():
+                if out.request_id in results:
+                    results[out.request_id].append(out.token_id)
+        return [results[i] for i in ids]

@@ -1,0 +1,135 @@
+"""Model runner: turns a ScheduledBatch into device tensors, runs the
+forward pass on the native ops, and samples next tokens."""
+from __future__ import annotations
+
+import torch
+
+from .. import ops
+from ..models.llama import ForwardMeta, LlamaForCausalLM
+from ..models.weights import load_safetensors, random_init
+from ..parallel import Communicator
+from .config import EngineConfig
+from .kv_cache import KVCache
+from .scheduler import ScheduledBatch
+from .sequence import Sequence
+
+
+class Sampler:
+    def __init__(self, device):
+        self.device = device
+        self.generator = None
+
+    def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:
+        if all(s.params.greedy for s in seqs):
+            return ops.greedy_sample(logits).tolist()
+        out: list[int] = [0] * len(seqs)
+        greedy_idx = [i for i, s in enumerate(seqs) if s.params.greedy]
+        rand_idx = [i for i, s in enumerate(seqs) if not s.params.greedy]
+        if greedy_idx:
+            ids = ops.greedy_sample(logits[greedy_idx])
+            for j, i in enumerate(greedy_idx):
+                out[i] = int(ids[j])
+        if rand_idx:
+            lg = logits[rand_idx].float()
+            temps = torch.tensor(
+                [seqs[i].params.temperature for i in rand_idx],
+                device=logits.device,
+            ).unsqueeze(1)
+            lg = lg / temps
+            probs = torch.softmax(lg, dim=-1)
+            for j, i in enumerate(rand_idx):
+                p = seqs[i].params
+                row = probs[j]
+                if p.top_k > 0 and p.top_k < row.shape[-1]:
+                    vals, idx = torch.topk(row, p.top_k)
+                    row = torch.zeros_like(row).scatter_(0, idx, vals)
+                if p.top_p < 1.0:
+                    sorted_p, sorted_i = torch.sort(row, descending=True)
+                    cum = torch.cumsum(sorted_p, dim=-1)
+                    keep = cum - sorted_p <= p.top_p
+                    keep[0] = True
+                    row = torch.zeros_like(row).scatter_(0, sorted_i[keep], sorted_p[keep])
+                row = row / row.sum()
+                if self.generator is None:
+                    self.generator = torch.Generator(device=logits.device)
+                if p.seed is not None:
+                    self.generator.manual_seed(p.seed + len(seqs[i].output_token_ids))
+                out[i] = int(torch.multinomial(row, 1, generator=self.generator))
+        return out
+
+
+class ModelRunner:
+    def __init__(self, cfg: EngineConfig, comm: Communicator | None = None):
+        self.cfg = cfg
+        self.comm = comm or Communicator()
+        self.device = torch.device(cfg.device)
+        self.model = LlamaForCausalLM(cfg, self.comm, self.device)
+        if cfg.model_dir and not cfg.enforce_random_weights:
+            load_safetensors(self.model, cfg, cfg.model_dir)
+        else:
+            random_init(self.model, cfg)
+        self.sampler = Sampler(self.device)
+        self.kv: KVCache | None = None
+
+    def init_kv_cache(self) -> KVCache:
+        cfg = self.cfg
+        if cfg.kv_cache_blocks is not None:
+            nblocks = cfg.kv_cache_blocks
+        elif self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            nblocks = KVCache.compute_num_blocks(cfg, free)
+        else:
+            nblocks = 512
+        max_needed = cfg.max_num_seqs * (
+            (cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
+        )
+        nblocks = min(nblocks, max_needed)
+        self.kv = KVCache(cfg, nblocks, self.device)
+        return self.kv
+
+    def _meta(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
+        dev = self.device
+        tokens = torch.tensor(batch.token_ids, dtype=torch.long, device=dev)
+        positions = torch.tensor(batch.positions, dtype=torch.long, device=dev)
+        slots = torch.tensor(batch.slot_mapping, dtype=torch.long, device=dev)
+        if batch.is_prefill:
+            # last token of each sequence produces the next-token logits
+            idx, off = [], 0
+            for L in batch.seq_lens:
+                idx.append(off + L - 1)
+                off += L
+            tiles = ops.build_prefill_tiles(batch.seq_lens, dev)
+            meta = ForwardMeta(
+                is_prefill=True,
+                positions=positions,
+                slot_mapping=slots,
+                logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
+                seq_lens_list=batch.seq_lens,
+                tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
+            )
+        else:
+            maxb = max(len(s.block_table) for s in batch.seqs)
+            bt = torch.zeros(len(batch.seqs), maxb, dtype=torch.int32)
+            for i, s in enumerate(batch.seqs):
+                bt[i, : len(s.block_table)] = torch.tensor(s.block_table, dtype=torch.int32)
+            meta = ForwardMeta(
+                is_prefill=False,
+                positions=positions,
+                slot_mapping=slots,
+                logits_indices=torch.arange(len(batch.seqs), dtype=torch.long, device=dev),
+                block_tables=bt.to(dev),
+                seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
+            )
+        return tokens, meta
+
+    @torch.inference_mode()
+    def execute(self, batch: ScheduledBatch) -> list[int]:
+        tokens, meta = self._meta(batch)
+        logits = self.model(tokens, meta, self.kv)
+        token_ids = self.sampler.sample(logits, batch.seqs)
+        if self.comm.tp_size > 1:
+            # ranks must agree on sampled tokens; rank 0 decides
+            t = torch.tensor(token_ids, dtype=torch.long, device=self.device)
+            self.comm.broadcast(t, src=0)
+            token_ids = t.tolist()
+        return token_ids

@@ -1,0 +1,78 @@
+"""Request / sequence state for the continuous-batching engine."""
+from __future__ import annotations
+
+import enum
+import time
+from dataclasses import dataclass, field
+
+
+class SeqStatus(enum.Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    FINISHED = "finished"
+
+
+@dataclass
+class SamplingParams:
+    temperature: float = 0.0        # 0 => greedy
+    top_p: float = 1.0
+    top_k: int = 0                  # 0 => disabled
+    max_tokens: int = 128
+    ignore_eos: bool = False
+    stop_token_ids: tuple[int, ...] = ()
+    seed: int | None = None
+
+    @property
+    def greedy(self) -> bool:
+        return self.temperature <= 0.0
+
+
+@dataclass
+class Sequence:
+    request_id: str
+    prompt_token_ids: list[int]
+    params: SamplingParams = field(default_factory=SamplingParams)
+    output_token_ids: list[int] = field(default_factory=list)
+    status: SeqStatus = SeqStatus.WAITING
+    block_table: list[int] = field(default_factory=list)
+    num_cached_tokens: int = 0      # tokens whose KV is already in the pool
+    arrival_time: float = field(default_factory=time.monotonic)
+    first_token_time: float | None = None
+    finish_time: float | None = None
+    finish_reason: str | None = None
+    preemptions: int = 0
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.prompt_token_ids) + len(self.output_token_ids)
+
+    @property
+    def all_token_ids(self) -> list[int]:
+        return self.prompt_token_ids + self.output_token_ids
+
+    @property
+    def num_prompt_tokens(self) -> int:
+        return len(self.prompt_token_ids)
+
+    def record_first_token(self) -> None:
+        if self.first_token_time is None:
+            self.first_token_time = time.monotonic()
+
+    def finish(self, reason: str) -> None:
+        self.status = SeqStatus.FINISHED
+        self.finish_reason = reason
+        self.finish_time = time.monotonic()
+
+    @property
+    def ttft(self) -> float | None:
+        if self.first_token_time is None:
+            return None
+        return self.first_token_time - self.arrival_time
+
+
+@dataclass
+class StepOutput:
+    request_id: str
+    token_id: int
+    finished: bool
+    finish_reason: str | None = None

@@ -1,0 +1,170 @@
+"""Llama-family decoder (Llama 2/3, Qwen2.x, Qwen3 dense) on the native ops.
+
+Weights live as fused bf16 parameters shaped for serving GEMMs
+(qkv / gate_up fused, hipBLASLt via F.linear), attention + norms + rope +
+activation run on the hand-written CDNA4 kernels (gpustack_amd.ops), and
+tensor parallelism shards heads/intermediate across the node's xGMI-linked
+GPUs with RCCL all-reduce after o_proj and down_proj.
+
+Replaces the engine-side model code the reference delegates to vLLM
+(SURVEY.md §2.9 #1). Covers ModelSpec.attention_bias (Qwen2) and
+ModelSpec.qk_norm (Qwen3).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from .. import ops
+from ..engine.config import EngineConfig, ModelSpec
+from ..parallel import Communicator
+
+
+@dataclass
+class ForwardMeta:
+    """Per-step tensors describing the batch (built by the model runner)."""
+
+    is_prefill: bool
+    positions: torch.Tensor       # [T] int64
+    slot_mapping: torch.Tensor    # [T] int64
+    logits_indices: torch.Tensor  # [N] int64 rows of x to project to logits
+    # prefill
+    seq_lens_list: list[int] | None = None
+    tile_start: torch.Tensor | None = None
+    tile_q0: torch.Tensor | None = None
+    tile_len: torch.Tensor | None = None
+    # decode
+    block_tables: torch.Tensor | None = None  # [N, maxb] int32
+    seq_lens: torch.Tensor | None = None      # [N] int32
+
+
+class Attention(nn.Module):
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+        super().__init__()
+        self.spec = spec
+        self.comm = comm
+        self.hq = spec.num_heads // tp_size
+        self.hkv = max(1, spec.num_kv_heads // tp_size)
+        self.d = spec.head_dim
+        self.scale = self.d ** -0.5
+        h = spec.hidden_size
+        qkv_out = (self.hq + 2 * self.hkv) * self.d
+        self.qkv_w = nn.Parameter(torch.empty(qkv_out, h, dtype=dtype), requires_grad=False)
+        self.qkv_b = (
+            nn.Parameter(torch.empty(qkv_out, dtype=dtype), requires_grad=False)
+            if spec.attention_bias else None
+        )
+        self.o_w = nn.Parameter(torch.empty(h, self.hq * self.d, dtype=dtype), requires_grad=False)
+        if spec.qk_norm:
+            self.q_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
+            self.k_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
+
+    def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
+        T = x.shape[0]
+        qkv = F.linear(x, self.qkv_w, self.qkv_b)
+        q, k, v = qkv.split(
+            [self.hq * self.d, self.hkv * self.d, self.hkv * self.d], dim=-1
+        )
+        q = q.contiguous().view(T, self.hq, self.d)
+        k = k.contiguous().view(T, self.hkv, self.d)
+        v = v.contiguous().view(T, self.hkv, self.d)
+        if self.spec.qk_norm:
+            ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
+            ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
+        ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d, self.d)
+        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        out = torch.empty_like(q)
+        if meta.is_prefill:
+            ops.varlen_prefill_attn(
+                out, q, k, v, meta.seq_lens_list, self.scale,
+                tiles=(meta.tile_start, meta.tile_q0, meta.tile_len),
+            )
+        else:
+            ops.paged_attn_decode(
+                out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale
+            )
+        o = F.linear(out.view(T, -1), self.o_w)
+        return self.comm.all_reduce(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+        super().__init__()
+        self.comm = comm
+        h = spec.hidden_size
+        self.i = spec.intermediate_size // tp_size
+        self.gate_up_w = nn.Parameter(torch.empty(2 * self.i, h, dtype=dtype), requires_grad=False)
+        self.down_w = nn.Parameter(torch.empty(h, self.i, dtype=dtype), requires_grad=False)
+
+    def forward(self, x):
+        gu = F.linear(x, self.gate_up_w)
+        act = torch.empty(x.shape[0], self.i, dtype=x.dtype, device=x.device)
+        ops.silu_and_mul(act, gu)
+        return self.comm.all_reduce(F.linear(act, self.down_w))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+        super().__init__()
+        self.spec = spec
+        self.attn = Attention(spec, tp_size, comm, dtype)
+        self.mlp = MLP(spec, tp_size, comm, dtype)
+        h = spec.hidden_size
+        self.input_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
+        self.post_attn_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
+
+    def forward(self, x, residual, meta, cos_sin, k_cache, v_cache):
+        eps = self.spec.rms_norm_eps
+        if residual is None:
+            residual = x
+            h = torch.empty_like(x)
+            ops.rms_norm(h, x, self.input_norm, eps)
+        else:
+            h = x
+            ops.fused_add_rms_norm(h, residual, self.input_norm, eps)
+        a = self.attn(h, meta, cos_sin, k_cache, v_cache)
+        ops.fused_add_rms_norm(a, residual, self.post_attn_norm, eps)
+        m = self.mlp(a)
+        return m, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: EngineConfig, comm: Communicator, device):
+        super().__init__()
+        spec = cfg.spec
+        self.spec = spec
+        self.cfg = cfg
+        dtype = getattr(torch, cfg.dtype)
+        self.dtype = dtype
+        self.embed = nn.Parameter(
+            torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.layers = nn.ModuleList(
+            [DecoderLayer(spec, cfg.tp_size, comm, dtype) for _ in range(spec.num_layers)]
+        )
+        self.final_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
+        if spec.tie_word_embeddings:
+            self.lm_head = self.embed
+        else:
+            self.lm_head = nn.Parameter(
+                torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype), requires_grad=False
+            )
+        cache = ops.build_cos_sin_cache(
+            spec.head_dim, spec.head_dim, cfg.max_model_len,
+            base=spec.rope_theta, scaling=spec.rope_scaling,
+        )
+        self.register_buffer("cos_sin", cache.to(device), persistent=False)
+        self.to(device)
+
+    @torch.inference_mode()
+    def forward(self, token_ids: torch.Tensor, meta: ForwardMeta, kv) -> torch.Tensor:
+        x = F.embedding(token_ids, self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
+        ops.fused_add_rms_norm(x, residual, self.final_norm, self.spec.rms_norm_eps)
+        hidden = x[meta.logits_indices]
+        return F.linear(hidden, self.lm_head)

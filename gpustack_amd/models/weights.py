@@ -1,0 +1,134 @@
+"""Weight init / loading for the native engine.
+
+Two paths (reference parity: the worker's model-file manager downloads HF
+safetensors — gpustack/worker/model_file_manager.py:59; here loading is
+first-party):
+
+  * random init — deterministic per (seed, tensor) and TP-consistent: full
+    tensors are generated then sliced to the local shard, so TP=N shards
+    always compose to the same TP=1 weights (bench uses this: no network).
+  * safetensors  — maps HF checkpoint names onto the fused serving layout.
+"""
+from __future__ import annotations
+
+import json
+from pathlib import Path
+
+import torch
+
+from ..engine.config import EngineConfig
+
+
+def _gen(shape, seed_key: str, base_seed: int, dtype, device, std=0.02):
+    g = torch.Generator(device="cpu")
+    g.manual_seed((base_seed * 1000003 + hash(seed_key)) % (2**63))
+    t = torch.randn(*shape, generator=g, dtype=torch.float32) * std
+    return t.to(dtype).to(device)
+
+
+def random_init(model, cfg: EngineConfig) -> None:
+    spec = model.spec
+    tp, rank = cfg.tp_size, cfg.tp_rank
+    dtype, device = model.dtype, model.embed.device
+    seed = cfg.seed
+    hq, hkv, d = spec.num_heads // tp, max(1, spec.num_kv_heads // tp), spec.head_dim
+    i_loc = spec.intermediate_size // tp
+
+    model.embed.copy_(_gen((spec.vocab_size, spec.hidden_size), "embed", seed, dtype, device))
+    if not spec.tie_word_embeddings:
+        model.lm_head.copy_(_gen((spec.vocab_size, spec.hidden_size), "lm_head", seed, dtype, device))
+    model.final_norm.fill_(1.0)
+
+    for li, layer in enumerate(model.layers):
+        q_full = _gen((spec.num_heads * d, spec.hidden_size), f"{li}.q", seed, dtype, device)
+        k_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.k", seed, dtype, device)
+        v_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.v", seed, dtype, device)
+        layer.attn.qkv_w.copy_(torch.cat([
+            q_full[rank * hq * d:(rank + 1) * hq * d],
+            k_full[rank * hkv * d:(rank + 1) * hkv * d],
+            v_full[rank * hkv * d:(rank + 1) * hkv * d],
+        ]))
+        if layer.attn.qkv_b is not None:
+            qb = _gen((spec.num_heads * d,), f"{li}.qb", seed, dtype, device)
+            kb = _gen((spec.num_kv_heads * d,), f"{li}.kb", seed, dtype, device)
+            vb = _gen((spec.num_kv_heads * d,), f"{li}.vb", seed, dtype, device)
+            layer.attn.qkv_b.copy_(torch.cat([
+                qb[rank * hq * d:(rank + 1) * hq * d],
+                kb[rank * hkv * d:(rank + 1) * hkv * d],
+                vb[rank * hkv * d:(rank + 1) * hkv * d],
+            ]))
+        o_full = _gen((spec.hidden_size, spec.num_heads * d), f"{li}.o", seed, dtype, device)
+        layer.attn.o_w.copy_(o_full[:, rank * hq * d:(rank + 1) * hq * d])
+        if spec.qk_norm:
+            layer.attn.q_norm.fill_(1.0)
+            layer.attn.k_norm.fill_(1.0)
+        gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
+        up = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.up", seed, dtype, device)
+        layer.mlp.gate_up_w.copy_(torch.cat([
+            gate[rank * i_loc:(rank + 1) * i_loc],
+            up[rank * i_loc:(rank + 1) * i_loc],
+        ]))
+        down = _gen((spec.hidden_size, spec.intermediate_size), f"{li}.down", seed, dtype, device)
+        layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
+        layer.input_norm.fill_(1.0)
+        layer.post_attn_norm.fill_(1.0)
+
+
+def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
+    """Load an HF Llama/Qwen checkpoint into the fused layout, sharded for TP."""
+    from safetensors import safe_open
+
+    spec = model.spec
+    tp, rank = cfg.tp_size, cfg.tp_rank
+    d = spec.head_dim
+    hq, hkv = spec.num_heads // tp, max(1, spec.num_kv_heads // tp)
+    i_loc = spec.intermediate_size // tp
+    model_dir = Path(model_dir)
+    files = sorted(model_dir.glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no safetensors under {model_dir}")
+
+    def row_shard(t, n):  # column-parallel: shard output rows
+        per = t.shape[0] // tp if n is None else n
+        return t[rank * per:(rank + 1) * per]
+
+    tensors: dict[str, torch.Tensor] = {}
+    for f in files:
+        with safe_open(str(f), framework="pt") as sf:
+            for name in sf.keys():
+                tensors[name] = sf.get_tensor(name)
+
+    def get(name):
+        t = tensors[name]
+        return t.to(model.dtype)
+
+    pre = "model."
+    model.embed.copy_(get(pre + "embed_tokens.weight"))
+    model.final_norm.copy_(get(pre + "norm.weight"))
+    if not spec.tie_word_embeddings:
+        model.lm_head.copy_(get("lm_head.weight"))
+    for li, layer in enumerate(model.layers):
+        p = f"{pre}layers.{li}."
+        q = row_shard(get(p + "self_attn.q_proj.weight"), hq * d)
+        k = row_shard(get(p + "self_attn.k_proj.weight"), hkv * d)
+        v = row_shard(get(p + "self_attn.v_proj.weight"), hkv * d)
+        layer.attn.qkv_w.copy_(torch.cat([q, k, v]))
+        if layer.attn.qkv_b is not None:
+            layer.attn.qkv_b.copy_(torch.cat([
+                row_shard(get(p + "self_attn.q_proj.bias"), hq * d),
+                row_shard(get(p + "self_attn.k_proj.bias"), hkv * d),
+                row_shard(get(p + "self_attn.v_proj.bias"), hkv * d),
+            ]))
+        o = get(p + "self_attn.o_proj.weight")
+        layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
+        if spec.qk_norm:
+            layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
+            layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
+        layer.mlp.gate_up_w.copy_(torch.cat([
+            row_shard(get(p + "mlp.gate_proj.weight"), i_loc),
+            row_shard(get(p + "mlp.up_proj.weight"), i_loc),
+        ]))
+        dn = get(p + "mlp.down_proj.weight")
+        layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
+        layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+        layer.post_attn_norm.copy_(get(p + "post_attention_layernorm.weight"))

@@ -132,11 +132,12 @@ def build_prefill_tiles(seq_lens: list[int], device) -> tuple[torch.Tensor, torc
     return mk(starts), mk(q0s), mk(lens)
 
 
-def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float) -> None:
+def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float, tiles=None) -> None:
     hip = _backend(q)
     if hip is not None:
-        ts, tq, tl = build_prefill_tiles(seq_lens, q.device)
-        hip.flash_prefill(out, q, k, v, ts, tq, tl, scale)
+        if tiles is None or tiles[0] is None:
+            tiles = build_prefill_tiles(seq_lens, q.device)
+        hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale)
     else:
         torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale)
 
