@@ -1,0 +1,175 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark (driver contract).
+
+Measures the BASELINE.json metric — output tokens/sec (+ p50 TTFT) for
+Llama-3-8B serving on MI355X — with one engine replica per GPU (the
+reference's headline deployment shape: N replicas bin-packed on N GPUs,
+BASELINE.json config 3). Synthetic random-token requests, random-init bf16
+weights (no network), closed-loop at fixed concurrency so the continuous
+batcher runs saturated; a "step" is one engine iteration (one varlen
+prefill batch or one decode batch over all running sequences).
+
+Usage:  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches this via torch.distributed.run with one rank
+per GPU; ranks are independent replicas (weak scaling) synchronized only
+at the timing barriers.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=192)
+    ap.add_argument("--warmup", type=int, default=48)
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--concurrency", type=int, default=128, help="target in-flight requests per GPU")
+    ap.add_argument("--isl", type=int, default=256, help="synthetic prompt length")
+    ap.add_argument("--osl", type=int, default=128, help="max output tokens per request")
+    ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--device", default=None)
+    return ap.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_cuda = torch.cuda.is_available()
+    device = args.device or (f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    model = args.model
+    cfg = EngineConfig(
+        model=model,
+        device=device,
+        max_model_len=args.max_model_len,
+        max_num_seqs=max(args.concurrency, 8),
+        seed=0,
+    )
+    if not use_cuda:  # CPU smoke path: shrink everything
+        cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=256, max_model_len=512)
+        args.concurrency = min(args.concurrency, 8)
+        args.isl, args.osl = 32, 16
+        model = "tiny"
+    eng = LLMEngine(cfg)
+
+    rng = random.Random(1234 + rank)
+    vocab = cfg.spec.vocab_size
+    params = SamplingParams(max_tokens=args.osl, ignore_eos=True)
+
+    def refill():
+        while eng.scheduler.num_unfinished < args.concurrency:
+            toks = [rng.randrange(2, vocab) for _ in range(args.isl)]
+            eng.add_request(toks, params)
+
+    def one_step() -> int:
+        refill()
+        outs = eng.step()
+        return len(outs)
+
+    def barrier_sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+            if use_cuda:
+                torch.cuda.synchronize()
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        one_step()
+    barrier_sync()
+
+    # ---- timed region: exactly K steps ----
+    first_token_seen: set[str] = {
+        rid for rid, s in eng.seqs.items() if s.first_token_time is not None
+    }
+    ttfts: list[float] = []
+    t0 = time.perf_counter()
+    tokens = 0
+    for _ in range(args.steps):
+        refill()
+        step_t = time.perf_counter()
+        outs = eng.step()
+        tokens += len(outs)
+        for o in outs:
+            if o.request_id not in first_token_seen:
+                first_token_seen.add(o.request_id)
+                seq = eng.seqs.get(o.request_id)
+                if seq is not None and seq.ttft is not None:
+                    ttfts.append(seq.ttft)
+                else:
+                    ttfts.append(time.perf_counter() - step_t)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # aggregate across ranks: total tokens, max elapsed
+    if dist:
+        dev = device if use_cuda else "cpu"
+        tt = torch.tensor([float(tokens)], device=dev)
+        dist.all_reduce(tt)
+        tot_tokens = int(tt.item())
+        te = torch.tensor([elapsed], device=dev)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+    else:
+        tot_tokens = tokens
+
+    ttfts.sort()
+    p50_ttft_ms = (ttfts[len(ttfts) // 2] * 1000) if ttfts else None
+
+    value = tot_tokens / elapsed if elapsed > 0 else 0.0
+    if rank == 0:
+        print(json.dumps({
+            "metric": "output tokens/sec (serving, continuous batching)",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world if world > 1 else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000 / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": cfg.dtype if use_cuda else "float32-cpu-smoke",
+            "data": "synthetic random-token prompts, random-init weights",
+            "config": {
+                "model": model,
+                "global_batch": args.concurrency * (world if world > 1 else 1),
+                "seq_len": args.isl + args.osl,
+                "parallelism": f"dp{world if world > 1 else args.gpus}",
+                "isl": args.isl,
+                "osl": args.osl,
+                "concurrency_per_gpu": args.concurrency,
+                "p50_ttft_ms": round(p50_ttft_ms, 2) if p50_ttft_ms else None,
+                "timed_output_tokens": tot_tokens,
+            },
+        }))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

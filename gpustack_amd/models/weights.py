@@ -20,8 +20,17 @@ from ..engine.config import EngineConfig
 
 
 def _gen(shape, seed_key: str, base_seed: int, dtype, device, std=0.02):
+    import zlib
+
+    seed = (base_seed * 1000003 + zlib.crc32(seed_key.encode())) % (2**63)
+    dev = torch.device(device)
+    if dev.type == "cuda":  # device-side randn: 8B init in seconds
+        g = torch.Generator(device=dev)
+        g.manual_seed(seed)
+        t = torch.randn(*shape, generator=g, dtype=torch.float32, device=dev) * std
+        return t.to(dtype)
     g = torch.Generator(device="cpu")
-    g.manual_seed((base_seed * 1000003 + hash(seed_key)) % (2**63))
+    g.manual_seed(seed)
     t = torch.randn(*shape, generator=g, dtype=torch.float32) * std
     return t.to(dtype).to(device)
 
