@@ -1,0 +1,122 @@
+"""hipGraph capture of the decode step.
+
+The profile (profiles/r01) showed the decode path dominated by host gaps
+and per-launch overhead (~300 dispatches/step across 32 layers, GPU busy
+~14%). Instead of a tracing compiler, the whole decode forward is captured
+once per batch-size bucket into a hipGraph (torch.cuda.CUDAGraph == hipGraph
+on ROCm) with static buffers; each step writes the batch into the buffers
+and replays one graph.
+
+Padding safety: rows beyond the live batch point at a reserved pad block
+(last block of the pool) so the captured reshape_and_cache writes land in
+scratch, never in live KV.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from ..models.llama import ForwardMeta
+from .scheduler import ScheduledBatch
+
+BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+
+
+class DecodeGraphRunner:
+    def __init__(self, runner):
+        self.runner = runner
+        cfg = runner.cfg
+        dev = runner.device
+        self.max_bs = min(cfg.max_num_seqs, BUCKETS[-1])
+        self.buckets = [b for b in BUCKETS if b <= self.max_bs]
+        if self.buckets[-1] != self.max_bs:
+            self.buckets.append(self.max_bs)
+        self.max_blocks = (cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
+        kv = runner.kv
+        self.pad_block = kv.pad_block
+        pad_slot = self.pad_block * cfg.block_size
+
+        mb = self.max_bs
+        self.tokens = torch.zeros(mb, dtype=torch.long, device=dev)
+        self.positions = torch.zeros(mb, dtype=torch.long, device=dev)
+        self.slots = torch.full((mb,), pad_slot, dtype=torch.long, device=dev)
+        self.block_tables = torch.full(
+            (mb, self.max_blocks), self.pad_block, dtype=torch.int32, device=dev
+        )
+        self.seq_lens = torch.ones(mb, dtype=torch.int32, device=dev)
+        # pinned host staging
+        pin = dev.type == "cuda"
+        self.h_tokens = torch.zeros(mb, dtype=torch.long, pin_memory=pin)
+        self.h_positions = torch.zeros(mb, dtype=torch.long, pin_memory=pin)
+        self.h_slots = torch.zeros(mb, dtype=torch.long, pin_memory=pin)
+        self.h_bt = torch.zeros(mb, self.max_blocks, dtype=torch.int32, pin_memory=pin)
+        self.h_seq_lens = torch.ones(mb, dtype=torch.int32, pin_memory=pin)
+        self._pad_slot = pad_slot
+        self._prev_bs = 0
+        self.graphs: dict[int, torch.cuda.CUDAGraph] = {}
+        self.outs: dict[int, torch.Tensor] = {}
+
+    def _meta(self, bs: int) -> ForwardMeta:
+        return ForwardMeta(
+            is_prefill=False,
+            positions=self.positions[:bs],
+            slot_mapping=self.slots[:bs],
+            logits_indices=torch.arange(bs, dtype=torch.long, device=self.runner.device),
+            block_tables=self.block_tables[:bs],
+            seq_lens=self.seq_lens[:bs],
+        )
+
+    def capture(self) -> None:
+        model, kv = self.runner.model, self.runner.kv
+        pool = torch.cuda.graph_pool_handle()
+        for bs in reversed(self.buckets):  # largest first reserves the pool
+            meta = self._meta(bs)
+            model(self.tokens[:bs], meta, kv)  # eager warmup (blas workspaces)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                out = model(self.tokens[:bs], meta, kv)
+            self.graphs[bs] = g
+            self.outs[bs] = out
+        torch.cuda.synchronize()
+
+    def can_run(self, batch: ScheduledBatch) -> bool:
+        return (not batch.is_prefill) and len(batch.seqs) <= self.max_bs and self.graphs
+
+    def run(self, batch: ScheduledBatch) -> torch.Tensor:
+        bs = len(batch.seqs)
+        bucket = next(b for b in self.buckets if b >= bs)
+        # host staging
+        self.h_tokens[:bs] = torch.tensor(batch.token_ids, dtype=torch.long)
+        self.h_positions[:bs] = torch.tensor(batch.positions, dtype=torch.long)
+        self.h_slots[:bs] = torch.tensor(batch.slot_mapping, dtype=torch.long)
+        self.h_seq_lens[:bs] = torch.tensor(batch.seq_lens, dtype=torch.int32)
+        maxb = 0
+        for i, s in enumerate(batch.seqs):
+            nb = len(s.block_table)
+            self.h_bt[i, :nb] = torch.tensor(s.block_table, dtype=torch.int32)
+            maxb = max(maxb, nb)
+        # pad rows dirtied by a previous (larger) batch
+        hi = max(self._prev_bs, bucket)
+        if hi > bs:
+            self.h_tokens[bs:hi] = 0
+            self.h_positions[bs:hi] = 0
+            self.h_slots[bs:hi] = self._pad_slot
+            self.h_seq_lens[bs:hi] = 1
+            self.h_bt[bs:hi, 0] = self.pad_block
+        self._prev_bs = bucket
+        n = hi
+        self.tokens[:n].copy_(self.h_tokens[:n], non_blocking=True)
+        self.positions[:n].copy_(self.h_positions[:n], non_blocking=True)
+        self.slots[:n].copy_(self.h_slots[:n], non_blocking=True)
+        self.seq_lens[:n].copy_(self.h_seq_lens[:n], non_blocking=True)
+        self.block_tables[:n, :max(maxb, 1)].copy_(
+            self.h_bt[:n, :max(maxb, 1)], non_blocking=True
+        )
+        self.graphs[bucket].replay()
+        return self.outs[bucket][:bs]
+
+
+def graphs_enabled() -> bool:
+    return os.environ.get("GPUSTACK_AMD_NO_GRAPHS", "0") != "1"

@@ -46,7 +46,10 @@ class KVCache:
         self.v_caches = [
             torch.zeros(shape, dtype=dtype, device=device) for _ in range(spec.num_layers)
         ]
-        self.allocator = BlockAllocator(num_blocks)
+        # last block reserved as the graph-padding scratch block (see
+        # engine/graph_runner.py): padded rows write/read there, never live KV
+        self.pad_block = num_blocks - 1
+        self.allocator = BlockAllocator(max(1, num_blocks - 1))
 
     @staticmethod
     def compute_num_blocks(cfg: EngineConfig, free_bytes: int) -> int:

@@ -63,6 +63,12 @@ class ModelRunner:
         self.cfg = cfg
         self.comm = comm or Communicator()
         self.device = torch.device(cfg.device)
+        if self.device.type == "cuda":
+            try:  # hipBLASLt picks much better skinny-GEMM kernels than rocBLAS
+                torch.backends.cuda.preferred_blas_library("cublaslt")
+            except Exception:  # noqa: BLE001
+                pass
+        self.graph_runner = None
         self.model = LlamaForCausalLM(cfg, self.comm, self.device)
         if cfg.model_dir and not cfg.enforce_random_weights:
             load_safetensors(self.model, cfg, cfg.model_dir)
@@ -85,6 +91,12 @@ class ModelRunner:
         )
         nblocks = min(nblocks, max_needed)
         self.kv = KVCache(cfg, nblocks, self.device)
+        if self.device.type == "cuda":
+            from .graph_runner import DecodeGraphRunner, graphs_enabled
+
+            if graphs_enabled():
+                self.graph_runner = DecodeGraphRunner(self)
+                self.graph_runner.capture()
         return self.kv
 
     def _meta(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
@@ -124,8 +136,11 @@ class ModelRunner:
 
     @torch.inference_mode()
     def execute(self, batch: ScheduledBatch) -> list[int]:
-        tokens, meta = self._meta(batch)
-        logits = self.model(tokens, meta, self.kv)
+        if self.graph_runner is not None and self.graph_runner.can_run(batch):
+            logits = self.graph_runner.run(batch)
+        else:
+            tokens, meta = self._meta(batch)
+            logits = self.model(tokens, meta, self.kv)
         token_ids = self.sampler.sample(logits, batch.seqs)
         if self.comm.tp_size > 1:
             # ranks must agree on sampled tokens; rank 0 decides

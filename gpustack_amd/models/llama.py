@@ -65,13 +65,16 @@ class Attention(nn.Module):
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
         T = x.shape[0]
         qkv = F.linear(x, self.qkv_w, self.qkv_b)
-        q, k, v = qkv.split(
-            [self.hq * self.d, self.hkv * self.d, self.hkv * self.d], dim=-1
-        )
-        q = q.contiguous().view(T, self.hq, self.d)
-        k = k.contiguous().view(T, self.hkv, self.d)
-        v = v.contiguous().view(T, self.hkv, self.d)
+        nq, nk = self.hq * self.d, self.hkv * self.d
+        # strided views into the fused buffer — the HIP kernels take row
+        # strides, so no layout copies on the hot path
+        q = qkv[:, :nq].unflatten(1, (self.hq, self.d))
+        k = qkv[:, nq:nq + nk].unflatten(1, (self.hkv, self.d))
+        v = qkv[:, nq + nk:].unflatten(1, (self.hkv, self.d))
         if self.spec.qk_norm:
+            q = q.contiguous()
+            k = k.contiguous()
+            v = v.contiguous()
             ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
             ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
         ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d, self.d)

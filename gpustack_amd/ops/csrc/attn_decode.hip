@@ -30,7 +30,7 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     const unsigned short* __restrict__ vc,   // [B, Hkv, BS, D]
     const int* __restrict__ block_tables,    // [N, max_blocks]
     const int* __restrict__ seq_lens,        // [N]
-    int Hkv, int max_blocks, float scale) {
+    int Hkv, int max_blocks, float scale, long q_stride) {
   constexpr int LPG = 16;           // lanes per token-group
   constexpr int DV = D / LPG;       // dims per lane (8 for D=128)
   static_assert(DV == 8, "decode kernel assumes D = 128");
@@ -51,7 +51,7 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
   for (int gq = 0; gq < GQ; ++gq) {
     const unsigned short* qp =
-        q + ((long)seq * Hq + (long)h * GQ + gq) * D + sub * DV;
+        q + (long)seq * q_stride + ((long)h * GQ + gq) * D + sub * DV;
     u16x8 u = *reinterpret_cast<const u16x8*>(qp);
     bf8_to_f32(u, qr[gq]);
   }
@@ -169,7 +169,7 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 void paged_attn_decode_launch(void* out, const void* q, const void* kc,
                               const void* vc, const int* block_tables,
                               const int* seq_lens, int N, int Hq, int Hkv,
-                              int D, int max_blocks, float scale,
+                              int D, int max_blocks, float scale, long q_stride,
                               int* err_unsupported, hipStream_t s) {
   const int GQ = Hq / Hkv;
   dim3 grid(N, Hkv);
@@ -180,7 +180,7 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
   hipLaunchKernelGGL((paged_attn_decode_kernel<128, G>), grid, block, 0, s,  \
                      (unsigned short*)out, (const unsigned short*)q,         \
                      (const unsigned short*)kc, (const unsigned short*)vc,   \
-                     block_tables, seq_lens, Hkv, max_blocks, scale)
+                     block_tables, seq_lens, Hkv, max_blocks, scale, q_stride)
   switch (GQ) {
     case 1: LAUNCH_GQ(1); break;
     case 2: LAUNCH_GQ(2); break;

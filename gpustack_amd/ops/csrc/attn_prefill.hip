@@ -53,7 +53,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     const int* __restrict__ tile_start,    // [ntiles] seq start (global row)
     const int* __restrict__ tile_q0,       // [ntiles] q-tile offset in seq
     const int* __restrict__ tile_len,      // [ntiles] seq length
-    int Hq, int Hkv, float scale) {
+    int Hq, int Hkv, float scale,
+    long qs, long ks, long vs) {           // row strides (elements)
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
   const int kvh = qh / (Hq / Hkv);
@@ -76,8 +77,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   u16x8 qfrag[4];
 #pragma unroll
   for (int kk = 0; kk < 4; ++kk) {
-    const unsigned short* qp = q + ((long)(seq0 + qrow_clamped) * Hq + qh) * PF_D +
-                               kk * 32 + lg * 8;
+    const unsigned short* qp = q + (long)(seq0 + qrow_clamped) * qs +
+                               (long)qh * PF_D + kk * 32 + lg * 8;
     qfrag[kk] = *reinterpret_cast<const u16x8*>(qp);
   }
 
@@ -104,7 +105,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
         u16x8 val{0, 0, 0, 0, 0, 0, 0, 0};
         if (kv0 + kv < len) {
           val = *reinterpret_cast<const u16x8*>(
-              k + ((long)(seq0 + kv0 + kv) * Hkv + kvh) * PF_D + d0);
+              k + (long)(seq0 + kv0 + kv) * ks + (long)kvh * PF_D + d0);
         }
         *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
                                   kswz(kv, d0 * 2)) = val;
@@ -118,7 +119,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
         for (int j = 0; j < 8; ++j) {
           const int kv = kv0 + kvc + j;
           tmp[j] = (kv < len)
-                       ? v[((long)(seq0 + kv) * Hkv + kvh) * PF_D + d]
+                       ? v[(long)(seq0 + kv) * vs + (long)kvh * PF_D + d]
                        : (unsigned short)0;
         }
         *reinterpret_cast<u16x8*>(&VTl[d][kvc]) =
@@ -236,6 +237,7 @@ void flash_prefill_launch(void* out, const void* q, const void* k,
                           const void* v, const int* tile_start,
                           const int* tile_q0, const int* tile_len, int ntiles,
                           int Hq, int Hkv, int D, float scale,
+                          long qs, long ks, long vs,
                           int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
   if (D != 128 || Hq % Hkv != 0) { *err_unsupported = 1; return; }
@@ -243,7 +245,7 @@ void flash_prefill_launch(void* out, const void* q, const void* k,
   hipLaunchKernelGGL(flash_prefill_kernel, grid, dim3(PF_THREADS), 0, s,
                      (unsigned short*)out, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
-                     tile_start, tile_q0, tile_len, Hq, Hkv, scale);
+                     tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks, vs);
 }
 
 // ---------------------------------------------------------------------------

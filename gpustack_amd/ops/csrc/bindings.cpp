@@ -7,12 +7,12 @@
 // launchers (defined in the .hip translation units)
 void rms_norm_launch(void*, const void*, const void*, float, int, int, hipStream_t);
 void fused_add_rms_norm_launch(void*, void*, const void*, float, int, int, hipStream_t);
-void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, int, int, hipStream_t);
+void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, int, int, long, long, hipStream_t);
 void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
-void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, hipStream_t);
+void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, hipStream_t);
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
-void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, int*, hipStream_t);
-void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, int*, hipStream_t);
+void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int*, hipStream_t);
+void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
@@ -32,6 +32,17 @@ void check_bf16(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// [T, H, D] tensor allowed a non-contiguous row stride (view into a fused
+// qkv buffer); heads and dims must stay contiguous.
+long row_stride_3d(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 3, name, " must be [T, H, D]");
+  TORCH_CHECK(t.stride(2) == 1 && t.stride(1) == t.size(2),
+              name, " heads/dims must be contiguous");
+  return t.stride(0);
 }
 
 void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
@@ -58,16 +69,18 @@ void fused_add_rms_norm(at::Tensor x, at::Tensor residual, at::Tensor weight,
 
 void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
                       at::Tensor cos_sin, long head_dim, long rot_dim) {
-  check_bf16(q, "q"); check_bf16(k, "k");
+  const long qs = row_stride_3d(q, "q");
+  const long ks = row_stride_3d(k, "k");
   TORCH_CHECK(positions.scalar_type() == at::kLong && positions.is_contiguous());
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
   const int T = positions.size(0);
   const int D = (int)head_dim, R = (int)rot_dim;
-  const int Hq = q.numel() / ((long)T * D);
-  const int Hk = k.numel() / ((long)T * D);
+  const int Hq = q.size(1);
+  const int Hk = k.size(1);
   TORCH_CHECK((R / 2) % 8 == 0, "rot_dim/2 must be a multiple of 8");
   rope_neox_launch(positions.data_ptr<long>(), q.data_ptr(), k.data_ptr(),
-                   cos_sin.data_ptr<float>(), T, Hq, Hk, D, R, cur_stream(q));
+                   cos_sin.data_ptr<float>(), T, Hq, Hk, D, R, qs, ks,
+                   cur_stream(q));
   HIP_CHECK_LAST();
 }
 
@@ -82,7 +95,8 @@ void silu_and_mul(at::Tensor out, at::Tensor x) {
 
 void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slots) {
-  check_bf16(k, "k"); check_bf16(v, "v");
+  const long ks = row_stride_3d(k, "k");
+  const long vs = row_stride_3d(v, "v");
   check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
   TORCH_CHECK(slots.scalar_type() == at::kLong && slots.is_contiguous());
   const int T = k.size(0), Hkv = k.size(1), D = k.size(2);
@@ -90,7 +104,7 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   TORCH_CHECK(k_cache.size(1) == Hkv && k_cache.size(3) == D && D % 8 == 0);
   reshape_and_cache_launch(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
                            v_cache.data_ptr(), slots.data_ptr<long>(), T, Hkv,
-                           D, BS, cur_stream(k));
+                           D, BS, ks, vs, cur_stream(k));
   HIP_CHECK_LAST();
 }
 
@@ -106,7 +120,8 @@ void greedy_sample(at::Tensor out, at::Tensor logits) {
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
                        at::Tensor seq_lens, double scale) {
-  check_bf16(out, "out"); check_bf16(q, "q");
+  check_bf16(out, "out");
+  const long qstride = row_stride_3d(q, "q");
   check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt && block_tables.is_contiguous());
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
@@ -119,7 +134,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   paged_attn_decode_launch(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                            v_cache.data_ptr(), block_tables.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), N, Hq, Hkv, D, max_blocks,
-                           (float)scale, &err, cur_stream(q));
+                           (float)scale, qstride, &err, cur_stream(q));
   TORCH_CHECK(!err, "paged_attn_decode: unsupported head_dim/GQ combination: D=",
               D, " Hq=", Hq, " Hkv=", Hkv);
   HIP_CHECK_LAST();
@@ -128,7 +143,10 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                    at::Tensor tile_start, at::Tensor tile_q0,
                    at::Tensor tile_len, double scale) {
-  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(out, "out");
+  const long qs = row_stride_3d(q, "q");
+  const long ks = row_stride_3d(k, "k");
+  const long vs = row_stride_3d(v, "v");
   TORCH_CHECK(tile_start.scalar_type() == at::kInt && tile_start.is_contiguous());
   const int ntiles = tile_start.size(0);
   const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
@@ -136,7 +154,7 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   flash_prefill_launch(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        tile_start.data_ptr<int>(), tile_q0.data_ptr<int>(),
                        tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D,
-                       (float)scale, &err, cur_stream(q));
+                       (float)scale, qs, ks, vs, &err, cur_stream(q));
   TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D);
   HIP_CHECK_LAST();
 }

@@ -132,7 +132,8 @@ __global__ void rope_neox_kernel(const long* __restrict__ positions,
                                  unsigned short* __restrict__ q,
                                  unsigned short* __restrict__ k,
                                  const float* __restrict__ cos_sin,
-                                 int T, int Hq, int Hk, int D, int R) {
+                                 int T, int Hq, int Hk, int D, int R,
+                                 long qs, long ks) {  // row strides (elements)
   const int half = R / 2;
   const int chunks_per_head = half / 8; // R/2 % 8 == 0 enforced on host
   const long total = (long)T * (Hq + Hk) * chunks_per_head;
@@ -144,7 +145,8 @@ __global__ void rope_neox_kernel(const long* __restrict__ positions,
     const long pos = positions[t];
     const float* cs = cos_sin + pos * R + c * 8;
     unsigned short* base =
-        (h < Hq) ? q + ((long)t * Hq + h) * D : k + ((long)t * Hk + (h - Hq)) * D;
+        (h < Hq) ? q + (long)t * qs + (long)h * D
+                 : k + (long)t * ks + (long)(h - Hq) * D;
     u16x8 u1 = *reinterpret_cast<const u16x8*>(base + c * 8);
     u16x8 u2 = *reinterpret_cast<const u16x8*>(base + half + c * 8);
     float o1[8], o2[8];
@@ -195,7 +197,8 @@ __global__ void reshape_and_cache_kernel(const unsigned short* __restrict__ k,
                                          unsigned short* __restrict__ kc,
                                          unsigned short* __restrict__ vc,
                                          const long* __restrict__ slots,
-                                         int T, int Hkv, int D, int BS) {
+                                         int T, int Hkv, int D, int BS,
+                                         long ks, long vs) {
   const int chunks = D / 8;
   const long total = (long)T * Hkv * chunks;
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
@@ -207,9 +210,11 @@ __global__ void reshape_and_cache_kernel(const unsigned short* __restrict__ k,
     if (slot < 0) continue;
     const long blk = slot / BS, off = slot % BS;
     const long dst = ((blk * Hkv + h) * BS + off) * D + c * 8;
-    const long src = ((long)t * Hkv + h) * D + c * 8;
-    *reinterpret_cast<u16x8*>(kc + dst) = *reinterpret_cast<const u16x8*>(k + src);
-    *reinterpret_cast<u16x8*>(vc + dst) = *reinterpret_cast<const u16x8*>(v + src);
+    const long hoff = (long)h * D + c * 8;
+    *reinterpret_cast<u16x8*>(kc + dst) =
+        *reinterpret_cast<const u16x8*>(k + (long)t * ks + hoff);
+    *reinterpret_cast<u16x8*>(vc + dst) =
+        *reinterpret_cast<const u16x8*>(v + (long)t * vs + hoff);
   }
 }
 
@@ -291,11 +296,11 @@ void fused_add_rms_norm_launch(void* x, void* residual, const void* w,
 
 void rope_neox_launch(const long* positions, void* q, void* k,
                       const float* cos_sin, int T, int Hq, int Hk, int D,
-                      int R, hipStream_t s) {
+                      int R, long qs, long ks, hipStream_t s) {
   long total = (long)T * (Hq + Hk) * (R / 2 / 8);
   hipLaunchKernelGGL(rope_neox_kernel, dim3(pw_grid(total, 256)), dim3(256), 0,
                      s, positions, (unsigned short*)q, (unsigned short*)k,
-                     cos_sin, T, Hq, Hk, D, R);
+                     cos_sin, T, Hq, Hk, D, R, qs, ks);
 }
 
 void silu_and_mul_launch(void* out, const void* x, long T, int I,
@@ -307,12 +312,12 @@ void silu_and_mul_launch(void* out, const void* x, long T, int I,
 
 void reshape_and_cache_launch(const void* k, const void* v, void* kc, void* vc,
                               const long* slots, int T, int Hkv, int D, int BS,
-                              hipStream_t s) {
+                              long ks, long vs, hipStream_t s) {
   long total = (long)T * Hkv * (D / 8);
   hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(pw_grid(total, 256)),
                      dim3(256), 0, s, (const unsigned short*)k,
                      (const unsigned short*)v, (unsigned short*)kc,
-                     (unsigned short*)vc, slots, T, Hkv, D, BS);
+                     (unsigned short*)vc, slots, T, Hkv, D, BS, ks, vs);
 }
 
 void greedy_sample_launch(long* out, const void* logits, int N, int V,

@@ -74,7 +74,7 @@ def rotary_embedding(
     sin = cs[:, half:].unsqueeze(1)
     for x in (q, k):
         T = x.shape[0]
-        xs = x.view(T, -1, head_dim)
+        xs = x if x.dim() == 3 else x.view(T, -1, head_dim)
         x1 = xs[..., :half].float()
         x2 = xs[..., half : 2 * half].float()
         o1 = x1 * cos - x2 * sin
