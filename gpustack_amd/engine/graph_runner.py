@@ -54,15 +54,19 @@ class DecodeGraphRunner:
         self.h_seq_lens = torch.ones(mb, dtype=torch.int32, pin_memory=pin)
         self._pad_slot = pad_slot
         self._prev_bs = 0
+        # persistent — every tensor a captured graph references must outlive
+        # the graph (an ephemeral arange here caused replay-time faults)
+        self.logits_idx = torch.arange(mb, dtype=torch.long, device=dev)
         self.graphs: dict[int, torch.cuda.CUDAGraph] = {}
         self.outs: dict[int, torch.Tensor] = {}
+        self.metas: dict[int, ForwardMeta] = {}
 
     def _meta(self, bs: int) -> ForwardMeta:
         return ForwardMeta(
             is_prefill=False,
             positions=self.positions[:bs],
             slot_mapping=self.slots[:bs],
-            logits_indices=torch.arange(bs, dtype=torch.long, device=self.runner.device),
+            logits_indices=self.logits_idx[:bs],
             block_tables=self.block_tables[:bs],
             seq_lens=self.seq_lens[:bs],
         )
@@ -79,6 +83,7 @@ class DecodeGraphRunner:
                 out = model(self.tokens[:bs], meta, kv)
             self.graphs[bs] = g
             self.outs[bs] = out
+            self.metas[bs] = meta  # keep every referenced tensor alive
         torch.cuda.synchronize()
 
     def can_run(self, batch: ScheduledBatch) -> bool:

@@ -63,7 +63,9 @@ class ModelRunner:
         self.cfg = cfg
         self.comm = comm or Communicator()
         self.device = torch.device(cfg.device)
-        if self.device.type == "cuda":
+        import os
+
+        if self.device.type == "cuda" and os.environ.get("GPUSTACK_AMD_BLASLT", "1") == "1":
             try:  # hipBLASLt picks much better skinny-GEMM kernels than rocBLAS
                 torch.backends.cuda.preferred_blas_library("cublaslt")
             except Exception:  # noqa: BLE001

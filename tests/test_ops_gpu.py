@@ -147,3 +147,77 @@ def test_greedy_sample():
     logits = torch.randn(64, 128256, dtype=torch.bfloat16, device="cuda")
     got = ops.greedy_sample(logits)
     assert torch.equal(got.cpu(), logits.float().argmax(-1).cpu())
+
+
+# ---- strided (fused qkv view) paths: exactly what the model runs ----
+
+def _fused_qkv(T, Hq, Hkv, D):
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device="cuda")
+    nq, nk = Hq * D, Hkv * D
+    q = qkv[:, :nq].unflatten(1, (Hq, D))
+    k = qkv[:, nq:nq + nk].unflatten(1, (Hkv, D))
+    v = qkv[:, nq + nk:].unflatten(1, (Hkv, D))
+    return qkv, q, k, v
+
+
+def test_rope_strided_matches_contiguous():
+    T, Hq, Hkv, D = 33, 32, 8, 128
+    qkv, q, k, v = _fused_qkv(T, Hq, Hkv, D)
+    pos = torch.randint(0, 1024, (T,), device="cuda")
+    cache = ops.build_cos_sin_cache(D, D, 2048, device="cuda")
+    qc, kc = q.contiguous(), k.contiguous()
+    ops.rotary_embedding(pos, q, k, cache, D, D)
+    ops.rotary_embedding(pos, qc, kc, cache, D, D)
+    _close(q, qc, atol=0, rtol=0)
+    _close(k, kc, atol=0, rtol=0)
+
+
+def test_reshape_and_cache_strided():
+    T, Hq, Hkv, D, BS, B = 48, 32, 8, 128, 16, 8
+    _, _, k, v = _fused_qkv(T, Hq, Hkv, D)
+    kc = torch.zeros(B, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    slots = torch.randperm(B * BS, device="cuda")[:T]
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    ops.reshape_and_cache(k.contiguous(), v.contiguous(), kc2, vc2, slots)
+    assert torch.equal(kc, kc2) and torch.equal(vc, vc2)
+
+
+def test_paged_decode_strided_q():
+    import math
+    Hq, Hkv, D, BS = 32, 8, 128, 16
+    lens = [7, 40, 180]
+    N = len(lens)
+    nb = sum((l + BS - 1) // BS for l in lens)
+    kcache = torch.randn(nb, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    vcache = torch.randn(nb, Hkv, BS, D, dtype=torch.bfloat16, device="cuda")
+    maxb = (max(lens) + BS - 1) // BS
+    bt = torch.zeros(N, maxb, dtype=torch.int32, device="cuda")
+    nxt = 0
+    for i, l in enumerate(lens):
+        n = (l + BS - 1) // BS
+        bt[i, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    _, q, _, _ = _fused_qkv(N, Hq, Hkv, D)
+    sl = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    scale = 1 / math.sqrt(D)
+    out = torch.empty(N, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out2 = torch.empty_like(out)
+    ops.paged_attn_decode(out, q, kcache, vcache, bt, sl, scale)
+    ops.paged_attn_decode(out2, q.contiguous(), kcache, vcache, bt, sl, scale)
+    assert torch.equal(out, out2)
+
+
+def test_flash_prefill_strided():
+    import math
+    Hq, Hkv, D = 32, 8, 128
+    lens = [70, 130, 5]
+    T = sum(lens)
+    _, q, k, v = _fused_qkv(T, Hq, Hkv, D)
+    out = torch.empty(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out2 = torch.empty_like(out)
+    scale = 1 / math.sqrt(D)
+    ops.varlen_prefill_attn(out, q, k, v, lens, scale)
+    ops.varlen_prefill_attn(out2, q.contiguous(), k.contiguous(), v.contiguous(), lens, scale)
+    assert torch.equal(out, out2)
