@@ -54,6 +54,10 @@ class DecodeGraphRunner:
         self.h_seq_lens = torch.ones(mb, dtype=torch.int32, pin_memory=pin)
         self._pad_slot = pad_slot
         self._prev_bs = 0
+        # per-row cache for incremental block-table staging: (request_id,
+        # nblocks) — decode batches are stable between steps, so most rows
+        # need no host work at all
+        self._row_state: list[tuple[str, int] | None] = [None] * mb
         # persistent — every tensor a captured graph references must outlive
         # the graph (an ephemeral arange here caused replay-time faults)
         self.logits_idx = torch.arange(mb, dtype=torch.long, device=dev)
@@ -98,10 +102,16 @@ class DecodeGraphRunner:
         self.h_slots[:bs] = torch.tensor(batch.slot_mapping, dtype=torch.long)
         self.h_seq_lens[:bs] = torch.tensor(batch.seq_lens, dtype=torch.int32)
         maxb = 0
+        dirty_lo, dirty_hi = self.max_bs, -1
         for i, s in enumerate(batch.seqs):
             nb = len(s.block_table)
-            self.h_bt[i, :nb] = torch.tensor(s.block_table, dtype=torch.int32)
             maxb = max(maxb, nb)
+            state = (s.request_id, nb)
+            if self._row_state[i] != state:
+                self.h_bt[i, :nb] = torch.tensor(s.block_table, dtype=torch.int32)
+                self._row_state[i] = state
+                dirty_lo = min(dirty_lo, i)
+                dirty_hi = max(dirty_hi, i)
         # pad rows dirtied by a previous (larger) batch
         hi = max(self._prev_bs, bucket)
         if hi > bs:
@@ -110,15 +120,20 @@ class DecodeGraphRunner:
             self.h_slots[bs:hi] = self._pad_slot
             self.h_seq_lens[bs:hi] = 1
             self.h_bt[bs:hi, 0] = self.pad_block
+            for i in range(bs, hi):
+                self._row_state[i] = None
+            dirty_lo = min(dirty_lo, bs)
+            dirty_hi = max(dirty_hi, hi - 1)
         self._prev_bs = bucket
         n = hi
         self.tokens[:n].copy_(self.h_tokens[:n], non_blocking=True)
         self.positions[:n].copy_(self.h_positions[:n], non_blocking=True)
         self.slots[:n].copy_(self.h_slots[:n], non_blocking=True)
         self.seq_lens[:n].copy_(self.h_seq_lens[:n], non_blocking=True)
-        self.block_tables[:n, :max(maxb, 1)].copy_(
-            self.h_bt[:n, :max(maxb, 1)], non_blocking=True
-        )
+        if dirty_hi >= 0:
+            self.block_tables[dirty_lo:dirty_hi + 1, :max(maxb, 1)].copy_(
+                self.h_bt[dirty_lo:dirty_hi + 1, :max(maxb, 1)], non_blocking=True
+            )
         self.graphs[bucket].replay()
         return self.outs[bucket][:bs]
 

@@ -70,6 +70,8 @@ class ModelRunner:
                 torch.backends.cuda.preferred_blas_library("cublaslt")
             except Exception:  # noqa: BLE001
                 pass
+        if self.device.type == "cuda":
+            self._init_tunableop()
         self.graph_runner = None
         self.model = LlamaForCausalLM(cfg, self.comm, self.device)
         if cfg.model_dir and not cfg.enforce_random_weights:
@@ -78,6 +80,34 @@ class ModelRunner:
             random_init(self.model, cfg)
         self.sampler = Sampler(self.device)
         self.kv: KVCache | None = None
+
+    def _init_tunableop(self) -> None:
+        """hipBLASLt algorithm selection via torch TunableOp.
+
+        The shipped per-shape tuning table (ops/tunableop_gfx950.csv, produced
+        by scripts/tune_gemms once on an MI355X) is loaded read-only; set
+        GPUSTACK_AMD_TUNE=1 to re-tune and write a fresh table.
+        """
+        import os
+        from pathlib import Path
+
+        try:
+            tunable = torch.cuda.tunable
+        except AttributeError:
+            return
+        tuning = os.environ.get("GPUSTACK_AMD_TUNE", "0") == "1"
+        shipped = Path(__file__).resolve().parent.parent / "ops" / "tunableop_gfx950.csv"
+        if tuning:
+            out = os.environ.get("GPUSTACK_AMD_TUNE_OUT", "gpurun_out/tunableop_gfx950.csv")
+            Path(out).parent.mkdir(parents=True, exist_ok=True)
+            tunable.set_filename(out)
+            tunable.enable(True)
+            tunable.tuning_enable(True)
+        elif shipped.exists() and os.environ.get("GPUSTACK_AMD_TUNABLEOP", "1") == "1":
+            tunable.set_filename(str(shipped))
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.read_file(str(shipped))
 
     def init_kv_cache(self) -> KVCache:
         cfg = self.cfg
