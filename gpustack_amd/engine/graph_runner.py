@@ -52,6 +52,11 @@ class DecodeGraphRunner:
         self.h_slots = torch.zeros(mb, dtype=torch.long, pin_memory=pin)
         self.h_bt = torch.zeros(mb, self.max_blocks, dtype=torch.int32, pin_memory=pin)
         self.h_seq_lens = torch.ones(mb, dtype=torch.int32, pin_memory=pin)
+        # cached numpy views of the pinned staging (shared memory)
+        self.n_tokens = self.h_tokens.numpy()
+        self.n_positions = self.h_positions.numpy()
+        self.n_slots = self.h_slots.numpy()
+        self.n_seq_lens = self.h_seq_lens.numpy()
         self._pad_slot = pad_slot
         self._prev_bs = 0
         # per-row cache for incremental block-table staging: (request_id,
@@ -94,13 +99,15 @@ class DecodeGraphRunner:
         return (not batch.is_prefill) and len(batch.seqs) <= self.max_bs and self.graphs
 
     def run(self, batch: ScheduledBatch) -> torch.Tensor:
+        import numpy as np
+
         bs = len(batch.seqs)
         bucket = next(b for b in self.buckets if b >= bs)
-        # host staging
-        self.h_tokens[:bs] = torch.tensor(batch.token_ids, dtype=torch.long)
-        self.h_positions[:bs] = torch.tensor(batch.positions, dtype=torch.long)
-        self.h_slots[:bs] = torch.tensor(batch.slot_mapping, dtype=torch.long)
-        self.h_seq_lens[:bs] = torch.tensor(batch.seq_lens, dtype=torch.int32)
+        # host staging (numpy views over pinned memory)
+        self.n_tokens[:bs] = batch.token_ids
+        self.n_positions[:bs] = batch.positions
+        self.n_slots[:bs] = batch.slot_mapping
+        self.n_seq_lens[:bs] = np.asarray(batch.seq_lens, dtype=np.int32)
         maxb = 0
         dirty_lo, dirty_hi = self.max_bs, -1
         for i, s in enumerate(batch.seqs):
@@ -115,10 +122,10 @@ class DecodeGraphRunner:
         # pad rows dirtied by a previous (larger) batch
         hi = max(self._prev_bs, bucket)
         if hi > bs:
-            self.h_tokens[bs:hi] = 0
-            self.h_positions[bs:hi] = 0
-            self.h_slots[bs:hi] = self._pad_slot
-            self.h_seq_lens[bs:hi] = 1
+            self.n_tokens[bs:hi] = 0
+            self.n_positions[bs:hi] = 0
+            self.n_slots[bs:hi] = self._pad_slot
+            self.n_seq_lens[bs:hi] = 1
             self.h_bt[bs:hi, 0] = self.pad_block
             for i in range(bs, hi):
                 self._row_state[i] = None

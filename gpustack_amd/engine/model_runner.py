@@ -133,9 +133,9 @@ class ModelRunner:
 
     def _meta(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
         dev = self.device
-        tokens = torch.tensor(batch.token_ids, dtype=torch.long, device=dev)
-        positions = torch.tensor(batch.positions, dtype=torch.long, device=dev)
-        slots = torch.tensor(batch.slot_mapping, dtype=torch.long, device=dev)
+        tokens = torch.as_tensor(batch.token_ids, dtype=torch.long).to(dev)
+        positions = torch.as_tensor(batch.positions, dtype=torch.long).to(dev)
+        slots = torch.as_tensor(batch.slot_mapping, dtype=torch.long).to(dev)
         if batch.is_prefill:
             # last token of each sequence produces the next-token logits
             idx, off = [], 0

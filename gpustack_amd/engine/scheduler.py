@@ -11,6 +11,8 @@ from __future__ import annotations
 from collections import deque
 from dataclasses import dataclass, field
 
+import numpy as np
+
 from .config import EngineConfig
 from .kv_cache import KVCache
 from .sequence import Sequence, SeqStatus
@@ -21,9 +23,11 @@ class ScheduledBatch:
     is_prefill: bool
     seqs: list[Sequence] = field(default_factory=list)
     # flat token ids / positions / KV-write slots for the whole batch
-    token_ids: list[int] = field(default_factory=list)
-    positions: list[int] = field(default_factory=list)
-    slot_mapping: list[int] = field(default_factory=list)
+    # (lists for prefill; int64 numpy arrays for decode — single-pass host
+    # assembly keeps the per-step gap small)
+    token_ids: "list[int] | np.ndarray" = field(default_factory=list)
+    positions: "list[int] | np.ndarray" = field(default_factory=list)
+    slot_mapping: "list[int] | np.ndarray" = field(default_factory=list)
     seq_lens: list[int] = field(default_factory=list)   # context length per seq
 
     @property
@@ -123,13 +127,24 @@ class Scheduler:
             i += 1
         if not self.running:
             return None
-        for seq in self.running:
+        n = len(self.running)
+        bs = self.kv.block_size
+        toks = np.empty(n, dtype=np.int64)
+        poss = np.empty(n, dtype=np.int64)
+        slots = np.empty(n, dtype=np.int64)
+        lens: list[int] = []
+        for i, seq in enumerate(self.running):
             pos = seq.num_tokens - 1
-            batch.seqs.append(seq)
-            batch.token_ids.append(seq.all_token_ids[-1])
-            batch.positions.append(pos)
-            batch.slot_mapping.extend(self.kv.slots_for(seq.block_table, pos, 1))
-            batch.seq_lens.append(seq.num_tokens)
+            out = seq.output_token_ids
+            toks[i] = out[-1] if out else seq.prompt_token_ids[-1]
+            poss[i] = pos
+            slots[i] = seq.block_table[pos // bs] * bs + pos % bs
+            lens.append(pos + 1)
+        batch.seqs = list(self.running)
+        batch.token_ids = toks
+        batch.positions = poss
+        batch.slot_mapping = slots
+        batch.seq_lens = lens
         return batch
 
     # -- lifecycle ---------------------------------------------------------
