@@ -1,0 +1,89 @@
+"""CLI entry (reference: gpustack/main.py + cmd/start.py).
+
+  python -m gpustack_amd start [--config FILE] [flags]           # server
+  python -m gpustack_amd start --server-url http://... --token T # worker
+  python -m gpustack_amd chat --model NAME --prompt "..."        # client
+  python -m gpustack_amd version
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import sys
+
+
+def _add_start_flags(p: argparse.ArgumentParser) -> None:
+    p.add_argument("--config", dest="config_file", default=None)
+    p.add_argument("--server-url", dest="server_url", default=None)
+    p.add_argument("--host", default=None)
+    p.add_argument("--port", type=int, default=None)
+    p.add_argument("--data-dir", dest="data_dir", default=None)
+    p.add_argument("--database-url", dest="database_url", default=None)
+    p.add_argument("--token", default=None)
+    p.add_argument("--worker-name", dest="worker_name", default=None)
+    p.add_argument("--worker-ip", dest="worker_ip", default=None)
+    p.add_argument("--worker-port", dest="worker_port", type=int, default=None)
+    p.add_argument("--bootstrap-password", dest="bootstrap_password", default=None)
+    p.add_argument("--disable-auth", dest="disable_auth", action="store_const", const=True, default=None)
+    p.add_argument("--debug", action="store_true")
+
+
+def main(argv: list[str] | None = None) -> int:
+    ap = argparse.ArgumentParser(prog="gpustack-amd")
+    sub = ap.add_subparsers(dest="cmd")
+    sp = sub.add_parser("start", help="run server (default) or worker (--server-url)")
+    _add_start_flags(sp)
+    cp = sub.add_parser("chat", help="quick chat against a served model")
+    cp.add_argument("--url", default="http://127.0.0.1:8080")
+    cp.add_argument("--api-key", default=None)
+    cp.add_argument("--model", required=True)
+    cp.add_argument("--prompt", required=True)
+    sub.add_parser("version")
+    args = ap.parse_args(argv)
+
+    if args.cmd == "version":
+        from . import __version__
+
+        print(__version__)
+        return 0
+    if args.cmd == "chat":
+        import httpx
+
+        headers = {"Authorization": f"Bearer {args.api_key}"} if args.api_key else {}
+        r = httpx.post(f"{args.url}/v1/chat/completions", headers=headers, json={
+            "model": args.model,
+            "messages": [{"role": "user", "content": args.prompt}],
+            "max_tokens": 64,
+        }, timeout=120)
+        print(r.json())
+        return 0
+    if args.cmd != "start":
+        ap.print_help()
+        return 1
+
+    logging.basicConfig(
+        level=logging.DEBUG if args.debug else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s: %(message)s",
+    )
+    from .config import load_config
+
+    overrides = {
+        k: getattr(args, k)
+        for k in ("server_url", "host", "port", "data_dir", "database_url", "token",
+                  "worker_name", "worker_ip", "worker_port", "bootstrap_password",
+                  "disable_auth")
+    }
+    cfg = load_config(args.config_file, overrides)
+    if cfg.server_role == "worker":
+        from .worker.agent import run_worker
+
+        run_worker(cfg)
+    else:
+        from .server.app import run_server
+
+        run_server(cfg)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,174 @@
+"""Placement policies (reference: gpustack/policies/*).
+
+Filter chain -> MI355X resource-fit selector -> scorer chain, with a
+claim-based allocation model: allocatable VRAM = device total - sum of
+DB-recorded claims - system reserved (reference: policies/utils.py:91-150),
+and an analytic VRAM claim for the native engine:
+weights x 1.2 + framework overhead + KV pool share
+(reference memory model: policies/utils.py:384-470 estimate_model_vram,
+re-derived for the first-party engine on 288 GB HBM3E devices).
+"""
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass, field
+
+from ..engine.config import PRESETS, ModelSpec
+from ..schemas import Model, ModelInstance, PlacementStrategy, Worker, WorkerState
+
+logger = logging.getLogger(__name__)
+
+FRAMEWORK_OVERHEAD = 2 << 30      # engine runtime + activations (LLM)
+WEIGHT_FUDGE = 1.2
+MIN_KV_BYTES = 4 << 30            # refuse placements with <4 GiB KV pool
+
+
+@dataclass
+class Candidate:
+    worker: dict
+    gpu_indexes: list[int]
+    score: float = 0.0
+    vram_claim: dict[int, int] = field(default_factory=dict)  # gpu idx -> bytes
+
+
+def model_spec_for(model: Model | dict) -> ModelSpec | None:
+    source = model["source"] if isinstance(model, dict) else model.source
+    ref = model["model_ref"] if isinstance(model, dict) else model.model_ref
+    if source == "preset":
+        return PRESETS.get(ref)
+    try:
+        return ModelSpec.from_dir(ref)
+    except Exception:  # noqa: BLE001
+        return None
+
+
+def estimate_vram_claim(model: Model | dict, spec: ModelSpec | None, tp: int) -> int:
+    """Per-GPU VRAM claim (bytes) for one replica shard."""
+    gmu = (model["gpu_memory_utilization"] if isinstance(model, dict)
+           else model.gpu_memory_utilization) or 0.9
+    if spec is None:
+        return 16 << 30
+    weights = int(spec.weight_bytes() * WEIGHT_FUDGE) // tp + FRAMEWORK_OVERHEAD
+    # the engine grabs gmu x free for KV; claim the weights + a KV floor and
+    # let the scorer prefer roomier devices
+    return weights + MIN_KV_BYTES
+
+
+# ---- filters (reference: policies/worker_filters/*) -----------------------
+
+def status_filter(workers: list[dict], model: dict) -> list[dict]:
+    return [w for w in workers if w.get("state") == WorkerState.READY.value]
+
+
+def label_filter(workers: list[dict], model: dict) -> list[dict]:
+    sel = model.get("worker_selector") or {}
+    if not sel:
+        return workers
+    return [
+        w for w in workers
+        if all((w.get("labels") or {}).get(k) == v for k, v in sel.items())
+    ]
+
+
+def gpu_arch_filter(workers: list[dict], model: dict) -> list[dict]:
+    """MI355X-native engine runs on gfx950 (rocm) devices."""
+    out = []
+    for w in workers:
+        devs = (w.get("status") or {}).get("gpu_devices", [])
+        if any(d.get("type") == "rocm" for d in devs):
+            out.append(w)
+    return out
+
+
+FILTER_CHAIN = [status_filter, label_filter, gpu_arch_filter]
+
+
+# ---- allocation accounting ------------------------------------------------
+
+def worker_allocatable(worker: dict, instances: list[dict]) -> dict[int, int]:
+    """Free VRAM per GPU index after subtracting recorded claims
+    (claim-based, not live — reference policies/utils.py:91-150)."""
+    reserved = (worker.get("system_reserved") or {}).get("vram", 0)
+    out: dict[int, int] = {}
+    for d in (worker.get("status") or {}).get("gpu_devices", []):
+        total = (d.get("memory") or {}).get("total", 0)
+        out[d.get("index", 0)] = max(0, total - reserved)
+    for inst in instances:
+        if inst.get("worker_id") != worker.get("id"):
+            continue
+        claim = (inst.get("computed_resource_claim") or {}).get("vram", {})
+        for idx_str, bytes_ in claim.items():
+            idx = int(idx_str)
+            if idx in out:
+                out[idx] = max(0, out[idx] - bytes_)
+    return out
+
+
+# ---- selector -------------------------------------------------------------
+
+def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -> list[Candidate]:
+    spec = model_spec_for(model)
+    tp = max(1, model.get("gpus_per_replica") or 1)
+    claim = estimate_vram_claim(model, spec, tp)
+    manual = model.get("gpu_selector") or None
+    out: list[Candidate] = []
+    for w in workers:
+        alloc = worker_allocatable(w, instances)
+        if manual:
+            ids = manual.get("gpu_ids", [])
+            picks = []
+            for gid in ids:
+                parts = str(gid).split(":")
+                if parts[0] in (str(w.get("id")), w.get("name")):
+                    picks.append(int(parts[-1]))
+            if len(picks) >= tp and all(alloc.get(i, 0) >= claim for i in picks[:tp]):
+                out.append(Candidate(w, picks[:tp], vram_claim={i: claim for i in picks[:tp]}))
+            continue
+        fits = sorted(
+            (i for i, free in alloc.items() if free >= claim),
+            key=lambda i: -alloc[i],
+        )
+        if len(fits) >= tp:
+            picks = fits[:tp]
+            out.append(Candidate(w, picks, vram_claim={i: claim for i in picks}))
+    return out
+
+
+# ---- scorers (reference: policies/scorers/*) ------------------------------
+
+def placement_score(cand: Candidate, model: dict, instances: list[dict]) -> float:
+    """binpack: prefer busier workers; spread: prefer emptier ones."""
+    strategy = model.get("placement_strategy", PlacementStrategy.BINPACK.value)
+    alloc = worker_allocatable(cand.worker, instances)
+    total = sum(
+        (d.get("memory") or {}).get("total", 1)
+        for d in (cand.worker.get("status") or {}).get("gpu_devices", [])
+    ) or 1
+    free_frac = sum(alloc.values()) / total
+    if strategy == PlacementStrategy.SPREAD.value:
+        return free_frac * 100
+    return (1 - free_frac) * 100
+
+
+def replica_spread_score(cand: Candidate, model: dict, instances: list[dict]) -> float:
+    """Prefer workers with fewer replicas of the same model (spreads load
+    across the node set at equal placement scores)."""
+    n = sum(
+        1 for i in instances
+        if i.get("model_id") == model.get("id") and i.get("worker_id") == cand.worker.get("id")
+    )
+    return -n
+
+
+def pick_candidate(model: dict, workers: list[dict], instances: list[dict]) -> Candidate | None:
+    flt = workers
+    for f in FILTER_CHAIN:
+        flt = f(flt, model)
+        if not flt:
+            return None
+    cands = select_candidates(model, flt, instances)
+    if not cands:
+        return None
+    for c in cands:
+        c.score = placement_score(c, model, instances) + replica_spread_score(c, model, instances)
+    return max(cands, key=lambda c: c.score)

@@ -1,0 +1,113 @@
+"""Placement scheduler loop (reference: gpustack/scheduler/scheduler.py:85).
+
+Event-driven (instance CREATED) + periodic scan of PENDING instances; for
+each: ANALYZING (resolve ModelSpec / claims) -> find candidate via the
+policy chain -> write worker_id / gpu_indexes / claim -> SCHEDULED.
+Stale SCHEDULED instances (worker never picked them up) are retried
+(reference: scheduler.py:262-299)."""
+from __future__ import annotations
+
+import logging
+import queue
+import time
+
+from ..config import Config
+from ..db import EventType, ar_update, bus, get_session
+from ..schemas import Model, ModelInstance, ModelInstanceState, Worker
+from .policies import estimate_vram_claim, model_spec_for, pick_candidate
+
+logger = logging.getLogger(__name__)
+
+STALE_SCHEDULED_SECONDS = 180.0
+
+
+class PlacementScheduler:
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+        self._stop = False
+
+    def stop(self):
+        self._stop = True
+
+    def run(self) -> None:
+        q = bus.subscribe("model_instances")
+        last_scan = 0.0
+        while not self._stop:
+            try:
+                try:
+                    ev = q.get(timeout=5.0)
+                    if ev.type == EventType.CREATED:
+                        self.schedule_one(ev.data["id"])
+                except queue.Empty:
+                    pass
+                if time.time() - last_scan > 15.0:
+                    last_scan = time.time()
+                    self.scan()
+            except Exception:  # noqa: BLE001
+                logger.exception("scheduler cycle failed")
+                time.sleep(2)
+
+    def scan(self) -> None:
+        now = time.time()
+        with get_session() as s:
+            pending = [
+                i.id for i in s.query(ModelInstance)
+                .filter(ModelInstance.state.in_([
+                    ModelInstanceState.PENDING.value,
+                    ModelInstanceState.ANALYZING.value,
+                ])).all()
+            ]
+            stale = [
+                i.id for i in s.query(ModelInstance)
+                .filter_by(state=ModelInstanceState.SCHEDULED.value).all()
+                if now - i.updated_at > STALE_SCHEDULED_SECONDS
+            ]
+        for iid in pending + stale:
+            self.schedule_one(iid)
+
+    def schedule_one(self, instance_id: int) -> bool:
+        with get_session() as s:
+            inst = s.get(ModelInstance, instance_id)
+            if inst is None or inst.state not in (
+                ModelInstanceState.PENDING.value,
+                ModelInstanceState.ANALYZING.value,
+                ModelInstanceState.SCHEDULED.value,
+            ):
+                return False
+            model = s.get(Model, inst.model_id)
+            if model is None:
+                return False
+            model_d = model.to_dict()
+            spec = model_spec_for(model_d)
+            if spec is None:
+                inst.state = ModelInstanceState.ANALYZING.value
+                inst.state_message = f"cannot resolve model spec for {model.model_ref!r}"
+                ar_update(s, inst)
+                return False
+            workers = [w.to_dict() for w in s.query(Worker).all()]
+            others = [
+                i.to_dict() for i in s.query(ModelInstance).all() if i.id != inst.id
+            ]
+            cand = pick_candidate(model_d, workers, others)
+            if cand is None:
+                inst.state = ModelInstanceState.ANALYZING.value
+                inst.state_message = "no worker fits the resource claim"
+                ar_update(s, inst)
+                return False
+            tp = max(1, model.gpus_per_replica or 1)
+            claim = estimate_vram_claim(model_d, spec, tp)
+            inst.worker_id = cand.worker["id"]
+            inst.worker_ip = cand.worker.get("ip", "")
+            inst.gpu_indexes = cand.gpu_indexes
+            inst.computed_resource_claim = {
+                "vram": {str(i): claim for i in cand.gpu_indexes},
+                "ram": 2 << 30,
+            }
+            inst.state = ModelInstanceState.SCHEDULED.value
+            inst.state_message = ""
+            ar_update(s, inst)
+            logger.info(
+                "scheduled %s -> worker %s gpus %s",
+                inst.name, cand.worker.get("name"), cand.gpu_indexes,
+            )
+            return True

@@ -1,0 +1,129 @@
+"""Pydantic API payloads + re-exported table models."""
+from __future__ import annotations
+
+from pydantic import BaseModel, Field
+
+from .tables import (  # noqa: F401
+    ApiKey,
+    Model,
+    ModelFile,
+    ModelInstance,
+    ModelInstanceState,
+    ModelRoute,
+    ModelUsage,
+    PlacementStrategy,
+    RegistrationToken,
+    SourceEnum,
+    SystemLoad,
+    User,
+    Worker,
+    WorkerState,
+)
+
+
+class UserCreate(BaseModel):
+    username: str
+    password: str
+    is_admin: bool = False
+    full_name: str = ""
+
+
+class UserPublic(BaseModel):
+    id: int
+    username: str
+    is_admin: bool
+    full_name: str = ""
+
+
+class LoginRequest(BaseModel):
+    username: str
+    password: str
+
+
+class ApiKeyCreate(BaseModel):
+    name: str
+    expires_in: float | None = None
+
+
+class ModelCreate(BaseModel):
+    name: str
+    source: str = SourceEnum.PRESET.value
+    model_ref: str = "llama-3-8b"
+    description: str = ""
+    replicas: int = 1
+    categories: list[str] = Field(default_factory=lambda: ["llm"])
+    placement_strategy: str = PlacementStrategy.BINPACK.value
+    worker_selector: dict = Field(default_factory=dict)
+    gpu_selector: dict | None = None
+    gpus_per_replica: int = 1
+    backend_parameters: dict = Field(default_factory=dict)
+    env: dict = Field(default_factory=dict)
+    max_model_len: int | None = None
+    gpu_memory_utilization: float = 0.9
+    speculative_config: dict | None = None
+    extended_kv_cache: dict | None = None
+    distributed_inference_across_workers: bool = False
+    restart_on_error: bool = True
+
+
+class ModelUpdate(BaseModel):
+    replicas: int | None = None
+    description: str | None = None
+    backend_parameters: dict | None = None
+    gpu_memory_utilization: float | None = None
+    max_model_len: int | None = None
+
+
+class GPUDeviceInfo(BaseModel):
+    uuid: str = ""
+    name: str = "AMD Instinct MI355X"
+    vendor: str = "AMD"
+    index: int = 0
+    device_index: int = 0
+    device_chip_index: int = 0
+    arch_family: str = "gfx950"
+    compute_capability: str = "gfx950"
+    driver_version: str = ""
+    runtime_version: str = ""
+    type: str = "rocm"
+    core: dict = Field(default_factory=lambda: {"total": 256, "utilization_rate": 0.0})
+    memory: dict = Field(
+        default_factory=lambda: {
+            "total": 288 * 1024**3,
+            "used": 0,
+            "allocated": 0,
+            "is_unified_memory": False,
+        }
+    )
+    temperature: float = 0.0
+    power_usage: float = 0.0
+
+
+class WorkerRegister(BaseModel):
+    name: str
+    hostname: str = ""
+    ip: str = ""
+    port: int = 10150
+    metrics_port: int = 10152
+    labels: dict = Field(default_factory=dict)
+    status: dict = Field(default_factory=dict)
+    system_reserved: dict = Field(default_factory=dict)
+    token: str = ""
+
+
+class WorkerStatusUpdate(BaseModel):
+    status: dict = Field(default_factory=dict)
+    state_message: str = ""
+
+
+class ModelInstanceUpdate(BaseModel):
+    state: str | None = None
+    state_message: str | None = None
+    port: int | None = None
+    pid: int | None = None
+    restart_count: int | None = None
+
+
+class ModelRouteCreate(BaseModel):
+    name: str
+    targets: list[dict] = Field(default_factory=list)

@@ -1,0 +1,199 @@
+"""SQLAlchemy tables — the data model (reference: gpustack/schemas/*).
+
+Key parity points:
+  * ModelInstance state machine (schemas/models.py:591-607):
+    PENDING -> ANALYZING -> SCHEDULED -> INITIALIZING -> DOWNLOADING ->
+    STARTING -> RUNNING (+ ERROR, UNREACHABLE)
+  * Worker.status carries GPUDeviceInfo entries with type="rocm",
+    arch_family="gfx950", 288 GiB HBM3E memory totals (schemas/workers.py:100)
+  * Model carries replicas / placement / gpu_selector / backend params /
+    speculative + extended-KV config (schemas/models.py:500)
+"""
+from __future__ import annotations
+
+import enum
+import time
+
+from sqlalchemy import JSON, Boolean, Column, Float, ForeignKey, Integer, String, Text
+
+from ..db import Base
+
+
+class TimestampMixin:
+    created_at = Column(Float, default=time.time, nullable=False)
+    updated_at = Column(Float, default=time.time, onupdate=time.time, nullable=False)
+
+
+class SerializeMixin:
+    def to_dict(self) -> dict:
+        return {c.name: getattr(self, c.name) for c in self.__table__.columns}
+
+    def update_from(self, data: dict) -> None:
+        for k, v in data.items():
+            if hasattr(self, k) and k != "id":
+                setattr(self, k, v)
+
+
+class WorkerState(str, enum.Enum):
+    NOT_READY = "not_ready"
+    READY = "ready"
+    UNREACHABLE = "unreachable"
+
+
+class ModelInstanceState(str, enum.Enum):
+    PENDING = "pending"
+    ANALYZING = "analyzing"
+    SCHEDULED = "scheduled"
+    INITIALIZING = "initializing"
+    DOWNLOADING = "downloading"
+    STARTING = "starting"
+    RUNNING = "running"
+    ERROR = "error"
+    UNREACHABLE = "unreachable"
+
+
+class SourceEnum(str, enum.Enum):
+    HUGGING_FACE = "huggingface"
+    LOCAL_PATH = "local_path"
+    PRESET = "preset"  # random-init named architecture (no-network serving)
+
+
+class PlacementStrategy(str, enum.Enum):
+    SPREAD = "spread"
+    BINPACK = "binpack"
+
+
+class User(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "users"
+    id = Column(Integer, primary_key=True)
+    username = Column(String(128), unique=True, nullable=False, index=True)
+    hashed_password = Column(String(256), nullable=False)
+    is_admin = Column(Boolean, default=False)
+    full_name = Column(String(256), default="")
+    require_password_change = Column(Boolean, default=False)
+
+
+class ApiKey(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "api_keys"
+    id = Column(Integer, primary_key=True)
+    user_id = Column(Integer, ForeignKey("users.id"), nullable=False)
+    name = Column(String(128), nullable=False)
+    access_key = Column(String(64), unique=True, index=True, nullable=False)
+    hashed_secret = Column(String(256), nullable=False)
+    expires_at = Column(Float, nullable=True)
+
+
+class RegistrationToken(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "registration_tokens"
+    id = Column(Integer, primary_key=True)
+    token = Column(String(128), unique=True, index=True, nullable=False)
+    description = Column(String(256), default="")
+
+
+class Worker(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "workers"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False)
+    hostname = Column(String(256), default="")
+    ip = Column(String(64), default="")
+    port = Column(Integer, default=10150)
+    metrics_port = Column(Integer, default=10152)
+    state = Column(String(32), default=WorkerState.NOT_READY.value)
+    state_message = Column(Text, default="")
+    labels = Column(JSON, default=dict)
+    # WorkerStatus: cpu/memory/swap/filesystem/os/kernel/gpu_devices
+    # (GPUDeviceInfo: uuid,name,vendor,index,arch_family,compute_capability,
+    #  core{total,utilization_rate}, memory{total,used,allocated}, temperature,
+    #  type="rocm")
+    status = Column(JSON, default=dict)
+    system_reserved = Column(JSON, default=dict)
+    heartbeat_time = Column(Float, default=0.0)
+    unreachable = Column(Boolean, default=False)
+
+
+class Model(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "models"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    description = Column(Text, default="")
+    source = Column(String(32), default=SourceEnum.PRESET.value)
+    # PRESET: preset name (llama-3-8b...); LOCAL_PATH: dir with safetensors;
+    # HUGGING_FACE: repo id (downloaded by the worker model-file manager)
+    model_ref = Column(String(512), nullable=False)
+    replicas = Column(Integer, default=1)
+    categories = Column(JSON, default=lambda: ["llm"])
+    placement_strategy = Column(String(32), default=PlacementStrategy.BINPACK.value)
+    worker_selector = Column(JSON, default=dict)     # label matching
+    gpu_selector = Column(JSON, default=None)        # manual worker:gpu ids
+    gpus_per_replica = Column(Integer, default=1)    # TP degree per replica
+    backend_parameters = Column(JSON, default=dict)  # engine kwargs overrides
+    env = Column(JSON, default=dict)
+    max_model_len = Column(Integer, default=None, nullable=True)
+    gpu_memory_utilization = Column(Float, default=0.9)
+    speculative_config = Column(JSON, default=None)  # {method: ngram/eagle3, ...}
+    extended_kv_cache = Column(JSON, default=None)   # {ram_size/ram_ratio, ...}
+    distributed_inference_across_workers = Column(Boolean, default=False)
+    restart_on_error = Column(Boolean, default=True)
+
+
+class ModelInstance(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "model_instances"
+    id = Column(Integer, primary_key=True)
+    model_id = Column(Integer, ForeignKey("models.id"), nullable=False, index=True)
+    model_name = Column(String(256), nullable=False)
+    name = Column(String(256), unique=True, nullable=False)
+    worker_id = Column(Integer, ForeignKey("workers.id"), nullable=True, index=True)
+    worker_ip = Column(String(64), default="")
+    gpu_indexes = Column(JSON, default=list)
+    state = Column(String(32), default=ModelInstanceState.PENDING.value, index=True)
+    state_message = Column(Text, default="")
+    # {vram: {gpu_idx: bytes}, ram: bytes, kv_blocks: int} (schemas/models.py:623)
+    computed_resource_claim = Column(JSON, default=dict)
+    port = Column(Integer, nullable=True)
+    pid = Column(Integer, nullable=True)
+    restart_count = Column(Integer, default=0)
+    distributed_servers = Column(JSON, default=None)  # subordinate workers
+
+
+class ModelFile(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "model_files"
+    id = Column(Integer, primary_key=True)
+    worker_id = Column(Integer, ForeignKey("workers.id"), nullable=False, index=True)
+    source = Column(String(32), nullable=False)
+    model_ref = Column(String(512), nullable=False)
+    local_path = Column(String(1024), default="")
+    state = Column(String(32), default="pending")   # pending/downloading/ready/error
+    state_message = Column(Text, default="")
+    size_bytes = Column(Integer, default=0)
+    download_progress = Column(Float, default=0.0)
+
+
+class ModelRoute(Base, TimestampMixin, SerializeMixin):
+    """Decouples the published model name from deployments
+    (reference: schemas/model_routes.py)."""
+    __tablename__ = "model_routes"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    targets = Column(JSON, default=list)  # [{model_name, weight}]
+
+
+class ModelUsage(Base, TimestampMixin, SerializeMixin):
+    __tablename__ = "model_usage"
+    id = Column(Integer, primary_key=True)
+    user_id = Column(Integer, index=True)
+    model_id = Column(Integer, index=True)
+    model_name = Column(String(256), index=True)
+    date = Column(String(16), index=True)  # YYYY-MM-DD
+    prompt_tokens = Column(Integer, default=0)
+    completion_tokens = Column(Integer, default=0)
+    request_count = Column(Integer, default=0)
+
+
+class SystemLoad(Base, SerializeMixin):
+    __tablename__ = "system_load"
+    id = Column(Integer, primary_key=True)
+    timestamp = Column(Float, default=time.time, index=True)
+    cpu = Column(Float, default=0.0)
+    ram = Column(Float, default=0.0)
+    gpu = Column(Float, default=0.0)
+    vram = Column(Float, default=0.0)

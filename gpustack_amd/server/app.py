@@ -1,0 +1,138 @@
+"""FastAPI app factory + server bootstrap (reference: gpustack/server/server.py:242).
+
+Startup: init DB, seed admin user + default registration token, start the
+scheduler + controllers, mount /v2 management, /v1 OpenAI, auth and
+Prometheus /metrics routes.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+
+from fastapi import APIRouter, Depends, FastAPI, HTTPException, Response
+from fastapi.responses import JSONResponse
+
+from ..config import Config
+from ..db import ar_create, get_session, init_db
+from ..schemas import LoginRequest, RegistrationToken, User
+from ..security import generate_registration_token, hash_password, jwt_encode, verify_password
+from . import deps
+from .deps import COOKIE_NAME, get_current_user
+from .routes_openai import router as openai_router
+from .routes_v2 import router as v2_router
+
+logger = logging.getLogger(__name__)
+
+auth_router = APIRouter(prefix="/auth")
+
+
+@auth_router.post("/login")
+def login(body: LoginRequest, response: Response):
+    cfg = deps.get_config()
+    with get_session() as s:
+        user = s.query(User).filter_by(username=body.username).first()
+        if not user or not verify_password(body.password, user.hashed_password):
+            raise HTTPException(401, "bad credentials")
+    token = jwt_encode({"sub": body.username}, cfg.get_jwt_secret())
+    response.set_cookie(COOKIE_NAME, token, httponly=True, samesite="lax")
+    return {"token": token}
+
+
+@auth_router.post("/logout")
+def logout(response: Response):
+    response.delete_cookie(COOKIE_NAME)
+    return {"ok": True}
+
+
+@auth_router.get("/me")
+def me(user: User = Depends(get_current_user)):
+    return {"id": user.id, "username": user.username, "is_admin": user.is_admin}
+
+
+def bootstrap_data(cfg: Config) -> dict:
+    """Admin user + default registration token (reference: server.py:381,750)."""
+    out = {}
+    with get_session() as s:
+        admin = s.query(User).filter_by(username="admin").first()
+        if admin is None:
+            import secrets
+
+            password = cfg.bootstrap_password or secrets.token_urlsafe(12)
+            admin = User(username="admin", hashed_password=hash_password(password),
+                         is_admin=True)
+            ar_create(s, admin)
+            out["admin_password"] = password
+            logger.info("bootstrap admin user created (password: %s)", password)
+        tok = s.query(RegistrationToken).first()
+        if tok is None:
+            value = cfg.token or generate_registration_token()
+            tok = RegistrationToken(token=value, description="default")
+            ar_create(s, tok)
+        out["registration_token"] = tok.token
+    return out
+
+
+def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
+    cfg.ensure_dirs()
+    init_db(cfg.resolved_database_url())
+    deps.init_auth(cfg)
+    boot = bootstrap_data(cfg)
+    if cfg.token is None:
+        cfg.token = boot["registration_token"]
+
+    app = FastAPI(title="gpustack_amd", version="0.1.0")
+    app.state.config = cfg
+    app.state.bootstrap = boot
+
+    app.include_router(auth_router)
+    app.include_router(v2_router)
+    app.include_router(openai_router)
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok"}
+
+    @app.get("/readyz")
+    def readyz():
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    def metrics():
+        from .exporter import render_metrics
+
+        return Response(render_metrics(), media_type="text/plain; version=0.0.4")
+
+    @app.exception_handler(Exception)
+    async def unhandled(request, exc):
+        logger.exception("unhandled error: %s", exc)
+        return JSONResponse({"error": {"message": str(exc)}}, status_code=500)
+
+    if start_background:
+        start_background_tasks(cfg, app)
+    return app
+
+
+def start_background_tasks(cfg: Config, app: FastAPI) -> None:
+    from ..scheduler.scheduler import PlacementScheduler
+    from .controllers import ModelController, SystemLoadCollector, WorkerMonitor
+
+    sched = PlacementScheduler(cfg)
+    app.state.scheduler = sched
+    threads = [
+        threading.Thread(target=sched.run, name="scheduler", daemon=True),
+        threading.Thread(target=ModelController(cfg).run, name="model-controller", daemon=True),
+        threading.Thread(target=WorkerMonitor(cfg).run, name="worker-monitor", daemon=True),
+        threading.Thread(target=SystemLoadCollector(cfg).run, name="sysload", daemon=True),
+    ]
+    for t in threads:
+        t.start()
+    app.state.background_threads = threads
+
+
+def run_server(cfg: Config) -> None:
+    import uvicorn
+
+    app = create_app(cfg)
+    logger.info("server listening on %s:%d", cfg.host, cfg.port)
+    uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="info")

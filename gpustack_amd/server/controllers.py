@@ -1,0 +1,169 @@
+"""Reconcilers (reference: gpustack/server/controllers.py).
+
+ModelController.sync_replicas (controllers.py:300): creates/deletes
+ModelInstances to match Model.replicas.
+WorkerMonitor (worker_syncer.py:51-134 + controllers.py:1399): flips stale
+workers NOT_READY/UNREACHABLE and marks their instances UNREACHABLE.
+SystemLoadCollector (system_load.py:113): periodic cluster utilization
+snapshots.
+"""
+from __future__ import annotations
+
+import logging
+import queue
+import time
+
+from ..config import Config
+from ..db import EventType, ar_create, ar_delete, ar_update, bus, get_session
+from ..schemas import (
+    Model, ModelInstance, ModelInstanceState, SystemLoad, Worker, WorkerState,
+)
+
+logger = logging.getLogger(__name__)
+
+HEARTBEAT_TIMEOUT = 60.0
+UNREACHABLE_TIMEOUT = 180.0
+
+
+class ModelController:
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+
+    def run(self) -> None:
+        q = bus.subscribe("models")
+        iq = bus.subscribe("model_instances")
+        self.reconcile_all()
+        while True:
+            try:
+                ev = q.get(timeout=10.0)
+                if ev.type in (EventType.CREATED, EventType.UPDATED):
+                    self.sync_replicas(ev.data["id"])
+            except queue.Empty:
+                pass
+            # drain instance deletions (e.g. user deleted an instance -> recreate)
+            try:
+                while True:
+                    iev = iq.get_nowait()
+                    if iev.type == EventType.DELETED and iev.data.get("model_id"):
+                        self.sync_replicas(iev.data["model_id"])
+            except queue.Empty:
+                pass
+
+    def reconcile_all(self) -> None:
+        with get_session() as s:
+            ids = [m.id for m in s.query(Model).all()]
+        for mid in ids:
+            try:
+                self.sync_replicas(mid)
+            except Exception:  # noqa: BLE001
+                logger.exception("sync_replicas(%s) failed", mid)
+
+    def sync_replicas(self, model_id: int) -> None:
+        with get_session() as s:
+            model = s.get(Model, model_id)
+            if model is None:
+                return
+            insts = s.query(ModelInstance).filter_by(model_id=model_id).all()
+            want, have = model.replicas, len(insts)
+            if have < want:
+                used = {i.name for i in insts}
+                for n in range(want * 2):
+                    if have >= want:
+                        break
+                    name = f"{model.name}-{n}"
+                    if name in used:
+                        continue
+                    inst = ModelInstance(
+                        model_id=model.id, model_name=model.name, name=name,
+                        state=ModelInstanceState.PENDING.value,
+                    )
+                    ar_create(s, inst)
+                    have += 1
+            elif have > want:
+                # scale down: prefer non-RUNNING victims (StatusScorer semantics)
+                order = {st.value: i for i, st in enumerate([
+                    ModelInstanceState.ERROR, ModelInstanceState.UNREACHABLE,
+                    ModelInstanceState.PENDING, ModelInstanceState.ANALYZING,
+                    ModelInstanceState.SCHEDULED, ModelInstanceState.INITIALIZING,
+                    ModelInstanceState.DOWNLOADING, ModelInstanceState.STARTING,
+                    ModelInstanceState.RUNNING,
+                ])}
+                victims = sorted(insts, key=lambda i: order.get(i.state, 9))
+                for v in victims[: have - want]:
+                    ar_delete(s, v)
+
+
+class WorkerMonitor:
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+
+    def run(self) -> None:
+        while True:
+            try:
+                self.check_once()
+            except Exception:  # noqa: BLE001
+                logger.exception("worker monitor cycle failed")
+            time.sleep(15.0)
+
+    def check_once(self) -> None:
+        now = time.time()
+        with get_session() as s:
+            for w in s.query(Worker).all():
+                age = now - (w.heartbeat_time or 0)
+                if age > UNREACHABLE_TIMEOUT and w.state != WorkerState.UNREACHABLE.value:
+                    w.state = WorkerState.UNREACHABLE.value
+                    w.state_message = f"no heartbeat for {int(age)}s"
+                    ar_update(s, w)
+                    self._mark_instances(s, w.id)
+                elif HEARTBEAT_TIMEOUT < age <= UNREACHABLE_TIMEOUT and w.state == WorkerState.READY.value:
+                    w.state = WorkerState.NOT_READY.value
+                    w.state_message = f"no heartbeat for {int(age)}s"
+                    ar_update(s, w)
+
+    def _mark_instances(self, s, worker_id: int) -> None:
+        for inst in s.query(ModelInstance).filter_by(worker_id=worker_id).all():
+            if inst.state == ModelInstanceState.RUNNING.value:
+                inst.state = ModelInstanceState.UNREACHABLE.value
+                inst.state_message = "worker unreachable"
+                ar_update(s, inst)
+
+
+class SystemLoadCollector:
+    def __init__(self, cfg: Config, interval: float = 60.0):
+        self.cfg = cfg
+        self.interval = interval
+
+    def run(self) -> None:
+        while True:
+            try:
+                self.collect_once()
+            except Exception:  # noqa: BLE001
+                logger.exception("system load collection failed")
+            time.sleep(self.interval)
+
+    def collect_once(self) -> None:
+        with get_session() as s:
+            workers = s.query(Worker).filter_by(state=WorkerState.READY.value).all()
+            if not workers:
+                return
+            cpu = ram = gpu = vram = 0.0
+            n_gpu = 0
+            for w in workers:
+                st = w.status or {}
+                cpu += st.get("cpu", {}).get("utilization_rate", 0.0)
+                mem = st.get("memory", {})
+                if mem.get("total"):
+                    ram += mem.get("used", 0) / mem["total"] * 100
+                for d in st.get("gpu_devices", []):
+                    n_gpu += 1
+                    gpu += d.get("core", {}).get("utilization_rate", 0.0)
+                    dm = d.get("memory", {})
+                    if dm.get("total"):
+                        vram += dm.get("used", 0) / dm["total"] * 100
+            nw = len(workers)
+            s.add(SystemLoad(
+                cpu=cpu / nw, ram=ram / nw,
+                gpu=gpu / n_gpu if n_gpu else 0.0,
+                vram=vram / n_gpu if n_gpu else 0.0,
+            ))
+            s.commit()

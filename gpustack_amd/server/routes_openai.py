@@ -1,0 +1,183 @@
+"""OpenAI-compatible inference API + proxy (reference: gpustack/routes/openai.py:185).
+
+Resolves the requested model through model routes (weighted targets) or
+directly by model name, picks a RUNNING instance round-robin
+(reference LB: gpustack/http_proxy/load_balancer.py:7), relays the request
+to the worker-side engine server, streams SSE chunks back, and records
+token usage (reference: api/middlewares.py:81 ModelUsageMiddleware).
+"""
+from __future__ import annotations
+
+import itertools
+import json
+import logging
+import random
+import time
+
+import httpx
+from fastapi import APIRouter, Depends, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from ..db import get_session
+from ..schemas import Model, ModelInstance, ModelInstanceState, ModelRoute, ModelUsage, User
+from .deps import get_current_user
+
+logger = logging.getLogger(__name__)
+
+router = APIRouter()
+
+_rr = itertools.count()
+
+PROXY_TIMEOUT = 1800.0
+
+# one registry shared with the gateway (reference: gateway/utils.py:167-194)
+OPENAI_PATHS = [
+    "/v1/chat/completions",
+    "/v1/completions",
+    "/v1/embeddings",
+    "/v1/rerank",
+    "/v1/score",
+]
+
+
+def _resolve_model_name(name: str) -> str:
+    """Model-route indirection with weighted targets."""
+    with get_session() as s:
+        route = s.query(ModelRoute).filter_by(name=name).first()
+        if route and route.targets:
+            weights = [t.get("weight", 1) for t in route.targets]
+            pick = random.choices(route.targets, weights=weights)[0]
+            return pick.get("model_name", name)
+    return name
+
+
+def _pick_instance(model_name: str) -> tuple[Model, dict]:
+    with get_session() as s:
+        model = s.query(Model).filter_by(name=model_name).first()
+        if not model:
+            raise HTTPException(404, f"model {model_name!r} not found")
+        insts = (
+            s.query(ModelInstance)
+            .filter_by(model_id=model.id, state=ModelInstanceState.RUNNING.value)
+            .all()
+        )
+        if not insts:
+            raise HTTPException(
+                503, f"no running instances for model {model_name!r}"
+            )
+        pick = insts[next(_rr) % len(insts)]  # round-robin
+        return model, pick.to_dict()
+
+
+def _record_usage(user: User, model: Model, usage: dict | None) -> None:
+    if not usage:
+        return
+    try:
+        with get_session() as s:
+            date = time.strftime("%Y-%m-%d")
+            row = (
+                s.query(ModelUsage)
+                .filter_by(user_id=user.id, model_id=model.id, date=date)
+                .first()
+            )
+            if row is None:
+                row = ModelUsage(user_id=user.id, model_id=model.id,
+                                 model_name=model.name, date=date)
+                s.add(row)
+            row.prompt_tokens += usage.get("prompt_tokens", 0)
+            row.completion_tokens += usage.get("completion_tokens", 0)
+            row.request_count += 1
+            s.commit()
+    except Exception:  # noqa: BLE001
+        logger.exception("usage recording failed")
+
+
+async def _proxy(request: Request, path: str, user: User):
+    body_bytes = await request.body()
+    try:
+        body = json.loads(body_bytes or b"{}")
+    except json.JSONDecodeError:
+        raise HTTPException(400, "invalid JSON body")
+    name = body.get("model")
+    if not name:
+        raise HTTPException(400, "missing 'model'")
+    target_name = _resolve_model_name(name)
+    model, inst = _pick_instance(target_name)
+    body["model"] = target_name
+    url = f"http://{inst['worker_ip']}:{inst['port']}{path}"
+    stream = bool(body.get("stream"))
+    client = httpx.AsyncClient(timeout=PROXY_TIMEOUT)
+    req = client.build_request("POST", url, json=body)
+    try:
+        resp = await client.send(req, stream=stream)
+    except httpx.ConnectError:
+        await client.aclose()
+        raise HTTPException(502, "instance unreachable")
+    if not stream:
+        data = await resp.aread()
+        await client.aclose()
+        try:
+            payload = json.loads(data)
+            _record_usage(user, model, payload.get("usage"))
+        except Exception:  # noqa: BLE001
+            payload = None
+        return JSONResponse(
+            content=payload if payload is not None else {"raw": data.decode(errors="replace")},
+            status_code=resp.status_code,
+        )
+
+    async def relay():
+        usage = None
+        try:
+            async for chunk in resp.aiter_lines():
+                if chunk.startswith("data:"):
+                    frag = chunk[5:].strip()
+                    if frag and frag != "[DONE]":
+                        try:
+                            u = json.loads(frag).get("usage")
+                            if u:
+                                usage = u
+                        except json.JSONDecodeError:
+                            pass
+                yield (chunk + "\n\n").encode()
+        except httpx.HTTPError as e:  # mid-stream error frame (openai.py:423-483)
+            yield f"data: {json.dumps({'error': {'message': str(e)}})}\n\n".encode()
+        finally:
+            await resp.aclose()
+            await client.aclose()
+            _record_usage(user, model, usage)
+
+    return StreamingResponse(relay(), media_type="text/event-stream",
+                             status_code=resp.status_code)
+
+
+@router.get("/v1/models")
+def list_models_v1(user: User = Depends(get_current_user)):
+    with get_session() as s:
+        models = s.query(Model).all()
+        routes = s.query(ModelRoute).all()
+        items = [
+            {"id": m.name, "object": "model", "created": int(m.created_at),
+             "owned_by": "gpustack_amd"}
+            for m in models
+        ] + [
+            {"id": r.name, "object": "model", "created": int(r.created_at),
+             "owned_by": "gpustack_amd/route"}
+            for r in routes
+        ]
+    return {"object": "list", "data": items}
+
+
+@router.post("/v1/chat/completions")
+async def chat_completions(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/chat/completions", user)
+
+
+@router.post("/v1/completions")
+async def completions(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/completions", user)
+
+
+@router.post("/v1/embeddings")
+async def embeddings(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/embeddings", user)

@@ -1,0 +1,344 @@
+"""/v2 management API (reference: gpustack/routes/*).
+
+CRUD for users/api-keys/workers/models/instances/routes plus the
+`?watch=true` NDJSON streams that drive worker and controller sync
+(reference watch mechanism: mixins/active_record.py:840,
+client/generated_model_instance_client.py:165).
+"""
+from __future__ import annotations
+
+import json
+import time
+
+from fastapi import APIRouter, Depends, HTTPException, Query, Request
+from fastapi.responses import StreamingResponse
+
+from ..db import Event, EventType, ar_create, ar_delete, ar_update, bus, get_session
+from ..schemas import (
+    ApiKey, ApiKeyCreate, Model, ModelCreate, ModelInstance, ModelInstanceState,
+    ModelInstanceUpdate, ModelRoute, ModelRouteCreate, ModelUpdate, ModelUsage,
+    RegistrationToken, SystemLoad, User, UserCreate, Worker, WorkerRegister,
+    WorkerState, WorkerStatusUpdate,
+)
+from ..security import generate_api_key, generate_registration_token, hash_password
+from .deps import get_admin_user, get_current_user, verify_worker_token
+
+router = APIRouter(prefix="/v2")
+
+
+def watch_ndjson(table: str, snapshot_rows: list[dict], flt):
+    q = bus.subscribe(table)
+    try:
+        for row in snapshot_rows:
+            yield json.dumps({"type": "CREATED", "data": row}) + "\n"
+        import queue as _q
+
+        while True:
+            try:
+                ev: Event = q.get(timeout=15.0)
+            except _q.Empty:
+                yield json.dumps({"type": "HEARTBEAT", "data": {}}) + "\n"
+                continue
+            if ev.type == EventType.HEARTBEAT or (flt and not flt(ev.data)):
+                continue
+            yield json.dumps({"type": ev.type.value, "data": ev.data}) + "\n"
+    finally:
+        bus.unsubscribe(table, q)
+
+
+def _watch_stream(table: str, snapshot_rows: list[dict], flt):
+    return StreamingResponse(watch_ndjson(table, snapshot_rows, flt),
+                             media_type="application/x-ndjson")
+
+
+# ---- users ---------------------------------------------------------------
+
+@router.get("/users")
+def list_users(_: User = Depends(get_admin_user)):
+    with get_session() as s:
+        return {"items": [u.to_dict() | {"hashed_password": None} for u in s.query(User).all()]}
+
+
+@router.post("/users", status_code=201)
+def create_user(body: UserCreate, _: User = Depends(get_admin_user)):
+    with get_session() as s:
+        if s.query(User).filter_by(username=body.username).first():
+            raise HTTPException(409, "username exists")
+        u = User(username=body.username, hashed_password=hash_password(body.password),
+                 is_admin=body.is_admin, full_name=body.full_name)
+        ar_create(s, u)
+        return u.to_dict() | {"hashed_password": None}
+
+
+@router.delete("/users/{user_id}")
+def delete_user(user_id: int, _: User = Depends(get_admin_user)):
+    with get_session() as s:
+        u = s.get(User, user_id)
+        if not u:
+            raise HTTPException(404)
+        ar_delete(s, u)
+        return {"ok": True}
+
+
+# ---- api keys ------------------------------------------------------------
+
+@router.get("/api_keys")
+def list_api_keys(user: User = Depends(get_current_user)):
+    with get_session() as s:
+        keys = s.query(ApiKey).filter_by(user_id=user.id).all()
+        return {"items": [k.to_dict() | {"hashed_secret": None} for k in keys]}
+
+
+@router.post("/api_keys", status_code=201)
+def create_api_key(body: ApiKeyCreate, user: User = Depends(get_current_user)):
+    full, access, hashed = generate_api_key()
+    with get_session() as s:
+        k = ApiKey(user_id=user.id, name=body.name, access_key=access,
+                   hashed_secret=hashed,
+                   expires_at=(time.time() + body.expires_in) if body.expires_in else None)
+        ar_create(s, k)
+        return {"id": k.id, "name": k.name, "value": full}  # secret shown once
+
+
+@router.delete("/api_keys/{key_id}")
+def delete_api_key(key_id: int, user: User = Depends(get_current_user)):
+    with get_session() as s:
+        k = s.get(ApiKey, key_id)
+        if not k or (k.user_id != user.id and not user.is_admin):
+            raise HTTPException(404)
+        ar_delete(s, k)
+        return {"ok": True}
+
+
+# ---- registration tokens -------------------------------------------------
+
+@router.get("/tokens")
+def list_tokens(_: User = Depends(get_admin_user)):
+    with get_session() as s:
+        return {"items": [t.to_dict() for t in s.query(RegistrationToken).all()]}
+
+
+@router.post("/tokens", status_code=201)
+def create_token(_: User = Depends(get_admin_user)):
+    with get_session() as s:
+        t = RegistrationToken(token=generate_registration_token())
+        ar_create(s, t)
+        return t.to_dict()
+
+
+# ---- workers -------------------------------------------------------------
+
+@router.get("/workers")
+def list_workers(request: Request, watch: bool = Query(False),
+                 user: User = Depends(get_current_user)):
+    with get_session() as s:
+        rows = [w.to_dict() for w in s.query(Worker).all()]
+    if watch:
+        return _watch_stream("workers", rows, None)
+    return {"items": rows}
+
+
+@router.post("/workers/register")
+def register_worker(body: WorkerRegister, request: Request,
+                    _=Depends(verify_worker_token)):
+    with get_session() as s:
+        w = s.query(Worker).filter_by(name=body.name).first()
+        if w is None:
+            w = Worker(name=body.name)
+        w.hostname = body.hostname
+        w.ip = body.ip or (request.client.host if request.client else "")
+        w.port = body.port
+        w.metrics_port = body.metrics_port
+        w.labels = body.labels
+        w.status = body.status
+        w.system_reserved = body.system_reserved
+        w.heartbeat_time = time.time()
+        w.state = WorkerState.READY.value
+        if w.id is None:
+            ar_create(s, w)
+        else:
+            ar_update(s, w)
+        return w.to_dict()
+
+
+@router.post("/workers/{worker_id}/status")
+def worker_status(worker_id: int, body: WorkerStatusUpdate,
+                  _=Depends(verify_worker_token)):
+    with get_session() as s:
+        w = s.get(Worker, worker_id)
+        if not w:
+            raise HTTPException(404, "worker not found (re-register)")
+        w.status = body.status or w.status
+        w.heartbeat_time = time.time()
+        if w.state != WorkerState.READY.value:
+            w.state = WorkerState.READY.value
+            ar_update(s, w)
+        else:
+            s.commit()
+        return {"ok": True}
+
+
+@router.post("/workers/{worker_id}/heartbeat")
+def worker_heartbeat(worker_id: int, _=Depends(verify_worker_token)):
+    with get_session() as s:
+        w = s.get(Worker, worker_id)
+        if not w:
+            raise HTTPException(404, "worker not found (re-register)")
+        w.heartbeat_time = time.time()
+        s.commit()
+        return {"ok": True}
+
+
+@router.delete("/workers/{worker_id}")
+def delete_worker(worker_id: int, _: User = Depends(get_admin_user)):
+    with get_session() as s:
+        w = s.get(Worker, worker_id)
+        if not w:
+            raise HTTPException(404)
+        ar_delete(s, w)
+        return {"ok": True}
+
+
+# ---- models ----------------------------------------------------------------
+
+@router.get("/models")
+def list_models(watch: bool = Query(False), user: User = Depends(get_current_user)):
+    with get_session() as s:
+        rows = [m.to_dict() for m in s.query(Model).all()]
+    if watch:
+        return _watch_stream("models", rows, None)
+    return {"items": rows}
+
+
+@router.post("/models", status_code=201)
+def create_model(body: ModelCreate, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        if s.query(Model).filter_by(name=body.name).first():
+            raise HTTPException(409, "model name exists")
+        m = Model(**body.model_dump())
+        ar_create(s, m)
+        return m.to_dict()
+
+
+@router.get("/models/{model_id}")
+def get_model(model_id: int, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        m = s.get(Model, model_id)
+        if not m:
+            raise HTTPException(404)
+        return m.to_dict()
+
+
+@router.patch("/models/{model_id}")
+def update_model(model_id: int, body: ModelUpdate, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        m = s.get(Model, model_id)
+        if not m:
+            raise HTTPException(404)
+        m.update_from({k: v for k, v in body.model_dump().items() if v is not None})
+        ar_update(s, m)
+        return m.to_dict()
+
+
+@router.delete("/models/{model_id}")
+def delete_model(model_id: int, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        m = s.get(Model, model_id)
+        if not m:
+            raise HTTPException(404)
+        for inst in s.query(ModelInstance).filter_by(model_id=model_id).all():
+            ar_delete(s, inst)
+        ar_delete(s, m)
+        return {"ok": True}
+
+
+# ---- model instances -------------------------------------------------------
+
+@router.get("/model_instances")
+def list_instances(watch: bool = Query(False), worker_id: int | None = Query(None),
+                   model_id: int | None = Query(None),
+                   user: User = Depends(get_current_user)):
+    with get_session() as s:
+        q = s.query(ModelInstance)
+        if model_id is not None:
+            q = q.filter_by(model_id=model_id)
+        rows = [i.to_dict() for i in q.all()]
+    if worker_id is not None:
+        # workers watch every instance event and filter locally on
+        # worker_id so SCHEDULED-assignment events reach them
+        flt = None
+        rows = [r for r in rows if r.get("worker_id") == worker_id]
+    else:
+        flt = None
+    if watch:
+        return _watch_stream("model_instances", rows, flt)
+    return {"items": rows}
+
+
+@router.patch("/model_instances/{instance_id}")
+def update_instance(instance_id: int, body: ModelInstanceUpdate,
+                    request: Request, _=Depends(verify_worker_token)):
+    with get_session() as s:
+        inst = s.get(ModelInstance, instance_id)
+        if not inst:
+            raise HTTPException(404)
+        data = {k: v for k, v in body.model_dump().items() if v is not None}
+        inst.update_from(data)
+        ar_update(s, inst)
+        return inst.to_dict()
+
+
+@router.delete("/model_instances/{instance_id}")
+def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        inst = s.get(ModelInstance, instance_id)
+        if not inst:
+            raise HTTPException(404)
+        ar_delete(s, inst)
+        return {"ok": True}
+
+
+# ---- model routes ----------------------------------------------------------
+
+@router.get("/model_routes")
+def list_routes(_: User = Depends(get_current_user)):
+    with get_session() as s:
+        return {"items": [r.to_dict() for r in s.query(ModelRoute).all()]}
+
+
+@router.post("/model_routes", status_code=201)
+def create_route(body: ModelRouteCreate, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        if s.query(ModelRoute).filter_by(name=body.name).first():
+            raise HTTPException(409, "route exists")
+        r = ModelRoute(name=body.name, targets=body.targets)
+        ar_create(s, r)
+        return r.to_dict()
+
+
+@router.delete("/model_routes/{route_id}")
+def delete_route(route_id: int, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        r = s.get(ModelRoute, route_id)
+        if not r:
+            raise HTTPException(404)
+        ar_delete(s, r)
+        return {"ok": True}
+
+
+# ---- usage / system load ---------------------------------------------------
+
+@router.get("/usage")
+def usage(user: User = Depends(get_current_user)):
+    with get_session() as s:
+        q = s.query(ModelUsage)
+        if not user.is_admin:
+            q = q.filter_by(user_id=user.id)
+        return {"items": [u.to_dict() for u in q.all()]}
+
+
+@router.get("/system_load")
+def system_load(_: User = Depends(get_current_user)):
+    with get_session() as s:
+        rows = s.query(SystemLoad).order_by(SystemLoad.timestamp.desc()).limit(120).all()
+        return {"items": [r.to_dict() for r in rows]}

@@ -1,0 +1,160 @@
+"""Worker agent (reference: gpustack/worker/worker.py:65).
+
+Registers with the server (retrying), runs heartbeat + status-sync
+threads, the ServeManager watch/health loops, and a small worker API
+(health, per-instance logs, Prometheus node/GPU metrics — reference
+worker/exporter.py:43)."""
+from __future__ import annotations
+
+import logging
+import socket
+import threading
+import time
+from pathlib import Path
+
+from fastapi import FastAPI, HTTPException
+from fastapi.responses import PlainTextResponse, Response
+
+from ..client import ServerClient
+from ..config import Config
+from .detector import collect_system_status
+from .serve_manager import ServeManager
+
+logger = logging.getLogger(__name__)
+
+
+def _default_ip(server_url: str) -> str:
+    try:
+        host = server_url.split("//")[-1].split("/")[0].split(":")[0]
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        s.connect((host, 80))
+        ip = s.getsockname()[0]
+        s.close()
+        return ip
+    except OSError:
+        return "127.0.0.1"
+
+
+class WorkerAgent:
+    def __init__(self, cfg: Config):
+        assert cfg.server_url, "worker needs --server-url"
+        self.cfg = cfg
+        self.client = ServerClient(cfg.server_url, token=cfg.token)
+        self.worker_id: int | None = None
+        self.serve_manager: ServeManager | None = None
+        self._stop = False
+
+    def register(self) -> None:
+        cfg = self.cfg
+        name = cfg.worker_name or socket.gethostname()
+        ip = cfg.worker_ip or _default_ip(cfg.server_url)
+        status = collect_system_status(cfg.gpu_devices)
+        delay = 2.0
+        while not self._stop:
+            try:
+                w = self.client.register_worker({
+                    "name": name,
+                    "hostname": socket.gethostname(),
+                    "ip": ip,
+                    "port": cfg.worker_port,
+                    "metrics_port": cfg.worker_metrics_port,
+                    "labels": cfg.labels,
+                    "status": status,
+                    "system_reserved": cfg.system_reserved,
+                })
+                self.worker_id = w["id"]
+                logger.info("registered as worker id=%s name=%s ip=%s", w["id"], name, ip)
+                return
+            except Exception as e:  # noqa: BLE001
+                logger.warning("registration failed (%s); retrying in %.0fs", e, delay)
+                time.sleep(delay)
+                delay = min(delay * 2, 30)
+
+    def heartbeat_loop(self) -> None:
+        while not self._stop:
+            try:
+                self.client.worker_heartbeat(self.worker_id)
+            except Exception as e:  # noqa: BLE001
+                logger.warning("heartbeat failed: %s", e)
+                if "404" in str(e):
+                    self.register()
+            time.sleep(self.cfg.heartbeat_interval)
+
+    def status_loop(self) -> None:
+        while not self._stop:
+            try:
+                status = collect_system_status(self.cfg.gpu_devices)
+                self.client.worker_status(self.worker_id, status)
+            except Exception as e:  # noqa: BLE001
+                logger.warning("status sync failed: %s", e)
+            time.sleep(self.cfg.worker_status_interval)
+
+    def create_api(self) -> FastAPI:
+        app = FastAPI(title="gpustack_amd-worker")
+
+        @app.get("/healthz")
+        def healthz():
+            return {"status": "ok", "worker_id": self.worker_id}
+
+        @app.get("/logs/{instance_name}")
+        def logs(instance_name: str, tail: int = 200):
+            path = Path(self.cfg.data_dir) / "log" / "instances" / f"{instance_name}.log"
+            if not path.exists():
+                raise HTTPException(404, "no log for instance")
+            lines = path.read_text(errors="replace").splitlines()
+            return PlainTextResponse("\n".join(lines[-tail:]))
+
+        @app.get("/metrics")
+        def metrics():
+            from prometheus_client import CollectorRegistry, Gauge, generate_latest
+
+            reg = CollectorRegistry()
+            st = collect_system_status(self.cfg.gpu_devices)
+            g_cpu = Gauge("gpustack_worker_cpu_utilization_rate", "cpu util", registry=reg)
+            g_cpu.set(st.get("cpu", {}).get("utilization_rate", 0))
+            g_mem = Gauge("gpustack_worker_memory_used_bytes", "ram used", registry=reg)
+            g_mem.set(st.get("memory", {}).get("used", 0))
+            g_util = Gauge("gpustack_worker_gpu_utilization_rate", "gpu util",
+                           ["index"], registry=reg)
+            g_vt = Gauge("gpustack_worker_gpu_vram_total_bytes", "vram total",
+                         ["index"], registry=reg)
+            g_vu = Gauge("gpustack_worker_gpu_vram_used_bytes", "vram used",
+                         ["index"], registry=reg)
+            g_temp = Gauge("gpustack_worker_gpu_temperature_celsius", "temp",
+                           ["index"], registry=reg)
+            for d in st.get("gpu_devices", []):
+                idx = str(d["index"])
+                g_util.labels(idx).set(d.get("core", {}).get("utilization_rate", 0))
+                g_vt.labels(idx).set(d.get("memory", {}).get("total", 0))
+                g_vu.labels(idx).set(d.get("memory", {}).get("used", 0))
+                g_temp.labels(idx).set(d.get("temperature", 0))
+            return Response(generate_latest(reg), media_type="text/plain; version=0.0.4")
+
+        return app
+
+    def start(self) -> None:
+        self.register()
+        self.serve_manager = ServeManager(self.cfg, self.client, self.worker_id)
+        threads = [
+            threading.Thread(target=self.heartbeat_loop, name="heartbeat", daemon=True),
+            threading.Thread(target=self.status_loop, name="status", daemon=True),
+            threading.Thread(target=self.serve_manager.watch_loop, name="serve-watch", daemon=True),
+            threading.Thread(target=self.serve_manager.health_loop, name="serve-health", daemon=True),
+        ]
+        for t in threads:
+            t.start()
+        import uvicorn
+
+        app = self.create_api()
+        logger.info("worker API on :%d", self.cfg.worker_port)
+        uvicorn.run(app, host="0.0.0.0", port=self.cfg.worker_port, log_level="warning")
+
+    def stop(self) -> None:
+        self._stop = True
+        if self.serve_manager:
+            self.serve_manager.stop()
+
+
+def run_worker(cfg: Config) -> None:
+    cfg.ensure_dirs()
+    WorkerAgent(cfg).start()

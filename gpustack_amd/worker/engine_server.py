@@ -1,0 +1,376 @@
+"""Per-instance OpenAI-compatible engine server.
+
+The first-party replacement for the vLLM/SGLang containers the reference
+launches: the worker's serve manager starts this as a
+subprocess per ModelInstance (`python -m gpustack_amd.worker.engine_server`)
+with HIP_VISIBLE_DEVICES pinned to the scheduled GPUs.
+
+A dedicated engine thread runs the continuous-batching loop; request
+handlers feed prompts in and stream sampled tokens out through per-request
+asyncio queues. Endpoints: /v1/chat/completions, /v1/completions,
+/v1/models, /health, /metrics.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import logging
+import threading
+import time
+import uuid
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse, Response, StreamingResponse
+
+logger = logging.getLogger(__name__)
+
+
+class ByteTokenizer:
+    """Reversible fallback tokenizer for preset/random-init serving (no
+    checkpoint, no vocab files): UTF-8 bytes offset past the special ids."""
+
+    OFFSET = 16
+
+    def __init__(self, vocab_size: int):
+        self.vocab_size = vocab_size
+        self.eos_token_id = 1
+
+    def encode(self, text: str) -> list[int]:
+        return [b % (self.vocab_size - self.OFFSET) + self.OFFSET
+                for b in text.encode("utf-8")]
+
+    def decode(self, ids: list[int]) -> str:
+        return bytes((i - self.OFFSET) % 256 for i in ids if i >= self.OFFSET).decode(
+            "utf-8", errors="replace"
+        )
+
+    def apply_chat_template(self, messages: list[dict]) -> str:
+        parts = [f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages]
+        return "\n".join(parts) + "\nassistant:"
+
+
+def load_tokenizer(model_dir: str | None, vocab_size: int):
+    if model_dir:
+        try:
+            from transformers import AutoTokenizer
+
+            return AutoTokenizer.from_pretrained(model_dir)
+        except Exception:  # noqa: BLE001
+            logger.warning("falling back to ByteTokenizer for %s", model_dir)
+    return ByteTokenizer(vocab_size)
+
+
+class EngineRunner:
+    """Owns the LLMEngine + step loop thread; bridges to asyncio."""
+
+    def __init__(self, engine_cfg, served_name: str):
+        from ..engine import LLMEngine
+
+        self.engine = LLMEngine(engine_cfg)
+        self.served_name = served_name
+        self.tokenizer = load_tokenizer(engine_cfg.model_dir, engine_cfg.spec.vocab_size)
+        self.loop: asyncio.AbstractEventLoop | None = None
+        self._queues: dict[str, asyncio.Queue] = {}
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        self._stop = False
+        self.stats = {"requests": 0, "generated_tokens": 0, "prompt_tokens": 0}
+        self.thread = threading.Thread(target=self._run, name="engine-loop", daemon=True)
+
+    def start(self, loop: asyncio.AbstractEventLoop) -> None:
+        self.loop = loop
+        self.thread.start()
+
+    def stop(self) -> None:
+        self._stop = True
+        self._wake.set()
+
+    def _run(self) -> None:
+        while not self._stop:
+            if not self.engine.has_unfinished():
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
+            try:
+                outputs = self.engine.step()
+            except Exception as e:  # noqa: BLE001
+                logger.exception("engine step failed")
+                with self._lock:
+                    for q in self._queues.values():
+                        self._push(q, {"error": str(e), "finished": True})
+                time.sleep(1)
+                continue
+            for out in outputs:
+                with self._lock:
+                    q = self._queues.get(out.request_id)
+                if q is not None:
+                    self.stats["generated_tokens"] += 1
+                    self._push(q, {
+                        "token_id": out.token_id,
+                        "finished": out.finished,
+                        "finish_reason": out.finish_reason,
+                    })
+
+    def _push(self, q: asyncio.Queue, item: dict) -> None:
+        assert self.loop is not None
+        self.loop.call_soon_threadsafe(q.put_nowait, item)
+
+    def submit(self, token_ids: list[int], params) -> tuple[str, asyncio.Queue]:
+        rid = f"req-{uuid.uuid4().hex[:16]}"
+        q: asyncio.Queue = asyncio.Queue()
+        with self._lock:
+            self._queues[rid] = q
+        self.engine.add_request(token_ids, params, request_id=rid)
+        self.stats["requests"] += 1
+        self.stats["prompt_tokens"] += len(token_ids)
+        self._wake.set()
+        return rid, q
+
+    def release(self, rid: str) -> None:
+        with self._lock:
+            self._queues.pop(rid, None)
+
+    def abort(self, rid: str) -> None:
+        self.engine.abort_request(rid)
+        self.release(rid)
+
+
+def _sampling_params(body: dict, eos_token_id: int):
+    from ..engine import SamplingParams
+
+    mt = body.get("max_tokens") or body.get("max_completion_tokens") or 256
+    return SamplingParams(
+        temperature=float(body.get("temperature", 1.0) or 0.0)
+        if body.get("temperature") is not None else 1.0,
+        top_p=float(body.get("top_p", 1.0)),
+        top_k=int(body.get("top_k", 0)),
+        max_tokens=int(mt),
+        ignore_eos=bool(body.get("ignore_eos", False)),
+        seed=body.get("seed"),
+    )
+
+
+def create_app(runner: EngineRunner) -> FastAPI:
+    app = FastAPI(title="gpustack_amd-engine")
+
+    @app.on_event("startup")
+    async def _startup():
+        runner.start(asyncio.get_running_loop())
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok", "model": runner.served_name}
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list", "data": [
+            {"id": runner.served_name, "object": "model", "owned_by": "gpustack_amd"}
+        ]}
+
+    @app.get("/metrics")
+    async def metrics():
+        e = runner.engine
+        lines = [
+            "# TYPE gpustack_engine_requests_total counter",
+            f"gpustack_engine_requests_total {runner.stats['requests']}",
+            "# TYPE gpustack_engine_generated_tokens_total counter",
+            f"gpustack_engine_generated_tokens_total {runner.stats['generated_tokens']}",
+            "# TYPE gpustack_engine_prompt_tokens_total counter",
+            f"gpustack_engine_prompt_tokens_total {runner.stats['prompt_tokens']}",
+            "# TYPE gpustack_engine_num_running gauge",
+            f"gpustack_engine_num_running {e.num_running}",
+            "# TYPE gpustack_engine_num_waiting gauge",
+            f"gpustack_engine_num_waiting {e.num_waiting}",
+            "# TYPE gpustack_engine_kv_blocks_free gauge",
+            f"gpustack_engine_kv_blocks_free {e.scheduler.kv.allocator.num_free}",
+            "# TYPE gpustack_engine_kv_blocks_total gauge",
+            f"gpustack_engine_kv_blocks_total {e.scheduler.kv.allocator.num_blocks}",
+        ]
+        return Response("\n".join(lines) + "\n", media_type="text/plain; version=0.0.4")
+
+    async def _generate(request: Request, body: dict, prompt_ids: list[int],
+                        kind: str, echo_text_prefix: str = ""):
+        params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
+        rid, q = runner.submit(prompt_ids, params)
+        created = int(time.time())
+        stream = bool(body.get("stream"))
+        model_name = runner.served_name
+
+        def chat_chunk(delta: dict, finish: str | None):
+            return {
+                "id": rid, "object": "chat.completion.chunk", "created": created,
+                "model": model_name,
+                "choices": [{"index": 0, "delta": delta, "finish_reason": finish}],
+            }
+
+        def text_chunk(text: str, finish: str | None):
+            return {
+                "id": rid, "object": "text_completion", "created": created,
+                "model": model_name,
+                "choices": [{"index": 0, "text": text, "finish_reason": finish}],
+            }
+
+        if stream:
+            async def gen():
+                tokens: list[int] = []
+                sent_len = 0
+                try:
+                    if kind == "chat":
+                        yield f"data: {json.dumps(chat_chunk({'role': 'assistant'}, None))}\n\n"
+                    while True:
+                        if await request.is_disconnected():
+                            runner.abort(rid)
+                            return
+                        item = await q.get()
+                        if "error" in item:
+                            yield f"data: {json.dumps({'error': {'message': item['error']}})}\n\n"
+                            break
+                        tokens.append(item["token_id"])
+                        text = runner.tokenizer.decode(tokens)
+                        new = text[sent_len:]
+                        # hold back partial unicode replacement chars
+                        if new and not new.endswith("�"):
+                            sent_len = len(text)
+                            payload = (chat_chunk({"content": new}, None)
+                                       if kind == "chat" else text_chunk(new, None))
+                            yield f"data: {json.dumps(payload)}\n\n"
+                        if item["finished"]:
+                            fin = item.get("finish_reason") or "stop"
+                            usage = {
+                                "prompt_tokens": len(prompt_ids),
+                                "completion_tokens": len(tokens),
+                                "total_tokens": len(prompt_ids) + len(tokens),
+                            }
+                            payload = (chat_chunk({}, fin) if kind == "chat"
+                                       else text_chunk("", fin))
+                            payload["usage"] = usage
+                            yield f"data: {json.dumps(payload)}\n\n"
+                            yield "data: [DONE]\n\n"
+                            break
+                finally:
+                    runner.release(rid)
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        tokens: list[int] = []
+        finish = "stop"
+        try:
+            while True:
+                item = await q.get()
+                if "error" in item:
+                    raise HTTPException(500, item["error"])
+                tokens.append(item["token_id"])
+                if item["finished"]:
+                    finish = item.get("finish_reason") or "stop"
+                    break
+        finally:
+            runner.release(rid)
+        text = runner.tokenizer.decode(tokens)
+        usage = {
+            "prompt_tokens": len(prompt_ids),
+            "completion_tokens": len(tokens),
+            "total_tokens": len(prompt_ids) + len(tokens),
+        }
+        if kind == "chat":
+            return JSONResponse({
+                "id": rid, "object": "chat.completion", "created": created,
+                "model": model_name,
+                "choices": [{"index": 0, "message": {"role": "assistant", "content": text},
+                             "finish_reason": finish}],
+                "usage": usage,
+            })
+        return JSONResponse({
+            "id": rid, "object": "text_completion", "created": created,
+            "model": model_name,
+            "choices": [{"index": 0, "text": echo_text_prefix + text,
+                         "finish_reason": finish}],
+            "usage": usage,
+        })
+
+    @app.post("/v1/chat/completions")
+    async def chat(request: Request):
+        body = await request.json()
+        messages = body.get("messages") or []
+        tok = runner.tokenizer
+        if hasattr(tok, "apply_chat_template"):
+            try:
+                prompt = tok.apply_chat_template(messages, tokenize=False,
+                                                 add_generation_prompt=True)
+            except TypeError:
+                prompt = tok.apply_chat_template(messages)
+        else:
+            prompt = "\n".join(f"{m['role']}: {m['content']}" for m in messages)
+        ids = tok.encode(prompt)
+        if hasattr(ids, "ids"):
+            ids = ids.ids
+        return await _generate(request, body, list(ids), "chat")
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        tok = runner.tokenizer
+        if isinstance(prompt, str):
+            ids = tok.encode(prompt)
+            if hasattr(ids, "ids"):
+                ids = ids.ids
+            ids = list(ids)
+        else:
+            ids = [int(t) for t in prompt]  # pre-tokenized
+        echo = body.get("echo") and isinstance(prompt, str)
+        return await _generate(request, body, ids, "text",
+                               echo_text_prefix=prompt if echo else "")
+
+    return app
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--served-name", required=True)
+    ap.add_argument("--source", default="preset")
+    ap.add_argument("--model-ref", required=True)
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--port", type=int, required=True)
+    ap.add_argument("--max-model-len", type=int, default=8192)
+    ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--gpu-memory-utilization", type=float, default=0.9)
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--kv-cache-blocks", type=int, default=None)
+    ap.add_argument("--backend-parameters", default="{}")
+    args = ap.parse_args()
+
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(levelname)s %(name)s: %(message)s")
+    import torch
+
+    from ..engine import EngineConfig
+
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    extra = json.loads(args.backend_parameters)
+    cfg_kwargs = dict(
+        model=args.model_ref,
+        device=device,
+        max_model_len=args.max_model_len,
+        max_num_seqs=args.max_num_seqs,
+        gpu_memory_utilization=args.gpu_memory_utilization,
+        kv_cache_blocks=args.kv_cache_blocks,
+    )
+    if device == "cpu" and args.kv_cache_blocks is None:
+        cfg_kwargs["kv_cache_blocks"] = 1024
+    cfg_kwargs.update({k: v for k, v in extra.items() if k in EngineConfig.__dataclass_fields__})
+    ecfg = EngineConfig(**cfg_kwargs)
+    if args.source == "local_path":
+        ecfg.model_dir = args.model_ref
+        ecfg.enforce_random_weights = False
+    runner = EngineRunner(ecfg, args.served_name)
+    app = create_app(runner)
+    import uvicorn
+
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,267 @@
+"""Worker-side instance lifecycle engine
+(reference: gpustack/worker/serve_manager.py:184).
+
+Watches ModelInstance events for this worker and drives the state machine:
+SCHEDULED -> INITIALIZING -> (DOWNLOADING) -> STARTING -> RUNNING, with
+port assignment from the configured range, per-instance log capture,
+health probing, and crash restarts with exponential backoff capped at
+300 s (reference: serve_manager.py:1842-1885).
+
+Engine processes are subprocesses of this agent (`python -m
+gpustack_amd.worker.engine_server`) with HIP_VISIBLE_DEVICES pinned to the
+scheduled GPUs — the MI355X-native replacement for the reference's
+Docker/K8s workload creation.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import socket
+import subprocess
+import sys
+import threading
+import time
+from pathlib import Path
+
+import httpx
+
+from ..client import ServerClient
+from ..config import Config
+from ..schemas import ModelInstanceState as S
+
+logger = logging.getLogger(__name__)
+
+RESTART_BASE = 10.0
+RESTART_CAP = 300.0
+
+
+class InstanceProcess:
+    def __init__(self, instance: dict, proc: subprocess.Popen, port: int, log_path: Path):
+        self.instance = instance
+        self.proc = proc
+        self.port = port
+        self.log_path = log_path
+        self.started_at = time.time()
+        self.healthy = False
+
+
+class ServeManager:
+    def __init__(self, cfg: Config, client: ServerClient, worker_id: int):
+        self.cfg = cfg
+        self.client = client
+        self.worker_id = worker_id
+        self.processes: dict[int, InstanceProcess] = {}  # instance_id -> proc
+        self._used_ports: set[int] = set()
+        self._lock = threading.Lock()
+        self._restart_at: dict[int, float] = {}
+        self._stop = False
+
+    # ---- main loops ------------------------------------------------------
+
+    def watch_loop(self) -> None:
+        while not self._stop:
+            try:
+                for frame in self.client.watch_instances(self.worker_id):
+                    if self._stop:
+                        return
+                    t = frame.get("type")
+                    data = frame.get("data", {})
+                    if t == "HEARTBEAT":
+                        continue
+                    if data.get("worker_id") != self.worker_id:
+                        if t == "DELETED" and data.get("id") in self.processes:
+                            self._stop_instance(data["id"])
+                        continue
+                    self.dispatch(t, data)
+            except Exception as e:  # noqa: BLE001
+                logger.warning("watch stream broken (%s); reconnecting", e)
+                time.sleep(3)
+
+    def health_loop(self) -> None:
+        while not self._stop:
+            try:
+                self.check_health()
+            except Exception:  # noqa: BLE001
+                logger.exception("health cycle failed")
+            time.sleep(5.0)
+
+    def stop(self) -> None:
+        self._stop = True
+        for iid in list(self.processes):
+            self._stop_instance(iid)
+
+    # ---- event dispatch (serve_manager.py:968) ---------------------------
+
+    def dispatch(self, event_type: str, inst: dict) -> None:
+        iid = inst["id"]
+        state = inst.get("state")
+        if event_type == "DELETED":
+            self._stop_instance(iid)
+            return
+        if state == S.SCHEDULED.value and iid not in self.processes:
+            try:
+                self._start_instance(inst)
+            except Exception as e:  # noqa: BLE001
+                logger.exception("failed to start instance %s", inst.get("name"))
+                self._safe_update(iid, state=S.ERROR.value, state_message=str(e))
+
+    # ---- lifecycle -------------------------------------------------------
+
+    def _assign_port(self) -> int:
+        lo, hi = self.cfg.engine_port_range()
+        with self._lock:
+            for p in range(lo, hi):
+                if p in self._used_ports:
+                    continue
+                with socket.socket() as s:
+                    try:
+                        s.bind(("", p))
+                    except OSError:
+                        continue
+                self._used_ports.add(p)
+                return p
+        raise RuntimeError("no free ports in range")
+
+    def _start_instance(self, inst: dict) -> None:
+        iid = inst["id"]
+        model = self.client.get_model(inst["model_id"])
+        self._safe_update(iid, state=S.INITIALIZING.value)
+
+        source = model.get("source", "preset")
+        ref = model["model_ref"]
+        if source == "local_path" and not Path(ref).exists():
+            self._safe_update(iid, state=S.ERROR.value,
+                              state_message=f"local path {ref} not found")
+            return
+        if source == "huggingface":
+            local = self._download_model(iid, ref)
+            if local is None:
+                return
+            source, ref = "local_path", local
+
+        port = self._assign_port()
+        log_dir = Path(self.cfg.data_dir) / "log" / "instances"
+        log_dir.mkdir(parents=True, exist_ok=True)
+        log_path = log_dir / f"{inst['name']}.log"
+
+        env = dict(os.environ)
+        gpus = inst.get("gpu_indexes") or []
+        if gpus:
+            env["HIP_VISIBLE_DEVICES"] = ",".join(str(g) for g in gpus)
+            env["CUDA_VISIBLE_DEVICES"] = env["HIP_VISIBLE_DEVICES"]
+        env.update(model.get("env") or {})
+
+        bp = dict(model.get("backend_parameters") or {})
+        args = [
+            sys.executable, "-m", "gpustack_amd.worker.engine_server",
+            "--served-name", model["name"],
+            "--source", source,
+            "--model-ref", ref,
+            "--port", str(port),
+            "--gpu-memory-utilization", str(model.get("gpu_memory_utilization") or 0.9),
+        ]
+        if model.get("max_model_len"):
+            args += ["--max-model-len", str(model["max_model_len"])]
+        if bp:
+            args += ["--backend-parameters", json.dumps(bp)]
+
+        logf = open(log_path, "ab")
+        proc = subprocess.Popen(args, env=env, stdout=logf, stderr=subprocess.STDOUT,
+                                start_new_session=True)
+        self.processes[iid] = InstanceProcess(inst, proc, port, log_path)
+        self._safe_update(iid, state=S.STARTING.value, port=port, pid=proc.pid)
+        logger.info("instance %s starting: pid=%d port=%d gpus=%s",
+                    inst["name"], proc.pid, port, gpus)
+
+    def _download_model(self, iid: int, repo: str) -> str | None:
+        self._safe_update(iid, state=S.DOWNLOADING.value)
+        try:
+            from huggingface_hub import snapshot_download
+
+            cache = Path(self.cfg.cache_dir or Path(self.cfg.data_dir) / "cache")
+            path = snapshot_download(repo, cache_dir=str(cache))
+            return path
+        except Exception as e:  # noqa: BLE001
+            self._safe_update(iid, state=S.ERROR.value,
+                              state_message=f"download failed: {e}")
+            return None
+
+    def _stop_instance(self, iid: int) -> None:
+        ip = self.processes.pop(iid, None)
+        self._restart_at.pop(iid, None)
+        if ip is None:
+            return
+        logger.info("stopping instance %s (pid %s)", ip.instance.get("name"), ip.proc.pid)
+        try:
+            os.killpg(ip.proc.pid, 15)
+        except ProcessLookupError:
+            pass
+        try:
+            ip.proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            try:
+                os.killpg(ip.proc.pid, 9)
+            except ProcessLookupError:
+                pass
+        with self._lock:
+            self._used_ports.discard(ip.port)
+
+    # ---- health / restart ------------------------------------------------
+
+    def check_health(self) -> None:
+        for iid, ip in list(self.processes.items()):
+            rc = ip.proc.poll()
+            if rc is not None:
+                self._handle_exit(iid, ip, rc)
+                continue
+            if not ip.healthy:
+                try:
+                    r = httpx.get(f"http://127.0.0.1:{ip.port}/health", timeout=3.0)
+                    if r.status_code == 200:
+                        ip.healthy = True
+                        self._safe_update(iid, state=S.RUNNING.value, state_message="")
+                        logger.info("instance %s RUNNING", ip.instance.get("name"))
+                except httpx.HTTPError:
+                    if time.time() - ip.started_at > 600:
+                        self._safe_update(iid, state=S.ERROR.value,
+                                          state_message="startup timeout")
+                        self._stop_instance(iid)
+
+    def _handle_exit(self, iid: int, ip: InstanceProcess, rc: int) -> None:
+        self.processes.pop(iid, None)
+        with self._lock:
+            self._used_ports.discard(ip.port)
+        inst = ip.instance
+        restart_count = (inst.get("restart_count") or 0) + 1
+        model = None
+        try:
+            model = self.client.get_model(inst["model_id"])
+        except httpx.HTTPError:
+            pass
+        restart = (model or {}).get("restart_on_error", True)
+        msg = f"engine exited with code {rc}"
+        logger.warning("instance %s: %s", inst.get("name"), msg)
+        if restart:
+            delay = min(RESTART_CAP, RESTART_BASE * 2 ** min(restart_count - 1, 8))
+            self._safe_update(iid, state=S.ERROR.value, state_message=msg,
+                              restart_count=restart_count)
+            inst2 = dict(inst, restart_count=restart_count)
+            self._restart_at[iid] = time.time() + delay
+            threading.Timer(delay, self._try_restart, args=(iid, inst2)).start()
+        else:
+            self._safe_update(iid, state=S.ERROR.value, state_message=msg)
+
+    def _try_restart(self, iid: int, inst: dict) -> None:
+        if self._stop or iid in self.processes:
+            return
+        try:
+            self._start_instance(inst)
+        except Exception as e:  # noqa: BLE001
+            self._safe_update(iid, state=S.ERROR.value, state_message=str(e))
+
+    def _safe_update(self, iid: int, **fields) -> None:
+        try:
+            self.client.update_instance(iid, **fields)
+        except httpx.HTTPError as e:
+            logger.warning("instance %d update failed: %s", iid, e)
