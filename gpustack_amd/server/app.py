@@ -118,16 +118,21 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
     from .controllers import ModelController, SystemLoadCollector, WorkerMonitor
 
     sched = PlacementScheduler(cfg)
+    tasks = [sched, ModelController(cfg), WorkerMonitor(cfg), SystemLoadCollector(cfg)]
     app.state.scheduler = sched
+    app.state.background_tasks = tasks
     threads = [
-        threading.Thread(target=sched.run, name="scheduler", daemon=True),
-        threading.Thread(target=ModelController(cfg).run, name="model-controller", daemon=True),
-        threading.Thread(target=WorkerMonitor(cfg).run, name="worker-monitor", daemon=True),
-        threading.Thread(target=SystemLoadCollector(cfg).run, name="sysload", daemon=True),
+        threading.Thread(target=t.run, name=type(t).__name__, daemon=True)
+        for t in tasks
     ]
     for t in threads:
         t.start()
     app.state.background_threads = threads
+
+
+def stop_background_tasks(app: FastAPI) -> None:
+    for t in getattr(app.state, "background_tasks", []):
+        t.stop()
 
 
 def run_server(cfg: Config) -> None:

@@ -28,14 +28,20 @@ UNREACHABLE_TIMEOUT = 180.0
 class ModelController:
     def __init__(self, cfg: Config):
         self.cfg = cfg
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
 
     def run(self) -> None:
         q = bus.subscribe("models")
         iq = bus.subscribe("model_instances")
         self.reconcile_all()
-        while True:
+        while not self._stop:
             try:
                 ev = q.get(timeout=10.0)
+                if self._stop:
+                    return
                 if ev.type in (EventType.CREATED, EventType.UPDATED):
                     self.sync_replicas(ev.data["id"])
             except queue.Empty:
@@ -96,14 +102,21 @@ class ModelController:
 class WorkerMonitor:
     def __init__(self, cfg: Config):
         self.cfg = cfg
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
 
     def run(self) -> None:
-        while True:
+        while not self._stop:
             try:
                 self.check_once()
             except Exception:  # noqa: BLE001
                 logger.exception("worker monitor cycle failed")
-            time.sleep(15.0)
+            for _ in range(15):
+                if self._stop:
+                    return
+                time.sleep(1.0)
 
     def check_once(self) -> None:
         now = time.time()
@@ -132,14 +145,21 @@ class SystemLoadCollector:
     def __init__(self, cfg: Config, interval: float = 60.0):
         self.cfg = cfg
         self.interval = interval
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
 
     def run(self) -> None:
-        while True:
+        while not self._stop:
             try:
                 self.collect_once()
             except Exception:  # noqa: BLE001
                 logger.exception("system load collection failed")
-            time.sleep(self.interval)
+            for _ in range(int(self.interval)):
+                if self._stop:
+                    return
+                time.sleep(1.0)
 
     def collect_once(self) -> None:
         with get_session() as s:

@@ -65,6 +65,19 @@ def resolve_user(request: Request) -> User | None:
         return s.query(User).filter_by(username=payload.get("sub", "")).first()
 
 
+def _system_principal(request: Request) -> User | None:
+    """Registration-token bearers act as a system principal (reference:
+    api/auth.py:262-289 worker/cluster system principals)."""
+    token = _bearer(request)
+    if not token or not token.startswith("tok_"):
+        return None
+    with get_session() as s:
+        if s.query(RegistrationToken).filter_by(token=token).first():
+            return User(id=0, username="system:worker", is_admin=True,
+                        hashed_password="")
+    return None
+
+
 def get_current_user(request: Request) -> User:
     cfg = get_config()
     if cfg.disable_auth:
@@ -72,7 +85,7 @@ def get_current_user(request: Request) -> User:
             u = s.query(User).filter_by(is_admin=True).first()
             if u:
                 return u
-    user = resolve_user(request)
+    user = resolve_user(request) or _system_principal(request)
     if user is None:
         raise HTTPException(401, "not authenticated")
     return user

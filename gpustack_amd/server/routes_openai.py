@@ -82,7 +82,9 @@ def _record_usage(user: User, model: Model, usage: dict | None) -> None:
             )
             if row is None:
                 row = ModelUsage(user_id=user.id, model_id=model.id,
-                                 model_name=model.name, date=date)
+                                 model_name=model.name, date=date,
+                                 prompt_tokens=0, completion_tokens=0,
+                                 request_count=0)
                 s.add(row)
             row.prompt_tokens += usage.get("prompt_tokens", 0)
             row.completion_tokens += usage.get("completion_tokens", 0)

@@ -1,0 +1,161 @@
+"""End-to-end cluster test on CPU (BASELINE.json config 1: server +
+scheduler + gateway with 1 CPU worker serving a tiny model greedily).
+
+Boots a real uvicorn server, a real worker agent (static GPU override so
+the fit filter passes on a GPU-less host), deploys a model through the API,
+waits for the full PENDING -> SCHEDULED -> STARTING -> RUNNING loop
+(engine subprocess included), then round-trips /v1/chat/completions and
+/v1/completions through the server-side proxy, checks usage metering and
+instance logs, and scales to zero.
+"""
+import socket
+import tempfile
+import threading
+import time
+
+import httpx
+import pytest
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture(scope="module")
+def cluster():
+    import uvicorn
+
+    from gpustack_amd.config import Config
+    from gpustack_amd.server.app import create_app
+    from gpustack_amd.worker.agent import WorkerAgent
+
+    sport, wport = _free_port(), _free_port()
+    lo = _free_port()
+    cfg = Config(
+        data_dir=tempfile.mkdtemp(), bootstrap_password="pw",
+        host="127.0.0.1", port=sport,
+    )
+    app = create_app(cfg, start_background=True)
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=sport,
+                                           log_level="warning"))
+    st = threading.Thread(target=server.run, daemon=True)
+    st.start()
+    base = f"http://127.0.0.1:{sport}"
+    for _ in range(100):
+        try:
+            httpx.get(base + "/healthz", timeout=1)
+            break
+        except httpx.HTTPError:
+            time.sleep(0.1)
+
+    wcfg = Config(
+        data_dir=tempfile.mkdtemp(),
+        server_url=base,
+        token=app.state.bootstrap["registration_token"],
+        worker_name="cpu-worker-0",
+        worker_ip="127.0.0.1",
+        worker_port=wport,
+        port_range=f"{lo}-{lo + 50}",
+        gpu_devices=[{"index": 0, "name": "AMD Instinct MI355X",
+                      "memory": {"total": 288 * 1024**3}}],
+        heartbeat_interval=2.0,
+        worker_status_interval=5.0,
+    )
+    agent = WorkerAgent(wcfg)
+    wt = threading.Thread(target=agent.start, daemon=True)
+    wt.start()
+
+    client = httpx.Client(base_url=base, timeout=30)
+    tok = client.post("/auth/login", json={"username": "admin", "password": "pw"}).json()["token"]
+    client.headers["Authorization"] = f"Bearer {tok}"
+    for _ in range(100):
+        if client.get("/v2/workers").json()["items"]:
+            break
+        time.sleep(0.2)
+    yield client, agent
+    from gpustack_amd.server.app import stop_background_tasks
+
+    stop_background_tasks(app)
+    agent.stop()
+    server.should_exit = True
+
+
+@pytest.mark.timeout(180)
+def test_deploy_and_chat(cluster):
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-chat", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201, r.text
+
+    state = None
+    for _ in range(240):  # engine subprocess needs to import torch
+        insts = client.get("/v2/model_instances").json()["items"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-chat",
+        "messages": [{"role": "user", "content": "hello"}],
+        "max_tokens": 8, "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["choices"][0]["message"]["content"] is not None
+    assert data["usage"]["completion_tokens"] == 8
+
+    # streaming
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "tiny-chat",
+        "messages": [{"role": "user", "content": "stream test"}],
+        "max_tokens": 4, "stream": True, "ignore_eos": True,
+    }) as resp:
+        assert resp.status_code == 200
+        frames = [l for l in resp.iter_lines() if l.startswith("data:")]
+    assert frames[-1].strip() == "data: [DONE]"
+
+    # completions endpoint
+    r = client.post("/v1/completions", json={
+        "model": "tiny-chat", "prompt": "abc", "max_tokens": 4, "ignore_eos": True,
+    })
+    assert r.status_code == 200
+    assert r.json()["usage"]["completion_tokens"] == 4
+
+    # usage metering recorded
+    usage = client.get("/v2/usage").json()["items"]
+    assert usage and usage[0]["completion_tokens"] >= 12
+
+    # instance logs reachable on the worker API
+    insts = client.get("/v2/model_instances").json()["items"]
+    wr = httpx.get(
+        f"http://127.0.0.1:{agent.cfg.worker_port}/logs/{insts[0]['name']}",
+        timeout=10)
+    assert wr.status_code == 200
+
+    # /v1/models lists it
+    ids = [m["id"] for m in client.get("/v1/models").json()["data"]]
+    assert "tiny-chat" in ids
+
+    # scale to zero -> instance deleted -> engine stopped
+    mid = client.get("/v2/models").json()["items"][0]["id"]
+    client.patch(f"/v2/models/{mid}", json={"replicas": 0})
+    for _ in range(60):
+        if not client.get("/v2/model_instances").json()["items"]:
+            break
+        time.sleep(0.5)
+    assert not client.get("/v2/model_instances").json()["items"]
+    for _ in range(40):
+        if not agent.serve_manager.processes:
+            break
+        time.sleep(0.5)
+    assert not agent.serve_manager.processes
