@@ -36,6 +36,9 @@ def parse_args():
     ap.add_argument("--osl", type=int, default=128, help="max output tokens per request")
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree: world becomes ONE replica "
+                         "sharded over RCCL/xGMI (default: dp replicas)")
     return ap.parse_args()
 
 
@@ -51,12 +54,19 @@ def main():
         torch.cuda.set_device(local_rank)
 
     dist = None
+    comm = None
+    tp = max(1, args.tp)
     if world > 1:
         import torch.distributed as tdist
 
         dist = tdist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+        if tp > 1:
+            assert world == tp, "bench --tp requires world_size == tp"
+            from gpustack_amd.parallel import Communicator
+
+            comm = Communicator(tp, rank)
 
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 
@@ -67,22 +77,31 @@ def main():
         max_model_len=args.max_model_len,
         max_num_seqs=max(args.concurrency, 8),
         seed=0,
+        tp_size=tp if comm else 1,
+        tp_rank=rank if comm else 0,
     )
     if not use_cuda:  # CPU smoke path: shrink everything
-        cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=256, max_model_len=512)
+        cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=256, max_model_len=512,
+                           tp_size=tp if comm else 1, tp_rank=rank if comm else 0)
         args.concurrency = min(args.concurrency, 8)
         args.isl, args.osl = 32, 16
         model = "tiny"
-    eng = LLMEngine(cfg)
+    eng = LLMEngine(cfg, comm)
+    is_driver = comm is None or rank == 0
 
     rng = random.Random(1234 + rank)
     vocab = cfg.spec.vocab_size
     params = SamplingParams(max_tokens=args.osl, ignore_eos=True)
 
     def refill():
-        while eng.scheduler.num_unfinished < args.concurrency:
+        if not is_driver:
+            return
+        target = args.concurrency
+        known = eng.scheduler.num_unfinished + len(eng._pending_ops)
+        while known < target:
             toks = [rng.randrange(2, vocab) for _ in range(args.isl)]
             eng.add_request(toks, params)
+            known += 1
 
     def one_step() -> int:
         refill()
@@ -128,12 +147,15 @@ def main():
     # aggregate across ranks: total tokens, max elapsed
     if dist:
         dev = device if use_cuda else "cpu"
-        tt = torch.tensor([float(tokens)], device=dev)
-        dist.all_reduce(tt)
-        tot_tokens = int(tt.item())
         te = torch.tensor([elapsed], device=dev)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         elapsed = float(te.item())
+        if comm is not None:
+            tot_tokens = tokens  # TP: one replica, every rank saw the batch
+        else:
+            tt = torch.tensor([float(tokens)], device=dev)
+            dist.all_reduce(tt)
+            tot_tokens = int(tt.item())
     else:
         tot_tokens = tokens
 
@@ -151,15 +173,16 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed * 1000 / args.steps, 3),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if comm is not None else "weak",
             "vs_baseline": None,
             "dtype": cfg.dtype if use_cuda else "float32-cpu-smoke",
             "data": "synthetic random-token prompts, random-init weights",
             "config": {
                 "model": model,
-                "global_batch": args.concurrency * (world if world > 1 else 1),
+                "global_batch": args.concurrency * (1 if comm is not None else (world if world > 1 else 1)),
                 "seq_len": args.isl + args.osl,
-                "parallelism": f"dp{world if world > 1 else args.gpus}",
+                "parallelism": (f"tp{world}" if comm is not None
+                                else f"dp{world if world > 1 else args.gpus}"),
                 "isl": args.isl,
                 "osl": args.osl,
                 "concurrency_per_gpu": args.concurrency,

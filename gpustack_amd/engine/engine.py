@@ -23,10 +23,14 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig, comm: Communicator | None = None):
         self.cfg = cfg
         self.runner = ModelRunner(cfg, comm)
+        self.comm = self.runner.comm
         kv = self.runner.init_kv_cache()
         self.scheduler = Scheduler(cfg, kv)
         self.seqs: dict[str, Sequence] = {}
         self._counter = itertools.count()
+        # TP: rank 0 buffers add/abort ops; step() broadcasts them so every
+        # rank replays the same scheduler state deterministically
+        self._pending_ops: list[tuple] = []
         logger.info(
             "engine ready: model=%s kv_blocks=%d (%.1f GiB KV pool)",
             cfg.model, kv.num_blocks,
@@ -42,20 +46,63 @@ class LLMEngine:
     ) -> str:
         rid = request_id or f"req-{next(self._counter)}"
         params = params or SamplingParams()
+        if self.comm.tp_size > 1:
+            assert self.comm.tp_rank == 0, "requests enter through TP rank 0"
+            self._pending_ops.append(("add", list(prompt_token_ids),
+                                      params.__dict__.copy(), rid))
+            return rid
+        self._apply_add(prompt_token_ids, params, rid)
+        return rid
+
+    def _apply_add(self, prompt_token_ids, params: SamplingParams, rid: str) -> None:
         if len(prompt_token_ids) > self.cfg.max_model_len - 1:
             prompt_token_ids = prompt_token_ids[-(self.cfg.max_model_len - 1):]
         seq = Sequence(rid, list(prompt_token_ids), params)
         self.seqs[rid] = seq
         self.scheduler.add(seq)
-        return rid
 
     def abort_request(self, request_id: str) -> bool:
+        if self.comm.tp_size > 1:
+            assert self.comm.tp_rank == 0
+            self._pending_ops.append(("abort", request_id))
+            return True
         ok = self.scheduler.abort(request_id)
         self.seqs.pop(request_id, None)
         return ok
 
+    def _sync_tp_ops(self) -> None:
+        """Broadcast buffered add/abort ops from rank 0 and apply on all
+        ranks (rank bootstrap + op replication over the RCCL/gloo group)."""
+        import torch.distributed as dist
+
+        ops = [self._pending_ops] if self.comm.tp_rank == 0 else [None]
+        dist.broadcast_object_list(ops, src=0, group=self.comm.group)
+        self._pending_ops = []
+        for op in ops[0]:
+            if op[0] == "add":
+                _, toks, params_d, rid = op
+                self._apply_add(toks, SamplingParams(**params_d), rid)
+            elif op[0] == "abort":
+                rid = op[1]
+                self.scheduler.abort(rid)
+                self.seqs.pop(rid, None)
+
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()
+
+    def tp_active(self) -> bool:
+        """Coordinated loop condition: every TP rank keeps stepping while
+        rank 0 has work (requests enter only through rank 0)."""
+        if self.comm.tp_size == 1:
+            return self.has_unfinished()
+        import torch.distributed as dist
+
+        if self.comm.tp_rank == 0:
+            flag = [self.has_unfinished() or bool(self._pending_ops)]
+        else:
+            flag = [None]
+        dist.broadcast_object_list(flag, src=0, group=self.comm.group)
+        return bool(flag[0])
 
     @property
     def num_running(self) -> int:
@@ -66,6 +113,8 @@ class LLMEngine:
         return len(self.scheduler.waiting)
 
     def step(self) -> list[StepOutput]:
+        if self.comm.tp_size > 1:
+            self._sync_tp_ops()
         batch = self.scheduler.schedule()
         if batch is None:
             return []
