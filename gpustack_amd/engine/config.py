@@ -131,6 +131,9 @@ class EngineConfig:
     model_dir: str | None = None
     # host-DRAM KV offload tier (extended_kv_cache in the reference schema)
     kv_offload_gb: float = 0.0
+    # speculative decoding (reference speculative_config schema):
+    # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
+    speculative: dict | None = None
 
     spec: ModelSpec = field(default_factory=ModelSpec)
 

@@ -122,14 +122,28 @@ class LLMEngine:
         if batch.is_prefill:
             self.scheduler.on_prefill_done(batch)
         outputs: list[StepOutput] = []
-        for seq, tok in zip(batch.seqs, token_ids):
+        rps = 1 if batch.is_prefill else batch.rows_per_seq
+        for i, seq in enumerate(batch.seqs):
+            if rps == 1:
+                emitted = [token_ids[i]]
+            else:
+                rows = token_ids[i * rps:(i + 1) * rps]
+                if seq.params.greedy and batch.k_eff[i] > 0:
+                    from .spec import accept_tokens
+
+                    ke = batch.k_eff[i]
+                    emitted = accept_tokens(batch.drafts[i][:ke], rows[:ke + 1])
+                else:
+                    emitted = [rows[0]]
             seq.record_first_token()
-            seq.output_token_ids.append(tok)
-            reason = self._finish_reason(seq, tok)
-            if reason:
-                self.scheduler.finish_seq(seq, reason)
-                self.seqs.pop(seq.request_id, None)
-            outputs.append(StepOutput(seq.request_id, tok, reason is not None, reason))
+            for tok in emitted:
+                seq.output_token_ids.append(tok)
+                reason = self._finish_reason(seq, tok)
+                outputs.append(StepOutput(seq.request_id, tok, reason is not None, reason))
+                if reason:
+                    self.scheduler.finish_seq(seq, reason)
+                    self.seqs.pop(seq.request_id, None)
+                    break
         return outputs
 
     def _finish_reason(self, seq: Sequence, tok: int) -> str | None:

@@ -20,7 +20,8 @@ import torch
 from ..models.llama import ForwardMeta
 from .scheduler import ScheduledBatch
 
-BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
+           320, 384, 448, 512]
 
 
 class DecodeGraphRunner:
@@ -28,7 +29,9 @@ class DecodeGraphRunner:
         self.runner = runner
         cfg = runner.cfg
         dev = runner.device
-        self.max_bs = min(cfg.max_num_seqs, BUCKETS[-1])
+        spec = getattr(cfg, "speculative", None)
+        rps = 1 + int(spec.get("num_draft_tokens", 3)) if spec else 1
+        self.max_bs = min(cfg.max_num_seqs * rps, BUCKETS[-1])
         self.buckets = [b for b in BUCKETS if b <= self.max_bs]
         if self.buckets[-1] != self.max_bs:
             self.buckets.append(self.max_bs)
@@ -96,12 +99,15 @@ class DecodeGraphRunner:
         torch.cuda.synchronize()
 
     def can_run(self, batch: ScheduledBatch) -> bool:
-        return (not batch.is_prefill) and len(batch.seqs) <= self.max_bs and self.graphs
+        return ((not batch.is_prefill)
+                and len(batch.seqs) * batch.rows_per_seq <= self.max_bs
+                and bool(self.graphs))
 
     def run(self, batch: ScheduledBatch) -> torch.Tensor:
         import numpy as np
 
-        bs = len(batch.seqs)
+        rps = batch.rows_per_seq
+        bs = len(batch.seqs) * rps
         bucket = next(b for b in self.buckets if b >= bs)
         # host staging (numpy views over pinned memory)
         self.n_tokens[:bs] = batch.token_ids
@@ -110,15 +116,17 @@ class DecodeGraphRunner:
         self.n_seq_lens[:bs] = np.asarray(batch.seq_lens, dtype=np.int32)
         maxb = 0
         dirty_lo, dirty_hi = self.max_bs, -1
-        for i, s in enumerate(batch.seqs):
+        for si, s in enumerate(batch.seqs):
             nb = len(s.block_table)
             maxb = max(maxb, nb)
             state = (s.request_id, nb)
-            if self._row_state[i] != state:
-                self.h_bt[i, :nb] = torch.tensor(s.block_table, dtype=torch.int32)
-                self._row_state[i] = state
-                dirty_lo = min(dirty_lo, i)
-                dirty_hi = max(dirty_hi, i)
+            for j in range(rps):
+                i = si * rps + j
+                if self._row_state[i] != state:
+                    self.h_bt[i, :nb] = torch.tensor(s.block_table, dtype=torch.int32)
+                    self._row_state[i] = state
+                    dirty_lo = min(dirty_lo, i)
+                    dirty_hi = max(dirty_hi, i)
         # pad rows dirtied by a previous (larger) batch
         hi = max(self._prev_bs, bucket)
         if hi > bs:

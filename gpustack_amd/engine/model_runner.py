@@ -152,15 +152,19 @@ class ModelRunner:
                 tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
             )
         else:
+            rps = batch.rows_per_seq
+            nrows = len(batch.seqs) * rps
             maxb = max(len(s.block_table) for s in batch.seqs)
-            bt = torch.zeros(len(batch.seqs), maxb, dtype=torch.int32)
+            bt = torch.zeros(nrows, maxb, dtype=torch.int32)
             for i, s in enumerate(batch.seqs):
-                bt[i, : len(s.block_table)] = torch.tensor(s.block_table, dtype=torch.int32)
+                row = torch.tensor(s.block_table, dtype=torch.int32)
+                for j in range(rps):
+                    bt[i * rps + j, : len(s.block_table)] = row
             meta = ForwardMeta(
                 is_prefill=False,
                 positions=positions,
                 slot_mapping=slots,
-                logits_indices=torch.arange(len(batch.seqs), dtype=torch.long, device=dev),
+                logits_indices=torch.arange(nrows, dtype=torch.long, device=dev),
                 block_tables=bt.to(dev),
                 seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
             )
@@ -173,7 +177,10 @@ class ModelRunner:
         else:
             tokens, meta = self._meta(batch)
             logits = self.model(tokens, meta, self.kv)
-        token_ids = self.sampler.sample(logits, batch.seqs)
+        row_seqs = batch.seqs
+        if not batch.is_prefill and batch.rows_per_seq > 1:
+            row_seqs = [s for s in batch.seqs for _ in range(batch.rows_per_seq)]
+        token_ids = self.sampler.sample(logits, row_seqs)
         if self.comm.tp_size > 1:
             # ranks must agree on sampled tokens; rank 0 decides
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)

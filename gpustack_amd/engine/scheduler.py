@@ -28,7 +28,12 @@ class ScheduledBatch:
     token_ids: "list[int] | np.ndarray" = field(default_factory=list)
     positions: "list[int] | np.ndarray" = field(default_factory=list)
     slot_mapping: "list[int] | np.ndarray" = field(default_factory=list)
-    seq_lens: list[int] = field(default_factory=list)   # context length per seq
+    seq_lens: list[int] = field(default_factory=list)   # context length per ROW
+    # speculative decoding: each seq contributes 1+K rows (fed window =
+    # last token + K drafts); see engine/spec.py
+    rows_per_seq: int = 1
+    drafts: list[list[int]] | None = None
+    k_eff: list[int] | None = None
 
     @property
     def num_tokens(self) -> int:
@@ -41,6 +46,18 @@ class Scheduler:
         self.kv = kv
         self.waiting: deque[Sequence] = deque()
         self.running: list[Sequence] = []
+        self.proposer = None
+        self.spec_k = 0
+        spec = getattr(cfg, "speculative", None)
+        if spec and spec.get("method", "ngram") == "ngram":
+            from .spec import NgramProposer
+
+            self.spec_k = int(spec.get("num_draft_tokens", 3))
+            self.proposer = NgramProposer(
+                self.spec_k,
+                int(spec.get("ngram_max", 3)),
+                int(spec.get("ngram_min", 1)),
+            )
 
     # -- queue ops ---------------------------------------------------------
     def add(self, seq: Sequence) -> None:
@@ -107,13 +124,18 @@ class Scheduler:
         batch = ScheduledBatch(is_prefill=False)
         # Ensure every running seq has a slot for its next token; preempt from
         # the back (most recent) on pool exhaustion.
+        k = self.spec_k
         i = 0
         while i < len(self.running):
             seq = self.running[i]
             pos = seq.num_tokens - 1          # position of the token to feed
-            if (pos + 1) > len(seq.block_table) * self.kv.block_size:
+            k_eff = min(k, self.cfg.max_model_len - 1 - pos)
+            k_eff = max(0, k_eff)
+            if (pos + 1 + k_eff) > len(seq.block_table) * self.kv.block_size:
+                need = (pos + 1 + k_eff + self.kv.block_size - 1) // self.kv.block_size \
+                    - len(seq.block_table)
                 try:
-                    seq.block_table.extend(self.kv.allocator.allocate(1))
+                    seq.block_table.extend(self.kv.allocator.allocate(need))
                 except RuntimeError:
                     victim = self.running.pop()  # preempt newest
                     self._release(victim)
@@ -129,22 +151,51 @@ class Scheduler:
             return None
         n = len(self.running)
         bs = self.kv.block_size
-        toks = np.empty(n, dtype=np.int64)
-        poss = np.empty(n, dtype=np.int64)
-        slots = np.empty(n, dtype=np.int64)
+        rps = 1 + self.spec_k
+        toks = np.empty(n * rps, dtype=np.int64)
+        poss = np.empty(n * rps, dtype=np.int64)
+        slots = np.empty(n * rps, dtype=np.int64)
         lens: list[int] = []
+        drafts: list[list[int]] = []
+        k_effs: list[int] = []
         for i, seq in enumerate(self.running):
             pos = seq.num_tokens - 1
             out = seq.output_token_ids
-            toks[i] = out[-1] if out else seq.prompt_token_ids[-1]
-            poss[i] = pos
-            slots[i] = seq.block_table[pos // bs] * bs + pos % bs
-            lens.append(pos + 1)
+            last = out[-1] if out else seq.prompt_token_ids[-1]
+            if rps == 1:
+                toks[i] = last
+                poss[i] = pos
+                slots[i] = seq.block_table[pos // bs] * bs + pos % bs
+                lens.append(pos + 1)
+                continue
+            k_eff = max(0, min(self.spec_k, self.cfg.max_model_len - 1 - pos))
+            if seq.params.greedy and k_eff > 0:
+                draft = self.proposer.propose(seq)
+            else:
+                draft = [last] * self.spec_k
+                k_eff = 0
+            fed = [last] + draft
+            for j in range(rps):
+                r = i * rps + j
+                toks[r] = fed[j]
+                valid = j <= k_eff
+                pj = pos + j if valid else pos + k_eff
+                poss[r] = pj
+                # overflow rows skip the KV write (slot -1) so they cannot
+                # clobber live positions
+                slots[r] = (seq.block_table[pj // bs] * bs + pj % bs) if valid else -1
+                lens.append(pj + 1)
+            drafts.append(draft)
+            k_effs.append(k_eff)
         batch.seqs = list(self.running)
         batch.token_ids = toks
         batch.positions = poss
         batch.slot_mapping = slots
         batch.seq_lens = lens
+        batch.rows_per_seq = rps
+        if rps > 1:
+            batch.drafts = drafts
+            batch.k_eff = k_effs
         return batch
 
     # -- lifecycle ---------------------------------------------------------
