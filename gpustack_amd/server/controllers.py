@@ -50,6 +50,8 @@ class ModelController:
             try:
                 while True:
                     iev = iq.get_nowait()
+                    if self._stop:
+                        return
                     if iev.type == EventType.DELETED and iev.data.get("model_id"):
                         self.sync_replicas(iev.data["model_id"])
             except queue.Empty:
