@@ -8,9 +8,15 @@ logical page, so one block table drives all layers).
 """
 from __future__ import annotations
 
+import contextlib
+
 import torch
 
 from .config import EngineConfig
+
+
+def _nullctx():
+    return contextlib.nullcontext()
 
 
 class BlockAllocator:
@@ -38,6 +44,9 @@ class KVCache:
         self.block_size = cfg.block_size
         self.num_blocks = num_blocks
         kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
+        self.kv_heads = kv_heads
+        self.head_dim = spec.head_dim
+        self.num_layers = spec.num_layers
         shape = (num_blocks, kv_heads, cfg.block_size, spec.head_dim)
         dtype = getattr(torch, cfg.dtype)
         self.k_caches = [
@@ -50,6 +59,67 @@ class KVCache:
         # engine/graph_runner.py): padded rows write/read there, never live KV
         self.pad_block = num_blocks - 1
         self.allocator = BlockAllocator(max(1, num_blocks - 1))
+
+        # ---- host-DRAM offload tier (extended_kv_cache) -----------------
+        # Pinned host pool; swap-out/in run as async copies on a side HIP
+        # stream (reference semantics: LMCache/HiCache tiering,
+        # schemas/models.py:204-215, re-done natively per SURVEY.md §5.7).
+        self.is_cuda = torch.device(device).type == "cuda"
+        host_blocks = 0
+        if cfg.kv_offload_gb > 0:
+            per_block = (2 * spec.num_layers * kv_heads * cfg.block_size
+                         * spec.head_dim * 2)
+            host_blocks = int(cfg.kv_offload_gb * 2**30) // per_block
+        self.host_blocks = host_blocks
+        if host_blocks > 0:
+            self.host_pool = torch.zeros(
+                (host_blocks, spec.num_layers, 2, kv_heads, cfg.block_size,
+                 spec.head_dim),
+                dtype=dtype, pin_memory=self.is_cuda,
+            )
+            self.host_allocator = BlockAllocator(host_blocks)
+            self.side_stream = torch.cuda.Stream() if self.is_cuda else None
+        else:
+            self.host_pool = None
+            self.host_allocator = None
+            self.side_stream = None
+
+    # ---- offload tier ----------------------------------------------------
+
+    def can_swap_out(self, n: int) -> bool:
+        return self.host_allocator is not None and self.host_allocator.num_free >= n
+
+    def swap_out(self, gpu_blocks: list[int]) -> list[int]:
+        """Copy blocks D2H on the side stream and free the GPU blocks."""
+        hblocks = self.host_allocator.allocate(len(gpu_blocks))
+        if self.side_stream is not None:
+            # D2H must observe the compute stream's latest KV writes
+            self.side_stream.wait_stream(torch.cuda.current_stream())
+        ctx = torch.cuda.stream(self.side_stream) if self.side_stream else _nullctx()
+        with ctx:
+            for g, h in zip(gpu_blocks, hblocks):
+                for li in range(self.num_layers):
+                    self.host_pool[h, li, 0].copy_(self.k_caches[li][g], non_blocking=True)
+                    self.host_pool[h, li, 1].copy_(self.v_caches[li][g], non_blocking=True)
+        if self.side_stream is not None:
+            self.side_stream.synchronize()  # blocks must land before reuse
+        self.allocator.free(gpu_blocks)
+        return hblocks
+
+    def swap_in(self, host_blocks: list[int]) -> list[int]:
+        """Allocate GPU blocks and copy H2D; frees the host blocks."""
+        gpu_blocks = self.allocator.allocate(len(host_blocks))
+        ctx = torch.cuda.stream(self.side_stream) if self.side_stream else _nullctx()
+        with ctx:
+            for h, g in zip(host_blocks, gpu_blocks):
+                for li in range(self.num_layers):
+                    self.k_caches[li][g].copy_(self.host_pool[h, li, 0], non_blocking=True)
+                    self.v_caches[li][g].copy_(self.host_pool[h, li, 1], non_blocking=True)
+        if self.side_stream is not None:
+            # compute stream must observe the blocks before attention reads
+            torch.cuda.current_stream().wait_stream(self.side_stream)
+        self.host_allocator.free(host_blocks)
+        return gpu_blocks
 
     @staticmethod
     def compute_num_blocks(cfg: EngineConfig, free_bytes: int) -> int:

@@ -46,6 +46,7 @@ class Scheduler:
         self.kv = kv
         self.waiting: deque[Sequence] = deque()
         self.running: list[Sequence] = []
+        self.swapped: deque[Sequence] = deque()  # offloaded to host DRAM
         self.proposer = None
         self.spec_k = 0
         spec = getattr(cfg, "speculative", None)
@@ -75,21 +76,40 @@ class Scheduler:
                 s.finish("abort")
                 self.waiting.remove(s)
                 return True
+        for s in list(self.swapped):
+            if s.request_id == request_id:
+                if s.host_block_table:
+                    self.kv.host_allocator.free(s.host_block_table)
+                    s.host_block_table = []
+                s.finish("abort")
+                self.swapped.remove(s)
+                return True
         return False
 
     @property
     def num_unfinished(self) -> int:
-        return len(self.waiting) + len(self.running)
+        return len(self.waiting) + len(self.running) + len(self.swapped)
 
     def has_work(self) -> bool:
         return self.num_unfinished > 0
 
     # -- scheduling --------------------------------------------------------
     def schedule(self) -> ScheduledBatch | None:
+        self._swap_in_ready()
         batch = self._schedule_prefill()
         if batch is not None:
             return batch
         return self._schedule_decode()
+
+    def _swap_in_ready(self) -> None:
+        while (self.swapped
+               and len(self.running) < self.cfg.max_num_seqs
+               and len(self.swapped[0].host_block_table) <= self.kv.allocator.num_free):
+            seq = self.swapped.popleft()
+            seq.block_table = self.kv.swap_in(seq.host_block_table)
+            seq.host_block_table = []
+            seq.status = SeqStatus.RUNNING
+            self.running.append(seq)
 
     def _schedule_prefill(self) -> ScheduledBatch | None:
         if not self.waiting:
@@ -138,10 +158,19 @@ class Scheduler:
                     seq.block_table.extend(self.kv.allocator.allocate(need))
                 except RuntimeError:
                     victim = self.running.pop()  # preempt newest
-                    self._release(victim)
-                    victim.status = SeqStatus.WAITING
-                    victim.preemptions += 1
-                    self.waiting.appendleft(victim)
+                    if self.kv.can_swap_out(len(victim.block_table)):
+                        # offload tier: swap KV to pinned host DRAM instead
+                        # of recompute
+                        victim.host_block_table = self.kv.swap_out(victim.block_table)
+                        victim.block_table = []
+                        victim.status = SeqStatus.WAITING
+                        victim.swap_outs += 1
+                        self.swapped.append(victim)
+                    else:
+                        self._release(victim)
+                        victim.status = SeqStatus.WAITING
+                        victim.preemptions += 1
+                        self.waiting.appendleft(victim)
                     if victim is seq:
                         continue
                     i = min(i, len(self.running))

@@ -35,7 +35,9 @@ class Sequence:
     output_token_ids: list[int] = field(default_factory=list)
     status: SeqStatus = SeqStatus.WAITING
     block_table: list[int] = field(default_factory=list)
+    host_block_table: list[int] = field(default_factory=list)  # offload tier
     num_cached_tokens: int = 0      # tokens whose KV is already in the pool
+    swap_outs: int = 0
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None
     finish_time: float | None = None
