@@ -5,6 +5,7 @@ from pydantic import BaseModel, Field
 
 from .tables import (  # noqa: F401
     ApiKey,
+    Benchmark,
     Model,
     ModelFile,
     ModelInstance,
@@ -127,3 +128,13 @@ class ModelInstanceUpdate(BaseModel):
 class ModelRouteCreate(BaseModel):
     name: str
     targets: list[dict] = Field(default_factory=list)
+
+
+class BenchmarkCreate(BaseModel):
+    name: str
+    model_name: str
+    mode: str = "concurrency"      # "concurrency" | "qps"
+    value: float = 8
+    duration_s: float = 30.0
+    isl: int = 128
+    osl: int = 64

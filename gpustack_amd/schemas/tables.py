@@ -177,6 +177,20 @@ class ModelRoute(Base, TimestampMixin, SerializeMixin):
     targets = Column(JSON, default=list)  # [{model_name, weight}]
 
 
+class Benchmark(Base, TimestampMixin, SerializeMixin):
+    """In-product benchmark runs (reference: schemas/benchmark.py)."""
+    __tablename__ = "benchmarks"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), nullable=False)
+    model_name = Column(String(256), nullable=False)
+    worker_id = Column(Integer, nullable=True)
+    state = Column(String(32), default="pending")  # pending/running/completed/error
+    state_message = Column(Text, default="")
+    # {mode: qps|concurrency, value, duration_s, isl, osl}
+    config = Column(JSON, default=dict)
+    results = Column(JSON, default=None)
+
+
 class ModelUsage(Base, TimestampMixin, SerializeMixin):
     __tablename__ = "model_usage"
     id = Column(Integer, primary_key=True)

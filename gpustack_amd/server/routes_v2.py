@@ -15,10 +15,10 @@ from fastapi.responses import StreamingResponse
 
 from ..db import Event, EventType, ar_create, ar_delete, ar_update, bus, get_session
 from ..schemas import (
-    ApiKey, ApiKeyCreate, Model, ModelCreate, ModelInstance, ModelInstanceState,
-    ModelInstanceUpdate, ModelRoute, ModelRouteCreate, ModelUpdate, ModelUsage,
-    RegistrationToken, SystemLoad, User, UserCreate, Worker, WorkerRegister,
-    WorkerState, WorkerStatusUpdate,
+    ApiKey, ApiKeyCreate, Benchmark, BenchmarkCreate, Model, ModelCreate,
+    ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelRoute,
+    ModelRouteCreate, ModelUpdate, ModelUsage, RegistrationToken, SystemLoad,
+    User, UserCreate, Worker, WorkerRegister, WorkerState, WorkerStatusUpdate,
 )
 from ..security import generate_api_key, generate_registration_token, hash_password
 from .deps import get_admin_user, get_current_user, verify_worker_token
@@ -323,6 +323,62 @@ def delete_route(route_id: int, _: User = Depends(get_current_user)):
         if not r:
             raise HTTPException(404)
         ar_delete(s, r)
+        return {"ok": True}
+
+
+# ---- benchmarks ------------------------------------------------------------
+
+@router.get("/benchmarks")
+def list_benchmarks(watch: bool = Query(False), user: User = Depends(get_current_user)):
+    with get_session() as s:
+        rows = [b.to_dict() for b in s.query(Benchmark).all()]
+    if watch:
+        return _watch_stream("benchmarks", rows, None)
+    return {"items": rows}
+
+
+@router.post("/benchmarks", status_code=201)
+def create_benchmark(body: BenchmarkCreate, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        model = s.query(Model).filter_by(name=body.model_name).first()
+        if not model:
+            raise HTTPException(404, f"model {body.model_name!r} not found")
+        inst = (
+            s.query(ModelInstance)
+            .filter_by(model_id=model.id, state=ModelInstanceState.RUNNING.value)
+            .first()
+        )
+        if not inst:
+            raise HTTPException(409, "model has no running instance to benchmark")
+        b = Benchmark(
+            name=body.name, model_name=body.model_name, worker_id=inst.worker_id,
+            config={"mode": body.mode, "value": body.value,
+                    "duration_s": body.duration_s, "isl": body.isl, "osl": body.osl},
+        )
+        ar_create(s, b)
+        return b.to_dict()
+
+
+@router.patch("/benchmarks/{bench_id}")
+def update_benchmark(bench_id: int, body: dict, _=Depends(verify_worker_token)):
+    with get_session() as s:
+        b = s.get(Benchmark, bench_id)
+        if not b:
+            raise HTTPException(404)
+        for k in ("state", "state_message", "results"):
+            if k in body:
+                setattr(b, k, body[k])
+        ar_update(s, b)
+        return b.to_dict()
+
+
+@router.delete("/benchmarks/{bench_id}")
+def delete_benchmark(bench_id: int, _: User = Depends(get_current_user)):
+    with get_session() as s:
+        b = s.get(Benchmark, bench_id)
+        if not b:
+            raise HTTPException(404)
+        ar_delete(s, b)
         return {"ok": True}
 
 

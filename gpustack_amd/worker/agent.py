@@ -135,11 +135,17 @@ class WorkerAgent:
     def start(self) -> None:
         self.register()
         self.serve_manager = ServeManager(self.cfg, self.client, self.worker_id)
+        from .benchmark_manager import BenchmarkManager
+
+        self.benchmark_manager = BenchmarkManager(
+            self.cfg, self.client, self.worker_id, self.serve_manager
+        )
         threads = [
             threading.Thread(target=self.heartbeat_loop, name="heartbeat", daemon=True),
             threading.Thread(target=self.status_loop, name="status", daemon=True),
             threading.Thread(target=self.serve_manager.watch_loop, name="serve-watch", daemon=True),
             threading.Thread(target=self.serve_manager.health_loop, name="serve-health", daemon=True),
+            threading.Thread(target=self.benchmark_manager.poll_loop, name="benchmarks", daemon=True),
         ]
         for t in threads:
             t.start()
@@ -151,6 +157,8 @@ class WorkerAgent:
 
     def stop(self) -> None:
         self._stop = True
+        if getattr(self, "benchmark_manager", None):
+            self.benchmark_manager.stop()
         if self.serve_manager:
             self.serve_manager.stop()
 

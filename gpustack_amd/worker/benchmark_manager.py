@@ -1,0 +1,95 @@
+"""Worker-side benchmark execution (reference: gpustack/worker/benchmark_manager.py:89).
+
+Watches Benchmark rows assigned to this worker, drives the first-party
+load generator (gpustack_amd.bench.loadgen) against the local RUNNING
+instance of the target model, and posts the TTFT/TPOT/TPS aggregates back."""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+
+import httpx
+
+from ..bench.loadgen import LoadSpec, run_load
+from ..client import ServerClient
+from ..config import Config
+
+logger = logging.getLogger(__name__)
+
+
+class BenchmarkManager:
+    def __init__(self, cfg: Config, client: ServerClient, worker_id: int,
+                 serve_manager):
+        self.cfg = cfg
+        self.client = client
+        self.worker_id = worker_id
+        self.serve_manager = serve_manager
+        self._stop = False
+        self._running: set[int] = set()
+
+    def stop(self):
+        self._stop = True
+
+    def poll_loop(self) -> None:
+        while not self._stop:
+            try:
+                r = self.client._c.get("/v2/benchmarks")
+                r.raise_for_status()
+                for b in r.json()["items"]:
+                    if (b.get("worker_id") == self.worker_id
+                            and b.get("state") == "pending"
+                            and b["id"] not in self._running):
+                        self._running.add(b["id"])
+                        threading.Thread(target=self._run_one, args=(b,),
+                                         daemon=True).start()
+            except httpx.HTTPError as e:
+                logger.warning("benchmark poll failed: %s", e)
+            time.sleep(3.0)
+
+    def _find_instance_port(self, model_name: str) -> int | None:
+        for ip in self.serve_manager.processes.values():
+            if ip.instance.get("model_name") == model_name and ip.healthy:
+                return ip.port
+        # fall back: ask the server
+        try:
+            insts = self.client.list_instances(worker_id=self.worker_id)
+            for i in insts:
+                if i.get("model_name") == model_name and i.get("state") == "running":
+                    return i.get("port")
+        except httpx.HTTPError:
+            pass
+        return None
+
+    def _run_one(self, b: dict) -> None:
+        bid = b["id"]
+        try:
+            port = self._find_instance_port(b["model_name"])
+            if port is None:
+                raise RuntimeError("no local running instance for model")
+            self._update(bid, state="running")
+            cfgd = b.get("config") or {}
+            spec = LoadSpec(
+                mode=cfgd.get("mode", "concurrency"),
+                value=float(cfgd.get("value", 8)),
+                duration_s=float(cfgd.get("duration_s", 30)),
+                isl=int(cfgd.get("isl", 128)),
+                osl=int(cfgd.get("osl", 64)),
+                model=b["model_name"],
+            )
+            results = run_load(f"http://127.0.0.1:{port}", spec)
+            self._update(bid, state="completed", results=results)
+            logger.info("benchmark %s done: %s out-tok/s, p50 TTFT %s ms",
+                        b.get("name"), results.get("output_tps"),
+                        results.get("ttft_p50_ms"))
+        except Exception as e:  # noqa: BLE001
+            logger.exception("benchmark %s failed", bid)
+            self._update(bid, state="error", state_message=str(e))
+        finally:
+            self._running.discard(bid)
+
+    def _update(self, bid: int, **fields) -> None:
+        try:
+            self.client._c.patch(f"/v2/benchmarks/{bid}", json=fields).raise_for_status()
+        except httpx.HTTPError as e:
+            logger.warning("benchmark update failed: %s", e)

@@ -159,3 +159,41 @@ def test_deploy_and_chat(cluster):
             break
         time.sleep(0.5)
     assert not agent.serve_manager.processes
+
+
+@pytest.mark.timeout(180)
+def test_benchmark_flow(cluster):
+    """In-product benchmark subsystem: fixed-concurrency load against the
+    deployed instance, aggregated TTFT/TPOT/TPS posted back."""
+    client, agent = cluster
+    mid = client.get("/v2/models").json()["items"][0]["id"]
+    client.patch(f"/v2/models/{mid}", json={"replicas": 1})
+    for _ in range(240):
+        insts = client.get("/v2/model_instances").json()["items"]
+        if insts and insts[0]["state"] == "running":
+            break
+        time.sleep(0.5)
+    assert insts and insts[0]["state"] == "running"
+
+    r = client.post("/v2/benchmarks", json={
+        "name": "b1", "model_name": "tiny-chat", "mode": "concurrency",
+        "value": 2, "duration_s": 3.0, "isl": 32, "osl": 8,
+    })
+    assert r.status_code == 201, r.text
+    bid = r.json()["id"]
+    state = None
+    for _ in range(120):
+        b = [x for x in client.get("/v2/benchmarks").json()["items"] if x["id"] == bid][0]
+        state = b["state"]
+        if state in ("completed", "error"):
+            break
+        time.sleep(1.0)
+    assert state == "completed", b.get("state_message")
+    res = b["results"]
+    assert res["successful_requests"] > 0
+    assert res["output_tps"] > 0
+    assert res["ttft_p50_ms"] is not None
+    # benchmark against a model with no instance is rejected
+    r = client.post("/v2/benchmarks", json={
+        "name": "b2", "model_name": "missing", "duration_s": 1})
+    assert r.status_code == 404
