@@ -34,6 +34,9 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   constexpr int LPG = 16;           // lanes per token-group
   constexpr int DV = D / LPG;       // dims per lane (8 for D=128)
   static_assert(DV == 8, "decode kernel assumes D = 128");
+  // tokens batched per softmax update: trade VALU savings against VGPR
+  // pressure (GQ>=4 would spill at TB=4)
+  constexpr int TB = (GQ <= 2) ? (BS / 4) : ((GQ <= 5) ? 2 : 1);
 
   const int seq = blockIdx.x;
   const int h = blockIdx.y;        // kv head
@@ -46,7 +49,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   const int g = lane >> 4;         // token-group within wave [0,4)
   const int sub = lane & 15;       // dim-slice within group [0,16)
 
-  // Load q rows for this kv head's GQ query heads (f32 registers).
+  // Load q rows for this kv head's GQ query heads (f32, pre-scaled so the
+  // per-token dot needs no multiply).
   float qr[GQ][DV];
 #pragma unroll
   for (int gq = 0; gq < GQ; ++gq) {
@@ -54,6 +58,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         q + (long)seq * q_stride + ((long)h * GQ + gq) * D + sub * DV;
     u16x8 u = *reinterpret_cast<const u16x8*>(qp);
     bf8_to_f32(u, qr[gq]);
+#pragma unroll
+    for (int j = 0; j < DV; ++j) qr[gq][j] *= scale;
   }
 
   float m[GQ], s[GQ], acc[GQ][DV];
@@ -70,32 +76,61 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     const long blk = bt[page];
     const unsigned short* kbase = kc + ((blk * Hkv + h) * BS) * D;
     const unsigned short* vbase = vc + ((blk * Hkv + h) * BS) * D;
+    // Whole page per wave-iteration: the group's 4 tokens are processed
+    // together — 4 independent dots (ILP across the shuffle-reduce
+    // chains), then ONE softmax update per gq for all 4 (cuts the
+    // exp/rescale VALU work ~2.5x vs per-token online updates).
 #pragma unroll
-    for (int it = 0; it < BS / 4; ++it) {
-      const int tok = it * 4 + g;
-      const bool valid = page * BS + tok < len;
-      // Coalesced: the wave's 64 lanes cover 4 tokens x 128 dims = 1 KiB.
-      u16x8 ku = *reinterpret_cast<const u16x8*>(kbase + tok * D + sub * DV);
-      u16x8 vu = *reinterpret_cast<const u16x8*>(vbase + tok * D + sub * DV);
-      float kf[DV], vf[DV];
-      bf8_to_f32(ku, kf);
-      bf8_to_f32(vu, vf);
+    for (int tb = 0; tb < BS / 4; tb += TB) {
+      const int base_tok = page * BS + tb * 4 + g;
+      u16x8 vu[TB];
+      float kf[TB][DV];  // K converted once, reused across all GQ heads
+#pragma unroll
+      for (int it = 0; it < TB; ++it) {
+        const int tok = (tb + it) * 4 + g;
+        u16x8 ku = *reinterpret_cast<const u16x8*>(kbase + tok * D + sub * DV);
+        vu[it] = *reinterpret_cast<const u16x8*>(vbase + tok * D + sub * DV);
+        bf8_to_f32(ku, kf[it]);
+      }
 #pragma unroll
       for (int gq = 0; gq < GQ; ++gq) {
-        float d = 0.f;
+        float dot[TB];
 #pragma unroll
-        for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[j];
-        d = group16_reduce_sum(d) * scale;   // all 16 lanes get the full dot
-        if (valid) {
-          const float nm = fmaxf(m[gq], d);
-          const float corr = __expf(m[gq] - nm);  // exp(-inf - finite) = 0
-          const float p = __expf(d - nm);
-          s[gq] = s[gq] * corr + p;
+        for (int it = 0; it < TB; ++it) {
+          float d = 0.f;
 #pragma unroll
-          for (int j = 0; j < DV; ++j)
-            acc[gq][j] = acc[gq][j] * corr + p * vf[j];
-          m[gq] = nm;
+          for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[it][j];
+          dot[it] = d;
         }
+#pragma unroll
+        for (int it = 0; it < TB; ++it)
+          dot[it] = group16_reduce_sum(dot[it]);  // all 16 lanes get the dot
+        float pmax = -INFINITY;
+#pragma unroll
+        for (int it = 0; it < TB; ++it) {
+          const bool valid = base_tok + it * 4 < len;
+          dot[it] = valid ? dot[it] : -INFINITY;
+          pmax = fmaxf(pmax, dot[it]);
+        }
+        if (pmax == -INFINITY) continue;
+        const float nm = fmaxf(m[gq], pmax);
+        const float corr = __expf(m[gq] - nm);  // exp(-inf - finite) = 0
+        float p[TB];
+        float psum = 0.f;
+#pragma unroll
+        for (int it = 0; it < TB; ++it) {
+          p[it] = (dot[it] == -INFINITY) ? 0.f : __expf(dot[it] - nm);
+          psum += p[it];
+        }
+        s[gq] = s[gq] * corr + psum;
+#pragma unroll
+        for (int j = 0; j < DV; ++j) {
+          float a = acc[gq][j] * corr;
+#pragma unroll
+          for (int it = 0; it < TB; ++it) a += p[it] * bf2f(vu[it][j]);
+          acc[gq][j] = a;
+        }
+        m[gq] = nm;
       }
     }
   }
