@@ -31,7 +31,7 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=192)
     ap.add_argument("--warmup", type=int, default=48)
     ap.add_argument("--model", default="llama-3-8b")
-    ap.add_argument("--concurrency", type=int, default=128, help="target in-flight requests per GPU")
+    ap.add_argument("--concurrency", type=int, default=512, help="target in-flight requests per GPU")
     ap.add_argument("--isl", type=int, default=256, help="synthetic prompt length")
     ap.add_argument("--osl", type=int, default=128, help="max output tokens per request")
     ap.add_argument("--max-model-len", type=int, default=4096)
