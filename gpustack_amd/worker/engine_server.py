@@ -64,10 +64,10 @@ def load_tokenizer(model_dir: str | None, vocab_size: int):
 class EngineRunner:
     """Owns the LLMEngine + step loop thread; bridges to asyncio."""
 
-    def __init__(self, engine_cfg, served_name: str):
+    def __init__(self, engine_cfg, served_name: str, comm=None):
         from ..engine import LLMEngine
 
-        self.engine = LLMEngine(engine_cfg)
+        self.engine = LLMEngine(engine_cfg, comm)
         self.served_name = served_name
         self.tokenizer = load_tokenizer(engine_cfg.model_dir, engine_cfg.spec.vocab_size)
         self.loop: asyncio.AbstractEventLoop | None = None
@@ -87,8 +87,16 @@ class EngineRunner:
         self._wake.set()
 
     def _run(self) -> None:
+        tp = self.engine.comm.tp_size > 1
         while not self._stop:
-            if not self.engine.has_unfinished():
+            if tp:
+                # coordinated loop: the tp_active broadcast doubles as the
+                # idle heartbeat for follower ranks
+                if not self.engine.tp_active():
+                    self._wake.wait(timeout=0.02)
+                    self._wake.clear()
+                    continue
+            elif not self.engine.has_unfinished():
                 self._wake.wait(timeout=0.05)
                 self._wake.clear()
                 continue
@@ -327,6 +335,21 @@ def create_app(runner: EngineRunner) -> FastAPI:
     return app
 
 
+def _spawn_followers(argv_base: list[str], tp: int, master_port: int):
+    """Rank 0 spawns follower rank processes (the first-party replacement
+    for the reference's multi-process executor bootstrap —
+    serve_manager.py:1685-1737 port bands + vLLM --headless followers)."""
+    import subprocess
+    import sys
+
+    procs = []
+    for r in range(1, tp):
+        cmd = [sys.executable, "-m", "gpustack_amd.worker.engine_server",
+               *argv_base, "--tp-rank", str(r), "--master-port", str(master_port)]
+        procs.append(subprocess.Popen(cmd))
+    return procs
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--served-name", required=True)
@@ -340,6 +363,9 @@ def main():
     ap.add_argument("--device", default=None)
     ap.add_argument("--kv-cache-blocks", type=int, default=None)
     ap.add_argument("--backend-parameters", default="{}")
+    ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--tp-rank", type=int, default=0)
+    ap.add_argument("--master-port", type=int, default=None)
     args = ap.parse_args()
 
     logging.basicConfig(level=logging.INFO,
@@ -348,7 +374,40 @@ def main():
 
     from ..engine import EngineConfig
 
-    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    use_cuda = torch.cuda.is_available()
+    device = args.device or ("cuda" if use_cuda else "cpu")
+    comm = None
+    followers = []
+    if args.tp > 1:
+        from ..parallel import init_tp
+
+        if args.tp_rank == 0 and args.master_port is None:
+            import socket
+
+            s = socket.socket()
+            s.bind(("127.0.0.1", 0))
+            args.master_port = s.getsockname()[1]
+            s.close()
+            argv_base = [
+                "--served-name", args.served_name, "--source", args.source,
+                "--model-ref", args.model_ref, "--port", str(args.port),
+                "--max-model-len", str(args.max_model_len),
+                "--max-num-seqs", str(args.max_num_seqs),
+                "--gpu-memory-utilization", str(args.gpu_memory_utilization),
+                "--backend-parameters", args.backend_parameters,
+                "--tp", str(args.tp),
+            ]
+            if args.kv_cache_blocks:
+                argv_base += ["--kv-cache-blocks", str(args.kv_cache_blocks)]
+            if args.device:
+                argv_base += ["--device", args.device]
+            followers = _spawn_followers(argv_base, args.tp, args.master_port)
+        if use_cuda:
+            device = f"cuda:{args.tp_rank}"
+            torch.cuda.set_device(args.tp_rank)
+        comm = init_tp(args.tp, args.tp_rank, master_port=args.master_port,
+                       device_id=args.tp_rank if use_cuda else None)
+
     extra = json.loads(args.backend_parameters)
     cfg_kwargs = dict(
         model=args.model_ref,
@@ -357,6 +416,8 @@ def main():
         max_num_seqs=args.max_num_seqs,
         gpu_memory_utilization=args.gpu_memory_utilization,
         kv_cache_blocks=args.kv_cache_blocks,
+        tp_size=args.tp,
+        tp_rank=args.tp_rank,
     )
     if device == "cpu" and args.kv_cache_blocks is None:
         cfg_kwargs["kv_cache_blocks"] = 1024
@@ -365,11 +426,30 @@ def main():
     if args.source == "local_path":
         ecfg.model_dir = args.model_ref
         ecfg.enforce_random_weights = False
-    runner = EngineRunner(ecfg, args.served_name)
+
+    if args.tp > 1 and args.tp_rank > 0:
+        # follower rank: no HTTP; run the coordinated engine loop forever
+        import time as _time
+
+        from ..engine import LLMEngine
+
+        eng = LLMEngine(ecfg, comm)
+        logger.info("TP follower rank %d ready", args.tp_rank)
+        while True:
+            if eng.tp_active():
+                eng.step()
+            else:
+                _time.sleep(0.02)
+
+    runner = EngineRunner(ecfg, args.served_name, comm)
     app = create_app(runner)
     import uvicorn
 
-    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    try:
+        uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    finally:
+        for p in followers:
+            p.terminate()
 
 
 if __name__ == "__main__":

@@ -163,6 +163,9 @@ class ServeManager:
         ]
         if model.get("max_model_len"):
             args += ["--max-model-len", str(model["max_model_len"])]
+        if len(gpus) > 1:
+            # TP replica sharded over the scheduled GPUs (RCCL over xGMI)
+            args += ["--tp", str(len(gpus))]
         if bp:
             args += ["--backend-parameters", json.dumps(bp)]
 

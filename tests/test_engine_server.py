@@ -1,0 +1,64 @@
+"""Engine server process tests (OpenAI surface + TP spawn) on CPU."""
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _wait_health(port: int, proc, timeout=90) -> None:
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if proc.poll() is not None:
+            raise AssertionError(f"engine server exited {proc.returncode}")
+        try:
+            r = httpx.get(f"http://127.0.0.1:{port}/health", timeout=2)
+            if r.status_code == 200:
+                return
+        except httpx.HTTPError:
+            pass
+        time.sleep(0.5)
+    raise AssertionError("engine server never became healthy")
+
+
+@pytest.mark.timeout(240)
+def test_engine_server_tp2_cpu():
+    """gpus_per_replica=2 path: rank 0 spawns a follower, both step in
+    lockstep over gloo, OpenAI endpoint answers."""
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-tp", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256", "--tp", "2",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        _wait_health(port, proc)
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-tp", "prompt": "ab", "max_tokens": 6,
+            "ignore_eos": True, "temperature": 0,
+        }, timeout=60)
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 6
+        # second request exercises the idle->active->idle transition
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-tp", "prompt": "xyz", "max_tokens": 4,
+            "ignore_eos": True, "temperature": 0,
+        }, timeout=60)
+        assert r.status_code == 200
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
