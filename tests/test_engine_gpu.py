@@ -24,12 +24,14 @@ def _cfg(**kw):
 PROMPTS = [[11, 12, 13, 14, 15, 16] * 8, [101, 102, 103] * 5, [7] * 33]
 
 
-def test_decode_matches_prefill_gpu():
-    eng = LLMEngine(_cfg())
+@pytest.mark.parametrize("model", ["llama-3-8b", "qwen3-14b", "qwen2.5-7b"])
+def test_decode_matches_prefill_gpu(model):
+    # covers GQA ratios 4/5/7, qwen2 attention bias, qwen3 per-head qk-norm
+    eng = LLMEngine(_cfg(model=model))
     full = eng.generate(PROMPTS[:1], SamplingParams(max_tokens=8, ignore_eos=True))[0]
     del eng
     torch.cuda.empty_cache()
-    eng2 = LLMEngine(_cfg())
+    eng2 = LLMEngine(_cfg(model=model))
     cont = eng2.generate([PROMPTS[0] + full[:4]],
                          SamplingParams(max_tokens=4, ignore_eos=True))[0]
     assert cont == full[4:], f"{cont} != {full[4:]}"
