@@ -38,6 +38,10 @@ def main(argv: list[str] | None = None) -> int:
     cp.add_argument("--api-key", default=None)
     cp.add_argument("--model", required=True)
     cp.add_argument("--prompt", required=True)
+    rp = sub.add_parser("reset-admin-password")
+    rp.add_argument("--data-dir", dest="data_dir", default=None)
+    rp.add_argument("--database-url", dest="database_url", default=None)
+    rp.add_argument("--password", required=True)
     sub.add_parser("version")
     args = ap.parse_args(argv)
 
@@ -45,6 +49,24 @@ def main(argv: list[str] | None = None) -> int:
         from . import __version__
 
         print(__version__)
+        return 0
+    if args.cmd == "reset-admin-password":
+        from .config import load_config
+        from .db import get_session, init_db
+        from .schemas import User
+        from .security import hash_password
+
+        cfg = load_config(None, {"data_dir": args.data_dir,
+                                 "database_url": args.database_url})
+        init_db(cfg.resolved_database_url())
+        with get_session() as s:
+            admin = s.query(User).filter_by(username="admin").first()
+            if admin is None:
+                admin = User(username="admin", hashed_password="", is_admin=True)
+                s.add(admin)
+            admin.hashed_password = hash_password(args.password)
+            s.commit()
+        print("admin password reset")
         return 0
     if args.cmd == "chat":
         import httpx

@@ -65,6 +65,7 @@ class ModelCreate(BaseModel):
     extended_kv_cache: dict | None = None
     distributed_inference_across_workers: bool = False
     restart_on_error: bool = True
+    scaling_schedule: dict | None = None
 
 
 class ModelUpdate(BaseModel):

@@ -67,12 +67,16 @@ class ModelController:
                 logger.exception("sync_replicas(%s) failed", mid)
 
     def sync_replicas(self, model_id: int) -> None:
+        self.sync_replicas_to(model_id, None)
+
+    def sync_replicas_to(self, model_id: int, want_override: int | None) -> None:
         with get_session() as s:
             model = s.get(Model, model_id)
             if model is None:
                 return
             insts = s.query(ModelInstance).filter_by(model_id=model_id).all()
-            want, have = model.replicas, len(insts)
+            want = model.replicas if want_override is None else want_override
+            have = len(insts)
             if have < want:
                 used = {i.name for i in insts}
                 for n in range(want * 2):
@@ -141,6 +145,53 @@ class WorkerMonitor:
                 inst.state = ModelInstanceState.UNREACHABLE.value
                 inst.state_message = "worker unreachable"
                 ar_update(s, inst)
+
+
+class ScalingScheduler:
+    """Cron-window desired-replica computation (reference:
+    server/scaling_scheduler.py:19,96). While a window is active the
+    model's instance count follows the rule's replicas; outside windows
+    it returns to the baseline Model.replicas."""
+
+    def __init__(self, cfg: Config, interval: float = 30.0):
+        self.cfg = cfg
+        self.interval = interval
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
+
+    def desired_replicas(self, model) -> int:
+        from ..utils.cron import window_active
+
+        sched = model.scaling_schedule or {}
+        desired = model.replicas
+        for rule in sched.get("rules", []):
+            try:
+                if window_active(rule["cron"], int(rule.get("duration_minutes", 60))):
+                    desired = max(desired, int(rule.get("replicas", desired)))
+            except (KeyError, ValueError):
+                logger.warning("bad scaling rule on model %s: %s", model.name, rule)
+        return desired
+
+    def run(self) -> None:
+        mc = ModelController(self.cfg)
+        while not self._stop:
+            try:
+                with get_session() as s:
+                    models = [m for m in s.query(Model).all() if m.scaling_schedule]
+                for m in models:
+                    want = self.desired_replicas(m)
+                    with get_session() as s:
+                        have = s.query(ModelInstance).filter_by(model_id=m.id).count()
+                    if have != want:
+                        mc.sync_replicas_to(m.id, want)
+            except Exception:  # noqa: BLE001
+                logger.exception("scaling scheduler cycle failed")
+            for _ in range(int(self.interval)):
+                if self._stop:
+                    return
+                time.sleep(1.0)
 
 
 class SystemLoadCollector:

@@ -326,6 +326,42 @@ def delete_route(route_id: int, _: User = Depends(get_current_user)):
         return {"ok": True}
 
 
+# ---- model evaluations (compatibility pre-check) ---------------------------
+
+@router.post("/model-evaluations")
+def evaluate_model(body: ModelCreate, _: User = Depends(get_current_user)):
+    """Dry-run of the placement pipeline (reference: scheduler/evaluator.py):
+    reports whether the model would fit, on which worker, and why not."""
+    from ..scheduler.policies import (
+        estimate_vram_claim, model_spec_for, pick_candidate,
+    )
+
+    model_d = body.model_dump() | {"id": -1}
+    spec = model_spec_for(model_d)
+    if spec is None:
+        return {"compatible": False,
+                "messages": [f"cannot resolve model spec for {body.model_ref!r}"]}
+    tp = max(1, body.gpus_per_replica)
+    claim = estimate_vram_claim(model_d, spec, tp)
+    with get_session() as s:
+        workers = [w.to_dict() for w in s.query(Worker).all()]
+        insts = [i.to_dict() for i in s.query(ModelInstance).all()]
+    cand = pick_candidate(model_d, workers, insts)
+    if cand is None:
+        return {
+            "compatible": False,
+            "estimated_vram_per_gpu": claim,
+            "messages": ["no worker currently fits the resource claim"],
+        }
+    return {
+        "compatible": True,
+        "estimated_vram_per_gpu": claim,
+        "candidate": {"worker": cand.worker.get("name"),
+                      "gpu_indexes": cand.gpu_indexes},
+        "messages": [],
+    }
+
+
 # ---- benchmarks ------------------------------------------------------------
 
 @router.get("/benchmarks")
