@@ -1,4 +1,4 @@
-// Varlen causal flash-attention prefill for MI355X (gfx950) on MFMA.
+// Varlen causal flash-attention prefill for MI355X (gfx950) on MFMA — v2.
 //
 // Replaces the prefill attention the reference gets from vLLM/SGLang
 // (SURVEY.md §2.9 #1). CDNA4-first structure (not a CUDA port):
@@ -8,30 +8,30 @@
 //      B-frag: lane holds B[k = (l>>4)*8 + j][col = l&15]   (u16x8)
 //      C/D  : lane holds D[row = (l>>4)*4 + r][col = l&15]  (f32x4)
 //  - swapped QK^T (S^T = K · Q^T) so BOTH operands are row-contiguous
-//    u16x8 reads (K from LDS, Q from registers) — the HipKittens-style
-//    trick described in the CDNA4 guide (§B fused attention).
-//  - K tile LDS-staged with the ((row&7)<<4) byte-XOR swizzle (guide §6 G4:
-//    row-major [32][128] bf16 would be a 16-way bank conflict on b128 reads).
-//  - V tile staged TRANSPOSED (VT[128][32+4]) so PV B-fragments are
-//    contiguous b128 reads; P goes through a small per-wave LDS buffer to
-//    re-distribute S^T's lane layout into PV's A-fragment layout.
-//  - online softmax entirely in f32 registers; stats per q-row shared
-//    across lanes via 64-wide shuffles.
+//    u16x8 reads (K from LDS, Q from registers).
+//  - K tile LDS-staged with the ((row&7)<<4) byte-XOR swizzle (guide §6 G4).
+//  - V staged TRANSPOSED (VT[128][BK+pad]) so PV B-fragments are b128 reads.
+//  - v2: BK = 64 kv per tile (twice the MFMA work per barrier pair) and
+//    async-stage split (guide T14): each iteration ISSUES the next tile's
+//    global loads into registers BEFORE the MFMA work, then writes them to
+//    LDS after the barrier — HBM latency hides under compute.
+//  - online softmax in f32 registers; per-q-row stats shared via shuffles.
 //
-// Tiling: BQ = 64 q rows per workgroup (4 waves x 16 rows), BK = 32 kv.
-// v0 computes attention over the prefill chunk's own contiguous K/V
-// (full-prompt prefill); chunked prefill against the paged pool reuses the
-// decode path.
+// Tiling: BQ = 64 q rows per workgroup (4 waves x 16 rows), BK = 64 kv.
 #include "common.h"
 
 namespace {
 
 constexpr int BQ = 64;
-constexpr int BK = 32;
+constexpr int BK = 64;
 constexpr int PF_D = 128;
 constexpr int PF_WAVES = 4;
 constexpr int PF_THREADS = PF_WAVES * WAVE_SIZE;
-constexpr int VT_PAD = 4;  // elements; breaks the 16-row bank cycle
+constexpr int VT_PAD = 4;  // elements; breaks the bank cycle on VT/P rows
+constexpr int NST = BK / 16;       // S^T stiles per tile (4)
+constexpr int KC2 = BK / 32;       // PV k-chunks per tile (2)
+constexpr int KU = BK * (PF_D / 8) / PF_THREADS;  // K u16x8 units/thread (4)
+constexpr int VU = PF_D * (BK / 8) / PF_THREADS;  // VT units/thread (4)
 
 typedef __attribute__((ext_vector_type(8))) short s16x8;
 
@@ -66,13 +66,14 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   const int wave = threadIdx.x / WAVE_SIZE;
   const int lc = lane & 15;        // column / row-id within 16
   const int lg = lane >> 4;        // 4-lane group id
+  const int tid = threadIdx.x;
 
   __shared__ unsigned short Kl[BK * PF_D];            // swizzled
   __shared__ unsigned short VTl[PF_D][BK + VT_PAD];   // transposed V
   __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
 
   // Hoist this wave's 16 q rows into B-fragments (4 k-chunks of 32).
-  const int qrow_local = q0 + wave * 16 + lc;   // this lane's B-frag q row
+  const int qrow_local = q0 + wave * 16 + lc;
   const int qrow_clamped = (qrow_local < len) ? qrow_local : (len - 1);
   u16x8 qfrag[4];
 #pragma unroll
@@ -88,54 +89,74 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 #pragma unroll
   for (int nt = 0; nt < PF_D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int q_hi = q0 + BQ - 1;                    // last q row of this block
+  const int q_hi = q0 + BQ - 1;
   const int kv_end = min(len, q_hi + 1);           // causal bound
   const int ntiles_kv = (kv_end + BK - 1) / BK;
   const int wave_q_hi = q0 + wave * 16 + 15;       // this wave's causal bound
 
-  for (int kt = 0; kt < ntiles_kv; ++kt) {
-    const int kv0 = kt * BK;
-    // ---- cooperative staging: K (swizzled) and V^T ----
-    __syncthreads();
-    {
-      // K: 512 u16x8 chunks over 2 rounds; unit -> (kv, d0)
-      for (int u = threadIdx.x; u < BK * (PF_D / 8); u += PF_THREADS) {
-        const int kv = u / (PF_D / 8);
-        const int d0 = (u % (PF_D / 8)) * 8;
-        u16x8 val{0, 0, 0, 0, 0, 0, 0, 0};
-        if (kv0 + kv < len) {
-          val = *reinterpret_cast<const u16x8*>(
-              k + (long)(seq0 + kv0 + kv) * ks + (long)kvh * PF_D + d0);
-        }
-        *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
-                                  kswz(kv, d0 * 2)) = val;
-      }
-      // V^T: unit -> (d, kv-chunk of 8); lane-contiguous d for coalescing
-      for (int u = threadIdx.x; u < PF_D * (BK / 8); u += PF_THREADS) {
-        const int d = u % PF_D;
-        const int kvc = (u / PF_D) * 8;
-        unsigned short tmp[8];
+  // staging assignment (fixed per thread):
+  //   K: unit u = tid + r*256 -> (kv = u/16, d0 = (u%16)*8), r in [0,KU)
+  //   VT: unit u -> (d = u%128, kvc = (u/128)*8)
+  const int kst_kv[KU] = {tid / 16, (tid + 256) / 16, (tid + 512) / 16,
+                          (tid + 768) / 16};
+  const int kst_d0 = (tid % 16) * 8;
+  const int vst_d = tid % PF_D;
+  const int vst_kvc0 = (tid / PF_D) * 8;   // + r*16 per round
+
+  u16x8 kstage[KU];
+  unsigned short vstage[VU][8];
+
+  auto issue_loads = [&](int kv0) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int kv = kv0 + kvc + j;
-          tmp[j] = (kv < len)
-                       ? v[(long)(seq0 + kv) * vs + (long)kvh * PF_D + d]
-                       : (unsigned short)0;
-        }
-        *reinterpret_cast<u16x8*>(&VTl[d][kvc]) =
-            *reinterpret_cast<u16x8*>(tmp);
+    for (int r = 0; r < KU; ++r) {
+      const int kv = kv0 + kst_kv[r];
+      kstage[r] = (kv < len)
+          ? *reinterpret_cast<const u16x8*>(
+                k + (long)(seq0 + kv) * ks + (long)kvh * PF_D + kst_d0)
+          : u16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+#pragma unroll
+    for (int r = 0; r < VU; ++r) {
+      const int kvc = vst_kvc0 + r * 16;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kv = kv0 + kvc + j;
+        vstage[r][j] = (kv < len)
+            ? v[(long)(seq0 + kv) * vs + (long)kvh * PF_D + vst_d]
+            : (unsigned short)0;
       }
     }
-    __syncthreads();
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int r = 0; r < KU; ++r) {
+      *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
+                                kswz(kst_kv[r], kst_d0 * 2)) = kstage[r];
+    }
+#pragma unroll
+    for (int r = 0; r < VU; ++r) {
+      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * 16]) =
+          *reinterpret_cast<u16x8*>(vstage[r]);
+    }
+  };
+
+  issue_loads(0);
+  for (int kt = 0; kt < ntiles_kv; ++kt) {
+    const int kv0 = kt * BK;
+    __syncthreads();   // previous tile's LDS reads are done
+    write_lds();
+    __syncthreads();   // tile ready
+    if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);  // hide HBM under MFMA
 
     if (kv0 > wave_q_hi) continue;  // fully masked for this wave
 
-    // ---- S^T = K · Q^T  (2 stiles x 4 k-chunks of MFMA) ----
-    f32x4 st[2];
-    st[0] = f32x4{0.f, 0.f, 0.f, 0.f};
-    st[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+    // ---- S^T = K · Q^T  (NST stiles x 4 k-chunks) ----
+    f32x4 st[NST];
 #pragma unroll
-    for (int stile = 0; stile < 2; ++stile) {
+    for (int i = 0; i < NST; ++i) st[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int stile = 0; stile < NST; ++stile) {
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const u16x8 a = *reinterpret_cast<const u16x8*>(
@@ -146,11 +167,11 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     }
 
     // ---- mask + online softmax (stats per q row lc) ----
-    const int qpos = q0 + wave * 16 + lc;  // q row this lane's stats cover
-    float sv[8];
+    const int qpos = q0 + wave * 16 + lc;
+    float sv[NST * 4];
     float tmax = -INFINITY;
 #pragma unroll
-    for (int stile = 0; stile < 2; ++stile) {
+    for (int stile = 0; stile < NST; ++stile) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
@@ -167,18 +188,18 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 
     const float nm = fmaxf(mcol, tmax);
     float corr = 1.f, tsum = 0.f;
-    float pv[8];
+    float pv[NST * 4];
     if (nm != -INFINITY) {
-      corr = __expf(mcol - nm);  // mcol = -inf -> 0
+      corr = __expf(mcol - nm);
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
+      for (int i = 0; i < NST * 4; ++i) {
         pv[i] = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - nm);
         tsum += pv[i];
       }
       mcol = nm;
     } else {
 #pragma unroll
-      for (int i = 0; i < 8; ++i) pv[i] = 0.f;
+      for (int i = 0; i < NST * 4; ++i) pv[i] = 0.f;
     }
 #pragma unroll
     for (int msk = 16; msk <= 32; msk <<= 1)
@@ -187,7 +208,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 
     // ---- stage P (bf16) into this wave's LDS buffer: P[q=lc][kv] ----
 #pragma unroll
-    for (int stile = 0; stile < 2; ++stile) {
+    for (int stile = 0; stile < NST; ++stile) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         Pl[wave][lc][stile * 16 + lg * 4 + r] = f2bf(pv[stile * 4 + r]);
@@ -205,13 +226,16 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     }
 
     // ---- O += P · V  (A = P from LDS, B = V^T rows from LDS) ----
-    const u16x8 pa =
-        *reinterpret_cast<const u16x8*>(&Pl[wave][lc][lg * 8]);
 #pragma unroll
-    for (int nt = 0; nt < PF_D / 16; ++nt) {
-      const u16x8 b =
-          *reinterpret_cast<const u16x8*>(&VTl[nt * 16 + lc][lg * 8]);
-      o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
+    for (int kk2 = 0; kk2 < KC2; ++kk2) {
+      const u16x8 pa = *reinterpret_cast<const u16x8*>(
+          &Pl[wave][lc][kk2 * 32 + lg * 8]);
+#pragma unroll
+      for (int nt = 0; nt < PF_D / 16; ++nt) {
+        const u16x8 b = *reinterpret_cast<const u16x8*>(
+            &VTl[nt * 16 + lc][kk2 * 32 + lg * 8]);
+        o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
+      }
     }
   }
 
