@@ -134,6 +134,8 @@ class EngineConfig:
     # speculative decoding (reference speculative_config schema):
     # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
     speculative: dict | None = None
+    # LoRA adapters merged into the weights at load (reference lora_list)
+    lora_dirs: list[str] = field(default_factory=list)
 
     spec: ModelSpec = field(default_factory=ModelSpec)
 

@@ -78,6 +78,14 @@ class ModelRunner:
             load_safetensors(self.model, cfg, cfg.model_dir)
         else:
             random_init(self.model, cfg)
+        if cfg.lora_dirs:
+            from ..models.weights import merge_lora
+
+            for d in cfg.lora_dirs:
+                n = merge_lora(self.model, cfg, d)
+                import logging
+
+                logging.getLogger(__name__).info("merged LoRA %s (%d tensors)", d, n)
         self.sampler = Sampler(self.device)
         self.kv: KVCache | None = None
 

@@ -141,3 +141,94 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
         layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.copy_(get(p + "input_layernorm.weight"))
         layer.post_attn_norm.copy_(get(p + "post_attention_layernorm.weight"))
+
+
+def merge_lora(model, cfg: EngineConfig, adapter_dir: str | Path) -> int:
+    """Merge a PEFT LoRA adapter into the serving weights at load time
+    (reference parity: Model.lora_list mounting, ref schemas/models.py:468;
+    merged adapters add zero runtime cost — dynamic multi-adapter serving
+    is a later round).
+
+    Supports q/k/v/o/gate/up/down projection adapters in the standard PEFT
+    naming (`base_model.model.model.layers.N.self_attn.q_proj.lora_A.weight`
+    etc). Returns the number of merged tensors.
+    """
+    import json as _json
+
+    from safetensors import safe_open
+
+    adapter_dir = Path(adapter_dir)
+    cfg_path = adapter_dir / "adapter_config.json"
+    alpha, r = 16.0, 8.0
+    if cfg_path.exists():
+        with open(cfg_path) as f:
+            ac = _json.load(f)
+        alpha = float(ac.get("lora_alpha", alpha))
+        r = float(ac.get("r", r))
+    scaling = alpha / r
+
+    files = sorted(adapter_dir.glob("*.safetensors"))
+    if not files:
+        raise FileNotFoundError(f"no adapter safetensors under {adapter_dir}")
+    tensors: dict[str, torch.Tensor] = {}
+    for f in files:
+        with safe_open(str(f), framework="pt") as sf:
+            for name in sf.keys():
+                tensors[name] = sf.get_tensor(name)
+
+    spec = model.spec
+    tp, rank = cfg.tp_size, cfg.tp_rank
+    d = spec.head_dim
+    hq, hkv = spec.num_heads // tp, max(1, spec.num_kv_heads // tp)
+    i_loc = spec.intermediate_size // tp
+
+    def find(li: int, proj: str, which: str) -> torch.Tensor | None:
+        for key, t in tensors.items():
+            if f"layers.{li}." in key and f"{proj}." in key and f"lora_{which}" in key:
+                return t.float()
+        return None
+
+    merged = 0
+    for li, layer in enumerate(model.layers):
+        # fused qkv: rows [0,hq*d) = q shard, then k, then v
+        offsets = {
+            "q_proj": (0, hq * d, rank * hq * d),
+            "k_proj": (hq * d, hkv * d, rank * hkv * d),
+            "v_proj": ((hq + hkv) * d, hkv * d, rank * hkv * d),
+        }
+        for proj, (dst_off, rows, src_off) in offsets.items():
+            A, B = find(li, proj, "A"), find(li, proj, "B")
+            if A is None or B is None:
+                continue
+            delta = (B @ A) * scaling              # [out_full, in]
+            shard = delta[src_off:src_off + rows]
+            w = layer.attn.qkv_w.data
+            w[dst_off:dst_off + rows] += shard.to(w.dtype).to(w.device)
+            merged += 1
+        A, B = find(li, "o_proj", "A"), find(li, "o_proj", "B")
+        if A is not None and B is not None:
+            delta = (B @ A) * scaling              # [h, hq_full*d]
+            shard = delta[:, rank * hq * d:(rank + 1) * hq * d]
+            w = layer.attn.o_w.data
+            w += shard.to(w.dtype).to(w.device)
+            merged += 1
+        for proj, dst_off, rows, src_off in (
+            ("gate_proj", 0, i_loc, rank * i_loc),
+            ("up_proj", i_loc, i_loc, rank * i_loc),
+        ):
+            A, B = find(li, proj, "A"), find(li, proj, "B")
+            if A is None or B is None:
+                continue
+            delta = (B @ A) * scaling
+            shard = delta[src_off:src_off + rows]
+            w = layer.mlp.gate_up_w.data
+            w[dst_off:dst_off + rows] += shard.to(w.dtype).to(w.device)
+            merged += 1
+        A, B = find(li, "down_proj", "A"), find(li, "down_proj", "B")
+        if A is not None and B is not None:
+            delta = (B @ A) * scaling              # [h, i_full]
+            shard = delta[:, rank * i_loc:(rank + 1) * i_loc]
+            w = layer.mlp.down_w.data
+            w += shard.to(w.dtype).to(w.device)
+            merged += 1
+    return merged

@@ -66,6 +66,7 @@ class ModelCreate(BaseModel):
     distributed_inference_across_workers: bool = False
     restart_on_error: bool = True
     scaling_schedule: dict | None = None
+    lora_list: list[str] | None = None
 
 
 class ModelUpdate(BaseModel):

@@ -134,6 +134,7 @@ class Model(Base, TimestampMixin, SerializeMixin):
     extended_kv_cache = Column(JSON, default=None)   # {ram_size/ram_ratio, ...}
     distributed_inference_across_workers = Column(Boolean, default=False)
     restart_on_error = Column(Boolean, default=True)
+    lora_list = Column(JSON, default=None)  # adapter dirs merged at load
     # cron-window autoscaling (reference: schemas/models.py:237-321):
     # {"rules": [{"cron": "0 9 * * 1-5", "duration_minutes": 60, "replicas": 4}]}
     scaling_schedule = Column(JSON, default=None)

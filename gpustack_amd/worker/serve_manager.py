@@ -153,6 +153,8 @@ class ServeManager:
         env.update(model.get("env") or {})
 
         bp = dict(model.get("backend_parameters") or {})
+        if model.get("lora_list"):
+            bp.setdefault("lora_dirs", model["lora_list"])
         args = [
             sys.executable, "-m", "gpustack_amd.worker.engine_server",
             "--served-name", model["name"],
