@@ -197,3 +197,52 @@ def test_benchmark_flow(cluster):
     r = client.post("/v2/benchmarks", json={
         "name": "b2", "model_name": "missing", "duration_s": 1})
     assert r.status_code == 404
+
+
+@pytest.mark.timeout(180)
+def test_crash_restart(cluster):
+    """Failure detection: killed engine -> ERROR -> automatic restart with
+    backoff -> RUNNING again (reference: serve_manager.py:1842-1885)."""
+    import os
+    import signal
+
+    from gpustack_amd.worker import serve_manager as sm_mod
+
+    client, agent = cluster
+    mid = client.get("/v2/models").json()["items"][0]["id"]
+    client.patch(f"/v2/models/{mid}", json={"replicas": 1})
+    for _ in range(240):
+        insts = client.get("/v2/model_instances").json()["items"]
+        if insts and insts[0]["state"] == "running":
+            break
+        time.sleep(0.5)
+    assert insts[0]["state"] == "running"
+
+    # shrink backoff so the test is fast
+    old_base = sm_mod.RESTART_BASE
+    sm_mod.RESTART_BASE = 0.5
+    try:
+        iid = insts[0]["id"]
+        ip = agent.serve_manager.processes[iid]
+        os.killpg(ip.proc.pid, signal.SIGKILL)
+        saw_error = False
+        state = None
+        for _ in range(240):
+            inst = [i for i in client.get("/v2/model_instances").json()["items"]
+                    if i["id"] == iid][0]
+            state = inst["state"]
+            if state == "error":
+                saw_error = True
+            if saw_error and state == "running":
+                break
+            time.sleep(0.5)
+        assert saw_error, "crash never surfaced as ERROR"
+        assert state == "running", f"never restarted (state={state})"
+        assert inst["restart_count"] >= 1
+        # still serves
+        r = client.post("/v1/chat/completions", json={
+            "model": "tiny-chat", "messages": [{"role": "user", "content": "x"}],
+            "max_tokens": 2, "ignore_eos": True})
+        assert r.status_code == 200
+    finally:
+        sm_mod.RESTART_BASE = old_base
