@@ -145,19 +145,37 @@ class ModelRunner:
         positions = torch.as_tensor(batch.positions, dtype=torch.long).to(dev)
         slots = torch.as_tensor(batch.slot_mapping, dtype=torch.long).to(dev)
         if batch.is_prefill:
-            # last token of each sequence produces the next-token logits
+            # last token of each prefill sequence produces its logits row
+            n_pre = batch.n_prefill_seqs or len(batch.seqs)
+            pre_lens = batch.seq_lens[:n_pre]
             idx, off = [], 0
-            for L in batch.seq_lens:
+            for L in pre_lens:
                 idx.append(off + L - 1)
                 off += L
-            tiles = ops.build_prefill_tiles(batch.seq_lens, dev)
+            tp = batch.num_prefill_tokens or batch.num_tokens
+            n_dec = len(batch.seqs) - n_pre
+            # decode rows (mixed batch) produce logits directly
+            idx.extend(range(tp, tp + n_dec))
+            tiles = ops.build_prefill_tiles(pre_lens, dev)
+            bt = None
+            dec_lens = None
+            if n_dec:
+                maxb = max(len(s.block_table) for s in batch.seqs[n_pre:])
+                btc = torch.zeros(n_dec, maxb, dtype=torch.int32)
+                for i, s in enumerate(batch.seqs[n_pre:]):
+                    btc[i, : len(s.block_table)] = torch.tensor(s.block_table, dtype=torch.int32)
+                bt = btc.to(dev)
+                dec_lens = torch.tensor(batch.decode_seq_lens, dtype=torch.int32, device=dev)
             meta = ForwardMeta(
                 is_prefill=True,
                 positions=positions,
                 slot_mapping=slots,
                 logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
-                seq_lens_list=batch.seq_lens,
+                seq_lens_list=pre_lens,
                 tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
+                block_tables=bt,
+                seq_lens=dec_lens,
+                num_prefill_tokens=tp,
             )
         else:
             rps = batch.rows_per_seq

@@ -34,6 +34,11 @@ class ScheduledBatch:
     rows_per_seq: int = 1
     drafts: list[list[int]] | None = None
     k_eff: list[int] | None = None
+    # mixed batches: prefill tokens first, then decode rows (one per running
+    # seq) — decode piggybacks on the prefill step's weight stream
+    n_prefill_seqs: int = 0
+    num_prefill_tokens: int = 0
+    decode_seq_lens: list[int] | None = None
 
     @property
     def num_tokens(self) -> int:
@@ -98,8 +103,48 @@ class Scheduler:
         self._swap_in_ready()
         batch = self._schedule_prefill()
         if batch is not None:
+            self._append_decode_rows(batch)
             return batch
         return self._schedule_decode()
+
+    def _append_decode_rows(self, batch: ScheduledBatch) -> None:
+        """Mixed batching: running sequences decode one token inside the
+        prefill step (shared GEMM weight stream; no decode stall during
+        admission bursts). Spec decoding is not applied to mixed steps."""
+        batch.n_prefill_seqs = len(batch.seqs)
+        batch.num_prefill_tokens = batch.num_tokens
+        if not self.running:
+            return
+        # ensure slots exist (same preemption rules as decode-only)
+        i = 0
+        while i < len(self.running):
+            seq = self.running[i]
+            pos = seq.num_tokens - 1
+            if (pos + 1) > len(seq.block_table) * self.kv.block_size:
+                try:
+                    seq.block_table.extend(self.kv.allocator.allocate(1))
+                except RuntimeError:
+                    victim = self.running.pop()
+                    self._release(victim)
+                    victim.status = SeqStatus.WAITING
+                    victim.preemptions += 1
+                    self.waiting.appendleft(victim)
+                    if victim is seq:
+                        continue
+                    i = min(i, len(self.running))
+                    continue
+            i += 1
+        bs = self.kv.block_size
+        lens: list[int] = []
+        for seq in self.running:
+            pos = seq.num_tokens - 1
+            out = seq.output_token_ids
+            batch.seqs.append(seq)
+            batch.token_ids.append(out[-1] if out else seq.prompt_token_ids[-1])
+            batch.positions.append(pos)
+            batch.slot_mapping.append(seq.block_table[pos // bs] * bs + pos % bs)
+            lens.append(pos + 1)
+        batch.decode_seq_lens = lens
 
     def _swap_in_ready(self) -> None:
         while (self.swapped
@@ -229,7 +274,8 @@ class Scheduler:
 
     # -- lifecycle ---------------------------------------------------------
     def on_prefill_done(self, batch: ScheduledBatch) -> None:
-        for seq in batch.seqs:
+        n = batch.n_prefill_seqs or len(batch.seqs)
+        for seq in batch.seqs[:n]:
             seq.num_cached_tokens = seq.num_tokens
             self.running.append(seq)
 

@@ -39,6 +39,9 @@ class ForwardMeta:
     # decode
     block_tables: torch.Tensor | None = None  # [N, maxb] int32
     seq_lens: torch.Tensor | None = None      # [N] int32
+    # mixed batches: rows [0, num_prefill_tokens) are prefill, the rest are
+    # single-token decode rows
+    num_prefill_tokens: int = 0
 
 
 class Attention(nn.Module):
@@ -79,12 +82,18 @@ class Attention(nn.Module):
             ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
         ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d, self.d)
         ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
-        out = torch.empty_like(q)
+        out = torch.empty(T, self.hq, self.d, dtype=q.dtype, device=q.device)
         if meta.is_prefill:
+            tp = meta.num_prefill_tokens or T
             ops.varlen_prefill_attn(
-                out, q, k, v, meta.seq_lens_list, self.scale,
+                out[:tp], q[:tp], k[:tp], v[:tp], meta.seq_lens_list, self.scale,
                 tiles=(meta.tile_start, meta.tile_q0, meta.tile_len),
             )
+            if tp < T:  # mixed: decode rows ride the same forward
+                ops.paged_attn_decode(
+                    out[tp:], q[tp:], k_cache, v_cache,
+                    meta.block_tables, meta.seq_lens, self.scale,
+                )
         else:
             ops.paged_attn_decode(
                 out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale
