@@ -207,6 +207,20 @@ class ModelUsage(Base, TimestampMixin, SerializeMixin):
     request_count = Column(Integer, default=0)
 
 
+class ModelUsageArchive(Base, TimestampMixin, SerializeMixin):
+    """Cold tier for usage rows (reference hot+archive table pairs,
+    server/usage_archiver.py)."""
+    __tablename__ = "model_usage_archive"
+    id = Column(Integer, primary_key=True)
+    user_id = Column(Integer, index=True)
+    model_id = Column(Integer, index=True)
+    model_name = Column(String(256), index=True)
+    date = Column(String(16), index=True)
+    prompt_tokens = Column(Integer, default=0)
+    completion_tokens = Column(Integer, default=0)
+    request_count = Column(Integer, default=0)
+
+
 class SystemLoad(Base, SerializeMixin):
     __tablename__ = "system_load"
     id = Column(Integer, primary_key=True)

@@ -116,12 +116,13 @@ def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
 def start_background_tasks(cfg: Config, app: FastAPI) -> None:
     from ..scheduler.scheduler import PlacementScheduler
     from .controllers import (
-        ModelController, ScalingScheduler, SystemLoadCollector, WorkerMonitor,
+        ModelController, ScalingScheduler, SystemLoadCollector, UsageArchiver,
+        WorkerMonitor,
     )
 
     sched = PlacementScheduler(cfg)
     tasks = [sched, ModelController(cfg), WorkerMonitor(cfg),
-             SystemLoadCollector(cfg), ScalingScheduler(cfg)]
+             SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg)]
     app.state.scheduler = sched
     app.state.background_tasks = tasks
     threads = [

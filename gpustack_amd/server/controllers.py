@@ -194,6 +194,55 @@ class ScalingScheduler:
                 time.sleep(1.0)
 
 
+class UsageArchiver:
+    """Hot -> archive mover for usage rows older than `keep_days`
+    (reference: server/usage_archiver.py TableArchiver)."""
+
+    def __init__(self, cfg: Config, keep_days: int = 30, interval: float = 3600.0):
+        self.cfg = cfg
+        self.keep_days = keep_days
+        self.interval = interval
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
+
+    def archive_once(self) -> int:
+        import datetime as dt
+
+        from ..schemas import ModelUsage
+        from ..schemas.tables import ModelUsageArchive
+
+        cutoff = (dt.date.today() - dt.timedelta(days=self.keep_days)).isoformat()
+        moved = 0
+        with get_session() as s:
+            for row in s.query(ModelUsage).filter(ModelUsage.date < cutoff).all():
+                s.add(ModelUsageArchive(
+                    user_id=row.user_id, model_id=row.model_id,
+                    model_name=row.model_name, date=row.date,
+                    prompt_tokens=row.prompt_tokens,
+                    completion_tokens=row.completion_tokens,
+                    request_count=row.request_count,
+                ))
+                s.delete(row)
+                moved += 1
+            s.commit()
+        return moved
+
+    def run(self) -> None:
+        while not self._stop:
+            try:
+                n = self.archive_once()
+                if n:
+                    logger.info("archived %d usage rows", n)
+            except Exception:  # noqa: BLE001
+                logger.exception("usage archiver cycle failed")
+            for _ in range(int(self.interval)):
+                if self._stop:
+                    return
+                time.sleep(1.0)
+
+
 class SystemLoadCollector:
     def __init__(self, cfg: Config, interval: float = 60.0):
         self.cfg = cfg

@@ -96,6 +96,30 @@ class WorkerAgent:
         def healthz():
             return {"status": "ok", "worker_id": self.worker_id}
 
+        @app.get("/files/model-config")
+        def model_config(path: str):
+            """Read a local model dir's config.json for the server/scheduler
+            (reference: routes/worker/filesystem.py /files/model-config)."""
+            import json as _json
+
+            p = Path(path) / "config.json"
+            if not p.exists():
+                raise HTTPException(404, "config.json not found")
+            return _json.loads(p.read_text())
+
+        @app.get("/files/file-exists")
+        def file_exists(path: str):
+            p = Path(path)
+            return {"exists": p.exists(), "is_dir": p.is_dir(),
+                    "size": p.stat().st_size if p.is_file() else None}
+
+        @app.get("/files/model-weight-size")
+        def model_weight_size(path: str):
+            total = 0
+            for f in Path(path).glob("*.safetensors"):
+                total += f.stat().st_size
+            return {"weight_size_bytes": total}
+
         @app.get("/logs/{instance_name}")
         def logs(instance_name: str, tail: int = 200):
             path = Path(self.cfg.data_dir) / "log" / "instances" / f"{instance_name}.log"

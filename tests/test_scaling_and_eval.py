@@ -87,3 +87,24 @@ def test_reset_admin_password_cli():
     c = TestClient(app2)
     assert c.post("/auth/login", json={"username": "admin", "password": "newpw"}).status_code == 200
     assert c.post("/auth/login", json={"username": "admin", "password": "old"}).status_code == 401
+
+
+def test_usage_archiver_moves_old_rows():
+    from gpustack_amd.db import get_session
+    from gpustack_amd.schemas import ModelUsage
+    from gpustack_amd.schemas.tables import ModelUsageArchive
+    from gpustack_amd.server.controllers import UsageArchiver
+
+    c, app, cfg = _server()
+    with get_session() as s:
+        s.add(ModelUsage(user_id=1, model_id=1, model_name="m", date="2020-01-01",
+                         prompt_tokens=10, completion_tokens=5, request_count=1))
+        s.add(ModelUsage(user_id=1, model_id=1, model_name="m", date="2999-01-01",
+                         prompt_tokens=1, completion_tokens=1, request_count=1))
+        s.commit()
+    moved = UsageArchiver(cfg, keep_days=30).archive_once()
+    assert moved == 1
+    with get_session() as s:
+        assert s.query(ModelUsage).count() == 1
+        arch = s.query(ModelUsageArchive).all()
+        assert len(arch) == 1 and arch[0].prompt_tokens == 10
