@@ -136,6 +136,12 @@ class EngineConfig:
     speculative: dict | None = None
     # LoRA adapters merged into the weights at load (reference lora_list)
     lora_dirs: list[str] = field(default_factory=list)
+    # admission hysteresis: open a prefill step only when this many requests
+    # wait, one has waited admission_max_wait_s, or nothing is running —
+    # keeps steady-state decode on the hipGraph path instead of degrading
+    # every step to a tiny eager mixed batch
+    admission_min_seqs: int = 16
+    admission_max_wait_s: float = 0.1
 
     spec: ModelSpec = field(default_factory=ModelSpec)
 

@@ -103,7 +103,13 @@ class Scheduler:
         self._swap_in_ready()
         batch = self._schedule_prefill()
         if batch is not None:
-            self._append_decode_rows(batch)
+            import os
+
+            if os.environ.get("GPUSTACK_AMD_NO_MIXED", "0") != "1":
+                self._append_decode_rows(batch)
+            else:
+                batch.n_prefill_seqs = len(batch.seqs)
+                batch.num_prefill_tokens = batch.num_tokens
             return batch
         return self._schedule_decode()
 
@@ -159,6 +165,14 @@ class Scheduler:
     def _schedule_prefill(self) -> ScheduledBatch | None:
         if not self.waiting:
             return None
+        if self.running:
+            import time as _time
+
+            oldest = self.waiting[0].arrival_time
+            if (len(self.waiting) < self.cfg.admission_min_seqs
+                    and _time.monotonic() - oldest < self.cfg.admission_max_wait_s
+                    and len(self.running) >= self.cfg.admission_min_seqs):
+                return None  # let decode keep its graph cadence
         batch = ScheduledBatch(is_prefill=True)
         budget = self.cfg.max_prefill_tokens
         while self.waiting and len(self.running) + len(batch.seqs) < self.cfg.max_num_seqs:
