@@ -105,7 +105,10 @@ class Scheduler:
         if batch is not None:
             import os
 
-            if os.environ.get("GPUSTACK_AMD_NO_MIXED", "0") != "1":
+            # Mixed steps measured ~5% slower than keeping decode on the
+            # hipGraph path at the default workload (A/B b_hyst vs
+            # b_hyst_nomix) — opt-in only.
+            if os.environ.get("GPUSTACK_AMD_MIXED", "0") == "1":
                 self._append_decode_rows(batch)
             else:
                 batch.n_prefill_seqs = len(batch.seqs)
