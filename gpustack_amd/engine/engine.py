@@ -31,6 +31,17 @@ class LLMEngine:
         # TP: rank 0 buffers add/abort ops; step() broadcasts them so every
         # rank replays the same scheduler state deterministically
         self._pending_ops: list[tuple] = []
+        # async decode pipeline: (seqs, event) of the submitted-but-unread
+        # step; the host processes step N-1 while the GPU runs step N
+        import os
+
+        self._async_enabled = (
+            os.environ.get("GPUSTACK_AMD_ASYNC", "1") == "1"
+            and self.comm.tp_size == 1
+            and cfg.speculative is None
+        )
+        self._pending: tuple[list[Sequence], object] | None = None
+        self._carry_outputs: list[StepOutput] = []
         logger.info(
             "engine ready: model=%s kv_blocks=%d (%.1f GiB KV pool)",
             cfg.model, kv.num_blocks,
@@ -66,6 +77,10 @@ class LLMEngine:
             assert self.comm.tp_rank == 0
             self._pending_ops.append(("abort", request_id))
             return True
+        if self._pending is not None:
+            # an in-flight graph still writes this seq's KV slot; drain
+            # before the abort can release blocks for reuse
+            self._carry_outputs.extend(self._drain())
         ok = self.scheduler.abort(request_id)
         self.seqs.pop(request_id, None)
         return ok
@@ -112,16 +127,104 @@ class LLMEngine:
     def num_waiting(self) -> int:
         return len(self.scheduler.waiting)
 
+    # -- async decode pipeline --------------------------------------------
+
+    def _must_drain_before_schedule(self) -> bool:
+        sch = self.scheduler
+        if sch.would_admit():
+            return True  # admission changes composition
+        if sch.kv.allocator.num_free < len(sch.running) + 8:
+            return True  # preemption possible under pressure
+        for seq in self._pending[0]:
+            p = seq.params
+            if not p.ignore_eos:
+                return True  # unpredictable stop
+            if len(seq.output_token_ids) + seq.pending_tokens >= p.max_tokens:
+                return True  # imminent length finish
+            if seq.num_tokens + 1 >= self.cfg.max_model_len:
+                return True
+        return False
+
+    def _drain(self) -> list[StepOutput]:
+        entry = self._pending
+        self._pending = None
+        return self._drain_entry(entry)
+
+    def _drain_entry(self, entry) -> list[StepOutput]:
+        seqs, ev = entry
+        toks = self.runner.read_sampled(len(seqs), ev)
+        outputs: list[StepOutput] = []
+        for seq, tok in zip(seqs, toks):
+            seq.pending_tokens -= 1
+            if seq.status == SeqStatus.FINISHED:
+                continue  # aborted/finished while in flight
+            tok = int(tok)
+            seq.record_first_token()
+            seq.output_token_ids.append(tok)
+            reason = self._finish_reason(seq, tok)
+            outputs.append(StepOutput(seq.request_id, tok, reason is not None, reason))
+            if reason:
+                self.scheduler.finish_seq(seq, reason)
+                self.seqs.pop(seq.request_id, None)
+        return outputs
+
+    @staticmethod
+    def _same_seqs(a: list[Sequence], b: list[Sequence]) -> bool:
+        return len(a) == len(b) and all(x is y for x, y in zip(a, b))
+
+    def _refill_tokens(self, batch) -> None:
+        """Replace placeholder tokens with the real (now drained) values."""
+        if batch.is_prefill:
+            base = batch.num_prefill_tokens
+            dec = batch.seqs[batch.n_prefill_seqs:]
+        else:
+            base = 0
+            dec = batch.seqs
+        for i, seq in enumerate(dec):
+            out = seq.output_token_ids
+            batch.token_ids[base + i] = out[-1] if out else seq.prompt_token_ids[-1]
+
+    def _async_ok(self, batch) -> bool:
+        return (
+            self._async_enabled
+            and not batch.is_prefill
+            and batch.rows_per_seq == 1
+            and all(s.params.greedy for s in batch.seqs)
+        )
+
     def step(self) -> list[StepOutput]:
         if self.comm.tp_size > 1:
             self._sync_tp_ops()
+        outputs: list[StepOutput] = []
+        if self._carry_outputs:
+            outputs.extend(self._carry_outputs)
+            self._carry_outputs = []
+        if self._pending is not None and self._must_drain_before_schedule():
+            outputs.extend(self._drain())
         batch = self.scheduler.schedule()
         if batch is None:
-            return []
+            if self._pending is not None:
+                outputs.extend(self._drain())
+            return outputs
+        if self._async_ok(batch):
+            reuse = self._pending is not None and self._same_seqs(self._pending[0], batch.seqs)
+            if self._pending is not None and not reuse:
+                outputs.extend(self._drain())
+                self._refill_tokens(batch)
+            old = self._pending if reuse else None
+            ev = self.runner.execute_async(batch, reuse_tokens=reuse)
+            for seq in batch.seqs:
+                seq.pending_tokens += 1
+            self._pending = (list(batch.seqs), ev)
+            if old is not None:
+                outputs.extend(self._drain_entry(old))
+            return outputs
+        if self._pending is not None:  # sync fallback with stale placeholders
+            outputs.extend(self._drain())
+            self._refill_tokens(batch)
         token_ids = self.runner.execute(batch)
         if batch.is_prefill:
             self.scheduler.on_prefill_done(batch)
-        outputs: list[StepOutput] = []
         rps = 1 if batch.is_prefill else batch.rows_per_seq
         for i, seq in enumerate(batch.seqs):
             if rps == 1:

@@ -103,14 +103,15 @@ class DecodeGraphRunner:
                 and len(batch.seqs) * batch.rows_per_seq <= self.max_bs
                 and bool(self.graphs))
 
-    def run(self, batch: ScheduledBatch) -> torch.Tensor:
+    def run(self, batch: ScheduledBatch, token_src: torch.Tensor | None = None) -> torch.Tensor:
         import numpy as np
 
         rps = batch.rows_per_seq
         bs = len(batch.seqs) * rps
         bucket = next(b for b in self.buckets if b >= bs)
         # host staging (numpy views over pinned memory)
-        self.n_tokens[:bs] = batch.token_ids
+        if token_src is None:
+            self.n_tokens[:bs] = batch.token_ids
         self.n_positions[:bs] = batch.positions
         self.n_slots[:bs] = batch.slot_mapping
         self.n_seq_lens[:bs] = np.asarray(batch.seq_lens, dtype=np.int32)
@@ -141,7 +142,14 @@ class DecodeGraphRunner:
             dirty_hi = max(dirty_hi, hi - 1)
         self._prev_bs = bucket
         n = hi
-        self.tokens[:n].copy_(self.h_tokens[:n], non_blocking=True)
+        if token_src is not None:
+            # async pipeline: previous step's sampled ids feed this step
+            # device-to-device (same stream => ordered after the sampler)
+            self.tokens[:bs].copy_(token_src[:bs])
+            if n > bs:
+                self.tokens[bs:n].copy_(self.h_tokens[bs:n], non_blocking=True)
+        else:
+            self.tokens[:n].copy_(self.h_tokens[:n], non_blocking=True)
         self.positions[:n].copy_(self.h_positions[:n], non_blocking=True)
         self.slots[:n].copy_(self.h_slots[:n], non_blocking=True)
         self.seq_lens[:n].copy_(self.h_seq_lens[:n], non_blocking=True)

@@ -88,6 +88,14 @@ class ModelRunner:
                 logging.getLogger(__name__).info("merged LoRA %s (%d tensors)", d, n)
         self.sampler = Sampler(self.device)
         self.kv: KVCache | None = None
+        # async decode pipeline buffers (pinned side is double-buffered:
+        # step N+1's D2H must not overwrite step N before the host reads it)
+        mb = cfg.max_num_seqs
+        self._sampled_dev = torch.zeros(mb, dtype=torch.long, device=self.device)
+        pin = self.device.type == "cuda"
+        self._sampled_pin = [torch.zeros(mb, dtype=torch.long, pin_memory=pin)
+                             for _ in range(2)]
+        self._pin_idx = 0
 
     def _init_tunableop(self) -> None:
         """hipBLASLt algorithm selection via torch TunableOp.
@@ -195,6 +203,41 @@ class ModelRunner:
                 seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
             )
         return tokens, meta
+
+    @torch.inference_mode()
+    def execute_async(self, batch: ScheduledBatch, reuse_tokens: bool):
+        """Submit a greedy decode step without reading results back.
+
+        Samples on-device into the persistent buffer and starts an async
+        D2H copy; read_sampled() collects it one step later. With
+        reuse_tokens, the previous step's sampled ids feed this step's
+        token input device-side (no host round-trip at all)."""
+        bs = len(batch.seqs)
+        token_src = self._sampled_dev if reuse_tokens else None
+        if self.graph_runner is not None and self.graph_runner.can_run(batch):
+            logits = self.graph_runner.run(batch, token_src=token_src)
+        else:
+            tokens, meta = self._meta(batch)
+            if token_src is not None:
+                tokens = token_src[:bs].to(self.device)
+            logits = self.model(tokens, meta, self.kv)
+        from .. import ops
+
+        ops.greedy_sample_into(self._sampled_dev[:bs], logits)
+        slot = self._pin_idx
+        self._pin_idx ^= 1
+        self._sampled_pin[slot][:bs].copy_(self._sampled_dev[:bs], non_blocking=True)
+        if self.device.type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record()
+            return (ev, slot)
+        return (None, slot)
+
+    def read_sampled(self, bs: int, handle) -> list[int]:
+        ev, slot = handle
+        if ev is not None:
+            ev.synchronize()
+        return self._sampled_pin[slot][:bs].tolist()
 
     @torch.inference_mode()
     def execute(self, batch: ScheduledBatch) -> list[int]:

@@ -165,17 +165,28 @@ class Scheduler:
             seq.status = SeqStatus.RUNNING
             self.running.append(seq)
 
+    def would_admit(self) -> bool:
+        """True if the next schedule() may open a prefill step (shared
+        predicate with _schedule_prefill; also used by the async decode
+        pipeline to decide when it must drain)."""
+        if not self.waiting and not self.swapped:
+            return False
+        if not self.running:
+            return True
+        import time as _time
+
+        if not self.waiting:
+            return True  # swapped re-admission
+        oldest = self.waiting[0].arrival_time
+        return (len(self.waiting) >= self.cfg.admission_min_seqs
+                or _time.monotonic() - oldest >= self.cfg.admission_max_wait_s
+                or len(self.running) < self.cfg.admission_min_seqs)
+
     def _schedule_prefill(self) -> ScheduledBatch | None:
         if not self.waiting:
             return None
-        if self.running:
-            import time as _time
-
-            oldest = self.waiting[0].arrival_time
-            if (len(self.waiting) < self.cfg.admission_min_seqs
-                    and _time.monotonic() - oldest < self.cfg.admission_max_wait_s
-                    and len(self.running) >= self.cfg.admission_min_seqs):
-                return None  # let decode keep its graph cadence
+        if self.running and not self.would_admit():
+            return None  # let decode keep its graph cadence
         batch = ScheduledBatch(is_prefill=True)
         budget = self.cfg.max_prefill_tokens
         while self.waiting and len(self.running) + len(batch.seqs) < self.cfg.max_num_seqs:
@@ -252,7 +263,10 @@ class Scheduler:
         for i, seq in enumerate(self.running):
             pos = seq.num_tokens - 1
             out = seq.output_token_ids
-            last = out[-1] if out else seq.prompt_token_ids[-1]
+            if seq.pending_tokens:
+                last = 0  # placeholder: async path feeds tokens from device
+            else:
+                last = out[-1] if out else seq.prompt_token_ids[-1]
             if rps == 1:
                 toks[i] = last
                 poss[i] = pos

@@ -38,6 +38,7 @@ class Sequence:
     host_block_table: list[int] = field(default_factory=list)  # offload tier
     num_cached_tokens: int = 0      # tokens whose KV is already in the pool
     swap_outs: int = 0
+    pending_tokens: int = 0         # async decode: sampled on device, not yet read back
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None
     finish_time: float | None = None
@@ -46,7 +47,7 @@ class Sequence:
 
     @property
     def num_tokens(self) -> int:
-        return len(self.prompt_token_ids) + len(self.output_token_ids)
+        return len(self.prompt_token_ids) + len(self.output_token_ids) + self.pending_tokens
 
     @property
     def all_token_ids(self) -> list[int]:

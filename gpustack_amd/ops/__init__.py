@@ -97,6 +97,14 @@ def reshape_and_cache(k, v, k_cache, v_cache, slots) -> None:
         torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slots)
 
 
+def greedy_sample_into(out, logits) -> None:
+    hip = _backend(logits)
+    if hip is not None:
+        hip.greedy_sample(out, logits)
+    else:
+        torch_ref.greedy_sample(out, logits)
+
+
 def greedy_sample(logits) -> torch.Tensor:
     out = torch.empty(logits.shape[0], dtype=torch.long, device=logits.device)
     hip = _backend(logits)
