@@ -205,6 +205,39 @@ class ModelRunner:
         return tokens, meta
 
     @torch.inference_mode()
+    def embed(self, prompts: list[list[int]], pooling: str = "last") -> list[list[float]]:
+        """Embedding forward (reference category "embedding",
+        schemas/models.py:51): prefill with KV writes disabled
+        (slot_mapping = -1), pooled + L2-normalized hidden states."""
+        dev = self.device
+        lens = [min(len(p), self.cfg.max_model_len) for p in prompts]
+        flat: list[int] = []
+        positions: list[int] = []
+        for p, L in zip(prompts, lens):
+            flat.extend(p[:L])
+            positions.extend(range(L))
+        tokens = torch.tensor(flat, dtype=torch.long, device=dev)
+        pos = torch.tensor(positions, dtype=torch.long, device=dev)
+        slots = torch.full((len(flat),), -1, dtype=torch.long, device=dev)
+        tiles = ops.build_prefill_tiles(lens, dev)
+        meta = ForwardMeta(
+            is_prefill=True, positions=pos, slot_mapping=slots,
+            logits_indices=torch.zeros(1, dtype=torch.long, device=dev),
+            seq_lens_list=lens,
+            tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
+        )
+        hidden = self.model(tokens, meta, self.kv, return_hidden=True).float()
+        out: list[list[float]] = []
+        off = 0
+        for L in lens:
+            seg = hidden[off:off + L]
+            vec = seg.mean(dim=0) if pooling == "mean" else seg[-1]
+            vec = vec / (vec.norm() + 1e-12)
+            out.append(vec.cpu().tolist())
+            off += L
+        return out
+
+    @torch.inference_mode()
     def execute_async(self, batch: ScheduledBatch, reuse_tokens: bool):
         """Submit a greedy decode step without reading results back.
 

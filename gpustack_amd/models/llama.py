@@ -172,11 +172,14 @@ class LlamaForCausalLM(nn.Module):
         self.to(device)
 
     @torch.inference_mode()
-    def forward(self, token_ids: torch.Tensor, meta: ForwardMeta, kv) -> torch.Tensor:
+    def forward(self, token_ids: torch.Tensor, meta: ForwardMeta, kv,
+                return_hidden: bool = False) -> torch.Tensor:
         x = F.embedding(token_ids, self.embed)
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
         ops.fused_add_rms_norm(x, residual, self.final_norm, self.spec.rms_norm_eps)
+        if return_hidden:
+            return x  # all rows, post final norm (embedding serving)
         hidden = x[meta.logits_indices]
         return F.linear(hidden, self.lm_head)

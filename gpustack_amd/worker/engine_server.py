@@ -72,6 +72,7 @@ class EngineRunner:
         self.tokenizer = load_tokenizer(engine_cfg.model_dir, engine_cfg.spec.vocab_size)
         self.loop: asyncio.AbstractEventLoop | None = None
         self._queues: dict[str, asyncio.Queue] = {}
+        self._aux_jobs: "list[tuple]" = []  # (fn, args, asyncio.Future)
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._stop = False
@@ -86,9 +87,30 @@ class EngineRunner:
         self._stop = True
         self._wake.set()
 
+    def _run_aux_jobs(self) -> None:
+        with self._lock:
+            jobs, self._aux_jobs = self._aux_jobs, []
+        for fn, args, fut in jobs:
+            try:
+                result = fn(*args)
+                self.loop.call_soon_threadsafe(fut.set_result, result)
+            except Exception as e:  # noqa: BLE001
+                self.loop.call_soon_threadsafe(fut.set_exception, e)
+
+    async def run_aux(self, fn, *args):
+        """Run fn on the engine thread (between steps) and await the result
+        — used by non-generative endpoints like /v1/embeddings."""
+        fut = asyncio.get_running_loop().create_future()
+        with self._lock:
+            self._aux_jobs.append((fn, args, fut))
+        self._wake.set()
+        return await fut
+
     def _run(self) -> None:
         tp = self.engine.comm.tp_size > 1
         while not self._stop:
+            if self._aux_jobs:
+                self._run_aux_jobs()
             if tp:
                 # coordinated loop: the tp_active broadcast doubles as the
                 # idle heartbeat for follower ranks
@@ -313,6 +335,33 @@ def create_app(runner: EngineRunner) -> FastAPI:
         if hasattr(ids, "ids"):
             ids = ids.ids
         return await _generate(request, body, list(ids), "chat")
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        body = await request.json()
+        inputs = body.get("input", [])
+        if isinstance(inputs, str):
+            inputs = [inputs]
+        tok = runner.tokenizer
+        prompts = []
+        for item in inputs:
+            if isinstance(item, str):
+                ids = tok.encode(item)
+                if hasattr(ids, "ids"):
+                    ids = ids.ids
+                prompts.append(list(ids) or [0])
+            else:
+                prompts.append([int(t) for t in item] or [0])
+        pooling = body.get("pooling", "last")
+        vecs = await runner.run_aux(runner.engine.runner.embed, prompts, pooling)
+        return {
+            "object": "list",
+            "model": runner.served_name,
+            "data": [{"object": "embedding", "index": i, "embedding": v}
+                     for i, v in enumerate(vecs)],
+            "usage": {"prompt_tokens": sum(len(p) for p in prompts),
+                      "total_tokens": sum(len(p) for p in prompts)},
+        }
 
     @app.post("/v1/completions")
     async def completions(request: Request):
