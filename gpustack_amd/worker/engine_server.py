@@ -166,6 +166,15 @@ class EngineRunner:
         self.release(rid)
 
 
+def _stop_strings(body: dict) -> list[str]:
+    stop = body.get("stop")
+    if stop is None:
+        return []
+    if isinstance(stop, str):
+        return [stop]
+    return [s for s in stop if isinstance(s, str)]
+
+
 def _sampling_params(body: dict, eos_token_id: int):
     from ..engine import SamplingParams
 
@@ -222,6 +231,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
     async def _generate(request: Request, body: dict, prompt_ids: list[int],
                         kind: str, echo_text_prefix: str = ""):
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
+        stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
         created = int(time.time())
         stream = bool(body.get("stream"))
@@ -291,12 +301,22 @@ def create_app(runner: EngineRunner) -> FastAPI:
                 if "error" in item:
                     raise HTTPException(500, item["error"])
                 tokens.append(item["token_id"])
+                if stop_strs:
+                    t = runner.tokenizer.decode(tokens)
+                    if any(ss in t for ss in stop_strs):
+                        runner.abort(rid)
+                        finish = "stop"
+                        break
                 if item["finished"]:
                     finish = item.get("finish_reason") or "stop"
                     break
         finally:
             runner.release(rid)
         text = runner.tokenizer.decode(tokens)
+        if stop_strs:
+            cuts = [text.find(ss) for ss in stop_strs if ss in text]
+            if cuts:
+                text = text[:min(cuts)]
         usage = {
             "prompt_tokens": len(prompt_ids),
             "completion_tokens": len(tokens),

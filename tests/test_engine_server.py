@@ -62,3 +62,42 @@ def test_engine_server_tp2_cpu():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_engine_server_stop_strings():
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-s", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        _wait_health(port, proc)
+        # find what the model actually generates, then stop on a substring
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-s", "prompt": "hello", "max_tokens": 20,
+            "ignore_eos": True, "temperature": 0,
+        }, timeout=60)
+        full = r.json()["choices"][0]["text"]
+        if len(full) > 4:
+            stop = full[2:4]
+            r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+                "model": "tiny-s", "prompt": "hello", "max_tokens": 20,
+                "ignore_eos": True, "temperature": 0, "stop": [stop],
+            }, timeout=60)
+            text = r.json()["choices"][0]["text"]
+            assert stop not in text
+            assert len(text) <= len(full)
+        # embeddings endpoint responds too
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/embeddings", json={
+            "model": "tiny-s", "input": ["abc", "def"]}, timeout=60)
+        assert r.status_code == 200
+        assert len(r.json()["data"]) == 2
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
