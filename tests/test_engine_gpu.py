@@ -65,3 +65,17 @@ def test_spec_matches_plain_gpu():
     del eng2
     torch.cuda.empty_cache()
     assert spec == plain
+
+
+def test_embed_gpu():
+    eng = LLMEngine(_cfg())
+    vecs = eng.runner.embed([[1, 2, 3, 4] * 10, [7, 8, 9]])
+    assert len(vecs) == 2 and len(vecs[0]) == eng.cfg.spec.hidden_size
+    import math
+
+    for v in vecs:
+        assert abs(math.sqrt(sum(x * x for x in v)) - 1.0) < 1e-3
+    # KV pool untouched
+    assert eng.scheduler.kv.allocator.num_free == eng.scheduler.kv.allocator.num_blocks
+    del eng
+    torch.cuda.empty_cache()
