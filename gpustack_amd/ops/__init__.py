@@ -150,6 +150,38 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float, tiles=N
         torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale)
 
 
+_SG_WORKSPACES: dict = {}
+
+
+def skinny_gemm(x, w, out=None, splitk: int | None = None):
+    """Split-K decode GEMM: out[M,N] = x[M,K] @ w[N,K]^T (bf16, f32 acc).
+    GPU-only (falls back to F.linear elsewhere or for unsupported shapes)."""
+    import torch.nn.functional as F
+
+    M, K = x.shape
+    N = w.shape[0]
+    if not x.is_cuda or N % 128 != 0 or K % 64 != 0:
+        return F.linear(x, w, None)
+    if splitk is None:
+        mt = (M + 255) // 256
+        nt = N // 128
+        splitk = 1
+        while (mt * nt * splitk < 400 and splitk < 8
+               and K % (64 * splitk * 2) == 0):
+            splitk *= 2
+    if out is None:
+        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    ws = None
+    if splitk > 1:
+        key = (splitk, M, N, x.device.index)
+        ws = _SG_WORKSPACES.get(key)
+        if ws is None:
+            ws = torch.empty(splitk * M * N, dtype=torch.float32, device=x.device)
+            _SG_WORKSPACES[key] = ws
+    _load_hip().skinny_gemm(out, x, w, ws, splitk)
+    return out
+
+
 def mfma_probe(a, b):
     return _load_hip().mfma_probe(a, b)
 

@@ -14,6 +14,7 @@ void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
 void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int*, hipStream_t);
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
+void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -159,6 +160,25 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   HIP_CHECK_LAST();
 }
 
+void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
+                 c10::optional<at::Tensor> workspace, long splitk) {
+  check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(N % 128 == 0, "N must be a multiple of 128");
+  TORCH_CHECK(K % (64 * splitk) == 0, "K must be a multiple of 64*splitk");
+  void* ws = nullptr;
+  if (splitk > 1) {
+    TORCH_CHECK(workspace.has_value(), "splitk>1 needs an f32 workspace");
+    TORCH_CHECK(workspace->scalar_type() == at::kFloat &&
+                workspace->numel() >= (long)splitk * M * N);
+    ws = workspace->data_ptr();
+  }
+  skinny_gemm_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), ws, M, N, K,
+                     (int)splitk, cur_stream(x));
+  HIP_CHECK_LAST();
+}
+
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a"); check_bf16(b, "b");
   auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
@@ -181,4 +201,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention");
   m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");
 }

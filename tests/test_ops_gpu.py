@@ -221,3 +221,21 @@ def test_flash_prefill_strided():
     ops.varlen_prefill_attn(out, q, k, v, lens, scale)
     ops.varlen_prefill_attn(out2, q.contiguous(), k.contiguous(), v.contiguous(), lens, scale)
     assert torch.equal(out, out2)
+
+
+@pytest.mark.parametrize("M,N,K,splitk", [
+    (512, 6144, 4096, None),    # qkv
+    (512, 4096, 4096, 4),       # o_proj, forced split
+    (256, 28672, 4096, 1),      # gate_up, no split
+    (512, 4096, 14336, 4),      # down_proj
+    (320, 4096, 4096, 2),       # M not a tile multiple (clamped rows)
+    (16, 4096, 4096, 8),        # tiny M
+])
+def test_skinny_gemm(M, N, K, splitk):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    got = ops.skinny_gemm(x, w, splitk=splitk)
+    ref = (x.float() @ w.float().T)
+    assert torch.allclose(got.float(), ref, atol=0.5, rtol=3e-2), (
+        (got.float() - ref).abs().max().item()
+    )
