@@ -1,4 +1,9 @@
 """Microbench: skinny_gemm vs hipBLASLt (F.linear) on decode shapes."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 import torch
 import torch.nn.functional as F
 
