@@ -190,7 +190,7 @@ void skinny_gemm_launch(void* out, const void* x, const void* w,
                        workspace, (const unsigned short*)x,
                        (const unsigned short*)w, M, N, K, splitk);
     long MN = (long)M * N;
-    int rgrid = (int)((MN / 256 + 255) / 256);
+    int rgrid = (int)((MN + 255) / 256);
     if (rgrid > 2048) rgrid = 2048;
     if (rgrid < 1) rgrid = 1;
     hipLaunchKernelGGL(sg_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
