@@ -115,7 +115,7 @@ class MLP(nn.Module):
         gu = F.linear(x, self.gate_up_w)
         act = torch.empty(x.shape[0], self.i, dtype=x.dtype, device=x.device)
         ops.silu_and_mul(act, gu)
-        return self.comm.all_reduce(F.linear(act, self.down_w))
+        return self.comm.all_reduce(ops.linear_auto(act, self.down_w))
 
 
 class DecoderLayer(nn.Module):

@@ -182,6 +182,25 @@ def skinny_gemm(x, w, out=None, splitk: int | None = None):
     return out
 
 
+def linear_auto(x, w):
+    """Shape-gated GEMM dispatch: the split-K skinny kernel where it
+    measured faster than hipBLASLt (scripts/bench_skinny_gemm.py on
+    MI355X: down_proj-like N=4096 K>=8192 M>=384 at splitk=4, 824 TF vs
+    hipBLASLt 680), F.linear everywhere else."""
+    import os
+
+    import torch.nn.functional as F
+
+    if (x.is_cuda and x.dim() == 2
+            and os.environ.get("GPUSTACK_AMD_SKINNY_GEMM", "1") == "1"):
+        M, K = x.shape
+        N = w.shape[0]
+        if (N == 4096 and K >= 8192 and 384 <= M <= 1024
+                and K % 256 == 0 and x.stride(1) == 1):
+            return skinny_gemm(x, w, splitk=4)
+    return F.linear(x, w)
+
+
 def mfma_probe(a, b):
     return _load_hip().mfma_probe(a, b)
 
