@@ -153,7 +153,7 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float, tiles=N
 _SG_WORKSPACES: dict = {}
 
 
-def skinny_gemm(x, w, out=None, splitk: int | None = None):
+def skinny_gemm(x, w, out=None, splitk: int | None = None, version: int = 1):
     """Split-K decode GEMM: out[M,N] = x[M,K] @ w[N,K]^T (bf16, f32 acc).
     GPU-only (falls back to F.linear elsewhere or for unsupported shapes)."""
     import torch.nn.functional as F
@@ -178,7 +178,7 @@ def skinny_gemm(x, w, out=None, splitk: int | None = None):
         if ws is None:
             ws = torch.empty(splitk * M * N, dtype=torch.float32, device=x.device)
             _SG_WORKSPACES[key] = ws
-    _load_hip().skinny_gemm(out, x, w, ws, splitk)
+    _load_hip().skinny_gemm(out, x, w, ws, splitk, version)
     return out
 
 

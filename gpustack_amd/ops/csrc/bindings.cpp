@@ -15,6 +15,7 @@ void paged_attn_decode_launch(void*, const void*, const void*, const void*, cons
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
+void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -161,7 +162,8 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
 }
 
 void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
-                 c10::optional<at::Tensor> workspace, long splitk) {
+                 c10::optional<at::Tensor> workspace, long splitk,
+                 long version) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   const int M = x.size(0), K = x.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
@@ -174,8 +176,14 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
                 workspace->numel() >= (long)splitk * M * N);
     ws = workspace->data_ptr();
   }
-  skinny_gemm_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), ws, M, N, K,
-                     (int)splitk, cur_stream(x));
+  if (version == 2) {
+    TORCH_CHECK(N % 256 == 0, "v2 needs N % 256 == 0");
+    skinny_gemm_v2_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), ws, M,
+                          N, K, (int)splitk, cur_stream(x));
+  } else {
+    skinny_gemm_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), ws, M, N,
+                       K, (int)splitk, cur_stream(x));
+  }
   HIP_CHECK_LAST();
 }
 

@@ -39,5 +39,9 @@ for M, N, K in SHAPES:
             continue
         t = time_fn(lambda: ops.skinny_gemm(x, w, splitk=sk))
         tf = 2 * M * N * K / (t * 1e-6) / 1e12
-        line += f"  sk{sk}={t:7.1f}us({tf:5.1f}TF)"
+        line += f" sk{sk}={t:6.1f}({tf:4.0f}TF)"
+        if N % 256 == 0:
+            t2 = time_fn(lambda: ops.skinny_gemm(x, w, splitk=sk, version=2))
+            tf2 = 2 * M * N * K / (t2 * 1e-6) / 1e12
+            line += f" v2:{t2:6.1f}({tf2:4.0f}TF)"
     print(line)
