@@ -189,7 +189,7 @@ class LLMEngine:
             self._async_enabled
             and not batch.is_prefill
             and batch.rows_per_seq == 1
-            and all(s.params.greedy for s in batch.seqs)
+            and all(s.params.greedy and not s.params.logprobs for s in batch.seqs)
         )
 
     def step(self) -> list[StepOutput]:
@@ -223,14 +223,17 @@ class LLMEngine:
             outputs.extend(self._drain())
             self._refill_tokens(batch)
         token_ids = self.runner.execute(batch)
+        lps = getattr(self.runner, "last_logprobs", None)
         if batch.is_prefill:
             self.scheduler.on_prefill_done(batch)
         rps = 1 if batch.is_prefill else batch.rows_per_seq
         for i, seq in enumerate(batch.seqs):
             if rps == 1:
                 emitted = [token_ids[i]]
+                row0 = i
             else:
                 rows = token_ids[i * rps:(i + 1) * rps]
+                row0 = i * rps
                 if seq.params.greedy and batch.k_eff[i] > 0:
                     from .spec import accept_tokens
 
@@ -239,10 +242,13 @@ class LLMEngine:
                 else:
                     emitted = [rows[0]]
             seq.record_first_token()
-            for tok in emitted:
+            for j, tok in enumerate(emitted):
                 seq.output_token_ids.append(tok)
                 reason = self._finish_reason(seq, tok)
-                outputs.append(StepOutput(seq.request_id, tok, reason is not None, reason))
+                lp = (lps[row0 + j] if (lps is not None and seq.params.logprobs)
+                      else None)
+                outputs.append(StepOutput(seq.request_id, tok,
+                                          reason is not None, reason, lp))
                 if reason:
                     self.scheduler.finish_seq(seq, reason)
                     self.seqs.pop(seq.request_id, None)

@@ -283,6 +283,15 @@ class ModelRunner:
         if not batch.is_prefill and batch.rows_per_seq > 1:
             row_seqs = [s for s in batch.seqs for _ in range(batch.rows_per_seq)]
         token_ids = self.sampler.sample(logits, row_seqs)
+        self.last_logprobs = None
+        if any(s.params.logprobs for s in batch.seqs):
+            lf = logits.float()
+            lse = torch.logsumexp(lf, dim=-1)
+            chosen = lf.gather(
+                1, torch.as_tensor(token_ids, dtype=torch.long,
+                                   device=logits.device).unsqueeze(1)
+            ).squeeze(1)
+            self.last_logprobs = (chosen - lse).tolist()
         if self.comm.tp_size > 1:
             # ranks must agree on sampled tokens; rank 0 decides
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)

@@ -21,6 +21,7 @@ class SamplingParams:
     ignore_eos: bool = False
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None
+    logprobs: bool = False
 
     @property
     def greedy(self) -> bool:
@@ -79,3 +80,4 @@ class StepOutput:
     token_id: int
     finished: bool
     finish_reason: str | None = None
+    logprob: float | None = None

@@ -140,6 +140,7 @@ class EngineRunner:
                         "token_id": out.token_id,
                         "finished": out.finished,
                         "finish_reason": out.finish_reason,
+                        "logprob": out.logprob,
                     })
 
     def _push(self, q: asyncio.Queue, item: dict) -> None:
@@ -187,6 +188,7 @@ def _sampling_params(body: dict, eos_token_id: int):
         max_tokens=int(mt),
         ignore_eos=bool(body.get("ignore_eos", False)),
         seed=body.get("seed"),
+        logprobs=bool(body.get("logprobs")),
     )
 
 
@@ -294,6 +296,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         tokens: list[int] = []
+        lps: list[float | None] = []
         finish = "stop"
         try:
             while True:
@@ -301,6 +304,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
                 if "error" in item:
                     raise HTTPException(500, item["error"])
                 tokens.append(item["token_id"])
+                lps.append(item.get("logprob"))
                 if stop_strs:
                     t = runner.tokenizer.decode(tokens)
                     if any(ss in t for ss in stop_strs):
@@ -322,20 +326,30 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "completion_tokens": len(tokens),
             "total_tokens": len(prompt_ids) + len(tokens),
         }
+        want_lp = bool(body.get("logprobs")) and any(x is not None for x in lps)
         if kind == "chat":
+            choice = {"index": 0,
+                      "message": {"role": "assistant", "content": text},
+                      "finish_reason": finish}
+            if want_lp:
+                choice["logprobs"] = {"content": [
+                    {"token": runner.tokenizer.decode([t]), "logprob": lp}
+                    for t, lp in zip(tokens, lps)
+                ]}
             return JSONResponse({
                 "id": rid, "object": "chat.completion", "created": created,
-                "model": model_name,
-                "choices": [{"index": 0, "message": {"role": "assistant", "content": text},
-                             "finish_reason": finish}],
-                "usage": usage,
+                "model": model_name, "choices": [choice], "usage": usage,
             })
+        choice = {"index": 0, "text": echo_text_prefix + text,
+                  "finish_reason": finish}
+        if want_lp:
+            choice["logprobs"] = {
+                "tokens": [runner.tokenizer.decode([t]) for t in tokens],
+                "token_logprobs": lps,
+            }
         return JSONResponse({
             "id": rid, "object": "text_completion", "created": created,
-            "model": model_name,
-            "choices": [{"index": 0, "text": echo_text_prefix + text,
-                         "finish_reason": finish}],
-            "usage": usage,
+            "model": model_name, "choices": [choice], "usage": usage,
         })
 
     @app.post("/v1/chat/completions")

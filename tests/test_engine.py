@@ -147,3 +147,25 @@ def test_random_sampling_with_seed_reproducible():
     a = make_engine().generate([[1, 2, 3]], p)
     b = make_engine().generate([[1, 2, 3]], p)
     assert a == b
+
+
+def test_logprobs():
+    import math
+
+    eng = make_engine()
+    rid = eng.add_request([1, 2, 3, 4], SamplingParams(max_tokens=4, ignore_eos=True,
+                                                      logprobs=True))
+    outs = []
+    while eng.has_unfinished():
+        outs += [o for o in eng.step() if o.request_id == rid]
+    assert len(outs) == 4
+    for o in outs:
+        assert o.logprob is not None and o.logprob <= 1e-6
+        assert math.isfinite(o.logprob)
+    # without the flag, no logprobs are computed
+    eng2 = make_engine()
+    rid2 = eng2.add_request([1, 2, 3, 4], SamplingParams(max_tokens=2, ignore_eos=True))
+    outs2 = []
+    while eng2.has_unfinished():
+        outs2 += eng2.step()
+    assert all(o.logprob is None for o in outs2)
