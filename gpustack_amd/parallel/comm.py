@@ -52,7 +52,8 @@ def get_communicator() -> Communicator:
 
 
 def init_tp(tp_size: int, tp_rank: int, master_port: int | None = None,
-            backend: str | None = None, device_id: int | None = None) -> Communicator:
+            backend: str | None = None, device_id: int | None = None,
+            master_addr: str | None = None) -> Communicator:
     """Initialize the TP process group (rank bootstrap via TCP store on
     127.0.0.1, replacing the master-port scheme the reference's port
     allocator models — serve_manager.py:1685-1737)."""
@@ -63,6 +64,8 @@ def init_tp(tp_size: int, tp_rank: int, master_port: int | None = None,
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if not dist.is_initialized():
+        if master_addr:
+            os.environ["MASTER_ADDR"] = master_addr
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         if master_port is not None:
             os.environ["MASTER_PORT"] = str(master_port)

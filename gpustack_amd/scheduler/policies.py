@@ -29,6 +29,9 @@ class Candidate:
     gpu_indexes: list[int]
     score: float = 0.0
     vram_claim: dict[int, int] = field(default_factory=dict)  # gpu idx -> bytes
+    # multi-worker TP (reference: subordinate workers,
+    # vllm_resource_fit_selector.py:800-867): [(worker_dict, gpu_indexes)]
+    subordinates: list = field(default_factory=list)
 
 
 def model_spec_for(model: Model | dict) -> ModelSpec | None:
@@ -131,7 +134,38 @@ def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -
         if len(fits) >= tp:
             picks = fits[:tp]
             out.append(Candidate(w, picks, vram_claim={i: claim for i in picks}))
+    if not out and model.get("distributed_inference_across_workers"):
+        cand = _multi_worker_candidate(tp, claim, workers, instances)
+        if cand is not None:
+            out.append(cand)
     return out
+
+
+def _multi_worker_candidate(tp: int, claim: int, workers: list[dict],
+                            instances: list[dict]) -> Candidate | None:
+    """Equal-GPU-count worker groups (reference selector semantics): find
+    the smallest worker count nw where tp/nw GPUs fit on each of nw
+    workers; first worker is the main (rank 0), the rest subordinate."""
+    fits_per_worker = []
+    for w in workers:
+        alloc = worker_allocatable(w, instances)
+        fits = sorted((i for i, free in alloc.items() if free >= claim),
+                      key=lambda i: -alloc[i])
+        if fits:
+            fits_per_worker.append((w, fits))
+    for nw in range(2, len(fits_per_worker) + 1):
+        if tp % nw:
+            continue
+        per = tp // nw
+        group = [(w, f[:per]) for w, f in fits_per_worker if len(f) >= per]
+        if len(group) >= nw:
+            group = group[:nw]
+            main_w, main_g = group[0]
+            cand = Candidate(main_w, main_g,
+                             vram_claim={i: claim for i in main_g})
+            cand.subordinates = group[1:]
+            return cand
+    return None
 
 
 # ---- scorers (reference: policies/scorers/*) ------------------------------

@@ -103,6 +103,26 @@ class PlacementScheduler:
                 "vram": {str(i): claim for i in cand.gpu_indexes},
                 "ram": 2 << 30,
             }
+            if cand.subordinates:
+                # cross-worker TP: rank layout + rendezvous port
+                # (reference: serve_manager.py:1643-1739 port bands +
+                # distributed_servers subordinate list)
+                ranks = []
+                base = len(cand.gpu_indexes)
+                for w, gpus in cand.subordinates:
+                    ranks.append({
+                        "worker_id": w["id"],
+                        "worker_ip": w.get("ip", ""),
+                        "gpu_indexes": gpus,
+                        "rank_base": base,
+                    })
+                    base += len(gpus)
+                inst.distributed_servers = {
+                    "tp": tp,
+                    "master_ip": cand.worker.get("ip", ""),
+                    "master_port": 45000 + (inst.id % 1000),
+                    "subordinates": ranks,
+                }
             inst.state = ModelInstanceState.SCHEDULED.value
             inst.state_message = ""
             ar_update(s, inst)

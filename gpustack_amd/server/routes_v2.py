@@ -265,9 +265,17 @@ def list_instances(watch: bool = Query(False), worker_id: int | None = Query(Non
         rows = [i.to_dict() for i in q.all()]
     if worker_id is not None:
         # workers watch every instance event and filter locally on
-        # worker_id so SCHEDULED-assignment events reach them
+        # worker_id so SCHEDULED-assignment events reach them; the snapshot
+        # includes rows where this worker is a distributed subordinate
+        def _mine(r):
+            if r.get("worker_id") == worker_id:
+                return True
+            ds = r.get("distributed_servers") or {}
+            return any(x.get("worker_id") == worker_id
+                       for x in ds.get("subordinates", []))
+
         flt = None
-        rows = [r for r in rows if r.get("worker_id") == worker_id]
+        rows = [r for r in rows if _mine(r)]
     else:
         flt = None
     if watch:

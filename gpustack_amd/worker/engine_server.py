@@ -418,15 +418,16 @@ def create_app(runner: EngineRunner) -> FastAPI:
     return app
 
 
-def _spawn_followers(argv_base: list[str], tp: int, master_port: int):
-    """Rank 0 spawns follower rank processes (the first-party replacement
-    for the reference's multi-process executor bootstrap —
-    serve_manager.py:1685-1737 port bands + vLLM --headless followers)."""
+def _spawn_followers(argv_base: list[str], ranks: list[int], master_port: int):
+    """The per-worker parent spawns follower processes for its local ranks
+    (the first-party replacement for the reference's multi-process executor
+    bootstrap — serve_manager.py:1685-1737 port bands + vLLM --headless
+    followers / ranktables)."""
     import subprocess
     import sys
 
     procs = []
-    for r in range(1, tp):
+    for r in ranks:
         cmd = [sys.executable, "-m", "gpustack_amd.worker.engine_server",
                *argv_base, "--tp-rank", str(r), "--master-port", str(master_port)]
         procs.append(subprocess.Popen(cmd))
@@ -447,7 +448,14 @@ def main():
     ap.add_argument("--kv-cache-blocks", type=int, default=None)
     ap.add_argument("--backend-parameters", default="{}")
     ap.add_argument("--tp", type=int, default=1)
-    ap.add_argument("--tp-rank", type=int, default=0)
+    ap.add_argument("--tp-rank", type=int, default=None,
+                    help="absolute rank of THIS process (set for spawned "
+                         "followers; unset = per-worker parent)")
+    ap.add_argument("--rank-base", type=int, default=0,
+                    help="first rank hosted on this worker")
+    ap.add_argument("--local-ranks", type=int, default=None,
+                    help="ranks hosted on this worker (default: tp)")
+    ap.add_argument("--master-addr", default="127.0.0.1")
     ap.add_argument("--master-port", type=int, default=None)
     args = ap.parse_args()
 
@@ -461,16 +469,20 @@ def main():
     device = args.device or ("cuda" if use_cuda else "cpu")
     comm = None
     followers = []
+    if args.local_ranks is None:
+        args.local_ranks = args.tp if args.tp_rank is None else 1
     if args.tp > 1:
         from ..parallel import init_tp
 
-        if args.tp_rank == 0 and args.master_port is None:
-            import socket
+        if args.tp_rank is None:
+            # per-worker parent: hosts ranks [rank_base, rank_base+local)
+            if args.master_port is None:
+                import socket
 
-            s = socket.socket()
-            s.bind(("127.0.0.1", 0))
-            args.master_port = s.getsockname()[1]
-            s.close()
+                s = socket.socket()
+                s.bind(("127.0.0.1", 0))
+                args.master_port = s.getsockname()[1]
+                s.close()
             argv_base = [
                 "--served-name", args.served_name, "--source", args.source,
                 "--model-ref", args.model_ref, "--port", str(args.port),
@@ -479,17 +491,24 @@ def main():
                 "--gpu-memory-utilization", str(args.gpu_memory_utilization),
                 "--backend-parameters", args.backend_parameters,
                 "--tp", str(args.tp),
+                "--rank-base", str(args.rank_base),
+                "--master-addr", args.master_addr,
             ]
             if args.kv_cache_blocks:
                 argv_base += ["--kv-cache-blocks", str(args.kv_cache_blocks)]
             if args.device:
                 argv_base += ["--device", args.device]
-            followers = _spawn_followers(argv_base, args.tp, args.master_port)
+            follow_ranks = list(range(args.rank_base + 1,
+                                      args.rank_base + args.local_ranks))
+            followers = _spawn_followers(argv_base, follow_ranks, args.master_port)
+            args.tp_rank = args.rank_base
+        local_ordinal = args.tp_rank - args.rank_base
         if use_cuda:
-            device = f"cuda:{args.tp_rank}"
-            torch.cuda.set_device(args.tp_rank)
+            device = f"cuda:{local_ordinal}"
+            torch.cuda.set_device(local_ordinal)
         comm = init_tp(args.tp, args.tp_rank, master_port=args.master_port,
-                       device_id=args.tp_rank if use_cuda else None)
+                       device_id=local_ordinal if use_cuda else None,
+                       master_addr=args.master_addr)
 
     extra = json.loads(args.backend_parameters)
     cfg_kwargs = dict(
@@ -510,7 +529,7 @@ def main():
         ecfg.model_dir = args.model_ref
         ecfg.enforce_random_weights = False
 
-    if args.tp > 1 and args.tp_rank > 0:
+    if args.tp > 1 and args.tp_rank != 0:
         # follower rank: no HTTP; run the coordinated engine loop forever
         import time as _time
 

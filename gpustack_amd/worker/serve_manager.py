@@ -44,6 +44,7 @@ class InstanceProcess:
         self.log_path = log_path
         self.started_at = time.time()
         self.healthy = False
+        self.role = "main"
 
 
 class ServeManager:
@@ -69,11 +70,12 @@ class ServeManager:
                     data = frame.get("data", {})
                     if t == "HEARTBEAT":
                         continue
-                    if data.get("worker_id") != self.worker_id:
+                    role = self._role_for(data)
+                    if role is None:
                         if t == "DELETED" and data.get("id") in self.processes:
                             self._stop_instance(data["id"])
                         continue
-                    self.dispatch(t, data)
+                    self.dispatch(t, data, role)
             except Exception as e:  # noqa: BLE001
                 logger.warning("watch stream broken (%s); reconnecting", e)
                 time.sleep(3)
@@ -93,7 +95,19 @@ class ServeManager:
 
     # ---- event dispatch (serve_manager.py:968) ---------------------------
 
-    def dispatch(self, event_type: str, inst: dict) -> None:
+    def _role_for(self, inst: dict) -> str | None:
+        """main | sub | None — cross-worker TP instances involve this
+        worker either as rank-0 host or as a subordinate rank host
+        (reference: distributed_servers / coordinate modes)."""
+        if inst.get("worker_id") == self.worker_id:
+            return "main"
+        ds = inst.get("distributed_servers") or {}
+        for sub in ds.get("subordinates", []):
+            if sub.get("worker_id") == self.worker_id:
+                return "sub"
+        return None
+
+    def dispatch(self, event_type: str, inst: dict, role: str = "main") -> None:
         iid = inst["id"]
         state = inst.get("state")
         if event_type == "DELETED":
@@ -101,7 +115,7 @@ class ServeManager:
             return
         if state == S.SCHEDULED.value and iid not in self.processes:
             try:
-                self._start_instance(inst)
+                self._start_instance(inst, role)
             except Exception as e:  # noqa: BLE001
                 logger.exception("failed to start instance %s", inst.get("name"))
                 self._safe_update(iid, state=S.ERROR.value, state_message=str(e))
@@ -123,10 +137,11 @@ class ServeManager:
                 return p
         raise RuntimeError("no free ports in range")
 
-    def _start_instance(self, inst: dict) -> None:
+    def _start_instance(self, inst: dict, role: str = "main") -> None:
         iid = inst["id"]
         model = self.client.get_model(inst["model_id"])
-        self._safe_update(iid, state=S.INITIALIZING.value)
+        if role == "main":
+            self._safe_update(iid, state=S.INITIALIZING.value)
 
         source = model.get("source", "preset")
         ref = model["model_ref"]
@@ -146,7 +161,12 @@ class ServeManager:
         log_path = log_dir / f"{inst['name']}.log"
 
         env = dict(os.environ)
-        gpus = inst.get("gpu_indexes") or []
+        ds = inst.get("distributed_servers") or None
+        my_sub = None
+        if ds and role == "sub":
+            my_sub = next(x for x in ds["subordinates"]
+                          if x["worker_id"] == self.worker_id)
+        gpus = (my_sub["gpu_indexes"] if my_sub else inst.get("gpu_indexes")) or []
         if gpus:
             env["HIP_VISIBLE_DEVICES"] = ",".join(str(g) for g in gpus)
             env["CUDA_VISIBLE_DEVICES"] = env["HIP_VISIBLE_DEVICES"]
@@ -165,7 +185,14 @@ class ServeManager:
         ]
         if model.get("max_model_len"):
             args += ["--max-model-len", str(model["max_model_len"])]
-        if len(gpus) > 1:
+        if ds:
+            # cross-worker TP: rank group layout from the scheduler
+            args += ["--tp", str(ds["tp"]),
+                     "--local-ranks", str(len(gpus)),
+                     "--rank-base", str(my_sub["rank_base"] if my_sub else 0),
+                     "--master-addr", ds["master_ip"] or "127.0.0.1",
+                     "--master-port", str(ds["master_port"])]
+        elif len(gpus) > 1:
             # TP replica sharded over the scheduled GPUs (RCCL over xGMI)
             args += ["--tp", str(len(gpus))]
         if bp:
@@ -174,10 +201,13 @@ class ServeManager:
         logf = open(log_path, "ab")
         proc = subprocess.Popen(args, env=env, stdout=logf, stderr=subprocess.STDOUT,
                                 start_new_session=True)
-        self.processes[iid] = InstanceProcess(inst, proc, port, log_path)
-        self._safe_update(iid, state=S.STARTING.value, port=port, pid=proc.pid)
-        logger.info("instance %s starting: pid=%d port=%d gpus=%s",
-                    inst["name"], proc.pid, port, gpus)
+        ip = InstanceProcess(inst, proc, port, log_path)
+        ip.role = role
+        self.processes[iid] = ip
+        if role == "main":
+            self._safe_update(iid, state=S.STARTING.value, port=port, pid=proc.pid)
+        logger.info("instance %s starting (%s): pid=%d port=%d gpus=%s",
+                    inst["name"], role, proc.pid, port, gpus)
 
     def _download_model(self, iid: int, repo: str) -> str | None:
         self._safe_update(iid, state=S.DOWNLOADING.value)
@@ -220,6 +250,8 @@ class ServeManager:
             if rc is not None:
                 self._handle_exit(iid, ip, rc)
                 continue
+            if getattr(ip, "role", "main") == "sub":
+                continue  # subordinate rank hosts have no HTTP endpoint
             if not ip.healthy:
                 try:
                     r = httpx.get(f"http://127.0.0.1:{ip.port}/health", timeout=3.0)
@@ -238,6 +270,12 @@ class ServeManager:
         with self._lock:
             self._used_ports.discard(ip.port)
         inst = ip.instance
+        if getattr(ip, "role", "main") == "sub":
+            # a dead subordinate stalls the TP group; surface the error and
+            # leave recovery to delete/recreate (round-2: group restart)
+            self._safe_update(iid, state=S.ERROR.value,
+                              state_message=f"subordinate rank host exited {rc}")
+            return
         restart_count = (inst.get("restart_count") or 0) + 1
         model = None
         try:
