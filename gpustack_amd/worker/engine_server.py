@@ -519,7 +519,7 @@ def main():
         gpu_memory_utilization=args.gpu_memory_utilization,
         kv_cache_blocks=args.kv_cache_blocks,
         tp_size=args.tp,
-        tp_rank=args.tp_rank,
+        tp_rank=args.tp_rank or 0,
     )
     if device == "cpu" and args.kv_cache_blocks is None:
         cfg_kwargs["kv_cache_blocks"] = 1024
