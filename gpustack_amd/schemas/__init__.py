@@ -10,6 +10,7 @@ from .tables import (  # noqa: F401
     ModelFile,
     ModelInstance,
     ModelInstanceState,
+    ModelProvider,
     ModelRoute,
     ModelUsage,
     PlacementStrategy,
@@ -140,3 +141,11 @@ class BenchmarkCreate(BaseModel):
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64
+
+
+class ModelProviderCreate(BaseModel):
+    name: str
+    base_url: str
+    api_key: str = ""
+    models: list[str] | None = None
+    enabled: bool = True

@@ -181,6 +181,18 @@ class ModelRoute(Base, TimestampMixin, SerializeMixin):
     targets = Column(JSON, default=list)  # [{model_name, weight}]
 
 
+class ModelProvider(Base, TimestampMixin, SerializeMixin):
+    """External OpenAI-compatible providers routable through the gateway
+    (reference: schemas/model_provider.py)."""
+    __tablename__ = "model_providers"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False)
+    base_url = Column(String(512), nullable=False)
+    api_key = Column(String(512), default="")
+    models = Column(JSON, default=None)  # None = any model name
+    enabled = Column(Boolean, default=True)
+
+
 class Benchmark(Base, TimestampMixin, SerializeMixin):
     """In-product benchmark runs (reference: schemas/benchmark.py)."""
     __tablename__ = "benchmarks"

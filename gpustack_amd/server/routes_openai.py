@@ -19,7 +19,10 @@ from fastapi import APIRouter, Depends, HTTPException, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
 from ..db import get_session
-from ..schemas import Model, ModelInstance, ModelInstanceState, ModelRoute, ModelUsage, User
+from ..schemas import (
+    Model, ModelInstance, ModelInstanceState, ModelProvider, ModelRoute,
+    ModelUsage, User,
+)
 from .deps import get_current_user
 
 logger = logging.getLogger(__name__)
@@ -51,6 +54,16 @@ def _resolve_model_name(name: str) -> str:
     return name
 
 
+def _find_provider(model_name: str):
+    """External provider fallback (reference: schemas/model_provider.py —
+    providers routable through the same gateway)."""
+    with get_session() as s:
+        for p in s.query(ModelProvider).filter_by(enabled=True).all():
+            if p.models is None or model_name in p.models:
+                return p.to_dict()
+    return None
+
+
 def _pick_instance(model_name: str) -> tuple[Model, dict]:
     with get_session() as s:
         model = s.query(Model).filter_by(name=model_name).first()
@@ -77,7 +90,7 @@ def _record_usage(user: User, model: Model, usage: dict | None) -> None:
             date = time.strftime("%Y-%m-%d")
             row = (
                 s.query(ModelUsage)
-                .filter_by(user_id=user.id, model_id=model.id, date=date)
+                .filter_by(user_id=user.id, model_name=model.name, date=date)
                 .first()
             )
             if row is None:
@@ -104,12 +117,26 @@ async def _proxy(request: Request, path: str, user: User):
     if not name:
         raise HTTPException(400, "missing 'model'")
     target_name = _resolve_model_name(name)
-    model, inst = _pick_instance(target_name)
+    headers = {}
+    with get_session() as s:
+        local = s.query(Model).filter_by(name=target_name).first() is not None
+    if local:
+        model, inst = _pick_instance(target_name)
+        url = f"http://{inst['worker_ip']}:{inst['port']}{path}"
+    else:
+        provider = _find_provider(target_name)
+        if provider is None:
+            raise HTTPException(404, f"model {target_name!r} not found")
+        model = Model(id=0, name=f"{provider['name']}/{target_name}")
+        url = provider["base_url"].rstrip("/") + path.replace("/v1", "", 1) \
+            if provider["base_url"].rstrip("/").endswith("/v1") \
+            else provider["base_url"].rstrip("/") + path
+        if provider.get("api_key"):
+            headers["Authorization"] = f"Bearer {provider['api_key']}"
     body["model"] = target_name
-    url = f"http://{inst['worker_ip']}:{inst['port']}{path}"
     stream = bool(body.get("stream"))
     client = httpx.AsyncClient(timeout=PROXY_TIMEOUT)
-    req = client.build_request("POST", url, json=body)
+    req = client.build_request("POST", url, json=body, headers=headers)
     try:
         resp = await client.send(req, stream=stream)
     except httpx.ConnectError:
@@ -166,6 +193,11 @@ def list_models_v1(user: User = Depends(get_current_user)):
             {"id": r.name, "object": "model", "created": int(r.created_at),
              "owned_by": "gpustack_amd/route"}
             for r in routes
+        ] + [
+            {"id": m, "object": "model", "created": int(p.created_at),
+             "owned_by": f"provider/{p.name}"}
+            for p in s.query(ModelProvider).filter_by(enabled=True).all()
+            for m in (p.models or [])
         ]
     return {"object": "list", "data": items}
 

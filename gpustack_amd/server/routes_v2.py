@@ -16,9 +16,10 @@ from fastapi.responses import StreamingResponse
 from ..db import Event, EventType, ar_create, ar_delete, ar_update, bus, get_session
 from ..schemas import (
     ApiKey, ApiKeyCreate, Benchmark, BenchmarkCreate, Model, ModelCreate,
-    ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelRoute,
-    ModelRouteCreate, ModelUpdate, ModelUsage, RegistrationToken, SystemLoad,
-    User, UserCreate, Worker, WorkerRegister, WorkerState, WorkerStatusUpdate,
+    ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
+    ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
+    ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
+    WorkerRegister, WorkerState, WorkerStatusUpdate,
 )
 from ..security import generate_api_key, generate_registration_token, hash_password
 from .deps import get_admin_user, get_current_user, verify_worker_token
@@ -423,6 +424,35 @@ def delete_benchmark(bench_id: int, _: User = Depends(get_current_user)):
         if not b:
             raise HTTPException(404)
         ar_delete(s, b)
+        return {"ok": True}
+
+
+# ---- model providers (external OpenAI-compatible backends) -----------------
+
+@router.get("/model_providers")
+def list_providers(_: User = Depends(get_current_user)):
+    with get_session() as s:
+        return {"items": [p.to_dict() | {"api_key": "***" if p.api_key else ""}
+                          for p in s.query(ModelProvider).all()]}
+
+
+@router.post("/model_providers", status_code=201)
+def create_provider(body: ModelProviderCreate, _: User = Depends(get_admin_user)):
+    with get_session() as s:
+        if s.query(ModelProvider).filter_by(name=body.name).first():
+            raise HTTPException(409, "provider exists")
+        p = ModelProvider(**body.model_dump())
+        ar_create(s, p)
+        return p.to_dict() | {"api_key": "***" if p.api_key else ""}
+
+
+@router.delete("/model_providers/{provider_id}")
+def delete_provider(provider_id: int, _: User = Depends(get_admin_user)):
+    with get_session() as s:
+        p = s.get(ModelProvider, provider_id)
+        if not p:
+            raise HTTPException(404)
+        ar_delete(s, p)
         return {"ok": True}
 
 

@@ -230,3 +230,60 @@ def test_model_route_resolution(server):
 
     assert _resolve_model_name("public-name") == "backend-a"
     assert _resolve_model_name("backend-a") == "backend-a"
+
+
+def test_model_provider_routing(server):
+    """External provider fallback: unknown local model routed to a
+    registered OpenAI-compatible provider (stub) with its API key."""
+    import threading
+    import socket
+    import time as _t
+
+    import uvicorn
+    from fastapi import FastAPI as _F, Request as _R
+
+    client, app, cfg, reg = server
+    stub = _F()
+    seen = {}
+
+    @stub.post("/v1/chat/completions")
+    async def chat(request: _R):
+        seen["auth"] = request.headers.get("authorization")
+        body = await request.json()
+        return {"id": "x", "object": "chat.completion",
+                "choices": [{"index": 0, "message": {"role": "assistant",
+                                                     "content": "from-provider"},
+                             "finish_reason": "stop"}],
+                "usage": {"prompt_tokens": 3, "completion_tokens": 2,
+                          "total_tokens": 5}}
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0)); port = s.getsockname()[1]; s.close()
+    server_u = uvicorn.Server(uvicorn.Config(stub, host="127.0.0.1", port=port,
+                                             log_level="warning"))
+    threading.Thread(target=server_u.run, daemon=True).start()
+    import httpx as _h
+    for _ in range(100):
+        try:
+            _h.post(f"http://127.0.0.1:{port}/v1/chat/completions", json={}, timeout=1)
+            break
+        except _h.HTTPError:
+            _t.sleep(0.1)
+
+    r = client.post("/v2/model_providers", json={
+        "name": "ext", "base_url": f"http://127.0.0.1:{port}",
+        "api_key": "sk-secret", "models": ["gpt-x"]})
+    assert r.status_code == 201
+    assert r.json()["api_key"] == "***"  # never echoed
+
+    r = client.post("/v1/chat/completions", json={
+        "model": "gpt-x", "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 200, r.text
+    assert r.json()["choices"][0]["message"]["content"] == "from-provider"
+    assert seen["auth"] == "Bearer sk-secret"
+    # listed in /v1/models
+    assert any(m["id"] == "gpt-x" for m in client.get("/v1/models").json()["data"])
+    # usage metered under provider/model name
+    usage = client.get("/v2/usage").json()["items"]
+    assert any(u["model_name"] == "ext/gpt-x" and u["completion_tokens"] == 2
+               for u in usage)
+    server_u.should_exit = True
