@@ -28,6 +28,7 @@ class Config(BaseModel):
     port: int = 8080
     metrics_port: int = 10151
     # auth
+    ha_leases: bool = False                # force lease-based leader election
     bootstrap_password: str | None = None
     jwt_secret: str | None = None
     disable_auth: bool = False

@@ -25,6 +25,10 @@ class PlacementScheduler:
     def __init__(self, cfg: Config):
         self.cfg = cfg
         self._stop = False
+        self.coordinator = None  # set by the server; None = always leader
+
+    def _is_leader(self) -> bool:
+        return self.coordinator is None or self.coordinator.is_leader
 
     def stop(self):
         self._stop = True
@@ -36,10 +40,12 @@ class PlacementScheduler:
             try:
                 try:
                     ev = q.get(timeout=5.0)
-                    if ev.type == EventType.CREATED:
+                    if ev.type == EventType.CREATED and self._is_leader():
                         self.schedule_one(ev.data["id"])
                 except queue.Empty:
                     pass
+                if not self._is_leader():
+                    continue
                 if time.time() - last_scan > 15.0:
                     last_scan = time.time()
                     self.scan()

@@ -119,10 +119,23 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
         ModelController, ScalingScheduler, SystemLoadCollector, UsageArchiver,
         WorkerMonitor,
     )
+    from .coordinator import LeaseCoordinator, LocalCoordinator
+
+    # leader election only matters with a shared external DB; a lease over
+    # SQLite still works for tests / local HA pairs
+    if getattr(cfg, "ha_leases", False) or cfg.database_url:
+        coord = LeaseCoordinator()
+        coord.try_acquire()
+        coord.start()
+    else:
+        coord = LocalCoordinator()
+    app.state.coordinator = coord
 
     sched = PlacementScheduler(cfg)
     tasks = [sched, ModelController(cfg), WorkerMonitor(cfg),
              SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg)]
+    for t in tasks:
+        t.coordinator = coord  # leader-only gating (checked per cycle)
     app.state.scheduler = sched
     app.state.background_tasks = tasks
     threads = [
@@ -137,6 +150,9 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
 def stop_background_tasks(app: FastAPI) -> None:
     for t in getattr(app.state, "background_tasks", []):
         t.stop()
+    coord = getattr(app.state, "coordinator", None)
+    if coord is not None:
+        coord.stop()
 
 
 def run_server(cfg: Config) -> None:

@@ -25,7 +25,14 @@ HEARTBEAT_TIMEOUT = 60.0
 UNREACHABLE_TIMEOUT = 180.0
 
 
-class ModelController:
+class _LeaderGated:
+    coordinator = None  # set by the server; None = always leader
+
+    def _is_leader(self) -> bool:
+        return self.coordinator is None or self.coordinator.is_leader
+
+
+class ModelController(_LeaderGated):
     def __init__(self, cfg: Config):
         self.cfg = cfg
         self._stop = False
@@ -42,7 +49,7 @@ class ModelController:
                 ev = q.get(timeout=10.0)
                 if self._stop:
                     return
-                if ev.type in (EventType.CREATED, EventType.UPDATED):
+                if ev.type in (EventType.CREATED, EventType.UPDATED) and self._is_leader():
                     self.sync_replicas(ev.data["id"])
             except queue.Empty:
                 pass
@@ -105,7 +112,7 @@ class ModelController:
                     ar_delete(s, v)
 
 
-class WorkerMonitor:
+class WorkerMonitor(_LeaderGated):
     def __init__(self, cfg: Config):
         self.cfg = cfg
         self._stop = False
@@ -116,7 +123,8 @@ class WorkerMonitor:
     def run(self) -> None:
         while not self._stop:
             try:
-                self.check_once()
+                if self._is_leader():
+                    self.check_once()
             except Exception:  # noqa: BLE001
                 logger.exception("worker monitor cycle failed")
             for _ in range(15):
@@ -147,7 +155,7 @@ class WorkerMonitor:
                 ar_update(s, inst)
 
 
-class ScalingScheduler:
+class ScalingScheduler(_LeaderGated):
     """Cron-window desired-replica computation (reference:
     server/scaling_scheduler.py:19,96). While a window is active the
     model's instance count follows the rule's replicas; outside windows
@@ -178,6 +186,8 @@ class ScalingScheduler:
         mc = ModelController(self.cfg)
         while not self._stop:
             try:
+                if not self._is_leader():
+                    raise StopIteration  # skip cycle, keep sleeping
                 with get_session() as s:
                     models = [m for m in s.query(Model).all() if m.scaling_schedule]
                 for m in models:
@@ -186,6 +196,8 @@ class ScalingScheduler:
                         have = s.query(ModelInstance).filter_by(model_id=m.id).count()
                     if have != want:
                         mc.sync_replicas_to(m.id, want)
+            except StopIteration:
+                pass
             except Exception:  # noqa: BLE001
                 logger.exception("scaling scheduler cycle failed")
             for _ in range(int(self.interval)):
@@ -194,7 +206,7 @@ class ScalingScheduler:
                 time.sleep(1.0)
 
 
-class UsageArchiver:
+class UsageArchiver(_LeaderGated):
     """Hot -> archive mover for usage rows older than `keep_days`
     (reference: server/usage_archiver.py TableArchiver)."""
 
@@ -232,7 +244,7 @@ class UsageArchiver:
     def run(self) -> None:
         while not self._stop:
             try:
-                n = self.archive_once()
+                n = self.archive_once() if self._is_leader() else 0
                 if n:
                     logger.info("archived %d usage rows", n)
             except Exception:  # noqa: BLE001
@@ -243,7 +255,7 @@ class UsageArchiver:
                 time.sleep(1.0)
 
 
-class SystemLoadCollector:
+class SystemLoadCollector(_LeaderGated):
     def __init__(self, cfg: Config, interval: float = 60.0):
         self.cfg = cfg
         self.interval = interval
@@ -255,7 +267,8 @@ class SystemLoadCollector:
     def run(self) -> None:
         while not self._stop:
             try:
-                self.collect_once()
+                if self._is_leader():
+                    self.collect_once()
             except Exception:  # noqa: BLE001
                 logger.exception("system load collection failed")
             for _ in range(int(self.interval)):
