@@ -40,6 +40,7 @@ class Config(BaseModel):
     worker_name: str | None = None
     labels: dict[str, str] = Field(default_factory=dict)
     gpu_devices: list[dict] | None = None  # static override (air-gapped)
+    proxy_mode: str = "direct"             # direct | tunnel (NAT workers)
     system_reserved: dict = Field(default_factory=lambda: {"ram": 2 << 30, "vram": 1 << 30})
     # engine defaults
     gpu_memory_utilization: float = 0.9

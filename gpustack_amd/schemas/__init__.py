@@ -113,6 +113,7 @@ class WorkerRegister(BaseModel):
     status: dict = Field(default_factory=dict)
     system_reserved: dict = Field(default_factory=dict)
     token: str = ""
+    proxy_mode: str = "direct"
 
 
 class WorkerStatusUpdate(BaseModel):

@@ -109,6 +109,7 @@ class Worker(Base, TimestampMixin, SerializeMixin):
     system_reserved = Column(JSON, default=dict)
     heartbeat_time = Column(Float, default=0.0)
     unreachable = Column(Boolean, default=False)
+    proxy_mode = Column(String(16), default="direct")  # direct | tunnel
 
 
 class Model(Base, TimestampMixin, SerializeMixin):

@@ -21,7 +21,7 @@ from fastapi.responses import JSONResponse, StreamingResponse
 from ..db import get_session
 from ..schemas import (
     Model, ModelInstance, ModelInstanceState, ModelProvider, ModelRoute,
-    ModelUsage, User,
+    ModelUsage, User, Worker,
 )
 from .deps import get_current_user
 
@@ -120,8 +120,13 @@ async def _proxy(request: Request, path: str, user: User):
     headers = {}
     with get_session() as s:
         local = s.query(Model).filter_by(name=target_name).first() is not None
+    tunnel_worker = None
     if local:
         model, inst = _pick_instance(target_name)
+        with get_session() as s:
+            w = s.get(Worker, inst["worker_id"]) if inst.get("worker_id") else None
+            if w is not None and w.proxy_mode == "tunnel":
+                tunnel_worker = (w.id, inst["port"])
         url = f"http://{inst['worker_ip']}:{inst['port']}{path}"
     else:
         provider = _find_provider(target_name)
@@ -135,6 +140,8 @@ async def _proxy(request: Request, path: str, user: User):
             headers["Authorization"] = f"Bearer {provider['api_key']}"
     body["model"] = target_name
     stream = bool(body.get("stream"))
+    if tunnel_worker is not None:
+        return await _proxy_via_tunnel(tunnel_worker, path, body, user, model, stream)
     client = httpx.AsyncClient(timeout=PROXY_TIMEOUT)
     req = client.build_request("POST", url, json=body, headers=headers)
     try:
@@ -178,6 +185,51 @@ async def _proxy(request: Request, path: str, user: User):
 
     return StreamingResponse(relay(), media_type="text/event-stream",
                              status_code=resp.status_code)
+
+
+async def _proxy_via_tunnel(tw, path: str, body: dict, user: User,
+                            model: Model, stream: bool):
+    """Relay through the worker-initiated long-poll tunnel
+    (reference: websocket_proxy/proxy_server.py semantics)."""
+    from .tunnel import hub
+
+    worker_id, port = tw
+    payload = json.dumps(body).encode()
+    try:
+        status, ctype, chunks = await hub.request(
+            worker_id, port, "POST", path, payload,
+            {"content-type": "application/json"},
+        )
+    except TimeoutError:
+        raise HTTPException(502, "tunnel worker did not answer")
+    if not stream:
+        data = b"".join([c async for c in chunks])
+        try:
+            parsed = json.loads(data)
+            _record_usage(user, model, parsed.get("usage"))
+        except Exception:  # noqa: BLE001
+            parsed = {"raw": data.decode(errors="replace")}
+        return JSONResponse(parsed, status_code=status)
+
+    async def relay():
+        buf = b""
+        usage = None
+        async for c in chunks:
+            buf += c
+            yield c
+        for line in buf.decode(errors="replace").splitlines():
+            if line.startswith("data:"):
+                frag = line[5:].strip()
+                if frag and frag != "[DONE]":
+                    try:
+                        u = json.loads(frag).get("usage")
+                        if u:
+                            usage = u
+                    except json.JSONDecodeError:
+                        pass
+        _record_usage(user, model, usage)
+
+    return StreamingResponse(relay(), media_type=ctype, status_code=status)
 
 
 @router.get("/v1/models")

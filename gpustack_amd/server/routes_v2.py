@@ -153,6 +153,7 @@ def register_worker(body: WorkerRegister, request: Request,
         w.labels = body.labels
         w.status = body.status
         w.system_reserved = body.system_reserved
+        w.proxy_mode = body.proxy_mode or "direct"
         w.heartbeat_time = time.time()
         w.state = WorkerState.READY.value
         if w.id is None:
@@ -425,6 +426,36 @@ def delete_benchmark(bench_id: int, _: User = Depends(get_current_user)):
             raise HTTPException(404)
         ar_delete(s, b)
         return {"ok": True}
+
+
+# ---- tunnel proxy (NAT workers; reference websocket_proxy/) ----------------
+
+@router.get("/tunnel/jobs")
+async def tunnel_jobs(worker_id: int, _=Depends(verify_worker_token)):
+    from .tunnel import hub
+
+    job = await hub.next_job(worker_id)
+    if job is None:
+        from fastapi.responses import Response as _Resp
+
+        return _Resp(status_code=204)
+    return job
+
+
+@router.post("/tunnel/reply/{req_id}")
+async def tunnel_reply(req_id: str, request: Request, _=Depends(verify_worker_token)):
+    from .tunnel import hub
+
+    status = int(request.headers.get("x-tunnel-status", "502"))
+    ctype = request.headers.get("x-tunnel-content-type", "application/json")
+    q = await hub.begin_reply(req_id, status, ctype)
+    if q is None:
+        raise HTTPException(410, "request no longer waiting")
+    async for chunk in request.stream():
+        if chunk:
+            await hub.push_chunk(req_id, chunk)
+    await hub.end_reply(req_id)
+    return {"ok": True}
 
 
 # ---- model providers (external OpenAI-compatible backends) -----------------

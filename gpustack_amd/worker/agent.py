@@ -61,6 +61,7 @@ class WorkerAgent:
                     "labels": cfg.labels,
                     "status": status,
                     "system_reserved": cfg.system_reserved,
+                    "proxy_mode": cfg.proxy_mode,
                 })
                 self.worker_id = w["id"]
                 logger.info("registered as worker id=%s name=%s ip=%s", w["id"], name, ip)
@@ -88,6 +89,65 @@ class WorkerAgent:
             except Exception as e:  # noqa: BLE001
                 logger.warning("status sync failed: %s", e)
             time.sleep(self.cfg.worker_status_interval)
+
+    def tunnel_loop(self) -> None:
+        """NAT mode: long-poll the server for tunneled requests and stream
+        local engine responses back (reference: websocket_proxy/message_client)."""
+        import base64
+
+        import httpx as _h
+
+        headers = {"Authorization": f"Bearer {self.cfg.token}"}
+        while not self._stop:
+            try:
+                r = self.client._c.get("/v2/tunnel/jobs",
+                                       params={"worker_id": self.worker_id},
+                                       timeout=_h.Timeout(30.0, read=40.0))
+                if r.status_code != 200:
+                    continue
+                job = r.json()
+                threading.Thread(target=self._serve_tunnel_job, args=(job,),
+                                 daemon=True).start()
+            except Exception as e:  # noqa: BLE001
+                logger.debug("tunnel poll error: %s", e)
+                time.sleep(1)
+
+    def _serve_tunnel_job(self, job: dict) -> None:
+        import base64
+
+        import httpx as _h
+
+        body = base64.b64decode(job.get("body_b64", ""))
+        url = f"http://127.0.0.1:{job['port']}{job['path']}"
+        try:
+            with _h.stream(job.get("method", "POST"), url, content=body,
+                           headers=job.get("headers") or {},
+                           timeout=_h.Timeout(30.0, read=None)) as resp:
+                def gen():
+                    for chunk in resp.iter_bytes():
+                        yield chunk
+
+                self.client._c.post(
+                    f"/v2/tunnel/reply/{job['id']}",
+                    content=gen(),
+                    headers={
+                        "X-Tunnel-Status": str(resp.status_code),
+                        "X-Tunnel-Content-Type": resp.headers.get(
+                            "content-type", "application/json"),
+                    },
+                    timeout=_h.Timeout(30.0, read=120.0),
+                )
+        except Exception as e:  # noqa: BLE001
+            logger.warning("tunnel job %s failed: %s", job.get("id"), e)
+            try:
+                self.client._c.post(
+                    f"/v2/tunnel/reply/{job['id']}",
+                    content=b'{"error": {"message": "tunnel upstream failed"}}',
+                    headers={"X-Tunnel-Status": "502",
+                             "X-Tunnel-Content-Type": "application/json"},
+                )
+            except Exception:  # noqa: BLE001
+                pass
 
     def create_api(self) -> FastAPI:
         app = FastAPI(title="gpustack_amd-worker")
@@ -171,6 +231,9 @@ class WorkerAgent:
             threading.Thread(target=self.serve_manager.health_loop, name="serve-health", daemon=True),
             threading.Thread(target=self.benchmark_manager.poll_loop, name="benchmarks", daemon=True),
         ]
+        if self.cfg.proxy_mode == "tunnel":
+            threads.append(threading.Thread(target=self.tunnel_loop,
+                                            name="tunnel", daemon=True))
         for t in threads:
             t.start()
         import uvicorn
