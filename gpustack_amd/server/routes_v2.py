@@ -487,6 +487,24 @@ def delete_provider(provider_id: int, _: User = Depends(get_admin_user)):
         return {"ok": True}
 
 
+# ---- catalog / version -----------------------------------------------------
+
+@router.get("/catalog")
+def catalog(_: User = Depends(get_current_user)):
+    """Built-in model catalog (reference: server/catalog.py)."""
+    import json as _json
+    from pathlib import Path as _P
+
+    return _json.loads((_P(__file__).parent / "catalog.json").read_text())
+
+
+@router.get("/version")
+def version():
+    from .. import __version__
+
+    return {"version": __version__}
+
+
 # ---- usage / system load ---------------------------------------------------
 
 @router.get("/usage")

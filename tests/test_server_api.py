@@ -287,3 +287,10 @@ def test_model_provider_routing(server):
     assert any(u["model_name"] == "ext/gpt-x" and u["completion_tokens"] == 2
                for u in usage)
     server_u.should_exit = True
+
+
+def test_catalog_and_version(server):
+    client, app, cfg, reg = server
+    cat = client.get("/v2/catalog").json()
+    assert any(m["model_ref"] == "llama-3-8b" for m in cat["models"])
+    assert "version" in client.get("/v2/version").json()
