@@ -35,6 +35,8 @@ def parse_args():
     ap.add_argument("--isl", type=int, default=256, help="synthetic prompt length")
     ap.add_argument("--osl", type=int, default=128, help="max output tokens per request")
     ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"],
+                    help="KV cache dtype (fp8 e4m3 halves KV bytes; compute stays bf16)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree: world becomes ONE replica "
@@ -77,6 +79,7 @@ def main():
         max_model_len=args.max_model_len,
         max_num_seqs=max(args.concurrency, 8),
         seed=0,
+        kv_cache_dtype=args.kv_dtype,
         tp_size=tp if comm else 1,
         tp_rank=rank if comm else 0,
     )

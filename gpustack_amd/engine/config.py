@@ -117,6 +117,7 @@ PRESETS: dict[str, ModelSpec] = {
 class EngineConfig:
     model: str = "llama-3-8b"              # preset name or model dir
     dtype: str = "bfloat16"
+    kv_cache_dtype: str = "bf16"           # bf16 | fp8 (e4m3, halves KV bytes)
     block_size: int = 16
     gpu_memory_utilization: float = 0.90
     max_num_seqs: int = 256

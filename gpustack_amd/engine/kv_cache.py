@@ -48,7 +48,11 @@ class KVCache:
         self.head_dim = spec.head_dim
         self.num_layers = spec.num_layers
         shape = (num_blocks, kv_heads, cfg.block_size, spec.head_dim)
-        dtype = getattr(torch, cfg.dtype)
+        if cfg.kv_cache_dtype == "fp8":
+            dtype = torch.float8_e4m3fn
+        else:
+            dtype = getattr(torch, cfg.dtype)
+        self.kv_dtype = dtype
         self.k_caches = [
             torch.zeros(shape, dtype=dtype, device=device) for _ in range(spec.num_layers)
         ]
@@ -75,7 +79,7 @@ class KVCache:
             self.host_pool = torch.zeros(
                 (host_blocks, spec.num_layers, 2, kv_heads, cfg.block_size,
                  spec.head_dim),
-                dtype=dtype, pin_memory=self.is_cuda,
+                dtype=self.kv_dtype, pin_memory=self.is_cuda,
             )
             self.host_allocator = BlockAllocator(host_blocks)
             self.side_stream = torch.cuda.Stream() if self.is_cuda else None
@@ -125,8 +129,9 @@ class KVCache:
     def compute_num_blocks(cfg: EngineConfig, free_bytes: int) -> int:
         spec = cfg.spec
         kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
+        esize = 1 if cfg.kv_cache_dtype == "fp8" else 2
         per_block = (
-            2 * spec.num_layers * kv_heads * cfg.block_size * spec.head_dim * 2
+            2 * spec.num_layers * kv_heads * cfg.block_size * spec.head_dim * esize
         )
         return max(1, int(free_bytes * cfg.gpu_memory_utilization) // per_block)
 

@@ -16,18 +16,20 @@
 // KV pool layout: [num_blocks, Hkv, BS, D] bf16, BS = 16 tokens.
 #include "common.h"
 
+#include <type_traits>
+
 namespace {
 
 constexpr int BS = 16;       // tokens per KV block (page)
 constexpr int NWAVES = 4;    // waves per workgroup
 constexpr int THREADS = NWAVES * WAVE_SIZE;
 
-template <int D, int GQ>
+template <int D, int GQ, bool FP8>
 __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,        // [N, Hq, D]
     const unsigned short* __restrict__ q,    // [N, Hq, D]
-    const unsigned short* __restrict__ kc,   // [B, Hkv, BS, D]
-    const unsigned short* __restrict__ vc,   // [B, Hkv, BS, D]
+    const void* __restrict__ kc,             // [B, Hkv, BS, D] bf16|fp8
+    const void* __restrict__ vc,             // [B, Hkv, BS, D]
     const int* __restrict__ block_tables,    // [N, max_blocks]
     const int* __restrict__ seq_lens,        // [N]
     int Hkv, int max_blocks, float scale, long q_stride) {
@@ -71,11 +73,12 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     for (int j = 0; j < DV; ++j) acc[gq][j] = 0.f;
   }
 
+  using KVT = std::conditional_t<FP8, unsigned char, unsigned short>;
   const int* bt = block_tables + (long)seq * max_blocks;
   for (int page = wave; page < npages; page += NWAVES) {
     const long blk = bt[page];
-    const unsigned short* kbase = kc + ((blk * Hkv + h) * BS) * D;
-    const unsigned short* vbase = vc + ((blk * Hkv + h) * BS) * D;
+    const KVT* kbase = (const KVT*)kc + ((blk * Hkv + h) * BS) * D;
+    const KVT* vbase = (const KVT*)vc + ((blk * Hkv + h) * BS) * D;
     // Whole page per wave-iteration: the group's 4 tokens are processed
     // together — 4 independent dots (ILP across the shuffle-reduce
     // chains), then ONE softmax update per gq for all 4 (cuts the
@@ -83,14 +86,16 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
     for (int tb = 0; tb < BS / 4; tb += TB) {
       const int base_tok = page * BS + tb * 4 + g;
-      u16x8 vu[TB];
+      using VecT = std::conditional_t<FP8, u8x8, u16x8>;
+      VecT vu[TB];
       float kf[TB][DV];  // K converted once, reused across all GQ heads
 #pragma unroll
       for (int it = 0; it < TB; ++it) {
         const int tok = (tb + it) * 4 + g;
-        u16x8 ku = *reinterpret_cast<const u16x8*>(kbase + tok * D + sub * DV);
-        vu[it] = *reinterpret_cast<const u16x8*>(vbase + tok * D + sub * DV);
-        bf8_to_f32(ku, kf[it]);
+        VecT ku = *reinterpret_cast<const VecT*>(kbase + tok * D + sub * DV);
+        vu[it] = *reinterpret_cast<const VecT*>(vbase + tok * D + sub * DV);
+        if constexpr (FP8) fp8x8_to_f32(ku, kf[it]);
+        else bf8_to_f32(ku, kf[it]);
       }
 #pragma unroll
       for (int gq = 0; gq < GQ; ++gq) {
@@ -127,7 +132,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         for (int j = 0; j < DV; ++j) {
           float a = acc[gq][j] * corr;
 #pragma unroll
-          for (int it = 0; it < TB; ++it) a += p[it] * bf2f(vu[it][j]);
+          for (int it = 0; it < TB; ++it)
+            a += p[it] * (FP8 ? fp8_to_f32(vu[it][j]) : bf2f(vu[it][j]));
           acc[gq][j] = a;
         }
         m[gq] = nm;
@@ -205,17 +211,21 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
                               const void* vc, const int* block_tables,
                               const int* seq_lens, int N, int Hq, int Hkv,
                               int D, int max_blocks, float scale, long q_stride,
-                              int* err_unsupported, hipStream_t s) {
+                              int fp8, int* err_unsupported, hipStream_t s) {
   const int GQ = Hq / Hkv;
   dim3 grid(N, Hkv);
   dim3 block(THREADS);
   *err_unsupported = 0;
   if (D != 128) { *err_unsupported = 1; return; }
-#define LAUNCH_GQ(G)                                                         \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<128, G>), grid, block, 0, s,  \
-                     (unsigned short*)out, (const unsigned short*)q,         \
-                     (const unsigned short*)kc, (const unsigned short*)vc,   \
+#define LAUNCH_GQ2(G, F)                                                       \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<128, G, F>), grid, block, 0, s, \
+                     (unsigned short*)out, (const unsigned short*)q, kc, vc,   \
                      block_tables, seq_lens, Hkv, max_blocks, scale, q_stride)
+#define LAUNCH_GQ(G)                                                          \
+  do {                                                                        \
+    if (fp8) LAUNCH_GQ2(G, true);                                             \
+    else LAUNCH_GQ2(G, false);                                                \
+  } while (0)
   switch (GQ) {
     case 1: LAUNCH_GQ(1); break;
     case 2: LAUNCH_GQ(2); break;
@@ -227,4 +237,5 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
     default: *err_unsupported = 1; return;
   }
 #undef LAUNCH_GQ
+#undef LAUNCH_GQ2
 }

@@ -9,9 +9,9 @@ void rms_norm_launch(void*, const void*, const void*, float, int, int, hipStream
 void fused_add_rms_norm_launch(void*, void*, const void*, float, int, int, hipStream_t);
 void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, int, int, long, long, hipStream_t);
 void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
-void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, hipStream_t);
+void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, int, hipStream_t);
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
-void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int*, hipStream_t);
+void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, int*, hipStream_t);
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
@@ -33,6 +33,17 @@ hipStream_t cur_stream(const at::Tensor& t) {
 void check_bf16(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+bool is_fp8_cache(const at::Tensor& t) {
+  return t.scalar_type() == at::kFloat8_e4m3fn || t.scalar_type() == at::kByte;
+}
+
+void check_cache(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 || is_fp8_cache(t),
+              name, " must be bf16 or fp8-e4m3");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
@@ -99,14 +110,15 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slots) {
   const long ks = row_stride_3d(k, "k");
   const long vs = row_stride_3d(v, "v");
-  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  check_cache(k_cache, "k_cache"); check_cache(v_cache, "v_cache");
   TORCH_CHECK(slots.scalar_type() == at::kLong && slots.is_contiguous());
   const int T = k.size(0), Hkv = k.size(1), D = k.size(2);
   const int BS = k_cache.size(2);
   TORCH_CHECK(k_cache.size(1) == Hkv && k_cache.size(3) == D && D % 8 == 0);
   reshape_and_cache_launch(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
                            v_cache.data_ptr(), slots.data_ptr<long>(), T, Hkv,
-                           D, BS, ks, vs, cur_stream(k));
+                           D, BS, ks, vs, is_fp8_cache(k_cache) ? 1 : 0,
+                           cur_stream(k));
   HIP_CHECK_LAST();
 }
 
@@ -124,7 +136,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor seq_lens, double scale) {
   check_bf16(out, "out");
   const long qstride = row_stride_3d(q, "q");
-  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  check_cache(k_cache, "k_cache"); check_cache(v_cache, "v_cache");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt && block_tables.is_contiguous());
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
   const int N = q.size(0), Hq = q.size(1), D = q.size(2);
@@ -136,7 +148,8 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   paged_attn_decode_launch(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                            v_cache.data_ptr(), block_tables.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), N, Hq, Hkv, D, max_blocks,
-                           (float)scale, qstride, &err, cur_stream(q));
+                           (float)scale, qstride,
+                           is_fp8_cache(k_cache) ? 1 : 0, &err, cur_stream(q));
   TORCH_CHECK(!err, "paged_attn_decode: unsupported head_dim/GQ combination: D=",
               D, " Hq=", Hq, " Hkv=", Hkv);
   HIP_CHECK_LAST();

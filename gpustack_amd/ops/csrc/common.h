@@ -6,6 +6,7 @@
 //  - memory-bound kernels vectorize loads as 16 B/lane (u16x8 for bf16).
 #pragma once
 
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
@@ -34,6 +35,25 @@ __device__ __forceinline__ unsigned short f2bf(float f) {
   if ((x & 0x7fffffffu) > 0x7f800000u) return 0x7fc0; // NaN
   unsigned int round = 0x7fffu + ((x >> 16) & 1u);
   return (unsigned short)((x + round) >> 16);
+}
+
+typedef __attribute__((ext_vector_type(8))) unsigned char u8x8;
+
+// OCP fp8 e4m3 (gfx950-native v_cvt_f32_fp8 / v_cvt_pk_fp8_f32)
+__device__ __forceinline__ float fp8_to_f32(unsigned char u) {
+  __hip_fp8_e4m3 v;
+  v.__x = u;
+  return (float)v;
+}
+
+__device__ __forceinline__ unsigned char f32_to_fp8(float f) {
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
+}
+
+__device__ __forceinline__ void fp8x8_to_f32(const u8x8 v, float* out) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) out[i] = fp8_to_f32(v[i]);
 }
 
 __device__ __forceinline__ void bf8_to_f32(const u16x8 v, float* out) {

@@ -192,6 +192,38 @@ __global__ void silu_and_mul_kernel(unsigned short* __restrict__ out,
 // k/v: [T, Hkv, D] bf16 ; caches: [num_blocks, Hkv, BS, D] ; slot[t] = global
 // slot index (block*BS + offset), -1 = skip.
 // ---------------------------------------------------------------------------
+__global__ void reshape_and_cache_fp8_kernel(const unsigned short* __restrict__ k,
+                                             const unsigned short* __restrict__ v,
+                                             unsigned char* __restrict__ kc,
+                                             unsigned char* __restrict__ vc,
+                                             const long* __restrict__ slots,
+                                             int T, int Hkv, int D, int BS,
+                                             long ks, long vs) {
+  const int chunks = D / 8;
+  const long total = (long)T * Hkv * chunks;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int c = idx % chunks;
+    const int h = (idx / chunks) % Hkv;
+    const int t = idx / ((long)chunks * Hkv);
+    const long slot = slots[t];
+    if (slot < 0) continue;
+    const long blk = slot / BS, off = slot % BS;
+    const long dst = ((blk * Hkv + h) * BS + off) * D + c * 8;
+    const long hoff = (long)h * D + c * 8;
+    u16x8 kv8 = *reinterpret_cast<const u16x8*>(k + (long)t * ks + hoff);
+    u16x8 vv8 = *reinterpret_cast<const u16x8*>(v + (long)t * vs + hoff);
+    u8x8 ko, vo;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      ko[i] = f32_to_fp8(bf2f(kv8[i]));
+      vo[i] = f32_to_fp8(bf2f(vv8[i]));
+    }
+    *reinterpret_cast<u8x8*>(kc + dst) = ko;
+    *reinterpret_cast<u8x8*>(vc + dst) = vo;
+  }
+}
+
 __global__ void reshape_and_cache_kernel(const unsigned short* __restrict__ k,
                                          const unsigned short* __restrict__ v,
                                          unsigned short* __restrict__ kc,
@@ -312,12 +344,19 @@ void silu_and_mul_launch(void* out, const void* x, long T, int I,
 
 void reshape_and_cache_launch(const void* k, const void* v, void* kc, void* vc,
                               const long* slots, int T, int Hkv, int D, int BS,
-                              long ks, long vs, hipStream_t s) {
+                              long ks, long vs, int fp8, hipStream_t s) {
   long total = (long)T * Hkv * (D / 8);
-  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(pw_grid(total, 256)),
-                     dim3(256), 0, s, (const unsigned short*)k,
-                     (const unsigned short*)v, (unsigned short*)kc,
-                     (unsigned short*)vc, slots, T, Hkv, D, BS, ks, vs);
+  if (fp8) {
+    hipLaunchKernelGGL(reshape_and_cache_fp8_kernel, dim3(pw_grid(total, 256)),
+                       dim3(256), 0, s, (const unsigned short*)k,
+                       (const unsigned short*)v, (unsigned char*)kc,
+                       (unsigned char*)vc, slots, T, Hkv, D, BS, ks, vs);
+  } else {
+    hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(pw_grid(total, 256)),
+                       dim3(256), 0, s, (const unsigned short*)k,
+                       (const unsigned short*)v, (unsigned short*)kc,
+                       (unsigned short*)vc, slots, T, Hkv, D, BS, ks, vs);
+  }
 }
 
 void greedy_sample_launch(long* out, const void* logits, int N, int V,

@@ -102,8 +102,8 @@ def reshape_and_cache(
     idx = slots[mask]
     blk = torch.div(idx, bs, rounding_mode="floor")
     off = idx % bs
-    k_cache[blk, :, off] = k[mask]
-    v_cache[blk, :, off] = v[mask]
+    k_cache[blk, :, off] = k[mask].to(k_cache.dtype)
+    v_cache[blk, :, off] = v[mask].to(v_cache.dtype)
 
 
 def greedy_sample(out: torch.Tensor, logits: torch.Tensor) -> None:
@@ -128,10 +128,10 @@ def paged_attn_decode(
         L = int(seq_lens[i])
         nblk = (L + BS - 1) // BS
         blocks = block_tables[i, :nblk].long()
-        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
-        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
-        keys = keys.repeat_interleave(GQ, dim=0).float()  # [Hq, L, D]
-        vals = vals.repeat_interleave(GQ, dim=0).float()
+        keys = k_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        vals = v_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        keys = keys.repeat_interleave(GQ, dim=0)  # [Hq, L, D]
+        vals = vals.repeat_interleave(GQ, dim=0)
         qi = q[i].float().unsqueeze(1)  # [Hq, 1, D]
         att = torch.softmax((qi @ keys.transpose(1, 2)) * scale, dim=-1)
         out[i] = (att @ vals).squeeze(1).to(out.dtype)

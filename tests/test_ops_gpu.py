@@ -242,3 +242,42 @@ def test_skinny_gemm(M, N, K, splitk, version):
     assert torch.allclose(got.float(), ref, atol=0.5, rtol=3e-2), (
         (got.float() - ref).abs().max().item()
     )
+
+
+@pytest.mark.parametrize("lens", [[17, 5, 160], [2048]])
+def test_paged_attn_decode_fp8(lens):
+    Hq, Hkv, D, BS = 32, 8, 128, 16
+    N = len(lens)
+    maxb = (max(lens) + BS - 1) // BS
+    nb = sum((l + BS - 1) // BS for l in lens) + 1
+    kc = (torch.randn(nb, Hkv, BS, D, device="cuda") * 2).to(torch.float8_e4m3fn)
+    vc = (torch.randn(nb, Hkv, BS, D, device="cuda") * 2).to(torch.float8_e4m3fn)
+    bt = torch.zeros(N, maxb, dtype=torch.int32, device="cuda")
+    nxt = 0
+    for i, l in enumerate(lens):
+        n = (l + BS - 1) // BS
+        bt[i, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    q = torch.randn(N, Hq, D, dtype=torch.bfloat16, device="cuda")
+    sl = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    ref = torch.empty_like(q)
+    scale = 1 / math.sqrt(D)
+    ops.paged_attn_decode(out, q, kc, vc, bt, sl, scale)
+    R.paged_attn_decode(ref, q, kc, vc, bt, sl, scale)
+    _close(out, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_reshape_and_cache_fp8():
+    T, Hkv, D, BS, B = 33, 8, 128, 16, 8
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(B, Hkv, BS, D, device="cuda").to(torch.float8_e4m3fn)
+    vc = torch.zeros(B, Hkv, BS, D, device="cuda").to(torch.float8_e4m3fn)
+    kc2, vc2 = kc.clone(), vc.clone()
+    slots = torch.randperm(B * BS, device="cuda")[:T]
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    R.reshape_and_cache(k, v, kc2, vc2, slots)
+    # HW cvt and torch cvt are both OCP e4m3 RNE: bit-identical expected
+    assert torch.equal(kc.view(torch.uint8), kc2.view(torch.uint8))
+    assert torch.equal(vc.view(torch.uint8), vc2.view(torch.uint8))
