@@ -40,6 +40,10 @@ def model_spec_for(model: Model | dict) -> ModelSpec | None:
     if source == "preset":
         return PRESETS.get(ref)
     try:
+        if str(ref).endswith(".gguf"):
+            from ..utils.gguf import spec_from_gguf
+
+            return spec_from_gguf(ref)
         return ModelSpec.from_dir(ref)
     except Exception:  # noqa: BLE001
         return None

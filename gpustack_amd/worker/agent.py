@@ -176,9 +176,39 @@ class WorkerAgent:
         @app.get("/files/model-weight-size")
         def model_weight_size(path: str):
             total = 0
-            for f in Path(path).glob("*.safetensors"):
-                total += f.stat().st_size
+            p = Path(path)
+            pats = ("*.safetensors",) if p.is_dir() else ()
+            if p.is_file():
+                total = p.stat().st_size
+            else:
+                for pat in pats + ("*.gguf",):
+                    for f in p.glob(pat):
+                        total += f.stat().st_size
             return {"weight_size_bytes": total}
+
+        @app.get("/files/parse-gguf")
+        def parse_gguf(path: str):
+            """First-party in-process GGUF header parse (reference runs the
+            gguf-parser Go binary remotely via the worker filesystem API,
+            routes/worker/filesystem.py:105-309 + scheduler/calculator.py:
+            622-682 — we parse natively, utils/gguf.py)."""
+            from ..utils.gguf import read_gguf
+
+            p = Path(path)
+            if not p.is_file():
+                raise HTTPException(404, "gguf file not found")
+            try:
+                info = read_gguf(p)
+            except ValueError as e:
+                raise HTTPException(422, str(e))
+            meta = {k: v for k, v in info.metadata.items()
+                    if not isinstance(v, list)}
+            return {"version": info.version,
+                    "architecture": info.architecture,
+                    "n_params": info.n_params,
+                    "weight_bytes": info.weight_bytes,
+                    "n_tensors": len(info.tensors),
+                    "metadata": meta}
 
         @app.get("/logs/{instance_name}")
         def logs(instance_name: str, tail: int = 200):
