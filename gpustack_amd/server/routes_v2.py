@@ -160,6 +160,7 @@ def register_worker(body: WorkerRegister, request: Request,
             ar_create(s, w)
         else:
             ar_update(s, w)
+        get_eval_cache().invalidate()  # placement landscape changed
         return w.to_dict()
 
 
@@ -198,6 +199,7 @@ def delete_worker(worker_id: int, _: User = Depends(get_admin_user)):
         if not w:
             raise HTTPException(404)
         ar_delete(s, w)
+        get_eval_cache().invalidate()
         return {"ok": True}
 
 
@@ -338,6 +340,22 @@ def delete_route(route_id: int, _: User = Depends(get_current_user)):
 
 # ---- model evaluations (compatibility pre-check) ---------------------------
 
+# evaluation results are deterministic for a given (model spec, worker set);
+# cache them briefly (reference: scheduler/evaluator.py TTL result cache) and
+# invalidate cluster-wide when workers change (see worker routes).
+_eval_cache = None
+
+
+def get_eval_cache():
+    global _eval_cache
+    if _eval_cache is None:
+        from .cache import TTLCache
+
+        _eval_cache = TTLCache(ttl=30.0, maxsize=256,
+                               distributed_name="model_evaluations")
+    return _eval_cache
+
+
 @router.post("/model-evaluations")
 def evaluate_model(body: ModelCreate, _: User = Depends(get_current_user)):
     """Dry-run of the placement pipeline (reference: scheduler/evaluator.py):
@@ -345,6 +363,13 @@ def evaluate_model(body: ModelCreate, _: User = Depends(get_current_user)):
     from ..scheduler.policies import (
         estimate_vram_claim, model_spec_for, pick_candidate,
     )
+
+    cache = get_eval_cache()
+    ck = (body.model_ref, body.source, body.gpus_per_replica,
+          body.gpu_memory_utilization)
+    hit = cache.get(ck)
+    if hit is not None:
+        return hit
 
     model_d = body.model_dump() | {"id": -1}
     spec = model_spec_for(model_d)
@@ -358,18 +383,22 @@ def evaluate_model(body: ModelCreate, _: User = Depends(get_current_user)):
         insts = [i.to_dict() for i in s.query(ModelInstance).all()]
     cand = pick_candidate(model_d, workers, insts)
     if cand is None:
-        return {
+        result = {
             "compatible": False,
             "estimated_vram_per_gpu": claim,
             "messages": ["no worker currently fits the resource claim"],
         }
-    return {
+        cache.set(ck, result)
+        return result
+    result = {
         "compatible": True,
         "estimated_vram_per_gpu": claim,
         "candidate": {"worker": cand.worker.get("name"),
                       "gpu_indexes": cand.gpu_indexes},
         "messages": [],
     }
+    cache.set(ck, result)
+    return result
 
 
 # ---- benchmarks ------------------------------------------------------------
