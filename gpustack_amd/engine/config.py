@@ -104,6 +104,13 @@ PRESETS: dict[str, ModelSpec] = {
         intermediate_size=18944, num_layers=28, num_heads=28, num_kv_heads=4,
         rope_theta=1000000.0, attention_bias=True, eos_token_id=151645,
     ),
+    # Mistral v0.3: llama-compatible compute graph (no SWA since v0.1;
+    # GQA 32/8, theta 1e6) — runs on the same CDNA4 kernel set
+    "mistral-7b": ModelSpec(
+        architecture="MistralForCausalLM", vocab_size=32768, hidden_size=4096,
+        intermediate_size=14336, num_layers=32, num_heads=32, num_kv_heads=8,
+        rope_theta=1000000.0, max_position_embeddings=32768,
+    ),
     # tiny CPU-testable model (OPT-125m-scale plumbing per BASELINE.json cfg 1)
     "tiny": ModelSpec(
         vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,

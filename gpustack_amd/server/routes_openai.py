@@ -267,3 +267,8 @@ async def completions(request: Request, user: User = Depends(get_current_user)):
 @router.post("/v1/embeddings")
 async def embeddings(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/embeddings", user)
+
+
+@router.post("/v1/rerank")
+async def rerank(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/rerank", user)

@@ -397,6 +397,46 @@ def create_app(runner: EngineRunner) -> FastAPI:
                       "total_tokens": sum(len(p) for p in prompts)},
         }
 
+    @app.post("/v1/rerank")
+    @app.post("/rerank")
+    async def rerank(request: Request):
+        """Query-document relevance ranking (reference serves rerank models
+        via vox-box/vLLM backends, routes/openai_compatible: /v1/rerank;
+        first-party design: score = cosine similarity of last-token pooled
+        embeddings from the same engine — no second model process)."""
+        body = await request.json()
+        query = body.get("query", "")
+        docs = body.get("documents", [])
+        top_n = int(body.get("top_n", len(docs)) or len(docs))
+        tok = runner.tokenizer
+
+        def enc(text):
+            ids = tok.encode(text)
+            if hasattr(ids, "ids"):
+                ids = ids.ids
+            return list(ids) or [0]
+
+        prompts = [enc(query)] + [enc(d) for d in docs]
+        vecs = await runner.run_aux(runner.engine.runner.embed, prompts, "mean")
+        import math
+
+        def cos(a, b):
+            num = sum(x * y for x, y in zip(a, b))
+            den = math.sqrt(sum(x * x for x in a)) * math.sqrt(sum(y * y for y in b))
+            return num / den if den else 0.0
+
+        qv = vecs[0]
+        scored = [{"index": i, "relevance_score": cos(qv, v),
+                   "document": {"text": docs[i]}}
+                  for i, v in enumerate(vecs[1:])]
+        scored.sort(key=lambda r: -r["relevance_score"])
+        return {
+            "model": runner.served_name,
+            "results": scored[:top_n],
+            "usage": {"prompt_tokens": sum(len(p) for p in prompts),
+                      "total_tokens": sum(len(p) for p in prompts)},
+        }
+
     @app.post("/v1/completions")
     async def completions(request: Request):
         body = await request.json()

@@ -169,3 +169,19 @@ def test_logprobs():
     while eng2.has_unfinished():
         outs2 += eng2.step()
     assert all(o.logprob is None for o in outs2)
+
+
+def test_mistral_family():
+    """MistralForCausalLM maps onto the llama-compatible compute graph."""
+    from gpustack_amd.engine.config import PRESETS, ModelSpec
+
+    spec = PRESETS["mistral-7b"]
+    assert spec.architecture == "MistralForCausalLM"
+    assert not spec.attention_bias and not spec.qk_norm
+    hf = ModelSpec.from_hf_config({
+        "architectures": ["MistralForCausalLM"], "vocab_size": 32768,
+        "hidden_size": 4096, "intermediate_size": 14336,
+        "num_hidden_layers": 32, "num_attention_heads": 32,
+        "num_key_value_heads": 8, "rope_theta": 1000000.0,
+    })
+    assert hf.num_kv_heads == 8 and not hf.attention_bias

@@ -95,6 +95,15 @@ def test_engine_server_stop_strings():
             "model": "tiny-s", "input": ["abc", "def"]}, timeout=60)
         assert r.status_code == 200
         assert len(r.json()["data"]) == 2
+        # rerank: identical doc must out-score an unrelated one
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/rerank", json={
+            "model": "tiny-s", "query": "abcabc",
+            "documents": ["abcabc", "zzqqwwx"], "top_n": 2}, timeout=60)
+        assert r.status_code == 200
+        res = r.json()["results"]
+        assert len(res) == 2
+        assert res[0]["index"] == 0  # self-similarity ranks first
+        assert res[0]["relevance_score"] >= res[1]["relevance_score"]
     finally:
         proc.terminate()
         try:
