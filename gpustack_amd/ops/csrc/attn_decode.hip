@@ -51,17 +51,26 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   const int g = lane >> 4;         // token-group within wave [0,4)
   const int sub = lane & 15;       // dim-slice within group [0,16)
 
-  // Load q rows for this kv head's GQ query heads (f32, pre-scaled so the
-  // per-token dot needs no multiply).
-  float qr[GQ][DV];
+  // Load q rows for this kv head's GQ query heads. bf16 path: keep q PACKED
+  // (u16x8 = 4 VGPRs/head vs 8 for f32) and dot with the native
+  // v_dot2c_f32_bf16 instruction — halves both q/k register pressure and
+  // dot-product instruction count (occupancy was the bottleneck: 2-3
+  // waves/SIMD at 168-200 VGPRs, profiles/r02_optimization_log.md).
+  // fp8 path: f32 q pre-scaled as before (k dequants through f32 anyway).
+  float qr[FP8 ? GQ : 1][DV];
+  u16x8 qb[FP8 ? 1 : GQ];
 #pragma unroll
   for (int gq = 0; gq < GQ; ++gq) {
     const unsigned short* qp =
         q + (long)seq * q_stride + ((long)h * GQ + gq) * D + sub * DV;
     u16x8 u = *reinterpret_cast<const u16x8*>(qp);
-    bf8_to_f32(u, qr[gq]);
+    if constexpr (FP8) {
+      bf8_to_f32(u, qr[gq]);
 #pragma unroll
-    for (int j = 0; j < DV; ++j) qr[gq][j] *= scale;
+      for (int j = 0; j < DV; ++j) qr[gq][j] *= scale;
+    } else {
+      qb[gq] = u;
+    }
   }
 
   float m[GQ], s[GQ], acc[GQ][DV];
@@ -88,14 +97,14 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
       const int base_tok = page * BS + tb * 4 + g;
       using VecT = std::conditional_t<FP8, u8x8, u16x8>;
       VecT vu[TB];
-      float kf[TB][DV];  // K converted once, reused across all GQ heads
+      VecT ku[TB];              // bf16: K stays packed (dot2 consumes it)
+      float kf[FP8 ? TB : 1][DV];  // fp8: K dequanted once, reused per gq
 #pragma unroll
       for (int it = 0; it < TB; ++it) {
         const int tok = (tb + it) * 4 + g;
-        VecT ku = *reinterpret_cast<const VecT*>(kbase + tok * D + sub * DV);
+        ku[it] = *reinterpret_cast<const VecT*>(kbase + tok * D + sub * DV);
         vu[it] = *reinterpret_cast<const VecT*>(vbase + tok * D + sub * DV);
-        if constexpr (FP8) fp8x8_to_f32(ku, kf[it]);
-        else bf8_to_f32(ku, kf[it]);
+        if constexpr (FP8) fp8x8_to_f32(ku[it], kf[it]);
       }
 #pragma unroll
       for (int gq = 0; gq < GQ; ++gq) {
@@ -103,8 +112,17 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
         for (int it = 0; it < TB; ++it) {
           float d = 0.f;
+          if constexpr (FP8) {
 #pragma unroll
-          for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[it][j];
+            for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[it][j];
+          } else {
+            const bf16x2* qa = reinterpret_cast<const bf16x2*>(&qb[gq]);
+            const bf16x2* ka = reinterpret_cast<const bf16x2*>(&ku[it]);
+#pragma unroll
+            for (int j = 0; j < DV / 2; ++j)
+              d = __builtin_amdgcn_fdot2_f32_bf16(qa[j], ka[j], d, false);
+            d *= scale;  // q not pre-scaled on this path (kept packed)
+          }
           dot[it] = d;
         }
 #pragma unroll
