@@ -82,7 +82,9 @@ class LLMEngine:
             # before the abort can release blocks for reuse
             self._carry_outputs.extend(self._drain())
         ok = self.scheduler.abort(request_id)
-        self.seqs.pop(request_id, None)
+        seq = self.seqs.pop(request_id, None)
+        if seq is not None and self.runner.eagle is not None:
+            self.runner.eagle.drop(seq)
         return ok
 
     def _sync_tp_ops(self) -> None:
@@ -100,7 +102,9 @@ class LLMEngine:
             elif op[0] == "abort":
                 rid = op[1]
                 self.scheduler.abort(rid)
-                self.seqs.pop(rid, None)
+                seq = self.seqs.pop(rid, None)
+                if seq is not None and self.runner.eagle is not None:
+                    self.runner.eagle.drop(seq)
 
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()
@@ -227,6 +231,7 @@ class LLMEngine:
         if batch.is_prefill:
             self.scheduler.on_prefill_done(batch)
         rps = 1 if batch.is_prefill else batch.rows_per_seq
+        emitted_all: list[list[int]] = []
         for i, seq in enumerate(batch.seqs):
             if rps == 1:
                 emitted = [token_ids[i]]
@@ -241,6 +246,7 @@ class LLMEngine:
                     emitted = accept_tokens(batch.drafts[i][:ke], rows[:ke + 1])
                 else:
                     emitted = [rows[0]]
+            emitted_all.append(emitted)
             seq.record_first_token()
             for j, tok in enumerate(emitted):
                 seq.output_token_ids.append(tok)
@@ -253,7 +259,26 @@ class LLMEngine:
                     self.scheduler.finish_seq(seq, reason)
                     self.seqs.pop(seq.request_id, None)
                     break
+        self._eagle_post_step(batch, token_ids, emitted_all)
         return outputs
+
+    def _eagle_post_step(self, batch, token_ids, emitted_all) -> None:
+        """Draft-model speculative: extend draft KV + propose the next
+        window from the hiddens the verify step produced."""
+        eagle = self.runner.eagle
+        if eagle is None:
+            return
+        alive = [s.status != SeqStatus.FINISHED for s in batch.seqs]
+        hidden = self.runner.last_hidden
+        if batch.is_prefill:
+            n_pre = batch.n_prefill_seqs or len(batch.seqs)
+            if n_pre == len(batch.seqs) and hidden is not None:
+                eagle.seed_from_prefill(batch, hidden, token_ids, alive)
+        elif batch.rows_per_seq > 1 and hidden is not None:
+            eagle.step(batch, hidden, emitted_all, alive)
+        for ok, seq in zip(alive, batch.seqs):
+            if not ok:
+                eagle.drop(seq)
 
     def _finish_reason(self, seq: Sequence, tok: int) -> str | None:
         p = seq.params

@@ -139,10 +139,20 @@ class ModelRunner:
         )
         nblocks = min(nblocks, max_needed)
         self.kv = KVCache(cfg, nblocks, self.device)
+        self.eagle = None
+        spec = getattr(cfg, "speculative", None)
+        if spec and spec.get("method") in ("eagle", "eagle3"):
+            from .eagle import EagleProposer
+
+            k = int(spec.get("num_draft_tokens", 3))
+            draft_blocks = min(nblocks, cfg.max_num_seqs * (
+                (cfg.max_model_len + k + cfg.block_size) // cfg.block_size + 1))
+            self.eagle = EagleProposer(cfg, self.model, self.comm,
+                                       self.device, draft_blocks)
         if self.device.type == "cuda":
             from .graph_runner import DecodeGraphRunner, graphs_enabled
 
-            if graphs_enabled():
+            if graphs_enabled() and self.eagle is None:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv
@@ -274,8 +284,21 @@ class ModelRunner:
 
     @torch.inference_mode()
     def execute(self, batch: ScheduledBatch) -> list[int]:
+        self.last_hidden = None
         if self.graph_runner is not None and self.graph_runner.can_run(batch):
             logits = self.graph_runner.run(batch)
+        elif self.eagle is not None:
+            # draft-model speculative: the verify step must also surface the
+            # target hidden states that condition the next draft window
+            tokens, meta = self._meta(batch)
+            if batch.is_prefill:
+                hidden_all = self.model(tokens, meta, self.kv, return_hidden=True)
+                logits = torch.nn.functional.linear(
+                    hidden_all[meta.logits_indices], self.model.lm_head)
+                self.last_hidden = hidden_all
+            else:
+                logits, self.last_hidden = self.model(tokens, meta, self.kv,
+                                                      return_both=True)
         else:
             tokens, meta = self._meta(batch)
             logits = self.model(tokens, meta, self.kv)

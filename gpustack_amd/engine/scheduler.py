@@ -54,16 +54,25 @@ class Scheduler:
         self.swapped: deque[Sequence] = deque()  # offloaded to host DRAM
         self.proposer = None
         self.spec_k = 0
+        self.spec_method = None
         spec = getattr(cfg, "speculative", None)
-        if spec and spec.get("method", "ngram") == "ngram":
+        method = spec.get("method", "ngram") if spec else None
+        if method == "ngram":
             from .spec import NgramProposer
 
+            self.spec_method = method
             self.spec_k = int(spec.get("num_draft_tokens", 3))
             self.proposer = NgramProposer(
                 self.spec_k,
                 int(spec.get("ngram_max", 3)),
                 int(spec.get("ngram_min", 1)),
             )
+        elif method in ("eagle", "eagle3"):
+            # draft-model speculative: proposals are computed on-GPU by the
+            # runner's EagleProposer each step (engine/eagle.py) and arrive
+            # via seq.next_draft
+            self.spec_method = method
+            self.spec_k = int(spec.get("num_draft_tokens", 3))
 
     # -- queue ops ---------------------------------------------------------
     def add(self, seq: Sequence) -> None:
@@ -274,8 +283,11 @@ class Scheduler:
                 lens.append(pos + 1)
                 continue
             k_eff = max(0, min(self.spec_k, self.cfg.max_model_len - 1 - pos))
-            if seq.params.greedy and k_eff > 0:
+            if seq.params.greedy and k_eff > 0 and self.proposer is not None:
                 draft = self.proposer.propose(seq)
+            elif seq.params.greedy and k_eff > 0 and seq.next_draft:
+                draft = (list(seq.next_draft) + [last] * self.spec_k)[:self.spec_k]
+                seq.next_draft = None
             else:
                 draft = [last] * self.spec_k
                 k_eff = 0

@@ -40,6 +40,7 @@ class Sequence:
     num_cached_tokens: int = 0      # tokens whose KV is already in the pool
     swap_outs: int = 0
     pending_tokens: int = 0         # async decode: sampled on device, not yet read back
+    next_draft: list | None = None  # draft-model speculative window (engine/eagle.py)
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None
     finish_time: float | None = None

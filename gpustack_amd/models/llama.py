@@ -173,7 +173,7 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.inference_mode()
     def forward(self, token_ids: torch.Tensor, meta: ForwardMeta, kv,
-                return_hidden: bool = False) -> torch.Tensor:
+                return_hidden: bool = False, return_both: bool = False):
         x = F.embedding(token_ids, self.embed)
         residual = None
         for i, layer in enumerate(self.layers):
@@ -182,4 +182,7 @@ class LlamaForCausalLM(nn.Module):
         if return_hidden:
             return x  # all rows, post final norm (embedding serving)
         hidden = x[meta.logits_indices]
-        return F.linear(hidden, self.lm_head)
+        logits = F.linear(hidden, self.lm_head)
+        if return_both:  # draft-model speculative needs the features too
+            return logits, hidden
+        return logits

@@ -79,3 +79,17 @@ def test_embed_gpu():
     assert eng.scheduler.kv.allocator.num_free == eng.scheduler.kv.allocator.num_blocks
     del eng
     torch.cuda.empty_cache()
+
+
+def test_eagle_matches_plain_gpu():
+    # draft-model speculative on the HIP path: identical output to plain
+    eng = LLMEngine(_cfg())
+    plain = eng.generate(PROMPTS, SamplingParams(max_tokens=10, ignore_eos=True))
+    del eng
+    torch.cuda.empty_cache()
+    eng2 = LLMEngine(_cfg(speculative={"method": "eagle", "num_draft_tokens": 3}))
+    assert eng2.runner.eagle is not None
+    out = eng2.generate(PROMPTS, SamplingParams(max_tokens=10, ignore_eos=True))
+    del eng2
+    torch.cuda.empty_cache()
+    assert out == plain
