@@ -35,6 +35,9 @@ def parse_args():
     ap.add_argument("--isl", type=int, default=256, help="synthetic prompt length")
     ap.add_argument("--osl", type=int, default=128, help="max output tokens per request")
     ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--speculative", default=None, choices=["ngram", "eagle", "eagle3"],
+                    help="speculative decoding method (BASELINE config 5)")
+    ap.add_argument("--draft-tokens", type=int, default=3)
     ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"],
                     help="KV cache dtype (fp8 e4m3 halves KV bytes; compute stays bf16)")
     ap.add_argument("--device", default=None)
@@ -80,6 +83,9 @@ def main():
         max_num_seqs=max(args.concurrency, 8),
         seed=0,
         kv_cache_dtype=args.kv_dtype,
+        speculative=({"method": args.speculative,
+                      "num_draft_tokens": args.draft_tokens}
+                     if args.speculative else None),
         tp_size=tp if comm else 1,
         tp_rank=rank if comm else 0,
     )
