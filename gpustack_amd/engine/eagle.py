@@ -68,9 +68,20 @@ class EagleProposer:
         self.states: dict[str, _DraftState] = {}
 
         h = spec.hidden_size
-        self.fc_w = torch.empty(h, 2 * h, dtype=dtype, device=device)
-        self.layer = DecoderLayer(spec, cfg.tp_size, comm, dtype).to(device)
-        self.norm = torch.empty(h, dtype=dtype, device=device)
+        # "eagle"/"eagle3": ONE draft head chained k times.
+        # "mtp": k chained heads with distinct weights, head m predicting
+        # draft position m (DeepSeek-style multi-token prediction modules,
+        # sharing one draft KV — an engineering simplification vs per-head
+        # KV streams; greedy acceptance keeps outputs exact regardless).
+        method = cfg.speculative.get("method", "eagle")
+        self.n_heads = self.k if method == "mtp" else 1
+        self.fc_ws: list[torch.Tensor] = []
+        self.layers: list[DecoderLayer] = []
+        self.norms: list[torch.Tensor] = []
+        for _ in range(self.n_heads):
+            self.fc_ws.append(torch.empty(h, 2 * h, dtype=dtype, device=device))
+            self.layers.append(DecoderLayer(spec, cfg.tp_size, comm, dtype).to(device))
+            self.norms.append(torch.empty(h, dtype=dtype, device=device))
         self._init_weights()
 
     def _init_weights(self) -> None:
@@ -83,34 +94,37 @@ class EagleProposer:
         d = spec.head_dim
         hq, hkv = spec.num_heads // tp, max(1, spec.num_kv_heads // tp)
         i_loc = spec.intermediate_size // tp
-        dt, dev, seed = self.fc_w.dtype, self.device, cfg.seed
-        self.fc_w.copy_(_gen((spec.hidden_size, 2 * spec.hidden_size),
-                             "eagle.fc", seed, dt, dev))
-        self.norm.fill_(1.0)
-        la = self.layer.attn
-        q = _gen((spec.num_heads * d, spec.hidden_size), "eagle.q", seed, dt, dev)
-        k = _gen((spec.num_kv_heads * d, spec.hidden_size), "eagle.k", seed, dt, dev)
-        v = _gen((spec.num_kv_heads * d, spec.hidden_size), "eagle.v", seed, dt, dev)
-        la.qkv_w.copy_(torch.cat([
-            q[rank * hq * d:(rank + 1) * hq * d],
-            k[rank * hkv * d:(rank + 1) * hkv * d],
-            v[rank * hkv * d:(rank + 1) * hkv * d]]))
-        if la.qkv_b is not None:
-            la.qkv_b.zero_()
-        o = _gen((spec.hidden_size, spec.num_heads * d), "eagle.o", seed, dt, dev)
-        la.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
-        if spec.qk_norm:
-            la.q_norm.fill_(1.0)
-            la.k_norm.fill_(1.0)
-        gate = _gen((spec.intermediate_size, spec.hidden_size), "eagle.gate", seed, dt, dev)
-        up = _gen((spec.intermediate_size, spec.hidden_size), "eagle.up", seed, dt, dev)
-        self.layer.mlp.gate_up_w.copy_(torch.cat([
-            gate[rank * i_loc:(rank + 1) * i_loc],
-            up[rank * i_loc:(rank + 1) * i_loc]]))
-        down = _gen((spec.hidden_size, spec.intermediate_size), "eagle.down", seed, dt, dev)
-        self.layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
-        self.layer.input_norm.fill_(1.0)
-        self.layer.post_attn_norm.fill_(1.0)
+        dt, dev, seed = self.fc_ws[0].dtype, self.device, cfg.seed
+        for hi in range(self.n_heads):
+            pfx = f"draft{hi}"
+            self.fc_ws[hi].copy_(_gen((spec.hidden_size, 2 * spec.hidden_size),
+                                      f"{pfx}.fc", seed, dt, dev))
+            self.norms[hi].fill_(1.0)
+            layer = self.layers[hi]
+            la = layer.attn
+            q = _gen((spec.num_heads * d, spec.hidden_size), f"{pfx}.q", seed, dt, dev)
+            k = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{pfx}.k", seed, dt, dev)
+            v = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{pfx}.v", seed, dt, dev)
+            la.qkv_w.copy_(torch.cat([
+                q[rank * hq * d:(rank + 1) * hq * d],
+                k[rank * hkv * d:(rank + 1) * hkv * d],
+                v[rank * hkv * d:(rank + 1) * hkv * d]]))
+            if la.qkv_b is not None:
+                la.qkv_b.zero_()
+            o = _gen((spec.hidden_size, spec.num_heads * d), f"{pfx}.o", seed, dt, dev)
+            la.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
+            if spec.qk_norm:
+                la.q_norm.fill_(1.0)
+                la.k_norm.fill_(1.0)
+            gate = _gen((spec.intermediate_size, spec.hidden_size), f"{pfx}.gate", seed, dt, dev)
+            up = _gen((spec.intermediate_size, spec.hidden_size), f"{pfx}.up", seed, dt, dev)
+            layer.mlp.gate_up_w.copy_(torch.cat([
+                gate[rank * i_loc:(rank + 1) * i_loc],
+                up[rank * i_loc:(rank + 1) * i_loc]]))
+            down = _gen((spec.hidden_size, spec.intermediate_size), f"{pfx}.down", seed, dt, dev)
+            layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
+            layer.input_norm.fill_(1.0)
+            layer.post_attn_norm.fill_(1.0)
 
     # -- draft-KV bookkeeping ---------------------------------------------
 
@@ -135,11 +149,12 @@ class EagleProposer:
     # -- draft forward -----------------------------------------------------
 
     @torch.inference_mode()
-    def _forward(self, tokens, h_prev, positions, slots, seq_lens, block_tables):
+    def _forward(self, tokens, h_prev, positions, slots, seq_lens, block_tables,
+                 head: int = 0):
         """One draft pass in spec-row decode mode; returns draft hidden."""
         dev = self.device
         e = F.embedding(tokens, self.model.embed)
-        x = F.linear(torch.cat([e, h_prev], dim=-1), self.fc_w)
+        x = F.linear(torch.cat([e, h_prev], dim=-1), self.fc_ws[head])
         max_pos = self.cfg.max_model_len - 1
         meta = ForwardMeta(
             is_prefill=False,
@@ -149,16 +164,16 @@ class EagleProposer:
             block_tables=block_tables,
             seq_lens=seq_lens,
         )
-        x, residual = self.layer(x, None, meta, self.model.cos_sin,
-                                 self.k_cache, self.v_cache)
-        ops.fused_add_rms_norm(x, residual, self.norm, self.spec.rms_norm_eps)
+        x, residual = self.layers[head](x, None, meta, self.model.cos_sin,
+                                        self.k_cache, self.v_cache)
+        ops.fused_add_rms_norm(x, residual, self.norms[head], self.spec.rms_norm_eps)
         return x
 
     @torch.inference_mode()
     def _forward_prefill(self, tokens, h_prev, positions, slots, lens):
         dev = self.device
         e = F.embedding(tokens, self.model.embed)
-        x = F.linear(torch.cat([e, h_prev], dim=-1), self.fc_w)
+        x = F.linear(torch.cat([e, h_prev], dim=-1), self.fc_ws[0])
         tiles = ops.build_prefill_tiles(lens, dev)
         meta = ForwardMeta(
             is_prefill=True,
@@ -168,9 +183,9 @@ class EagleProposer:
             seq_lens_list=lens,
             tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
         )
-        x, residual = self.layer(x, None, meta, self.model.cos_sin,
-                                 self.k_cache, self.v_cache)
-        ops.fused_add_rms_norm(x, residual, self.norm, self.spec.rms_norm_eps)
+        x, residual = self.layers[0](x, None, meta, self.model.cos_sin,
+                                     self.k_cache, self.v_cache)
+        ops.fused_add_rms_norm(x, residual, self.norms[0], self.spec.rms_norm_eps)
         return x
 
     def _argmax_token(self, hidden) -> torch.Tensor:
@@ -216,7 +231,9 @@ class EagleProposer:
                  for i in range(n)], dtype=torch.long, device=dev)
             seq_lens = torch.tensor([feed_pos[i] + m + 1 for i in range(n)],
                                     dtype=torch.int32, device=dev)
-            h_chain = self._forward(tok, h_chain, positions, slots, seq_lens, bt)
+            head = min(len(drafts), self.n_heads - 1)  # MTP: head per position
+            h_chain = self._forward(tok, h_chain, positions, slots, seq_lens,
+                                    bt, head=head)
             tok = self._argmax_token(h_chain)
             drafts.append(tok)
             m += 1

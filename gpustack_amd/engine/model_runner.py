@@ -141,7 +141,7 @@ class ModelRunner:
         self.kv = KVCache(cfg, nblocks, self.device)
         self.eagle = None
         spec = getattr(cfg, "speculative", None)
-        if spec and spec.get("method") in ("eagle", "eagle3"):
+        if spec and spec.get("method") in ("eagle", "eagle3", "mtp"):
             from .eagle import EagleProposer
 
             k = int(spec.get("num_draft_tokens", 3))

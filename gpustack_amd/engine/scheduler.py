@@ -67,7 +67,7 @@ class Scheduler:
                 int(spec.get("ngram_max", 3)),
                 int(spec.get("ngram_min", 1)),
             )
-        elif method in ("eagle", "eagle3"):
+        elif method in ("eagle", "eagle3", "mtp"):
             # draft-model speculative: proposals are computed on-GPU by the
             # runner's EagleProposer each step (engine/eagle.py) and arrive
             # via seq.next_draft

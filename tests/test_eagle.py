@@ -80,3 +80,11 @@ def test_eagle_survives_preemption():
     eng = LLMEngine(cfg)
     out = eng.generate(prompts, p)
     assert out == plain
+
+
+def test_mtp_matches_plain():
+    # MTP: k chained draft heads with distinct weights
+    _, plain = _gen(None)
+    eng, out = _gen({"method": "mtp", "num_draft_tokens": 3})
+    assert eng.runner.eagle.n_heads == 3
+    assert out == plain
