@@ -40,6 +40,7 @@ OPENAI_PATHS = [
     "/v1/embeddings",
     "/v1/rerank",
     "/v1/score",
+    "/v1/messages",
 ]
 
 
@@ -272,3 +273,10 @@ async def embeddings(request: Request, user: User = Depends(get_current_user)):
 @router.post("/v1/rerank")
 async def rerank(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/rerank", user)
+
+
+@router.post("/v1/messages")
+async def anthropic_messages(request: Request, user: User = Depends(get_current_user)):
+    """Anthropic-style Messages API, proxied to the placed instance
+    (reference gateway routes /v1/messages: gateway/__init__.py:70-75)."""
+    return await _proxy(request, "/v1/messages", user)

@@ -370,6 +370,137 @@ def create_app(runner: EngineRunner) -> FastAPI:
             ids = ids.ids
         return await _generate(request, body, list(ids), "chat")
 
+    @app.post("/v1/messages")
+    async def anthropic_messages(request: Request):
+        """Anthropic-style Messages API (reference routes /v1/messages
+        through its gateway: gateway/utils.py anthropic_model_exact +
+        gateway/__init__.py supported_anthropic_routes)."""
+        body = await request.json()
+        messages = body.get("messages") or []
+        system = body.get("system")
+        norm = []
+        if system:
+            if isinstance(system, list):  # content-block form
+                system = "".join(b.get("text", "") for b in system
+                                 if isinstance(b, dict))
+            norm.append({"role": "system", "content": system})
+        for m in messages:
+            c = m.get("content", "")
+            if isinstance(c, list):
+                c = "".join(b.get("text", "") for b in c
+                            if isinstance(b, dict) and b.get("type") == "text")
+            norm.append({"role": m.get("role", "user"), "content": c})
+        tok = runner.tokenizer
+        if hasattr(tok, "apply_chat_template"):
+            try:
+                prompt = tok.apply_chat_template(norm, tokenize=False,
+                                                 add_generation_prompt=True)
+            except TypeError:
+                prompt = tok.apply_chat_template(norm)
+        else:
+            prompt = "\n".join(f"{m['role']}: {m['content']}" for m in norm)
+        ids = tok.encode(prompt)
+        if hasattr(ids, "ids"):
+            ids = ids.ids
+        ids = list(ids)
+        oa = {
+            "max_tokens": body.get("max_tokens", 1024),
+            "temperature": body.get("temperature", 1.0),
+            "top_p": body.get("top_p", 1.0),
+            "stop": body.get("stop_sequences"),
+        }
+        params = _sampling_params(oa, runner.engine.cfg.spec.eos_token_id)
+        stop_strs = _stop_strings(oa)
+        rid, q = runner.submit(ids, params)
+        model_name = runner.served_name
+        msg_id = f"msg_{rid}"
+
+        def stop_reason(fin: str) -> str:
+            return {"length": "max_tokens", "stop": "end_turn"}.get(fin, "end_turn")
+
+        if body.get("stream"):
+            async def gen():
+                tokens: list[int] = []
+                sent_len = 0
+                fin = "end_turn"
+                try:
+                    start = {"type": "message_start", "message": {
+                        "id": msg_id, "type": "message", "role": "assistant",
+                        "model": model_name, "content": [],
+                        "usage": {"input_tokens": len(ids), "output_tokens": 0}}}
+                    yield f"event: message_start\ndata: {json.dumps(start)}\n\n"
+                    cbs = {"type": "content_block_start", "index": 0,
+                           "content_block": {"type": "text", "text": ""}}
+                    yield f"event: content_block_start\ndata: {json.dumps(cbs)}\n\n"
+                    while True:
+                        if await request.is_disconnected():
+                            runner.abort(rid)
+                            return
+                        item = await q.get()
+                        if "error" in item:
+                            err = {"type": "error",
+                                   "error": {"type": "api_error",
+                                             "message": item["error"]}}
+                            yield f"event: error\ndata: {json.dumps(err)}\n\n"
+                            return
+                        tokens.append(item["token_id"])
+                        text = runner.tokenizer.decode(tokens)
+                        new = text[sent_len:]
+                        if new and not new.endswith("\ufffd"):
+                            sent_len = len(text)
+                            d = {"type": "content_block_delta", "index": 0,
+                                 "delta": {"type": "text_delta", "text": new}}
+                            yield f"event: content_block_delta\ndata: {json.dumps(d)}\n\n"
+                        if item["finished"]:
+                            fin = stop_reason(item.get("finish_reason") or "stop")
+                            break
+                    yield ('event: content_block_stop\ndata: '
+                           + json.dumps({"type": "content_block_stop", "index": 0})
+                           + "\n\n")
+                    md = {"type": "message_delta",
+                          "delta": {"stop_reason": fin, "stop_sequence": None},
+                          "usage": {"output_tokens": len(tokens)}}
+                    yield f"event: message_delta\ndata: {json.dumps(md)}\n\n"
+                    yield ('event: message_stop\ndata: '
+                           + json.dumps({"type": "message_stop"}) + "\n\n")
+                finally:
+                    runner.release(rid)
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        tokens = []
+        fin = "stop"
+        try:
+            while True:
+                item = await q.get()
+                if "error" in item:
+                    raise HTTPException(500, item["error"])
+                tokens.append(item["token_id"])
+                if stop_strs:
+                    t = runner.tokenizer.decode(tokens)
+                    if any(ss in t for ss in stop_strs):
+                        runner.abort(rid)
+                        fin = "stop_sequence"
+                        break
+                if item["finished"]:
+                    fin = item.get("finish_reason") or "stop"
+                    break
+        finally:
+            runner.release(rid)
+        text = runner.tokenizer.decode(tokens)
+        if stop_strs:
+            cuts = [text.find(ss) for ss in stop_strs if ss in text]
+            if cuts:
+                text = text[:min(cuts)]
+        sr = ("stop_sequence" if fin == "stop_sequence" else stop_reason(fin))
+        return JSONResponse({
+            "id": msg_id, "type": "message", "role": "assistant",
+            "model": model_name,
+            "content": [{"type": "text", "text": text}],
+            "stop_reason": sr, "stop_sequence": None,
+            "usage": {"input_tokens": len(ids), "output_tokens": len(tokens)},
+        })
+
     @app.post("/v1/embeddings")
     async def embeddings(request: Request):
         body = await request.json()

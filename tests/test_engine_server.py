@@ -110,3 +110,44 @@ def test_engine_server_stop_strings():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_anthropic_messages_endpoint():
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-a", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        _wait_health(port, proc)
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/messages", json={
+            "model": "tiny-a", "max_tokens": 8,
+            "system": "be brief",
+            "messages": [{"role": "user", "content": "hello"}],
+        }, timeout=60)
+        assert r.status_code == 200
+        j = r.json()
+        assert j["type"] == "message" and j["role"] == "assistant"
+        assert j["content"][0]["type"] == "text"
+        assert j["stop_reason"] in ("end_turn", "max_tokens")
+        assert j["usage"]["output_tokens"] > 0
+        # streaming: anthropic event sequence
+        with httpx.stream("POST", f"http://127.0.0.1:{port}/v1/messages", json={
+            "model": "tiny-a", "max_tokens": 6, "stream": True,
+            "messages": [{"role": "user",
+                          "content": [{"type": "text", "text": "hi"}]}],
+        }, timeout=60) as r:
+            events = [ln.split(" ", 1)[1] for ln in r.iter_lines()
+                      if ln.startswith("event: ")]
+        assert events[0] == "message_start"
+        assert "content_block_delta" in events
+        assert events[-1] == "message_stop"
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
