@@ -98,7 +98,16 @@ def init_db(database_url: str) -> None:
     _SessionLocal = sessionmaker(bind=_engine, expire_on_commit=False)
     from ..schemas import tables  # noqa: F401  (register models)
 
+    from sqlalchemy import inspect as _inspect
+
+    fresh = not _inspect(_engine).get_table_names()
     Base.metadata.create_all(_engine)
+    from . import migrations
+
+    if fresh:
+        migrations.stamp_head(_engine)   # new DB is already at head
+    else:
+        migrations.migrate(_engine)      # DB from an older build: upgrade
 
 
 def get_engine():

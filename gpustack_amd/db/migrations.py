@@ -1,0 +1,86 @@
+"""Versioned schema migrations (reference: alembic revisions under
+gpustack/migrations/ + the `migrate` CLI, cmd/db_migration.py).
+
+Re-designed without alembic: an ordered list of (version, description,
+fn) steps and a `schema_version` table. `create_all` brings a FRESH
+database straight to head (SQLAlchemy models are the source of truth);
+migrations exist for databases created by OLDER builds, where
+`create_all` does not alter existing tables. Each step must therefore be
+written additively (ADD COLUMN / CREATE TABLE IF NOT EXISTS) and be a
+no-op on a database that already has the change.
+"""
+from __future__ import annotations
+
+import logging
+
+from sqlalchemy import inspect, text
+
+logger = logging.getLogger(__name__)
+
+
+def _add_column(conn, table: str, column: str, ddl: str) -> None:
+    cols = [c["name"] for c in inspect(conn).get_columns(table)]
+    if column not in cols:
+        conn.execute(text(f"ALTER TABLE {table} ADD COLUMN {column} {ddl}"))
+
+
+# ---- migration steps -------------------------------------------------------
+# Append-only: never renumber or edit a shipped step.
+
+def _m001_worker_proxy_mode(conn):
+    _add_column(conn, "workers", "proxy_mode", "VARCHAR(16) DEFAULT 'direct'")
+
+
+def _m002_model_kv_features(conn):
+    _add_column(conn, "models", "extended_kv_cache", "JSON")
+    _add_column(conn, "models", "speculative_config", "JSON")
+    _add_column(conn, "models", "scaling_schedule", "JSON")
+
+
+def _m003_instance_distributed_servers(conn):
+    _add_column(conn, "model_instances", "distributed_servers", "JSON")
+
+
+MIGRATIONS: list[tuple[int, str, object]] = [
+    (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
+    (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
+    (3, "instance cross-worker rank layout", _m003_instance_distributed_servers),
+]
+
+HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0
+
+
+def current_version(conn) -> int:
+    conn.execute(text(
+        "CREATE TABLE IF NOT EXISTS schema_version (version INTEGER NOT NULL)"))
+    row = conn.execute(text("SELECT version FROM schema_version")).fetchone()
+    if row is None:
+        conn.execute(text("INSERT INTO schema_version (version) VALUES (0)"))
+        return 0
+    return int(row[0])
+
+
+def _set_version(conn, v: int) -> None:
+    conn.execute(text("UPDATE schema_version SET version = :v"), {"v": v})
+
+
+def stamp_head(engine) -> None:
+    """Mark a freshly-created (already at head) schema as up to date."""
+    with engine.begin() as conn:
+        current_version(conn)
+        _set_version(conn, HEAD)
+
+
+def migrate(engine) -> list[int]:
+    """Apply pending steps; returns the versions applied."""
+    applied: list[int] = []
+    with engine.begin() as conn:
+        v = current_version(conn)
+        for ver, desc, fn in MIGRATIONS:
+            if ver <= v:
+                continue
+            logger.info("migrating schema to v%d: %s", ver, desc)
+            fn(conn)
+            _set_version(conn, ver)
+            applied.append(ver)
+    return applied

@@ -42,6 +42,9 @@ def main(argv: list[str] | None = None) -> int:
     rp.add_argument("--data-dir", dest="data_dir", default=None)
     rp.add_argument("--database-url", dest="database_url", default=None)
     rp.add_argument("--password", required=True)
+    mp = sub.add_parser("migrate", help="apply pending DB schema migrations")
+    mp.add_argument("--database-url", default=None)
+    mp.add_argument("--data-dir", default=None)
     sub.add_parser("version")
     args = ap.parse_args(argv)
 
@@ -67,6 +70,18 @@ def main(argv: list[str] | None = None) -> int:
             admin.hashed_password = hash_password(args.password)
             s.commit()
         print("admin password reset")
+        return 0
+    if args.cmd == "migrate":
+        from .config import load_config
+        from .db import get_engine, init_db
+        from .db.migrations import HEAD, current_version, migrate
+
+        cfg = load_config(None, {"data_dir": args.data_dir,
+                                 "database_url": args.database_url})
+        init_db(cfg.resolved_database_url())  # runs pending migrations itself
+        with get_engine().begin() as conn:
+            v = current_version(conn)
+        print(f"schema at v{v} (head v{HEAD})")
         return 0
     if args.cmd == "chat":
         import httpx
