@@ -280,3 +280,8 @@ async def anthropic_messages(request: Request, user: User = Depends(get_current_
     """Anthropic-style Messages API, proxied to the placed instance
     (reference gateway routes /v1/messages: gateway/__init__.py:70-75)."""
     return await _proxy(request, "/v1/messages", user)
+
+
+@router.post("/v1/score")
+async def score(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/score", user)

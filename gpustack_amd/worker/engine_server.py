@@ -370,6 +370,44 @@ def create_app(runner: EngineRunner) -> FastAPI:
             ids = ids.ids
         return await _generate(request, body, list(ids), "chat")
 
+    @app.post("/v1/score")
+    @app.post("/score")
+    async def score(request: Request):
+        """Sentence-pair scoring (reference routes /v1/score to engines that
+        serve cross-encoder models; first-party design: embedding cosine
+        from the same engine, matching /v1/rerank)."""
+        body = await request.json()
+        t1 = body.get("text_1") or body.get("query") or ""
+        t2 = body.get("text_2") or body.get("documents") or []
+        if isinstance(t2, str):
+            t2 = [t2]
+        tok = runner.tokenizer
+
+        def enc(text):
+            ids = tok.encode(text)
+            if hasattr(ids, "ids"):
+                ids = ids.ids
+            return list(ids) or [0]
+
+        prompts = [enc(t1)] + [enc(d) for d in t2]
+        vecs = await runner.run_aux(runner.engine.runner.embed, prompts, "mean")
+        import math
+
+        def cos(a, b):
+            num = sum(x * y for x, y in zip(a, b))
+            den = math.sqrt(sum(x * x for x in a)) * math.sqrt(sum(y * y for y in b))
+            return num / den if den else 0.0
+
+        qv = vecs[0]
+        return {
+            "object": "list",
+            "model": runner.served_name,
+            "data": [{"object": "score", "index": i, "score": cos(qv, v)}
+                     for i, v in enumerate(vecs[1:])],
+            "usage": {"prompt_tokens": sum(len(p) for p in prompts),
+                      "total_tokens": sum(len(p) for p in prompts)},
+        }
+
     @app.post("/v1/messages")
     async def anthropic_messages(request: Request):
         """Anthropic-style Messages API (reference routes /v1/messages

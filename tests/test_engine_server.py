@@ -104,6 +104,12 @@ def test_engine_server_stop_strings():
         assert len(res) == 2
         assert res[0]["index"] == 0  # self-similarity ranks first
         assert res[0]["relevance_score"] >= res[1]["relevance_score"]
+        # score: same-text pair scores ~1.0
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/score", json={
+            "text_1": "abcabc", "text_2": ["abcabc", "zzqq"]}, timeout=60)
+        assert r.status_code == 200
+        d = r.json()["data"]
+        assert d[0]["score"] > 0.99 and d[0]["score"] >= d[1]["score"]
     finally:
         proc.terminate()
         try:
