@@ -131,6 +131,22 @@ def test_deploy_and_chat(cluster):
     assert r.status_code == 200
     assert r.json()["usage"]["completion_tokens"] == 4
 
+    # anthropic-style messages endpoint through the gateway
+    r = client.post("/v1/messages", json={
+        "model": "tiny-chat", "max_tokens": 4,
+        "messages": [{"role": "user", "content": "hello"}],
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["type"] == "message"
+    assert r.json()["usage"]["output_tokens"] > 0
+
+    # rerank through the gateway
+    r = client.post("/v1/rerank", json={
+        "model": "tiny-chat", "query": "abc", "documents": ["abc", "zzz"],
+    })
+    assert r.status_code == 200, r.text
+    assert len(r.json()["results"]) == 2
+
     # usage metering recorded
     usage = client.get("/v2/usage").json()["items"]
     assert usage and usage[0]["completion_tokens"] >= 12
