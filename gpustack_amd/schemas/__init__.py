@@ -139,6 +139,7 @@ class BenchmarkCreate(BaseModel):
     model_name: str
     mode: str = "concurrency"      # "concurrency" | "qps"
     value: float = 8
+    sweep: list[float] | None = None   # multi-point profile (one run per value)
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64

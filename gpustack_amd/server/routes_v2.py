@@ -428,6 +428,7 @@ def create_benchmark(body: BenchmarkCreate, _: User = Depends(get_current_user))
         b = Benchmark(
             name=body.name, model_name=body.model_name, worker_id=inst.worker_id,
             config={"mode": body.mode, "value": body.value,
+                    "sweep": body.sweep,
                     "duration_s": body.duration_s, "isl": body.isl, "osl": body.osl},
         )
         ar_create(s, b)

@@ -69,15 +69,26 @@ class BenchmarkManager:
                 raise RuntimeError("no local running instance for model")
             self._update(bid, state="running")
             cfgd = b.get("config") or {}
-            spec = LoadSpec(
-                mode=cfgd.get("mode", "concurrency"),
-                value=float(cfgd.get("value", 8)),
-                duration_s=float(cfgd.get("duration_s", 30)),
-                isl=int(cfgd.get("isl", 128)),
-                osl=int(cfgd.get("osl", 64)),
-                model=b["model_name"],
-            )
-            results = run_load(f"http://127.0.0.1:{port}", spec)
+            # sweep profiles (reference: multi-point performance profiles):
+            # config.sweep = [v1, v2, ...] runs one load point per value and
+            # stores the whole curve under results.profile
+            values = cfgd.get("sweep") or [cfgd.get("value", 8)]
+            profile = []
+            for v in values:
+                spec = LoadSpec(
+                    mode=cfgd.get("mode", "concurrency"),
+                    value=float(v),
+                    duration_s=float(cfgd.get("duration_s", 30)),
+                    isl=int(cfgd.get("isl", 128)),
+                    osl=int(cfgd.get("osl", 64)),
+                    model=b["model_name"],
+                )
+                point = run_load(f"http://127.0.0.1:{port}", spec)
+                point["value"] = float(v)
+                profile.append(point)
+            results = dict(max(profile, key=lambda r: r.get("output_tps") or 0))
+            if len(profile) > 1:
+                results["profile"] = profile
             self._update(bid, state="completed", results=results)
             logger.info("benchmark %s done: %s out-tok/s, p50 TTFT %s ms",
                         b.get("name"), results.get("output_tps"),

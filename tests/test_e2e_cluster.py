@@ -209,6 +209,23 @@ def test_benchmark_flow(cluster):
     assert res["successful_requests"] > 0
     assert res["output_tps"] > 0
     assert res["ttft_p50_ms"] is not None
+    # sweep profile: one point per value, curve in results.profile
+    r = client.post("/v2/benchmarks", json={
+        "name": "b-sweep", "model_name": "tiny-chat", "mode": "concurrency",
+        "sweep": [1, 2], "duration_s": 2.0, "isl": 16, "osl": 4,
+    })
+    assert r.status_code == 201, r.text
+    bid2 = r.json()["id"]
+    for _ in range(120):
+        b2 = [x for x in client.get("/v2/benchmarks").json()["items"]
+              if x["id"] == bid2][0]
+        if b2["state"] in ("completed", "error"):
+            break
+        time.sleep(1.0)
+    assert b2["state"] == "completed", b2.get("state_message")
+    prof = b2["results"]["profile"]
+    assert len(prof) == 2 and {p["value"] for p in prof} == {1.0, 2.0}
+
     # benchmark against a model with no instance is rejected
     r = client.post("/v2/benchmarks", json={
         "name": "b2", "model_name": "missing", "duration_s": 1})
