@@ -29,6 +29,12 @@ class ModelSpec:
     attention_bias: bool = False  # qwen2-style qkv bias
     qk_norm: bool = False         # qwen3-style per-head q/k RMSNorm
     eos_token_id: int = 128001
+    # MoE (Qwen3-MoE / Mixtral): num_experts > 0 makes every decoder
+    # layer's MLP a router + expert bank (models/llama.py MoEMLP)
+    num_experts: int = 0
+    num_experts_per_tok: int = 0
+    moe_intermediate_size: int = 0
+    norm_topk_prob: bool = True
 
     @property
     def gqa_ratio(self) -> int:
@@ -41,7 +47,10 @@ class ModelSpec:
         h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
         qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
         o = self.num_heads * self.head_dim * h
-        mlp = 3 * h * i
+        if self.num_experts > 0:
+            mlp = self.num_experts * (3 * h * self.moe_intermediate_size)                 + self.num_experts * h  # router
+        else:
+            mlp = 3 * h * i
         per_layer = qkv + o + mlp + 2 * h
         emb = v * h * (1 if self.tie_word_embeddings else 2)
         return (per_layer * self.num_layers + emb + h) * dtype_size
@@ -71,6 +80,13 @@ class ModelSpec:
             attention_bias=arch.startswith("Qwen2"),
             qk_norm=arch.startswith("Qwen3"),
             eos_token_id=eos,
+            num_experts=cfg.get("num_experts",
+                                cfg.get("num_local_experts", 0)) or 0,
+            num_experts_per_tok=cfg.get("num_experts_per_tok", 0) or 0,
+            moe_intermediate_size=cfg.get("moe_intermediate_size",
+                                          cfg.get("intermediate_size", 0))
+            if (cfg.get("num_experts") or cfg.get("num_local_experts")) else 0,
+            norm_topk_prob=bool(cfg.get("norm_topk_prob", True)),
         )
 
     @classmethod
@@ -103,6 +119,29 @@ PRESETS: dict[str, ModelSpec] = {
         architecture="Qwen2ForCausalLM", vocab_size=152064, hidden_size=3584,
         intermediate_size=18944, num_layers=28, num_heads=28, num_kv_heads=4,
         rope_theta=1000000.0, attention_bias=True, eos_token_id=151645,
+    ),
+    # Qwen3-30B-A3B: 128-expert top-8 MoE, 3B active params
+    "qwen3-30b-a3b": ModelSpec(
+        architecture="Qwen3MoeForCausalLM", vocab_size=151936,
+        hidden_size=2048, intermediate_size=6144, num_layers=48,
+        num_heads=32, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
+        max_position_embeddings=40960, qk_norm=True, eos_token_id=151645,
+        num_experts=128, num_experts_per_tok=8, moe_intermediate_size=768,
+    ),
+    # Mixtral 8x7B: 8-expert top-2 MoE on the llama graph
+    "mixtral-8x7b": ModelSpec(
+        architecture="MixtralForCausalLM", vocab_size=32000,
+        hidden_size=4096, intermediate_size=14336, num_layers=32,
+        num_heads=32, num_kv_heads=8, rope_theta=1000000.0,
+        max_position_embeddings=32768, eos_token_id=2,
+        num_experts=8, num_experts_per_tok=2, moe_intermediate_size=14336,
+    ),
+    # tiny MoE (CPU-testable routing/dispatch plumbing)
+    "tiny-moe": ModelSpec(
+        vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,
+        num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
+        rope_theta=10000.0, eos_token_id=1,
+        num_experts=8, num_experts_per_tok=2, moe_intermediate_size=64,
     ),
     # Mistral v0.3: llama-compatible compute graph (no SWA since v0.1;
     # GQA 32/8, theta 1e6) — runs on the same CDNA4 kernel set

@@ -118,12 +118,71 @@ class MLP(nn.Module):
         return self.comm.all_reduce(ops.linear_auto(act, self.down_w))
 
 
+class MoEMLP(nn.Module):
+    """Router + expert bank (Qwen3-MoE `Qwen3MoeSparseMoeBlock`, Mixtral).
+
+    Experts live as stacked fused tensors [E, 2*I_loc, h] / [E, h, I_loc];
+    TP shards each expert's intermediate dim (same all-reduce pattern as
+    the dense MLP). Dispatch is sort-free eager: per-expert index_select +
+    index_add over the top-k assignment table — a fused grouped-GEMM
+    kernel is the round-2 optimization; this path is correct under graphs
+    capture (fixed shapes per expert loop iteration are not required
+    because MoE models currently run eager decode).
+    """
+
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+        super().__init__()
+        self.comm = comm
+        self.spec = spec
+        h = spec.hidden_size
+        self.e = spec.num_experts
+        self.top_k = spec.num_experts_per_tok
+        self.i = spec.moe_intermediate_size // tp_size
+        self.router_w = nn.Parameter(torch.empty(self.e, h, dtype=dtype),
+                                     requires_grad=False)
+        self.gate_up_w = nn.Parameter(
+            torch.empty(self.e, 2 * self.i, h, dtype=dtype), requires_grad=False)
+        self.down_w = nn.Parameter(
+            torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
+
+    def forward(self, x):
+        T = x.shape[0]
+        logits = F.linear(x.float(), self.router_w.float())      # [T, E]
+        if self.spec.norm_topk_prob:
+            # softmax over all experts -> top-k -> renormalize == softmax
+            # restricted to the top-k logits (Qwen3-MoE default, Mixtral)
+            weights, experts = torch.topk(logits, self.top_k, dim=-1)
+            weights = torch.softmax(weights, dim=-1)
+        else:
+            # un-renormalized: keep the full-softmax probabilities
+            probs = torch.softmax(logits, dim=-1)
+            weights, experts = torch.topk(probs, self.top_k, dim=-1)
+        out = torch.zeros_like(x)
+        flat_exp = experts.reshape(-1)                            # [T*k]
+        flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
+        flat_w = weights.reshape(-1).to(x.dtype)
+        hit = torch.bincount(flat_exp, minlength=self.e)
+        for e in torch.nonzero(hit, as_tuple=False).flatten().tolist():
+            sel = flat_exp == e
+            idx = flat_tok[sel]
+            xe = x.index_select(0, idx)
+            gu = F.linear(xe, self.gate_up_w[e])
+            act = torch.empty(xe.shape[0], self.i, dtype=x.dtype, device=x.device)
+            ops.silu_and_mul(act, gu)
+            he = F.linear(act, self.down_w[e])
+            out.index_add_(0, idx, he * flat_w[sel].unsqueeze(1))
+        return self.comm.all_reduce(out)
+
+
 class DecoderLayer(nn.Module):
     def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
         super().__init__()
         self.spec = spec
         self.attn = Attention(spec, tp_size, comm, dtype)
-        self.mlp = MLP(spec, tp_size, comm, dtype)
+        if spec.num_experts > 0:
+            self.mlp = MoEMLP(spec, tp_size, comm, dtype)
+        else:
+            self.mlp = MLP(spec, tp_size, comm, dtype)
         h = spec.hidden_size
         self.input_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
         self.post_attn_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)

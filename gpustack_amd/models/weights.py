@@ -71,16 +71,36 @@ def random_init(model, cfg: EngineConfig) -> None:
         if spec.qk_norm:
             layer.attn.q_norm.fill_(1.0)
             layer.attn.k_norm.fill_(1.0)
-        gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
-        up = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.up", seed, dtype, device)
-        layer.mlp.gate_up_w.copy_(torch.cat([
-            gate[rank * i_loc:(rank + 1) * i_loc],
-            up[rank * i_loc:(rank + 1) * i_loc],
-        ]))
-        down = _gen((spec.hidden_size, spec.intermediate_size), f"{li}.down", seed, dtype, device)
-        layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
+        if spec.num_experts > 0:
+            _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank)
+        else:
+            gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
+            up = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.up", seed, dtype, device)
+            layer.mlp.gate_up_w.copy_(torch.cat([
+                gate[rank * i_loc:(rank + 1) * i_loc],
+                up[rank * i_loc:(rank + 1) * i_loc],
+            ]))
+            down = _gen((spec.hidden_size, spec.intermediate_size), f"{li}.down", seed, dtype, device)
+            layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.fill_(1.0)
         layer.post_attn_norm.fill_(1.0)
+
+
+def _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank):
+    mi_loc = spec.moe_intermediate_size // tp
+    layer.mlp.router_w.copy_(_gen((spec.num_experts, spec.hidden_size),
+                                  f"{li}.router", seed, dtype, device))
+    for e in range(spec.num_experts):
+        gate = _gen((spec.moe_intermediate_size, spec.hidden_size),
+                    f"{li}.e{e}.gate", seed, dtype, device)
+        up = _gen((spec.moe_intermediate_size, spec.hidden_size),
+                  f"{li}.e{e}.up", seed, dtype, device)
+        layer.mlp.gate_up_w[e].copy_(torch.cat([
+            gate[rank * mi_loc:(rank + 1) * mi_loc],
+            up[rank * mi_loc:(rank + 1) * mi_loc]]))
+        down = _gen((spec.hidden_size, spec.moe_intermediate_size),
+                    f"{li}.e{e}.down", seed, dtype, device)
+        layer.mlp.down_w[e].copy_(down[:, rank * mi_loc:(rank + 1) * mi_loc])
 
 
 def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
@@ -133,12 +153,47 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
         if spec.qk_norm:
             layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
             layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
-        layer.mlp.gate_up_w.copy_(torch.cat([
-            row_shard(get(p + "mlp.gate_proj.weight"), i_loc),
-            row_shard(get(p + "mlp.up_proj.weight"), i_loc),
-        ]))
-        dn = get(p + "mlp.down_proj.weight")
-        layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
+        if spec.num_experts > 0:
+            # MoE (Qwen3-MoE `mlp.gate`, Mixtral `block_sparse_moe.gate`)
+            router = (p + "mlp.gate.weight" if p + "mlp.gate.weight" in tensors
+                      else p + "block_sparse_moe.gate.weight")
+            layer.mlp.router_w.copy_(get(router))
+            mi_loc = spec.moe_intermediate_size // tp
+            mi = spec.moe_intermediate_size
+            if p + "mlp.experts.gate_up_proj" in tensors:
+                # transformers >=5 fused layout [E, 2mi, h] / [E, h, mi]
+                gu = get(p + "mlp.experts.gate_up_proj")
+                dn = get(p + "mlp.experts.down_proj")
+                layer.mlp.gate_up_w.copy_(torch.cat([
+                    gu[:, rank * mi_loc:(rank + 1) * mi_loc],
+                    gu[:, mi + rank * mi_loc:mi + (rank + 1) * mi_loc],
+                ], dim=1))
+                layer.mlp.down_w.copy_(
+                    dn[:, :, rank * mi_loc:(rank + 1) * mi_loc])
+                layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+                layer.post_attn_norm.copy_(
+                    get(p + "post_attention_layernorm.weight"))
+                continue
+            for e in range(spec.num_experts):
+                if f"{p}mlp.experts.{e}.gate_proj.weight" in tensors:
+                    ep = f"{p}mlp.experts.{e}."
+                    gname, uname, dname = "gate_proj", "up_proj", "down_proj"
+                else:  # mixtral naming: w1=gate, w3=up, w2=down
+                    ep = f"{p}block_sparse_moe.experts.{e}."
+                    gname, uname, dname = "w1", "w3", "w2"
+                layer.mlp.gate_up_w[e].copy_(torch.cat([
+                    row_shard(get(ep + gname + ".weight"), mi_loc),
+                    row_shard(get(ep + uname + ".weight"), mi_loc),
+                ]))
+                dn = get(ep + dname + ".weight")
+                layer.mlp.down_w[e].copy_(dn[:, rank * mi_loc:(rank + 1) * mi_loc])
+        else:
+            layer.mlp.gate_up_w.copy_(torch.cat([
+                row_shard(get(p + "mlp.gate_proj.weight"), i_loc),
+                row_shard(get(p + "mlp.up_proj.weight"), i_loc),
+            ]))
+            dn = get(p + "mlp.down_proj.weight")
+            layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.copy_(get(p + "input_layernorm.weight"))
         layer.post_attn_norm.copy_(get(p + "post_attention_layernorm.weight"))
 

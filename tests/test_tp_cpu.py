@@ -24,16 +24,17 @@ def _free_port() -> int:
 PROMPTS = [[3, 1, 4, 1, 5, 9, 2, 6], [11, 22, 33]]
 
 
-def _single_proc_result() -> list[list[int]]:
+def _single_proc_result(model: str = "tiny") -> list[list[int]]:
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 
-    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+    cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
                        max_model_len=128, seed=0)
     eng = LLMEngine(cfg)
     return eng.generate(PROMPTS, SamplingParams(max_tokens=6, ignore_eos=True))
 
 
-def _tp_rank_main(rank: int, world: int, port: int, out_path: str):
+def _tp_rank_main(rank: int, world: int, port: int, out_path: str,
+                  model: str = "tiny"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -42,7 +43,7 @@ def _tp_rank_main(rank: int, world: int, port: int, out_path: str):
     from gpustack_amd.parallel import init_tp
 
     comm = init_tp(world, rank, master_port=port, backend="gloo")
-    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+    cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
                        max_model_len=128, seed=0, tp_size=world, tp_rank=rank)
     eng = LLMEngine(cfg, comm)
     results: dict[str, list[int]] = {}
@@ -64,13 +65,11 @@ def _tp_rank_main(rank: int, world: int, port: int, out_path: str):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
-def test_tp2_matches_tp1_greedy():
-    expected = _single_proc_result()
+def _run_tp2(model: str) -> list[list[int]]:
     port = _free_port()
     out_path = tempfile.mktemp(suffix=".json")
     ctx = mp.get_context("spawn")
-    procs = [ctx.Process(target=_tp_rank_main, args=(r, 2, port, out_path))
+    procs = [ctx.Process(target=_tp_rank_main, args=(r, 2, port, out_path, model))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -78,5 +77,24 @@ def test_tp2_matches_tp1_greedy():
         p.join(timeout=240)
         assert p.exitcode == 0, f"rank process exited {p.exitcode}"
     with open(out_path) as f:
-        got = json.load(f)
+        return json.load(f)
+
+
+@pytest.mark.timeout(300)
+def test_tp2_matches_tp1_greedy():
+    expected = _single_proc_result("tiny")
+    got = _run_tp2("tiny")
     assert got == expected, f"{got} != {expected}"
+
+
+@pytest.mark.timeout(600)
+def test_tp2_moe_deterministic():
+    """MoE under TP: the all-reduce changes f32 summation order by ~1 ulp,
+    and the router top-k can flip a near-tie expert choice, so exact
+    TP2==TP1 token match is not guaranteed (unlike dense, where vocab
+    argmax gaps absorb the epsilon). The contract tested: TP2 is
+    deterministic run-to-run, and both ranks agree on every token."""
+    a = _run_tp2("tiny-moe")
+    b = _run_tp2("tiny-moe")
+    assert a == b, f"{a} != {b}"
+    assert all(len(x) == 6 for x in a)
