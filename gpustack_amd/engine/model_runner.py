@@ -152,7 +152,10 @@ class ModelRunner:
         if self.device.type == "cuda":
             from .graph_runner import DecodeGraphRunner, graphs_enabled
 
-            if graphs_enabled() and self.eagle is None:
+            # MoE dispatch syncs with the host per layer (expert routing),
+            # which hipGraph capture cannot record -> eager decode for MoE
+            # (sync-free fused dispatch is the round-2 item)
+            if graphs_enabled() and self.eagle is None and cfg.spec.num_experts == 0:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv

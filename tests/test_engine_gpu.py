@@ -96,12 +96,13 @@ def test_eagle_matches_plain_gpu():
 
 
 def test_moe_decode_matches_prefill_gpu():
-    # MoE routing/dispatch on the HIP path (qwen3-moe-style tiny config)
-    eng = LLMEngine(_cfg(model="tiny-moe", max_model_len=256))
+    # MoE routing/dispatch on the HIP path (D=128 decode kernel; 4 layers
+    # of the real qwen3-30b-a3b expert geometry)
+    eng = LLMEngine(_cfg(model="qwen3-30b-a3b", max_model_len=256))
     full = eng.generate(PROMPTS[:1], SamplingParams(max_tokens=8, ignore_eos=True))[0]
     del eng
     torch.cuda.empty_cache()
-    eng2 = LLMEngine(_cfg(model="tiny-moe", max_model_len=256))
+    eng2 = LLMEngine(_cfg(model="qwen3-30b-a3b", max_model_len=256))
     cont = eng2.generate([PROMPTS[0] + full[:4]],
                          SamplingParams(max_tokens=4, ignore_eos=True))[0]
     assert cont == full[4:], f"{cont} != {full[4:]}"
