@@ -121,3 +121,24 @@ def test_moe_from_hf_config():
         "num_experts_per_tok": 2, "intermediate_size": 14336,
     })
     assert mix.num_experts == 8 and mix.moe_intermediate_size == 14336
+
+
+def test_moe_dispatch_paths_agree():
+    # the padded-bmm decode path must match the per-expert loop exactly
+    # (same per-row math; padding rows are discarded)
+    eng = LLMEngine(EngineConfig(model="tiny-moe", device="cpu",
+                                 kv_cache_blocks=64, max_model_len=128))
+    mlp = eng.runner.model.layers[0].mlp
+    torch.manual_seed(0)
+    x = torch.randn(9, 128, dtype=mlp.gate_up_w.dtype)
+    logits = torch.nn.functional.linear(x.float(), mlp.router_w.float())
+    w, e = torch.topk(logits, mlp.top_k, dim=-1)
+    w = torch.softmax(w, dim=-1)
+    fe = e.reshape(-1)
+    ft = torch.arange(9).repeat_interleave(mlp.top_k)
+    fw = w.reshape(-1).to(x.dtype)
+    a = torch.zeros_like(x)
+    b = torch.zeros_like(x)
+    mlp._loop_dispatch(x, a, fe, ft, fw)
+    mlp._bmm_dispatch(x, b, fe, ft, fw)
+    torch.testing.assert_close(a, b, atol=2e-2, rtol=2e-2)
