@@ -157,10 +157,15 @@ class MoEMLP(nn.Module):
             # un-renormalized: keep the full-softmax probabilities
             probs = torch.softmax(logits, dim=-1)
             weights, experts = torch.topk(probs, self.top_k, dim=-1)
-        out = torch.zeros_like(x)
         flat_exp = experts.reshape(-1)                            # [T*k]
         flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
         flat_w = weights.reshape(-1).to(x.dtype)
+        # Per-assignment contributions land at UNIQUE rows of a [T*k, h]
+        # buffer, then reduce over the fixed k axis — index_add_ over
+        # duplicated token rows would use atomics, whose order (and thus
+        # the bf16 rounding) varies run to run and flips the next layer's
+        # router on near-ties (observed on HW).
+        contrib = x.new_zeros(T * self.top_k, x.shape[1])
         # Two dispatch paths:
         #  - decode-shaped (few tokens/expert, many experts): ONE padded
         #    strided-batched GEMM pair — the per-expert loop would be
@@ -168,29 +173,29 @@ class MoEMLP(nn.Module):
         #  - prefill-shaped: per-expert GEMMs are large enough that the
         #    loop is GEMM-bound and avoids the padding overcompute
         if self.e >= 16 and flat_exp.numel() < 32 * self.e:
-            self._bmm_dispatch(x, out, flat_exp, flat_tok, flat_w)
+            self._bmm_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
         else:
-            self._loop_dispatch(x, out, flat_exp, flat_tok, flat_w)
+            self._loop_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
+        out = contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
         return self.comm.all_reduce(out)
 
-    def _loop_dispatch(self, x, out, flat_exp, flat_tok, flat_w):
+    def _loop_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w):
         hit = torch.bincount(flat_exp, minlength=self.e)
         for e in torch.nonzero(hit, as_tuple=False).flatten().tolist():
-            sel = flat_exp == e
-            idx = flat_tok[sel]
+            rows = torch.nonzero(flat_exp == e, as_tuple=False).flatten()
+            idx = flat_tok[rows]
             xe = x.index_select(0, idx)
             gu = F.linear(xe, self.gate_up_w[e])
             act = torch.empty(xe.shape[0], self.i, dtype=x.dtype, device=x.device)
             ops.silu_and_mul(act, gu)
             he = F.linear(act, self.down_w[e])
-            out.index_add_(0, idx, he * flat_w[sel].unsqueeze(1))
+            contrib[rows] = he * flat_w[rows].unsqueeze(1)
 
-    def _bmm_dispatch(self, x, out, flat_exp, flat_tok, flat_w):
+    def _bmm_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w):
         TK = flat_exp.numel()
         order = torch.argsort(flat_exp, stable=True)
         s_exp = flat_exp[order]
         s_tok = flat_tok[order]
-        s_w = flat_w[order]
         counts = torch.bincount(s_exp, minlength=self.e)
         cap = int(counts.max())          # one host sync per layer
         if cap == 0:
@@ -204,7 +209,7 @@ class MoEMLP(nn.Module):
         ops.silu_and_mul(act, gu.reshape(self.e * cap, 2 * self.i))
         hd = torch.bmm(act.view(self.e, cap, self.i),
                        self.down_w.transpose(1, 2))            # [E, cap, h]
-        out.index_add_(0, s_tok, hd[s_exp, pos] * s_w.unsqueeze(1))
+        contrib[order] = hd[s_exp, pos] * flat_w[order].unsqueeze(1)
 
 
 class DecoderLayer(nn.Module):

@@ -115,8 +115,8 @@ def test_moe_gpu_deterministic_and_paths_agree():
     fe = e.reshape(-1)
     ft = torch.arange(40, device="cuda").repeat_interleave(mlp.top_k)
     fw = w.reshape(-1).to(x.dtype)
-    pa = torch.zeros_like(x)
-    pb = torch.zeros_like(x)
+    pa = x.new_zeros(x.shape[0] * mlp.top_k, x.shape[1])
+    pb = x.new_zeros(x.shape[0] * mlp.top_k, x.shape[1])
     mlp._loop_dispatch(x, pa, fe, ft, fw)
     mlp._bmm_dispatch(x, pb, fe, ft, fw)
     torch.testing.assert_close(pa.float(), pb.float(), atol=3e-2, rtol=3e-2)

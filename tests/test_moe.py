@@ -137,8 +137,8 @@ def test_moe_dispatch_paths_agree():
     fe = e.reshape(-1)
     ft = torch.arange(9).repeat_interleave(mlp.top_k)
     fw = w.reshape(-1).to(x.dtype)
-    a = torch.zeros_like(x)
-    b = torch.zeros_like(x)
+    a = x.new_zeros(x.shape[0] * mlp.top_k, x.shape[1])
+    b = x.new_zeros(x.shape[0] * mlp.top_k, x.shape[1])
     mlp._loop_dispatch(x, a, fe, ft, fw)
     mlp._bmm_dispatch(x, b, fe, ft, fw)
     torch.testing.assert_close(a, b, atol=2e-2, rtol=2e-2)
