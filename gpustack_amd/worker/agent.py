@@ -242,7 +242,29 @@ class WorkerAgent:
                 g_vt.labels(idx).set(d.get("memory", {}).get("total", 0))
                 g_vu.labels(idx).set(d.get("memory", {}).get("used", 0))
                 g_temp.labels(idx).set(d.get("temperature", 0))
-            return Response(generate_latest(reg), media_type="text/plain; version=0.0.4")
+            body = generate_latest(reg).decode()
+            # re-export each local engine's runtime metrics with an
+            # instance label (reference: worker metrics aggregation of
+            # backend runtime metrics)
+            import httpx as _hx
+
+            if self.serve_manager is not None:
+                for ip in list(self.serve_manager.processes.values()):
+                    if not ip.healthy:
+                        continue
+                    try:
+                        r = _hx.get(f"http://127.0.0.1:{ip.port}/metrics",
+                                    timeout=2)
+                    except _hx.HTTPError:
+                        continue
+                    inst = ip.instance.get("name", "")
+                    for ln in r.text.splitlines():
+                        if ln.startswith("#"):
+                            body += ln + "\n"
+                        elif ln.strip():
+                            name, _, val = ln.partition(" ")
+                            body += f'{name}{{instance="{inst}"}} {val}\n'
+            return Response(body, media_type="text/plain; version=0.0.4")
 
         return app
 

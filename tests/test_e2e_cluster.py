@@ -151,6 +151,11 @@ def test_deploy_and_chat(cluster):
     usage = client.get("/v2/usage").json()["items"]
     assert usage and usage[0]["completion_tokens"] >= 12
 
+    # worker /metrics re-exports engine runtime metrics with instance label
+    wm = httpx.get(f"http://127.0.0.1:{agent.cfg.worker_port}/metrics", timeout=10)
+    assert wm.status_code == 200
+    assert 'gpustack_engine_generated_tokens_total{instance="' in wm.text
+
     # instance logs reachable on the worker API
     insts = client.get("/v2/model_instances").json()["items"]
     wr = httpx.get(
