@@ -153,6 +153,26 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float, tiles=N
 _SG_WORKSPACES: dict = {}
 
 
+def gemm8(x, w, out=None, safe: bool = False):
+    """8-phase pipelined 256x256 MFMA GEMM: out[M,N] = x[M,K] @ w[N,K]^T
+    (bf16, f32 accumulate). GPU-only; requires N%256==0, K%128==0 —
+    falls back to F.linear otherwise. `safe=True` runs the fully-drained
+    variant (race isolation for tests)."""
+    import torch.nn.functional as F
+
+    M, K = x.shape
+    N = w.shape[0]
+    if not x.is_cuda or N % 256 != 0 or K % 128 != 0:
+        return F.linear(x, w, None)
+    if out is None:
+        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    hip = _backend(x)
+    if hip is None:
+        return F.linear(x, w, None)
+    hip.gemm8(out, x.contiguous(), w, 1 if safe else 0)
+    return out
+
+
 def skinny_gemm(x, w, out=None, splitk: int | None = None, version: int = 1):
     """Split-K decode GEMM: out[M,N] = x[M,K] @ w[N,K]^T (bf16, f32 acc).
     GPU-only (falls back to F.linear elsewhere or for unsupported shapes)."""

@@ -15,6 +15,7 @@ void paged_attn_decode_launch(void*, const void*, const void*, const void*, cons
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
+void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
@@ -174,6 +175,18 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   HIP_CHECK_LAST();
 }
 
+void gemm8(at::Tensor out, at::Tensor x, at::Tensor w, long safe) {
+  check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  int err = 0;
+  gemm8_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), M, N, K,
+               (int)safe, &err, cur_stream(x));
+  TORCH_CHECK(!err, "gemm8: unsupported shape M=", M, " N=", N, " K=", K,
+              " (need N%256==0, K%128==0)");
+  HIP_CHECK_LAST();
+}
+
 void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
                  c10::optional<at::Tensor> workspace, long splitk,
                  long version) {
@@ -223,4 +236,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");
+  m.def("gemm8", &gemm8, "8-phase pipelined 256x256 GEMM (bf16, f32 accum)");
 }

@@ -281,3 +281,25 @@ def test_reshape_and_cache_fp8():
     # HW cvt and torch cvt are both OCP e4m3 RNE: bit-identical expected
     assert torch.equal(kc.view(torch.uint8), kc2.view(torch.uint8))
     assert torch.equal(vc.view(torch.uint8), vc2.view(torch.uint8))
+
+
+@pytest.mark.parametrize("safe", [True, False])
+@pytest.mark.parametrize("shape", [(256, 256, 128), (320, 512, 512),
+                                   (512, 1024, 896), (100, 256, 256)])
+def test_gemm8(shape, safe):
+    M, N, K = shape
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+    out = ops.gemm8(x, w, safe=safe)
+    ref = (x.float() @ w.float().t())
+    _close(out, ref.to(torch.bfloat16), atol=8e-2, rtol=8e-2)
+
+
+def test_gemm8_matches_safe_variant():
+    # the pipelined schedule must agree bit-for-bit with the drained one
+    torch.manual_seed(7)
+    x = torch.randn(512, 4096, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(4096, 4096, dtype=torch.bfloat16, device="cuda")
+    a = ops.gemm8(x, w, safe=False)
+    b = ops.gemm8(x, w, safe=True)
+    assert torch.equal(a, b)
