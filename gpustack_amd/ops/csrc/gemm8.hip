@@ -230,5 +230,4 @@ void gemm8_launch(void* out, const void* x, const void* w, int M, int N,
                        (unsigned short*)out, (const unsigned short*)x,
                        (const unsigned short*)w, M, N, K);
   }
-  HIP_CHECK_KERNEL();
 }

@@ -1,4 +1,9 @@
 """A/B the 8-phase GEMM vs hipBLASLt on its target shapes."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 import torch
 
 from gpustack_amd import ops
