@@ -144,8 +144,11 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
       afr[rt] = *reinterpret_cast<const u16x8*>(
           reinterpret_cast<const char*>(Al[buf][kh]) + a_off(mh, rt));
     if (sh >= 0) stage(sh);
+    // ONE barrier per phase: paces the waves (LDS-slot safety additionally
+    // rides the >=600ns global-load latency, see profiles notes); the
+    // compiler inserts fine-grained lgkmcnt waits per MFMA operand, so
+    // early MFMAs overlap the tail of the ds_read burst.
     __builtin_amdgcn_s_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)");
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int rt = 0; rt < 4; ++rt)
@@ -154,7 +157,6 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
         acc[mh][rt][ct] = g8_mfma(afr[rt], bfr[ct], acc[mh][rt][ct]);
     __builtin_amdgcn_s_setprio(0);
     if (wait6 && !SAFE) asm volatile("s_waitcnt vmcnt(6)");
-    __builtin_amdgcn_s_barrier();
   };
 
   // ---- prologue: stage tile0 fully + 3 halves of tile1 (7 ahead)
