@@ -88,6 +88,12 @@ def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
     app.include_router(auth_router)
     app.include_router(v2_router)
     app.include_router(openai_router)
+    from ..extension import apply_routers, load_plugins, run_start_hooks
+
+    plugins = load_plugins()
+    app.state.plugins = plugins
+    apply_routers(app, plugins)
+    run_start_hooks(app, cfg, plugins)
 
     @app.get("/healthz")
     def healthz():
@@ -102,6 +108,26 @@ def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
         from .exporter import render_metrics
 
         return Response(render_metrics(), media_type="text/plain; version=0.0.4")
+
+    @app.get("/metrics/targets")
+    def metrics_targets():
+        """Prometheus http_sd discovery of every worker's /metrics endpoint
+        (reference: exporter/exporter.py:272-295)."""
+        from ..db import get_session as _gs
+        from ..schemas import Worker, WorkerState
+
+        targets = []
+        with _gs() as s:
+            for w in s.query(Worker).all():
+                if w.state != WorkerState.READY.value or not w.ip:
+                    continue
+                port = w.metrics_port or w.port
+                targets.append({
+                    "targets": [f"{w.ip}:{port}"],
+                    "labels": {"worker": w.name,
+                               "instance": f"{w.ip}:{port}"},
+                })
+        return targets
 
     @app.exception_handler(Exception)
     async def unhandled(request, exc):
@@ -123,12 +149,16 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
 
     # leader election only matters with a shared external DB; a lease over
     # SQLite still works for tests / local HA pairs
-    if getattr(cfg, "ha_leases", False) or cfg.database_url:
-        coord = LeaseCoordinator()
-        coord.try_acquire()
-        coord.start()
-    else:
-        coord = LocalCoordinator()
+    from ..extension import pick_coordinator
+
+    coord = pick_coordinator(cfg, getattr(app.state, "plugins", []))
+    if coord is None:
+        if getattr(cfg, "ha_leases", False) or cfg.database_url:
+            coord = LeaseCoordinator()
+            coord.try_acquire()
+            coord.start()
+        else:
+            coord = LocalCoordinator()
     app.state.coordinator = coord
 
     sched = PlacementScheduler(cfg)
