@@ -158,6 +158,10 @@ class ModelInstance(Base, TimestampMixin, SerializeMixin):
     pid = Column(Integer, nullable=True)
     restart_count = Column(Integer, default=0)
     distributed_servers = Column(JSON, default=None)  # subordinate workers
+    # hash of the model's serving-relevant fields at creation; the
+    # controller replaces instances whose hash no longer matches (model
+    # updates redeploy, reference: model spec changes recreate instances)
+    spec_hash = Column(String(64), default="")
 
 
 class ModelFile(Base, TimestampMixin, SerializeMixin):

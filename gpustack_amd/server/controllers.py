@@ -32,6 +32,26 @@ class _LeaderGated:
         return self.coordinator is None or self.coordinator.is_leader
 
 
+def model_spec_hash(model) -> str:
+    """Hash of every field that changes how an instance must be served;
+    instances carrying a stale hash are replaced by the controller."""
+    import hashlib
+    import json as _json
+
+    payload = _json.dumps({
+        "source": model.source, "model_ref": model.model_ref,
+        "gpus_per_replica": model.gpus_per_replica,
+        "backend_parameters": model.backend_parameters,
+        "env": model.env, "max_model_len": model.max_model_len,
+        "gpu_memory_utilization": model.gpu_memory_utilization,
+        "speculative_config": model.speculative_config,
+        "extended_kv_cache": model.extended_kv_cache,
+        "lora_list": model.lora_list,
+        "distributed": model.distributed_inference_across_workers,
+    }, sort_keys=True, default=str)
+    return hashlib.sha256(payload.encode()).hexdigest()[:16]
+
+
 class ModelController(_LeaderGated):
     def __init__(self, cfg: Config):
         self.cfg = cfg
@@ -83,6 +103,13 @@ class ModelController(_LeaderGated):
                 return
             insts = s.query(ModelInstance).filter_by(model_id=model_id).all()
             want = model.replicas if want_override is None else want_override
+            # model updates redeploy: drop instances built from a stale spec
+            # (their deletion events re-enter this loop and recreate them)
+            cur_hash = model_spec_hash(model)
+            stale = [i for i in insts if i.spec_hash and i.spec_hash != cur_hash]
+            for v in stale:
+                ar_delete(s, v)
+            insts = [i for i in insts if i not in stale]
             have = len(insts)
             if have < want:
                 used = {i.name for i in insts}
@@ -95,6 +122,7 @@ class ModelController(_LeaderGated):
                     inst = ModelInstance(
                         model_id=model.id, model_name=model.name, name=name,
                         state=ModelInstanceState.PENDING.value,
+                        spec_hash=cur_hash,
                     )
                     ar_create(s, inst)
                     have += 1

@@ -294,3 +294,56 @@ def test_catalog_and_version(server):
     cat = client.get("/v2/catalog").json()
     assert any(m["model_ref"] == "llama-3-8b" for m in cat["models"])
     assert "version" in client.get("/v2/version").json()
+
+
+def test_model_update_redeploys_instances(server):
+    from gpustack_amd.db import get_session
+    from gpustack_amd.schemas import ModelInstance
+    from gpustack_amd.server.controllers import ModelController
+
+    client, app, cfg, reg = server
+    r = client.post("/v2/models", json={"name": "redeploy-m",
+                                        "model_ref": "tiny", "replicas": 2})
+    assert r.status_code == 201
+    mid = r.json()["id"]
+    from gpustack_amd.db import EventType, bus
+
+    mc = ModelController(cfg)
+    mc.sync_replicas(mid)
+    with get_session() as s:
+        before = {i.name: i.spec_hash for i in
+                  s.query(ModelInstance).filter_by(model_id=mid).all()}
+    assert len(before) == 2 and all(before.values())
+    # serving-relevant update -> instances torn down and recreated
+    # (DELETED + CREATED events drive the worker restart; SQLite may reuse
+    # row ids, so assert on the event stream + spec_hash)
+    q = bus.subscribe("model_instances")
+    client.patch(f"/v2/models/{mid}", json={"max_model_len": 2048})
+    mc.sync_replicas(mid)
+    events = []
+    import queue as _q
+
+    try:
+        while True:
+            events.append(q.get_nowait())
+    except _q.Empty:
+        pass
+    bus.unsubscribe("model_instances", q)
+    assert sum(1 for e in events if e.type == EventType.DELETED) == 2
+    assert sum(1 for e in events if e.type == EventType.CREATED) == 2
+    with get_session() as s:
+        after = s.query(ModelInstance).filter_by(model_id=mid).all()
+    assert len(after) == 2
+    old_hash = list(before.values())[0]
+    assert all(i.spec_hash and i.spec_hash != old_hash for i in after)
+    # replicas-only change does NOT replace
+    q = bus.subscribe("model_instances")
+    client.patch(f"/v2/models/{mid}", json={"replicas": 2})
+    mc.sync_replicas(mid)
+    try:
+        while True:
+            assert q.get_nowait().type not in (EventType.DELETED,
+                                               EventType.CREATED)
+    except _q.Empty:
+        pass
+    bus.unsubscribe("model_instances", q)
