@@ -193,7 +193,9 @@ class LLMEngine:
             self._async_enabled
             and not batch.is_prefill
             and batch.rows_per_seq == 1
-            and all(s.params.greedy and not s.params.logprobs for s in batch.seqs)
+            and all(s.params.greedy and not s.params.logprobs
+                    and not s.params.needs_logit_processing
+                    for s in batch.seqs)
         )
 
     def step(self) -> list[StepOutput]:

@@ -19,16 +19,48 @@ class Sampler:
         self.device = device
         self.generator = None
 
+    @staticmethod
+    def _process_logits(row: torch.Tensor, seq: Sequence) -> torch.Tensor:
+        """OpenAI/HF-style logit processors (presence/frequency/repetition
+        penalties + logit_bias), applied on the fp32 logits row."""
+        p = seq.params
+        seen = seq.output_token_ids
+        if seen and (p.presence_penalty or p.frequency_penalty
+                     or p.repetition_penalty != 1.0):
+            ids = torch.tensor(seen, dtype=torch.long, device=row.device)
+            uniq, counts = torch.unique(ids, return_counts=True)
+            if p.repetition_penalty != 1.0:
+                vals = row[uniq]
+                row[uniq] = torch.where(vals > 0, vals / p.repetition_penalty,
+                                        vals * p.repetition_penalty)
+            if p.presence_penalty:
+                row[uniq] -= p.presence_penalty
+            if p.frequency_penalty:
+                row[uniq] -= p.frequency_penalty * counts.to(row.dtype)
+        if p.logit_bias:
+            for tid, b in p.logit_bias.items():
+                t = int(tid)
+                if 0 <= t < row.shape[-1]:
+                    row[t] += float(b)
+        return row
+
     def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:
-        if all(s.params.greedy for s in seqs):
+        if all(s.params.greedy and not s.params.needs_logit_processing
+               for s in seqs):
             return ops.greedy_sample(logits).tolist()
         out: list[int] = [0] * len(seqs)
-        greedy_idx = [i for i, s in enumerate(seqs) if s.params.greedy]
+        greedy_idx = [i for i, s in enumerate(seqs)
+                      if s.params.greedy and not s.params.needs_logit_processing]
+        proc_greedy = [i for i, s in enumerate(seqs)
+                       if s.params.greedy and s.params.needs_logit_processing]
         rand_idx = [i for i, s in enumerate(seqs) if not s.params.greedy]
         if greedy_idx:
             ids = ops.greedy_sample(logits[greedy_idx])
             for j, i in enumerate(greedy_idx):
                 out[i] = int(ids[j])
+        for i in proc_greedy:
+            row = self._process_logits(logits[i].float(), seqs[i])
+            out[i] = int(row.argmax())
         if rand_idx:
             lg = logits[rand_idx].float()
             temps = torch.tensor(
@@ -39,7 +71,13 @@ class Sampler:
             probs = torch.softmax(lg, dim=-1)
             for j, i in enumerate(rand_idx):
                 p = seqs[i].params
+                if p.needs_logit_processing:
+                    r = self._process_logits(logits[i].float(), seqs[i])
+                    probs[j] = torch.softmax(r / max(p.temperature, 1e-5), dim=-1)
                 row = probs[j]
+                if p.min_p > 0.0:
+                    row = torch.where(row >= p.min_p * row.max(), row,
+                                      torch.zeros_like(row))
                 if p.top_k > 0 and p.top_k < row.shape[-1]:
                     vals, idx = torch.topk(row, p.top_k)
                     row = torch.zeros_like(row).scatter_(0, idx, vals)

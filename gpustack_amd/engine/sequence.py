@@ -17,11 +17,21 @@ class SamplingParams:
     temperature: float = 0.0        # 0 => greedy
     top_p: float = 1.0
     top_k: int = 0                  # 0 => disabled
+    min_p: float = 0.0              # vLLM-style min-p nucleus floor
+    presence_penalty: float = 0.0   # OpenAI semantics: subtract if seen
+    frequency_penalty: float = 0.0  # OpenAI semantics: subtract per count
+    repetition_penalty: float = 1.0  # HF semantics: divide/multiply seen
+    logit_bias: dict | None = None  # {token_id: bias}
     max_tokens: int = 128
     ignore_eos: bool = False
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None
     logprobs: bool = False
+
+    @property
+    def needs_logit_processing(self) -> bool:
+        return bool(self.logit_bias) or self.presence_penalty != 0.0 \
+            or self.frequency_penalty != 0.0 or self.repetition_penalty != 1.0
 
     @property
     def greedy(self) -> bool:

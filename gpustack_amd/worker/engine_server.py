@@ -185,6 +185,11 @@ def _sampling_params(body: dict, eos_token_id: int):
         if body.get("temperature") is not None else 1.0,
         top_p=float(body.get("top_p", 1.0)),
         top_k=int(body.get("top_k", 0)),
+        min_p=float(body.get("min_p", 0.0)),
+        presence_penalty=float(body.get("presence_penalty", 0.0)),
+        frequency_penalty=float(body.get("frequency_penalty", 0.0)),
+        repetition_penalty=float(body.get("repetition_penalty", 1.0)),
+        logit_bias=body.get("logit_bias"),
         max_tokens=int(mt),
         ignore_eos=bool(body.get("ignore_eos", False)),
         seed=body.get("seed"),
