@@ -38,6 +38,8 @@ def parse_args():
     ap.add_argument("--speculative", default=None, choices=["ngram", "eagle", "eagle3"],
                     help="speculative decoding method (BASELINE config 5)")
     ap.add_argument("--draft-tokens", type=int, default=3)
+    ap.add_argument("--prefix-caching", action="store_true",
+                    help="enable automatic prefix caching (shared-prefix workloads)")
     ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"],
                     help="KV cache dtype (fp8 e4m3 halves KV bytes; compute stays bf16)")
     ap.add_argument("--device", default=None)
@@ -83,6 +85,7 @@ def main():
         max_num_seqs=max(args.concurrency, 8),
         seed=0,
         kv_cache_dtype=args.kv_dtype,
+        enable_prefix_caching=args.prefix_caching,
         speculative=({"method": args.speculative,
                       "num_draft_tokens": args.draft_tokens}
                      if args.speculative else None),

@@ -178,6 +178,11 @@ class EngineConfig:
     model_dir: str | None = None
     # host-DRAM KV offload tier (extended_kv_cache in the reference schema)
     kv_offload_gb: float = 0.0
+    # automatic prefix caching: cache-hit prompts recompute only the suffix
+    # (through the paged-decode row path); suffixes longer than the cap
+    # fall back to full prefill (still registering blocks for later hits)
+    enable_prefix_caching: bool = False
+    prefix_cache_suffix_cap: int = 512
     # speculative decoding (reference speculative_config schema):
     # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
     speculative: dict | None = None

@@ -192,6 +192,7 @@ class LLMEngine:
         return (
             self._async_enabled
             and not batch.is_prefill
+            and not batch.is_suffix
             and batch.rows_per_seq == 1
             and all(s.params.greedy and not s.params.logprobs
                     and not s.params.needs_logit_processing
@@ -230,9 +231,9 @@ class LLMEngine:
             self._refill_tokens(batch)
         token_ids = self.runner.execute(batch)
         lps = getattr(self.runner, "last_logprobs", None)
-        if batch.is_prefill:
+        if batch.is_prefill or batch.is_suffix:
             self.scheduler.on_prefill_done(batch)
-        rps = 1 if batch.is_prefill else batch.rows_per_seq
+        rps = 1 if (batch.is_prefill or batch.is_suffix) else batch.rows_per_seq
         emitted_all: list[list[int]] = []
         for i, seq in enumerate(batch.seqs):
             if rps == 1:

@@ -100,6 +100,7 @@ class DecodeGraphRunner:
 
     def can_run(self, batch: ScheduledBatch) -> bool:
         return ((not batch.is_prefill)
+                and not batch.is_suffix
                 and len(batch.seqs) * batch.rows_per_seq <= self.max_bs
                 and bool(self.graphs))
 

@@ -38,6 +38,99 @@ class BlockAllocator:
         self.free_list.extend(reversed(blocks))
 
 
+class CachingBlockAllocator(BlockAllocator):
+    """Ref-counted allocator with a prefix cache (vLLM-style automatic
+    prefix caching, re-designed for this engine): blocks released by a
+    sequence stay resident (LRU-evictable) while registered under their
+    content hash; a later prompt sharing the prefix re-acquires them with
+    a refcount instead of recomputing the KV. Shared blocks are read-only
+    by construction: the block holding position n-1 of any prompt is never
+    registered, so every KV write lands in an owned block."""
+
+    def __init__(self, num_blocks: int):
+        super().__init__(num_blocks)
+        self.ref: dict[int, int] = {}
+        self.by_hash: dict[int, int] = {}
+        self.hash_of: dict[int, int] = {}
+        from collections import OrderedDict
+
+        self.lru: "OrderedDict[int, None]" = OrderedDict()
+        self.hits = 0
+        self.misses = 0
+
+    @property
+    def num_free(self) -> int:
+        return len(self.free_list) + len(self.lru)
+
+    def _evict_one(self) -> int:
+        blk, _ = self.lru.popitem(last=False)
+        h = self.hash_of.pop(blk, None)
+        if h is not None:
+            self.by_hash.pop(h, None)
+        return blk
+
+    def allocate(self, n: int) -> list[int]:
+        out: list[int] = []
+        for _ in range(n):
+            if self.free_list:
+                b = self.free_list.pop()
+            elif self.lru:
+                b = self._evict_one()
+            else:
+                self.free_list.extend(reversed(out))  # roll back
+                for b2 in out:
+                    self.ref.pop(b2, None)
+                raise RuntimeError("out of KV blocks")
+            self.ref[b] = 1
+            out.append(b)
+        return out
+
+    def free(self, blocks: list[int]) -> None:
+        for b in blocks:
+            r = self.ref.get(b, 1) - 1
+            if r > 0:
+                self.ref[b] = r
+                continue
+            self.ref.pop(b, None)
+            if b in self.hash_of:
+                self.lru[b] = None        # evictable, cache entry kept
+                self.lru.move_to_end(b)
+            else:
+                self.free_list.append(b)
+
+    def acquire_cached(self, h: int) -> int | None:
+        b = self.by_hash.get(h)
+        if b is None:
+            self.misses += 1
+            return None
+        if b in self.lru:                 # resurrect from evictable set
+            del self.lru[b]
+            self.ref[b] = 1
+        else:
+            self.ref[b] = self.ref.get(b, 0) + 1
+        self.hits += 1
+        return b
+
+    def register(self, block: int, h: int) -> None:
+        if h in self.by_hash or block in self.hash_of:
+            return  # first writer wins; duplicates stay plain
+        self.by_hash[h] = block
+        self.hash_of[block] = h
+
+
+def block_hashes(tokens, block_size: int) -> list[int]:
+    """Content-chain hashes for every FULL block of `tokens`."""
+    import zlib
+
+    out: list[int] = []
+    h = 0
+    for i in range(len(tokens) // block_size):
+        blk = tokens[i * block_size:(i + 1) * block_size]
+        h = zlib.crc32(repr(blk).encode(), h)
+        out.append(h)
+    return out
+
+
 class KVCache:
     def __init__(self, cfg: EngineConfig, num_blocks: int, device: str | torch.device):
         spec = cfg.spec
@@ -62,7 +155,10 @@ class KVCache:
         # last block reserved as the graph-padding scratch block (see
         # engine/graph_runner.py): padded rows write/read there, never live KV
         self.pad_block = num_blocks - 1
-        self.allocator = BlockAllocator(max(1, num_blocks - 1))
+        if getattr(cfg, "enable_prefix_caching", False):
+            self.allocator = CachingBlockAllocator(max(1, num_blocks - 1))
+        else:
+            self.allocator = BlockAllocator(max(1, num_blocks - 1))
 
         # ---- host-DRAM offload tier (extended_kv_cache) -----------------
         # Pinned host pool; swap-out/in run as async copies on a side HIP

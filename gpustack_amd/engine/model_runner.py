@@ -236,6 +236,28 @@ class ModelRunner:
                 seq_lens=dec_lens,
                 num_prefill_tokens=tp,
             )
+        elif batch.is_suffix:
+            # prefix-cache suffix rows: variable rows/seq through the paged
+            # decode path; only each seq's last row projects to logits
+            nrows = sum(batch.suffix_rows)
+            maxb = max(len(s.block_table) for s in batch.seqs)
+            bt = torch.zeros(nrows, maxb, dtype=torch.int32)
+            idx = []
+            r = 0
+            for s, nr in zip(batch.seqs, batch.suffix_rows):
+                row = torch.tensor(s.block_table, dtype=torch.int32)
+                for _ in range(nr):
+                    bt[r, : len(s.block_table)] = row
+                    r += 1
+                idx.append(r - 1)
+            meta = ForwardMeta(
+                is_prefill=False,
+                positions=positions,
+                slot_mapping=slots,
+                logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
+                block_tables=bt.to(dev),
+                seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
+            )
         else:
             rps = batch.rows_per_seq
             nrows = len(batch.seqs) * rps

@@ -39,6 +39,10 @@ class ScheduledBatch:
     n_prefill_seqs: int = 0
     num_prefill_tokens: int = 0
     decode_seq_lens: list[int] | None = None
+    # prefix-cache suffix batch: decode-style rows, variable count per seq,
+    # one sampled token per seq (engine/kv_cache.py CachingBlockAllocator)
+    is_suffix: bool = False
+    suffix_rows: list[int] | None = None   # rows per seq
 
     @property
     def num_tokens(self) -> int:
@@ -110,6 +114,10 @@ class Scheduler:
     # -- scheduling --------------------------------------------------------
     def schedule(self) -> ScheduledBatch | None:
         self._swap_in_ready()
+        if getattr(self.cfg, "enable_prefix_caching", False):
+            sb = self._schedule_suffix()
+            if sb is not None:
+                return sb
         batch = self._schedule_prefill()
         if batch is not None:
             import os
@@ -191,6 +199,81 @@ class Scheduler:
                 or _time.monotonic() - oldest >= self.cfg.admission_max_wait_s
                 or len(self.running) < self.cfg.admission_min_seqs)
 
+    def _cache_plan(self, seq: Sequence):
+        """(cached_blocks, hashes) for the longest cached prefix chain.
+        Never covers the block holding position n-1 (it will be written)."""
+        from .kv_cache import block_hashes
+
+        tokens = seq.all_token_ids
+        n = len(tokens)
+        bs = self.kv.block_size
+        hashes = block_hashes(tokens, bs)
+        max_cached = (n - 1) // bs
+        blocks: list[int] = []
+        for h in hashes[:max_cached]:
+            b = self.kv.allocator.acquire_cached(h)
+            if b is None:
+                break
+            blocks.append(b)
+        return blocks, hashes
+
+    def _schedule_suffix(self) -> ScheduledBatch | None:
+        """Admit cache-hit prompts: acquire shared prefix blocks and run
+        only the suffix tokens as decode-style rows."""
+        if not self.waiting:
+            return None
+        if self.running and not self.would_admit():
+            return None
+        batch = ScheduledBatch(is_prefill=False, is_suffix=True)
+        batch.suffix_rows = []
+        budget = self.cfg.max_prefill_tokens
+        toks: list[int] = []
+        poss: list[int] = []
+        slots: list[int] = []
+        lens: list[int] = []
+        scanned = 0
+        while (self.waiting and scanned < len(self.waiting) + 8
+               and len(self.running) + len(batch.seqs) < self.cfg.max_num_seqs):
+            seq = self.waiting[0]
+            n = seq.num_tokens
+            cached_blocks, hashes = self._cache_plan(seq)
+            suffix = n - len(cached_blocks) * self.kv.block_size
+            if (not cached_blocks
+                    or suffix > self.cfg.prefix_cache_suffix_cap):
+                self.kv.allocator.free(cached_blocks)  # drop refs
+                break  # head-of-line is a plain prefill; let it run first
+            if batch.seqs and sum(batch.suffix_rows) + suffix > budget:
+                self.kv.allocator.free(cached_blocks)
+                break
+            need = self.kv.blocks_needed(n) - len(cached_blocks)
+            try:
+                own = self.kv.allocator.allocate(need)
+            except RuntimeError:
+                self.kv.allocator.free(cached_blocks)
+                break
+            self.waiting.popleft()
+            scanned += 1
+            seq.block_table = cached_blocks + own
+            seq.block_hashes = hashes
+            seq.num_cached_tokens = len(cached_blocks) * self.kv.block_size
+            seq.status = SeqStatus.RUNNING
+            start = seq.num_cached_tokens
+            tokens = seq.all_token_ids
+            batch.seqs.append(seq)
+            batch.suffix_rows.append(suffix)
+            toks.extend(tokens[start:])
+            poss.extend(range(start, n))
+            slots.extend(self.kv.slots_for(seq.block_table, start, suffix))
+            lens.extend(range(start + 1, n + 1))  # row r attends 0..start+r
+        if not batch.seqs:
+            return None
+        batch.token_ids = toks
+        batch.positions = poss
+        batch.slot_mapping = slots
+        batch.seq_lens = lens
+        batch.n_prefill_seqs = len(batch.seqs)
+        return batch
+
     def _schedule_prefill(self) -> ScheduledBatch | None:
         if not self.waiting:
             return None
@@ -210,6 +293,11 @@ class Scheduler:
             seq.block_table = self.kv.allocator.allocate(nblocks)
             seq.num_cached_tokens = 0
             seq.status = SeqStatus.RUNNING
+            if getattr(self.cfg, "enable_prefix_caching", False):
+                from .kv_cache import block_hashes
+
+                seq.block_hashes = block_hashes(seq.all_token_ids,
+                                                self.kv.block_size)
             tokens = seq.all_token_ids
             batch.seqs.append(seq)
             batch.token_ids.extend(tokens)
@@ -318,8 +406,16 @@ class Scheduler:
     # -- lifecycle ---------------------------------------------------------
     def on_prefill_done(self, batch: ScheduledBatch) -> None:
         n = batch.n_prefill_seqs or len(batch.seqs)
+        caching = getattr(self.cfg, "enable_prefix_caching", False)
         for seq in batch.seqs[:n]:
             seq.num_cached_tokens = seq.num_tokens
+            if caching and getattr(seq, "block_hashes", None):
+                # register this prompt's full blocks (minus the one holding
+                # position n-1, which decode will write) for later sharing
+                limit = (seq.num_tokens - 1) // self.kv.block_size
+                for i, h in enumerate(seq.block_hashes[:limit]):
+                    self.kv.allocator.register(seq.block_table[i], h)
+                seq.block_hashes = None
             self.running.append(seq)
 
     def finish_seq(self, seq: Sequence, reason: str) -> None:

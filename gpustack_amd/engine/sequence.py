@@ -51,6 +51,7 @@ class Sequence:
     swap_outs: int = 0
     pending_tokens: int = 0         # async decode: sampled on device, not yet read back
     next_draft: list | None = None  # draft-model speculative window (engine/eagle.py)
+    block_hashes: list | None = None  # prefix-cache chain (engine/kv_cache.py)
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None
     finish_time: float | None = None

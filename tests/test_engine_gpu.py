@@ -127,3 +127,21 @@ def test_moe_gpu_deterministic_and_paths_agree():
     assert a == b, f"{a} != {b}"
     del eng2
     torch.cuda.empty_cache()
+
+
+def test_prefix_cache_gpu():
+    # cached-prefix suffix path on the HIP decode kernel: outputs identical
+    prefix = [11 + i for i in range(40)]
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    plain_eng = LLMEngine(_cfg())
+    plain = [plain_eng.generate([prefix + [100]], p)[0],
+             plain_eng.generate([prefix + [101, 102]], p)[0]]
+    del plain_eng
+    torch.cuda.empty_cache()
+    eng = LLMEngine(_cfg(enable_prefix_caching=True))
+    a = eng.generate([prefix + [100]], p)[0]
+    b = eng.generate([prefix + [101, 102]], p)[0]
+    assert eng.scheduler.kv.allocator.hits > 0
+    assert [a, b] == plain
+    del eng
+    torch.cuda.empty_cache()
