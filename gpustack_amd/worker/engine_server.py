@@ -237,6 +237,10 @@ def create_app(runner: EngineRunner) -> FastAPI:
 
     async def _generate(request: Request, body: dict, prompt_ids: list[int],
                         kind: str, echo_text_prefix: str = ""):
+        n = max(1, int(body.get("n", 1) or 1))
+        if n > 1:
+            return await _generate_n(request, body, prompt_ids, kind, n,
+                                     echo_text_prefix)
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
@@ -374,6 +378,129 @@ def create_app(runner: EngineRunner) -> FastAPI:
         if hasattr(ids, "ids"):
             ids = ids.ids
         return await _generate(request, body, list(ids), "chat")
+
+    async def _generate_n(request: Request, body: dict, prompt_ids: list[int],
+                          kind: str, n: int, echo_text_prefix: str = ""):
+        """OpenAI `n` choices: n independent engine requests batched by the
+        continuous-batching scheduler; seeded requests get seed+i so the
+        choices differ."""
+        import dataclasses
+
+        base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
+        stop_strs = _stop_strings(body)
+        subs = []
+        for i in range(n):
+            p = dataclasses.replace(
+                base, seed=(base.seed + i) if base.seed is not None else None)
+            subs.append(runner.submit(prompt_ids, p))
+        created = int(time.time())
+        model_name = runner.served_name
+        if body.get("stream"):
+            async def gen():
+                merged: asyncio.Queue = asyncio.Queue()
+
+                async def pump(idx, q):
+                    while True:
+                        item = await q.get()
+                        await merged.put((idx, item))
+                        if item.get("finished") or "error" in item:
+                            return
+
+                tasks = [asyncio.ensure_future(pump(i, q))
+                         for i, (_, q) in enumerate(subs)]
+                toks = [[] for _ in range(n)]
+                sent = [0] * n
+                done = 0
+                try:
+                    while done < n:
+                        if await request.is_disconnected():
+                            for rid, _ in subs:
+                                runner.abort(rid)
+                            return
+                        idx, item = await merged.get()
+                        if "error" in item:
+                            yield f"data: {json.dumps({'error': {'message': item['error']}})}\n\n"
+                            done += 1
+                            continue
+                        toks[idx].append(item["token_id"])
+                        text = runner.tokenizer.decode(toks[idx])
+                        new = text[sent[idx]:]
+                        fin = item.get("finished")
+                        if new and not new.endswith("�"):
+                            sent[idx] = len(text)
+                            delta = ({"content": new} if kind == "chat"
+                                     else None)
+                            ch = {"index": idx,
+                                  "finish_reason": (item.get("finish_reason")
+                                                    or "stop") if fin else None}
+                            if kind == "chat":
+                                ch["delta"] = delta
+                            else:
+                                ch["text"] = new
+                            payload = {"id": subs[0][0], "created": created,
+                                       "model": model_name,
+                                       "object": ("chat.completion.chunk"
+                                                  if kind == "chat"
+                                                  else "text_completion"),
+                                       "choices": [ch]}
+                            yield f"data: {json.dumps(payload)}\n\n"
+                        if fin:
+                            done += 1
+                    yield "data: [DONE]\n\n"
+                finally:
+                    for t in tasks:
+                        t.cancel()
+                    for rid, _ in subs:
+                        runner.release(rid)
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        choices = []
+        total_out = 0
+        try:
+            for i, (rid, q) in enumerate(subs):
+                toks = []
+                finish = "stop"
+                while True:
+                    item = await q.get()
+                    if "error" in item:
+                        raise HTTPException(500, item["error"])
+                    toks.append(item["token_id"])
+                    if stop_strs:
+                        t = runner.tokenizer.decode(toks)
+                        if any(ss in t for ss in stop_strs):
+                            runner.abort(rid)
+                            break
+                    if item["finished"]:
+                        finish = item.get("finish_reason") or "stop"
+                        break
+                text = runner.tokenizer.decode(toks)
+                if stop_strs:
+                    cuts = [text.find(ss) for ss in stop_strs if ss in text]
+                    if cuts:
+                        text = text[:min(cuts)]
+                total_out += len(toks)
+                if kind == "chat":
+                    choices.append({"index": i,
+                                    "message": {"role": "assistant",
+                                                "content": text},
+                                    "finish_reason": finish})
+                else:
+                    choices.append({"index": i,
+                                    "text": echo_text_prefix + text,
+                                    "finish_reason": finish})
+        finally:
+            for rid, _ in subs:
+                runner.release(rid)
+        usage = {"prompt_tokens": len(prompt_ids) * n,
+                 "completion_tokens": total_out,
+                 "total_tokens": len(prompt_ids) * n + total_out}
+        return JSONResponse({
+            "id": subs[0][0],
+            "object": "chat.completion" if kind == "chat" else "text_completion",
+            "created": created, "model": model_name,
+            "choices": choices, "usage": usage,
+        })
 
     @app.post("/v1/score")
     @app.post("/score")

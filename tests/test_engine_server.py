@@ -104,6 +104,16 @@ def test_engine_server_stop_strings():
         assert len(res) == 2
         assert res[0]["index"] == 0  # self-similarity ranks first
         assert res[0]["relevance_score"] >= res[1]["relevance_score"]
+        # n>1 choices (independent samples through the batcher)
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-s", "prompt": "hi", "max_tokens": 5, "n": 3,
+            "ignore_eos": True, "temperature": 0.8, "seed": 1}, timeout=60)
+        assert r.status_code == 200
+        ch = r.json()["choices"]
+        assert [c["index"] for c in ch] == [0, 1, 2]
+        assert r.json()["usage"]["completion_tokens"] == 15
+        assert len({c["text"] for c in ch}) > 1  # seeds differ per choice
+
         # score: same-text pair scores ~1.0
         r = httpx.post(f"http://127.0.0.1:{port}/v1/score", json={
             "text_1": "abcabc", "text_2": ["abcabc", "zzqq"]}, timeout=60)
