@@ -42,6 +42,26 @@ class Sampler:
                 t = int(tid)
                 if 0 <= t < row.shape[-1]:
                     row[t] += float(b)
+        if p.guided_token_seqs:
+            out = seq.output_token_ids
+            allowed = set()
+            done = False
+            for choice in p.guided_token_seqs:
+                c = list(choice)
+                if len(out) < len(c) and c[:len(out)] == out:
+                    allowed.add(c[len(out)])
+                elif c == out:
+                    done = True
+            mask = torch.full_like(row, float("-inf"))
+            if allowed and not done:
+                ids = torch.tensor(sorted(allowed), dtype=torch.long,
+                                   device=row.device)
+                mask[ids] = 0.0
+                row = row + mask
+            else:  # complete (or dead-end): force EOS
+                eos = getattr(seq.params, "_eos_token_id", 0)
+                mask[eos] = 0.0
+                row = row + mask
         return row
 
     def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:

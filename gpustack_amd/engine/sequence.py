@@ -22,6 +22,10 @@ class SamplingParams:
     frequency_penalty: float = 0.0  # OpenAI semantics: subtract per count
     repetition_penalty: float = 1.0  # HF semantics: divide/multiply seen
     logit_bias: dict | None = None  # {token_id: bias}
+    # guided decoding: output must be exactly one of these token sequences
+    # (constrained per-step logit masking; engine_server encodes the
+    # user-facing `guided_choice` strings)
+    guided_token_seqs: tuple | None = None
     max_tokens: int = 128
     ignore_eos: bool = False
     stop_token_ids: tuple[int, ...] = ()
@@ -30,7 +34,8 @@ class SamplingParams:
 
     @property
     def needs_logit_processing(self) -> bool:
-        return bool(self.logit_bias) or self.presence_penalty != 0.0 \
+        return bool(self.logit_bias) or bool(self.guided_token_seqs) \
+            or self.presence_penalty != 0.0 \
             or self.frequency_penalty != 0.0 or self.repetition_penalty != 1.0
 
     @property

@@ -176,11 +176,20 @@ def _stop_strings(body: dict) -> list[str]:
     return [s for s in stop if isinstance(s, str)]
 
 
-def _sampling_params(body: dict, eos_token_id: int):
+def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
     from ..engine import SamplingParams
 
     mt = body.get("max_tokens") or body.get("max_completion_tokens") or 256
-    return SamplingParams(
+    guided = None
+    if body.get("guided_choice") and tokenizer is not None:
+        seqs = []
+        for choice in body["guided_choice"]:
+            ids = tokenizer.encode(choice)
+            if hasattr(ids, "ids"):
+                ids = ids.ids
+            seqs.append(tuple(ids))
+        guided = tuple(seqs)
+    sp = SamplingParams(
         temperature=float(body.get("temperature", 1.0) or 0.0)
         if body.get("temperature") is not None else 1.0,
         top_p=float(body.get("top_p", 1.0)),
@@ -194,7 +203,13 @@ def _sampling_params(body: dict, eos_token_id: int):
         ignore_eos=bool(body.get("ignore_eos", False)),
         seed=body.get("seed"),
         logprobs=bool(body.get("logprobs")),
+        guided_token_seqs=guided,
     )
+    sp._eos_token_id = eos_token_id
+    if guided:
+        sp.ignore_eos = False
+        sp.max_tokens = max(len(c) for c in guided) + 1
+    return sp
 
 
 def create_app(runner: EngineRunner) -> FastAPI:
@@ -241,7 +256,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         if n > 1:
             return await _generate_n(request, body, prompt_ids, kind, n,
                                      echo_text_prefix)
-        params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
+        params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
+                                  runner.tokenizer)
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
         created = int(time.time())
@@ -386,7 +402,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         choices differ."""
         import dataclasses
 
-        base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id)
+        base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
+                                runner.tokenizer)
         stop_strs = _stop_strings(body)
         subs = []
         for i in range(n):

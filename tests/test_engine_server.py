@@ -114,6 +114,13 @@ def test_engine_server_stop_strings():
         assert r.json()["usage"]["completion_tokens"] == 15
         assert len({c["text"] for c in ch}) > 1  # seeds differ per choice
 
+        # guided_choice: output constrained to one of the given strings
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-s", "prompt": "pick", "max_tokens": 20,
+            "guided_choice": ["yes", "no"]}, timeout=60)
+        assert r.status_code == 200
+        assert r.json()["choices"][0]["text"] in ("yes", "no")
+
         # score: same-text pair scores ~1.0
         r = httpx.post(f"http://127.0.0.1:{port}/v1/score", json={
             "text_1": "abcabc", "text_2": ["abcabc", "zzqq"]}, timeout=60)
