@@ -59,8 +59,7 @@ class Sampler:
                 mask[ids] = 0.0
                 row = row + mask
             else:  # complete (or dead-end): force EOS
-                eos = getattr(seq.params, "_eos_token_id", 0)
-                mask[eos] = 0.0
+                mask[p.eos_token_id] = 0.0
                 row = row + mask
         return row
 

@@ -26,6 +26,7 @@ class SamplingParams:
     # (constrained per-step logit masking; engine_server encodes the
     # user-facing `guided_choice` strings)
     guided_token_seqs: tuple | None = None
+    eos_token_id: int = 0           # used by guided decoding to terminate
     max_tokens: int = 128
     ignore_eos: bool = False
     stop_token_ids: tuple[int, ...] = ()

@@ -204,8 +204,8 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         seed=body.get("seed"),
         logprobs=bool(body.get("logprobs")),
         guided_token_seqs=guided,
+        eos_token_id=eos_token_id,
     )
-    sp._eos_token_id = eos_token_id
     if guided:
         sp.ignore_eos = False
         sp.max_tokens = max(len(c) for c in guided) + 1

@@ -68,8 +68,8 @@ def test_min_p_restricts_support():
 def test_guided_choice_exact():
     eng = _eng()
     choices = ((41, 42, 43), (44, 45))
-    p = SamplingParams(max_tokens=10, guided_token_seqs=choices)
-    p._eos_token_id = 1
+    p = SamplingParams(max_tokens=10, guided_token_seqs=choices,
+                       eos_token_id=1)
     out = eng.generate([[1, 2, 3]], p)[0]
     body = out[:-1] if out and out[-1] == 1 else out
     assert tuple(body) in choices
@@ -77,7 +77,7 @@ def test_guided_choice_exact():
 
 def test_guided_choice_single_forces_sequence():
     eng = _eng()
-    p = SamplingParams(max_tokens=10, guided_token_seqs=((9, 8, 7, 6),))
-    p._eos_token_id = 1
+    p = SamplingParams(max_tokens=10, guided_token_seqs=((9, 8, 7, 6),),
+                       eos_token_id=1)
     out = eng.generate([[5]], p)[0]
     assert out[:4] == [9, 8, 7, 6]
