@@ -183,6 +183,12 @@ class EngineConfig:
     # fall back to full prefill (still registering blocks for later hits)
     enable_prefix_caching: bool = False
     prefix_cache_suffix_cap: int = 512
+    # chunked prefill (reference: vLLM --enable-chunked-prefill): a prompt
+    # longer than max_prefill_tokens is split into budget-sized chunks
+    # (chunk 0 = capped prefill batch, continuations = paged-decode rows),
+    # alternating 1:1 with decode steps so running sequences keep a bounded
+    # time-between-tokens during long-prompt admission
+    enable_chunked_prefill: bool = False
     # speculative decoding (reference speculative_config schema):
     # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
     speculative: dict | None = None
@@ -206,3 +212,9 @@ class EngineConfig:
             self.model_dir = self.model
             self.spec = ModelSpec.from_dir(self.model)
         self.max_model_len = min(self.max_model_len, self.spec.max_position_embeddings)
+        if self.enable_chunked_prefill and self.speculative and \
+                self.speculative.get("method") in ("eagle", "eagle3", "mtp"):
+            # draft-model speculation seeds from full-prompt prefill hiddens,
+            # which chunked admission does not produce in one step
+            raise ValueError("chunked prefill is incompatible with "
+                             "draft-model speculative decoding")

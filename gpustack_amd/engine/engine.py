@@ -178,6 +178,8 @@ class LLMEngine:
 
     def _refill_tokens(self, batch) -> None:
         """Replace placeholder tokens with the real (now drained) values."""
+        if batch.is_suffix:
+            return  # suffix/chunk rows carry prompt tokens, not placeholders
         if batch.is_prefill:
             base = batch.num_prefill_tokens
             dec = batch.seqs[batch.n_prefill_seqs:]
@@ -234,8 +236,12 @@ class LLMEngine:
         if batch.is_prefill or batch.is_suffix:
             self.scheduler.on_prefill_done(batch)
         rps = 1 if (batch.is_prefill or batch.is_suffix) else batch.rows_per_seq
+        finals = batch.chunk_final
         emitted_all: list[list[int]] = []
         for i, seq in enumerate(batch.seqs):
+            if finals is not None and i < len(finals) and not finals[i]:
+                emitted_all.append([])
+                continue  # intermediate prefill chunk: discard sampled token
             if rps == 1:
                 emitted = [token_ids[i]]
                 row0 = i

@@ -43,6 +43,10 @@ class ScheduledBatch:
     # one sampled token per seq (engine/kv_cache.py CachingBlockAllocator)
     is_suffix: bool = False
     suffix_rows: list[int] | None = None   # rows per seq
+    # chunked prefill: per prefill-seq flag — False means this step computes
+    # an intermediate chunk whose sampled token must be discarded and whose
+    # seq must not yet join the running set
+    chunk_final: list[bool] | None = None
 
     @property
     def num_tokens(self) -> int:
@@ -56,6 +60,10 @@ class Scheduler:
         self.waiting: deque[Sequence] = deque()
         self.running: list[Sequence] = []
         self.swapped: deque[Sequence] = deque()  # offloaded to host DRAM
+        # chunked prefill: the one sequence currently being admitted in
+        # budget-sized chunks; chunk steps alternate 1:1 with decode steps
+        self._chunking: Sequence | None = None
+        self._chunk_decode_turn = False
         self.proposer = None
         self.spec_k = 0
         self.spec_method = None
@@ -102,11 +110,19 @@ class Scheduler:
                 s.finish("abort")
                 self.swapped.remove(s)
                 return True
+        if (self._chunking is not None
+                and self._chunking.request_id == request_id):
+            s = self._chunking
+            self._chunking = None
+            self._release(s)
+            s.finish("abort")
+            return True
         return False
 
     @property
     def num_unfinished(self) -> int:
-        return len(self.waiting) + len(self.running) + len(self.swapped)
+        return (len(self.waiting) + len(self.running) + len(self.swapped)
+                + (1 if self._chunking is not None else 0))
 
     def has_work(self) -> bool:
         return self.num_unfinished > 0
@@ -114,6 +130,19 @@ class Scheduler:
     # -- scheduling --------------------------------------------------------
     def schedule(self) -> ScheduledBatch | None:
         self._swap_in_ready()
+        if self._chunking is not None:
+            # alternate chunk and decode steps so running seqs keep a
+            # bounded time-between-tokens during long-prompt admission
+            if self._chunk_decode_turn and self.running:
+                self._chunk_decode_turn = False
+                b = self._schedule_decode()
+                if b is not None:
+                    return b
+            b = self._schedule_chunk()
+            if b is not None:
+                self._chunk_decode_turn = True
+                return b
+            return self._schedule_decode()  # KV pressure: let decode drain
         if getattr(self.cfg, "enable_prefix_caching", False):
             sb = self._schedule_suffix()
             if sb is not None:
@@ -284,6 +313,29 @@ class Scheduler:
         while self.waiting and len(self.running) + len(batch.seqs) < self.cfg.max_num_seqs:
             seq = self.waiting[0]
             n = seq.num_tokens  # prompt + any generated tokens (preempted seqs)
+            if (not batch.seqs and n > budget
+                    and getattr(self.cfg, "enable_chunked_prefill", False)):
+                # chunk 0: prefill only the first budget tokens; the rest
+                # continues through _schedule_chunk (paged-decode rows)
+                nblocks = self.kv.blocks_needed(budget)
+                if nblocks > self.kv.allocator.num_free:
+                    break
+                self.waiting.popleft()
+                seq.block_table = self.kv.allocator.allocate(nblocks)
+                seq.num_cached_tokens = budget
+                seq.block_hashes = None
+                seq.status = SeqStatus.RUNNING
+                self._chunking = seq
+                self._chunk_decode_turn = True
+                tokens = seq.all_token_ids
+                batch.seqs.append(seq)
+                batch.token_ids.extend(tokens[:budget])
+                batch.positions.extend(range(budget))
+                batch.slot_mapping.extend(
+                    self.kv.slots_for(seq.block_table, 0, budget))
+                batch.seq_lens.append(budget)
+                batch.chunk_final = [False]
+                return batch  # chunk steps run exclusively
             if batch.seqs and batch.num_tokens + n > budget:
                 break
             nblocks = self.kv.blocks_needed(n)
@@ -306,6 +358,36 @@ class Scheduler:
             batch.seq_lens.append(n)
         if not batch.seqs:
             return None
+        return batch
+
+    def _schedule_chunk(self) -> ScheduledBatch | None:
+        """Continue the in-progress chunked prefill: the next budget-sized
+        window of prompt tokens runs as paged-decode rows (the same path
+        prefix-cache suffixes use — each row attends to all prior KV)."""
+        seq = self._chunking
+        p = seq.num_cached_tokens
+        n = seq.num_tokens
+        take = min(self.cfg.max_prefill_tokens, n - p)
+        need = self.kv.blocks_needed(p + take) - len(seq.block_table)
+        if need > 0:
+            try:
+                seq.block_table.extend(self.kv.allocator.allocate(need))
+            except RuntimeError:
+                return None  # pool exhausted; decode steps may free blocks
+        batch = ScheduledBatch(is_prefill=False, is_suffix=True)
+        tokens = seq.all_token_ids
+        batch.seqs = [seq]
+        batch.suffix_rows = [take]
+        batch.token_ids = tokens[p:p + take]
+        batch.positions = list(range(p, p + take))
+        batch.slot_mapping = self.kv.slots_for(seq.block_table, p, take)
+        batch.seq_lens = list(range(p + 1, p + take + 1))
+        batch.n_prefill_seqs = 1
+        final = p + take == n
+        batch.chunk_final = [final]
+        seq.num_cached_tokens = p + take
+        if final:
+            self._chunking = None
         return batch
 
     def _schedule_decode(self) -> ScheduledBatch | None:
@@ -407,7 +489,10 @@ class Scheduler:
     def on_prefill_done(self, batch: ScheduledBatch) -> None:
         n = batch.n_prefill_seqs or len(batch.seqs)
         caching = getattr(self.cfg, "enable_prefix_caching", False)
-        for seq in batch.seqs[:n]:
+        finals = batch.chunk_final
+        for i, seq in enumerate(batch.seqs[:n]):
+            if finals is not None and i < len(finals) and not finals[i]:
+                continue  # intermediate chunk: seq stays with the scheduler
             seq.num_cached_tokens = seq.num_tokens
             if caching and getattr(seq, "block_hashes", None):
                 # register this prompt's full blocks (minus the one holding
