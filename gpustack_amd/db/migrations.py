@@ -45,11 +45,16 @@ def _m004_instance_spec_hash(conn):
     _add_column(conn, "model_instances", "spec_hash", "VARCHAR(64) DEFAULT ''")
 
 
+def _m005_model_lora_adapters(conn):
+    _add_column(conn, "models", "lora_adapters", "JSON")
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
     (3, "instance cross-worker rank layout", _m003_instance_distributed_servers),
     (4, "instance spec_hash for update-triggered redeploy", _m004_instance_spec_hash),
+    (5, "model.lora_adapters for dynamic multi-LoRA", _m005_model_lora_adapters),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

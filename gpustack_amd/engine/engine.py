@@ -69,8 +69,36 @@ class LLMEngine:
         if len(prompt_token_ids) > self.cfg.max_model_len - 1:
             prompt_token_ids = prompt_token_ids[-(self.cfg.max_model_len - 1):]
         seq = Sequence(rid, list(prompt_token_ids), params)
+        if params.lora_name:
+            slot = self.runner.lora_bank.slot_of(params.lora_name)
+            if slot is None:
+                logger.warning("unknown LoRA adapter %r; serving base model",
+                               params.lora_name)
+            else:
+                seq.lora_slot = slot
         self.seqs[rid] = seq
         self.scheduler.add(seq)
+
+    # -- dynamic multi-LoRA (reference: vLLM /v1/load_lora_adapter, gpustack
+    # per-LoRA model routes) ------------------------------------------------
+    def add_lora(self, name: str, adapter_dir: str) -> None:
+        if self.comm.tp_size > 1:
+            assert self.comm.tp_rank == 0
+            self._pending_ops.append(("lora_add", name, adapter_dir))
+            return
+        self.runner.add_lora(name, adapter_dir)
+
+    def remove_lora(self, name: str) -> bool:
+        if self.comm.tp_size > 1:
+            assert self.comm.tp_rank == 0
+            self._pending_ops.append(("lora_rm", name))
+            return True
+        return self.runner.remove_lora(name)
+
+    def lora_names(self) -> list[str]:
+        names = list(self.runner.lora_bank.names())
+        names.extend(op[1] for op in self._pending_ops if op[0] == "lora_add")
+        return names
 
     def abort_request(self, request_id: str) -> bool:
         if self.comm.tp_size > 1:
@@ -105,6 +133,10 @@ class LLMEngine:
                 seq = self.seqs.pop(rid, None)
                 if seq is not None and self.runner.eagle is not None:
                     self.runner.eagle.drop(seq)
+            elif op[0] == "lora_add":
+                self.runner.add_lora(op[1], op[2])
+            elif op[0] == "lora_rm":
+                self.runner.remove_lora(op[1])
 
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()

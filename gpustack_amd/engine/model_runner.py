@@ -144,6 +144,9 @@ class ModelRunner:
 
                 logging.getLogger(__name__).info("merged LoRA %s (%d tensors)", d, n)
         self.sampler = Sampler(self.device)
+        from ..models.lora import LoraBank
+
+        self.lora_bank = LoraBank(cfg.spec, cfg)
         self.kv: KVCache | None = None
         # async decode pipeline buffers (pinned side is double-buffered:
         # step N+1's D2H must not overwrite step N before the host reads it)
@@ -216,6 +219,39 @@ class ModelRunner:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv
+
+    def add_lora(self, name: str, adapter_dir: str) -> int:
+        return self.lora_bank.add(name, adapter_dir, self.device,
+                                  getattr(torch, self.cfg.dtype))
+
+    def remove_lora(self, name: str) -> bool:
+        return self.lora_bank.remove(name)
+
+    def _row_lora_slots(self, batch: ScheduledBatch) -> list[int] | None:
+        """Per-row adapter slots, aligned with the batch's flat row order
+        (prefill tokens first, then decode rows). None when no row uses an
+        adapter — the common case pays one all() over the seq list."""
+        if not self.lora_bank.adapters:
+            return None
+        if all(s.lora_slot == 0 for s in batch.seqs):
+            return None
+        slots: list[int] = []
+        if batch.is_prefill:
+            n_pre = batch.n_prefill_seqs or len(batch.seqs)
+            for s, L in zip(batch.seqs[:n_pre], batch.seq_lens[:n_pre]):
+                slots.extend([s.lora_slot] * L)
+            slots.extend(s.lora_slot for s in batch.seqs[n_pre:])
+        elif batch.is_suffix:
+            for s, nr in zip(batch.seqs, batch.suffix_rows):
+                slots.extend([s.lora_slot] * nr)
+        else:
+            for s in batch.seqs:
+                slots.extend([s.lora_slot] * batch.rows_per_seq)
+        return slots
+
+    def batch_uses_lora(self, batch: ScheduledBatch) -> bool:
+        return bool(self.lora_bank.adapters) and any(
+            s.lora_slot != 0 for s in batch.seqs)
 
     def _meta(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
         dev = self.device
@@ -294,6 +330,11 @@ class ModelRunner:
                 block_tables=bt.to(dev),
                 seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
             )
+        slots = self._row_lora_slots(batch)
+        if slots is not None:
+            from ..models.lora import BatchLora
+
+            meta.lora = BatchLora.from_rows(self.lora_bank, slots, dev)
         return tokens, meta
 
     @torch.inference_mode()
@@ -339,7 +380,8 @@ class ModelRunner:
         token input device-side (no host round-trip at all)."""
         bs = len(batch.seqs)
         token_src = self._sampled_dev if reuse_tokens else None
-        if self.graph_runner is not None and self.graph_runner.can_run(batch):
+        if (self.graph_runner is not None and self.graph_runner.can_run(batch)
+                and not self.batch_uses_lora(batch)):
             logits = self.graph_runner.run(batch, token_src=token_src)
         else:
             tokens, meta = self._meta(batch)
@@ -367,7 +409,8 @@ class ModelRunner:
     @torch.inference_mode()
     def execute(self, batch: ScheduledBatch) -> list[int]:
         self.last_hidden = None
-        if self.graph_runner is not None and self.graph_runner.can_run(batch):
+        if (self.graph_runner is not None and self.graph_runner.can_run(batch)
+                and not self.batch_uses_lora(batch)):
             logits = self.graph_runner.run(batch)
         elif self.eagle is not None:
             # draft-model speculative: the verify step must also surface the

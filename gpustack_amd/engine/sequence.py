@@ -32,6 +32,9 @@ class SamplingParams:
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None
     logprobs: bool = False
+    # dynamic multi-LoRA: name of a live adapter (engine.add_lora) applied
+    # to this request's rows only (models/lora.py)
+    lora_name: str | None = None
 
     @property
     def needs_logit_processing(self) -> bool:
@@ -57,6 +60,7 @@ class Sequence:
     swap_outs: int = 0
     pending_tokens: int = 0         # async decode: sampled on device, not yet read back
     next_draft: list | None = None  # draft-model speculative window (engine/eagle.py)
+    lora_slot: int = 0              # 0 = no adapter (models/lora.py LoraBank)
     block_hashes: list | None = None  # prefix-cache chain (engine/kv_cache.py)
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None

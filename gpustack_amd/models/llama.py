@@ -42,6 +42,9 @@ class ForwardMeta:
     # mixed batches: rows [0, num_prefill_tokens) are prefill, the rest are
     # single-token decode rows
     num_prefill_tokens: int = 0
+    # dynamic multi-LoRA: per-batch adapter row groups (models/lora.py
+    # BatchLora); None when no row in the batch uses an adapter
+    lora: object | None = None
 
 
 class Attention(nn.Module):
@@ -61,6 +64,11 @@ class Attention(nn.Module):
             if spec.attention_bias else None
         )
         self.o_w = nn.Parameter(torch.empty(h, self.hq * self.d, dtype=dtype), requires_grad=False)
+        self.layer_idx = 0  # set by LlamaForCausalLM
+        nq, nk = self.hq * self.d, self.hkv * self.d
+        self._qkv_projs = [("q_proj", 0, nq), ("k_proj", nq, nk),
+                           ("v_proj", nq + nk, nk)]
+        self._o_projs = [("o_proj", 0, h)]
         if spec.qk_norm:
             self.q_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
             self.k_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
@@ -68,6 +76,9 @@ class Attention(nn.Module):
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
         T = x.shape[0]
         qkv = F.linear(x, self.qkv_w, self.qkv_b)
+        if meta.lora is not None:
+            qkv = qkv.contiguous()
+            meta.lora.apply(self.layer_idx, x, qkv, self._qkv_projs)
         nq, nk = self.hq * self.d, self.hkv * self.d
         # strided views into the fused buffer — the HIP kernels take row
         # strides, so no layout copies on the hot path
@@ -99,6 +110,8 @@ class Attention(nn.Module):
                 out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale
             )
         o = F.linear(out.view(T, -1), self.o_w)
+        if meta.lora is not None:
+            meta.lora.apply(self.layer_idx, out.view(T, -1), o, self._o_projs)
         return self.comm.all_reduce(o)
 
 
@@ -110,12 +123,20 @@ class MLP(nn.Module):
         self.i = spec.intermediate_size // tp_size
         self.gate_up_w = nn.Parameter(torch.empty(2 * self.i, h, dtype=dtype), requires_grad=False)
         self.down_w = nn.Parameter(torch.empty(h, self.i, dtype=dtype), requires_grad=False)
+        self.layer_idx = 0  # set by LlamaForCausalLM
+        self._gu_projs = [("gate_proj", 0, self.i), ("up_proj", self.i, self.i)]
+        self._down_projs = [("down_proj", 0, h)]
 
-    def forward(self, x):
+    def forward(self, x, meta: ForwardMeta | None = None):
         gu = F.linear(x, self.gate_up_w)
+        if meta is not None and meta.lora is not None:
+            meta.lora.apply(self.layer_idx, x, gu, self._gu_projs)
         act = torch.empty(x.shape[0], self.i, dtype=x.dtype, device=x.device)
         ops.silu_and_mul(act, gu)
-        return self.comm.all_reduce(ops.linear_auto(act, self.down_w))
+        down = ops.linear_auto(act, self.down_w)
+        if meta is not None and meta.lora is not None:
+            meta.lora.apply(self.layer_idx, act, down, self._down_projs)
+        return self.comm.all_reduce(down)
 
 
 class MoEMLP(nn.Module):
@@ -145,7 +166,9 @@ class MoEMLP(nn.Module):
         self.down_w = nn.Parameter(
             torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
 
-    def forward(self, x):
+    def forward(self, x, meta: ForwardMeta | None = None):
+        # MoE expert MLPs do not take LoRA (rejected at adapter load);
+        # attention adapters still apply upstream
         T = x.shape[0]
         logits = F.linear(x.float(), self.router_w.float())      # [T, E]
         if self.spec.norm_topk_prob:
@@ -236,7 +259,7 @@ class DecoderLayer(nn.Module):
             ops.fused_add_rms_norm(h, residual, self.input_norm, eps)
         a = self.attn(h, meta, cos_sin, k_cache, v_cache)
         ops.fused_add_rms_norm(a, residual, self.post_attn_norm, eps)
-        m = self.mlp(a)
+        m = self.mlp(a, meta)
         return m, residual
 
 
@@ -254,6 +277,9 @@ class LlamaForCausalLM(nn.Module):
         self.layers = nn.ModuleList(
             [DecoderLayer(spec, cfg.tp_size, comm, dtype) for _ in range(spec.num_layers)]
         )
+        for i, layer in enumerate(self.layers):
+            layer.attn.layer_idx = i
+            layer.mlp.layer_idx = i
         self.final_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
         if spec.tie_word_embeddings:
             self.lm_head = self.embed

@@ -68,6 +68,7 @@ class ModelCreate(BaseModel):
     restart_on_error: bool = True
     scaling_schedule: dict | None = None
     lora_list: list[str] | None = None
+    lora_adapters: list[dict] | None = None  # [{name, path}] dynamic LoRA
 
 
 class ModelUpdate(BaseModel):

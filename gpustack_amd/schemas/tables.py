@@ -136,6 +136,10 @@ class Model(Base, TimestampMixin, SerializeMixin):
     distributed_inference_across_workers = Column(Boolean, default=False)
     restart_on_error = Column(Boolean, default=True)
     lora_list = Column(JSON, default=None)  # adapter dirs merged at load
+    # dynamic multi-LoRA: [{"name": ..., "path": ...}] served unmerged and
+    # routable by adapter name (reference: per-LoRA child model routes,
+    # gpustack/server/lora_model_routes.py)
+    lora_adapters = Column(JSON, default=None)
     # cron-window autoscaling (reference: schemas/models.py:237-321):
     # {"rules": [{"cron": "0 9 * * 1-5", "duration_minutes": 60, "replicas": 4}]}
     scaling_schedule = Column(JSON, default=None)

@@ -47,6 +47,7 @@ def model_spec_hash(model) -> str:
         "speculative_config": model.speculative_config,
         "extended_kv_cache": model.extended_kv_cache,
         "lora_list": model.lora_list,
+        "lora_adapters": model.lora_adapters,
         "distributed": model.distributed_inference_across_workers,
     }, sort_keys=True, default=str)
     return hashlib.sha256(payload.encode()).hexdigest()[:16]

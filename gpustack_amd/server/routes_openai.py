@@ -44,15 +44,24 @@ OPENAI_PATHS = [
 ]
 
 
-def _resolve_model_name(name: str) -> str:
-    """Model-route indirection with weighted targets."""
+def _resolve_model_name(name: str) -> tuple[str, str | None]:
+    """Model-route indirection with weighted targets, plus per-LoRA child
+    routes (reference: gpustack/server/lora_model_routes.py): requesting an
+    adapter name from some model's `lora_adapters` routes to that model's
+    instances with the adapter applied per-request. Returns
+    (target model name, lora adapter name or None)."""
     with get_session() as s:
         route = s.query(ModelRoute).filter_by(name=name).first()
         if route and route.targets:
             weights = [t.get("weight", 1) for t in route.targets]
             pick = random.choices(route.targets, weights=weights)[0]
-            return pick.get("model_name", name)
-    return name
+            return pick.get("model_name", name), None
+        if s.query(Model).filter_by(name=name).first() is None:
+            for m in s.query(Model).filter(Model.lora_adapters.isnot(None)).all():
+                for ad in m.lora_adapters or []:
+                    if ad.get("name") == name:
+                        return m.name, name
+    return name, None
 
 
 def _find_provider(model_name: str):
@@ -117,7 +126,7 @@ async def _proxy(request: Request, path: str, user: User):
     name = body.get("model")
     if not name:
         raise HTTPException(400, "missing 'model'")
-    target_name = _resolve_model_name(name)
+    target_name, lora_name = _resolve_model_name(name)
     headers = {}
     with get_session() as s:
         local = s.query(Model).filter_by(name=target_name).first() is not None
@@ -140,6 +149,8 @@ async def _proxy(request: Request, path: str, user: User):
         if provider.get("api_key"):
             headers["Authorization"] = f"Bearer {provider['api_key']}"
     body["model"] = target_name
+    if lora_name:
+        body["lora_name"] = lora_name
     stream = bool(body.get("stream"))
     if tunnel_worker is not None:
         return await _proxy_via_tunnel(tunnel_worker, path, body, user, model, stream)
@@ -242,6 +253,11 @@ def list_models_v1(user: User = Depends(get_current_user)):
             {"id": m.name, "object": "model", "created": int(m.created_at),
              "owned_by": "gpustack_amd"}
             for m in models
+        ] + [
+            {"id": ad["name"], "object": "model", "created": int(m.created_at),
+             "owned_by": "gpustack_amd", "parent": m.name}
+            for m in models for ad in (m.lora_adapters or [])
+            if ad.get("name")
         ] + [
             {"id": r.name, "object": "model", "created": int(r.created_at),
              "owned_by": "gpustack_amd/route"}

@@ -212,6 +212,17 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
     return sp
 
 
+def _apply_lora_routing(sp, body: dict, runner: "EngineRunner") -> None:
+    """Per-request adapter resolution (reference: per-LoRA child model
+    routes, gpustack/server/lora_model_routes.py): requesting a live
+    adapter's name as `model` — or an explicit `lora_name` — serves the
+    base model with that adapter applied to this request's rows."""
+    name = body.get("lora_name") or body.get("model")
+    if name and name != runner.served_name \
+            and name in runner.engine.lora_names():
+        sp.lora_name = name
+
+
 def create_app(runner: EngineRunner) -> FastAPI:
     app = FastAPI(title="gpustack_amd-engine")
 
@@ -225,9 +236,40 @@ def create_app(runner: EngineRunner) -> FastAPI:
 
     @app.get("/v1/models")
     async def models():
-        return {"object": "list", "data": [
-            {"id": runner.served_name, "object": "model", "owned_by": "gpustack_amd"}
-        ]}
+        data = [{"id": runner.served_name, "object": "model",
+                 "owned_by": "gpustack_amd"}]
+        data.extend({"id": n, "object": "model", "owned_by": "gpustack_amd",
+                     "parent": runner.served_name}
+                    for n in runner.engine.lora_names())
+        return {"object": "list", "data": data}
+
+    # vLLM-compatible dynamic adapter management (reference: gpustack mounts
+    # per-LoRA child routes over these, server/lora_model_routes.py)
+    @app.post("/v1/load_lora_adapter")
+    async def load_lora(request: Request):
+        body = await request.json()
+        name, path = body.get("lora_name"), body.get("lora_path")
+        if not name or not path:
+            raise HTTPException(400, "lora_name and lora_path are required")
+        if name == runner.served_name:
+            raise HTTPException(400, "adapter name collides with served model")
+        try:
+            # on the engine thread: the bank mutates between steps only
+            await runner.run_aux(runner.engine.add_lora, name, path)
+        except (FileNotFoundError, ValueError) as e:
+            raise HTTPException(400, str(e))
+        return {"status": "ok", "lora_name": name}
+
+    @app.post("/v1/unload_lora_adapter")
+    async def unload_lora(request: Request):
+        body = await request.json()
+        name = body.get("lora_name")
+        if not name:
+            raise HTTPException(400, "lora_name is required")
+        ok = await runner.run_aux(runner.engine.remove_lora, name)
+        if not ok:
+            raise HTTPException(404, f"no adapter named {name!r}")
+        return {"status": "ok", "lora_name": name}
 
     @app.get("/metrics")
     async def metrics():
@@ -258,6 +300,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
                                      echo_text_prefix)
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                   runner.tokenizer)
+        _apply_lora_routing(params, body, runner)
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
         created = int(time.time())
@@ -404,6 +447,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
 
         base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                 runner.tokenizer)
+        _apply_lora_routing(base, body, runner)
         stop_strs = _stop_strings(body)
         subs = []
         for i in range(n):
@@ -597,6 +641,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "stop": body.get("stop_sequences"),
         }
         params = _sampling_params(oa, runner.engine.cfg.spec.eos_token_id)
+        _apply_lora_routing(params, body, runner)
         stop_strs = _stop_strings(oa)
         rid, q = runner.submit(ids, params)
         model_name = runner.served_name
@@ -902,6 +947,12 @@ def main():
                 _time.sleep(0.02)
 
     runner = EngineRunner(ecfg, args.served_name, comm)
+    for ad in (extra.get("lora_adapters") or []):
+        try:
+            runner.engine.add_lora(ad["name"], ad["path"])
+            logger.info("mounted LoRA adapter %s from %s", ad["name"], ad["path"])
+        except Exception:  # noqa: BLE001
+            logger.exception("failed to load LoRA adapter %s", ad.get("name"))
     app = create_app(runner)
     import uvicorn
 

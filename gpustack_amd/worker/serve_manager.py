@@ -175,6 +175,8 @@ class ServeManager:
         bp = dict(model.get("backend_parameters") or {})
         if model.get("lora_list"):
             bp.setdefault("lora_dirs", model["lora_list"])
+        if model.get("lora_adapters"):
+            bp.setdefault("lora_adapters", model["lora_adapters"])
         args = [
             sys.executable, "-m", "gpustack_amd.worker.engine_server",
             "--served-name", model["name"],

@@ -228,8 +228,25 @@ def test_model_route_resolution(server):
     assert r.status_code == 201
     from gpustack_amd.server.routes_openai import _resolve_model_name
 
-    assert _resolve_model_name("public-name") == "backend-a"
-    assert _resolve_model_name("backend-a") == "backend-a"
+    assert _resolve_model_name("public-name") == ("backend-a", None)
+    assert _resolve_model_name("backend-a") == ("backend-a", None)
+
+
+def test_lora_adapter_route_resolution(server):
+    """Per-LoRA child routes: an adapter name from Model.lora_adapters
+    resolves to the parent model with the adapter name attached."""
+    client, app, cfg, reg = server
+    r = client.post("/v2/models", json={
+        "name": "base-m", "model_ref": "tiny",
+        "lora_adapters": [{"name": "sql-tuned", "path": "/adapters/sql"}]})
+    assert r.status_code == 201
+    from gpustack_amd.server.routes_openai import _resolve_model_name
+
+    assert _resolve_model_name("sql-tuned") == ("base-m", "sql-tuned")
+    assert _resolve_model_name("base-m") == ("base-m", None)
+    # adapters appear in /v1/models with their parent
+    items = {m["id"]: m for m in client.get("/v1/models").json()["data"]}
+    assert items["sql-tuned"]["parent"] == "base-m"
 
 
 def test_model_provider_routing(server):
