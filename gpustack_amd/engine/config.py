@@ -194,6 +194,8 @@ class EngineConfig:
     speculative: dict | None = None
     # LoRA adapters merged into the weights at load (reference lora_list)
     lora_dirs: list[str] = field(default_factory=list)
+    # GGUF checkpoint execution: dequantized to bf16 at load (utils/gguf.py)
+    gguf_path: str | None = None
     # admission hysteresis: open a prefill step only when this many requests
     # wait, one has waited admission_max_wait_s, or nothing is running —
     # keeps steady-state decode on the hipGraph path instead of degrading
@@ -208,6 +210,11 @@ class EngineConfig:
             self.spec = PRESETS[self.model]
         elif self.model_dir:
             self.spec = ModelSpec.from_dir(self.model_dir)
+        elif self.model.endswith(".gguf") and Path(self.model).is_file():
+            from ..utils.gguf import spec_from_gguf
+
+            self.gguf_path = self.model
+            self.spec = spec_from_gguf(self.model)
         elif Path(self.model).is_dir():
             self.model_dir = self.model
             self.spec = ModelSpec.from_dir(self.model)

@@ -131,7 +131,11 @@ class ModelRunner:
             self._init_tunableop()
         self.graph_runner = None
         self.model = LlamaForCausalLM(cfg, self.comm, self.device)
-        if cfg.model_dir and not cfg.enforce_random_weights:
+        if cfg.gguf_path and not cfg.enforce_random_weights:
+            from ..models.weights import load_gguf
+
+            load_gguf(self.model, cfg, cfg.gguf_path)
+        elif cfg.model_dir and not cfg.enforce_random_weights:
             load_safetensors(self.model, cfg, cfg.model_dir)
         else:
             random_init(self.model, cfg)

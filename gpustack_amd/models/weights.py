@@ -287,3 +287,76 @@ def merge_lora(model, cfg: EngineConfig, adapter_dir: str | Path) -> int:
             w += shard.to(w.dtype).to(w.device)
             merged += 1
     return merged
+
+def load_gguf(model, cfg: EngineConfig, gguf_path: str | Path) -> None:
+    """Load a GGUF checkpoint (llama.cpp naming: blk.N.attn_q / ffn_gate /
+    token_embd / output ...) into the fused bf16 serving layout, dequantizing
+    Q8_0/Q4_0/Q4_1/Q4_K/Q6_K blocks at load time (utils/gguf.py).
+
+    Reference parity: the reference schedules GGUF models but delegates
+    execution to llama-box containers; here the first-
+    party engine executes them on the same MFMA bf16 path as safetensors.
+
+    llama-arch GGUFs store q/k projections PERMUTED for ggml's interleaved
+    rope (convert_hf_to_gguf.py LlamaModel.permute); our rope kernel uses
+    the HF half-split convention, so those rows are un-permuted here.
+    qwen2/qwen3-arch GGUFs use neox rope and need no permute.
+    """
+    from ..utils.gguf import read_gguf, read_tensor
+
+    spec = model.spec
+    tp, rank = cfg.tp_size, cfg.tp_rank
+    d = spec.head_dim
+    hq, hkv = spec.num_heads // tp, max(1, spec.num_kv_heads // tp)
+    i_loc = spec.intermediate_size // tp
+    info = read_gguf(gguf_path)
+    by_name = {t.name: t for t in info.tensors}
+    permuted_qk = info.architecture == "llama"
+
+    def get(name, unpermute_heads: int | None = None):
+        t = by_name[name]
+        w = read_tensor(gguf_path, info, t)
+        if unpermute_heads:
+            # inverse of convert_hf_to_gguf permute:
+            #   permute  = reshape(h, 2, d/2, in).swapaxes(1, 2)
+            #   inverse  = reshape(h, d/2, 2, in).swapaxes(1, 2)
+            nh = unpermute_heads
+            w = w.reshape(nh, d // 2, 2, *w.shape[1:]).transpose(1, 2) \
+                 .reshape(nh * d, *w.shape[1:])
+        return w.to(model.dtype)
+
+    def row_shard(t, per):
+        return t[rank * per:(rank + 1) * per]
+
+    model.embed.copy_(get("token_embd.weight"))
+    model.final_norm.copy_(get("output_norm.weight"))
+    if not spec.tie_word_embeddings:
+        name = "output.weight" if "output.weight" in by_name else "token_embd.weight"
+        model.lm_head.copy_(get(name))
+    up_q = spec.num_heads if permuted_qk else None
+    up_k = spec.num_kv_heads if permuted_qk else None
+    for li, layer in enumerate(model.layers):
+        p = f"blk.{li}."
+        q = row_shard(get(p + "attn_q.weight", up_q), hq * d)
+        k = row_shard(get(p + "attn_k.weight", up_k), hkv * d)
+        v = row_shard(get(p + "attn_v.weight"), hkv * d)
+        layer.attn.qkv_w.copy_(torch.cat([q, k, v]))
+        if layer.attn.qkv_b is not None:
+            layer.attn.qkv_b.copy_(torch.cat([
+                row_shard(get(p + "attn_q.bias", up_q), hq * d),
+                row_shard(get(p + "attn_k.bias", up_k), hkv * d),
+                row_shard(get(p + "attn_v.bias"), hkv * d),
+            ]))
+        o = get(p + "attn_output.weight")
+        layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
+        if spec.qk_norm:
+            layer.attn.q_norm.copy_(get(p + "attn_q_norm.weight"))
+            layer.attn.k_norm.copy_(get(p + "attn_k_norm.weight"))
+        layer.mlp.gate_up_w.copy_(torch.cat([
+            row_shard(get(p + "ffn_gate.weight"), i_loc),
+            row_shard(get(p + "ffn_up.weight"), i_loc),
+        ]))
+        dn = get(p + "ffn_down.weight")
+        layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
+        layer.input_norm.copy_(get(p + "attn_norm.weight"))
+        layer.post_attn_norm.copy_(get(p + "ffn_norm.weight"))

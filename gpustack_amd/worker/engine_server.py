@@ -929,8 +929,11 @@ def main():
     cfg_kwargs.update({k: v for k, v in extra.items() if k in EngineConfig.__dataclass_fields__})
     ecfg = EngineConfig(**cfg_kwargs)
     if args.source == "local_path":
-        ecfg.model_dir = args.model_ref
         ecfg.enforce_random_weights = False
+        if args.model_ref.endswith(".gguf"):
+            ecfg.gguf_path = args.model_ref  # dequant-on-load execution
+        else:
+            ecfg.model_dir = args.model_ref
 
     if args.tp > 1 and args.tp_rank != 0:
         # follower rank: no HTTP; run the coordinated engine loop forever

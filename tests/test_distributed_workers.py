@@ -80,7 +80,7 @@ def test_cross_worker_tp2():
 
         state = None
         inst = None
-        for _ in range(240):
+        for _ in range(480):  # generous: full-suite CPU load slows gloo rendezvous
             insts = client.get("/v2/model_instances").json()["items"]
             if insts:
                 inst = insts[0]

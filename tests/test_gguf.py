@@ -88,3 +88,223 @@ def test_bad_magic(tmp_path):
 def test_unknown_type_fallback():
     t = GGUFTensorInfo("x", (100,), 99, 0)
     assert t.nbytes == 100  # 1 byte/element fallback
+
+
+# ---- execution: dequantization + load_gguf ---------------------------------
+
+def test_q8_0_roundtrip():
+    import numpy as np
+
+    from gpustack_amd.utils.gguf import GGML_Q8_0, dequantize, quantize_q8_0
+
+    rng = np.random.default_rng(0)
+    x = rng.standard_normal(32 * 40).astype(np.float32)
+    y = dequantize(quantize_q8_0(x), GGML_Q8_0, x.size)
+    step = np.abs(x).reshape(-1, 32).max(axis=1) / 127.0
+    assert np.abs(y - x).reshape(-1, 32).max(axis=1).max() <= step.max() * 0.51
+
+
+def test_q4_0_roundtrip():
+    import numpy as np
+
+    from gpustack_amd.utils.gguf import GGML_Q4_0, dequantize, quantize_q4_0
+
+    rng = np.random.default_rng(1)
+    x = rng.standard_normal(32 * 40).astype(np.float32)
+    y = dequantize(quantize_q4_0(x), GGML_Q4_0, x.size)
+    step = np.abs(x).reshape(-1, 32).max(axis=1) / 8.0
+    err = np.abs(y - x).reshape(-1, 32)
+    assert (err <= step[:, None] * 1.01).all()
+
+
+def _scalar_q4_k(block):
+    """Direct translation of ggml dequantize_row_q4_K (independent ref)."""
+    import numpy as np
+
+    d = float(np.frombuffer(block[0:2], np.float16)[0])
+    dmin = float(np.frombuffer(block[2:4], np.float16)[0])
+    scales = block[4:16]
+    qs = block[16:144]
+
+    def get_scale_min(j):
+        if j < 4:
+            return scales[j] & 63, scales[j + 4] & 63
+        return ((scales[j + 4] & 0xF) | ((scales[j - 4] >> 6) << 4),
+                (scales[j + 4] >> 4) | ((scales[j] >> 6) << 4))
+
+    y = [0.0] * 256
+    idx, is_ = 0, 0
+    for j in range(0, 256, 64):
+        sc1, m1 = get_scale_min(is_)
+        sc2, m2 = get_scale_min(is_ + 1)
+        for l in range(32):
+            y[j + l] = d * sc1 * (qs[idx + l] & 0xF) - dmin * m1
+            y[j + 32 + l] = d * sc2 * (qs[idx + l] >> 4) - dmin * m2
+        idx += 32
+        is_ += 2
+    return y
+
+
+def _scalar_q6_k(block):
+    """Direct translation of ggml dequantize_row_q6_K."""
+    import numpy as np
+
+    ql = block[0:128]
+    qh = block[128:192]
+    sc = np.frombuffer(block[192:208], np.int8)
+    d = float(np.frombuffer(block[208:210], np.float16)[0])
+    y = [0.0] * 256
+    for half in range(2):
+        yb, qlb, qhb, scb = 128 * half, 64 * half, 32 * half, 8 * half
+        for l in range(32):
+            is_ = l // 16
+            q1 = ((ql[qlb + l] & 0xF) | (((qh[qhb + l] >> 0) & 3) << 4)) - 32
+            q2 = ((ql[qlb + l + 32] & 0xF) | (((qh[qhb + l] >> 2) & 3) << 4)) - 32
+            q3 = ((ql[qlb + l] >> 4) | (((qh[qhb + l] >> 4) & 3) << 4)) - 32
+            q4 = ((ql[qlb + l + 32] >> 4) | (((qh[qhb + l] >> 6) & 3) << 4)) - 32
+            y[yb + l] = d * int(sc[scb + is_ + 0]) * q1
+            y[yb + l + 32] = d * int(sc[scb + is_ + 2]) * q2
+            y[yb + l + 64] = d * int(sc[scb + is_ + 4]) * q3
+            y[yb + l + 96] = d * int(sc[scb + is_ + 6]) * q4
+    return y
+
+
+def test_q4_k_matches_scalar_reference():
+    import numpy as np
+
+    from gpustack_amd.utils.gguf import GGML_Q4_K, dequantize
+
+    rng = np.random.default_rng(2)
+    nb = 5
+    blocks = bytearray()
+    for _ in range(nb):
+        blocks += np.float16(rng.uniform(0.01, 0.1)).tobytes()
+        blocks += np.float16(rng.uniform(0.01, 0.1)).tobytes()
+        blocks += rng.integers(0, 256, 12, dtype=np.uint8).tobytes()
+        blocks += rng.integers(0, 256, 128, dtype=np.uint8).tobytes()
+    got = dequantize(bytes(blocks), GGML_Q4_K, nb * 256)
+    want = np.array([v for i in range(nb)
+                     for v in _scalar_q4_k(bytes(blocks[144 * i:144 * (i + 1)]))],
+                    dtype=np.float32)
+    assert np.allclose(got, want, atol=1e-5)
+
+
+def test_q6_k_matches_scalar_reference():
+    import numpy as np
+
+    from gpustack_amd.utils.gguf import GGML_Q6_K, dequantize
+
+    rng = np.random.default_rng(3)
+    nb = 5
+    blocks = bytearray()
+    for _ in range(nb):
+        blocks += rng.integers(0, 256, 128, dtype=np.uint8).tobytes()  # ql
+        blocks += rng.integers(0, 256, 64, dtype=np.uint8).tobytes()   # qh
+        blocks += rng.integers(-60, 60, 16, dtype=np.int8).tobytes()   # scales
+        blocks += np.float16(rng.uniform(0.01, 0.1)).tobytes()         # d
+    got = dequantize(bytes(blocks), GGML_Q6_K, nb * 256)
+    want = np.array([v for i in range(nb)
+                     for v in _scalar_q6_k(bytes(blocks[210 * i:210 * (i + 1)]))],
+                    dtype=np.float32)
+    assert np.allclose(got, want, atol=1e-5)
+
+
+def _export_tiny_gguf(path, eng, quant=None):
+    """Export a tiny engine's weights as a llama-arch GGUF, applying the
+    convert_hf_to_gguf q/k permute (the loader must undo it)."""
+    import numpy as np
+
+    from gpustack_amd.utils.gguf import (
+        GGML_F32, GGML_Q8_0, quantize_q8_0, write_gguf_with_data,
+    )
+
+    spec = eng.cfg.spec
+    d = spec.head_dim
+
+    def permute(w, nh):
+        return (w.reshape(nh, 2, d // 2, w.shape[-1])
+                 .transpose(1, 2).reshape(nh * d, w.shape[-1]))
+
+    tensors = []
+
+    def add(name, w, q=False):
+        a = w.float().cpu().numpy()
+        shape = tuple(reversed(a.shape))
+        if q:
+            tensors.append((name, shape, GGML_Q8_0, quantize_q8_0(a.reshape(-1))))
+        else:
+            tensors.append((name, shape, GGML_F32,
+                            np.ascontiguousarray(a).tobytes()))
+
+    m = eng.runner.model
+    add("token_embd.weight", m.embed)
+    add("output_norm.weight", m.final_norm)
+    add("output.weight", m.lm_head)
+    nq, nk = spec.num_heads * d, spec.num_kv_heads * d
+    for li, layer in enumerate(m.layers):
+        p = f"blk.{li}."
+        qkv = layer.attn.qkv_w.data
+        add(p + "attn_q.weight", permute(qkv[:nq], spec.num_heads), quant)
+        add(p + "attn_k.weight", permute(qkv[nq:nq + nk], spec.num_kv_heads), quant)
+        add(p + "attn_v.weight", qkv[nq + nk:], quant)
+        add(p + "attn_output.weight", layer.attn.o_w, quant)
+        i = spec.intermediate_size
+        add(p + "ffn_gate.weight", layer.mlp.gate_up_w[:i], quant)
+        add(p + "ffn_up.weight", layer.mlp.gate_up_w[i:], quant)
+        add(p + "ffn_down.weight", layer.mlp.down_w, quant)
+        add(p + "attn_norm.weight", layer.input_norm)
+        add(p + "ffn_norm.weight", layer.post_attn_norm)
+    meta = {
+        "general.architecture": "llama",
+        "llama.attention.head_count": spec.num_heads,
+        "llama.attention.head_count_kv": spec.num_kv_heads,
+        "llama.attention.key_length": spec.head_dim,
+        "llama.embedding_length": spec.hidden_size,
+        "llama.feed_forward_length": spec.intermediate_size,
+        "llama.block_count": spec.num_layers,
+        "llama.vocab_size": spec.vocab_size,
+        "llama.context_length": 512,
+        "llama.rope.freq_base": float(spec.rope_theta),
+        "llama.attention.layer_norm_rms_epsilon": float(spec.rms_norm_eps),
+    }
+    write_gguf_with_data(path, meta, tensors)
+
+
+def test_load_gguf_f32_exact(tmp_path):
+    """F32 GGUF of the tiny model's weights reproduces its outputs exactly
+    (incl. the llama q/k un-permute)."""
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    ref = LLMEngine(EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64))
+    path = tmp_path / "tiny-f32.gguf"
+    _export_tiny_gguf(path, ref)
+    eng = LLMEngine(EngineConfig(model=str(path), device="cpu",
+                                 kv_cache_blocks=64,
+                                 enforce_random_weights=False))
+    assert eng.cfg.gguf_path == str(path)
+    assert eng.cfg.spec.num_layers == ref.cfg.spec.num_layers
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    assert eng.generate([[1, 2, 3, 4, 5]], p) == ref.generate([[1, 2, 3, 4, 5]], p)
+
+
+def test_load_gguf_q8_0_close(tmp_path):
+    """Q8_0-quantized projections: weights land within quantization error
+    and the engine decodes."""
+    import torch
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    ref = LLMEngine(EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64))
+    path = tmp_path / "tiny-q8.gguf"
+    _export_tiny_gguf(path, ref, quant=True)
+    eng = LLMEngine(EngineConfig(model=str(path), device="cpu",
+                                 kv_cache_blocks=64,
+                                 enforce_random_weights=False))
+    w_ref = ref.runner.model.layers[0].mlp.down_w.float()
+    w_got = eng.runner.model.layers[0].mlp.down_w.float()
+    scale = w_ref.abs().max()
+    assert (w_got - w_ref).abs().max() <= scale / 127 + 1e-2
+    assert torch.equal(ref.runner.model.embed, eng.runner.model.embed)
+    out = eng.generate([[1, 2, 3, 4, 5]], SamplingParams(max_tokens=8,
+                                                         ignore_eos=True))[0]
+    assert len(out) == 8
