@@ -95,6 +95,11 @@ class LLMEngine:
             return True
         return self.runner.remove_lora(name)
 
+    def set_token_table(self, table: list[str]) -> None:
+        """Per-token-id decoded strings for guided-JSON decoding (rank 0
+        only; follower picks are overwritten by the rank-0 broadcast)."""
+        self.runner.sampler.token_table = table
+
     def lora_names(self) -> list[str]:
         names = list(self.runner.lora_bank.names())
         names.extend(op[1] for op in self._pending_ops if op[0] == "lora_add")
@@ -280,7 +285,7 @@ class LLMEngine:
             else:
                 rows = token_ids[i * rps:(i + 1) * rps]
                 row0 = i * rps
-                if seq.params.greedy and batch.k_eff[i] > 0:
+                if seq.params.spec_safe and batch.k_eff[i] > 0:
                     from .spec import accept_tokens
 
                     ke = batch.k_eff[i]

@@ -1,0 +1,447 @@
+"""Grammar-constrained JSON decoding (reference: vLLM guided_json via
+outlines/xgrammar, passed through gpustack backend_parameters).
+
+First-party design: a character-level machine validates the generated text
+incrementally; sampling picks the highest-probability token whose decoded
+string keeps the output a valid prefix (top-k candidate rejection — the
+top few tokens are almost always valid, so the per-step cost is a handful
+of short string walks, not a vocab scan).
+
+Two grammars:
+  * `guided_json: true` — any syntactically valid JSON value (full PDA:
+    nested objects/arrays, strings with escapes, numbers, literals).
+  * `guided_json: {schema}` — a canonical template compiled from a JSON
+    Schema subset (flat or nested objects with string/number/integer/
+    boolean/null/enum/array-typed properties, all emitted in property
+    order): {"name":<string>,"age":<integer>} — literal segments are
+    forced, value slots run the matching value machine.
+
+Machines implement `advance(ch) -> bool` (consume or reject) and
+`.complete` (the text so far is a finished value). Sequencing retries a
+rejected char on the next segment when the current one is complete.
+"""
+from __future__ import annotations
+
+import copy
+
+_WS = " \t\n\r"
+_DIGITS = "0123456789"
+_HEX = "0123456789abcdefABCDEF"
+
+
+class StringM:
+    """A JSON string including both quotes."""
+
+    def __init__(self):
+        self.state = 0  # 0=expect open ", 1=body, 2=escape, 3..6=\uXXXX, 7=done
+
+    @property
+    def complete(self) -> bool:
+        return self.state == 7
+
+    def advance(self, ch: str) -> bool:
+        s = self.state
+        if s == 0:
+            if ch == '"':
+                self.state = 1
+                return True
+            return False
+        if s == 1:
+            if ch == '"':
+                self.state = 7
+                return True
+            if ch == "\\":
+                self.state = 2
+                return True
+            return ord(ch) >= 0x20
+        if s == 2:
+            if ch in '"\\/bfnrt':
+                self.state = 1
+                return True
+            if ch == "u":
+                self.state = 3
+                return True
+            return False
+        if 3 <= s <= 6:
+            if ch in _HEX:
+                self.state = 1 if s == 6 else s + 1
+                return True
+            return False
+        return False  # done: no more chars
+
+
+class NumberM:
+    """-?(0|[1-9][0-9]*)(\\.[0-9]+)?([eE][+-]?[0-9]+)?; integer_only drops
+    the fraction/exponent parts."""
+
+    def __init__(self, integer_only: bool = False):
+        self.integer_only = integer_only
+        self.state = 0  # 0=start 1=after- 2=after0 3=int 4=dot 5=frac 6=e 7=esign 8=exp
+
+    @property
+    def complete(self) -> bool:
+        return self.state in (2, 3, 5, 8)
+
+    def advance(self, ch: str) -> bool:
+        s = self.state
+        if s == 0:
+            if ch == "-":
+                self.state = 1
+                return True
+            if ch == "0":
+                self.state = 2
+                return True
+            if ch in "123456789":
+                self.state = 3
+                return True
+            return False
+        if s == 1:
+            if ch == "0":
+                self.state = 2
+                return True
+            if ch in "123456789":
+                self.state = 3
+                return True
+            return False
+        if s in (2, 3):
+            if s == 3 and ch in _DIGITS:
+                return True
+            if self.integer_only:
+                return False
+            if ch == ".":
+                self.state = 4
+                return True
+            if ch in "eE":
+                self.state = 6
+                return True
+            return False
+        if s in (4, 5):
+            if ch in _DIGITS:
+                self.state = 5
+                return True
+            if s == 5 and ch in "eE":
+                self.state = 6
+                return True
+            return False
+        if s == 6:
+            if ch in "+-":
+                self.state = 7
+                return True
+            if ch in _DIGITS:
+                self.state = 8
+                return True
+            return False
+        if s == 7:
+            if ch in _DIGITS:
+                self.state = 8
+                return True
+            return False
+        if s == 8:
+            return ch in _DIGITS
+        return False
+
+
+class LitM:
+    """One of several fixed strings (true/false, null, enum values)."""
+
+    def __init__(self, options: list[str]):
+        self.options = options
+        self.pos = 0
+        self.alive = list(range(len(options)))
+
+    @property
+    def complete(self) -> bool:
+        return any(len(self.options[i]) == self.pos for i in self.alive)
+
+    def advance(self, ch: str) -> bool:
+        nxt = [i for i in self.alive
+               if self.pos < len(self.options[i]) and self.options[i][self.pos] == ch]
+        if not nxt:
+            return False
+        self.alive = nxt
+        self.pos += 1
+        return True
+
+
+class JsonPDA:
+    """Any syntactically valid JSON value (single top-level value)."""
+
+    def __init__(self):
+        self.stack: list[str] = []   # 'o' object, 'a' array
+        self.mode = "value"          # value|key|colon|comma|string|number|lit
+        self.sub = None              # active StringM/NumberM/LitM
+        self.sub_is_key = False
+        self.done_value = False      # a complete top-level value was consumed
+
+    @property
+    def complete(self) -> bool:
+        if self.stack:
+            return False
+        if self.mode == "number" and self.sub is not None:
+            return self.sub.complete
+        return self.done_value
+
+    def _finish_value(self) -> None:
+        self.sub = None
+        if not self.stack:
+            self.mode = "end"
+            self.done_value = True
+        else:
+            self.mode = "comma"
+
+    def advance(self, ch: str) -> bool:
+        m = self.mode
+        if m in ("string", "lit"):
+            if self.sub.advance(ch):
+                if self.sub.complete and m == "string" and self.sub_is_key:
+                    if self.sub.state == 7:
+                        self.mode = "colon"
+                        self.sub = None
+                elif self.sub.complete:
+                    self._finish_value()
+                return True
+            return False
+        if m == "number":
+            if self.sub.advance(ch):
+                return True
+            if self.sub.complete:  # number ended; re-dispatch ch
+                self._finish_value()
+                return self.advance(ch)
+            return False
+        if ch in _WS:
+            return m in ("value", "key", "colon", "comma", "end")
+        if m == "value":
+            if ch == "{":
+                self.stack.append("o")
+                self.mode = "key"
+                return True
+            if ch == "[":
+                self.stack.append("a")
+                self.mode = "value"
+                self.first_in_container = True
+                return True
+            if ch == "]" and self.stack and self.stack[-1] == "a" \
+                    and getattr(self, "first_in_container", False):
+                self.stack.pop()
+                self.first_in_container = False
+                self._finish_value()
+                return True
+            if ch == '"':
+                self.sub = StringM()
+                self.sub.advance('"')
+                self.sub_is_key = False
+                self.mode = "string"
+                return True
+            if ch in "-0123456789":
+                self.sub = NumberM()
+                self.mode = "number"
+                return self.sub.advance(ch)
+            if ch in "tfn":
+                self.sub = LitM(["true", "false", "null"])
+                self.mode = "lit"
+                return self.sub.advance(ch)
+            return False
+        if m == "key":
+            if ch == '"':
+                self.sub = StringM()
+                self.sub.advance('"')
+                self.sub_is_key = True
+                self.mode = "string"
+                return True
+            if ch == "}" and self.stack and self.stack[-1] == "o" \
+                    and getattr(self, "first_in_container", True):
+                self.stack.pop()
+                self._finish_value()
+                return True
+            return False
+        if m == "colon":
+            if ch == ":":
+                self.mode = "value"
+                self.first_in_container = False
+                return True
+            return False
+        if m == "comma":
+            top = self.stack[-1] if self.stack else None
+            if ch == "," and top == "o":
+                self.mode = "key"
+                self.first_in_container = False
+                return True
+            if ch == "," and top == "a":
+                self.mode = "value"
+                self.first_in_container = False
+                return True
+            if ch == "}" and top == "o":
+                self.stack.pop()
+                self._finish_value()
+                return True
+            if ch == "]" and top == "a":
+                self.stack.pop()
+                self._finish_value()
+                return True
+            return False
+        return False  # end: nothing more
+
+
+class SeqM:
+    """Sequence of machines/literals; rejected chars retry on the next
+    segment once the current one is complete."""
+
+    def __init__(self, segments: list):
+        self.segments = segments
+        self.idx = 0
+
+    @property
+    def complete(self) -> bool:
+        i = self.idx
+        if i >= len(self.segments):
+            return True
+        # complete if every remaining segment is already complete (only the
+        # current can hold partial state)
+        return self.segments[i].complete and i == len(self.segments) - 1
+
+    def advance(self, ch: str) -> bool:
+        while self.idx < len(self.segments):
+            seg = self.segments[self.idx]
+            if seg.advance(ch):
+                return True
+            if seg.complete:
+                self.idx += 1
+                continue
+            return False
+        return False
+
+
+class ArrayM:
+    """[item(,item)*] with typed items from a factory."""
+
+    def __init__(self, item_factory):
+        self.factory = item_factory
+        self.state = 0  # 0=expect [ 1=first item or ] 2=in item 3=comma or ]
+        self.item = None
+
+    @property
+    def complete(self) -> bool:
+        return self.state == 4
+
+    def advance(self, ch: str) -> bool:
+        if self.state == 0:
+            if ch == "[":
+                self.state = 1
+                return True
+            return False
+        if self.state == 1:
+            if ch == "]":
+                self.state = 4
+                return True
+            self.item = self.factory()
+            self.state = 2
+            return self.advance(ch)
+        if self.state == 2:
+            if self.item.advance(ch):
+                if self.item.complete and not isinstance(self.item, NumberM):
+                    self.state = 3
+                return True
+            if self.item.complete:  # numbers end on the delimiter
+                self.state = 3
+                return self.advance(ch)
+            return False
+        if self.state == 3:
+            if ch == ",":
+                self.item = self.factory()
+                self.state = 2
+                return True
+            if ch == "]":
+                self.state = 4
+                return True
+            # item machine may accept more (e.g. string already closed: no)
+            return False
+        return False
+
+
+def _value_machine(schema: dict):
+    t = schema.get("type")
+    if "enum" in schema:
+        import json as _json
+
+        return lambda: LitM([_json.dumps(v) for v in schema["enum"]])
+    if t == "string":
+        return StringM
+    if t == "integer":
+        return lambda: NumberM(integer_only=True)
+    if t == "number":
+        return NumberM
+    if t == "boolean":
+        return lambda: LitM(["true", "false"])
+    if t == "null":
+        return lambda: LitM(["null"])
+    if t == "array":
+        inner = _value_machine(schema.get("items", {}))
+        return lambda: ArrayM(inner)
+    if t == "object" and schema.get("properties"):
+        return lambda: compile_schema(schema)
+    return JsonPDA  # unconstrained value
+
+
+class _Lit:
+    """Exact literal text segment."""
+
+    def __init__(self, text: str):
+        self.text = text
+        self.pos = 0
+
+    @property
+    def complete(self) -> bool:
+        return self.pos >= len(self.text)
+
+    def advance(self, ch: str) -> bool:
+        if self.pos < len(self.text) and self.text[self.pos] == ch:
+            self.pos += 1
+            return True
+        return False
+
+
+def compile_schema(schema: dict) -> SeqM:
+    """JSON Schema (object subset) -> canonical template machine.
+    Properties are emitted in declaration order, all of them, unquoted
+    whitespace-free: {"a":<v>,"b":<v>}."""
+    import json as _json
+
+    props = schema.get("properties") or {}
+    segs: list = [_Lit("{")]
+    for i, (key, sub) in enumerate(props.items()):
+        if i:
+            segs.append(_Lit(","))
+        segs.append(_Lit(_json.dumps(key) + ":"))
+        segs.append(_value_machine(sub)())
+    segs.append(_Lit("}"))
+    return SeqM(segs)
+
+
+class GuidedJsonState:
+    """Per-sequence guided-decoding state: the machine + commit/probe API."""
+
+    def __init__(self, schema):
+        if isinstance(schema, dict) and schema.get("type") == "object" \
+                and schema.get("properties"):
+            self.machine = compile_schema(schema)
+        elif isinstance(schema, dict) and (schema.get("type")
+                                           or "enum" in schema):
+            self.machine = _value_machine(schema)()
+        else:
+            self.machine = JsonPDA()
+
+    @property
+    def complete(self) -> bool:
+        return self.machine.complete
+
+    def try_advance(self, text: str):
+        """Probe: returns the advanced machine if `text` keeps the output a
+        valid prefix (None otherwise); caller commits via `commit`."""
+        m = copy.deepcopy(self.machine)
+        for ch in text:
+            if not m.advance(ch):
+                return None
+        return m
+
+    def commit(self, machine) -> None:
+        self.machine = machine

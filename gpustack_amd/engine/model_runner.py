@@ -18,6 +18,56 @@ class Sampler:
     def __init__(self, device):
         self.device = device
         self.generator = None
+        # guided-JSON decoding: per-token-id decoded strings (set by the
+        # engine server; None on TP followers, whose picks rank 0 overwrites)
+        self.token_table: list[str] | None = None
+
+    def _guided_pick(self, row: torch.Tensor, seq: Sequence) -> int:
+        """Grammar-constrained pick (engine/guided.py): sync the machine
+        with the emitted text, then take the best token whose string keeps
+        the output a valid JSON prefix (top-down candidate rejection)."""
+        from .guided import GuidedJsonState
+
+        p = seq.params
+        table = self.token_table
+        if table is None:
+            return int(row.argmax())
+        if seq.guided_sm is None:
+            schema = p.guided_json if isinstance(p.guided_json, dict) else None
+            seq.guided_sm = GuidedJsonState(schema)
+            seq.guided_consumed = 0
+        sm = seq.guided_sm
+        out = seq.output_token_ids
+        for tok in out[seq.guided_consumed:]:
+            m = sm.try_advance(table[tok] if tok < len(table) else "")
+            if m is not None:
+                sm.commit(m)
+        seq.guided_consumed = len(out)
+        if sm.complete:
+            return p.eos_token_id
+
+        def valid(tid: int) -> bool:
+            s = table[tid] if tid < len(table) else ""
+            return bool(s) and sm.try_advance(s) is not None
+
+        if not p.greedy:
+            k = min(256, row.shape[-1])
+            vals, idx = torch.topk(row, k)
+            ids = idx.tolist()
+            valid_j = [j for j in range(k) if valid(ids[j])]
+            if valid_j:
+                probs = torch.softmax(
+                    vals.float()[valid_j] / max(p.temperature, 1e-5), -1)
+                if self.generator is None:
+                    self.generator = torch.Generator(device=row.device)
+                if p.seed is not None:
+                    self.generator.manual_seed(p.seed + len(out))
+                j = int(torch.multinomial(probs, 1, generator=self.generator))
+                return ids[valid_j[j]]
+        for tid in torch.argsort(row, descending=True).tolist():
+            if valid(tid):
+                return tid
+        return p.eos_token_id  # vocab cannot extend the prefix: terminate
 
     @staticmethod
     def _process_logits(row: torch.Tensor, seq: Sequence) -> torch.Tensor:
@@ -68,11 +118,17 @@ class Sampler:
                for s in seqs):
             return ops.greedy_sample(logits).tolist()
         out: list[int] = [0] * len(seqs)
+        guided = {i for i, s in enumerate(seqs)
+                  if s.params.guided_json is not None}
         greedy_idx = [i for i, s in enumerate(seqs)
                       if s.params.greedy and not s.params.needs_logit_processing]
         proc_greedy = [i for i, s in enumerate(seqs)
-                       if s.params.greedy and s.params.needs_logit_processing]
-        rand_idx = [i for i, s in enumerate(seqs) if not s.params.greedy]
+                       if i not in guided
+                       and s.params.greedy and s.params.needs_logit_processing]
+        rand_idx = [i for i, s in enumerate(seqs)
+                    if i not in guided and not s.params.greedy]
+        for i in guided:
+            out[i] = self._guided_pick(logits[i].float(), seqs[i])
         if greedy_idx:
             ids = ops.greedy_sample(logits[greedy_idx])
             for j, i in enumerate(greedy_idx):

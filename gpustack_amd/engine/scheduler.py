@@ -453,9 +453,9 @@ class Scheduler:
                 lens.append(pos + 1)
                 continue
             k_eff = max(0, min(self.spec_k, self.cfg.max_model_len - 1 - pos))
-            if seq.params.greedy and k_eff > 0 and self.proposer is not None:
+            if seq.params.spec_safe and k_eff > 0 and self.proposer is not None:
                 draft = self.proposer.propose(seq)
-            elif seq.params.greedy and k_eff > 0 and seq.next_draft:
+            elif seq.params.spec_safe and k_eff > 0 and seq.next_draft:
                 draft = (list(seq.next_draft) + [last] * self.spec_k)[:self.spec_k]
                 seq.next_draft = None
             else:

@@ -26,6 +26,9 @@ class SamplingParams:
     # (constrained per-step logit masking; engine_server encodes the
     # user-facing `guided_choice` strings)
     guided_token_seqs: tuple | None = None
+    # grammar-constrained JSON output (engine/guided.py): True = any valid
+    # JSON; a dict = JSON-Schema subset compiled to a template machine
+    guided_json: "bool | dict | None" = None
     eos_token_id: int = 0           # used by guided decoding to terminate
     max_tokens: int = 128
     ignore_eos: bool = False
@@ -39,8 +42,15 @@ class SamplingParams:
     @property
     def needs_logit_processing(self) -> bool:
         return bool(self.logit_bias) or bool(self.guided_token_seqs) \
+            or self.guided_json is not None \
             or self.presence_penalty != 0.0 \
             or self.frequency_penalty != 0.0 or self.repetition_penalty != 1.0
+
+    @property
+    def spec_safe(self) -> bool:
+        """Speculative drafts verify greedily row by row; stateful guided
+        decoding cannot validate draft rows, so such seqs decode plain."""
+        return self.greedy and self.guided_json is None
 
     @property
     def greedy(self) -> bool:
@@ -62,6 +72,8 @@ class Sequence:
     next_draft: list | None = None  # draft-model speculative window (engine/eagle.py)
     lora_slot: int = 0              # 0 = no adapter (models/lora.py LoraBank)
     block_hashes: list | None = None  # prefix-cache chain (engine/kv_cache.py)
+    guided_sm: object | None = None   # guided-JSON machine (engine/guided.py)
+    guided_consumed: int = 0          # output tokens already fed to guided_sm
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: float | None = None
     finish_time: float | None = None

@@ -158,6 +158,20 @@ class EngineRunner:
         self._wake.set()
         return rid, q
 
+    def ensure_token_table(self) -> None:
+        """Build the id->string table guided-JSON decoding probes (lazy:
+        one decode pass over the vocab, first guided request only)."""
+        if getattr(self, "_token_table_done", False):
+            return
+        self._token_table_done = True
+        table = []
+        for i in range(self.engine.cfg.spec.vocab_size):
+            try:
+                table.append(self.tokenizer.decode([i]))
+            except Exception:  # noqa: BLE001
+                table.append("")
+        self.engine.set_token_table(table)
+
     def release(self, rid: str) -> None:
         with self._lock:
             self._queues.pop(rid, None)
@@ -180,6 +194,14 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
     from ..engine import SamplingParams
 
     mt = body.get("max_tokens") or body.get("max_completion_tokens") or 256
+    # guided JSON: explicit guided_json (bool | schema) or the OpenAI
+    # response_format surface (json_object / json_schema)
+    gj = body.get("guided_json")
+    rf = body.get("response_format") or {}
+    if gj is None and rf.get("type") == "json_object":
+        gj = True
+    elif gj is None and rf.get("type") == "json_schema":
+        gj = (rf.get("json_schema") or {}).get("schema") or True
     guided = None
     if body.get("guided_choice") and tokenizer is not None:
         seqs = []
@@ -204,8 +226,11 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         seed=body.get("seed"),
         logprobs=bool(body.get("logprobs")),
         guided_token_seqs=guided,
+        guided_json=gj,
         eos_token_id=eos_token_id,
     )
+    if gj is not None:
+        sp.ignore_eos = False
     if guided:
         sp.ignore_eos = False
         sp.max_tokens = max(len(c) for c in guided) + 1
@@ -301,6 +326,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                   runner.tokenizer)
         _apply_lora_routing(params, body, runner)
+        if params.guided_json is not None:
+            runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
         created = int(time.time())
@@ -448,6 +475,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                 runner.tokenizer)
         _apply_lora_routing(base, body, runner)
+        if base.guided_json is not None:
+            runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         subs = []
         for i in range(n):

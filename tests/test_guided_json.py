@@ -1,0 +1,196 @@
+"""Guided JSON decoding: char-level PDA, schema templates, engine + server
+integration (reference: vLLM guided_json through gpustack backend params)."""
+import json
+
+import pytest
+
+from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+from gpustack_amd.engine.guided import (
+    GuidedJsonState, JsonPDA, compile_schema,
+)
+
+
+def _accepts(machine, text: str) -> bool:
+    for ch in text:
+        if not machine.advance(ch):
+            return False
+    return True
+
+
+@pytest.mark.parametrize("text", [
+    '{"a": 1}',
+    '[1, 2.5, {"x": [true, null]}]',
+    '"hi\\n there \\u00e9"',
+    '-0.5e3',
+    'true',
+    '{}',
+    '[]',
+    '{"nested": {"deep": [[]]}}',
+    '  {"ws" : [ 1 , 2 ] }',
+])
+def test_pda_accepts_valid(text):
+    m = JsonPDA()
+    assert _accepts(m, text)
+    assert m.complete
+
+
+@pytest.mark.parametrize("text", [
+    '{,',
+    '[1,]',
+    '{"a" 1}',
+    '01',
+    'tru x',
+    '{"a":}',
+    '}',
+    '{"a": 1,}',
+])
+def test_pda_rejects_invalid(text):
+    assert not _accepts(JsonPDA(), text)
+
+
+def test_pda_incomplete_prefixes():
+    m = JsonPDA()
+    assert _accepts(m, '{"key": [1, 2')
+    assert not m.complete
+    assert _accepts(m, ']}')
+    assert m.complete
+
+
+SCHEMA = {
+    "type": "object",
+    "properties": {
+        "name": {"type": "string"},
+        "age": {"type": "integer"},
+        "ok": {"type": "boolean"},
+        "tags": {"type": "array", "items": {"type": "string"}},
+        "mode": {"enum": ["fast", "slow"]},
+    },
+}
+
+
+def test_schema_template_accepts_canonical():
+    m = compile_schema(SCHEMA)
+    good = '{"name":"bo","age":42,"ok":true,"tags":["a","b"],"mode":"slow"}'
+    assert _accepts(m, good)
+    assert m.complete
+    assert json.loads(good)
+
+
+@pytest.mark.parametrize("bad", [
+    '{"age":42',                  # wrong first key
+    '{"name":42',                 # wrong type
+    '{"name":"bo","age":4.5',     # float for integer
+    '{"name":"bo","age":42,"ok":"y"',   # string for boolean
+    '{"name":"bo","age":42,"ok":true,"tags":[1',  # non-string item
+    '{"name":"bo","age":42,"ok":true,"tags":[],"mode":"mid"',  # bad enum
+])
+def test_schema_template_rejects(bad):
+    assert not _accepts(compile_schema(SCHEMA), bad)
+
+
+def _engine_with_table(**kw):
+    from gpustack_amd.worker.engine_server import ByteTokenizer
+
+    kw.setdefault("model", "tiny")
+    kw.setdefault("device", "cpu")
+    kw.setdefault("kv_cache_blocks", 64)
+    eng = LLMEngine(EngineConfig(**kw))
+    tok = ByteTokenizer(eng.cfg.spec.vocab_size)
+    eng.set_token_table([tok.decode([i]) for i in range(eng.cfg.spec.vocab_size)])
+    return eng, tok
+
+
+def test_engine_guided_any_json_prefix_valid():
+    """Every emitted token keeps the output a valid JSON prefix; if the
+    request finished by stop, the whole output parses."""
+    eng, tok = _engine_with_table()
+    p = SamplingParams(max_tokens=48, guided_json=True, eos_token_id=1)
+    rid = eng.add_request([20, 21, 22], p)
+    toks, reason = [], None
+    while eng.has_unfinished():
+        for o in eng.step():
+            if o.request_id == rid:
+                if o.finish_reason:
+                    reason = o.finish_reason
+                toks.append(o.token_id)
+    text = tok.decode([t for t in toks if t != 1])
+    m = JsonPDA()
+    assert _accepts(m, text), f"invalid prefix: {text!r}"
+    if reason == "stop":
+        json.loads(text)
+
+
+def test_engine_guided_schema_exact():
+    eng, tok = _engine_with_table()
+    schema = {"type": "object",
+              "properties": {"ok": {"type": "boolean"},
+                             "mode": {"enum": ["a", "b"]}}}
+    p = SamplingParams(max_tokens=40, guided_json=schema, eos_token_id=1)
+    out = eng.generate([[30, 31]], p)[0]
+    text = tok.decode([t for t in out if t != 1])
+    doc = json.loads(text)
+    assert set(doc) == {"ok", "mode"}
+    assert isinstance(doc["ok"], bool)
+    assert doc["mode"] in ("a", "b")
+
+
+def test_engine_guided_sampled_schema():
+    """Temperature sampling stays inside the grammar too."""
+    eng, tok = _engine_with_table()
+    schema = {"type": "object", "properties": {"ok": {"type": "boolean"}}}
+    p = SamplingParams(max_tokens=30, temperature=0.9, seed=3,
+                       guided_json=schema, eos_token_id=1)
+    out = eng.generate([[40]], p)[0]
+    doc = json.loads(tok.decode([t for t in out if t != 1]))
+    assert isinstance(doc["ok"], bool)
+
+
+@pytest.mark.timeout(240)
+def test_engine_server_response_format():
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-g", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 90:
+            if proc.poll() is not None:
+                raise AssertionError(f"engine server exited {proc.returncode}")
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health",
+                             timeout=2).status_code == 200:
+                    break
+            except httpx.HTTPError:
+                pass
+            time.sleep(0.5)
+        else:
+            raise AssertionError("engine server never became healthy")
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+            "model": "tiny-g", "prompt": "give me json", "max_tokens": 40,
+            "temperature": 0,
+            "response_format": {"type": "json_schema", "json_schema": {
+                "schema": {"type": "object",
+                           "properties": {"ok": {"type": "boolean"}}}}},
+        }, timeout=120)
+        assert r.status_code == 200
+        doc = json.loads(r.json()["choices"][0]["text"])
+        assert isinstance(doc["ok"], bool)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
