@@ -83,6 +83,10 @@ class EagleProposer:
             self.layers.append(DecoderLayer(spec, cfg.tp_size, comm, dtype).to(device))
             self.norms.append(torch.empty(h, dtype=dtype, device=device))
         self._init_weights()
+        draft_dir = cfg.speculative.get("draft_dir")
+        if draft_dir:
+            n = self.load_draft(draft_dir)
+            logger.info("loaded draft checkpoint %s (%d tensors)", draft_dir, n)
 
     def _init_weights(self) -> None:
         """Deterministic TP-consistent random init (models/weights.py
@@ -125,6 +129,105 @@ class EagleProposer:
             layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
             layer.input_norm.fill_(1.0)
             layer.post_attn_norm.fill_(1.0)
+
+    def load_draft(self, draft_dir) -> int:
+        """Load a published EAGLE/MTP draft checkpoint (speculative_config
+        `draft_dir`; reference: spec-decode model refs routed to engine
+        flags at worker/backends/vllm.py:532-566).
+
+        Accepts the EAGLE release naming (`fc.weight`,
+        `layers.0.self_attn.q_proj.weight`, optional `norm.weight`, with or
+        without a `model.` prefix); head hi of an MTP checkpoint reads
+        `layers.{hi}.*`. Tensors the checkpoint lacks keep their random
+        init (missing norms default to ones via init). Returns the number
+        of tensors loaded.
+        """
+        from pathlib import Path
+
+        from safetensors import safe_open
+
+        draft_dir = Path(draft_dir)
+        files = sorted(draft_dir.glob("*.safetensors"))
+        if not files:
+            raise FileNotFoundError(f"no safetensors under {draft_dir}")
+        tensors: dict[str, torch.Tensor] = {}
+        for f in files:
+            with safe_open(str(f), framework="pt") as sf:
+                for name in sf.keys():
+                    tensors[name] = sf.get_tensor(name)
+
+        cfg, spec = self.cfg, self.spec
+        tp, rank = cfg.tp_size, cfg.tp_rank
+        d = spec.head_dim
+        hq, hkv = spec.num_heads // tp, max(1, spec.num_kv_heads // tp)
+        i_loc = spec.intermediate_size // tp
+        dt, dev = self.fc_ws[0].dtype, self.device
+
+        def find(*frags: str) -> torch.Tensor | None:
+            for key, t in tensors.items():
+                if all(f in key for f in frags):
+                    return t.to(dt)
+            return None
+
+        loaded = 0
+
+        def copy_rows(dst, t, off, rows):
+            nonlocal loaded
+            dst.copy_(t[off:off + rows].to(dev))
+            loaded += 1
+
+        for hi in range(self.n_heads):
+            li = f"layers.{hi}." if self.n_heads > 1 else "layers.0."
+            fc = find("fc.weight") if hi == 0 or self.n_heads == 1 \
+                else find(f"{hi}.", "fc.weight")
+            if fc is not None:
+                self.fc_ws[hi].copy_(fc.to(dev))
+                loaded += 1
+            nrm = None
+            for cand in (f"{li}norm.weight", "norm.weight", "model.norm.weight"):
+                if cand in tensors:
+                    nrm = tensors[cand].to(dt)
+                    break
+            if nrm is not None and nrm.numel() == spec.hidden_size:
+                self.norms[hi].copy_(nrm.to(dev))
+                loaded += 1
+            layer = self.layers[hi]
+            la = layer.attn
+            q = find(li, "q_proj.weight")
+            k = find(li, "k_proj.weight")
+            v = find(li, "v_proj.weight")
+            if q is not None and k is not None and v is not None:
+                la.qkv_w.copy_(torch.cat([
+                    q[rank * hq * d:(rank + 1) * hq * d],
+                    k[rank * hkv * d:(rank + 1) * hkv * d],
+                    v[rank * hkv * d:(rank + 1) * hkv * d]]).to(dev))
+                loaded += 3
+            o = find(li, "o_proj.weight")
+            if o is not None:
+                la.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d].to(dev))
+                loaded += 1
+            g, u = find(li, "gate_proj.weight"), find(li, "up_proj.weight")
+            if g is not None and u is not None:
+                layer.mlp.gate_up_w.copy_(torch.cat([
+                    g[rank * i_loc:(rank + 1) * i_loc],
+                    u[rank * i_loc:(rank + 1) * i_loc]]).to(dev))
+                loaded += 2
+            dn = find(li, "down_proj.weight")
+            if dn is not None:
+                layer.mlp.down_w.copy_(
+                    dn[:, rank * i_loc:(rank + 1) * i_loc].to(dev))
+                loaded += 1
+            inorm = find(li, "input_layernorm.weight")
+            if inorm is not None:
+                layer.input_norm.copy_(inorm.to(dev))
+                loaded += 1
+            pnorm = find(li, "post_attention_layernorm.weight")
+            if pnorm is not None:
+                layer.post_attn_norm.copy_(pnorm.to(dev))
+                loaded += 1
+        if loaded == 0:
+            raise ValueError(f"{draft_dir}: no recognizable draft tensors")
+        return loaded
 
     # -- draft-KV bookkeeping ---------------------------------------------
 

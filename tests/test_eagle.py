@@ -88,3 +88,66 @@ def test_mtp_matches_plain():
     eng, out = _gen({"method": "mtp", "num_draft_tokens": 3})
     assert eng.runner.eagle.n_heads == 3
     assert out == plain
+
+
+def test_draft_checkpoint_loading(tmp_path):
+    """speculative_config draft_dir: published EAGLE-naming checkpoint loads
+    into the draft head; outputs stay exact vs plain greedy."""
+    import torch
+    from safetensors.torch import save_file
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    spec = EngineConfig(model="tiny").spec
+    d = spec.head_dim
+    torch.manual_seed(9)
+    ckpt = {
+        "fc.weight": torch.randn(spec.hidden_size, 2 * spec.hidden_size) * 0.02,
+        "layers.0.self_attn.q_proj.weight":
+            torch.randn(spec.num_heads * d, spec.hidden_size) * 0.02,
+        "layers.0.self_attn.k_proj.weight":
+            torch.randn(spec.num_kv_heads * d, spec.hidden_size) * 0.02,
+        "layers.0.self_attn.v_proj.weight":
+            torch.randn(spec.num_kv_heads * d, spec.hidden_size) * 0.02,
+        "layers.0.self_attn.o_proj.weight":
+            torch.randn(spec.hidden_size, spec.num_heads * d) * 0.02,
+        "layers.0.mlp.gate_proj.weight":
+            torch.randn(spec.intermediate_size, spec.hidden_size) * 0.02,
+        "layers.0.mlp.up_proj.weight":
+            torch.randn(spec.intermediate_size, spec.hidden_size) * 0.02,
+        "layers.0.mlp.down_proj.weight":
+            torch.randn(spec.hidden_size, spec.intermediate_size) * 0.02,
+        "layers.0.input_layernorm.weight": torch.ones(spec.hidden_size),
+        "layers.0.post_attention_layernorm.weight": torch.ones(spec.hidden_size),
+    }
+    save_file(ckpt, str(tmp_path / "draft.safetensors"))
+
+    def cfg(spec_cfg=None):
+        return EngineConfig(model="tiny", device="cpu", kv_cache_blocks=96,
+                            max_model_len=256, speculative=spec_cfg)
+
+    eng = LLMEngine(cfg({"method": "eagle", "num_draft_tokens": 2,
+                         "draft_dir": str(tmp_path)}))
+    # loaded weights match the checkpoint (TP=1: no shard offset)
+    got = eng.runner.eagle.layers[0].attn.qkv_w.float()
+    want = torch.cat([ckpt["layers.0.self_attn.q_proj.weight"],
+                      ckpt["layers.0.self_attn.k_proj.weight"],
+                      ckpt["layers.0.self_attn.v_proj.weight"]]).to(torch.bfloat16).float()
+    assert torch.equal(got, want)
+    assert torch.equal(eng.runner.eagle.fc_ws[0].float(),
+                       ckpt["fc.weight"].to(torch.bfloat16).float())
+    # exactness invariant holds with a loaded draft too
+    p = SamplingParams(max_tokens=12, ignore_eos=True)
+    plain = LLMEngine(cfg()).generate([[3, 4, 5]], p)
+    assert eng.generate([[3, 4, 5]], p) == plain
+
+
+def test_draft_checkpoint_bad_dir(tmp_path):
+    import pytest as _pytest
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine
+
+    with _pytest.raises(FileNotFoundError):
+        LLMEngine(EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                               speculative={"method": "eagle",
+                                            "draft_dir": str(tmp_path)}))
