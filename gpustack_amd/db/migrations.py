@@ -49,12 +49,18 @@ def _m005_model_lora_adapters(conn):
     _add_column(conn, "models", "lora_adapters", "JSON")
 
 
+def _m006_worker_pools(conn):
+    from ..schemas.tables import WorkerPool
+    WorkerPool.__table__.create(conn, checkfirst=True)
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
     (3, "instance cross-worker rank layout", _m003_instance_distributed_servers),
     (4, "instance spec_hash for update-triggered redeploy", _m004_instance_spec_hash),
     (5, "model.lora_adapters for dynamic multi-LoRA", _m005_model_lora_adapters),
+    (6, "worker_pools table for auto-provisioned capacity", _m006_worker_pools),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

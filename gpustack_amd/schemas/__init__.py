@@ -18,7 +18,7 @@ from .tables import (  # noqa: F401
     SourceEnum,
     SystemLoad,
     User,
-    Worker,
+    Worker, WorkerPool,
     WorkerState,
 )
 
@@ -144,6 +144,21 @@ class BenchmarkCreate(BaseModel):
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64
+
+
+class WorkerPoolCreate(BaseModel):
+    name: str
+    provider: str = "mock"
+    instance_type: str = "mi355x-8gpu"
+    replicas: int = 0
+    provider_config: dict = Field(default_factory=dict)
+    labels: dict = Field(default_factory=dict)
+
+
+class WorkerPoolUpdate(BaseModel):
+    replicas: int | None = None
+    provider_config: dict | None = None
+    labels: dict | None = None
 
 
 class ModelProviderCreate(BaseModel):

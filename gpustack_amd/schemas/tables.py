@@ -202,6 +202,24 @@ class ModelProvider(Base, TimestampMixin, SerializeMixin):
     enabled = Column(Boolean, default=True)
 
 
+class WorkerPool(Base, TimestampMixin, SerializeMixin):
+    """Auto-provisioned worker capacity (reference: schemas/clusters.py
+    WorkerPool + cloud_providers/; the provider abstraction is
+    server/providers.py — mock / command-hook providers replace the
+    reference's DigitalOcean droplets for bare-metal MI355X labs)."""
+    __tablename__ = "worker_pools"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    provider = Column(String(64), default="mock")
+    instance_type = Column(String(128), default="mi355x-8gpu")
+    replicas = Column(Integer, default=0)
+    provider_config = Column(JSON, default=dict)  # endpoint/commands/etc.
+    labels = Column(JSON, default=dict)           # stamped on pool workers
+    # provisioning records: [{instance_id, name, state, created_at}]
+    instances = Column(JSON, default=list)
+    state_message = Column(Text, default="")
+
+
 class Benchmark(Base, TimestampMixin, SerializeMixin):
     """In-product benchmark runs (reference: schemas/benchmark.py)."""
     __tablename__ = "benchmarks"

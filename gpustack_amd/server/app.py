@@ -143,7 +143,7 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
     from ..scheduler.scheduler import PlacementScheduler
     from .controllers import (
         ModelController, ScalingScheduler, SystemLoadCollector, UsageArchiver,
-        WorkerMonitor,
+        WorkerMonitor, WorkerPoolController,
     )
     from .coordinator import LeaseCoordinator, LocalCoordinator
 
@@ -163,7 +163,8 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
 
     sched = PlacementScheduler(cfg)
     tasks = [sched, ModelController(cfg), WorkerMonitor(cfg),
-             SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg)]
+             SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg),
+             WorkerPoolController(cfg)]
     for t in tasks:
         t.coordinator = coord  # leader-only gating (checked per cycle)
     app.state.scheduler = sched

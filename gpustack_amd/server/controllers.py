@@ -331,3 +331,114 @@ class SystemLoadCollector(_LeaderGated):
                 vram=vram / n_gpu if n_gpu else 0.0,
             ))
             s.commit()
+
+
+class WorkerPoolController(_LeaderGated):
+    """Reconciles WorkerPool.replicas against provisioned instances
+    (reference: WorkerPoolController + WorkerProvisioningController,
+    server/controllers.py:2352,2398 — cloud scale-out). Providers live in
+    server/providers.py; the bootstrap script is the cloud-init analog."""
+
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+        self._stop = False
+
+    def stop(self) -> None:
+        self._stop = True
+
+    def run(self) -> None:
+        from ..schemas import WorkerPool  # noqa: F401
+
+        q = bus.subscribe("worker_pools")
+        self.reconcile_all()
+        while not self._stop:
+            try:
+                ev = q.get(timeout=30.0)
+                if self._stop:
+                    return
+                if ev.type in (EventType.CREATED, EventType.UPDATED) and self._is_leader():
+                    self.reconcile_pool(ev.data["id"])
+            except queue.Empty:
+                if self._is_leader():
+                    self.reconcile_all()
+
+    def reconcile_all(self) -> None:
+        from ..schemas import WorkerPool
+
+        with get_session() as s:
+            ids = [p.id for p in s.query(WorkerPool).all()]
+        for pid in ids:
+            try:
+                self.reconcile_pool(pid)
+            except Exception:  # noqa: BLE001
+                logger.exception("worker pool %s reconcile failed", pid)
+
+    def _bootstrap(self, pool) -> str:
+        from ..schemas import RegistrationToken
+        from .providers import bootstrap_script
+
+        with get_session() as s:
+            row = s.query(RegistrationToken).first()
+            token = row.token if row else ""
+        server_url = f"http://{self.cfg.host}:{self.cfg.port}"
+        return bootstrap_script(server_url, token, pool.labels)
+
+    def reconcile_pool(self, pool_id: int) -> None:
+        from ..schemas import WorkerPool
+        from .providers import get_provider
+
+        with get_session() as s:
+            pool = s.get(WorkerPool, pool_id)
+            if pool is None:
+                return
+            try:
+                provider = get_provider(pool.provider, pool.provider_config)
+            except ValueError as e:
+                pool.state_message = str(e)
+                ar_update(s, pool)
+                return
+            # copy the dicts too: mutating the loaded JSON in place would
+            # erase the attribute history and skip the flush
+            records = [dict(r) for r in (pool.instances or [])]
+            want = pool.replicas or 0
+            changed = False
+            errors: list[str] = []
+            while len(records) < want:
+                name = f"{pool.name}-{len(records)}"
+                taken = {r["name"] for r in records}
+                n = 0
+                while name in taken:
+                    n += 1
+                    name = f"{pool.name}-{n + len(records)}"
+                try:
+                    iid = provider.create(name, pool.instance_type,
+                                          self._bootstrap(pool))
+                except Exception as e:  # noqa: BLE001
+                    errors.append(f"create failed: {e}")
+                    break
+                records.append({"instance_id": iid, "name": name,
+                                "state": "provisioning",
+                                "created_at": time.time()})
+                changed = True
+            while len(records) > want:
+                # scale down newest-first; workers that registered from the
+                # node disappear via the heartbeat monitor once it powers off
+                victim = records[-1]
+                try:
+                    provider.delete(victim["instance_id"])
+                except Exception as e:  # noqa: BLE001
+                    errors.append(f"delete failed: {e}")
+                    break
+                records.pop()
+                changed = True
+            # mark provisioned instances whose worker has registered READY
+            with_workers = {w.name for w in s.query(Worker).all()}
+            for r in records:
+                new_state = "ready" if r["name"] in with_workers else r["state"]
+                if new_state != r["state"]:
+                    r["state"] = new_state
+                    changed = True
+            if changed or errors:
+                pool.instances = records
+                pool.state_message = "; ".join(errors)
+                ar_update(s, pool)

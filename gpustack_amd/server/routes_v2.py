@@ -19,7 +19,8 @@ from ..schemas import (
     ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
     ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
     ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
-    WorkerRegister, WorkerState, WorkerStatusUpdate,
+    WorkerPoolCreate, WorkerPoolUpdate, WorkerRegister, WorkerState,
+    WorkerStatusUpdate,
 )
 from ..security import generate_api_key, generate_registration_token, hash_password
 from .deps import get_admin_user, get_current_user, verify_worker_token
@@ -311,6 +312,69 @@ def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
 
 
 # ---- model routes ----------------------------------------------------------
+
+@router.get("/worker_pools")
+def list_worker_pools(_: User = Depends(get_current_user)):
+    from ..schemas import WorkerPool
+
+    with get_session() as s:
+        return {"items": [p.to_dict() for p in s.query(WorkerPool).all()]}
+
+
+@router.post("/worker_pools", status_code=201)
+def create_worker_pool(body: WorkerPoolCreate,
+                       _: User = Depends(get_admin_user)):
+    from ..schemas import WorkerPool
+    from .providers import PROVIDERS
+
+    if body.provider not in PROVIDERS:
+        raise HTTPException(400, f"unknown provider {body.provider!r}")
+    with get_session() as s:
+        if s.query(WorkerPool).filter_by(name=body.name).first():
+            raise HTTPException(409, "pool exists")
+        p = WorkerPool(name=body.name, provider=body.provider,
+                       instance_type=body.instance_type,
+                       replicas=body.replicas,
+                       provider_config=body.provider_config,
+                       labels=body.labels, instances=[])
+        ar_create(s, p)
+        return p.to_dict()
+
+
+@router.put("/worker_pools/{pool_id}")
+def update_worker_pool(pool_id: int, body: WorkerPoolUpdate,
+                       _: User = Depends(get_admin_user)):
+    from ..schemas import WorkerPool
+
+    with get_session() as s:
+        p = s.get(WorkerPool, pool_id)
+        if not p:
+            raise HTTPException(404, "pool not found")
+        for k, v in body.model_dump().items():
+            if v is not None:
+                setattr(p, k, v)
+        ar_update(s, p)
+        return p.to_dict()
+
+
+@router.delete("/worker_pools/{pool_id}")
+def delete_worker_pool(pool_id: int, _: User = Depends(get_admin_user)):
+    from ..schemas import WorkerPool
+    from .providers import get_provider
+
+    with get_session() as s:
+        p = s.get(WorkerPool, pool_id)
+        if not p:
+            raise HTTPException(404, "pool not found")
+        try:
+            provider = get_provider(p.provider, p.provider_config)
+            for r in p.instances or []:
+                provider.delete(r["instance_id"])
+        except Exception:  # noqa: BLE001
+            pass  # deprovision is best-effort on delete
+        ar_delete(s, p)
+        return {"deleted": pool_id}
+
 
 @router.get("/model_routes")
 def list_routes(_: User = Depends(get_current_user)):
