@@ -29,6 +29,13 @@ class Config(BaseModel):
     metrics_port: int = 10151
     # auth
     ha_leases: bool = False                # force lease-based leader election
+    # external auth: OIDC authorization-code flow (reference:
+    # routes/auth.py:805,834 — OIDC login/callback + group sync)
+    oidc_issuer: str | None = None         # e.g. https://idp/realms/x
+    oidc_client_id: str | None = None
+    oidc_client_secret: str | None = None
+    oidc_username_claim: str = "preferred_username"
+    oidc_admin_group: str | None = None    # groups claim granting is_admin
     bootstrap_password: str | None = None
     jwt_secret: str | None = None
     disable_auth: bool = False

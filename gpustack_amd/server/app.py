@@ -10,7 +10,7 @@ import logging
 import threading
 import time
 
-from fastapi import APIRouter, Depends, FastAPI, HTTPException, Response
+from fastapi import APIRouter, Depends, FastAPI, HTTPException, Request, Response
 from fastapi.responses import JSONResponse
 
 from ..config import Config
@@ -43,6 +43,110 @@ def login(body: LoginRequest, response: Response):
 def logout(response: Response):
     response.delete_cookie(COOKIE_NAME)
     return {"ok": True}
+
+
+# -- OIDC authorization-code flow (reference: routes/auth.py:805-834 OIDC
+# login/callback, :1312 auth-config discovery). Token validation is
+# delegated to the IdP via the userinfo endpoint (the access token is
+# verified server-side there), so no local RS256/JWKS stack is needed.
+
+_oidc_discovery_cache: dict = {}
+
+
+def _oidc_discover(issuer: str) -> dict:
+    import httpx
+
+    doc = _oidc_discovery_cache.get(issuer)
+    if doc is None:
+        url = issuer.rstrip("/") + "/.well-known/openid-configuration"
+        doc = httpx.get(url, timeout=10).json()
+        _oidc_discovery_cache[issuer] = doc
+    return doc
+
+
+@auth_router.get("/config")
+def auth_config():
+    cfg = deps.get_config()
+    return {
+        "password_login": True,
+        "oidc": bool(cfg.oidc_issuer and cfg.oidc_client_id),
+        "oidc_login_url": "/auth/oidc/login" if cfg.oidc_issuer else None,
+    }
+
+
+@auth_router.get("/oidc/login")
+def oidc_login(request: Request, redirect_uri: str | None = None):
+    from urllib.parse import urlencode
+
+    cfg = deps.get_config()
+    if not (cfg.oidc_issuer and cfg.oidc_client_id):
+        raise HTTPException(404, "OIDC is not configured")
+    doc = _oidc_discover(cfg.oidc_issuer)
+    callback = redirect_uri or str(request.url_for("oidc_callback"))
+    # self-validating state: signed + expiring, no server-side session
+    state = jwt_encode({"cb": callback}, cfg.get_jwt_secret(), expires_in=600)
+    q = urlencode({
+        "response_type": "code",
+        "client_id": cfg.oidc_client_id,
+        "redirect_uri": callback,
+        "scope": "openid profile email",
+        "state": state,
+    })
+    from fastapi.responses import RedirectResponse
+
+    return RedirectResponse(f"{doc['authorization_endpoint']}?{q}")
+
+
+@auth_router.get("/oidc/callback")
+def oidc_callback(code: str, state: str, response: Response):
+    import httpx
+
+    from ..security import jwt_decode
+
+    cfg = deps.get_config()
+    if not (cfg.oidc_issuer and cfg.oidc_client_id):
+        raise HTTPException(404, "OIDC is not configured")
+    st = jwt_decode(state, cfg.get_jwt_secret())
+    if st is None:
+        raise HTTPException(400, "bad or expired state")
+    doc = _oidc_discover(cfg.oidc_issuer)
+    tok = httpx.post(doc["token_endpoint"], data={
+        "grant_type": "authorization_code",
+        "code": code,
+        "redirect_uri": st["cb"],
+        "client_id": cfg.oidc_client_id,
+        "client_secret": cfg.oidc_client_secret or "",
+    }, timeout=10)
+    if tok.status_code != 200:
+        raise HTTPException(401, f"token exchange failed: {tok.text[:200]}")
+    access = tok.json().get("access_token")
+    if not access:
+        raise HTTPException(401, "no access_token in IdP response")
+    ui = httpx.get(doc["userinfo_endpoint"],
+                   headers={"Authorization": f"Bearer {access}"}, timeout=10)
+    if ui.status_code != 200:
+        raise HTTPException(401, "userinfo rejected the access token")
+    claims = ui.json()
+    username = (claims.get(cfg.oidc_username_claim) or claims.get("email")
+                or claims.get("sub"))
+    if not username:
+        raise HTTPException(401, "no usable username claim")
+    groups = claims.get("groups") or []
+    is_admin = bool(cfg.oidc_admin_group and cfg.oidc_admin_group in groups)
+    with get_session() as s:
+        user = s.query(User).filter_by(username=username).first()
+        if user is None:  # JIT provisioning (reference group sync :726)
+            user = User(username=username, hashed_password=hash_password(
+                __import__("secrets").token_urlsafe(24)),
+                is_admin=is_admin,
+                full_name=claims.get("name") or "")
+            ar_create(s, user)
+        elif cfg.oidc_admin_group and user.is_admin != is_admin:
+            user.is_admin = is_admin
+            s.commit()
+    token = jwt_encode({"sub": username}, cfg.get_jwt_secret())
+    response.set_cookie(COOKIE_NAME, token, httponly=True, samesite="lax")
+    return {"token": token, "username": username, "is_admin": is_admin}
 
 
 @auth_router.get("/me")
