@@ -45,9 +45,23 @@ def main(argv: list[str] | None = None) -> int:
     mp = sub.add_parser("migrate", help="apply pending DB schema migrations")
     mp.add_argument("--database-url", default=None)
     mp.add_argument("--data-dir", default=None)
+    kp = sub.add_parser("manifests", help="print Kubernetes install manifests")
+    kp.add_argument("--namespace", default="gpustack")
+    kp.add_argument("--image", default="gpustack-amd:latest")
+    kp.add_argument("--server-url", default=None)
+    kp.add_argument("--bootstrap-password", default="admin")
+    kp.add_argument("--registration-token", default="tok_cluster")
+    kp.add_argument("--gpus-per-node", type=int, default=8)
     sub.add_parser("version")
     args = ap.parse_args(argv)
 
+    if args.cmd == "manifests":
+        from .utils.k8s_manifests import render_all
+
+        print(render_all(args.namespace, args.image, args.server_url,
+                         args.bootstrap_password, args.registration_token,
+                         args.gpus_per_node))
+        return 0
     if args.cmd == "version":
         from . import __version__
 

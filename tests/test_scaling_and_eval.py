@@ -108,3 +108,31 @@ def test_usage_archiver_moves_old_rows():
         assert s.query(ModelUsage).count() == 1
         arch = s.query(ModelUsageArchive).all()
         assert len(arch) == 1 and arch[0].prompt_tokens == 10
+
+
+def test_k8s_manifests_render():
+    """K8s install manifests (reference: gpustack/k8s/ daemonset.jinja):
+    server Deployment/Service/Secret + ROCm worker DaemonSet."""
+    import yaml
+
+    from gpustack_amd.utils.k8s_manifests import render_all
+
+    docs = list(yaml.safe_load_all(render_all(gpus_per_node=4)))
+    kinds = [d["kind"] for d in docs]
+    assert kinds == ["Namespace", "Secret", "Deployment", "Service",
+                     "DaemonSet"]
+    ds = docs[-1]["spec"]["template"]["spec"]
+    c = ds["containers"][0]
+    assert c["resources"]["limits"]["amd.com/gpu"] == 4
+    paths = {v["hostPath"]["path"] for v in ds["volumes"] if "hostPath" in v}
+    assert "/dev/kfd" in paths and "/dev/dri" in paths
+    env = {e["name"]: e for e in c["env"]}
+    assert env["HSA_ENABLE_IPC_MODE_LEGACY"]["value"] == "0"
+
+
+def test_k8s_manifests_cli(capsys):
+    from gpustack_amd.main import main
+
+    assert main(["manifests", "--namespace", "prod"]) == 0
+    out = capsys.readouterr().out
+    assert "kind: DaemonSet" in out and "namespace: prod" in out
