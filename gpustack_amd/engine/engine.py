@@ -37,7 +37,7 @@ class LLMEngine:
 
         self._async_enabled = (
             os.environ.get("GPUSTACK_AMD_ASYNC", "1") == "1"
-            and self.comm.tp_size == 1
+            and self.comm.world_size == 1
             and cfg.speculative is None
         )
         self._pending: tuple[list[Sequence], object] | None = None
@@ -57,8 +57,12 @@ class LLMEngine:
     ) -> str:
         rid = request_id or f"req-{next(self._counter)}"
         params = params or SamplingParams()
-        if self.comm.tp_size > 1:
-            assert self.comm.tp_rank == 0, "requests enter through TP rank 0"
+        if params.guided_json is not None and self.comm.pp_size > 1:
+            raise ValueError("guided_json is not supported with pipeline "
+                             "parallelism (sampling runs on the last stage, "
+                             "which has no token table)")
+        if self.comm.world_size > 1:
+            assert self.comm.world_rank == 0, "requests enter through rank 0"
             self._pending_ops.append(("add", list(prompt_token_ids),
                                       params.__dict__.copy(), rid))
             return rid
@@ -82,15 +86,15 @@ class LLMEngine:
     # -- dynamic multi-LoRA (reference: vLLM /v1/load_lora_adapter, gpustack
     # per-LoRA model routes) ------------------------------------------------
     def add_lora(self, name: str, adapter_dir: str) -> None:
-        if self.comm.tp_size > 1:
-            assert self.comm.tp_rank == 0
+        if self.comm.world_size > 1:
+            assert self.comm.world_rank == 0
             self._pending_ops.append(("lora_add", name, adapter_dir))
             return
         self.runner.add_lora(name, adapter_dir)
 
     def remove_lora(self, name: str) -> bool:
-        if self.comm.tp_size > 1:
-            assert self.comm.tp_rank == 0
+        if self.comm.world_size > 1:
+            assert self.comm.world_rank == 0
             self._pending_ops.append(("lora_rm", name))
             return True
         return self.runner.remove_lora(name)
@@ -106,8 +110,8 @@ class LLMEngine:
         return names
 
     def abort_request(self, request_id: str) -> bool:
-        if self.comm.tp_size > 1:
-            assert self.comm.tp_rank == 0
+        if self.comm.world_size > 1:
+            assert self.comm.world_rank == 0
             self._pending_ops.append(("abort", request_id))
             return True
         if self._pending is not None:
@@ -125,8 +129,9 @@ class LLMEngine:
         ranks (rank bootstrap + op replication over the RCCL/gloo group)."""
         import torch.distributed as dist
 
-        ops = [self._pending_ops] if self.comm.tp_rank == 0 else [None]
-        dist.broadcast_object_list(ops, src=0, group=self.comm.group)
+        # world-group broadcast: every TP AND PP rank replays the same ops
+        ops = [self._pending_ops] if self.comm.world_rank == 0 else [None]
+        dist.broadcast_object_list(ops, src=0)
         self._pending_ops = []
         for op in ops[0]:
             if op[0] == "add":
@@ -147,17 +152,17 @@ class LLMEngine:
         return self.scheduler.has_work()
 
     def tp_active(self) -> bool:
-        """Coordinated loop condition: every TP rank keeps stepping while
-        rank 0 has work (requests enter only through rank 0)."""
-        if self.comm.tp_size == 1:
+        """Coordinated loop condition: every rank (TP and PP) keeps
+        stepping while rank 0 has work (requests enter only through rank 0)."""
+        if self.comm.world_size == 1:
             return self.has_unfinished()
         import torch.distributed as dist
 
-        if self.comm.tp_rank == 0:
+        if self.comm.world_rank == 0:
             flag = [self.has_unfinished() or bool(self._pending_ops)]
         else:
             flag = [None]
-        dist.broadcast_object_list(flag, src=0, group=self.comm.group)
+        dist.broadcast_object_list(flag, src=0)
         return bool(flag[0])
 
     @property
@@ -239,7 +244,7 @@ class LLMEngine:
         )
 
     def step(self) -> list[StepOutput]:
-        if self.comm.tp_size > 1:
+        if self.comm.world_size > 1:
             self._sync_tp_ops()
         outputs: list[StepOutput] = []
         if self._carry_outputs:

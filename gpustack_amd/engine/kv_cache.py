@@ -132,14 +132,16 @@ def block_hashes(tokens, block_size: int) -> list[int]:
 
 
 class KVCache:
-    def __init__(self, cfg: EngineConfig, num_blocks: int, device: str | torch.device):
+    def __init__(self, cfg: EngineConfig, num_blocks: int,
+                 device: str | torch.device, num_layers: int | None = None):
         spec = cfg.spec
         self.block_size = cfg.block_size
         self.num_blocks = num_blocks
         kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
         self.kv_heads = kv_heads
         self.head_dim = spec.head_dim
-        self.num_layers = spec.num_layers
+        # pipeline stages hold KV only for their own layers
+        self.num_layers = num_layers if num_layers is not None else spec.num_layers
         shape = (num_blocks, kv_heads, cfg.block_size, spec.head_dim)
         if cfg.kv_cache_dtype == "fp8":
             dtype = torch.float8_e4m3fn
@@ -147,10 +149,10 @@ class KVCache:
             dtype = getattr(torch, cfg.dtype)
         self.kv_dtype = dtype
         self.k_caches = [
-            torch.zeros(shape, dtype=dtype, device=device) for _ in range(spec.num_layers)
+            torch.zeros(shape, dtype=dtype, device=device) for _ in range(self.num_layers)
         ]
         self.v_caches = [
-            torch.zeros(shape, dtype=dtype, device=device) for _ in range(spec.num_layers)
+            torch.zeros(shape, dtype=dtype, device=device) for _ in range(self.num_layers)
         ]
         # last block reserved as the graph-padding scratch block (see
         # engine/graph_runner.py): padded rows write/read there, never live KV
@@ -167,13 +169,13 @@ class KVCache:
         self.is_cuda = torch.device(device).type == "cuda"
         host_blocks = 0
         if cfg.kv_offload_gb > 0:
-            per_block = (2 * spec.num_layers * kv_heads * cfg.block_size
+            per_block = (2 * self.num_layers * kv_heads * cfg.block_size
                          * spec.head_dim * 2)
             host_blocks = int(cfg.kv_offload_gb * 2**30) // per_block
         self.host_blocks = host_blocks
         if host_blocks > 0:
             self.host_pool = torch.zeros(
-                (host_blocks, spec.num_layers, 2, kv_heads, cfg.block_size,
+                (host_blocks, self.num_layers, 2, kv_heads, cfg.block_size,
                  spec.head_dim),
                 dtype=self.kv_dtype, pin_memory=self.is_cuda,
             )
@@ -222,12 +224,14 @@ class KVCache:
         return gpu_blocks
 
     @staticmethod
-    def compute_num_blocks(cfg: EngineConfig, free_bytes: int) -> int:
+    def compute_num_blocks(cfg: EngineConfig, free_bytes: int,
+                           num_layers: int | None = None) -> int:
         spec = cfg.spec
+        nl = num_layers if num_layers is not None else spec.num_layers
         kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
         esize = 1 if cfg.kv_cache_dtype == "fp8" else 2
         per_block = (
-            2 * spec.num_layers * kv_heads * cfg.block_size * spec.head_dim * esize
+            2 * nl * kv_heads * cfg.block_size * spec.head_dim * esize
         )
         return max(1, int(free_bytes * cfg.gpu_memory_utilization) // per_block)
 

@@ -247,21 +247,25 @@ class ModelRunner:
 
     def init_kv_cache(self) -> KVCache:
         cfg = self.cfg
+        local_layers = self.model.num_local_layers
         if cfg.kv_cache_blocks is not None:
             nblocks = cfg.kv_cache_blocks
         elif self.device.type == "cuda":
             free, _total = torch.cuda.mem_get_info(self.device)
-            nblocks = KVCache.compute_num_blocks(cfg, free)
+            nblocks = KVCache.compute_num_blocks(cfg, free, local_layers)
         else:
             nblocks = 512
         max_needed = cfg.max_num_seqs * (
             (cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
         )
         nblocks = min(nblocks, max_needed)
-        self.kv = KVCache(cfg, nblocks, self.device)
+        self.kv = KVCache(cfg, nblocks, self.device, num_layers=local_layers)
         self.eagle = None
         spec = getattr(cfg, "speculative", None)
         if spec and spec.get("method") in ("eagle", "eagle3", "mtp"):
+            if self.comm.pp_size > 1:
+                raise ValueError("draft-model speculative decoding is not "
+                                 "supported with pipeline parallelism")
             from .eagle import EagleProposer
 
             k = int(spec.get("num_draft_tokens", 3))
@@ -275,7 +279,8 @@ class ModelRunner:
             # MoE dispatch syncs with the host per layer (expert routing),
             # which hipGraph capture cannot record -> eager decode for MoE
             # (sync-free fused dispatch is the round-2 item)
-            if graphs_enabled() and self.eagle is None and cfg.spec.num_experts == 0:
+            if graphs_enabled() and self.eagle is None \
+                    and cfg.spec.num_experts == 0 and self.comm.pp_size == 1:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv
@@ -419,7 +424,14 @@ class ModelRunner:
             seq_lens_list=lens,
             tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
         )
-        hidden = self.model(tokens, meta, self.kv, return_hidden=True).float()
+        hidden = self.model(tokens, meta, self.kv, return_hidden=True)
+        if self.comm.pp_size > 1:
+            if hidden is None:  # non-final stage: receive the broadcast
+                hidden = torch.empty(len(flat), self.cfg.spec.hidden_size,
+                                     dtype=getattr(torch, self.cfg.dtype),
+                                     device=dev)
+            self.comm.broadcast_world(hidden, src=self.comm.last_stage_rank)
+        hidden = hidden.float()
         out: list[list[float]] = []
         off = 0
         for L in lens:
@@ -490,6 +502,8 @@ class ModelRunner:
         row_seqs = batch.seqs
         if not batch.is_prefill and batch.rows_per_seq > 1:
             row_seqs = [s for s in batch.seqs for _ in range(batch.rows_per_seq)]
+        if self.comm.pp_size > 1:
+            return self._pp_finish(batch, logits, row_seqs)
         token_ids = self.sampler.sample(logits, row_seqs)
         self.last_logprobs = None
         if any(s.params.logprobs for s in batch.seqs):
@@ -505,4 +519,34 @@ class ModelRunner:
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)
             self.comm.broadcast(t, src=0)
             token_ids = t.tolist()
+        return token_ids
+
+    def _pp_finish(self, batch: ScheduledBatch, logits, row_seqs) -> list[int]:
+        """Pipeline epilogue: the last stage samples (its tp rank 0
+        decides under TPxPP) and every rank receives the token ids so the
+        replicated schedulers stay in lockstep."""
+        import torch.distributed as dist
+
+        n = len(row_seqs)
+        if logits is not None:
+            token_ids = self.sampler.sample(logits, row_seqs)
+            t = torch.tensor(token_ids, dtype=torch.long, device=self.device)
+        else:
+            t = torch.empty(n, dtype=torch.long, device=self.device)
+        self.comm.broadcast_world(t, src=self.comm.last_stage_rank)
+        token_ids = t.tolist()
+        self.last_logprobs = None
+        if any(s.params.logprobs for s in batch.seqs):
+            if logits is not None:
+                lf = logits.float()
+                lse = torch.logsumexp(lf, dim=-1)
+                chosen = lf.gather(
+                    1, torch.as_tensor(token_ids, dtype=torch.long,
+                                       device=logits.device).unsqueeze(1)
+                ).squeeze(1)
+                obj = [(chosen - lse).tolist()]
+            else:
+                obj = [None]
+            dist.broadcast_object_list(obj, src=self.comm.last_stage_rank)
+            self.last_logprobs = obj[0]
         return token_ids

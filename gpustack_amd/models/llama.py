@@ -269,24 +269,45 @@ class LlamaForCausalLM(nn.Module):
         spec = cfg.spec
         self.spec = spec
         self.cfg = cfg
+        self.comm = comm
+        self.device = torch.device(device)
         dtype = getattr(torch, cfg.dtype)
         self.dtype = dtype
-        self.embed = nn.Parameter(
-            torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype), requires_grad=False
-        )
+        # pipeline partition: this rank owns layers
+        # [layer_offset, layer_offset + num_local_layers); remainder layers
+        # go to the EARLY stages so the last stage (which also runs the
+        # lm_head GEMM and sampling) carries less per-step work
+        pp, pr = comm.pp_size, comm.pp_rank
+        counts = [spec.num_layers // pp + (1 if i < spec.num_layers % pp else 0)
+                  for i in range(pp)]
+        self.layer_offset = sum(counts[:pr])
+        self.num_local_layers = counts[pr]
+        first, last = comm.is_first_stage, comm.is_last_stage
+        if first or (last and spec.tie_word_embeddings):
+            self.embed = nn.Parameter(
+                torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype),
+                requires_grad=False)
+        else:
+            self.embed = None
         self.layers = nn.ModuleList(
-            [DecoderLayer(spec, cfg.tp_size, comm, dtype) for _ in range(spec.num_layers)]
+            [DecoderLayer(spec, cfg.tp_size, comm, dtype)
+             for _ in range(self.num_local_layers)]
         )
         for i, layer in enumerate(self.layers):
-            layer.attn.layer_idx = i
-            layer.mlp.layer_idx = i
-        self.final_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
-        if spec.tie_word_embeddings:
-            self.lm_head = self.embed
+            layer.attn.layer_idx = self.layer_offset + i
+            layer.mlp.layer_idx = self.layer_offset + i
+        if last:
+            self.final_norm = nn.Parameter(
+                torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
+            if spec.tie_word_embeddings:
+                self.lm_head = self.embed
+            else:
+                self.lm_head = nn.Parameter(
+                    torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype),
+                    requires_grad=False)
         else:
-            self.lm_head = nn.Parameter(
-                torch.empty(spec.vocab_size, spec.hidden_size, dtype=dtype), requires_grad=False
-            )
+            self.final_norm = None
+            self.lm_head = None
         cache = ops.build_cos_sin_cache(
             spec.head_dim, spec.head_dim, cfg.max_model_len,
             base=spec.rope_theta, scaling=spec.rope_scaling,
@@ -297,10 +318,23 @@ class LlamaForCausalLM(nn.Module):
     @torch.inference_mode()
     def forward(self, token_ids: torch.Tensor, meta: ForwardMeta, kv,
                 return_hidden: bool = False, return_both: bool = False):
-        x = F.embedding(token_ids, self.embed)
+        if self.comm.pp_size > 1 and not self.comm.is_first_stage:
+            # previous stage sends the summed residual stream; starting the
+            # local stack with residual=None reproduces the single-rank
+            # fused_add_rms_norm numerics exactly (it stores x+residual in
+            # bf16 before normalizing)
+            x = self.comm.recv_hidden(
+                (token_ids.shape[0], self.spec.hidden_size),
+                self.dtype, self.device)
+        else:
+            x = F.embedding(token_ids, self.embed)
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
+        if self.comm.pp_size > 1 and not self.comm.is_last_stage:
+            s = (x.float() + residual.float()).to(self.dtype)
+            self.comm.send_hidden(s)
+            return None
         ops.fused_add_rms_norm(x, residual, self.final_norm, self.spec.rms_norm_eps)
         if return_hidden:
             return x  # all rows, post final norm (embedding serving)

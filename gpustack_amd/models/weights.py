@@ -38,17 +38,21 @@ def _gen(shape, seed_key: str, base_seed: int, dtype, device, std=0.02):
 def random_init(model, cfg: EngineConfig) -> None:
     spec = model.spec
     tp, rank = cfg.tp_size, cfg.tp_rank
-    dtype, device = model.dtype, model.embed.device
+    dtype, device = model.dtype, model.device
     seed = cfg.seed
     hq, hkv, d = spec.num_heads // tp, max(1, spec.num_kv_heads // tp), spec.head_dim
     i_loc = spec.intermediate_size // tp
+    off = getattr(model, "layer_offset", 0)  # pipeline stage-local layers
 
-    model.embed.copy_(_gen((spec.vocab_size, spec.hidden_size), "embed", seed, dtype, device))
-    if not spec.tie_word_embeddings:
+    if model.embed is not None:
+        model.embed.copy_(_gen((spec.vocab_size, spec.hidden_size), "embed", seed, dtype, device))
+    if model.lm_head is not None and not spec.tie_word_embeddings:
         model.lm_head.copy_(_gen((spec.vocab_size, spec.hidden_size), "lm_head", seed, dtype, device))
-    model.final_norm.fill_(1.0)
+    if model.final_norm is not None:
+        model.final_norm.fill_(1.0)
 
-    for li, layer in enumerate(model.layers):
+    for local_i, layer in enumerate(model.layers):
+        li = off + local_i
         q_full = _gen((spec.num_heads * d, spec.hidden_size), f"{li}.q", seed, dtype, device)
         k_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.k", seed, dtype, device)
         v_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.v", seed, dtype, device)
@@ -132,11 +136,15 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
         return t.to(model.dtype)
 
     pre = "model."
-    model.embed.copy_(get(pre + "embed_tokens.weight"))
-    model.final_norm.copy_(get(pre + "norm.weight"))
-    if not spec.tie_word_embeddings:
+    off = getattr(model, "layer_offset", 0)
+    if model.embed is not None:
+        model.embed.copy_(get(pre + "embed_tokens.weight"))
+    if model.final_norm is not None:
+        model.final_norm.copy_(get(pre + "norm.weight"))
+    if model.lm_head is not None and not spec.tie_word_embeddings:
         model.lm_head.copy_(get("lm_head.weight"))
-    for li, layer in enumerate(model.layers):
+    for local_i, layer in enumerate(model.layers):
+        li = off + local_i
         p = f"{pre}layers.{li}."
         q = row_shard(get(p + "self_attn.q_proj.weight"), hq * d)
         k = row_shard(get(p + "self_attn.k_proj.weight"), hkv * d)
@@ -244,7 +252,9 @@ def merge_lora(model, cfg: EngineConfig, adapter_dir: str | Path) -> int:
         return None
 
     merged = 0
-    for li, layer in enumerate(model.layers):
+    off = getattr(model, "layer_offset", 0)
+    for local_i, layer in enumerate(model.layers):
+        li = off + local_i
         # fused qkv: rows [0,hq*d) = q shard, then k, then v
         offsets = {
             "q_proj": (0, hq * d, rank * hq * d),
@@ -328,14 +338,18 @@ def load_gguf(model, cfg: EngineConfig, gguf_path: str | Path) -> None:
     def row_shard(t, per):
         return t[rank * per:(rank + 1) * per]
 
-    model.embed.copy_(get("token_embd.weight"))
-    model.final_norm.copy_(get("output_norm.weight"))
-    if not spec.tie_word_embeddings:
+    off = getattr(model, "layer_offset", 0)
+    if model.embed is not None:
+        model.embed.copy_(get("token_embd.weight"))
+    if model.final_norm is not None:
+        model.final_norm.copy_(get("output_norm.weight"))
+    if model.lm_head is not None and not spec.tie_word_embeddings:
         name = "output.weight" if "output.weight" in by_name else "token_embd.weight"
         model.lm_head.copy_(get(name))
     up_q = spec.num_heads if permuted_qk else None
     up_k = spec.num_kv_heads if permuted_qk else None
-    for li, layer in enumerate(model.layers):
+    for local_i, layer in enumerate(model.layers):
+        li = off + local_i
         p = f"blk.{li}."
         q = row_shard(get(p + "attn_q.weight", up_q), hq * d)
         k = row_shard(get(p + "attn_k.weight", up_k), hkv * d)

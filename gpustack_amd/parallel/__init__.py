@@ -1,3 +1,3 @@
-from .comm import Communicator, get_communicator, init_tp
+from .comm import Communicator, get_communicator, init_parallel, init_tp
 
-__all__ = ["Communicator", "get_communicator", "init_tp"]
+__all__ = ["Communicator", "get_communicator", "init_parallel", "init_tp"]

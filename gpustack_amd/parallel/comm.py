@@ -19,12 +19,58 @@ import torch.distributed as dist
 
 
 class Communicator:
-    """TP group communicator; degenerates to no-ops at world size 1."""
+    """TP (+PP) communicator; degenerates to no-ops at world size 1.
 
-    def __init__(self, tp_size: int = 1, tp_rank: int = 0, group=None):
+    Rank layout: global = pp_rank * tp_size + tp_rank — TP groups stay
+    contiguous so their RCCL rings ride node-local xGMI links, while PP
+    boundaries (one point-to-point hidden-state transfer per stage per
+    step) are the only traffic that crosses nodes. `group` is the TP
+    subgroup; PP send/recv and world broadcasts use the default group."""
+
+    def __init__(self, tp_size: int = 1, tp_rank: int = 0, group=None,
+                 pp_size: int = 1, pp_rank: int = 0):
         self.tp_size = tp_size
         self.tp_rank = tp_rank
         self.group = group
+        self.pp_size = pp_size
+        self.pp_rank = pp_rank
+
+    @property
+    def world_size(self) -> int:
+        return self.tp_size * self.pp_size
+
+    @property
+    def world_rank(self) -> int:
+        return self.pp_rank * self.tp_size + self.tp_rank
+
+    @property
+    def is_first_stage(self) -> bool:
+        return self.pp_rank == 0
+
+    @property
+    def is_last_stage(self) -> bool:
+        return self.pp_rank == self.pp_size - 1
+
+    @property
+    def last_stage_rank(self) -> int:
+        """Global rank of (last stage, tp_rank 0) — the sampling rank."""
+        return (self.pp_size - 1) * self.tp_size
+
+    # -- pipeline point-to-point -------------------------------------------
+    def send_hidden(self, t: torch.Tensor) -> None:
+        dst = (self.pp_rank + 1) * self.tp_size + self.tp_rank
+        dist.send(t.contiguous(), dst=dst)
+
+    def recv_hidden(self, shape, dtype, device) -> torch.Tensor:
+        src = (self.pp_rank - 1) * self.tp_size + self.tp_rank
+        t = torch.empty(shape, dtype=dtype, device=device)
+        dist.recv(t, src=src)
+        return t
+
+    def broadcast_world(self, t: torch.Tensor, src: int) -> torch.Tensor:
+        if self.world_size > 1:
+            dist.broadcast(t, src=src)
+        return t
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.tp_size > 1:
@@ -54,11 +100,22 @@ def get_communicator() -> Communicator:
 def init_tp(tp_size: int, tp_rank: int, master_port: int | None = None,
             backend: str | None = None, device_id: int | None = None,
             master_addr: str | None = None) -> Communicator:
-    """Initialize the TP process group (rank bootstrap via TCP store on
+    """Initialize a pure-TP process group (rank bootstrap via TCP store on
     127.0.0.1, replacing the master-port scheme the reference's port
     allocator models — serve_manager.py:1685-1737)."""
+    return init_parallel(tp_size, 1, tp_rank, master_port=master_port,
+                         backend=backend, device_id=device_id,
+                         master_addr=master_addr)
+
+
+def init_parallel(tp_size: int, pp_size: int, global_rank: int,
+                  master_port: int | None = None, backend: str | None = None,
+                  device_id: int | None = None,
+                  master_addr: str | None = None) -> Communicator:
+    """Initialize a TP x PP process group (global = pp*tp_size + tp)."""
     global _COMM
-    if tp_size <= 1:
+    world = tp_size * pp_size
+    if world <= 1:
         _COMM = Communicator()
         return _COMM
     if backend is None:
@@ -70,8 +127,17 @@ def init_tp(tp_size: int, tp_rank: int, master_port: int | None = None,
         if master_port is not None:
             os.environ["MASTER_PORT"] = str(master_port)
         os.environ.setdefault("MASTER_PORT", "29500")
-        dist.init_process_group(backend=backend, world_size=tp_size, rank=tp_rank)
+        dist.init_process_group(backend=backend, world_size=world,
+                                rank=global_rank)
     if backend == "nccl" and device_id is not None:
         torch.cuda.set_device(device_id)
-    _COMM = Communicator(tp_size, tp_rank)
+    pp_rank, tp_rank = divmod(global_rank, tp_size)
+    tp_group = None
+    if pp_size > 1 and tp_size > 1:
+        # every rank must create every subgroup (collective contract)
+        for p in range(pp_size):
+            g = dist.new_group(list(range(p * tp_size, (p + 1) * tp_size)))
+            if p == pp_rank:
+                tp_group = g
+    _COMM = Communicator(tp_size, tp_rank, tp_group, pp_size, pp_rank)
     return _COMM

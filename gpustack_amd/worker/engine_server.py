@@ -107,7 +107,7 @@ class EngineRunner:
         return await fut
 
     def _run(self) -> None:
-        tp = self.engine.comm.tp_size > 1
+        tp = self.engine.comm.world_size > 1
         while not self._stop:
             if self._aux_jobs:
                 self._run_aux_jobs()
@@ -880,6 +880,9 @@ def main():
     ap.add_argument("--kv-cache-blocks", type=int, default=None)
     ap.add_argument("--backend-parameters", default="{}")
     ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--pp", type=int, default=1,
+                    help="pipeline stages (layers partitioned across ranks; "
+                         "global rank = pp_rank * tp + tp_rank)")
     ap.add_argument("--tp-rank", type=int, default=None,
                     help="absolute rank of THIS process (set for spawned "
                          "followers; unset = per-worker parent)")
@@ -901,10 +904,11 @@ def main():
     device = args.device or ("cuda" if use_cuda else "cpu")
     comm = None
     followers = []
+    world = args.tp * args.pp
     if args.local_ranks is None:
-        args.local_ranks = args.tp if args.tp_rank is None else 1
-    if args.tp > 1:
-        from ..parallel import init_tp
+        args.local_ranks = world if args.tp_rank is None else 1
+    if world > 1:
+        from ..parallel import init_parallel
 
         if args.tp_rank is None:
             # per-worker parent: hosts ranks [rank_base, rank_base+local)
@@ -922,7 +926,7 @@ def main():
                 "--max-num-seqs", str(args.max_num_seqs),
                 "--gpu-memory-utilization", str(args.gpu_memory_utilization),
                 "--backend-parameters", args.backend_parameters,
-                "--tp", str(args.tp),
+                "--tp", str(args.tp), "--pp", str(args.pp),
                 "--rank-base", str(args.rank_base),
                 "--master-addr", args.master_addr,
             ]
@@ -938,9 +942,10 @@ def main():
         if use_cuda:
             device = f"cuda:{local_ordinal}"
             torch.cuda.set_device(local_ordinal)
-        comm = init_tp(args.tp, args.tp_rank, master_port=args.master_port,
-                       device_id=local_ordinal if use_cuda else None,
-                       master_addr=args.master_addr)
+        comm = init_parallel(args.tp, args.pp, args.tp_rank,
+                             master_port=args.master_port,
+                             device_id=local_ordinal if use_cuda else None,
+                             master_addr=args.master_addr)
 
     extra = json.loads(args.backend_parameters)
     cfg_kwargs = dict(
@@ -951,7 +956,7 @@ def main():
         gpu_memory_utilization=args.gpu_memory_utilization,
         kv_cache_blocks=args.kv_cache_blocks,
         tp_size=args.tp,
-        tp_rank=args.tp_rank or 0,
+        tp_rank=(args.tp_rank or 0) % args.tp,
     )
     if device == "cpu" and args.kv_cache_blocks is None:
         cfg_kwargs["kv_cache_blocks"] = 1024
@@ -964,7 +969,7 @@ def main():
         else:
             ecfg.model_dir = args.model_ref
 
-    if args.tp > 1 and args.tp_rank != 0:
+    if world > 1 and args.tp_rank != 0:
         # follower rank: no HTTP; run the coordinated engine loop forever
         import time as _time
 

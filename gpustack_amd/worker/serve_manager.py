@@ -195,8 +195,14 @@ class ServeManager:
                      "--master-addr", ds["master_ip"] or "127.0.0.1",
                      "--master-port", str(ds["master_port"])]
         elif len(gpus) > 1:
-            # TP replica sharded over the scheduled GPUs (RCCL over xGMI)
-            args += ["--tp", str(len(gpus))]
+            # TP replica sharded over the scheduled GPUs (RCCL over xGMI);
+            # backend_parameters pp_size splits layers into pipeline stages
+            # instead (tp x pp = scheduled GPUs)
+            pp = int(bp.pop("pp_size", 1) or 1)
+            if pp > 1 and len(gpus) % pp == 0:
+                args += ["--tp", str(len(gpus) // pp), "--pp", str(pp)]
+            else:
+                args += ["--tp", str(len(gpus))]
         if bp:
             args += ["--backend-parameters", json.dumps(bp)]
 
