@@ -131,8 +131,12 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             for name in sf.keys():
                 tensors[name] = sf.get_tensor(name)
 
+    from .quantized import maybe_dequant, quant_config
+
+    qc = quant_config(model_dir)  # GPTQ/AWQ: dequant packed linears on load
+
     def get(name):
-        t = tensors[name]
+        t = maybe_dequant(tensors, name, qc)
         return t.to(model.dtype)
 
     pre = "model."
