@@ -30,10 +30,15 @@ _HEX = "0123456789abcdefABCDEF"
 
 
 class StringM:
-    """A JSON string including both quotes."""
+    """A JSON string including both quotes. Supports the JSON-Schema
+    maxLength/minLength bounds (counted in raw body characters, so escape
+    sequences count conservatively high — the cap still holds)."""
 
-    def __init__(self):
+    def __init__(self, max_len: int | None = None, min_len: int = 0):
         self.state = 0  # 0=expect open ", 1=body, 2=escape, 3..6=\uXXXX, 7=done
+        self.max_len = max_len
+        self.min_len = min_len
+        self.n = 0
 
     @property
     def complete(self) -> bool:
@@ -48,8 +53,13 @@ class StringM:
             return False
         if s == 1:
             if ch == '"':
+                if self.n < self.min_len:
+                    return False
                 self.state = 7
                 return True
+            if self.max_len is not None and self.n >= self.max_len:
+                return False  # only the closing quote may follow
+            self.n += 1
             if ch == "\\":
                 self.state = 2
                 return True
@@ -365,7 +375,8 @@ def _value_machine(schema: dict):
 
         return lambda: LitM([_json.dumps(v) for v in schema["enum"]])
     if t == "string":
-        return StringM
+        mx, mn = schema.get("maxLength"), schema.get("minLength", 0)
+        return lambda: StringM(mx, mn)
     if t == "integer":
         return lambda: NumberM(integer_only=True)
     if t == "number":

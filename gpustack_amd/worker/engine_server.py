@@ -181,6 +181,22 @@ class EngineRunner:
         self.release(rid)
 
 
+def _tool_call_message(rid: str, body: dict, text: str) -> dict:
+    """Shape a generated (guided) JSON into an OpenAI tool_calls message."""
+    if body.get("_tool_envelope"):
+        try:
+            doc = json.loads(text)
+            name = doc.get("name")
+            args = json.dumps(doc.get("arguments", {}))
+        except json.JSONDecodeError:
+            name, args = None, text
+    else:
+        name, args = body.get("_tool_name"), text
+    return {"role": "assistant", "content": None,
+            "tool_calls": [{"id": f"call_{rid}", "type": "function",
+                            "function": {"name": name, "arguments": args}}]}
+
+
 def _stop_strings(body: dict) -> list[str]:
     stop = body.get("stop")
     if stop is None:
@@ -348,6 +364,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
                 "choices": [{"index": 0, "text": text, "finish_reason": finish}],
             }
 
+        tool_mode = kind == "chat" and bool(body.get("_tool_name")
+                                            or body.get("_tool_envelope"))
         if stream:
             async def gen():
                 tokens: list[int] = []
@@ -367,11 +385,25 @@ def create_app(runner: EngineRunner) -> FastAPI:
                         text = runner.tokenizer.decode(tokens)
                         new = text[sent_len:]
                         # hold back partial unicode replacement chars
-                        if new and not new.endswith("�"):
+                        if new and not new.endswith("�") and not tool_mode:
                             sent_len = len(text)
                             payload = (chat_chunk({"content": new}, None)
                                        if kind == "chat" else text_chunk(new, None))
                             yield f"data: {json.dumps(payload)}\n\n"
+                        if item["finished"] and tool_mode:
+                            msg = _tool_call_message(rid, body,
+                                                     runner.tokenizer.decode(tokens))
+                            yield f"data: {json.dumps(chat_chunk({'tool_calls': msg['tool_calls']}, None))}\n\n"
+                            usage = {
+                                "prompt_tokens": len(prompt_ids),
+                                "completion_tokens": len(tokens),
+                                "total_tokens": len(prompt_ids) + len(tokens),
+                            }
+                            payload = chat_chunk({}, "tool_calls")
+                            payload["usage"] = usage
+                            yield f"data: {json.dumps(payload)}\n\n"
+                            yield "data: [DONE]\n\n"
+                            break
                         if item["finished"]:
                             fin = item.get("finish_reason") or "stop"
                             usage = {
@@ -422,6 +454,15 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "total_tokens": len(prompt_ids) + len(tokens),
         }
         want_lp = bool(body.get("logprobs")) and any(x is not None for x in lps)
+        if kind == "chat" and (body.get("_tool_name") or body.get("_tool_envelope")):
+            return JSONResponse({
+                "id": rid, "object": "chat.completion", "created": created,
+                "model": model_name,
+                "choices": [{"index": 0,
+                             "message": _tool_call_message(rid, body, text),
+                             "finish_reason": "tool_calls"}],
+                "usage": usage,
+            })
         if kind == "chat":
             choice = {"index": 0,
                       "message": {"role": "assistant", "content": text},
@@ -447,15 +488,45 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "model": model_name, "choices": [choice], "usage": usage,
         })
 
+    def _setup_tool_calling(body: dict) -> None:
+        """OpenAI tool calling (reference: engines' function-calling surface
+        proxied by gpustack). A forced tool_choice or "required" constrains
+        generation with the function's JSON-Schema via guided decoding;
+        "auto" stays unconstrained (plain content answer)."""
+        tools = body.get("tools") or []
+        tc = body.get("tool_choice")
+        funcs = {t["function"]["name"]: t["function"]
+                 for t in tools if t.get("type") == "function"
+                 and t.get("function", {}).get("name")}
+        if not funcs or tc == "none":
+            return
+        if isinstance(tc, dict) and tc.get("type") == "function":
+            name = (tc.get("function") or {}).get("name")
+            if name not in funcs:
+                raise HTTPException(400, f"unknown tool {name!r}")
+            body["guided_json"] = funcs[name].get("parameters") or True
+            body["_tool_name"] = name
+        elif tc == "required":
+            body["guided_json"] = {
+                "type": "object",
+                "properties": {
+                    "name": {"enum": sorted(funcs)},
+                    "arguments": {},  # any JSON value
+                },
+            }
+            body["_tool_envelope"] = True
+
     @app.post("/v1/chat/completions")
     async def chat(request: Request):
         body = await request.json()
         messages = body.get("messages") or []
+        _setup_tool_calling(body)
         tok = runner.tokenizer
         if hasattr(tok, "apply_chat_template"):
             try:
-                prompt = tok.apply_chat_template(messages, tokenize=False,
-                                                 add_generation_prompt=True)
+                prompt = tok.apply_chat_template(
+                    messages, tokenize=False, add_generation_prompt=True,
+                    tools=body.get("tools") or None)
             except TypeError:
                 prompt = tok.apply_chat_template(messages)
         else:

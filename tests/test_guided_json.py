@@ -194,3 +194,124 @@ def test_engine_server_response_format():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_tool_calling():
+    """OpenAI tool calling: forced tool_choice constrains arguments with the
+    function's JSON-Schema; "required" uses a name+arguments envelope."""
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-t", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    base = f"http://127.0.0.1:{port}"
+    # maxLength bounds keep the random-init model's strings finite
+    tools = [{"type": "function", "function": {
+        "name": "get_weather",
+        "parameters": {"type": "object",
+                       "properties": {"city": {"type": "string",
+                                               "maxLength": 12},
+                                      "celsius": {"type": "boolean"}}}}},
+        {"type": "function", "function": {
+            "name": "get_time",
+            "parameters": {"type": "object",
+                           "properties": {"tz": {"type": "string",
+                                                 "maxLength": 12}}}}}]
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 90:
+            if proc.poll() is not None:
+                raise AssertionError(f"engine server exited {proc.returncode}")
+            try:
+                if httpx.get(f"{base}/health", timeout=2).status_code == 200:
+                    break
+            except httpx.HTTPError:
+                pass
+            time.sleep(0.5)
+        else:
+            raise AssertionError("engine server never became healthy")
+
+        msg = [{"role": "user", "content": "weather in Oslo?"}]
+        # forced function
+        r = httpx.post(f"{base}/v1/chat/completions", json={
+            "model": "tiny-t", "messages": msg, "max_tokens": 60,
+            "temperature": 0, "tools": tools,
+            "tool_choice": {"type": "function",
+                            "function": {"name": "get_weather"}},
+        }, timeout=120)
+        assert r.status_code == 200, r.text
+        ch = r.json()["choices"][0]
+        assert ch["finish_reason"] == "tool_calls"
+        tc = ch["message"]["tool_calls"][0]
+        assert tc["function"]["name"] == "get_weather"
+        args = json.loads(tc["function"]["arguments"])
+        assert set(args) == {"city", "celsius"}
+        assert isinstance(args["city"], str) and isinstance(args["celsius"], bool)
+
+        # required: envelope picks one of the declared tools
+        r = httpx.post(f"{base}/v1/chat/completions", json={
+            "model": "tiny-t", "messages": msg, "max_tokens": 80,
+            "temperature": 0, "tools": tools, "tool_choice": "required",
+        }, timeout=120)
+        ch = r.json()["choices"][0]
+        assert ch["finish_reason"] == "tool_calls"
+        tc = ch["message"]["tool_calls"][0]
+        assert tc["function"]["name"] in ("get_weather", "get_time")
+        json.loads(tc["function"]["arguments"])
+
+        # tool_choice none: plain content answer
+        r = httpx.post(f"{base}/v1/chat/completions", json={
+            "model": "tiny-t", "messages": msg, "max_tokens": 8,
+            "temperature": 0, "ignore_eos": True,
+            "tools": tools, "tool_choice": "none",
+        }, timeout=120)
+        ch = r.json()["choices"][0]
+        assert ch["message"].get("content")
+        assert "tool_calls" not in ch["message"]
+
+        # unknown forced tool -> 400
+        r = httpx.post(f"{base}/v1/chat/completions", json={
+            "model": "tiny-t", "messages": msg, "tools": tools,
+            "tool_choice": {"type": "function", "function": {"name": "nope"}},
+        }, timeout=60)
+        assert r.status_code == 400
+
+        # streaming: single tool_calls delta then finish
+        deltas = []
+        with httpx.stream("POST", f"{base}/v1/chat/completions", json={
+            "model": "tiny-t", "messages": msg, "max_tokens": 60,
+            "temperature": 0, "stream": True, "tools": tools,
+            "tool_choice": {"type": "function",
+                            "function": {"name": "get_weather"}},
+        }, timeout=120) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:") and "[DONE]" not in line:
+                    deltas.append(json.loads(line[5:]))
+        tc_deltas = [d for d in deltas
+                     if d["choices"][0]["delta"].get("tool_calls")]
+        assert len(tc_deltas) == 1
+        args = json.loads(tc_deltas[0]["choices"][0]["delta"]["tool_calls"][0]
+                          ["function"]["arguments"])
+        assert set(args) == {"city", "celsius"}
+        assert deltas[-1]["choices"][0]["finish_reason"] == "tool_calls"
+        # no plain-content deltas leaked in tool mode
+        assert not any(d["choices"][0]["delta"].get("content") for d in deltas)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
