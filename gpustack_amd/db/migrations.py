@@ -54,6 +54,14 @@ def _m006_worker_pools(conn):
     WorkerPool.__table__.create(conn, checkfirst=True)
 
 
+def _m007_clusters(conn):
+    from ..schemas.tables import Cluster
+    Cluster.__table__.create(conn, checkfirst=True)
+    _add_column(conn, "workers", "cluster_id", "INTEGER")
+    _add_column(conn, "models", "cluster_id", "INTEGER")
+    _add_column(conn, "registration_tokens", "cluster_id", "INTEGER")
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -61,6 +69,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (4, "instance spec_hash for update-triggered redeploy", _m004_instance_spec_hash),
     (5, "model.lora_adapters for dynamic multi-LoRA", _m005_model_lora_adapters),
     (6, "worker_pools table for auto-provisioned capacity", _m006_worker_pools),
+    (7, "multi-cluster: clusters table + cluster_id columns", _m007_clusters),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

@@ -87,7 +87,17 @@ def gpu_arch_filter(workers: list[dict], model: dict) -> list[dict]:
     return out
 
 
-FILTER_CHAIN = [status_filter, label_filter, gpu_arch_filter]
+def cluster_filter(workers: list[dict], model: dict) -> list[dict]:
+    """Multi-cluster scoping (reference: policies/worker_filters
+    ClusterFilter): a model pinned to a cluster only places on that
+    cluster's workers; unpinned models see every worker."""
+    cid = model.get("cluster_id")
+    if not cid:
+        return workers
+    return [w for w in workers if w.get("cluster_id") == cid]
+
+
+FILTER_CHAIN = [cluster_filter, status_filter, label_filter, gpu_arch_filter]
 
 
 # ---- allocation accounting ------------------------------------------------

@@ -4,7 +4,7 @@ from __future__ import annotations
 from pydantic import BaseModel, Field
 
 from .tables import (  # noqa: F401
-    ApiKey,
+    ApiKey, Cluster,
     Benchmark,
     Model,
     ModelFile,
@@ -51,6 +51,7 @@ class ModelCreate(BaseModel):
     name: str
     source: str = SourceEnum.PRESET.value
     model_ref: str = "llama-3-8b"
+    cluster_id: int | None = None  # None = any cluster
     description: str = ""
     replicas: int = 1
     categories: list[str] = Field(default_factory=lambda: ["llm"])
@@ -144,6 +145,11 @@ class BenchmarkCreate(BaseModel):
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64
+
+
+class ClusterCreate(BaseModel):
+    name: str
+    description: str = ""
 
 
 class WorkerPoolCreate(BaseModel):

@@ -88,12 +88,27 @@ class RegistrationToken(Base, TimestampMixin, SerializeMixin):
     id = Column(Integer, primary_key=True)
     token = Column(String(128), unique=True, index=True, nullable=False)
     description = Column(String(256), default="")
+    cluster_id = Column(Integer, ForeignKey("clusters.id"), nullable=True)
+
+
+class Cluster(Base, TimestampMixin, SerializeMixin):
+    """Multi-cluster: workers register into a cluster via cluster-scoped
+    registration tokens; models deploy into one cluster (reference:
+    schemas/clusters.py — Docker/K8s/cloud clusters with per-cluster
+    tokens and system principals)."""
+    __tablename__ = "clusters"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    description = Column(Text, default="")
+    is_default = Column(Boolean, default=False)
 
 
 class Worker(Base, TimestampMixin, SerializeMixin):
     __tablename__ = "workers"
     id = Column(Integer, primary_key=True)
     name = Column(String(256), unique=True, nullable=False)
+    cluster_id = Column(Integer, ForeignKey("clusters.id"), nullable=True,
+                        index=True)
     hostname = Column(String(256), default="")
     ip = Column(String(64), default="")
     port = Column(Integer, default=10150)
@@ -121,6 +136,7 @@ class Model(Base, TimestampMixin, SerializeMixin):
     # PRESET: preset name (llama-3-8b...); LOCAL_PATH: dir with safetensors;
     # HUGGING_FACE: repo id (downloaded by the worker model-file manager)
     model_ref = Column(String(512), nullable=False)
+    cluster_id = Column(Integer, ForeignKey("clusters.id"), nullable=True)
     replicas = Column(Integer, default=1)
     categories = Column(JSON, default=lambda: ["llm"])
     placement_strategy = Column(String(32), default=PlacementStrategy.BINPACK.value)

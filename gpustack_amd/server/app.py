@@ -15,7 +15,7 @@ from fastapi.responses import JSONResponse
 
 from ..config import Config
 from ..db import ar_create, get_session, init_db
-from ..schemas import LoginRequest, RegistrationToken, User
+from ..schemas import Cluster, LoginRequest, RegistrationToken, User
 from ..security import generate_registration_token, hash_password, jwt_encode, verify_password
 from . import deps
 from .deps import COOKIE_NAME, get_current_user
@@ -168,12 +168,22 @@ def bootstrap_data(cfg: Config) -> dict:
             ar_create(s, admin)
             out["admin_password"] = password
             logger.info("bootstrap admin user created (password: %s)", password)
+        default = s.query(Cluster).filter_by(is_default=True).first()
+        if default is None:
+            default = Cluster(name="default", description="default cluster",
+                              is_default=True)
+            ar_create(s, default)
         tok = s.query(RegistrationToken).first()
         if tok is None:
             value = cfg.token or generate_registration_token()
-            tok = RegistrationToken(token=value, description="default")
+            tok = RegistrationToken(token=value, description="default",
+                                    cluster_id=default.id)
             ar_create(s, tok)
+        elif tok.cluster_id is None:
+            tok.cluster_id = default.id
+            s.commit()
         out["registration_token"] = tok.token
+        out["default_cluster_id"] = default.id
     return out
 
 

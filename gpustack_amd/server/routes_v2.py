@@ -15,7 +15,8 @@ from fastapi.responses import StreamingResponse
 
 from ..db import Event, EventType, ar_create, ar_delete, ar_update, bus, get_session
 from ..schemas import (
-    ApiKey, ApiKeyCreate, Benchmark, BenchmarkCreate, Model, ModelCreate,
+    ApiKey, ApiKeyCreate, Benchmark, BenchmarkCreate, ClusterCreate, Model,
+    ModelCreate,
     ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
     ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
     ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
@@ -140,13 +141,28 @@ def list_workers(request: Request, watch: bool = Query(False),
     return {"items": rows}
 
 
+def _token_cluster_id(request: Request) -> int | None:
+    """Cluster the presented registration token is scoped to (multi-cluster:
+    reference per-cluster registration tokens, schemas/clusters.py)."""
+    auth = request.headers.get("authorization", "")
+    token = auth.removeprefix("Bearer ").strip()
+    if not token:
+        return None
+    with get_session() as s:
+        row = s.query(RegistrationToken).filter_by(token=token).first()
+        return row.cluster_id if row else None
+
+
 @router.post("/workers/register")
 def register_worker(body: WorkerRegister, request: Request,
                     _=Depends(verify_worker_token)):
+    cluster_id = _token_cluster_id(request)
     with get_session() as s:
         w = s.query(Worker).filter_by(name=body.name).first()
         if w is None:
             w = Worker(name=body.name)
+        if cluster_id is not None:
+            w.cluster_id = cluster_id
         w.hostname = body.hostname
         w.ip = body.ip or (request.client.host if request.client else "")
         w.port = body.port
@@ -312,6 +328,59 @@ def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
 
 
 # ---- model routes ----------------------------------------------------------
+
+@router.get("/clusters")
+def list_clusters(_: User = Depends(get_current_user)):
+    from ..schemas import Cluster
+
+    with get_session() as s:
+        items = []
+        for c in s.query(Cluster).all():
+            d = c.to_dict()
+            d["workers"] = s.query(Worker).filter_by(cluster_id=c.id).count()
+            items.append(d)
+        return {"items": items}
+
+
+@router.post("/clusters", status_code=201)
+def create_cluster(body: ClusterCreate, _: User = Depends(get_admin_user)):
+    """Creates the cluster AND a registration token scoped to it — workers
+    registering with that token land in this cluster."""
+    from ..schemas import Cluster
+    from ..security import generate_registration_token
+
+    with get_session() as s:
+        if s.query(Cluster).filter_by(name=body.name).first():
+            raise HTTPException(409, "cluster exists")
+        c = Cluster(name=body.name, description=body.description)
+        ar_create(s, c)
+        tok = RegistrationToken(token=generate_registration_token(),
+                                description=f"cluster {body.name}",
+                                cluster_id=c.id)
+        ar_create(s, tok)
+        d = c.to_dict()
+        d["registration_token"] = tok.token
+        return d
+
+
+@router.delete("/clusters/{cluster_id}")
+def delete_cluster(cluster_id: int, _: User = Depends(get_admin_user)):
+    from ..schemas import Cluster
+
+    with get_session() as s:
+        c = s.get(Cluster, cluster_id)
+        if not c:
+            raise HTTPException(404, "cluster not found")
+        if c.is_default:
+            raise HTTPException(400, "cannot delete the default cluster")
+        if s.query(Worker).filter_by(cluster_id=cluster_id).count():
+            raise HTTPException(409, "cluster still has workers")
+        for t in s.query(RegistrationToken).filter_by(cluster_id=cluster_id).all():
+            s.delete(t)
+        s.commit()  # tokens must go before the cluster row (FK, no ORM rel)
+        ar_delete(s, c)
+        return {"deleted": cluster_id}
+
 
 @router.get("/worker_pools")
 def list_worker_pools(_: User = Depends(get_current_user)):

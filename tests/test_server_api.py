@@ -364,3 +364,42 @@ def test_model_update_redeploys_instances(server):
     except _q.Empty:
         pass
     bus.unsubscribe("model_instances", q)
+
+
+def test_multi_cluster_registration_and_placement(server):
+    """Clusters: token-scoped worker registration + scheduler cluster
+    filter (reference: schemas/clusters.py per-cluster tokens)."""
+    client, app, cfg, reg = server
+    # default cluster exists and has the bootstrap token
+    clusters = client.get("/v2/clusters").json()["items"]
+    assert any(c["is_default"] for c in clusters)
+    default_id = next(c["id"] for c in clusters if c["is_default"])
+
+    r = client.post("/v2/clusters", json={"name": "edge"})
+    assert r.status_code == 201
+    edge = r.json()
+    edge_token = edge["registration_token"]
+    assert edge_token and edge["id"] != default_id
+
+    # one worker per cluster: default token vs edge token
+    _register_worker(client, reg, name="w-default")
+    _register_worker(client, edge_token, name="w-edge")
+    ws = {w["name"]: w for w in client.get("/v2/workers").json()["items"]}
+    assert ws["w-default"]["cluster_id"] == default_id
+    assert ws["w-edge"]["cluster_id"] == edge["id"]
+
+    # placement respects the model's cluster pin
+    from gpustack_amd.scheduler.policies import cluster_filter
+
+    workers = list(ws.values())
+    got = cluster_filter(workers, {"cluster_id": edge["id"]})
+    assert [w["name"] for w in got] == ["w-edge"]
+    assert len(cluster_filter(workers, {"cluster_id": None})) == 2
+
+    # default cluster cannot be deleted; empty one can
+    rid = next(c["id"] for c in clusters if c["is_default"])
+    assert client.delete(f"/v2/clusters/{rid}").status_code == 400
+    assert client.delete(f"/v2/clusters/{edge['id']}").status_code == 409  # has workers
+    wid = ws["w-edge"]["id"]
+    client.delete(f"/v2/workers/{wid}")
+    assert client.delete(f"/v2/clusters/{edge['id']}").status_code == 200
