@@ -37,10 +37,12 @@ PROXY_TIMEOUT = 1800.0
 OPENAI_PATHS = [
     "/v1/chat/completions",
     "/v1/completions",
+    "/v1/responses",
     "/v1/embeddings",
     "/v1/rerank",
     "/v1/score",
     "/v1/messages",
+    "/v1/messages/count_tokens",
 ]
 
 
@@ -301,3 +303,13 @@ async def anthropic_messages(request: Request, user: User = Depends(get_current_
 @router.post("/v1/score")
 async def score(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/score", user)
+
+
+@router.post("/v1/responses")
+async def responses(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/responses", user)
+
+
+@router.post("/v1/messages/count_tokens")
+async def count_tokens(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/messages/count_tokens", user)

@@ -918,6 +918,81 @@ def create_app(runner: EngineRunner) -> FastAPI:
         return await _generate(request, body, ids, "text",
                                echo_text_prefix=prompt if echo else "")
 
+    @app.post("/v1/responses")
+    async def responses(request: Request):
+        """OpenAI Responses API, minimal non-stream shape (reference
+        endpoint registry includes /responses, gateway/utils.py:167-194):
+        `input` is a string or message list; answer comes back as one
+        output_text item."""
+        body = await request.json()
+        inp = body.get("input", "")
+        messages = []
+        if body.get("instructions"):
+            messages.append({"role": "system",
+                             "content": body["instructions"]})
+        if isinstance(inp, str):
+            messages.append({"role": "user", "content": inp})
+        else:
+            for m in inp or []:
+                content = m.get("content")
+                if isinstance(content, list):  # typed content parts
+                    content = "".join(p.get("text", "") for p in content)
+                messages.append({"role": m.get("role", "user"),
+                                 "content": content or ""})
+        tok = runner.tokenizer
+        if hasattr(tok, "apply_chat_template"):
+            try:
+                prompt = tok.apply_chat_template(messages, tokenize=False,
+                                                 add_generation_prompt=True)
+            except TypeError:
+                prompt = tok.apply_chat_template(messages)
+        else:
+            prompt = "\n".join(f"{m['role']}: {m['content']}" for m in messages)
+        ids = tok.encode(prompt)
+        if hasattr(ids, "ids"):
+            ids = ids.ids
+        inner = dict(body)
+        inner["max_tokens"] = body.get("max_output_tokens") or body.get(
+            "max_tokens") or 256
+        inner.pop("stream", None)  # minimal surface: non-streaming
+        resp = await _generate(request, inner, list(ids), "chat")
+        data = json.loads(bytes(resp.body))
+        msg = data["choices"][0]["message"]
+        return JSONResponse({
+            "id": data["id"].replace("req-", "resp_"),
+            "object": "response",
+            "created_at": data["created"],
+            "model": data["model"],
+            "status": "completed",
+            "output": [{
+                "type": "message", "role": "assistant",
+                "content": [{"type": "output_text",
+                             "text": msg.get("content") or ""}],
+            }],
+            "output_text": msg.get("content") or "",
+            "usage": {
+                "input_tokens": data["usage"]["prompt_tokens"],
+                "output_tokens": data["usage"]["completion_tokens"],
+                "total_tokens": data["usage"]["total_tokens"],
+            },
+        })
+
+    @app.post("/v1/messages/count_tokens")
+    async def count_tokens(request: Request):
+        """Anthropic count_tokens (reference endpoint surface)."""
+        body = await request.json()
+        tok = runner.tokenizer
+        text = "".join(
+            (m.get("content") if isinstance(m.get("content"), str)
+             else "".join(p.get("text", "") for p in m.get("content") or []))
+            for m in body.get("messages") or [])
+        if body.get("system"):
+            text = str(body["system"]) + text
+        ids = tok.encode(text)
+        if hasattr(ids, "ids"):
+            ids = ids.ids
+        return {"input_tokens": len(list(ids))}
+
     return app
 
 

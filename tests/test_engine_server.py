@@ -174,3 +174,52 @@ def test_anthropic_messages_endpoint():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_responses_and_count_tokens():
+    """OpenAI Responses API (minimal) + Anthropic count_tokens."""
+    import httpx
+
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-r", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    base = f"http://127.0.0.1:{port}"
+    try:
+        _wait_health(port, proc)
+        r = httpx.post(f"{base}/v1/responses", json={
+            "model": "tiny-r", "input": "hello there",
+            "max_output_tokens": 6, "temperature": 0, "ignore_eos": True,
+        }, timeout=60)
+        assert r.status_code == 200, r.text
+        d = r.json()
+        assert d["object"] == "response" and d["status"] == "completed"
+        assert d["output"][0]["content"][0]["type"] == "output_text"
+        assert d["output_text"] == d["output"][0]["content"][0]["text"]
+        assert d["usage"]["output_tokens"] == 6
+
+        # message-list input with typed content parts
+        r = httpx.post(f"{base}/v1/responses", json={
+            "model": "tiny-r", "instructions": "be brief",
+            "input": [{"role": "user",
+                       "content": [{"type": "input_text", "text": "hi"}]}],
+            "max_output_tokens": 4, "temperature": 0, "ignore_eos": True,
+        }, timeout=60)
+        assert r.status_code == 200
+        assert r.json()["usage"]["output_tokens"] == 4
+
+        r = httpx.post(f"{base}/v1/messages/count_tokens", json={
+            "model": "tiny-r",
+            "messages": [{"role": "user", "content": "hello"}]}, timeout=30)
+        assert r.status_code == 200
+        assert r.json()["input_tokens"] == 5  # ByteTokenizer: 1 token/byte
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
