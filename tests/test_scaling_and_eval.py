@@ -136,3 +136,34 @@ def test_k8s_manifests_cli(capsys):
     assert main(["manifests", "--namespace", "prod"]) == 0
     out = capsys.readouterr().out
     assert "kind: DaemonSet" in out and "namespace: prod" in out
+
+
+def test_grafana_dashboard_metric_names_exist():
+    """The shipped dashboard only references metric names our exporters
+    actually emit (keeps deploy/grafana honest as metrics evolve)."""
+    import json as _json
+    import re
+    from pathlib import Path
+
+    dash = _json.loads(Path("deploy/grafana/dashboards/gpustack-amd.json")
+                       .read_text())
+    exprs = [t["expr"] for p in dash["panels"] for t in p["targets"]]
+    used = set()
+    for e in exprs:
+        used.update(re.findall(r"gpustack_[a-z_]+", e))
+    emitted = set()
+    for src in ["gpustack_amd/server/exporter.py",
+                "gpustack_amd/worker/agent.py",
+                "gpustack_amd/worker/engine_server.py"]:
+        emitted.update(re.findall(r"gpustack_[a-z_]+", Path(src).read_text()))
+    missing = used - emitted
+    assert not missing, f"dashboard references unknown metrics: {missing}"
+
+
+def test_prometheus_sd_config_points_at_targets_endpoint():
+    import yaml
+    from pathlib import Path
+
+    cfg = yaml.safe_load(Path("deploy/prometheus.yml").read_text())
+    sd = cfg["scrape_configs"][0]["http_sd_configs"][0]
+    assert sd["url"].endswith("/metrics/targets")
