@@ -239,6 +239,7 @@ class LLMEngine:
             and not batch.is_suffix
             and batch.rows_per_seq == 1
             and all(s.params.greedy and not s.params.logprobs
+                    and not s.params.top_logprobs
                     and not s.params.needs_logit_processing
                     for s in batch.seqs)
         )
@@ -275,6 +276,7 @@ class LLMEngine:
             self._refill_tokens(batch)
         token_ids = self.runner.execute(batch)
         lps = getattr(self.runner, "last_logprobs", None)
+        tops = getattr(self.runner, "last_top_logprobs", None)
         if batch.is_prefill or batch.is_suffix:
             self.scheduler.on_prefill_done(batch)
         rps = 1 if (batch.is_prefill or batch.is_suffix) else batch.rows_per_seq
@@ -304,8 +306,11 @@ class LLMEngine:
                 reason = self._finish_reason(seq, tok)
                 lp = (lps[row0 + j] if (lps is not None and seq.params.logprobs)
                       else None)
+                top = (tops[row0 + j][:seq.params.top_logprobs]
+                       if (tops is not None and seq.params.top_logprobs)
+                       else None)
                 outputs.append(StepOutput(seq.request_id, tok,
-                                          reason is not None, reason, lp))
+                                          reason is not None, reason, lp, top))
                 if reason:
                     self.scheduler.finish_seq(seq, reason)
                     self.seqs.pop(seq.request_id, None)

@@ -506,7 +506,8 @@ class ModelRunner:
             return self._pp_finish(batch, logits, row_seqs)
         token_ids = self.sampler.sample(logits, row_seqs)
         self.last_logprobs = None
-        if any(s.params.logprobs for s in batch.seqs):
+        self.last_top_logprobs = None
+        if any(s.params.logprobs or s.params.top_logprobs for s in batch.seqs):
             lf = logits.float()
             lse = torch.logsumexp(lf, dim=-1)
             chosen = lf.gather(
@@ -514,6 +515,14 @@ class ModelRunner:
                                    device=logits.device).unsqueeze(1)
             ).squeeze(1)
             self.last_logprobs = (chosen - lse).tolist()
+            k = max((s.params.top_logprobs for s in batch.seqs), default=0)
+            if k > 0:
+                lsm = lf - lse.unsqueeze(1)
+                vals, idx = torch.topk(lsm, min(k, lf.shape[-1]), dim=-1)
+                self.last_top_logprobs = [
+                    list(zip(idx[r].tolist(), vals[r].tolist()))
+                    for r in range(lf.shape[0])
+                ]
         if self.comm.tp_size > 1:
             # ranks must agree on sampled tokens; rank 0 decides
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)

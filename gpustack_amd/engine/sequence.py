@@ -35,6 +35,7 @@ class SamplingParams:
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None
     logprobs: bool = False
+    top_logprobs: int = 0       # OpenAI top-k alternative logprobs per token
     # dynamic multi-LoRA: name of a live adapter (engine.add_lora) applied
     # to this request's rows only (models/lora.py)
     lora_name: str | None = None
@@ -115,3 +116,4 @@ class StepOutput:
     finished: bool
     finish_reason: str | None = None
     logprob: float | None = None
+    top_logprobs: "list[tuple[int, float]] | None" = None

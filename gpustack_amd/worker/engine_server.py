@@ -141,6 +141,7 @@ class EngineRunner:
                         "finished": out.finished,
                         "finish_reason": out.finish_reason,
                         "logprob": out.logprob,
+                        "top_logprobs": out.top_logprobs,
                     })
 
     def _push(self, q: asyncio.Queue, item: dict) -> None:
@@ -240,7 +241,8 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         max_tokens=int(mt),
         ignore_eos=bool(body.get("ignore_eos", False)),
         seed=body.get("seed"),
-        logprobs=bool(body.get("logprobs")),
+        logprobs=bool(body.get("logprobs")) or int(body.get("top_logprobs") or 0) > 0,
+        top_logprobs=int(body.get("top_logprobs") or 0),
         guided_token_seqs=guided,
         guided_json=gj,
         eos_token_id=eos_token_id,
@@ -424,6 +426,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
 
         tokens: list[int] = []
         lps: list[float | None] = []
+        tops: list[list | None] = []
         finish = "stop"
         try:
             while True:
@@ -432,6 +435,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
                     raise HTTPException(500, item["error"])
                 tokens.append(item["token_id"])
                 lps.append(item.get("logprob"))
+                tops.append(item.get("top_logprobs"))
                 if stop_strs:
                     t = runner.tokenizer.decode(tokens)
                     if any(ss in t for ss in stop_strs):
@@ -453,7 +457,9 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "completion_tokens": len(tokens),
             "total_tokens": len(prompt_ids) + len(tokens),
         }
-        want_lp = bool(body.get("logprobs")) and any(x is not None for x in lps)
+        want_lp = (bool(body.get("logprobs"))
+                   or int(body.get("top_logprobs") or 0) > 0) \
+            and any(x is not None for x in lps)
         if kind == "chat" and (body.get("_tool_name") or body.get("_tool_envelope")):
             return JSONResponse({
                 "id": rid, "object": "chat.completion", "created": created,
@@ -469,8 +475,12 @@ def create_app(runner: EngineRunner) -> FastAPI:
                       "finish_reason": finish}
             if want_lp:
                 choice["logprobs"] = {"content": [
-                    {"token": runner.tokenizer.decode([t]), "logprob": lp}
-                    for t, lp in zip(tokens, lps)
+                    {"token": runner.tokenizer.decode([t]), "logprob": lp,
+                     **({"top_logprobs": [
+                         {"token": runner.tokenizer.decode([tid]),
+                          "logprob": tlp} for tid, tlp in top]}
+                        if top else {})}
+                    for t, lp, top in zip(tokens, lps, tops)
                 ]}
             return JSONResponse({
                 "id": rid, "object": "chat.completion", "created": created,

@@ -81,3 +81,30 @@ def test_guided_choice_single_forces_sequence():
                        eos_token_id=1)
     out = eng.generate([[5]], p)[0]
     assert out[:4] == [9, 8, 7, 6]
+
+
+def test_top_logprobs():
+    """OpenAI top_logprobs: per-token top-k alternatives; the chosen
+    (greedy) token is the top-1 alternative with a matching logprob."""
+    import math
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                 kv_cache_blocks=64))
+    p = SamplingParams(max_tokens=4, ignore_eos=True, logprobs=True,
+                       top_logprobs=3)
+    rid = eng.add_request([1, 2, 3], p)
+    outs = []
+    while eng.has_unfinished():
+        outs.extend(o for o in eng.step() if o.request_id == rid)
+    assert len(outs) == 4
+    for o in outs:
+        assert o.logprob is not None
+        assert o.top_logprobs is not None and len(o.top_logprobs) == 3
+        ids = [t for t, _ in o.top_logprobs]
+        lps = [lp for _, lp in o.top_logprobs]
+        assert ids[0] == o.token_id          # greedy pick is argmax
+        assert math.isclose(lps[0], o.logprob, rel_tol=1e-5, abs_tol=1e-6)
+        assert lps == sorted(lps, reverse=True)
+        assert all(lp <= 0 for lp in lps)
