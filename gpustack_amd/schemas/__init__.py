@@ -142,6 +142,7 @@ class BenchmarkCreate(BaseModel):
     mode: str = "concurrency"      # "concurrency" | "qps"
     value: float = 8
     sweep: list[float] | None = None   # multi-point profile (one run per value)
+    sla: dict | None = None            # SLA threshold overrides (analysis.py)
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64

@@ -11,6 +11,7 @@ import time
 
 import httpx
 
+from ..bench.analysis import analyze_profile
 from ..bench.loadgen import LoadSpec, run_load
 from ..client import ServerClient
 from ..config import Config
@@ -89,6 +90,9 @@ class BenchmarkManager:
             results = dict(max(profile, key=lambda r: r.get("output_tps") or 0))
             if len(profile) > 1:
                 results["profile"] = profile
+            # SLA + saturation analysis (reference: benchmark SLA thresholds
+            # and saturation probe)
+            results["analysis"] = analyze_profile(profile, cfgd.get("sla"))
             self._update(bid, state="completed", results=results)
             logger.info("benchmark %s done: %s out-tok/s, p50 TTFT %s ms",
                         b.get("name"), results.get("output_tps"),
