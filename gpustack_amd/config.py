@@ -36,6 +36,8 @@ class Config(BaseModel):
     oidc_client_secret: str | None = None
     oidc_username_claim: str = "preferred_username"
     oidc_admin_group: str | None = None    # groups claim granting is_admin
+    # CAS external auth (reference: routes/auth.py:1019-1140)
+    cas_server_url: str | None = None      # e.g. https://cas.corp/cas
     bootstrap_password: str | None = None
     jwt_secret: str | None = None
     disable_auth: bool = False

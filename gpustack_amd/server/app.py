@@ -71,6 +71,8 @@ def auth_config():
         "password_login": True,
         "oidc": bool(cfg.oidc_issuer and cfg.oidc_client_id),
         "oidc_login_url": "/auth/oidc/login" if cfg.oidc_issuer else None,
+        "cas": bool(cfg.cas_server_url),
+        "cas_login_url": "/auth/cas/login" if cfg.cas_server_url else None,
     }
 
 
@@ -147,6 +149,62 @@ def oidc_callback(code: str, state: str, response: Response):
     token = jwt_encode({"sub": username}, cfg.get_jwt_secret())
     response.set_cookie(COOKIE_NAME, token, httponly=True, samesite="lax")
     return {"token": token, "username": username, "is_admin": is_admin}
+
+
+@auth_router.get("/cas/login")
+def cas_login(request: Request):
+    """CAS SSO (reference: routes/auth.py:1019-1140): redirect to the CAS
+    login with our callback as the service URL."""
+    from urllib.parse import urlencode
+
+    from fastapi.responses import RedirectResponse
+
+    cfg = deps.get_config()
+    if not cfg.cas_server_url:
+        raise HTTPException(404, "CAS is not configured")
+    service = str(request.url_for("cas_callback"))
+    return RedirectResponse(
+        f"{cfg.cas_server_url.rstrip('/')}/login?{urlencode({'service': service})}")
+
+
+@auth_router.get("/cas/callback")
+def cas_callback(ticket: str, request: Request, response: Response):
+    """Validate the service ticket via CAS /serviceValidate (XML)."""
+    import xml.etree.ElementTree as ET
+    from urllib.parse import urlencode
+
+    import httpx
+
+    cfg = deps.get_config()
+    if not cfg.cas_server_url:
+        raise HTTPException(404, "CAS is not configured")
+    service = str(request.url_for("cas_callback"))
+    r = httpx.get(
+        f"{cfg.cas_server_url.rstrip('/')}/serviceValidate?"
+        + urlencode({"ticket": ticket, "service": service}), timeout=10)
+    if r.status_code != 200:
+        raise HTTPException(401, "CAS validation failed")
+    ns = {"cas": "http://www.yale.edu/tp/cas"}
+    try:
+        root = ET.fromstring(r.text)
+    except ET.ParseError:
+        raise HTTPException(401, "CAS returned invalid XML")
+    ok = root.find("cas:authenticationSuccess", ns)
+    if ok is None:
+        raise HTTPException(401, "CAS rejected the ticket")
+    user_el = ok.find("cas:user", ns)
+    if user_el is None or not (user_el.text or "").strip():
+        raise HTTPException(401, "CAS response has no user")
+    username = user_el.text.strip()
+    with get_session() as s:
+        user = s.query(User).filter_by(username=username).first()
+        if user is None:
+            user = User(username=username, hashed_password=hash_password(
+                __import__("secrets").token_urlsafe(24)))
+            ar_create(s, user)
+    token = jwt_encode({"sub": username}, cfg.get_jwt_secret())
+    response.set_cookie(COOKIE_NAME, token, httponly=True, samesite="lax")
+    return {"token": token, "username": username}
 
 
 @auth_router.get("/me")

@@ -134,3 +134,63 @@ def test_oidc_unconfigured_404():
     assert client.get("/auth/oidc/login",
                       follow_redirects=False).status_code == 404
     assert client.get("/auth/config").json()["oidc"] is False
+
+
+@pytest.fixture()
+def cas():
+    """Minimal CAS server: /login redirect target + /serviceValidate XML."""
+    from fastapi.responses import PlainTextResponse
+
+    stub = FastAPI()
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    base = f"http://127.0.0.1:{port}"
+
+    @stub.get("/cas/serviceValidate")
+    def validate(ticket: str, service: str):
+        if ticket == "ST-good":
+            body = ('<cas:serviceResponse xmlns:cas="http://www.yale.edu/tp/cas">'
+                    "<cas:authenticationSuccess><cas:user>carol</cas:user>"
+                    "</cas:authenticationSuccess></cas:serviceResponse>")
+        else:
+            body = ('<cas:serviceResponse xmlns:cas="http://www.yale.edu/tp/cas">'
+                    '<cas:authenticationFailure code="INVALID_TICKET">bad'
+                    "</cas:authenticationFailure></cas:serviceResponse>")
+        return PlainTextResponse(body, media_type="application/xml")
+
+    srv = uvicorn.Server(uvicorn.Config(stub, host="127.0.0.1", port=port,
+                                        log_level="warning"))
+    threading.Thread(target=srv.run, daemon=True).start()
+    for _ in range(100):
+        try:
+            httpx.get(f"{base}/cas/serviceValidate?ticket=x&service=y",
+                      timeout=1)
+            break
+        except httpx.HTTPError:
+            time.sleep(0.1)
+    yield f"{base}/cas"
+    srv.should_exit = True
+
+
+def test_cas_flow(cas):
+    cfg = Config(data_dir=tempfile.mkdtemp(), bootstrap_password="pw123",
+                 cas_server_url=cas)
+    client = TestClient(create_app(cfg, start_background=False))
+    doc = client.get("/auth/config").json()
+    assert doc["cas"] is True
+    r = client.get("/auth/cas/login", follow_redirects=False)
+    assert r.status_code in (302, 307)
+    assert r.headers["location"].startswith(cas + "/login?service=")
+    r = client.get("/auth/cas/callback", params={"ticket": "ST-good"})
+    assert r.status_code == 200, r.text
+    tok = r.json()["token"]
+    assert r.json()["username"] == "carol"
+    c2 = TestClient(client.app)
+    c2.headers["Authorization"] = f"Bearer {tok}"
+    assert c2.get("/auth/me").json()["username"] == "carol"
+    # bad ticket rejected
+    assert client.get("/auth/cas/callback",
+                      params={"ticket": "ST-bad"}).status_code == 401
