@@ -310,6 +310,30 @@ async def responses(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/responses", user)
 
 
+# frozen legacy mount (reference: routes/openai.py:81-92 re-mounts a legacy
+# subset at /v1-openai for older SDK base_url conventions)
+@router.post("/v1-openai/chat/completions")
+async def legacy_chat(request: Request, user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/chat/completions", user)
+
+
+@router.post("/v1-openai/completions")
+async def legacy_completions(request: Request,
+                             user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/completions", user)
+
+
+@router.post("/v1-openai/embeddings")
+async def legacy_embeddings(request: Request,
+                            user: User = Depends(get_current_user)):
+    return await _proxy(request, "/v1/embeddings", user)
+
+
+@router.get("/v1-openai/models")
+def legacy_models(user: User = Depends(get_current_user)):
+    return list_models_v1(user)
+
+
 @router.post("/v1/messages/count_tokens")
 async def count_tokens(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/messages/count_tokens", user)

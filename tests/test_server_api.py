@@ -403,3 +403,15 @@ def test_multi_cluster_registration_and_placement(server):
     wid = ws["w-edge"]["id"]
     client.delete(f"/v2/workers/{wid}")
     assert client.delete(f"/v2/clusters/{edge['id']}").status_code == 200
+
+
+def test_v1_openai_legacy_mount(server):
+    """Frozen legacy subset at /v1-openai (reference: routes/openai.py:81-92)."""
+    client, app, cfg, reg = server
+    r = client.get("/v1-openai/models")
+    assert r.status_code == 200
+    assert r.json()["object"] == "list"
+    # proxy path resolves models the same way (404 for unknown)
+    r = client.post("/v1-openai/chat/completions", json={
+        "model": "nope", "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 404
