@@ -62,6 +62,13 @@ def _m007_clusters(conn):
     _add_column(conn, "registration_tokens", "cluster_id", "INTEGER")
 
 
+def _m008_orgs(conn):
+    from ..schemas.tables import Org
+    Org.__table__.create(conn, checkfirst=True)
+    _add_column(conn, "users", "org_id", "INTEGER")
+    _add_column(conn, "models", "org_id", "INTEGER")
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -70,6 +77,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (5, "model.lora_adapters for dynamic multi-LoRA", _m005_model_lora_adapters),
     (6, "worker_pools table for auto-provisioned capacity", _m006_worker_pools),
     (7, "multi-cluster: clusters table + cluster_id columns", _m007_clusters),
+    (8, "orgs table + user/model org scoping", _m008_orgs),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

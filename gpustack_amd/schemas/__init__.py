@@ -6,7 +6,7 @@ from pydantic import BaseModel, Field
 from .tables import (  # noqa: F401
     ApiKey, Cluster,
     Benchmark,
-    Model,
+    Model, Org,
     ModelFile,
     ModelInstance,
     ModelInstanceState,
@@ -28,6 +28,7 @@ class UserCreate(BaseModel):
     password: str
     is_admin: bool = False
     full_name: str = ""
+    org_id: int | None = None
 
 
 class UserPublic(BaseModel):
@@ -52,6 +53,7 @@ class ModelCreate(BaseModel):
     source: str = SourceEnum.PRESET.value
     model_ref: str = "llama-3-8b"
     cluster_id: int | None = None  # None = any cluster
+    org_id: int | None = None      # None = visible to all users
     description: str = ""
     replicas: int = 1
     categories: list[str] = Field(default_factory=lambda: ["llm"])
@@ -146,6 +148,11 @@ class BenchmarkCreate(BaseModel):
     duration_s: float = 30.0
     isl: int = 128
     osl: int = 64
+
+
+class OrgCreate(BaseModel):
+    name: str
+    description: str = ""
 
 
 class ClusterCreate(BaseModel):

@@ -63,6 +63,15 @@ class PlacementStrategy(str, enum.Enum):
     BINPACK = "binpack"
 
 
+class Org(Base, TimestampMixin, SerializeMixin):
+    """Tenant boundary (reference: schemas/users.py Org/Group multi-tenant
+    RBAC — platform admin vs org members; models scoped per org)."""
+    __tablename__ = "orgs"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    description = Column(Text, default="")
+
+
 class User(Base, TimestampMixin, SerializeMixin):
     __tablename__ = "users"
     id = Column(Integer, primary_key=True)
@@ -71,6 +80,7 @@ class User(Base, TimestampMixin, SerializeMixin):
     is_admin = Column(Boolean, default=False)
     full_name = Column(String(256), default="")
     require_password_change = Column(Boolean, default=False)
+    org_id = Column(Integer, ForeignKey("orgs.id"), nullable=True, index=True)
 
 
 class ApiKey(Base, TimestampMixin, SerializeMixin):
@@ -137,6 +147,8 @@ class Model(Base, TimestampMixin, SerializeMixin):
     # HUGGING_FACE: repo id (downloaded by the worker model-file manager)
     model_ref = Column(String(512), nullable=False)
     cluster_id = Column(Integer, ForeignKey("clusters.id"), nullable=True)
+    # tenancy: None = visible to every user; set = that org's members only
+    org_id = Column(Integer, ForeignKey("orgs.id"), nullable=True, index=True)
     replicas = Column(Integer, default=1)
     categories = Column(JSON, default=lambda: ["llm"])
     placement_strategy = Column(String(32), default=PlacementStrategy.BINPACK.value)

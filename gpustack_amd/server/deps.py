@@ -110,3 +110,16 @@ def verify_worker_token(request: Request) -> None:
         if s.query(RegistrationToken).filter_by(token=token).first():
             return
     raise HTTPException(401, "invalid worker token")
+
+
+def model_allowed_for_user(user, model) -> bool:
+    """Tenancy gate (reference: services.py:57 UserService
+    .model_allowed_for_user): platform admins see everything; org-scoped
+    models are visible to that org's members only; unscoped models are
+    public to every authenticated user."""
+    if getattr(user, "is_admin", False):
+        return True
+    org = getattr(model, "org_id", None)
+    if org is None and isinstance(model, dict):
+        org = model.get("org_id")
+    return org is None or org == getattr(user, "org_id", None)

@@ -76,10 +76,15 @@ def _find_provider(model_name: str):
     return None
 
 
-def _pick_instance(model_name: str) -> tuple[Model, dict]:
+def _pick_instance(model_name: str, user: User | None = None) -> tuple[Model, dict]:
+    from .deps import model_allowed_for_user
+
     with get_session() as s:
         model = s.query(Model).filter_by(name=model_name).first()
         if not model:
+            raise HTTPException(404, f"model {model_name!r} not found")
+        if user is not None and not model_allowed_for_user(user, model):
+            # 404 (not 403): tenants must not learn other orgs' model names
             raise HTTPException(404, f"model {model_name!r} not found")
         insts = (
             s.query(ModelInstance)
@@ -134,7 +139,7 @@ async def _proxy(request: Request, path: str, user: User):
         local = s.query(Model).filter_by(name=target_name).first() is not None
     tunnel_worker = None
     if local:
-        model, inst = _pick_instance(target_name)
+        model, inst = _pick_instance(target_name, user)
         with get_session() as s:
             w = s.get(Worker, inst["worker_id"]) if inst.get("worker_id") else None
             if w is not None and w.proxy_mode == "tunnel":
@@ -248,8 +253,11 @@ async def _proxy_via_tunnel(tw, path: str, body: dict, user: User,
 
 @router.get("/v1/models")
 def list_models_v1(user: User = Depends(get_current_user)):
+    from .deps import model_allowed_for_user
+
     with get_session() as s:
-        models = s.query(Model).all()
+        models = [m for m in s.query(Model).all()
+                  if model_allowed_for_user(user, m)]
         routes = s.query(ModelRoute).all()
         items = [
             {"id": m.name, "object": "model", "created": int(m.created_at),

@@ -16,7 +16,7 @@ from fastapi.responses import StreamingResponse
 from ..db import Event, EventType, ar_create, ar_delete, ar_update, bus, get_session
 from ..schemas import (
     ApiKey, ApiKeyCreate, Benchmark, BenchmarkCreate, ClusterCreate, Model,
-    ModelCreate,
+    ModelCreate, OrgCreate,
     ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
     ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
     ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
@@ -68,7 +68,8 @@ def create_user(body: UserCreate, _: User = Depends(get_admin_user)):
         if s.query(User).filter_by(username=body.username).first():
             raise HTTPException(409, "username exists")
         u = User(username=body.username, hashed_password=hash_password(body.password),
-                 is_admin=body.is_admin, full_name=body.full_name)
+                 is_admin=body.is_admin, full_name=body.full_name,
+                 org_id=body.org_id)
         ar_create(s, u)
         return u.to_dict() | {"hashed_password": None}
 
@@ -224,8 +225,11 @@ def delete_worker(worker_id: int, _: User = Depends(get_admin_user)):
 
 @router.get("/models")
 def list_models(watch: bool = Query(False), user: User = Depends(get_current_user)):
+    from .deps import model_allowed_for_user
+
     with get_session() as s:
-        rows = [m.to_dict() for m in s.query(Model).all()]
+        rows = [m.to_dict() for m in s.query(Model).all()
+                if model_allowed_for_user(user, m)]
     if watch:
         return _watch_stream("models", rows, None)
     return {"items": rows}
@@ -328,6 +332,40 @@ def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
 
 
 # ---- model routes ----------------------------------------------------------
+
+@router.get("/orgs")
+def list_orgs(_: User = Depends(get_current_user)):
+    from ..schemas import Org
+
+    with get_session() as s:
+        return {"items": [o.to_dict() for o in s.query(Org).all()]}
+
+
+@router.post("/orgs", status_code=201)
+def create_org(body: OrgCreate, _: User = Depends(get_admin_user)):
+    from ..schemas import Org
+
+    with get_session() as s:
+        if s.query(Org).filter_by(name=body.name).first():
+            raise HTTPException(409, "org exists")
+        o = Org(name=body.name, description=body.description)
+        ar_create(s, o)
+        return o.to_dict()
+
+
+@router.delete("/orgs/{org_id}")
+def delete_org(org_id: int, _: User = Depends(get_admin_user)):
+    from ..schemas import Org
+
+    with get_session() as s:
+        o = s.get(Org, org_id)
+        if not o:
+            raise HTTPException(404, "org not found")
+        if s.query(User).filter_by(org_id=org_id).count()                 or s.query(Model).filter_by(org_id=org_id).count():
+            raise HTTPException(409, "org still has users or models")
+        ar_delete(s, o)
+        return {"deleted": org_id}
+
 
 @router.get("/clusters")
 def list_clusters(_: User = Depends(get_current_user)):

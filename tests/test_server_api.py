@@ -415,3 +415,43 @@ def test_v1_openai_legacy_mount(server):
     r = client.post("/v1-openai/chat/completions", json={
         "model": "nope", "messages": [{"role": "user", "content": "hi"}]})
     assert r.status_code == 404
+
+
+def test_org_tenancy(server):
+    """Org-scoped model visibility (reference: UserService
+    .model_allowed_for_user, api/tenant.py org gates)."""
+    client, app, cfg, reg = server
+    org = client.post("/v2/orgs", json={"name": "team-a"}).json()
+    client.post("/v2/models", json={"name": "pub-model", "model_ref": "tiny"})
+    client.post("/v2/models", json={"name": "team-model", "model_ref": "tiny",
+                                    "org_id": org["id"]})
+    client.post("/v2/users", json={"username": "ann", "password": "pw",
+                                   "org_id": org["id"]})
+    client.post("/v2/users", json={"username": "bob", "password": "pw"})
+
+    def as_user(name):
+        c = TestClient(app)
+        r = c.post("/auth/login", json={"username": name, "password": "pw"})
+        c.headers["Authorization"] = f"Bearer {r.json()['token']}"
+        return c
+
+    ann, bob = as_user("ann"), as_user("bob")
+    assert {m["name"] for m in ann.get("/v2/models").json()["items"]} == \
+        {"pub-model", "team-model"}
+    assert {m["name"] for m in bob.get("/v2/models").json()["items"]} == \
+        {"pub-model"}
+    # /v1/models mirrors the same visibility
+    assert "team-model" not in {m["id"] for m in
+                                bob.get("/v1/models").json()["data"]}
+    assert "team-model" in {m["id"] for m in
+                            ann.get("/v1/models").json()["data"]}
+    # inference path: out-of-org model answers 404 (name disclosure safe)
+    r = bob.post("/v1/chat/completions", json={
+        "model": "team-model", "messages": [{"role": "user", "content": "x"}]})
+    assert r.status_code == 404
+    # in-org user reaches placement (503: no running instance yet)
+    r = ann.post("/v1/chat/completions", json={
+        "model": "team-model", "messages": [{"role": "user", "content": "x"}]})
+    assert r.status_code == 503
+    # admin sees everything; org with members cannot be deleted
+    assert client.delete(f"/v2/orgs/{org['id']}").status_code == 409
