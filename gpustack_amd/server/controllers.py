@@ -62,16 +62,22 @@ class ModelController(_LeaderGated):
         self._stop = True
 
     def run(self) -> None:
+        from .workqueue import WorkQueue
+
         q = bus.subscribe("models")
         iq = bus.subscribe("model_instances")
+        # rate-limited queue (reference: server/workqueue.py): coalesces
+        # bursts of events per model and retries failing reconciles with
+        # exponential backoff instead of hot-looping
+        self.wq = WorkQueue(base_delay=2.0, max_delay=300.0)
         self.reconcile_all()
         while not self._stop:
             try:
-                ev = q.get(timeout=10.0)
+                ev = q.get(timeout=0.5)
                 if self._stop:
                     return
                 if ev.type in (EventType.CREATED, EventType.UPDATED) and self._is_leader():
-                    self.sync_replicas(ev.data["id"])
+                    self.wq.add(ev.data["id"])
             except queue.Empty:
                 pass
             # drain instance deletions (e.g. user deleted an instance -> recreate)
@@ -81,9 +87,18 @@ class ModelController(_LeaderGated):
                     if self._stop:
                         return
                     if iev.type == EventType.DELETED and iev.data.get("model_id"):
-                        self.sync_replicas(iev.data["model_id"])
+                        self.wq.add(iev.data["model_id"])
             except queue.Empty:
                 pass
+            item = self.wq.get(timeout=0.01)
+            if item is None:
+                continue
+            try:
+                self.sync_replicas(item)
+                self.wq.done(item)
+            except Exception:  # noqa: BLE001
+                logger.exception("sync_replicas(%s) failed; backing off", item)
+                self.wq.done(item, requeue=True)
 
     def reconcile_all(self) -> None:
         with get_session() as s:
