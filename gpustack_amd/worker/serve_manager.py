@@ -58,9 +58,51 @@ class ServeManager:
         self._restart_at: dict[int, float] = {}
         self._stop = False
 
+    # ---- orphan cleanup (reference: worker/workload_cleaner.py — GC of
+    # containers a previous worker run left behind) -------------------------
+
+    def cleanup_orphans(self) -> None:
+        """Kill engine_server processes a PREVIOUS agent run left behind.
+
+        The DB records each instance's pid; a fresh agent has no entry in
+        self.processes for them, so a still-alive pid would double-serve
+        the GPU and squat the port. Identity is verified via
+        /proc/<pid>/cmdline (exact module + --served-name) before the kill
+        — never pattern-matched."""
+        import os
+        import signal
+
+        try:
+            insts = self.client.list_instances(worker_id=self.worker_id)
+        except Exception:  # noqa: BLE001
+            return
+        for inst in insts:
+            pid = inst.get("pid")
+            if not pid or inst.get("id") in self.processes:
+                continue
+            try:
+                with open(f"/proc/{pid}/cmdline", "rb") as f:
+                    argv = f.read().split(b"\0")
+            except OSError:
+                continue  # not running
+            if b"gpustack_amd.worker.engine_server" not in argv:
+                continue  # pid was recycled by an unrelated process
+            if inst.get("model_name", "").encode() not in argv:
+                continue
+            logger.warning("killing orphan engine process pid=%s (instance %s "
+                           "from a previous agent run)", pid, inst.get("name"))
+            try:
+                os.killpg(os.getpgid(pid), signal.SIGTERM)
+            except (OSError, ProcessLookupError):
+                try:
+                    os.kill(pid, signal.SIGTERM)
+                except (OSError, ProcessLookupError):
+                    continue
+
     # ---- main loops ------------------------------------------------------
 
     def watch_loop(self) -> None:
+        self.cleanup_orphans()
         while not self._stop:
             try:
                 for frame in self.client.watch_instances(self.worker_id):

@@ -108,3 +108,46 @@ def test_cross_worker_tp2():
         for a in agents:
             a.stop()
         server.should_exit = True
+
+
+def test_orphan_cleanup_kills_stale_engine_process(tmp_path):
+    """A fresh agent kills engine_server pids a previous run left behind
+    (identity verified via /proc cmdline — reference WorkloadCleaner)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time as _t
+
+    from gpustack_amd.config import Config
+    from gpustack_amd.worker.serve_manager import ServeManager
+
+    # a real orphan: engine_server for model "tiny-orphan" with no agent
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-orphan", "--source", "preset",
+        "--model-ref", "tiny", "--port", "0", "--device", "cpu",
+        "--kv-cache-blocks", "32", "--max-model-len", "128",
+    ], start_new_session=True)
+    try:
+        class _FakeClient:
+            def list_instances(self, worker_id=None):
+                return [{"id": 1, "pid": proc.pid, "model_name": "tiny-orphan",
+                         "name": "tiny-orphan-0"},
+                        {"id": 2, "pid": 999999, "model_name": "x",
+                         "name": "gone"},  # dead pid: skipped
+                        {"id": 3, "pid": os.getppid(), "model_name": "x",
+                         "name": "unrelated"}]  # live but wrong cmdline: kept
+
+        sm = ServeManager(Config(data_dir=str(tmp_path)), _FakeClient(), 1)
+        sm.cleanup_orphans()
+        for _ in range(100):
+            if proc.poll() is not None:
+                break
+            _t.sleep(0.1)
+        assert proc.poll() is not None  # orphan terminated
+        # the unrelated pid (our parent) is untouched
+        os.kill(os.getppid(), 0)
+    finally:
+        if proc.poll() is None:
+            os.killpg(os.getpgid(proc.pid), signal.SIGKILL)
