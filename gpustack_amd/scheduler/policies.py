@@ -208,7 +208,24 @@ def replica_spread_score(cand: Candidate, model: dict, instances: list[dict]) ->
     return -n
 
 
-def pick_candidate(model: dict, workers: list[dict], instances: list[dict]) -> Candidate | None:
+def file_locality_score(cand: Candidate, model: dict,
+                        model_files: list[dict] | None) -> float:
+    """Prefer workers that already hold the model's files (reference:
+    policies/scorers/model_file_locality_scorer.py) — skips a multi-GB
+    download before STARTING."""
+    if not model_files:
+        return 0.0
+    wid = cand.worker.get("id")
+    for f in model_files:
+        if (f.get("worker_id") == wid
+                and f.get("model_ref") == model.get("model_ref")
+                and f.get("source") == model.get("source")):
+            return 10.0
+    return 0.0
+
+
+def pick_candidate(model: dict, workers: list[dict], instances: list[dict],
+                   model_files: list[dict] | None = None) -> Candidate | None:
     flt = workers
     for f in FILTER_CHAIN:
         flt = f(flt, model)
@@ -218,5 +235,7 @@ def pick_candidate(model: dict, workers: list[dict], instances: list[dict]) -> C
     if not cands:
         return None
     for c in cands:
-        c.score = placement_score(c, model, instances) + replica_spread_score(c, model, instances)
+        c.score = (placement_score(c, model, instances)
+                   + replica_spread_score(c, model, instances)
+                   + file_locality_score(c, model, model_files))
     return max(cands, key=lambda c: c.score)

@@ -94,7 +94,10 @@ class PlacementScheduler:
             others = [
                 i.to_dict() for i in s.query(ModelInstance).all() if i.id != inst.id
             ]
-            cand = pick_candidate(model_d, workers, others)
+            from ..schemas import ModelFile
+
+            files = [f.to_dict() for f in s.query(ModelFile).all()]
+            cand = pick_candidate(model_d, workers, others, files)
             if cand is None:
                 inst.state = ModelInstanceState.ANALYZING.value
                 inst.state_message = "no worker fits the resource claim"

@@ -333,6 +333,37 @@ def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
 
 # ---- model routes ----------------------------------------------------------
 
+@router.get("/model_files")
+def list_model_files(_: User = Depends(get_current_user)):
+    from ..schemas import ModelFile
+
+    with get_session() as s:
+        return {"items": [f.to_dict() for f in s.query(ModelFile).all()]}
+
+
+@router.post("/model_files")
+def upsert_model_file(body: dict, request: Request,
+                      _=Depends(verify_worker_token)):
+    """Worker-reported local model artifacts (reference: ModelFile records,
+    schemas/model_files.py) — feeds the scheduler's locality scorer."""
+    from ..schemas import ModelFile
+
+    with get_session() as s:
+        row = s.query(ModelFile).filter_by(
+            worker_id=body.get("worker_id"), source=body.get("source"),
+            model_ref=body.get("model_ref")).first()
+        if row is None:
+            row = ModelFile(worker_id=body["worker_id"],
+                            source=body.get("source", "huggingface"),
+                            model_ref=body["model_ref"])
+            s.add(row)
+        row.local_path = body.get("local_path", row.local_path)
+        row.size_bytes = body.get("size_bytes", row.size_bytes)
+        row.state = body.get("state", "ready")
+        s.commit()
+        return row.to_dict()
+
+
 @router.get("/orgs")
 def list_orgs(_: User = Depends(get_current_user)):
     from ..schemas import Org

@@ -260,17 +260,44 @@ class ServeManager:
                     inst["name"], role, proc.pid, port, gpus)
 
     def _download_model(self, iid: int, repo: str) -> str | None:
+        """Resolve HF files under an OS file lock (reference: model-file
+        manager download locks, model_file_manager.py:375-427 — two replicas
+        of one model must not download concurrently) and record a ModelFile
+        row so the scheduler's locality scorer can prefer this worker."""
+        import fcntl
+        import hashlib
+
         self._safe_update(iid, state=S.DOWNLOADING.value)
         try:
             from huggingface_hub import snapshot_download
 
             cache = Path(self.cfg.cache_dir or Path(self.cfg.data_dir) / "cache")
-            path = snapshot_download(repo, cache_dir=str(cache))
+            cache.mkdir(parents=True, exist_ok=True)
+            lock_path = cache / f".{hashlib.sha256(repo.encode()).hexdigest()[:16]}.lock"
+            with open(lock_path, "w") as lf:
+                fcntl.flock(lf, fcntl.LOCK_EX)
+                try:
+                    path = snapshot_download(repo, cache_dir=str(cache))
+                finally:
+                    fcntl.flock(lf, fcntl.LOCK_UN)
+            self._record_model_file("huggingface", repo, path)
             return path
         except Exception as e:  # noqa: BLE001
             self._safe_update(iid, state=S.ERROR.value,
                               state_message=f"download failed: {e}")
             return None
+
+    def _record_model_file(self, source: str, ref: str, path: str) -> None:
+        try:
+            size = sum(f.stat().st_size for f in Path(path).rglob("*")
+                       if f.is_file())
+            self.client._c.post("/v2/model_files", json={
+                "worker_id": self.worker_id, "source": source,
+                "model_ref": ref, "local_path": str(path),
+                "size_bytes": size, "state": "ready",
+            }).raise_for_status()
+        except Exception as e:  # noqa: BLE001
+            logger.warning("model-file record failed: %s", e)
 
     def _stop_instance(self, iid: int) -> None:
         ip = self.processes.pop(iid, None)

@@ -117,3 +117,30 @@ def test_replicas_spread_across_gpus():
                      if (c := estimate_vram_claim(_model(), model_spec_for(_model()), 1))
                      else {})
     assert sorted(placed) == list(range(8))
+
+
+def test_file_locality_scorer_prefers_holder():
+    """ModelFileLocalityScorer analog: at equal fit, the worker already
+    holding the checkpoint wins placement."""
+    import sys
+    sys.path.insert(0, "tests")
+    from fixtures.workers.fixtures import mi355x_8g
+
+    from gpustack_amd.scheduler.policies import pick_candidate
+
+    w1, w2 = mi355x_8g(1), mi355x_8g(2)
+    w1["id"], w1["name"] = 1, "w1"
+    w2["id"], w2["name"] = 2, "w2"
+    w1["state"] = w2["state"] = "ready"
+    model = {"id": 9, "name": "m", "source": "huggingface",
+             "model_ref": "org/llama", "gpus_per_replica": 1,
+             "placement_strategy": "spread",
+             "gpu_memory_utilization": 0.9, "max_model_len": 2048,
+             "categories": ["llm"]}
+    files = [{"worker_id": 2, "source": "huggingface",
+              "model_ref": "org/llama", "state": "ready"}]
+    cand = pick_candidate(model, [w1, w2], [], files)
+    assert cand.worker["id"] == 2
+    # without the file record the tie breaks by id order (w1)
+    cand = pick_candidate(model, [w1, w2], [], [])
+    assert cand.worker["id"] == 1
