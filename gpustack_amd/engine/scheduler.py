@@ -323,7 +323,14 @@ class Scheduler:
                 self.waiting.popleft()
                 seq.block_table = self.kv.allocator.allocate(nblocks)
                 seq.num_cached_tokens = budget
-                seq.block_hashes = None
+                if getattr(self.cfg, "enable_prefix_caching", False):
+                    from .kv_cache import block_hashes
+
+                    # registered by on_prefill_done at the FINAL chunk
+                    seq.block_hashes = block_hashes(seq.all_token_ids,
+                                                    self.kv.block_size)
+                else:
+                    seq.block_hashes = None
                 seq.status = SeqStatus.RUNNING
                 self._chunking = seq
                 self._chunk_decode_turn = True

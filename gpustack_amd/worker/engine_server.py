@@ -385,6 +385,30 @@ def create_app(runner: EngineRunner) -> FastAPI:
                             break
                         tokens.append(item["token_id"])
                         text = runner.tokenizer.decode(tokens)
+                        if stop_strs:  # stop strings apply mid-stream too
+                            cuts = [text.find(ss) for ss in stop_strs
+                                    if ss in text]
+                            if cuts:
+                                cut = min(cuts)
+                                runner.abort(rid)
+                                tail = text[sent_len:cut]
+                                if tail and not tool_mode:
+                                    payload = (chat_chunk({"content": tail}, None)
+                                               if kind == "chat"
+                                               else text_chunk(tail, None))
+                                    yield f"data: {json.dumps(payload)}\n\n"
+                                usage = {
+                                    "prompt_tokens": len(prompt_ids),
+                                    "completion_tokens": len(tokens),
+                                    "total_tokens": len(prompt_ids) + len(tokens),
+                                }
+                                payload = (chat_chunk({}, "stop")
+                                           if kind == "chat"
+                                           else text_chunk("", "stop"))
+                                payload["usage"] = usage
+                                yield f"data: {json.dumps(payload)}\n\n"
+                                yield "data: [DONE]\n\n"
+                                break
                         new = text[sent_len:]
                         # hold back partial unicode replacement chars
                         if new and not new.endswith("�") and not tool_mode:

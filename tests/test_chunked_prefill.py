@@ -93,3 +93,22 @@ def test_chunked_with_ngram_spec_ok():
     out = eng.generate([LONG], p)[0]
     plain = LLMEngine(_cfg()).generate([LONG], p)[0]
     assert out == plain
+
+
+def test_chunked_with_prefix_caching():
+    """Chunked prefill + automatic prefix caching coexist: cache-hit
+    prompts still take the suffix path; cache-miss long prompts chunk."""
+    p = SamplingParams(max_tokens=6, ignore_eos=True)
+    plain = LLMEngine(_cfg())
+    want_long = plain.generate([LONG], p)[0]
+    short = [5, 6, 7] * 20  # 60 tokens, shares no prefix with LONG
+    want_short = plain.generate([short], p)[0]
+
+    eng = LLMEngine(_cfg(enable_chunked_prefill=True,
+                         enable_prefix_caching=True))
+    assert eng.generate([LONG], p)[0] == want_long
+    assert eng.generate([short], p)[0] == want_short
+    # re-running the long prompt hits the prefix cache (suffix path)
+    hits0 = eng.scheduler.kv.allocator.hits
+    assert eng.generate([LONG], p)[0] == want_long
+    assert eng.scheduler.kv.allocator.hits > hits0

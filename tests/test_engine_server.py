@@ -223,3 +223,49 @@ def test_responses_and_count_tokens():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_streaming_stop_strings():
+    """Stop strings truncate STREAMED output too (non-stream path already
+    covered by test_engine_server_stop_strings)."""
+    import json as _json
+
+    import httpx
+
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-ss", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    base = f"http://127.0.0.1:{port}"
+    try:
+        _wait_health(port, proc)
+        r = httpx.post(f"{base}/v1/completions", json={
+            "model": "tiny-ss", "prompt": "hello", "max_tokens": 20,
+            "ignore_eos": True, "temperature": 0}, timeout=60)
+        full = r.json()["choices"][0]["text"]
+        if len(full) <= 4:
+            pytest.skip("output too short to carve a stop string")
+        stop = full[2:4]
+        chunks = []
+        with httpx.stream("POST", f"{base}/v1/completions", json={
+            "model": "tiny-ss", "prompt": "hello", "max_tokens": 20,
+            "ignore_eos": True, "temperature": 0, "stream": True,
+            "stop": [stop],
+        }, timeout=60) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:") and "[DONE]" not in line:
+                    chunks.append(_json.loads(line[5:]))
+        text = "".join(c["choices"][0].get("text", "") for c in chunks)
+        assert stop not in text
+        assert text == full[:2]
+        assert chunks[-1]["choices"][0]["finish_reason"] == "stop"
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()

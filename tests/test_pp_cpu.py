@@ -192,3 +192,63 @@ def test_engine_server_pp2_cpu():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(300)
+def test_pp2_chunked_prefill_matches():
+    """Chunked prefill under PP: scheduler state is replicated, chunk and
+    decode steps alternate identically on both stages."""
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    long_prompt = [(11 * t + 3) % 500 for t in range(100)]
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, max_prefill_tokens=24,
+                       enable_chunked_prefill=True)
+    # prompt is clamped to max_model_len-1 internally; compare like for like
+    plain = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                   kv_cache_blocks=64, max_model_len=128,
+                                   seed=0)).generate(
+        [long_prompt], SamplingParams(max_tokens=6, ignore_eos=True))
+
+    port = _free_port()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pp_chunked_rank_main,
+                         args=(r, port, out_path, long_prompt))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    with open(out_path) as f:
+        assert json.load(f) == plain
+
+
+def _pp_chunked_rank_main(rank: int, port: int, out_path: str, prompt):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_parallel
+
+    comm = init_parallel(1, 2, rank, master_port=port, backend="gloo")
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, max_prefill_tokens=24,
+                       enable_chunked_prefill=True)
+    eng = LLMEngine(cfg, comm)
+    results, rids = {}, []
+    if rank == 0:
+        rids = [eng.add_request(prompt, SamplingParams(max_tokens=6,
+                                                       ignore_eos=True))]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
