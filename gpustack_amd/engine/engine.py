@@ -338,7 +338,9 @@ class LLMEngine:
 
     def _finish_reason(self, seq: Sequence, tok: int) -> str | None:
         p = seq.params
-        if not p.ignore_eos:
+        # min_tokens suppresses EOS/stop finishes (vLLM semantics);
+        # length finishes still apply
+        if not p.ignore_eos and len(seq.output_token_ids) > p.min_tokens:
             if tok == self.cfg.spec.eos_token_id or tok in p.stop_token_ids:
                 return "stop"
         if len(seq.output_token_ids) >= p.max_tokens:

@@ -31,6 +31,7 @@ class SamplingParams:
     guided_json: "bool | dict | None" = None
     eos_token_id: int = 0           # used by guided decoding to terminate
     max_tokens: int = 128
+    min_tokens: int = 0        # suppress EOS/stop until this many tokens
     ignore_eos: bool = False
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None

@@ -108,3 +108,20 @@ def test_top_logprobs():
         assert math.isclose(lps[0], o.logprob, rel_tol=1e-5, abs_tol=1e-6)
         assert lps == sorted(lps, reverse=True)
         assert all(lp <= 0 for lp in lps)
+
+
+def test_min_tokens_suppresses_eos():
+    """min_tokens (vLLM semantics): EOS/stop finishes are ignored until the
+    sequence has produced that many tokens."""
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                 kv_cache_blocks=64))
+    # force EOS every step via logit_bias toward the eos id
+    eos = eng.cfg.spec.eos_token_id
+    p = SamplingParams(max_tokens=10, logit_bias={eos: 100.0})
+    out = eng.generate([[2, 3, 4]], p)[0]
+    assert len(out) == 1 and out[0] == eos  # stops immediately without min
+    p = SamplingParams(max_tokens=10, min_tokens=5, logit_bias={eos: 100.0})
+    out = eng.generate([[2, 3, 4]], p)[0]
+    assert len(out) == 6  # 5 suppressed EOS emissions + the terminating one
