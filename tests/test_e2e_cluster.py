@@ -284,3 +284,55 @@ def test_crash_restart(cluster):
         assert r.status_code == 200
     finally:
         sm_mod.RESTART_BASE = old_base
+
+
+@pytest.mark.timeout(240)
+def test_deploy_with_lora_adapters(cluster, tmp_path_factory):
+    """Full dynamic-LoRA path: Model.lora_adapters -> serve manager backend
+    params -> engine server mounts the adapter -> gateway resolves the
+    adapter NAME as a servable model and routes it to the parent."""
+    import sys
+    sys.path.insert(0, "tests")
+    from test_lora_dynamic import _make_adapter
+
+    from gpustack_amd.engine import EngineConfig
+
+    client, agent = cluster
+    tmp = tmp_path_factory.mktemp("adapter")
+    _make_adapter(tmp, EngineConfig(model="tiny").spec)
+
+    r = client.post("/v2/models", json={
+        "name": "tiny-lora-host", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+        "lora_adapters": [{"name": "tuned-x", "path": str(tmp)}],
+    })
+    assert r.status_code == 201, r.text
+    state = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-lora-host"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+
+    # adapter name is listed and routable through the gateway
+    ids = {m["id"] for m in client.get("/v1/models").json()["data"]}
+    assert "tuned-x" in ids
+    body = {"prompt": "hello", "max_tokens": 8, "ignore_eos": True,
+            "temperature": 0}
+    base_text = client.post("/v1/completions", json={
+        **body, "model": "tiny-lora-host"}).json()["choices"][0]["text"]
+    r = client.post("/v1/completions", json={**body, "model": "tuned-x"})
+    assert r.status_code == 200, r.text
+    lora_text = r.json()["choices"][0]["text"]
+    assert lora_text != base_text  # adapter rows actually applied
+    # parent stays clean
+    again = client.post("/v1/completions", json={
+        **body, "model": "tiny-lora-host"}).json()["choices"][0]["text"]
+    assert again == base_text
+    client.delete("/v2/models/" + str(client.get("/v2/models").json()[
+        "items"][-1]["id"]))
