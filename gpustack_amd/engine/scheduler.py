@@ -88,7 +88,30 @@ class Scheduler:
 
     # -- queue ops ---------------------------------------------------------
     def add(self, seq: Sequence) -> None:
+        self._insert_waiting(seq, front=False)
+
+    def _insert_waiting(self, seq: Sequence, front: bool) -> None:
+        """Priority-ordered admission (vLLM priority scheduling analog):
+        higher priority enters ahead of lower; within a priority class,
+        new arrivals go last and preempted resumers (front=True) first."""
+        pr = seq.params.priority
+        if pr == 0 and not front and (not self.waiting
+                                      or self.waiting[-1].params.priority >= 0):
+            self.waiting.append(seq)  # common case: plain FIFO
+            return
+        for i, s in enumerate(self.waiting):
+            p = s.params.priority
+            if (p < pr) or (front and p <= pr):
+                self.waiting.insert(i, seq)
+                return
         self.waiting.append(seq)
+
+    def _pop_victim(self) -> tuple[int, Sequence]:
+        """Preemption victim: lowest priority loses; ties break to the
+        newest (latest-admitted) sequence."""
+        idx = min(range(len(self.running)),
+                  key=lambda j: (self.running[j].params.priority, -j))
+        return idx, self.running.pop(idx)
 
     def abort(self, request_id: str) -> bool:
         for i, s in enumerate(self.running):
@@ -179,13 +202,15 @@ class Scheduler:
                 try:
                     seq.block_table.extend(self.kv.allocator.allocate(1))
                 except RuntimeError:
-                    victim = self.running.pop()
+                    vidx, victim = self._pop_victim()
                     self._release(victim)
                     victim.status = SeqStatus.WAITING
                     victim.preemptions += 1
-                    self.waiting.appendleft(victim)
+                    self._insert_waiting(victim, front=True)
                     if victim is seq:
                         continue
+                    if vidx < i:
+                        i -= 1
                     i = min(i, len(self.running))
                     continue
             i += 1
@@ -416,7 +441,7 @@ class Scheduler:
                 try:
                     seq.block_table.extend(self.kv.allocator.allocate(need))
                 except RuntimeError:
-                    victim = self.running.pop()  # preempt newest
+                    vidx, victim = self._pop_victim()
                     if self.kv.can_swap_out(len(victim.block_table)):
                         # offload tier: swap KV to pinned host DRAM instead
                         # of recompute
@@ -429,9 +454,11 @@ class Scheduler:
                         self._release(victim)
                         victim.status = SeqStatus.WAITING
                         victim.preemptions += 1
-                        self.waiting.appendleft(victim)
+                        self._insert_waiting(victim, front=True)
                     if victim is seq:
                         continue
+                    if vidx < i:
+                        i -= 1
                     i = min(i, len(self.running))
                     continue
             i += 1

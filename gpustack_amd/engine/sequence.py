@@ -36,6 +36,7 @@ class SamplingParams:
     stop_token_ids: tuple[int, ...] = ()
     seed: int | None = None
     logprobs: bool = False
+    priority: int = 0           # higher = admitted sooner, preempted last
     top_logprobs: int = 0       # OpenAI top-k alternative logprobs per token
     # dynamic multi-LoRA: name of a live adapter (engine.add_lora) applied
     # to this request's rows only (models/lora.py)

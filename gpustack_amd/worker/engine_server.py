@@ -240,6 +240,7 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         logit_bias=body.get("logit_bias"),
         max_tokens=int(mt),
         min_tokens=int(body.get("min_tokens") or 0),
+        priority=int(body.get("priority") or 0),
         ignore_eos=bool(body.get("ignore_eos", False)),
         seed=body.get("seed"),
         logprobs=bool(body.get("logprobs")) or int(body.get("top_logprobs") or 0) > 0,
