@@ -275,6 +275,18 @@ def create_app(runner: EngineRunner) -> FastAPI:
     async def _startup():
         runner.start(asyncio.get_running_loop())
 
+    @app.on_event("shutdown")
+    async def _shutdown():
+        # graceful drain: stop the engine loop and fail pending requests
+        # cleanly instead of cutting streams mid-token (uvicorn delivers
+        # this on SIGTERM from the serve manager)
+        runner.stop()
+        with runner._lock:
+            queues = list(runner._queues.values())
+        for q in queues:
+            runner._push(q, {"error": "server shutting down",
+                             "finished": True})
+
     @app.get("/health")
     async def health():
         return {"status": "ok", "model": runner.served_name}
