@@ -348,10 +348,11 @@ class ArrayM:
             return self.advance(ch)
         if self.state == 2:
             if self.item.advance(ch):
-                if self.item.complete and not isinstance(self.item, NumberM):
-                    self.state = 3
                 return True
-            if self.item.complete:  # numbers end on the delimiter
+            # item can take no more; if it is complete the char must be the
+            # delimiter — uniform handling so ambiguous items (numbers,
+            # prefix-overlapping enums like "a"/"ab") can keep extending
+            if self.item.complete:
                 self.state = 3
                 return self.advance(ch)
             return False
@@ -363,7 +364,6 @@ class ArrayM:
             if ch == "]":
                 self.state = 4
                 return True
-            # item machine may accept more (e.g. string already closed: no)
             return False
         return False
 

@@ -315,3 +315,15 @@ def test_tool_calling():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+def test_array_of_overlapping_enums():
+    """Ambiguous enum items ("a" vs "ab") keep extending inside arrays."""
+    schema = {"type": "object",
+              "properties": {"xs": {"type": "array",
+                                    "items": {"enum": ["a", "ab"]}}}}
+    assert _accepts(compile_schema(schema), '{"xs":["ab","a","ab"]}')
+    assert _accepts(compile_schema(schema), '{"xs":[]}')
+    assert not _accepts(compile_schema(schema), '{"xs":["b"]}')
+    m = compile_schema(schema)
+    assert _accepts(m, '{"xs":["a"]}') and m.complete
