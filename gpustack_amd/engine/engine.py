@@ -57,7 +57,8 @@ class LLMEngine:
     ) -> str:
         rid = request_id or f"req-{next(self._counter)}"
         params = params or SamplingParams()
-        if params.guided_json is not None and self.comm.pp_size > 1:
+        if (params.guided_json is not None
+                or params.guided_regex is not None) and self.comm.pp_size > 1:
             raise ValueError("guided_json is not supported with pipeline "
                              "parallelism (sampling runs on the last stage, "
                              "which has no token table)")

@@ -456,3 +456,251 @@ class GuidedJsonState:
 
     def commit(self, machine) -> None:
         self.machine = machine
+
+
+# ---- guided_regex: regex subset -> NFA with incremental prefix matching ----
+# (reference surface: vLLM guided_regex). Supported: literals, escapes
+# (\d \w \s \D \W \S and escaped metachars), '.', character classes
+# [a-z0-9_] incl. negation, groups, alternation, and the quantifiers
+# * + ? {m} {m,} {m,n}. Thompson construction; the machine keeps the
+# current state set, so advance() is O(states) per character and
+# `complete` means "the text so far fully matches".
+
+class _RxAst:
+    def __init__(self, kind, **kw):
+        self.kind = kind
+        self.__dict__.update(kw)
+
+
+def _rx_parse(pattern: str) -> _RxAst:
+    pos = 0
+
+    def peek():
+        return pattern[pos] if pos < len(pattern) else None
+
+    def take():
+        nonlocal pos
+        ch = pattern[pos]
+        pos += 1
+        return ch
+
+    def parse_alt():
+        branches = [parse_concat()]
+        while peek() == "|":
+            take()
+            branches.append(parse_concat())
+        return branches[0] if len(branches) == 1 else _RxAst("alt", parts=branches)
+
+    def parse_concat():
+        parts = []
+        while peek() is not None and peek() not in "|)":
+            parts.append(parse_rep())
+        if not parts:
+            return _RxAst("empty")
+        return parts[0] if len(parts) == 1 else _RxAst("cat", parts=parts)
+
+    def parse_rep():
+        atom = parse_atom()
+        while True:
+            c = peek()
+            if c == "*":
+                take()
+                atom = _RxAst("star", inner=atom)
+            elif c == "+":
+                take()
+                atom = _RxAst("cat", parts=[atom, _RxAst("star", inner=atom)])
+            elif c == "?":
+                take()
+                atom = _RxAst("alt", parts=[atom, _RxAst("empty")])
+            elif c == "{":
+                take()
+                spec = ""
+                while peek() is not None and peek() != "}":
+                    spec += take()
+                if peek() != "}":
+                    raise ValueError("unterminated {quantifier}")
+                take()
+                if "," in spec:
+                    lo_s, hi_s = spec.split(",", 1)
+                    lo = int(lo_s or 0)
+                    hi = int(hi_s) if hi_s else None
+                else:
+                    lo = hi = int(spec)
+                parts = [atom] * lo
+                if hi is None:
+                    parts.append(_RxAst("star", inner=atom))
+                else:
+                    parts.extend(_RxAst("alt", parts=[atom, _RxAst("empty")])
+                                 for _ in range(hi - lo))
+                atom = (_RxAst("empty") if not parts
+                        else parts[0] if len(parts) == 1
+                        else _RxAst("cat", parts=parts))
+            else:
+                return atom
+
+    _CLASSES = {
+        "d": lambda c: c.isdigit(),
+        "D": lambda c: not c.isdigit(),
+        "w": lambda c: c.isalnum() or c == "_",
+        "W": lambda c: not (c.isalnum() or c == "_"),
+        "s": lambda c: c in " \t\n\r\f\v",
+        "S": lambda c: c not in " \t\n\r\f\v",
+    }
+    _ESCAPED = {"n": "\n", "t": "\t", "r": "\r"}
+
+    def escape_pred():
+        e = take()
+        if e in _CLASSES:
+            return _RxAst("pred", fn=_CLASSES[e], label=f"\\\\{e}")
+        lit = _ESCAPED.get(e, e)
+        return _RxAst("pred", fn=(lambda c, L=lit: c == L), label=lit)
+
+    def parse_atom():
+        c = take()
+        if c == "(":
+            inner = parse_alt()
+            if peek() != ")":
+                raise ValueError("unbalanced group")
+            take()
+            return inner
+        if c == ".":
+            return _RxAst("pred", fn=lambda ch: ch != "\n", label=".")
+        if c == "[":
+            neg = peek() == "^"
+            if neg:
+                take()
+            items = []  # (lo, hi) ranges or predicate fns
+            while peek() is not None and peek() != "]":
+                a = take()
+                if a == "\\\\" or a == "\\":
+                    e = take()
+                    if e in _CLASSES:
+                        items.append(_CLASSES[e])
+                        continue
+                    a = _ESCAPED.get(e, e)
+                if peek() == "-" and pos + 1 < len(pattern) \
+                        and pattern[pos + 1] != "]":
+                    take()
+                    b = take()
+                    items.append((a, b))
+                else:
+                    items.append((a, a))
+            if peek() != "]":
+                raise ValueError("unterminated character class")
+            take()
+
+            def in_class(ch, items=items, neg=neg):
+                hit = any(it(ch) if callable(it) else it[0] <= ch <= it[1]
+                          for it in items)
+                return hit != neg
+
+            return _RxAst("pred", fn=in_class, label="[class]")
+        if c in ("\\", "\\\\"):
+            return escape_pred()
+        if c in "*+?{}|)":
+            raise ValueError(f"unexpected {c!r}")
+        return _RxAst("pred", fn=(lambda ch, L=c: ch == L), label=c)
+
+    ast = parse_alt()
+    if pos != len(pattern):
+        raise ValueError(f"trailing regex input at {pos}")
+    return ast
+
+
+class RegexM:
+    """NFA machine over a regex subset (advance/complete protocol)."""
+
+    def __init__(self, pattern: str):
+        # compile: states are ints; trans[i] = [(pred, j)]; eps[i] = [j]
+        self.trans: list[list] = []
+        self.eps: list[list[int]] = []
+
+        def new_state():
+            self.trans.append([])
+            self.eps.append([])
+            return len(self.trans) - 1
+
+        def build(ast, start) -> int:
+            """Wire ast from `start`; returns its accepting state."""
+            if ast.kind == "empty":
+                return start
+            if ast.kind == "pred":
+                end = new_state()
+                self.trans[start].append((ast.fn, end))
+                return end
+            if ast.kind == "cat":
+                cur = start
+                for p in ast.parts:
+                    cur = build(p, cur)
+                return cur
+            if ast.kind == "alt":
+                end = new_state()
+                for p in ast.parts:
+                    s = new_state()
+                    self.eps[start].append(s)
+                    self.eps[build(p, s)].append(end)
+                return end
+            if ast.kind == "star":
+                hub = new_state()
+                self.eps[start].append(hub)
+                s = new_state()
+                self.eps[hub].append(s)
+                self.eps[build(ast.inner, s)].append(hub)
+                return hub
+            raise AssertionError(ast.kind)
+
+        s0 = new_state()
+        self.accept = build(_rx_parse(pattern), s0)
+        self.cur = self._closure({s0})
+
+    def _closure(self, states: set) -> frozenset:
+        stack, seen = list(states), set(states)
+        while stack:
+            for j in self.eps[stack.pop()]:
+                if j not in seen:
+                    seen.add(j)
+                    stack.append(j)
+        return frozenset(seen)
+
+    @property
+    def complete(self) -> bool:
+        return self.accept in self.cur
+
+    def advance(self, ch: str) -> bool:
+        nxt = {j for i in self.cur for fn, j in self.trans[i] if fn(ch)}
+        if not nxt:
+            return False
+        self.cur = self._closure(nxt)
+        return True
+
+
+def _regexm_clone(self) -> "RegexM":
+    m = RegexM.__new__(RegexM)
+    m.trans, m.eps, m.accept = self.trans, self.eps, self.accept  # static
+    m.cur = self.cur
+    return m
+
+
+RegexM.clone = _regexm_clone
+
+
+class GuidedRegexState:
+    """Per-sequence guided_regex state (same probe/commit API as
+    GuidedJsonState)."""
+
+    def __init__(self, pattern: str):
+        self.machine = RegexM(pattern)
+
+    @property
+    def complete(self) -> bool:
+        return self.machine.complete
+
+    def try_advance(self, text: str):
+        m = self.machine.clone()
+        for ch in text:
+            if not m.advance(ch):
+                return None
+        return m
+
+    def commit(self, machine) -> None:
+        self.machine = machine

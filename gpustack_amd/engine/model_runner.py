@@ -33,8 +33,13 @@ class Sampler:
         if table is None:
             return int(row.argmax())
         if seq.guided_sm is None:
-            schema = p.guided_json if isinstance(p.guided_json, dict) else None
-            seq.guided_sm = GuidedJsonState(schema)
+            if p.guided_regex is not None:
+                from .guided import GuidedRegexState
+
+                seq.guided_sm = GuidedRegexState(p.guided_regex)
+            else:
+                schema = p.guided_json if isinstance(p.guided_json, dict) else None
+                seq.guided_sm = GuidedJsonState(schema)
             seq.guided_consumed = 0
         sm = seq.guided_sm
         out = seq.output_token_ids
@@ -119,7 +124,8 @@ class Sampler:
             return ops.greedy_sample(logits).tolist()
         out: list[int] = [0] * len(seqs)
         guided = {i for i, s in enumerate(seqs)
-                  if s.params.guided_json is not None}
+                  if s.params.guided_json is not None
+                  or s.params.guided_regex is not None}
         greedy_idx = [i for i, s in enumerate(seqs)
                       if s.params.greedy and not s.params.needs_logit_processing]
         proc_greedy = [i for i, s in enumerate(seqs)

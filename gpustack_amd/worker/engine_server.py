@@ -247,9 +247,10 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         top_logprobs=int(body.get("top_logprobs") or 0),
         guided_token_seqs=guided,
         guided_json=gj,
+        guided_regex=body.get("guided_regex"),
         eos_token_id=eos_token_id,
     )
-    if gj is not None:
+    if gj is not None or sp.guided_regex is not None:
         sp.ignore_eos = False
     if guided:
         sp.ignore_eos = False
@@ -358,7 +359,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                   runner.tokenizer)
         _apply_lora_routing(params, body, runner)
-        if params.guided_json is not None:
+        if params.guided_json is not None or params.guided_regex is not None:
             runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
@@ -594,7 +595,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
         base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                 runner.tokenizer)
         _apply_lora_routing(base, body, runner)
-        if base.guided_json is not None:
+        if base.guided_json is not None or base.guided_regex is not None:
             runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         subs = []

@@ -327,3 +327,49 @@ def test_array_of_overlapping_enums():
     assert not _accepts(compile_schema(schema), '{"xs":["b"]}')
     m = compile_schema(schema)
     assert _accepts(m, '{"xs":["a"]}') and m.complete
+
+
+import pytest as _pytest
+
+
+@_pytest.mark.parametrize("pattern,good,bad", [
+    (r"[a-c]+x", "abcx", "abd"),
+    (r"(ab|cd){2}", "abcd", "abx"),
+    (r"\d{3}-\d{4}", "555-1234", "55-1234x"),
+    (r"yes|no|maybe", "maybe", "nope?"),
+    (r"a?b+c", "bbc", "ac?"),
+    (r"[^0-9]+", "hello", "h3"),
+    (r"\w+@\w+\.(com|org)", "a_1@b.org", "a@b.net?"),
+])
+def test_regex_machine(pattern, good, bad):
+    from gpustack_amd.engine.guided import RegexM
+
+    def feed(s):
+        m = RegexM(pattern)
+        for ch in s:
+            if not m.advance(ch):
+                return None
+        return m
+
+    m = feed(good)
+    assert m is not None and m.complete
+    assert feed(bad) is None or not feed(bad).complete
+
+
+def test_regex_prefix_validity():
+    from gpustack_amd.engine.guided import RegexM
+
+    m = RegexM(r"\d{3}-\d{4}")
+    for ch in "555-":
+        assert m.advance(ch)
+    assert not m.complete      # valid prefix, not a full match
+    assert not m.advance("x")  # invalid continuation rejected
+
+
+def test_engine_guided_regex():
+    eng, tok = _engine_with_table()
+    p = SamplingParams(max_tokens=20, guided_regex=r"(red|green|blue)!",
+                       eos_token_id=1)
+    out = eng.generate([[25, 26]], p)[0]
+    text = tok.decode([t for t in out if t != 1])
+    assert text in ("red!", "green!", "blue!")
