@@ -252,3 +252,54 @@ def _pp_chunked_rank_main(rank: int, port: int, out_path: str, prompt):
 
     dist.barrier()
     dist.destroy_process_group()
+
+
+def _tpxpp_rank_main(rank: int, port: int, out_path: str):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_parallel
+
+    comm = init_parallel(2, 2, rank, master_port=port, backend="gloo")
+    assert comm.world_size == 4 and comm.world_rank == rank
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, tp_size=2,
+                       tp_rank=comm.tp_rank)
+    eng = LLMEngine(cfg, comm)
+    assert eng.runner.model.num_local_layers == 1  # 2 layers over 2 stages
+    results, rids = {}, []
+    if rank == 0:
+        rids = [eng.add_request(p, SamplingParams(max_tokens=6,
+                                                  ignore_eos=True))
+                for p in PROMPTS]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2xpp2_matches_single_rank():
+    """Combined TPxPP (world 4, tp-contiguous layout): sharded weights per
+    stage + residual handoff reproduce the single-rank outputs exactly."""
+    plain = _single_proc_result()
+    port = _free_port()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_tpxpp_rank_main, args=(r, port, out_path))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0, f"rank exited {p.exitcode}"
+    with open(out_path) as f:
+        assert json.load(f) == plain
