@@ -103,6 +103,21 @@ PRESETS: dict[str, ModelSpec] = {
         hidden_size=8192, intermediate_size=28672, num_layers=80,
         num_heads=64, num_kv_heads=8,
     ),
+    # Llama-3.1: same graph as 3.0 plus llama3 rope scaling and 128k context
+    "llama-3.1-8b": ModelSpec(
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=131072,
+    ),
+    "llama-3.1-70b": ModelSpec(
+        hidden_size=8192, intermediate_size=28672, num_layers=80,
+        num_heads=64, num_kv_heads=8,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=131072,
+    ),
     "qwen3-32b": ModelSpec(
         architecture="Qwen3ForCausalLM", vocab_size=151936, hidden_size=5120,
         intermediate_size=25600, num_layers=64, num_heads=64, num_kv_heads=8,

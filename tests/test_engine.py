@@ -185,3 +185,31 @@ def test_mistral_family():
         "num_key_value_heads": 8, "rope_theta": 1000000.0,
     })
     assert hf.num_kv_heads == 8 and not hf.attention_bias
+
+
+def test_llama31_rope_scaling_preset():
+    """Llama-3.1 preset: llama3 rope scaling changes long-range frequencies
+    but leaves short-wavelength (high-frequency) components untouched."""
+    import torch
+
+    from gpustack_amd import ops
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    cfg = EngineConfig(model="llama-3.1-8b")
+    assert cfg.spec.rope_scaling["rope_type"] == "llama3"
+    assert cfg.max_model_len == 8192  # default cap, spec allows 131072
+    plain = ops.build_cos_sin_cache(128, 128, 256, base=500000.0)
+    scaled = ops.build_cos_sin_cache(128, 128, 256, base=500000.0,
+                                     scaling=cfg.spec.rope_scaling)
+    assert not torch.equal(plain, scaled)          # low-freq bands rescaled
+    assert torch.allclose(plain[:, :8], scaled[:, :8])  # high-freq preserved
+
+    # the preset serves end to end (2 layers for speed)
+    import dataclasses
+
+    small = EngineConfig(model="llama-3.1-8b", device="cpu",
+                         kv_cache_blocks=32, max_model_len=128)
+    small.spec = dataclasses.replace(small.spec, num_layers=2)
+    out = LLMEngine(small).generate(
+        [[1, 2, 3]], SamplingParams(max_tokens=4, ignore_eos=True))[0]
+    assert len(out) == 4
