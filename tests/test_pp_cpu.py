@@ -303,3 +303,69 @@ def test_tp2xpp2_matches_single_rank():
         assert p.exitcode == 0, f"rank exited {p.exitcode}"
     with open(out_path) as f:
         assert json.load(f) == plain
+
+
+def _pp_lora_rank_main(rank: int, port: int, out_path: str, adapter: str):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_parallel
+
+    comm = init_parallel(1, 2, rank, master_port=port, backend="gloo")
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0)
+    eng = LLMEngine(cfg, comm)
+    results, rids = {}, []
+    if rank == 0:
+        eng.add_lora("t", adapter)  # replicated to the other stage via ops
+        rids = [eng.add_request(PROMPTS[0],
+                                SamplingParams(max_tokens=6, ignore_eos=True,
+                                               lora_name="t"))]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pp2_dynamic_lora_matches_single():
+    """Dynamic LoRA under PP: the adapter op replicates to every stage and
+    each stage applies only its local layers' deltas."""
+    import sys
+    sys.path.insert(0, "tests")
+    from test_lora_dynamic import _make_adapter
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    tmp = tempfile.mkdtemp()
+    from pathlib import Path
+
+    _make_adapter(Path(tmp), EngineConfig(model="tiny").spec)
+    single = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                    kv_cache_blocks=64, max_model_len=128,
+                                    seed=0))
+    single.add_lora("t", tmp)
+    want = single.generate([PROMPTS[0]],
+                           SamplingParams(max_tokens=6, ignore_eos=True,
+                                          lora_name="t"))
+
+    port = _free_port()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pp_lora_rank_main,
+                         args=(r, port, out_path, tmp)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    with open(out_path) as f:
+        assert json.load(f) == want
