@@ -387,6 +387,7 @@ def create_app(runner: EngineRunner) -> FastAPI:
             async def gen():
                 tokens: list[int] = []
                 sent_len = 0
+                pending_lps: list[float] = []  # logprobs since the last emit
                 try:
                     if kind == "chat":
                         yield f"data: {json.dumps(chat_chunk({'role': 'assistant'}, None))}\n\n"
@@ -399,6 +400,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
                             yield f"data: {json.dumps({'error': {'message': item['error']}})}\n\n"
                             break
                         tokens.append(item["token_id"])
+                        if item.get("logprob") is not None:
+                            pending_lps.append(item["logprob"])
                         text = runner.tokenizer.decode(tokens)
                         if stop_strs:  # stop strings apply mid-stream too
                             cuts = [text.find(ss) for ss in stop_strs
@@ -430,6 +433,17 @@ def create_app(runner: EngineRunner) -> FastAPI:
                             sent_len = len(text)
                             payload = (chat_chunk({"content": new}, None)
                                        if kind == "chat" else text_chunk(new, None))
+                            if body.get("logprobs") and pending_lps:
+                                if kind == "chat":
+                                    payload["choices"][0]["logprobs"] = {
+                                        "content": [{"token": new,
+                                                     "logprob": lp}
+                                                    for lp in pending_lps]}
+                                else:
+                                    payload["choices"][0]["logprobs"] = {
+                                        "tokens": [new] + [""] * (len(pending_lps) - 1),
+                                        "token_logprobs": list(pending_lps)}
+                                pending_lps.clear()
                             yield f"data: {json.dumps(payload)}\n\n"
                         if item["finished"] and tool_mode:
                             msg = _tool_call_message(rid, body,

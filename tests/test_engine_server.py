@@ -269,3 +269,38 @@ def test_streaming_stop_strings():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+@pytest.mark.timeout(240)
+def test_streaming_logprobs():
+    import httpx
+    import json as _json
+
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-lp", "--source", "preset", "--model-ref", "tiny",
+        "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        _wait_health(port, proc)
+        lps = []
+        with httpx.stream("POST", f"http://127.0.0.1:{port}/v1/completions",
+                          json={"model": "tiny-lp", "prompt": "x",
+                                "max_tokens": 5, "ignore_eos": True,
+                                "temperature": 0, "stream": True,
+                                "logprobs": True}, timeout=60) as resp:
+            for line in resp.iter_lines():
+                if line.startswith("data:") and "[DONE]" not in line:
+                    ch = _json.loads(line[5:])["choices"][0]
+                    if ch.get("logprobs"):
+                        lps.extend(ch["logprobs"]["token_logprobs"])
+        assert len(lps) >= 4
+        assert all(isinstance(v, float) and v <= 0 for v in lps)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
