@@ -369,3 +369,9 @@ def test_pp2_dynamic_lora_matches_single():
         assert p.exitcode == 0
     with open(out_path) as f:
         assert json.load(f) == want
+
+
+@pytest.mark.timeout(300)
+def test_pp2_moe_matches_single_rank():
+    """MoE layers partition across pipeline stages like dense ones."""
+    assert _run_pp(2, model="tiny-moe") == _single_proc_result("tiny-moe")
