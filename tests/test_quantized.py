@@ -198,3 +198,70 @@ def test_quant_config_detection(tmp_path):
 def test_maybe_dequant_missing_raises():
     with pytest.raises(KeyError):
         maybe_dequant({}, "x.weight", {"method": "gptq", "bits": 4})
+
+
+def _quant_tp_rank_main(rank: int, port: int, out_path: str, model_dir: str):
+    import json
+    import os
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_tp
+
+    comm = init_tp(2, rank, master_port=port, backend="gloo")
+    cfg = EngineConfig(model=model_dir, device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, tp_size=2, tp_rank=rank,
+                       enforce_random_weights=False)
+    eng = LLMEngine(cfg, comm)
+    results, rids = {}, []
+    if rank == 0:
+        rids = [eng.add_request([1, 2, 3, 4],
+                                SamplingParams(max_tokens=6, ignore_eos=True))]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_gptq_tp2_matches_tp1(tmp_path):
+    """Dequant-on-load + TP row sharding compose (also exercises the
+    checkpoint loader's TP path end to end)."""
+    import json as _json
+    import multiprocessing as mp
+    import socket
+    import tempfile
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    _write_quant_checkpoint(tmp_path, "gptq")
+    want = LLMEngine(EngineConfig(model=str(tmp_path), device="cpu",
+                                  kv_cache_blocks=64, max_model_len=128,
+                                  enforce_random_weights=False)).generate(
+        [[1, 2, 3, 4]], SamplingParams(max_tokens=6, ignore_eos=True))
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_quant_tp_rank_main,
+                         args=(r, port, out_path, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    with open(out_path) as f:
+        assert _json.load(f) == want
