@@ -26,7 +26,14 @@ PROMPTS = [[3, 1, 4, 1, 5, 9, 2, 6], [11, 22, 33]]
 
 
 def _single_proc_result(model: str = "tiny", **params) -> list[list[int]]:
+    import torch
+
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    # single-thread math: CPU-parallel reductions reorder under load, and
+    # 1-ulp drift flips the MoE router's top-k (documented fragility) —
+    # both sides of the exactness comparison must reduce in the same order
+    torch.set_num_threads(1)
 
     cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
                        max_model_len=128, seed=0)
@@ -39,8 +46,12 @@ def _pp_rank_main(rank: int, pp: int, port: int, out_path: str,
                   model: str, params: dict):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
+    import torch
+
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
     from gpustack_amd.parallel import init_parallel
+
+    torch.set_num_threads(1)  # match _single_proc_result's reduction order
 
     comm = init_parallel(1, pp, rank, master_port=port, backend="gloo")
     cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
