@@ -222,7 +222,11 @@ class EngineConfig:
 
     def __post_init__(self):
         if self.model in PRESETS:
-            self.spec = PRESETS[self.model]
+            import dataclasses
+
+            # copy: configs may tweak their spec (tests shrink num_layers,
+            # qk_norm overrides) and must not mutate the shared preset
+            self.spec = dataclasses.replace(PRESETS[self.model])
         elif self.model_dir:
             self.spec = ModelSpec.from_dir(self.model_dir)
         elif self.model.endswith(".gguf") and Path(self.model).is_file():
