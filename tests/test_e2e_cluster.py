@@ -334,5 +334,6 @@ def test_deploy_with_lora_adapters(cluster, tmp_path_factory):
     again = client.post("/v1/completions", json={
         **body, "model": "tiny-lora-host"}).json()["choices"][0]["text"]
     assert again == base_text
-    client.delete("/v2/models/" + str(client.get("/v2/models").json()[
-        "items"][-1]["id"]))
+    mid = next(m["id"] for m in client.get("/v2/models").json()["items"]
+               if m["name"] == "tiny-lora-host")
+    client.delete(f"/v2/models/{mid}")
