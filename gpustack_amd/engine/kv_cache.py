@@ -133,7 +133,8 @@ def block_hashes(tokens, block_size: int) -> list[int]:
 
 class KVCache:
     def __init__(self, cfg: EngineConfig, num_blocks: int,
-                 device: str | torch.device, num_layers: int | None = None):
+                 device: str | torch.device, num_layers: int | None = None,
+                 host_blocks: int | None = None):
         spec = cfg.spec
         self.block_size = cfg.block_size
         self.num_blocks = num_blocks
@@ -167,11 +168,9 @@ class KVCache:
         # stream (reference semantics: LMCache/HiCache tiering,
         # schemas/models.py:204-215, re-done natively per SURVEY.md §5.7).
         self.is_cuda = torch.device(device).type == "cuda"
-        host_blocks = 0
-        if cfg.kv_offload_gb > 0:
-            per_block = (2 * self.num_layers * kv_heads * cfg.block_size
-                         * spec.head_dim * 2)
-            host_blocks = int(cfg.kv_offload_gb * 2**30) // per_block
+        if host_blocks is None:
+            host_blocks = self.compute_host_blocks(cfg, self.num_layers,
+                                                   kv_heads)
         self.host_blocks = host_blocks
         if host_blocks > 0:
             self.host_pool = torch.zeros(
@@ -222,6 +221,22 @@ class KVCache:
             torch.cuda.current_stream().wait_stream(self.side_stream)
         self.host_allocator.free(host_blocks)
         return gpu_blocks
+
+    @staticmethod
+    def compute_host_blocks(cfg: EngineConfig, num_layers: int,
+                            kv_heads: int | None = None) -> int:
+        """Host-DRAM offload pool size (blocks) for `kv_offload_gb`.
+
+        Exposed so the runner can take a cross-rank MIN before building
+        the pool (PP stages have different layer counts)."""
+        if cfg.kv_offload_gb <= 0:
+            return 0
+        spec = cfg.spec
+        if kv_heads is None:
+            kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
+        per_block = (2 * num_layers * kv_heads * cfg.block_size
+                     * spec.head_dim * 2)
+        return int(cfg.kv_offload_gb * 2**30) // per_block
 
     @staticmethod
     def compute_num_blocks(cfg: EngineConfig, free_bytes: int,

@@ -15,8 +15,12 @@ from .sequence import Sequence
 
 
 class Sampler:
-    def __init__(self, device):
+    def __init__(self, device, model_eos_token_id: int | None = None):
         self.device = device
+        # the model's EOS id (cfg.spec.eos_token_id) — the one the engine's
+        # finish check tests; params.eos_token_id is the guided-decoding
+        # terminator and may differ
+        self.model_eos_token_id = model_eos_token_id
         self.generator = None
         # guided-JSON decoding: per-token-id decoded strings (set by the
         # engine server; None on TP followers, whose picks rank 0 overwrites)
@@ -48,10 +52,17 @@ class Sampler:
             if m is not None:
                 sm.commit(m)
         seq.guided_consumed = len(out)
-        if sm.complete:
-            return p.eos_token_id
+        # `complete` means the text so far is a finished match — but the
+        # machine may still be extensible ('\d+' after one digit, a JSON
+        # number mid-stream). Forcing EOS here would give shortest-match
+        # semantics; instead EOS becomes a *candidate* competing on logits
+        # with tokens that keep the output matchable (reference guided-
+        # decoding behavior: generation continues while still matchable).
+        complete = sm.complete
 
         def valid(tid: int) -> bool:
+            if complete and tid == p.eos_token_id:
+                return True
             s = table[tid] if tid < len(table) else ""
             return bool(s) and sm.try_advance(s) is not None
 
@@ -60,6 +71,11 @@ class Sampler:
             vals, idx = torch.topk(row, k)
             ids = idx.tolist()
             valid_j = [j for j in range(k) if valid(ids[j])]
+            if complete and p.eos_token_id not in ids:
+                # make sure termination stays reachable under sampling
+                ids.append(p.eos_token_id)
+                vals = torch.cat([vals, row[p.eos_token_id].reshape(1)])
+                valid_j.append(len(ids) - 1)
             if valid_j:
                 probs = torch.softmax(
                     vals.float()[valid_j] / max(p.temperature, 1e-5), -1)
@@ -74,8 +90,7 @@ class Sampler:
                 return tid
         return p.eos_token_id  # vocab cannot extend the prefix: terminate
 
-    @staticmethod
-    def _process_logits(row: torch.Tensor, seq: Sequence) -> torch.Tensor:
+    def _process_logits(self, row: torch.Tensor, seq: Sequence) -> torch.Tensor:
         """OpenAI/HF-style logit processors (presence/frequency/repetition
         penalties + logit_bias), applied on the fp32 logits row."""
         p = seq.params
@@ -97,6 +112,19 @@ class Sampler:
                 t = int(tid)
                 if 0 <= t < row.shape[-1]:
                     row[t] += float(b)
+        if p.min_tokens > 0 and not p.ignore_eos \
+                and len(seen) < p.min_tokens:
+            # vLLM min-tokens semantics: EOS/stop ids are masked out of the
+            # distribution, not merely ignored by the finish check (an
+            # unmasked greedy model would emit EOS repeatedly into the
+            # user-visible output)
+            eos = (self.model_eos_token_id
+                   if self.model_eos_token_id is not None else p.eos_token_id)
+            if 0 <= eos < row.shape[-1]:
+                row[eos] = float("-inf")
+            for t in p.stop_token_ids:
+                if 0 <= t < row.shape[-1]:
+                    row[t] = float("-inf")
         if p.guided_token_seqs:
             out = seq.output_token_ids
             allowed = set()
@@ -209,7 +237,7 @@ class ModelRunner:
                 import logging
 
                 logging.getLogger(__name__).info("merged LoRA %s (%d tensors)", d, n)
-        self.sampler = Sampler(self.device)
+        self.sampler = Sampler(self.device, cfg.spec.eos_token_id)
         from ..models.lora import LoraBank
 
         self.lora_bank = LoraBank(cfg.spec, cfg)
@@ -265,7 +293,22 @@ class ModelRunner:
             (cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
         )
         nblocks = min(nblocks, max_needed)
-        self.kv = KVCache(cfg, nblocks, self.device, num_layers=local_layers)
+        host_blocks = KVCache.compute_host_blocks(cfg, local_layers)
+        if self.comm.world_size > 1:
+            # scheduling is replicated on every TP/PP rank and gates
+            # admission/preemption on allocator capacity; ranks measure
+            # different free memory (and PP stages hold different layer
+            # counts), so without consensus the lockstep schedulers diverge
+            # under KV pressure -> mismatched batches -> collective hang.
+            # Take the world-wide MIN of both pool sizes.
+            import torch.distributed as dist
+
+            t = torch.tensor([nblocks, host_blocks], dtype=torch.long,
+                             device=self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            nblocks, host_blocks = int(t[0]), int(t[1])
+        self.kv = KVCache(cfg, nblocks, self.device, num_layers=local_layers,
+                          host_blocks=host_blocks)
         self.eagle = None
         spec = getattr(cfg, "speculative", None)
         if spec and spec.get("method") in ("eagle", "eagle3", "mtp"):
@@ -551,7 +594,9 @@ class ModelRunner:
         self.comm.broadcast_world(t, src=self.comm.last_stage_rank)
         token_ids = t.tolist()
         self.last_logprobs = None
-        if any(s.params.logprobs for s in batch.seqs):
+        self.last_top_logprobs = None
+        if any(s.params.logprobs or s.params.top_logprobs
+               for s in batch.seqs):
             if logits is not None:
                 lf = logits.float()
                 lse = torch.logsumexp(lf, dim=-1)
@@ -559,9 +604,18 @@ class ModelRunner:
                     1, torch.as_tensor(token_ids, dtype=torch.long,
                                        device=logits.device).unsqueeze(1)
                 ).squeeze(1)
-                obj = [(chosen - lse).tolist()]
+                tops = None
+                k = max((s.params.top_logprobs for s in batch.seqs),
+                        default=0)
+                if k > 0:
+                    lsm = lf - lse.unsqueeze(1)
+                    vals, idx = torch.topk(lsm, min(k, lf.shape[-1]), dim=-1)
+                    tops = [list(zip(idx[r].tolist(), vals[r].tolist()))
+                            for r in range(lf.shape[0])]
+                obj = [((chosen - lse).tolist(), tops)]
             else:
                 obj = [None]
             dist.broadcast_object_list(obj, src=self.comm.last_stage_rank)
-            self.last_logprobs = obj[0]
+            if obj[0] is not None:
+                self.last_logprobs, self.last_top_logprobs = obj[0]
         return token_ids

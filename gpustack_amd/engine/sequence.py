@@ -49,7 +49,9 @@ class SamplingParams:
             or self.guided_json is not None \
             or self.guided_regex is not None \
             or self.presence_penalty != 0.0 \
-            or self.frequency_penalty != 0.0 or self.repetition_penalty != 1.0
+            or self.frequency_penalty != 0.0 \
+            or self.repetition_penalty != 1.0 \
+            or (self.min_tokens > 0 and not self.ignore_eos)
 
     @property
     def spec_safe(self) -> bool:

@@ -161,14 +161,41 @@ class EngineRunner:
 
     def ensure_token_table(self) -> None:
         """Build the id->string table guided-JSON decoding probes (lazy:
-        one decode pass over the vocab, first guided request only)."""
+        one decode pass over the vocab, first guided request only).
+
+        Strings are derived as decode([anchor, i]) minus decode([anchor])
+        rather than decode([i]): for SentencePiece / byte-level-BPE
+        tokenizers a lone-token decode strips the leading '▁'/space, so the
+        characters the guided machine validates would diverge from the final
+        decode(tokens) text and break the valid-JSON/regex guarantee. The
+        anchor context makes each token's string match what it contributes
+        mid-sequence. (Multi-token UTF-8 sequences still decode to U+FFFD
+        per partial token — an inherent limit of a static per-id table;
+        guided grammars constrain ASCII structure, where the diff is exact.)
+        """
         if getattr(self, "_token_table_done", False):
             return
         self._token_table_done = True
+        tok = self.tokenizer
+        anchor = None
+        prefix = ""
+        try:
+            ids = tok.encode("x")
+            ids = [i for i in ids] if not hasattr(ids, "ids") else list(ids.ids)
+            if ids:
+                anchor = ids[-1]
+                prefix = tok.decode([anchor])
+        except Exception:  # noqa: BLE001
+            anchor = None
         table = []
         for i in range(self.engine.cfg.spec.vocab_size):
             try:
-                table.append(self.tokenizer.decode([i]))
+                if anchor is not None:
+                    s = tok.decode([anchor, i])
+                    table.append(s[len(prefix):] if s.startswith(prefix)
+                                 else tok.decode([i]))
+                else:
+                    table.append(tok.decode([i]))
             except Exception:  # noqa: BLE001
                 table.append("")
         self.engine.set_token_table(table)

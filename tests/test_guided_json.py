@@ -373,3 +373,36 @@ def test_engine_guided_regex():
     out = eng.generate([[25, 26]], p)[0]
     text = tok.decode([t for t in out if t != 1])
     assert text in ("red!", "green!", "blue!")
+
+
+def test_guided_regex_not_shortest_match():
+    """when the machine is complete but still extensible, EOS
+    competes on logits with continuation tokens instead of being forced
+    ('\\d+' must be able to emit more than one digit)."""
+    import torch
+
+    from gpustack_amd.engine.model_runner import Sampler
+    from gpustack_amd.engine.sequence import SamplingParams, Sequence
+
+    table = ["<eos>", "1", "2", "a"]
+    s = Sampler("cpu")
+    s.token_table = table
+    p = SamplingParams(guided_regex=r"\d+", eos_token_id=0)
+
+    # digit logit above EOS: generation continues past the first digit
+    seq = Sequence("t", [9], p)
+    seq.output_token_ids = ["skip"]  # placeholder replaced below
+    seq.output_token_ids = [1]       # one digit emitted -> machine complete
+    row = torch.tensor([1.0, 5.0, 0.0, 9.0])  # 'a' highest but invalid
+    assert s._guided_pick(row, seq) == 1      # continues with '1'
+
+    # EOS logit above all digits: terminates (complete => EOS valid)
+    seq2 = Sequence("t2", [9], p)
+    seq2.output_token_ids = [1]
+    row2 = torch.tensor([5.0, 1.0, 0.0, 9.0])
+    assert s._guided_pick(row2, seq2) == 0
+
+    # machine NOT complete (no digit yet): EOS is not a valid pick
+    seq3 = Sequence("t3", [9], p)
+    row3 = torch.tensor([9.0, 1.0, 0.0, 5.0])
+    assert s._guided_pick(row3, seq3) == 1

@@ -36,11 +36,11 @@ def test_repetition_penalty_processor():
     s = Sequence("t", [1], SamplingParams(repetition_penalty=2.0))
     s.output_token_ids = [3]
     row = torch.tensor([0.0, 1.0, -1.0, 4.0])
-    out = Sampler._process_logits(row.clone(), s)
+    out = Sampler('cpu')._process_logits(row.clone(), s)
     assert out[3] == 2.0 and out[1] == 1.0
     s2 = Sequence("t", [1], SamplingParams(frequency_penalty=0.5))
     s2.output_token_ids = [2, 2, 2]
-    out2 = Sampler._process_logits(row.clone(), s2)
+    out2 = Sampler('cpu')._process_logits(row.clone(), s2)
     assert out2[2] == -1.0 - 1.5
 
 
@@ -124,4 +124,8 @@ def test_min_tokens_suppresses_eos():
     assert len(out) == 1 and out[0] == eos  # stops immediately without min
     p = SamplingParams(max_tokens=10, min_tokens=5, logit_bias={eos: 100.0})
     out = eng.generate([[2, 3, 4]], p)[0]
-    assert len(out) == 6  # 5 suppressed EOS emissions + the terminating one
+    assert len(out) == 6  # 5 masked-EOS steps + the terminating EOS
+    # EOS is masked out of the distribution (not merely ignored by the
+    # finish check): no EOS appears in user-visible output before the end
+    assert eos not in out[:5]
+    assert out[5] == eos
