@@ -115,9 +115,12 @@ def main():
             eng.add_request(toks, params)
             known += 1
 
+    finished_count = [0]
+
     def one_step() -> int:
         refill()
         outs = eng.step()
+        finished_count[0] += sum(1 for o in outs if o.finished)
         return len(outs)
 
     def barrier_sync():
@@ -153,7 +156,12 @@ def main():
             saturated_streak += 1
         else:
             saturated_streak = 0
-        done = (saturated_streak >= 3 or extra_cap <= 0
+        # churn equilibrium: the first admission wave must have cycled out
+        # (>= concurrency finishes) so the timed window sees the true serving
+        # mix of decode steps + refill prefills, and admitted-in-window
+        # requests yield steady-state TTFT samples
+        churned = finished_count[0] >= target
+        done = ((saturated_streak >= 3 and churned) or extra_cap <= 0
                 or time.perf_counter() > warmup_deadline)
         if dist is not None:
             # every rank must exit the same iteration (collectives inside

@@ -17,6 +17,7 @@ void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
+void gemm_lab_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -213,6 +214,18 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
   HIP_CHECK_LAST();
 }
 
+void gemm_lab(at::Tensor out, at::Tensor x, at::Tensor w, long mode) {
+  check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  int err = 0;
+  gemm_lab_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), M, N, K,
+                  (int)mode, &err, cur_stream(x));
+  TORCH_CHECK(!err, "gemm_lab: unsupported shape/mode M=", M, " N=", N,
+              " K=", K, " mode=", mode);
+  HIP_CHECK_LAST();
+}
+
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a"); check_bf16(b, "b");
   auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
@@ -237,4 +250,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");
   m.def("gemm8", &gemm8, "8-phase pipelined 256x256 GEMM (bf16, f32 accum)");
+  m.def("gemm_lab", &gemm_lab, "GEMM schedule lab variants (A/B vs hipBLASLt)");
 }
