@@ -116,11 +116,31 @@ def main():
             known += 1
 
     finished_count = [0]
+    # TTFT sampling: with identical ISL/OSL the closed loop is periodic
+    # (whole admission waves finish together every ~OSL steps), so a short
+    # timed window can miss all admissions. Samples are therefore collected
+    # from churn start onward — warmup tail AND timed region — which covers
+    # at least one full admission cycle of steady-state TTFTs.
+    ttft_track = [False]
+    first_token_seen: set[str] = set()
+    ttfts: list[float] = []
+
+    def record_first_tokens(outs) -> None:
+        if not ttft_track[0]:
+            return
+        for o in outs:
+            if o.request_id in first_token_seen:
+                continue
+            first_token_seen.add(o.request_id)
+            seq = eng.seqs.get(o.request_id)
+            if seq is not None and seq.ttft is not None:
+                ttfts.append(seq.ttft)
 
     def one_step() -> int:
         refill()
         outs = eng.step()
         finished_count[0] += sum(1 for o in outs if o.finished)
+        record_first_tokens(outs)
         return len(outs)
 
     def barrier_sync():
@@ -158,9 +178,13 @@ def main():
             saturated_streak = 0
         # churn equilibrium: the first admission wave must have cycled out
         # (>= concurrency finishes) so the timed window sees the true serving
-        # mix of decode steps + refill prefills, and admitted-in-window
-        # requests yield steady-state TTFT samples
+        # mix of decode steps + refill prefills
         churned = finished_count[0] >= target
+        if churned and not ttft_track[0]:
+            ttft_track[0] = True
+            for rid, s in eng.seqs.items():
+                if s.first_token_time is not None:
+                    first_token_seen.add(rid)
         done = ((saturated_streak >= 3 and churned) or extra_cap <= 0
                 or time.perf_counter() > warmup_deadline)
         if dist is not None:
@@ -173,25 +197,14 @@ def main():
     barrier_sync()
 
     # ---- timed region: exactly K steps ----
-    first_token_seen: set[str] = {
-        rid for rid, s in eng.seqs.items() if s.first_token_time is not None
-    }
-    ttfts: list[float] = []
+    ttft_track[0] = True
     t0 = time.perf_counter()
     tokens = 0
     for _ in range(args.steps):
         refill()
-        step_t = time.perf_counter()
         outs = eng.step()
         tokens += len(outs)
-        for o in outs:
-            if o.request_id not in first_token_seen:
-                first_token_seen.add(o.request_id)
-                seq = eng.seqs.get(o.request_id)
-                if seq is not None and seq.ttft is not None:
-                    ttfts.append(seq.ttft)
-                else:
-                    ttfts.append(time.perf_counter() - step_t)
+        record_first_tokens(outs)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 

@@ -13,6 +13,11 @@ import argparse
 import json
 import time
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 import torch
 import torch.nn.functional as F
 
@@ -85,5 +90,42 @@ def main():
     return results
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--skinny" not in __import__("sys").argv:
     main()
+
+
+def skinny_sweep(iters=50):
+    """skinny_gemm v1/v2 split-K sweep on the small-grid decode shapes."""
+    dev = "cuda:0"
+    for name, M, N, K in SHAPES:
+        if M > 1024:
+            continue
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) / 8
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) / 8
+        ref = F.linear(x, w)
+        flops = 2.0 * M * N * K
+        t_blt = time_fn(lambda: F.linear(x, w), iters)
+        row = {"shape": name, "blaslt_us": round(t_blt * 1e6, 1),
+               "blaslt_tf": round(flops / t_blt / 1e12, 0)}
+        for ver in (1, 2):
+            for sk in (1, 2, 4, 8, 16):
+                if K % (64 * sk) or (ver == 2 and N % 256):
+                    continue
+                try:
+                    out = ops.skinny_gemm(x, w, splitk=sk, version=ver)
+                    torch.cuda.synchronize()
+                except Exception:  # noqa: BLE001
+                    continue
+                rel = (out.float() - ref.float()).abs().max().item()
+                scale = ref.float().abs().max().item() + 1e-6
+                if rel / scale > 2e-2:
+                    row[f"v{ver}k{sk}"] = "ERR"
+                    continue
+                t = time_fn(lambda: ops.skinny_gemm(x, w, splitk=sk, version=ver), iters)
+                row[f"v{ver}k{sk}_tf"] = round(flops / t / 1e12, 0)
+        print(json.dumps(row), flush=True)
+
+
+import sys as _sys  # noqa: E402
+if "--skinny" in _sys.argv:
+    skinny_sweep()
