@@ -194,16 +194,19 @@ class EngineConfig:
     # host-DRAM KV offload tier (extended_kv_cache in the reference schema)
     kv_offload_gb: float = 0.0
     # automatic prefix caching: cache-hit prompts recompute only the suffix
-    # (through the paged-decode row path); suffixes longer than the cap
-    # fall back to full prefill (still registering blocks for later hits)
+    # through the paged prefill-with-history MFMA kernel; the r1 cap (the
+    # old paged-decode-row fallback cost ~3.5x per token) is retired — any
+    # suffix length now runs at flash-prefill efficiency
     enable_prefix_caching: bool = False
-    prefix_cache_suffix_cap: int = 512
+    prefix_cache_suffix_cap: int = 1 << 30
     # chunked prefill (reference: vLLM --enable-chunked-prefill): a prompt
     # longer than max_prefill_tokens is split into budget-sized chunks
-    # (chunk 0 = capped prefill batch, continuations = paged-decode rows),
-    # alternating 1:1 with decode steps so running sequences keep a bounded
-    # time-between-tokens during long-prompt admission
-    enable_chunked_prefill: bool = False
+    # (chunk 0 = capped prefill batch, continuations = paged prefill-with-
+    # history tiles), alternating 1:1 with decode steps so running
+    # sequences keep a bounded time-between-tokens during long-prompt
+    # admission. Default ON since r2: continuations run at flash-prefill
+    # efficiency (the r1 3.5x continuation penalty is gone)
+    enable_chunked_prefill: bool = True
     # speculative decoding (reference speculative_config schema):
     # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
     speculative: dict | None = None
@@ -241,6 +244,12 @@ class EngineConfig:
         if self.enable_chunked_prefill and self.speculative and \
                 self.speculative.get("method") in ("eagle", "eagle3", "mtp"):
             # draft-model speculation seeds from full-prompt prefill hiddens,
-            # which chunked admission does not produce in one step
-            raise ValueError("chunked prefill is incompatible with "
-                             "draft-model speculative decoding")
+            # which chunked admission does not produce in one step; chunked
+            # prefill (default-on since r2) yields to the explicit
+            # speculative config
+            import logging
+
+            logging.getLogger(__name__).info(
+                "disabling chunked prefill: incompatible with draft-model "
+                "speculative decoding")
+            self.enable_chunked_prefill = False

@@ -406,27 +406,64 @@ class ModelRunner:
                 num_prefill_tokens=tp,
             )
         elif batch.is_suffix:
-            # prefix-cache suffix rows: variable rows/seq through the paged
-            # decode path; only each seq's last row projects to logits
-            nrows = sum(batch.suffix_rows)
-            maxb = max(len(s.block_table) for s in batch.seqs)
-            bt = torch.zeros(nrows, maxb, dtype=torch.int32)
-            idx = []
-            r = 0
-            for s, nr in zip(batch.seqs, batch.suffix_rows):
-                row = torch.tensor(s.block_table, dtype=torch.int32)
-                for _ in range(nr):
-                    bt[r, : len(s.block_table)] = row
-                    r += 1
-                idx.append(r - 1)
-            meta = ForwardMeta(
-                is_prefill=False,
-                positions=positions,
-                slot_mapping=slots,
-                logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
-                block_tables=bt.to(dev),
-                seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32, device=dev),
-            )
+            # prefix-cache suffix / chunk-continuation rows. Preferred path:
+            # the paged prefill-with-history MFMA kernel (one flash pass per
+            # seq whose K/V stream starts from cached blocks). Fallback to
+            # per-row paged-decode only where the kernel doesn't apply
+            # (fp8 KV cache, head_dim != 128 on GPU).
+            hists = [L0 - 1 for L0 in
+                     (batch.seq_lens[sum(batch.suffix_rows[:i])]
+                      for i in range(len(batch.seqs)))]
+            news = list(batch.suffix_rows)
+            starts = []
+            off = 0
+            for nr in news:
+                starts.append(off)
+                off += nr
+            idx = [st + nr - 1 for st, nr in zip(starts, news)]
+            if dev.type == "cuda":
+                use_pp = (self.kv is not None
+                          and self.kv.kv_dtype == torch.bfloat16
+                          and self.cfg.spec.head_dim == 128)
+            else:  # torch_ref path handles any dtype/head_dim except fp8
+                use_pp = (self.kv is not None
+                          and self.kv.kv_dtype != torch.float8_e4m3fn)
+            if use_pp:
+                maxb = max(len(s.block_table) for s in batch.seqs)
+                bt = torch.zeros(len(batch.seqs), maxb, dtype=torch.int32)
+                for i, s in enumerate(batch.seqs):
+                    bt[i, : len(s.block_table)] = torch.tensor(
+                        s.block_table, dtype=torch.int32)
+                tiles = ops.build_paged_prefill_tiles(starts, hists, news, dev)
+                meta = ForwardMeta(
+                    is_prefill=False,
+                    positions=positions,
+                    slot_mapping=slots,
+                    logits_indices=torch.tensor(idx, dtype=torch.long,
+                                                device=dev),
+                    block_tables=bt.to(dev),
+                    suffix_meta=(tiles, starts, hists, news),
+                )
+            else:
+                nrows = sum(batch.suffix_rows)
+                maxb = max(len(s.block_table) for s in batch.seqs)
+                bt = torch.zeros(nrows, maxb, dtype=torch.int32)
+                r = 0
+                for s, nr in zip(batch.seqs, batch.suffix_rows):
+                    row = torch.tensor(s.block_table, dtype=torch.int32)
+                    for _ in range(nr):
+                        bt[r, : len(s.block_table)] = row
+                        r += 1
+                meta = ForwardMeta(
+                    is_prefill=False,
+                    positions=positions,
+                    slot_mapping=slots,
+                    logits_indices=torch.tensor(idx, dtype=torch.long,
+                                                device=dev),
+                    block_tables=bt.to(dev),
+                    seq_lens=torch.tensor(batch.seq_lens, dtype=torch.int32,
+                                          device=dev),
+                )
         else:
             rps = batch.rows_per_seq
             nrows = len(batch.seqs) * rps

@@ -42,6 +42,10 @@ class ForwardMeta:
     # mixed batches: rows [0, num_prefill_tokens) are prefill, the rest are
     # single-token decode rows
     num_prefill_tokens: int = 0
+    # prefix-cache suffix / chunked-prefill continuation rows running through
+    # the paged prefill-with-history kernel: (tiles 5-tuple, starts, hists,
+    # news); block_tables is then per-SEQ ([nseq, maxb]) instead of per-row
+    suffix_meta: tuple | None = None
     # dynamic multi-LoRA: per-batch adapter row groups (models/lora.py
     # BatchLora); None when no row in the batch uses an adapter
     lora: object | None = None
@@ -105,6 +109,11 @@ class Attention(nn.Module):
                     out[tp:], q[tp:], k_cache, v_cache,
                     meta.block_tables, meta.seq_lens, self.scale,
                 )
+        elif meta.suffix_meta is not None:
+            tiles, starts, hists, news = meta.suffix_meta
+            ops.paged_prefill_attn(out, q, k_cache, v_cache,
+                                   meta.block_tables, starts, hists, news,
+                                   self.scale, tiles=tiles)
         else:
             ops.paged_attn_decode(
                 out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale

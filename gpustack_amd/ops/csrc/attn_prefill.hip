@@ -255,7 +255,253 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Prefill-with-history: same MFMA flash structure, but the K/V stream is
+// gathered block-wise from the PAGED pool ([nblocks, Hkv, 16, D] bf16)
+// through the sequence's block table. Used for prefix-cache suffixes and
+// chunked-prefill continuations, whose rows previously ran as paged-decode
+// rows at ~3.5x the attention cost (profiles/r03 chunked A/B) — the
+// measured round-1 penalty this kernel removes. The new tokens' K/V are
+// already scattered into the pool (reshape_and_cache runs before
+// attention), so ALL positions stream from the pool uniformly.
+// ---------------------------------------------------------------------------
+namespace {
+
+constexpr int PP_BS = 16;  // pool block size (tokens per KV block)
+
+__global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
+    unsigned short* __restrict__ out,      // [T, Hq, D] (suffix rows)
+    const unsigned short* __restrict__ q,  // [T, Hq, D] (suffix rows)
+    const unsigned short* __restrict__ kc, // [nblocks, Hkv, BS, D]
+    const unsigned short* __restrict__ vc, // [nblocks, Hkv, BS, D]
+    const int* __restrict__ block_tables,  // [nseq, maxb]
+    const int* __restrict__ tile_qstart,   // [ntiles] seq's first row in q/out
+    const int* __restrict__ tile_q0,       // [ntiles] q-tile offset in suffix
+    const int* __restrict__ tile_hist,     // [ntiles] cached tokens (history)
+    const int* __restrict__ tile_new,      // [ntiles] suffix length
+    const int* __restrict__ tile_seq,      // [ntiles] row into block_tables
+    int Hq, int Hkv, int maxb, float scale, long qs) {
+  const int tile = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (Hq / Hkv);
+  const int qstart = tile_qstart[tile];
+  const int q0 = tile_q0[tile];
+  const int hist = tile_hist[tile];
+  const int nnew = tile_new[tile];
+  const int len = hist + nnew;  // total context length
+  const int* bt = block_tables + (long)tile_seq[tile] * maxb;
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lc = lane & 15;
+  const int lg = lane >> 4;
+  const int tid = threadIdx.x;
+
+  __shared__ unsigned short Kl[BK * PF_D];
+  __shared__ unsigned short VTl[PF_D][BK + VT_PAD];
+  __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
+
+  const int qrow_local = q0 + wave * 16 + lc;           // row within suffix
+  const int qrow_clamped = (qrow_local < nnew) ? qrow_local : (nnew - 1);
+  u16x8 qfrag[4];
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk) {
+    const unsigned short* qp = q + (long)(qstart + qrow_clamped) * qs +
+                               (long)qh * PF_D + kk * 32 + lg * 8;
+    qfrag[kk] = *reinterpret_cast<const u16x8*>(qp);
+  }
+
+  float mcol = -INFINITY;
+  float lcol = 0.f;
+  f32x4 o[PF_D / 16];
+#pragma unroll
+  for (int nt = 0; nt < PF_D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int q_hi_abs = hist + q0 + BQ - 1;              // absolute position
+  const int kv_end = min(len, q_hi_abs + 1);
+  const int ntiles_kv = (kv_end + BK - 1) / BK;
+  const int wave_q_hi_abs = hist + q0 + wave * 16 + 15;
+
+  const int kst_kv[KU] = {tid / 16, (tid + 256) / 16, (tid + 512) / 16,
+                          (tid + 768) / 16};
+  const int kst_d0 = (tid % 16) * 8;
+  const int vst_d = tid % PF_D;
+  const int vst_kvc0 = (tid / PF_D) * 8;
+
+  u16x8 kstage[KU];
+  unsigned short vstage[VU][8];
+
+  // pool row base (elements) for token position `kv`
+  auto pool_row = [&](int kv) {
+    const int blk = bt[kv / PP_BS];
+    return (((long)blk * Hkv + kvh) * PP_BS + kv % PP_BS) * PF_D;
+  };
+
+  auto issue_loads = [&](int kv0) {
+    // clamp BEFORE the block-table lookup: a speculated address compute
+    // for masked-out lanes must not index past the table row
+#pragma unroll
+    for (int r = 0; r < KU; ++r) {
+      const int kv = kv0 + kst_kv[r];
+      const bool ok = kv < len;
+      const u16x8 val = *reinterpret_cast<const u16x8*>(
+          kc + pool_row(ok ? kv : len - 1) + kst_d0);
+      kstage[r] = ok ? val : u16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+#pragma unroll
+    for (int r = 0; r < VU; ++r) {
+      const int kvc = vst_kvc0 + r * 16;
+      // 8 consecutive positions span at most 2 pool blocks; resolving the
+      // row base per element keeps the gather exact at block boundaries
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kv = kv0 + kvc + j;
+        const bool ok = kv < len;
+        const unsigned short val = vc[pool_row(ok ? kv : len - 1) + vst_d];
+        vstage[r][j] = ok ? val : (unsigned short)0;
+      }
+    }
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int r = 0; r < KU; ++r) {
+      *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
+                                kswz(kst_kv[r], kst_d0 * 2)) = kstage[r];
+    }
+#pragma unroll
+    for (int r = 0; r < VU; ++r) {
+      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * 16]) =
+          *reinterpret_cast<u16x8*>(vstage[r]);
+    }
+  };
+
+  issue_loads(0);
+  for (int kt = 0; kt < ntiles_kv; ++kt) {
+    const int kv0 = kt * BK;
+    __syncthreads();
+    write_lds();
+    __syncthreads();
+    if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);
+
+    if (kv0 > wave_q_hi_abs) continue;
+
+    f32x4 st[NST];
+#pragma unroll
+    for (int i = 0; i < NST; ++i) st[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int stile = 0; stile < NST; ++stile) {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const u16x8 a = *reinterpret_cast<const u16x8*>(
+            reinterpret_cast<char*>(Kl) +
+            kswz(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
+        st[stile] = mfma16x16x32_bf16(a, qfrag[kk], st[stile]);
+      }
+    }
+
+    const int qpos = hist + q0 + wave * 16 + lc;  // absolute position
+    const bool qvalid = (q0 + wave * 16 + lc) < nnew;
+    float sv[NST * 4];
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int stile = 0; stile < NST; ++stile) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kvpos = kv0 + stile * 16 + lg * 4 + r;
+        float x = st[stile][r] * scale;
+        const bool ok = (kvpos <= qpos) && (kvpos < len) && qvalid;
+        x = ok ? x : -INFINITY;
+        sv[stile * 4 + r] = x;
+        tmax = fmaxf(tmax, x);
+      }
+    }
+#pragma unroll
+    for (int msk = 16; msk <= 32; msk <<= 1)
+      tmax = fmaxf(tmax, __shfl_xor(tmax, msk, WAVE_SIZE));
+
+    const float nm = fmaxf(mcol, tmax);
+    float corr = 1.f, tsum = 0.f;
+    float pv[NST * 4];
+    if (nm != -INFINITY) {
+      corr = __expf(mcol - nm);
+#pragma unroll
+      for (int i = 0; i < NST * 4; ++i) {
+        pv[i] = (sv[i] == -INFINITY) ? 0.f : __expf(sv[i] - nm);
+        tsum += pv[i];
+      }
+      mcol = nm;
+    } else {
+#pragma unroll
+      for (int i = 0; i < NST * 4; ++i) pv[i] = 0.f;
+    }
+#pragma unroll
+    for (int msk = 16; msk <= 32; msk <<= 1)
+      tsum += __shfl_xor(tsum, msk, WAVE_SIZE);
+    lcol = lcol * corr + tsum;
+
+#pragma unroll
+    for (int stile = 0; stile < NST; ++stile) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        Pl[wave][lc][stile * 16 + lg * 4 + r] = f2bf(pv[stile * 4 + r]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int orow = lg * 4 + r;
+      const float c = __shfl(corr, orow, WAVE_SIZE);
+#pragma unroll
+      for (int nt = 0; nt < PF_D / 16; ++nt) o[nt][r] *= c;
+    }
+
+#pragma unroll
+    for (int kk2 = 0; kk2 < KC2; ++kk2) {
+      const u16x8 pa = *reinterpret_cast<const u16x8*>(
+          &Pl[wave][lc][kk2 * 32 + lg * 8]);
+#pragma unroll
+      for (int nt = 0; nt < PF_D / 16; ++nt) {
+        const u16x8 b = *reinterpret_cast<const u16x8*>(
+            &VTl[nt * 16 + lc][kk2 * 32 + lg * 8]);
+        o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int orow = lg * 4 + r;
+    const float denom = __shfl(lcol, orow, WAVE_SIZE);
+    const int qrow = q0 + wave * 16 + orow;
+    if (qrow >= nnew || denom <= 0.f) continue;
+    const float inv = 1.f / denom;
+#pragma unroll
+    for (int nt = 0; nt < PF_D / 16; ++nt) {
+      out[((long)(qstart + qrow) * Hq + qh) * PF_D + nt * 16 + lc] =
+          f2bf(o[nt][r] * inv);
+    }
+  }
+}
+
 }  // namespace
+
+void flash_prefill_paged_launch(
+    void* out, const void* q, const void* kc, const void* vc,
+    const int* block_tables, const int* tile_qstart, const int* tile_q0,
+    const int* tile_hist, const int* tile_new, const int* tile_seq,
+    int ntiles, int Hq, int Hkv, int D, int maxb, float scale, long qs,
+    int* err_unsupported, hipStream_t s) {
+  *err_unsupported = 0;
+  if (D != 128 || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  dim3 grid(ntiles, Hq);
+  hipLaunchKernelGGL(flash_prefill_paged_kernel, grid, dim3(PF_THREADS), 0, s,
+                     (unsigned short*)out, (const unsigned short*)q,
+                     (const unsigned short*)kc, (const unsigned short*)vc,
+                     block_tables, tile_qstart, tile_q0, tile_hist, tile_new,
+                     tile_seq, Hq, Hkv, maxb, scale, qs);
+}
 
 void flash_prefill_launch(void* out, const void* q, const void* k,
                           const void* v, const int* tile_start,

@@ -13,6 +13,7 @@ void reshape_and_cache_launch(const void*, const void*, void*, void*, const long
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
 void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, int*, hipStream_t);
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
+void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
@@ -176,6 +177,34 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   HIP_CHECK_LAST();
 }
 
+void flash_prefill_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                         at::Tensor v_cache, at::Tensor block_tables,
+                         at::Tensor tile_qstart, at::Tensor tile_q0,
+                         at::Tensor tile_hist, at::Tensor tile_new,
+                         at::Tensor tile_seq, double scale) {
+  check_bf16(out, "out");
+  const long qs = row_stride_3d(q, "q");
+  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  for (auto* t : {&block_tables, &tile_qstart, &tile_q0, &tile_hist,
+                  &tile_new, &tile_seq}) {
+    TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous());
+  }
+  const int ntiles = tile_qstart.size(0);
+  const int Hq = q.size(1), D = q.size(2);
+  const int Hkv = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "paged prefill assumes block_size 16");
+  const int maxb = block_tables.size(1);
+  int err = 0;
+  flash_prefill_paged_launch(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr<int>(), tile_qstart.data_ptr<int>(),
+      tile_q0.data_ptr<int>(), tile_hist.data_ptr<int>(),
+      tile_new.data_ptr<int>(), tile_seq.data_ptr<int>(), ntiles, Hq, Hkv, D,
+      maxb, (float)scale, qs, &err, cur_stream(q));
+  TORCH_CHECK(!err, "flash_prefill_paged: unsupported config D=", D);
+  HIP_CHECK_LAST();
+}
+
 void gemm8(at::Tensor out, at::Tensor x, at::Tensor w, long safe) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   const int M = x.size(0), K = x.size(1), N = w.size(0);
@@ -247,6 +276,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample, "argmax over vocab");
   m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention");
   m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention");
+  m.def("flash_prefill_paged", &flash_prefill_paged,
+        "MFMA prefill attention with paged-KV history (suffix/chunk rows)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");
   m.def("gemm8", &gemm8, "8-phase pipelined 256x256 GEMM (bf16, f32 accum)");

@@ -159,3 +159,46 @@ def varlen_prefill_attn(
         att = torch.softmax(att + mask, dim=-1)
         out[start : start + L] = (att @ vs).permute(1, 0, 2).to(out.dtype)
         start += L
+
+
+def paged_prefill_attn(
+    out: torch.Tensor,
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_starts: list[int],
+    seq_hists: list[int],
+    seq_news: list[int],
+    scale: float,
+) -> None:
+    """Prefill-with-history: q/out hold only the NEW (suffix) rows of each
+    sequence; K/V for positions [0, hist+new) are gathered from the paged
+    pool. Row r of sequence i attends causally to positions [0, hist+r].
+
+    q/out: [T, Hq, D]; caches [B, Hkv, BS, D]; block_tables [nseq, maxb].
+    """
+    Hq = q.shape[1]
+    D = q.shape[2]
+    Hkv = k_cache.shape[1]
+    BS = k_cache.shape[2]
+    GQ = Hq // Hkv
+    row = 0
+    for i, (hist, new) in enumerate(zip(seq_hists, seq_news)):
+        L = hist + new
+        nblk = (L + BS - 1) // BS
+        blocks = block_tables[i, :nblk].long()
+        keys = k_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        vals = v_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hkv, -1, D)[:, :L]
+        keys = keys.repeat_interleave(GQ, dim=0)  # [Hq, L, D]
+        vals = vals.repeat_interleave(GQ, dim=0)
+        start = seq_starts[i]
+        qs = q[start:start + new].float().permute(1, 0, 2)  # [Hq, new, D]
+        att = (qs @ keys.transpose(1, 2)) * scale           # [Hq, new, L]
+        pos = torch.arange(L, device=q.device)
+        qpos = hist + torch.arange(new, device=q.device)
+        mask = torch.where(pos.unsqueeze(0) <= qpos.unsqueeze(1), 0.0,
+                           float("-inf"))
+        att = torch.softmax(att + mask, dim=-1)
+        out[start:start + new] = (att @ vals).permute(1, 0, 2).to(out.dtype)
+        row += new

@@ -80,9 +80,11 @@ def test_block_accounting_and_abort_mid_chunk():
 
 
 def test_chunked_with_spec_model_rejected():
-    with pytest.raises(ValueError):
-        _cfg(enable_chunked_prefill=True,
-             speculative={"method": "eagle", "num_draft_tokens": 3})
+    # chunked prefill (default-on since r2) yields to an explicit
+    # draft-model speculative config instead of raising
+    cfg = _cfg(enable_chunked_prefill=True,
+               speculative={"method": "eagle", "num_draft_tokens": 3})
+    assert cfg.enable_chunked_prefill is False
 
 
 def test_chunked_with_ngram_spec_ok():

@@ -303,3 +303,84 @@ def test_gemm8_matches_safe_variant():
     a = ops.gemm8(x, w, safe=False)
     b = ops.gemm8(x, w, safe=True)
     assert torch.equal(a, b)
+
+
+@pytest.mark.parametrize("Hq,Hkv", [(32, 8), (8, 8)])
+@pytest.mark.parametrize("cases", [
+    [(0, 64)],                 # no history (pure paged prefill)
+    [(96, 64)],                # block-aligned history
+    [(100, 29)],               # unaligned history + odd suffix
+    [(1000, 513), (16, 1)],    # long history, multi-tile suffix, 1-row seq
+])
+def test_flash_prefill_paged(Hq, Hkv, cases):
+    """Prefill-with-history kernel vs torch_ref: K/V gathered from the
+    paged pool through shuffled block tables; history + suffix causality."""
+    D, BS = 128, 16
+    dev = "cuda"
+    nseq = len(cases)
+    tot_new = sum(n for _, n in cases)
+    maxb = max((h + n + BS - 1) // BS for h, n in cases)
+    nblocks = sum((h + n + BS - 1) // BS for h, n in cases) + 3
+    k_cache = torch.randn(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device=dev) / 4
+    v_cache = torch.randn(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device=dev) / 4
+    # shuffled block assignment (realistic non-contiguous tables)
+    perm = torch.randperm(nblocks - 1)[: sum((h + n + BS - 1) // BS for h, n in cases)] + 1
+    bt = torch.zeros(nseq, maxb, dtype=torch.int32, device=dev)
+    off = 0
+    for i, (h, n) in enumerate(cases):
+        nb = (h + n + BS - 1) // BS
+        bt[i, :nb] = perm[off:off + nb].to(torch.int32)
+        off += nb
+    q = torch.randn(tot_new, Hq, D, dtype=torch.bfloat16, device=dev) / 4
+    starts, hists, news = [], [], []
+    r = 0
+    for h, n in cases:
+        starts.append(r)
+        hists.append(h)
+        news.append(n)
+        r += n
+    out = torch.empty_like(q)
+    ops.paged_prefill_attn(out, q, k_cache, v_cache, bt, starts, hists, news,
+                           1.0 / math.sqrt(D))
+    ref = torch.empty_like(q)
+    R.paged_prefill_attn(ref, q, k_cache, v_cache, bt, starts, hists, news,
+                         1.0 / math.sqrt(D))
+    _close(out, ref)
+
+
+def test_flash_prefill_paged_matches_contiguous():
+    """hist=0 paged prefill == the contiguous varlen prefill kernel on the
+    same data (cross-kernel consistency)."""
+    D, BS, Hq, Hkv = 128, 16, 32, 8
+    lens = [64, 129, 7]
+    T = sum(lens)
+    dev = "cuda"
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev) / 4
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev) / 4
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev) / 4
+    out_c = torch.empty_like(q)
+    ops.varlen_prefill_attn(out_c, q, k, v, lens, 1.0 / math.sqrt(D))
+    # scatter k/v into a pool
+    nb_per = [(L + BS - 1) // BS for L in lens]
+    nblocks = sum(nb_per) + 1
+    k_cache = torch.zeros(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device=dev)
+    v_cache = torch.zeros(nblocks, Hkv, BS, D, dtype=torch.bfloat16, device=dev)
+    maxb = max(nb_per)
+    bt = torch.zeros(len(lens), maxb, dtype=torch.int32, device=dev)
+    blk = 1
+    row = 0
+    slots = []
+    for i, L in enumerate(lens):
+        for j in range(nb_per[i]):
+            bt[i, j] = blk + j
+        for p in range(L):
+            slots.append((blk + p // BS) * BS + p % BS)
+        blk += nb_per[i]
+        row += L
+    slots_t = torch.tensor(slots, dtype=torch.long, device=dev)
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots_t)
+    starts = [sum(lens[:i]) for i in range(len(lens))]
+    out_p = torch.empty_like(q)
+    ops.paged_prefill_attn(out_p, q, k_cache, v_cache, bt, starts,
+                           [0] * len(lens), lens, 1.0 / math.sqrt(D))
+    _close(out_p, out_c)
