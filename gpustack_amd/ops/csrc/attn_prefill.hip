@@ -255,6 +255,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   }
 }
 
+}  // namespace
+
 // ---------------------------------------------------------------------------
 // Prefill-with-history: same MFMA flash structure, but the K/V stream is
 // gathered block-wise from the PAGED pool ([nblocks, Hkv, 16, D] bf16)
