@@ -325,11 +325,17 @@ class ModelRunner:
         if self.device.type == "cuda":
             from .graph_runner import DecodeGraphRunner, graphs_enabled
 
-            # MoE dispatch syncs with the host per layer (expert routing),
-            # which hipGraph capture cannot record -> eager decode for MoE
-            # (sync-free fused dispatch is the round-2 item)
+            # MoE decode is graph-capturable since r2: the fused grouped-
+            # GEMM dispatch (ops/csrc/moe_gemm.hip) has no host syncs and
+            # fixed launch grids. Only the torch fallback paths (unaligned
+            # dims) still sync per layer -> eager for those.
+            moe_graph_ok = (cfg.spec.num_experts == 0
+                            or (cfg.spec.moe_intermediate_size
+                                // max(1, cfg.tp_size) % 64 == 0
+                                and cfg.spec.hidden_size % 64 == 0
+                                and ops.hip_available()))
             if graphs_enabled() and self.eagle is None \
-                    and cfg.spec.num_experts == 0 and self.comm.pp_size == 1:
+                    and moe_graph_ok and self.comm.pp_size == 1:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv

@@ -197,19 +197,54 @@ class MoEMLP(nn.Module):
         # duplicated token rows would use atomics, whose order (and thus
         # the bf16 rounding) varies run to run and flips the next layer's
         # router on near-ties (observed on HW).
+        # Three dispatch paths:
+        #  - fused grouped-GEMM HIP kernels (ops/csrc/moe_gemm.hip): no host
+        #    sync anywhere, fixed launch grids — hipGraph-capturable; the
+        #    default on GPU for MFMA-aligned dims
+        #  - decode-shaped torch fallback: ONE padded strided-batched GEMM
+        #    pair (pays a per-layer counts.max() host sync)
+        #  - prefill-shaped: per-expert loop GEMMs (large per-expert work)
+        if self._fused_ok(x):
+            out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w)
+            return self.comm.all_reduce(out)
         contrib = x.new_zeros(T * self.top_k, x.shape[1])
-        # Two dispatch paths:
-        #  - decode-shaped (few tokens/expert, many experts): ONE padded
-        #    strided-batched GEMM pair — the per-expert loop would be
-        #    launch/host-sync bound (measured 800 ms/step on qwen3-30b-a3b)
-        #  - prefill-shaped: per-expert GEMMs are large enough that the
-        #    loop is GEMM-bound and avoids the padding overcompute
         if self.e >= 16 and flat_exp.numel() < 32 * self.e:
             self._bmm_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
         else:
             self._loop_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
         out = contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
         return self.comm.all_reduce(out)
+
+    def _fused_ok(self, x) -> bool:
+        import os
+
+        return (x.is_cuda and x.dtype == torch.bfloat16
+                and os.environ.get("GPUSTACK_AMD_FUSED_MOE", "1") == "1"
+                and self.i % 64 == 0 and x.shape[1] % 64 == 0
+                and ops.hip_available())
+
+    def _fused_dispatch(self, x, flat_exp, flat_tok, flat_w):
+        """Sync-free grouped expert GEMMs: sort assignments by expert on
+        device, run the two fused kernels, reduce over the k axis. Every
+        tensor shape here depends only on (T, k, E) — safe under hipGraph
+        capture (reference capability: vLLM fused_moe; re-designed for
+        CDNA4 per SURVEY.md §2.9 #1)."""
+        T = x.shape[0]
+        TK = flat_exp.numel()
+        order = torch.argsort(flat_exp, stable=True)
+        s_tok = flat_tok[order].to(torch.int32)
+        # scatter_add instead of bincount: bincount computes max() on host
+        counts = torch.zeros(self.e, dtype=torch.int32, device=x.device)
+        counts.scatter_add_(0, flat_exp,
+                            torch.ones_like(flat_exp, dtype=torch.int32))
+        offs = (counts.cumsum(0, dtype=torch.int32) - counts).to(torch.int32)
+        hip = ops._load_hip()
+        act = x.new_empty(TK, self.i)
+        hip.moe_gate_up_silu(act, x, self.gate_up_w, s_tok, offs, counts)
+        contrib = x.new_empty(TK, x.shape[1])
+        hip.moe_down_scale(contrib, act, self.down_w, offs, counts,
+                           order.to(torch.int32), flat_w.float())
+        return contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
 
     def _loop_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w):
         hit = torch.bincount(flat_exp, minlength=self.e)

@@ -19,6 +19,8 @@ void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, i
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm_lab_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
+void moe_gate_up_silu_launch(void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int*, hipStream_t);
+void moe_down_scale_launch(void*, const void*, const void*, const int*, const int*, const int*, const float*, int, int, int, int*, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -243,6 +245,46 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
   HIP_CHECK_LAST();
 }
 
+void moe_gate_up_silu(at::Tensor act, at::Tensor x, at::Tensor w,
+                      at::Tensor s_tok, at::Tensor offs, at::Tensor counts) {
+  check_bf16(act, "act"); check_bf16(x, "x"); check_bf16(w, "w");
+  for (auto* t : {&s_tok, &offs, &counts}) {
+    TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous());
+  }
+  const int E = w.size(0), H = x.size(1);
+  const long I2 = w.size(1);
+  const int I = (int)(I2 / 2);
+  TORCH_CHECK(w.size(2) == H && act.size(1) == I);
+  TORCH_CHECK(act.size(0) == s_tok.size(0));
+  int err = 0;
+  moe_gate_up_silu_launch(act.data_ptr(), x.data_ptr(), w.data_ptr(),
+                          s_tok.data_ptr<int>(), offs.data_ptr<int>(),
+                          counts.data_ptr<int>(), E, H, I, &err,
+                          cur_stream(x));
+  TORCH_CHECK(!err, "moe_gate_up_silu: unsupported dims H=", H, " I=", I);
+  HIP_CHECK_LAST();
+}
+
+void moe_down_scale(at::Tensor contrib, at::Tensor act, at::Tensor w,
+                    at::Tensor offs, at::Tensor counts, at::Tensor order,
+                    at::Tensor flat_w) {
+  check_bf16(contrib, "contrib"); check_bf16(act, "act"); check_bf16(w, "w");
+  for (auto* t : {&offs, &counts, &order}) {
+    TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous());
+  }
+  TORCH_CHECK(flat_w.scalar_type() == at::kFloat && flat_w.is_contiguous());
+  const int E = w.size(0), H = w.size(1), I = w.size(2);
+  TORCH_CHECK(act.size(1) == I && contrib.size(1) == H);
+  TORCH_CHECK(contrib.size(0) == order.size(0));
+  int err = 0;
+  moe_down_scale_launch(contrib.data_ptr(), act.data_ptr(), w.data_ptr(),
+                        offs.data_ptr<int>(), counts.data_ptr<int>(),
+                        order.data_ptr<int>(), flat_w.data_ptr<float>(), E,
+                        H, I, &err, cur_stream(act));
+  TORCH_CHECK(!err, "moe_down_scale: unsupported dims H=", H, " I=", I);
+  HIP_CHECK_LAST();
+}
+
 void gemm_lab(at::Tensor out, at::Tensor x, at::Tensor w, long mode) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   const int M = x.size(0), K = x.size(1), N = w.size(0);
@@ -282,4 +324,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");
   m.def("gemm8", &gemm8, "8-phase pipelined 256x256 GEMM (bf16, f32 accum)");
   m.def("gemm_lab", &gemm_lab, "GEMM schedule lab variants (A/B vs hipBLASLt)");
+  m.def("moe_gate_up_silu", &moe_gate_up_silu,
+        "grouped MoE gate/up GEMM + SiLU (sorted assignments, sync-free)");
+  m.def("moe_down_scale", &moe_down_scale,
+        "grouped MoE down GEMM + routing-weight scale/scatter");
 }

@@ -1,0 +1,190 @@
+// Grouped MoE expert GEMMs for MI355X (gfx950) — the sync-free dispatch
+// path (round-2 item: remove the per-layer bincount/max host syncs and the
+// padded-bmm overcompute so MoE decode can run under hipGraphs).
+//
+// Two kernels over the expert-sorted assignment list (sorting/count/offset
+// tensors are computed on-device by torch ops in models/llama.py MoEMLP —
+// no host readback anywhere):
+//
+//   moe_gate_up_silu: act[j, :] = silu(x[tok_j] @ Wg[e_j]^T) * (x[tok_j] @ Wu[e_j]^T)
+//   moe_down_scale:   contrib[order[j], :] = (act[j] @ Wd[e_j]^T) * w_j
+//
+// Fixed launch grids (E x N-tiles) independent of the routing outcome:
+// workgroups for empty experts exit immediately. MFMA 16x16x32 tiles,
+// A = 16 gathered token rows staged through LDS in 512-element K chunks,
+// B = expert weight panels streamed from HBM (read once per 16-row m-tile;
+// typical decode occupancy is <= 2 m-tiles per expert).
+#include "common.h"
+
+namespace {
+
+constexpr int MOE_BN = 64;        // N columns per workgroup (16 per wave)
+constexpr int MOE_KC = 512;       // K chunk staged in LDS
+constexpr int MOE_THREADS = 256;  // 4 waves
+
+typedef __attribute__((ext_vector_type(8))) short m_s16x8;
+
+DEVICE_INLINE f32x4 moe_mfma(u16x8 a, u16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+      __builtin_bit_cast(m_s16x8, a), __builtin_bit_cast(m_s16x8, b), c, 0, 0, 0);
+}
+
+// stage rows [16][kc] of gathered x into LDS (row padded to KC+8 elements
+// to break the bank cycle on the A-frag reads)
+template <bool GATHER>
+DEVICE_INLINE void stage_rows(unsigned short* lds, const unsigned short* src,
+                              long src_stride, const int* tok, int base_row,
+                              int nrows, int k0, int kc, int tid) {
+  const int units = 16 * (kc / 8);         // u16x8 units
+  for (int u = tid; u < units; u += MOE_THREADS) {
+    const int row = u / (kc / 8);
+    const int kk = (u % (kc / 8)) * 8;
+    u16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (row < nrows) {
+      const long r = GATHER ? (long)tok[row] : (long)(base_row + row);
+      val = *reinterpret_cast<const u16x8*>(src + r * src_stride + k0 + kk);
+    }
+    *reinterpret_cast<u16x8*>(lds + row * (MOE_KC + 8) + kk) = val;
+  }
+}
+
+__global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
+    unsigned short* __restrict__ act,       // [TK, I] bf16 (sorted rows)
+    const unsigned short* __restrict__ x,   // [T, H] bf16
+    const unsigned short* __restrict__ w,   // [E, 2I, H] bf16 (gate then up)
+    const int* __restrict__ s_tok,          // [TK] token index per sorted row
+    const int* __restrict__ offs,           // [E]
+    const int* __restrict__ counts,         // [E]
+    int H, int I) {
+  const int nnt = I / MOE_BN;
+  const int e = blockIdx.x / nnt;
+  const int nt = blockIdx.x % nnt;
+  const int cnt = counts[e];
+  if (cnt == 0) return;
+  const int base = offs[e];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE_SIZE - 1);
+  const int wave = tid / WAVE_SIZE;
+  const int lc = lane & 15;
+  const int lg = lane >> 4;
+
+  __shared__ unsigned short Xl[16 * (MOE_KC + 8)];
+
+  const unsigned short* wg_panel =
+      w + ((long)e * 2 * I + nt * MOE_BN + wave * 16) * H;
+  const unsigned short* wu_panel =
+      w + ((long)e * 2 * I + I + nt * MOE_BN + wave * 16) * H;
+
+  for (int m0 = 0; m0 < cnt; m0 += 16) {
+    const int nrows = min(16, cnt - m0);
+    f32x4 ag{0.f, 0.f, 0.f, 0.f}, au{0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < H; k0 += MOE_KC) {
+      const int kc = min(MOE_KC, H - k0);
+      __syncthreads();
+      stage_rows<true>(Xl, x, H, s_tok + base + m0, 0, nrows, k0, kc, tid);
+      __syncthreads();
+      for (int kk = 0; kk < kc; kk += 32) {
+        const u16x8 a = *reinterpret_cast<const u16x8*>(
+            Xl + lc * (MOE_KC + 8) + kk + lg * 8);
+        const u16x8 bg = *reinterpret_cast<const u16x8*>(
+            wg_panel + (long)lc * H + k0 + kk + lg * 8);
+        const u16x8 bu = *reinterpret_cast<const u16x8*>(
+            wu_panel + (long)lc * H + k0 + kk + lg * 8);
+        ag = moe_mfma(a, bg, ag);
+        au = moe_mfma(a, bu, au);
+      }
+    }
+    // D[row = lg*4 + r][col = lc]; fuse SiLU(gate) * up and write
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = lg * 4 + r;
+      if (row >= nrows) continue;
+      const float g = ag[r];
+      const float v = g / (1.f + __expf(-g)) * au[r];
+      act[(long)(base + m0 + row) * I + nt * MOE_BN + wave * 16 + lc] = f2bf(v);
+    }
+  }
+}
+
+__global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
+    unsigned short* __restrict__ contrib,   // [TK, H] bf16 (original order)
+    const unsigned short* __restrict__ act, // [TK, I] bf16 (sorted rows)
+    const unsigned short* __restrict__ w,   // [E, H, I] bf16
+    const int* __restrict__ offs,           // [E]
+    const int* __restrict__ counts,         // [E]
+    const int* __restrict__ order,          // [TK] sorted -> original index
+    const float* __restrict__ flat_w,       // [TK] routing weight (original)
+    int H, int I) {
+  const int nnt = H / MOE_BN;
+  const int e = blockIdx.x / nnt;
+  const int nt = blockIdx.x % nnt;
+  const int cnt = counts[e];
+  if (cnt == 0) return;
+  const int base = offs[e];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE_SIZE - 1);
+  const int wave = tid / WAVE_SIZE;
+  const int lc = lane & 15;
+  const int lg = lane >> 4;
+
+  __shared__ unsigned short Al[16 * (MOE_KC + 8)];
+
+  const unsigned short* wd_panel =
+      w + ((long)e * H + nt * MOE_BN + wave * 16) * I;
+  const int first = base;  // act rows are consecutive in sorted space
+
+  for (int m0 = 0; m0 < cnt; m0 += 16) {
+    const int nrows = min(16, cnt - m0);
+    f32x4 acc{0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < I; k0 += MOE_KC) {
+      const int kc = min(MOE_KC, I - k0);
+      __syncthreads();
+      stage_rows<false>(Al, act, I, nullptr, first + m0, nrows, k0, kc, tid);
+      __syncthreads();
+      for (int kk = 0; kk < kc; kk += 32) {
+        const u16x8 a = *reinterpret_cast<const u16x8*>(
+            Al + lc * (MOE_KC + 8) + kk + lg * 8);
+        const u16x8 b = *reinterpret_cast<const u16x8*>(
+            wd_panel + (long)lc * I + k0 + kk + lg * 8);
+        acc = moe_mfma(a, b, acc);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = lg * 4 + r;
+      if (row >= nrows) continue;
+      const int oj = order[base + m0 + row];
+      contrib[(long)oj * H + nt * MOE_BN + wave * 16 + lc] =
+          f2bf(acc[r] * flat_w[oj]);
+    }
+  }
+}
+
+}  // namespace
+
+void moe_gate_up_silu_launch(void* act, const void* x, const void* w,
+                             const int* s_tok, const int* offs,
+                             const int* counts, int E, int H, int I,
+                             int* err_unsupported, hipStream_t s) {
+  *err_unsupported = 0;
+  if (I % MOE_BN != 0 || H % 32 != 0) { *err_unsupported = 1; return; }
+  dim3 grid(E * (I / MOE_BN));
+  hipLaunchKernelGGL(moe_gate_up_silu_kernel, grid, dim3(MOE_THREADS), 0, s,
+                     (unsigned short*)act, (const unsigned short*)x,
+                     (const unsigned short*)w, s_tok, offs, counts, H, I);
+}
+
+void moe_down_scale_launch(void* contrib, const void* act, const void* w,
+                           const int* offs, const int* counts,
+                           const int* order, const float* flat_w, int E,
+                           int H, int I, int* err_unsupported, hipStream_t s) {
+  *err_unsupported = 0;
+  if (H % MOE_BN != 0 || I % 32 != 0) { *err_unsupported = 1; return; }
+  dim3 grid(E * (H / MOE_BN));
+  hipLaunchKernelGGL(moe_down_scale_kernel, grid, dim3(MOE_THREADS), 0, s,
+                     (unsigned short*)contrib, (const unsigned short*)act,
+                     (const unsigned short*)w, offs, counts, order, flat_w,
+                     H, I);
+}

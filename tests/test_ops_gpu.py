@@ -384,3 +384,76 @@ def test_flash_prefill_paged_matches_contiguous():
     ops.paged_prefill_attn(out_p, q, k_cache, v_cache, bt, starts,
                            [0] * len(lens), lens, 1.0 / math.sqrt(D))
     _close(out_p, out_c)
+
+
+@pytest.mark.parametrize("T,E,topk,H,I", [
+    (7, 16, 2, 256, 128),     # tiny, many empty experts
+    (128, 128, 8, 2048, 768), # qwen3-30b-a3b decode shape
+    (64, 8, 2, 4096, 1792),   # mixtral-like (i//tp 8 -> 1792)
+])
+def test_fused_moe_kernels(T, E, topk, H, I):
+    """Fused grouped MoE kernels vs a plain per-expert torch loop."""
+    import torch.nn.functional as F
+
+    dev = "cuda"
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) / 8
+    w_gu = torch.randn(E, 2 * I, H, dtype=torch.bfloat16, device=dev) / 16
+    w_d = torch.randn(E, H, I, dtype=torch.bfloat16, device=dev) / 16
+    logits = torch.randn(T, E, device=dev)
+    weights, experts = torch.topk(torch.softmax(logits, -1), topk, dim=-1)
+    flat_exp = experts.reshape(-1)
+    flat_tok = torch.arange(T, device=dev).repeat_interleave(topk)
+    flat_w = weights.reshape(-1).float()
+
+    order = torch.argsort(flat_exp, stable=True)
+    s_tok = flat_tok[order].to(torch.int32)
+    counts = torch.zeros(E, dtype=torch.int32, device=dev)
+    counts.scatter_add_(0, flat_exp, torch.ones_like(flat_exp, dtype=torch.int32))
+    offs = (counts.cumsum(0, dtype=torch.int32) - counts).to(torch.int32)
+    hip = ops._load_hip()
+    act = x.new_empty(T * topk, I)
+    hip.moe_gate_up_silu(act, x, w_gu, s_tok, offs, counts)
+    contrib = x.new_empty(T * topk, H)
+    hip.moe_down_scale(contrib, act, w_d, offs, counts,
+                       order.to(torch.int32), flat_w)
+    got = contrib.view(T, topk, H).sum(1)
+
+    # reference: per-expert loop in fp32-ish (bf16 GEMM via F.linear)
+    ref = torch.zeros(T, H, dtype=torch.float32, device=dev)
+    for e in range(E):
+        rows = torch.nonzero(flat_exp == e).flatten()
+        if not rows.numel():
+            continue
+        xe = x[flat_tok[rows]]
+        gu = F.linear(xe, w_gu[e]).float()
+        a = (F.silu(gu[:, :I]) * gu[:, I:]).to(torch.bfloat16)
+        he = F.linear(a, w_d[e]).float() * flat_w[rows].unsqueeze(1)
+        ref.index_add_(0, flat_tok[rows], he)
+    _close(got, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_moe_model_fused_matches_fallback():
+    """End-to-end MoEMLP: fused HIP dispatch == torch fallback dispatch."""
+    import os
+
+    from gpustack_amd.engine.config import ModelSpec
+    from gpustack_amd.models.llama import MoEMLP
+    from gpustack_amd.parallel import Communicator
+
+    spec = ModelSpec(hidden_size=512, intermediate_size=1024, num_layers=1,
+                     num_heads=8, num_kv_heads=8, head_dim=64,
+                     num_experts=32, num_experts_per_tok=4,
+                     moe_intermediate_size=256, vocab_size=1000)
+    torch.manual_seed(7)
+    m = MoEMLP(spec, 1, Communicator(), torch.bfloat16)
+    for p in m.parameters():
+        p.data.normal_(0, 0.05)
+    m = m.to("cuda")
+    x = torch.randn(33, 512, dtype=torch.bfloat16, device="cuda") / 8
+    out_fused = m(x.clone())
+    os.environ["GPUSTACK_AMD_FUSED_MOE"] = "0"
+    try:
+        out_ref = m(x.clone())
+    finally:
+        os.environ["GPUSTACK_AMD_FUSED_MOE"] = "1"
+    _close(out_fused, out_ref, atol=3e-2, rtol=3e-2)
