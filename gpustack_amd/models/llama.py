@@ -204,7 +204,11 @@ class MoEMLP(nn.Module):
         #  - decode-shaped torch fallback: ONE padded strided-batched GEMM
         #    pair (pays a per-layer counts.max() host sync)
         #  - prefill-shaped: per-expert loop GEMMs (large per-expert work)
-        if self._fused_ok(x):
+        if self._fused_ok(x) and flat_exp.numel() <= 32 * self.e:
+            # decode-shaped only: the fused kernels re-stream each expert's
+            # weight panel per 16-row m-tile, which is free when experts
+            # hold <=32 rows but ruinous at prefill occupancy (hundreds of
+            # rows/expert) — prefill keeps the per-expert hipBLASLt loop
             out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w)
             return self.comm.all_reduce(out)
         contrib = x.new_zeros(T * self.top_k, x.shape[1])
