@@ -175,11 +175,26 @@ class MoEMLP(nn.Module):
         self.down_w = nn.Parameter(
             torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
 
+    def _routing_consts(self, T: int, device):
+        """Per-(T,device) cached routing constants — rebuilt tensors per
+        layer per step showed up as ~27% glue kernels in the r2 profile."""
+        c = getattr(self, "_rc", None)
+        if c is None or c[0] != T:
+            flat_tok = torch.arange(T, device=device).repeat_interleave(self.top_k)
+            ones = torch.ones(T * self.top_k, dtype=torch.int32, device=device)
+            self._rc = (T, flat_tok, ones)
+            c = self._rc
+        return c[1], c[2]
+
     def forward(self, x, meta: ForwardMeta | None = None):
         # MoE expert MLPs do not take LoRA (rejected at adapter load);
         # attention adapters still apply upstream
         T = x.shape[0]
-        logits = F.linear(x.float(), self.router_w.float())      # [T, E]
+        rw = getattr(self, "_router_w_f32", None)
+        if rw is None or rw.device != x.device:
+            self._router_w_f32 = self.router_w.float()
+            rw = self._router_w_f32
+        logits = F.linear(x.float(), rw)                          # [T, E]
         if self.spec.norm_topk_prob:
             # softmax over all experts -> top-k -> renormalize == softmax
             # restricted to the top-k logits (Qwen3-MoE default, Mixtral)
@@ -190,8 +205,8 @@ class MoEMLP(nn.Module):
             probs = torch.softmax(logits, dim=-1)
             weights, experts = torch.topk(probs, self.top_k, dim=-1)
         flat_exp = experts.reshape(-1)                            # [T*k]
-        flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
-        flat_w = weights.reshape(-1).to(x.dtype)
+        flat_tok, ones_i32 = self._routing_consts(T, x.device)
+        flat_w32 = weights.reshape(-1)                            # f32
         # Per-assignment contributions land at UNIQUE rows of a [T*k, h]
         # buffer, then reduce over the fixed k axis — index_add_ over
         # duplicated token rows would use atomics, whose order (and thus
@@ -209,8 +224,10 @@ class MoEMLP(nn.Module):
             # weight panel per 16-row m-tile, which is free when experts
             # hold <=32 rows but ruinous at prefill occupancy (hundreds of
             # rows/expert) — prefill keeps the per-expert hipBLASLt loop
-            out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w)
+            out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w32,
+                                       ones_i32)
             return self.comm.all_reduce(out)
+        flat_w = flat_w32.to(x.dtype)
         contrib = x.new_zeros(T * self.top_k, x.shape[1])
         if self.e >= 16 and flat_exp.numel() < 32 * self.e:
             self._bmm_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
@@ -224,10 +241,10 @@ class MoEMLP(nn.Module):
 
         return (x.is_cuda and x.dtype == torch.bfloat16
                 and os.environ.get("GPUSTACK_AMD_FUSED_MOE", "1") == "1"
-                and self.i % 64 == 0 and x.shape[1] % 64 == 0
+                and self.i % 128 == 0 and x.shape[1] % 128 == 0
                 and ops.hip_available())
 
-    def _fused_dispatch(self, x, flat_exp, flat_tok, flat_w):
+    def _fused_dispatch(self, x, flat_exp, flat_tok, flat_w32, ones_i32):
         """Sync-free grouped expert GEMMs: sort assignments by expert on
         device, run the two fused kernels, reduce over the k axis. Every
         tensor shape here depends only on (T, k, E) — safe under hipGraph
@@ -239,15 +256,14 @@ class MoEMLP(nn.Module):
         s_tok = flat_tok[order].to(torch.int32)
         # scatter_add instead of bincount: bincount computes max() on host
         counts = torch.zeros(self.e, dtype=torch.int32, device=x.device)
-        counts.scatter_add_(0, flat_exp,
-                            torch.ones_like(flat_exp, dtype=torch.int32))
+        counts.scatter_add_(0, flat_exp, ones_i32)
         offs = (counts.cumsum(0, dtype=torch.int32) - counts).to(torch.int32)
         hip = ops._load_hip()
         act = x.new_empty(TK, self.i)
         hip.moe_gate_up_silu(act, x, self.gate_up_w, s_tok, offs, counts)
         contrib = x.new_empty(TK, x.shape[1])
         hip.moe_down_scale(contrib, act, self.down_w, offs, counts,
-                           order.to(torch.int32), flat_w.float())
+                           order.to(torch.int32), flat_w32)
         return contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
 
     def _loop_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w):

@@ -76,6 +76,11 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
   const unsigned short* wu_panel =
       w + ((long)e * 2 * I + I + nt * MOE_BN + wave * 16) * H;
 
+  // The kernel is weight-stream bound; a wave must keep enough loads in
+  // flight to cover HBM latency (~600 ns). Weights are consumed in
+  // K-blocks of 128 (4 MFMA k-steps), software-pipelined one block ahead:
+  // 8 b128 loads (gate+up) issue while the previous block's 8 MFMAs run.
+  const long wrow = (long)lc * H;  // this lane's weight row offset
   for (int m0 = 0; m0 < cnt; m0 += 16) {
     const int nrows = min(16, cnt - m0);
     f32x4 ag{0.f, 0.f, 0.f, 0.f}, au{0.f, 0.f, 0.f, 0.f};
@@ -84,15 +89,27 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
       __syncthreads();
       stage_rows<true>(Xl, x, H, s_tok + base + m0, 0, nrows, k0, kc, tid);
       __syncthreads();
-      for (int kk = 0; kk < kc; kk += 32) {
-        const u16x8 a = *reinterpret_cast<const u16x8*>(
-            Xl + lc * (MOE_KC + 8) + kk + lg * 8);
-        const u16x8 bg = *reinterpret_cast<const u16x8*>(
-            wg_panel + (long)lc * H + k0 + kk + lg * 8);
-        const u16x8 bu = *reinterpret_cast<const u16x8*>(
-            wu_panel + (long)lc * H + k0 + kk + lg * 8);
-        ag = moe_mfma(a, bg, ag);
-        au = moe_mfma(a, bu, au);
+      u16x8 bg[2][4], bu[2][4];
+      auto load_w = [&](int buf, int kk) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          bg[buf][j] = *reinterpret_cast<const u16x8*>(
+              wg_panel + wrow + k0 + kk + j * 32 + lg * 8);
+          bu[buf][j] = *reinterpret_cast<const u16x8*>(
+              wu_panel + wrow + k0 + kk + j * 32 + lg * 8);
+        }
+      };
+      load_w(0, 0);
+      for (int kk = 0; kk < kc; kk += 128) {
+        const int cur = (kk >> 7) & 1;
+        if (kk + 128 < kc) load_w(cur ^ 1, kk + 128);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const u16x8 a = *reinterpret_cast<const u16x8*>(
+              Xl + lc * (MOE_KC + 8) + kk + j * 32 + lg * 8);
+          ag = moe_mfma(a, bg[cur][j], ag);
+          au = moe_mfma(a, bu[cur][j], au);
+        }
       }
     }
     // D[row = lg*4 + r][col = lc]; fuse SiLU(gate) * up and write
@@ -135,6 +152,7 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
       w + ((long)e * H + nt * MOE_BN + wave * 16) * I;
   const int first = base;  // act rows are consecutive in sorted space
 
+  const long wrow = (long)lc * I;
   for (int m0 = 0; m0 < cnt; m0 += 16) {
     const int nrows = min(16, cnt - m0);
     f32x4 acc{0.f, 0.f, 0.f, 0.f};
@@ -143,12 +161,23 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
       __syncthreads();
       stage_rows<false>(Al, act, I, nullptr, first + m0, nrows, k0, kc, tid);
       __syncthreads();
-      for (int kk = 0; kk < kc; kk += 32) {
-        const u16x8 a = *reinterpret_cast<const u16x8*>(
-            Al + lc * (MOE_KC + 8) + kk + lg * 8);
-        const u16x8 b = *reinterpret_cast<const u16x8*>(
-            wd_panel + (long)lc * I + k0 + kk + lg * 8);
-        acc = moe_mfma(a, b, acc);
+      u16x8 bd[2][4];
+      auto load_w = [&](int buf, int kk) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          bd[buf][j] = *reinterpret_cast<const u16x8*>(
+              wd_panel + wrow + k0 + kk + j * 32 + lg * 8);
+      };
+      load_w(0, 0);
+      for (int kk = 0; kk < kc; kk += 128) {
+        const int cur = (kk >> 7) & 1;
+        if (kk + 128 < kc) load_w(cur ^ 1, kk + 128);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const u16x8 a = *reinterpret_cast<const u16x8*>(
+              Al + lc * (MOE_KC + 8) + kk + j * 32 + lg * 8);
+          acc = moe_mfma(a, bd[cur][j], acc);
+        }
       }
     }
 #pragma unroll
@@ -169,7 +198,7 @@ void moe_gate_up_silu_launch(void* act, const void* x, const void* w,
                              const int* counts, int E, int H, int I,
                              int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
-  if (I % MOE_BN != 0 || H % 32 != 0) { *err_unsupported = 1; return; }
+  if (I % MOE_BN != 0 || H % 128 != 0) { *err_unsupported = 1; return; }
   dim3 grid(E * (I / MOE_BN));
   hipLaunchKernelGGL(moe_gate_up_silu_kernel, grid, dim3(MOE_THREADS), 0, s,
                      (unsigned short*)act, (const unsigned short*)x,
@@ -181,7 +210,7 @@ void moe_down_scale_launch(void* contrib, const void* act, const void* w,
                            const int* order, const float* flat_w, int E,
                            int H, int I, int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
-  if (H % MOE_BN != 0 || I % 32 != 0) { *err_unsupported = 1; return; }
+  if (H % MOE_BN != 0 || I % 128 != 0) { *err_unsupported = 1; return; }
   dim3 grid(E * (H / MOE_BN));
   hipLaunchKernelGGL(moe_down_scale_kernel, grid, dim3(MOE_THREADS), 0, s,
                      (unsigned short*)contrib, (const unsigned short*)act,
