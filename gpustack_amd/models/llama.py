@@ -176,15 +176,26 @@ class MoEMLP(nn.Module):
             torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
 
     def _routing_consts(self, T: int, device):
-        """Per-(T,device) cached routing constants — rebuilt tensors per
-        layer per step showed up as ~27% glue kernels in the r2 profile."""
-        c = getattr(self, "_rc", None)
-        if c is None or c[0] != T:
+        """Per-T cached routing constants — rebuilt tensors per layer per
+        step showed up as ~27% glue kernels in the r2 profile.
+
+        Entries are kept for the module's lifetime in a dict: a hipGraph
+        captured after an eager warmup references the cached tensor by
+        ADDRESS, so evicting/overwriting an entry would leave captured
+        graphs reading freed (re-used) memory — garbage token indices and
+        out-of-bounds gathers at replay. Only decode-bucket-sized T values
+        are cached (<=1024 seqs); large prefill shapes compute fresh."""
+        cache = getattr(self, "_rc", None)
+        if cache is None:
+            cache = self._rc = {}
+        c = cache.get(T)
+        if c is None:
             flat_tok = torch.arange(T, device=device).repeat_interleave(self.top_k)
             ones = torch.ones(T * self.top_k, dtype=torch.int32, device=device)
-            self._rc = (T, flat_tok, ones)
-            c = self._rc
-        return c[1], c[2]
+            if T > 1024:
+                return flat_tok, ones
+            c = cache[T] = (flat_tok, ones)
+        return c
 
     def forward(self, x, meta: ForwardMeta | None = None):
         # MoE expert MLPs do not take LoRA (rejected at adapter load);
