@@ -331,8 +331,10 @@ class ModelRunner:
             # dims) still sync per layer -> eager for those.
             moe_graph_ok = (cfg.spec.num_experts == 0
                             or (cfg.spec.moe_intermediate_size
-                                // max(1, cfg.tp_size) % 64 == 0
-                                and cfg.spec.hidden_size % 64 == 0
+                                // max(1, cfg.tp_size) % 128 == 0
+                                and cfg.spec.hidden_size % 128 == 0
+                                and os.environ.get("GPUSTACK_AMD_FUSED_MOE",
+                                                   "1") == "1"
                                 and ops.hip_available()))
             if graphs_enabled() and self.eagle is None \
                     and moe_graph_ok and self.comm.pp_size == 1:
