@@ -2,6 +2,8 @@
 forward pass on the native ops, and samples next tokens."""
 from __future__ import annotations
 
+import os
+
 import torch
 
 from .. import ops
