@@ -246,7 +246,10 @@ class MoEMLP(nn.Module):
         # padding overcompute (~cap/avg) is far cheaper at hipBLASLt batch
         # rates. The loop remains for few-expert models (big per-expert
         # GEMMs, negligible glue), CPU, and pathological imbalance.
-        use_bmm = self.e >= 16 and x.is_cuda
+        import os as _os
+
+        use_bmm = (self.e >= 16 and x.is_cuda
+                   and _os.environ.get("GPUSTACK_AMD_MOE_BMM", "1") == "1")
         if use_bmm:
             counts = torch.zeros(self.e, dtype=torch.long, device=x.device)
             counts.scatter_add_(0, flat_exp, torch.ones_like(flat_exp))
