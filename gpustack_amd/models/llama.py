@@ -307,6 +307,15 @@ class MoEMLP(nn.Module):
 
     def _bmm_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w,
                       cap: int | None = None):
+        import os as _os
+
+        dbg = _os.environ.get("GPUSTACK_AMD_MOE_DEBUG", "0") == "1"
+
+        def _ck(tag):
+            if dbg:
+                torch.cuda.synchronize()
+                print(f"bmm[{tag}] ok", flush=True)
+
         TK = flat_exp.numel()
         order = torch.argsort(flat_exp, stable=True)
         s_exp = flat_exp[order]
@@ -316,16 +325,22 @@ class MoEMLP(nn.Module):
             cap = int(counts.max())      # one host sync per layer
         if cap == 0:
             return
+        _ck(f"route cap={cap} TK={TK}")
         offs = counts.cumsum(0) - counts
         pos = torch.arange(TK, device=x.device) - offs[s_exp]
         xpad = x.new_zeros(self.e, cap, x.shape[1])
         xpad[s_exp, pos] = x[s_tok]
+        _ck("xpad")
         gu = torch.bmm(xpad, self.gate_up_w.transpose(1, 2))   # [E, cap, 2i]
+        _ck("bmm1")
         act = torch.empty(self.e * cap, self.i, dtype=x.dtype, device=x.device)
         ops.silu_and_mul(act, gu.reshape(self.e * cap, 2 * self.i))
+        _ck("silu")
         hd = torch.bmm(act.view(self.e, cap, self.i),
                        self.down_w.transpose(1, 2))            # [E, cap, h]
+        _ck("bmm2")
         contrib[order] = hd[s_exp, pos] * flat_w[order].unsqueeze(1)
+        _ck("combine")
 
 
 class DecoderLayer(nn.Module):
