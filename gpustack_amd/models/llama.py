@@ -325,6 +325,7 @@ class MoEMLP(nn.Module):
             cap = int(counts.max())      # one host sync per layer
         if cap == 0:
             return
+        cap = (cap + 63) // 64 * 64  # GEMM-friendly M; pad rows are zero
         _ck(f"route cap={cap} TK={TK}")
         offs = counts.cumsum(0) - counts
         pos = torch.arange(TK, device=x.device) - offs[s_exp]
