@@ -248,8 +248,13 @@ class MoEMLP(nn.Module):
         # GEMMs, negligible glue), CPU, and pathological imbalance.
         import os as _os
 
+        # default OFF: hipBLASLt strided-batched GEMM faults at large odd
+        # per-expert row counts inside the engine (reproduced at
+        # [E=128, M~800, K=2048] bf16 even with 64-aligned M; standalone
+        # repros at smaller caps pass) — the per-expert loop is the safe
+        # prefill path until the library issue is understood
         use_bmm = (self.e >= 16 and x.is_cuda
-                   and _os.environ.get("GPUSTACK_AMD_MOE_BMM", "1") == "1")
+                   and _os.environ.get("GPUSTACK_AMD_MOE_BMM", "0") == "1")
         if use_bmm:
             counts = torch.zeros(self.e, dtype=torch.long, device=x.device)
             counts.scatter_add_(0, flat_exp, torch.ones_like(flat_exp))
