@@ -40,6 +40,8 @@ def parse_args():
     ap.add_argument("--draft-tokens", type=int, default=3)
     ap.add_argument("--prefix-caching", action="store_true",
                     help="enable automatic prefix caching (shared-prefix workloads)")
+    ap.add_argument("--quantize", default=None, choices=["w4"],
+                    help="runtime weight quantization (packed int4 in HBM)")
     ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"],
                     help="KV cache dtype (fp8 e4m3 halves KV bytes; compute stays bf16)")
     ap.add_argument("--device", default=None)
@@ -85,6 +87,7 @@ def main():
         max_num_seqs=max(args.concurrency, 8),
         seed=0,
         kv_cache_dtype=args.kv_dtype,
+        quantize_runtime=args.quantize,
         enable_prefix_caching=args.prefix_caching,
         speculative=({"method": args.speculative,
                       "num_draft_tokens": args.draft_tokens}
@@ -239,7 +242,8 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if comm is not None else "weak",
             "vs_baseline": None,
-            "dtype": cfg.dtype if use_cuda else "float32-cpu-smoke",
+            "dtype": (f"{cfg.dtype}+w4-weights" if args.quantize == "w4"
+                      else cfg.dtype) if use_cuda else "float32-cpu-smoke",
             "data": "synthetic random-token prompts, random-init weights",
             "config": {
                 "model": model,

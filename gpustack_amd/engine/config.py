@@ -210,6 +210,11 @@ class EngineConfig:
     # speculative decoding (reference speculative_config schema):
     # {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3, "ngram_min": 1}
     speculative: dict | None = None
+    # W4 runtime quantization ("w4"): the big serving weights are packed
+    # int4 in HBM after load and decode GEMMs dequantize in-register
+    # (ops/csrc/w4_gemm.hip) — 70B-class models stay resident packed
+    # instead of inflating to bf16 (reference: vLLM --quantization)
+    quantize_runtime: str | None = None
     # LoRA adapters merged into the weights at load (reference lora_list)
     lora_dirs: list[str] = field(default_factory=list)
     # GGUF checkpoint execution: dequantized to bf16 at load (utils/gguf.py)

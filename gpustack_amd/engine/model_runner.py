@@ -239,6 +239,14 @@ class ModelRunner:
                 import logging
 
                 logging.getLogger(__name__).info("merged LoRA %s (%d tensors)", d, n)
+        if getattr(cfg, "quantize_runtime", None) == "w4":
+            from ..models.weights import convert_to_w4_runtime
+
+            n = convert_to_w4_runtime(self.model, cfg)
+            import logging
+
+            logging.getLogger(__name__).info(
+                "W4 runtime: %d weight tensors packed int4", n)
         self.sampler = Sampler(self.device, cfg.spec.eos_token_id)
         from ..models.lora import LoraBank
 
@@ -585,9 +593,11 @@ class ModelRunner:
             # target hidden states that condition the next draft window
             tokens, meta = self._meta(batch)
             if batch.is_prefill:
+                from ..models.llama import qlinear
+
                 hidden_all = self.model(tokens, meta, self.kv, return_hidden=True)
-                logits = torch.nn.functional.linear(
-                    hidden_all[meta.logits_indices], self.model.lm_head)
+                logits = qlinear(hidden_all[meta.logits_indices],
+                                 self.model.lm_head, self.model.lm_head_pack)
                 self.last_hidden = hidden_all
             else:
                 logits, self.last_hidden = self.model(tokens, meta, self.kv,

@@ -51,6 +51,38 @@ class ForwardMeta:
     lora: object | None = None
 
 
+class W4Pack:
+    """Runtime-packed int4 weight (models/quantized.py pack_w4_runtime):
+    qw u8 [N, K/2] frag-ordered, sc/zs bf16 [N, K/128]."""
+
+    __slots__ = ("qw", "sc", "zs", "shape")
+
+    def __init__(self, qw, sc, zs):
+        self.qw, self.sc, self.zs = qw, sc, zs
+        self.shape = (qw.shape[0], qw.shape[1] * 2)
+
+
+def qlinear(x, w, pack: "W4Pack | None", bias=None):
+    """F.linear with an optional W4 runtime pack. Decode-shaped M runs the
+    in-register dequant MFMA kernel (weight traffic halved vs bf16);
+    prefill-shaped M dequantizes to a transient bf16 tensor and uses
+    hipBLASLt (the 256-row kernel would re-stream weights per m-tile)."""
+    if pack is None:
+        return F.linear(x, w, bias)
+    if x.is_cuda and x.shape[0] <= 1024:
+        out = torch.empty(x.shape[0], pack.shape[0], dtype=x.dtype,
+                          device=x.device)
+        ops._load_hip().w4_gemm(out, x.contiguous(), pack.qw, pack.sc,
+                                pack.zs)
+        if bias is not None:
+            out += bias
+        return out
+    from .quantized import dequant_w4_runtime
+
+    wt = dequant_w4_runtime(pack.qw, pack.sc, pack.zs).to(x.dtype)
+    return F.linear(x, wt, bias)
+
+
 class Attention(nn.Module):
     def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
         super().__init__()
@@ -68,6 +100,8 @@ class Attention(nn.Module):
             if spec.attention_bias else None
         )
         self.o_w = nn.Parameter(torch.empty(h, self.hq * self.d, dtype=dtype), requires_grad=False)
+        self.qkv_pack: W4Pack | None = None   # W4 runtime (qlinear)
+        self.o_pack: W4Pack | None = None
         self.layer_idx = 0  # set by LlamaForCausalLM
         nq, nk = self.hq * self.d, self.hkv * self.d
         self._qkv_projs = [("q_proj", 0, nq), ("k_proj", nq, nk),
@@ -79,7 +113,7 @@ class Attention(nn.Module):
 
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
         T = x.shape[0]
-        qkv = F.linear(x, self.qkv_w, self.qkv_b)
+        qkv = qlinear(x, self.qkv_w, self.qkv_pack, self.qkv_b)
         if meta.lora is not None:
             qkv = qkv.contiguous()
             meta.lora.apply(self.layer_idx, x, qkv, self._qkv_projs)
@@ -118,7 +152,7 @@ class Attention(nn.Module):
             ops.paged_attn_decode(
                 out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale
             )
-        o = F.linear(out.view(T, -1), self.o_w)
+        o = qlinear(out.view(T, -1), self.o_w, self.o_pack)
         if meta.lora is not None:
             meta.lora.apply(self.layer_idx, out.view(T, -1), o, self._o_projs)
         return self.comm.all_reduce(o)
@@ -132,17 +166,21 @@ class MLP(nn.Module):
         self.i = spec.intermediate_size // tp_size
         self.gate_up_w = nn.Parameter(torch.empty(2 * self.i, h, dtype=dtype), requires_grad=False)
         self.down_w = nn.Parameter(torch.empty(h, self.i, dtype=dtype), requires_grad=False)
+        self.gate_up_pack: W4Pack | None = None  # W4 runtime (qlinear)
+        self.down_pack: W4Pack | None = None
         self.layer_idx = 0  # set by LlamaForCausalLM
         self._gu_projs = [("gate_proj", 0, self.i), ("up_proj", self.i, self.i)]
         self._down_projs = [("down_proj", 0, h)]
 
     def forward(self, x, meta: ForwardMeta | None = None):
-        gu = F.linear(x, self.gate_up_w)
+        gu = qlinear(x, self.gate_up_w, self.gate_up_pack)
         if meta is not None and meta.lora is not None:
             meta.lora.apply(self.layer_idx, x, gu, self._gu_projs)
         act = torch.empty(x.shape[0], self.i, dtype=x.dtype, device=x.device)
         ops.silu_and_mul(act, gu)
-        down = ops.linear_auto(act, self.down_w)
+        down = (qlinear(act, self.down_w, self.down_pack)
+                if self.down_pack is not None
+                else ops.linear_auto(act, self.down_w))
         if meta is not None and meta.lora is not None:
             meta.lora.apply(self.layer_idx, act, down, self._down_projs)
         return self.comm.all_reduce(down)
@@ -422,6 +460,7 @@ class LlamaForCausalLM(nn.Module):
         else:
             self.final_norm = None
             self.lm_head = None
+        self.lm_head_pack: W4Pack | None = None  # W4 runtime (qlinear)
         cache = ops.build_cos_sin_cache(
             spec.head_dim, spec.head_dim, cfg.max_model_len,
             base=spec.rope_theta, scaling=spec.rope_scaling,
@@ -453,7 +492,7 @@ class LlamaForCausalLM(nn.Module):
         if return_hidden:
             return x  # all rows, post final norm (embedding serving)
         hidden = x[meta.logits_indices]
-        logits = F.linear(hidden, self.lm_head)
+        logits = qlinear(hidden, self.lm_head, self.lm_head_pack)
         if return_both:  # draft-model speculative needs the features too
             return logits, hidden
         return logits

@@ -112,3 +112,63 @@ def maybe_dequant(tensors: dict, name: str, qc: dict | None):
         return dequant_gptq(qw, qz, sc, tensors.get(base + ".g_idx"),
                             qc["bits"])
     return dequant_awq(qw, qz, sc, qc["bits"])
+
+
+# ---------------------------------------------------------------------------
+# W4 RUNTIME format (round 2): weights stay packed int4 in HBM and the
+# ops/csrc/w4_gemm.hip kernel dequantizes in-register. Canonical layout:
+#   qw u8 [N, K/2] in FRAGMENT ORDER (see w4_gemm.hip header),
+#   sc/zs bf16 [N, K/128] (one group per 128-k block; zs = zero * scale).
+# ---------------------------------------------------------------------------
+
+W4_BLOCK = 128
+
+
+def _frag_perm(K: int, device=None) -> torch.Tensor:
+    """k-index permutation: position p in packed order -> source k.
+
+    Within each 128-k block, packed order is [lg(4)][chunk(4)][j(8)] while
+    the source order is [chunk][lg][j] — this puts the 32 values of lane
+    lg's four MFMA B-fragments into one contiguous 16-byte load."""
+    arr = torch.arange(K, device=device).view(-1, 4, 4, 8)  # [blk, c, lg, j]
+    return arr.permute(0, 2, 1, 3).reshape(-1)
+
+
+def pack_w4_runtime(q: torch.Tensor, scales: torch.Tensor,
+                    zeros: torch.Tensor, group_size: int):
+    """q: [N, K] uint4 values (0..15); scales/zeros: [N, K/group_size].
+
+    -> (qw u8 [N, K/2], sc bf16 [N, K/128], zs bf16 [N, K/128]) or None
+    if the shape/grouping cannot run packed (caller falls back to
+    dequant-at-load)."""
+    N, K = q.shape
+    if K % W4_BLOCK != 0:
+        return None
+    if group_size % W4_BLOCK != 0:
+        return None  # sub-block groups: scale changes inside a k-block
+    rep = group_size // W4_BLOCK
+    sc_f = scales.float().repeat_interleave(rep, dim=1)   # [N, K/128]
+    zr_f = zeros.float().repeat_interleave(rep, dim=1)
+    perm = _frag_perm(K, q.device)
+    qp = q.index_select(1, perm).to(torch.uint8)
+    qw = (qp[:, 0::2] | (qp[:, 1::2] << 4)).contiguous()
+    sc = sc_f.to(torch.bfloat16).contiguous()
+    zs = (zr_f * sc_f).to(torch.bfloat16).contiguous()
+    return qw, sc, zs
+
+
+def dequant_w4_runtime(qw: torch.Tensor, sc: torch.Tensor,
+                       zs: torch.Tensor) -> torch.Tensor:
+    """Packed runtime format -> bf16 [N, K] (prefill transient / CPU ref)."""
+    N = qw.shape[0]
+    K = qw.shape[1] * 2
+    lo = (qw & 0xF)
+    hi = (qw >> 4)
+    qp = torch.stack([lo, hi], dim=2).reshape(N, K)  # packed order
+    perm = _frag_perm(K, qw.device)
+    inv = torch.empty_like(perm)
+    inv[perm] = torch.arange(K, device=qw.device)
+    q = qp.index_select(1, inv).float()
+    s = sc.float().repeat_interleave(W4_BLOCK, dim=1)
+    z = zs.float().repeat_interleave(W4_BLOCK, dim=1)
+    return (q * s - z).to(torch.bfloat16)

@@ -378,3 +378,66 @@ def load_gguf(model, cfg: EngineConfig, gguf_path: str | Path) -> None:
         layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.copy_(get(p + "attn_norm.weight"))
         layer.post_attn_norm.copy_(get(p + "ffn_norm.weight"))
+
+
+def convert_to_w4_runtime(model, cfg: EngineConfig) -> int:
+    """Quantize the big serving weights to the packed W4 runtime format
+    (models/quantized.py) and free their bf16 parameters: weights stay
+    resident int4 in HBM and decode GEMMs run the in-register dequant
+    kernel (ops/csrc/w4_gemm.hip). Returns the number of packed tensors.
+
+    Uniform path for every weight source: random-init, safetensors, and
+    dequantized GPTQ/AWQ/GGUF checkpoints all quantize post-load
+    (asymmetric per-128 groups). Ineligible shapes (N%64 / K%128) keep
+    bf16 — mixed execution is fine. Reference capability: vLLM
+    --quantization gptq/awq serving, re-designed as a
+    CDNA4 fragment-ordered format.
+    """
+    from .llama import W4Pack
+    from .quantized import pack_w4_runtime
+
+    def quantize(w: torch.Tensor):
+        N, K = w.shape
+        if N % 64 or K % 128:
+            return None
+        wf = w.float().view(N, K // 128, 128)
+        mn = wf.amin(-1)
+        mx = wf.amax(-1)
+        scale = ((mx - mn) / 15).clamp_min(1e-8)
+        zero = (-mn / scale).round().clamp(0, 15)
+        q = ((wf / scale.unsqueeze(-1)) + zero.unsqueeze(-1)) \
+            .round().clamp(0, 15).view(N, K).to(torch.uint8)
+        return pack_w4_runtime(q, scale, zero, 128)
+
+    n_packed = 0
+
+    def pack_site(mod, wname: str, pname: str) -> None:
+        nonlocal n_packed
+        w = getattr(mod, wname)
+        if w is None or w.numel() == 0:
+            return
+        pk = quantize(w.data)
+        if pk is None:
+            return
+        setattr(mod, pname, W4Pack(*pk))
+        w.data = torch.empty(0, dtype=w.dtype, device=w.device)
+        n_packed += 1
+
+    for layer in model.layers:
+        pack_site(layer.attn, "qkv_w", "qkv_pack")
+        pack_site(layer.attn, "o_w", "o_pack")
+        if model.spec.num_experts == 0:
+            pack_site(layer.mlp, "gate_up_w", "gate_up_pack")
+            pack_site(layer.mlp, "down_w", "down_pack")
+    if model.lm_head is not None and not model.spec.tie_word_embeddings:
+        w = model.lm_head
+        pk = quantize(w.data)
+        if pk is not None:
+            from .llama import W4Pack as _P
+
+            model.lm_head_pack = _P(*pk)
+            w.data = torch.empty(0, dtype=w.dtype, device=w.device)
+            n_packed += 1
+    if model.device.type == "cuda":
+        torch.cuda.empty_cache()
+    return n_packed

@@ -21,6 +21,7 @@ void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int
 void gemm_lab_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 void moe_gate_up_silu_launch(void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int*, hipStream_t);
 void moe_down_scale_launch(void*, const void*, const void*, const int*, const int*, const int*, const float*, int, int, int, int*, hipStream_t);
+void w4_gemm_launch(void*, const void*, const void*, const void*, const void*, int, int, int, int*, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -285,6 +286,22 @@ void moe_down_scale(at::Tensor contrib, at::Tensor act, at::Tensor w,
   HIP_CHECK_LAST();
 }
 
+void w4_gemm(at::Tensor out, at::Tensor x, at::Tensor qw, at::Tensor sc,
+             at::Tensor zs) {
+  check_bf16(out, "out"); check_bf16(x, "x");
+  check_bf16(sc, "sc"); check_bf16(zs, "zs");
+  TORCH_CHECK(qw.scalar_type() == at::kByte && qw.is_contiguous());
+  const int M = x.size(0), K = x.size(1), N = qw.size(0);
+  TORCH_CHECK(qw.size(1) == K / 2 && out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(sc.size(0) == N && sc.size(1) == K / 128);
+  int err = 0;
+  w4_gemm_launch(out.data_ptr(), x.data_ptr(), qw.data_ptr(), sc.data_ptr(),
+                 zs.data_ptr(), M, N, K, &err, cur_stream(x));
+  TORCH_CHECK(!err, "w4_gemm: unsupported shape M=", M, " N=", N, " K=", K,
+              " (need N%64==0, K%128==0)");
+  HIP_CHECK_LAST();
+}
+
 void gemm_lab(at::Tensor out, at::Tensor x, at::Tensor w, long mode) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   const int M = x.size(0), K = x.size(1), N = w.size(0);
@@ -328,4 +345,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped MoE gate/up GEMM + SiLU (sorted assignments, sync-free)");
   m.def("moe_down_scale", &moe_down_scale,
         "grouped MoE down GEMM + routing-weight scale/scatter");
+  m.def("w4_gemm", &w4_gemm,
+        "W4A16 GEMM: packed-int4 weights dequantized in-register");
 }

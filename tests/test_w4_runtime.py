@@ -1,0 +1,55 @@
+"""W4 runtime quantization: packed int4 weights + in-register dequant GEMM.
+
+CPU tier: pack/dequant roundtrip + engine e2e on the torch fallback.
+GPU tier (test_ops_gpu.py): kernel vs dequant reference.
+"""
+import pytest
+import torch
+
+from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+from gpustack_amd.models.quantized import (dequant_w4_runtime,
+                                           pack_w4_runtime)
+
+
+def test_pack_roundtrip():
+    torch.manual_seed(0)
+    N, K, g = 64, 256, 128
+    q = torch.randint(0, 16, (N, K))
+    sc = torch.rand(N, K // g) * 0.1 + 0.01
+    zr = torch.randint(0, 16, (N, K // g)).float()
+    qw, s, zs = pack_w4_runtime(q, sc, zr, g)
+    assert qw.shape == (N, K // 2) and qw.dtype == torch.uint8
+    w = dequant_w4_runtime(qw, s, zs).float()
+    ref = (q.float() - zr.repeat_interleave(g, 1)) * sc.repeat_interleave(g, 1)
+    assert (w - ref).abs().max().item() < 2e-2
+
+
+def test_pack_group_multiple_and_reject():
+    q = torch.randint(0, 16, (64, 256))
+    assert pack_w4_runtime(q, torch.rand(64, 1), torch.zeros(64, 1), 256) is not None
+    assert pack_w4_runtime(q, torch.rand(64, 4), torch.zeros(64, 4), 64) is None
+    assert pack_w4_runtime(torch.randint(0, 16, (64, 100)),
+                           torch.rand(64, 1), torch.zeros(64, 1), 100) is None
+
+
+def test_engine_w4_runtime_cpu():
+    """Engine serves with packed weights on the CPU fallback path; packs
+    exist and the bf16 parameters are freed."""
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       quantize_runtime="w4")
+    eng = LLMEngine(cfg)
+    model = eng.runner.model
+    packed = sum(1 for layer in model.layers
+                 if layer.attn.qkv_pack is not None)
+    assert packed > 0
+    for layer in model.layers:
+        if layer.attn.qkv_pack is not None:
+            assert layer.attn.qkv_w.numel() == 0
+    out = eng.generate([[1, 2, 3, 4]],
+                       SamplingParams(max_tokens=6, ignore_eos=True))[0]
+    assert len(out) == 6
+    # determinism
+    eng2 = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                  kv_cache_blocks=64, quantize_runtime="w4"))
+    assert eng2.generate([[1, 2, 3, 4]],
+                         SamplingParams(max_tokens=6, ignore_eos=True))[0] == out
