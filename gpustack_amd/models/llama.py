@@ -69,14 +69,18 @@ def qlinear(x, w, pack: "W4Pack | None", bias=None):
     hipBLASLt (the 256-row kernel would re-stream weights per m-tile)."""
     if pack is None:
         return F.linear(x, w, bias)
-    if x.is_cuda and x.shape[0] <= 1024:
-        out = torch.empty(x.shape[0], pack.shape[0], dtype=x.dtype,
-                          device=x.device)
-        ops._load_hip().w4_gemm(out, x.contiguous(), pack.qw, pack.sc,
-                                pack.zs)
-        if bias is not None:
-            out += bias
-        return out
+    if x.is_cuda:
+        hip = ops._load_hip()
+        if x.shape[0] <= 1024:
+            out = torch.empty(x.shape[0], pack.shape[0], dtype=x.dtype,
+                              device=x.device)
+            hip.w4_gemm(out, x.contiguous(), pack.qw, pack.sc, pack.zs)
+            if bias is not None:
+                out += bias
+            return out
+        wt = torch.empty(pack.shape, dtype=torch.bfloat16, device=x.device)
+        hip.w4_dequant(wt, pack.qw, pack.sc, pack.zs)
+        return F.linear(x, wt, bias)
     from .quantized import dequant_w4_runtime
 
     wt = dequant_w4_runtime(pack.qw, pack.sc, pack.zs).to(x.dtype)

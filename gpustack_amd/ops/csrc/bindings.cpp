@@ -22,6 +22,7 @@ void gemm_lab_launch(void*, const void*, const void*, int, int, int, int, int*, 
 void moe_gate_up_silu_launch(void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int*, hipStream_t);
 void moe_down_scale_launch(void*, const void*, const void*, const int*, const int*, const int*, const float*, int, int, int, int*, hipStream_t);
 void w4_gemm_launch(void*, const void*, const void*, const void*, const void*, int, int, int, int*, hipStream_t);
+void w4_dequant_launch(void*, const void*, const void*, const void*, long, int, int*, hipStream_t);
 
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
@@ -302,6 +303,19 @@ void w4_gemm(at::Tensor out, at::Tensor x, at::Tensor qw, at::Tensor sc,
   HIP_CHECK_LAST();
 }
 
+void w4_dequant(at::Tensor out, at::Tensor qw, at::Tensor sc, at::Tensor zs) {
+  check_bf16(out, "out"); check_bf16(sc, "sc"); check_bf16(zs, "zs");
+  TORCH_CHECK(qw.scalar_type() == at::kByte && qw.is_contiguous());
+  const long N = qw.size(0);
+  const int K = (int)(qw.size(1) * 2);
+  TORCH_CHECK(out.size(0) == N && out.size(1) == K);
+  int err = 0;
+  w4_dequant_launch(out.data_ptr(), qw.data_ptr(), sc.data_ptr(),
+                    zs.data_ptr(), N, K, &err, cur_stream(out));
+  TORCH_CHECK(!err, "w4_dequant: K must be a multiple of 128");
+  HIP_CHECK_LAST();
+}
+
 void gemm_lab(at::Tensor out, at::Tensor x, at::Tensor w, long mode) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   const int M = x.size(0), K = x.size(1), N = w.size(0);
@@ -347,4 +361,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped MoE down GEMM + routing-weight scale/scatter");
   m.def("w4_gemm", &w4_gemm,
         "W4A16 GEMM: packed-int4 weights dequantized in-register");
+  m.def("w4_dequant", &w4_dequant,
+        "packed-int4 -> bf16 weight expansion (prefill transient)");
 }

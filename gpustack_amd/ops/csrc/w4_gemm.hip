@@ -123,7 +123,56 @@ __global__ __launch_bounds__(W4_THREADS) void w4_gemm_kernel(
   }
 }
 
+// Transient dequant for prefill-shaped GEMMs (bandwidth-bound): packed
+// frag-order nibbles -> bf16 [N, K] in canonical k order, so hipBLASLt
+// consumes the result directly. Each thread expands one 16-byte packed
+// chunk (lane-lg slice of a 128-k block) into its four 8-value octets at
+// k = blk*128 + c*32 + lg*8.
+__global__ __launch_bounds__(256) void w4_dequant_kernel(
+    unsigned short* __restrict__ out,      // [N, K] bf16
+    const unsigned char* __restrict__ qw,  // [N, K/2] frag-ordered
+    const unsigned short* __restrict__ sc, // [N, K/128]
+    const unsigned short* __restrict__ zs, // [N, K/128]
+    long N, int K) {
+  const long chunks = N * (K / 32);        // 16B packed chunks
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < chunks;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long row = idx / (K / 32);
+    const int ck = idx % (K / 32);         // chunk within row
+    const int blk = ck / 4;
+    const int lg = ck % 4;
+    const uint4 w = *reinterpret_cast<const uint4*>(
+        qw + row * (K / 2) + (long)blk * 64 + lg * 16);
+    const float s = bf2f(sc[row * (K / 128) + blk]);
+    const float z = bf2f(zs[row * (K / 128) + blk]);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const unsigned int d = (&w.x)[c];
+      unsigned short o[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf((float)((d >> (4 * j)) & 0xF) * s - z);
+      *reinterpret_cast<u16x8*>(out + row * K + blk * 128 + c * 32 + lg * 8) =
+          *reinterpret_cast<u16x8*>(o);
+    }
+  }
+}
+
 }  // namespace
+
+void w4_dequant_launch(void* out, const void* qw, const void* sc,
+                       const void* zs, long N, int K, int* err_unsupported,
+                       hipStream_t s) {
+  *err_unsupported = 0;
+  if (K % W4_KC != 0) { *err_unsupported = 1; return; }
+  const long chunks = N * (K / 32);
+  long grid = (chunks + 255) / 256;
+  if (grid > 16384) grid = 16384;
+  hipLaunchKernelGGL(w4_dequant_kernel, dim3((unsigned)grid), dim3(256), 0, s,
+                     (unsigned short*)out, (const unsigned char*)qw,
+                     (const unsigned short*)sc, (const unsigned short*)zs, N,
+                     K);
+}
 
 void w4_gemm_launch(void* out, const void* x, const void* qw, const void* sc,
                     const void* zs, int M, int N, int K,

@@ -495,3 +495,20 @@ def test_engine_w4_runtime_gpu(model):
     p = SamplingParams(max_tokens=8, ignore_eos=True)
     out = eng.generate([[1, 2, 3, 4, 5] * 10], p)[0]
     assert len(out) == 8
+
+
+def test_w4_dequant_kernel():
+    from gpustack_amd.models.quantized import (dequant_w4_runtime,
+                                               pack_w4_runtime)
+
+    torch.manual_seed(5)
+    dev = "cuda"
+    N, K = 192, 640
+    q = torch.randint(0, 16, (N, K), device=dev)
+    sc = torch.rand(N, K // 128, device=dev) * 0.05 + 0.01
+    zr = torch.randint(0, 16, (N, K // 128), device=dev).float()
+    qw, s, zs = pack_w4_runtime(q, sc, zr, 128)
+    ref = dequant_w4_runtime(qw, s, zs)
+    out = torch.empty(N, K, dtype=torch.bfloat16, device=dev)
+    ops._load_hip().w4_dequant(out, qw, s, zs)
+    assert torch.equal(out, ref) or (out.float() - ref.float()).abs().max() < 1e-2
