@@ -60,8 +60,6 @@ __global__ __launch_bounds__(W4_THREADS) void w4_gemm_kernel(
   const int lg = lane >> 4;
   const int col = n0 + wave * 16 + lc;  // this lane's output column
 
-  __shared__ unsigned short Xl[W4_BM * (W4_KC + W4_PAD)];
-
   f32x4 acc[W4_BM / 16];
 #pragma unroll
   for (int mf = 0; mf < W4_BM / 16; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -70,21 +68,19 @@ __global__ __launch_bounds__(W4_THREADS) void w4_gemm_kernel(
   const long qrow = (long)col * (K / 2);
   const long srow = (long)col * nblk;
 
+  // A-fragments read straight from L2 (decode-shaped X is a few MB and
+  // every N-workgroup re-reads the same panel): no LDS staging, no
+  // barriers, occupancy limited only by registers. Row base offsets are
+  // per-mf invariant; rows past M clamp to M-1 (stores are guarded).
+  long row_off[W4_BM / 16];
+#pragma unroll
+  for (int mf = 0; mf < W4_BM / 16; ++mf) {
+    int row = m0 + mf * 16 + lc;
+    if (row >= M) row = M - 1;
+    row_off[mf] = (long)row * K;
+  }
+
   for (int kb = 0; kb < nblk; ++kb) {
-    __syncthreads();
-    // stage X[m0..m0+256)[kb*128..+128) -> LDS (zero-fill past M)
-    {
-      const int units = W4_BM * (W4_KC / 8);  // u16x8 units
-      for (int u = tid; u < units; u += W4_THREADS) {
-        const int row = u / (W4_KC / 8);
-        const int kk = (u % (W4_KC / 8)) * 8;
-        u16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (m0 + row < M)
-          val = *reinterpret_cast<const u16x8*>(
-              x + (long)(m0 + row) * K + kb * W4_KC + kk);
-        *reinterpret_cast<u16x8*>(Xl + row * (W4_KC + W4_PAD) + kk) = val;
-      }
-    }
     // lane's packed weights for this block: 16B = 4 dwords = 32 nibbles
     const uint4 w = *reinterpret_cast<const uint4*>(
         qw + qrow + (long)kb * 64 + lg * 16);
@@ -100,13 +96,13 @@ __global__ __launch_bounds__(W4_THREADS) void w4_gemm_kernel(
         bfr[c][j] = f2bf(q4 * s - z);
       }
     }
-    __syncthreads();
+    const int k0 = kb * W4_KC;
 #pragma unroll
     for (int mf = 0; mf < W4_BM / 16; ++mf) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         const u16x8 a = *reinterpret_cast<const u16x8*>(
-            Xl + (mf * 16 + lc) * (W4_KC + W4_PAD) + c * 32 + lg * 8);
+            x + row_off[mf] + k0 + c * 32 + lg * 8);
         acc[mf] = w4_mfma(a, bfr[c], acc[mf]);
       }
     }
