@@ -63,15 +63,23 @@ class W4Pack:
 
 
 def qlinear(x, w, pack: "W4Pack | None", bias=None):
-    """F.linear with an optional W4 runtime pack. Decode-shaped M runs the
-    in-register dequant MFMA kernel (weight traffic halved vs bf16);
-    prefill-shaped M dequantizes to a transient bf16 tensor and uses
-    hipBLASLt (the 256-row kernel would re-stream weights per m-tile)."""
+    """F.linear with an optional W4 runtime pack.
+
+    Default GPU path: the w4_dequant HIP kernel expands the packed weight
+    into a transient bf16 tensor (allocator-cached) and hipBLASLt runs the
+    GEMM at full MFMA rate — weights stay RESIDENT packed (the capacity
+    goal) at a measured ~2.25x packed-read traffic per use. The in-register
+    dequant MFMA kernel (w4_gemm) is kept behind GPUSTACK_AMD_W4_KERNEL=1:
+    both its LDS-staged and direct-L2 variants measured slower than
+    dequant+hipBLASLt at serving shapes so far (profiles/r04)."""
     if pack is None:
         return F.linear(x, w, bias)
     if x.is_cuda:
+        import os as _os
+
         hip = ops._load_hip()
-        if x.shape[0] <= 1024:
+        if (x.shape[0] <= 1024
+                and _os.environ.get("GPUSTACK_AMD_W4_KERNEL", "0") == "1"):
             out = torch.empty(x.shape[0], pack.shape[0], dtype=x.dtype,
                               device=x.device)
             hip.w4_gemm(out, x.contiguous(), pack.qw, pack.sc, pack.zs)

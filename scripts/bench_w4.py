@@ -1,6 +1,7 @@
 import sys, time
 from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+import gpustack_amd.engine  # noqa: F401  (import-order: break package cycle)
 import torch
 import torch.nn.functional as F
 from gpustack_amd import ops
