@@ -7,6 +7,10 @@ REPO_ROOT = Path(__file__).resolve().parent.parent
 if str(REPO_ROOT) not in sys.path:
     sys.path.insert(0, str(REPO_ROOT))
 
+import gpustack_amd.engine  # noqa: F401,E402  (import order: engine before
+# models — a test importing gpustack_amd.models first would hit the
+# llama<->model_runner package cycle)
+
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs an MI355X (run via gpurun)")
