@@ -215,6 +215,11 @@ class EngineConfig:
     # (ops/csrc/w4_gemm.hip) — 70B-class models stay resident packed
     # instead of inflating to bf16 (reference: vLLM --quantization)
     quantize_runtime: str | None = None
+    # CPU weight offload (reference: vLLM --cpu-offload-gb, the GGUF
+    # partial-offload placement): ~this many GiB of trailing layers'
+    # weights live pinned in host DRAM and stream to a double-buffered
+    # device staging area one layer ahead of use (engine/offload.py)
+    cpu_offload_gb: float = 0.0
     # LoRA adapters merged into the weights at load (reference lora_list)
     lora_dirs: list[str] = field(default_factory=list)
     # GGUF checkpoint execution: dequantized to bf16 at load (utils/gguf.py)

@@ -247,6 +247,10 @@ class ModelRunner:
 
             logging.getLogger(__name__).info(
                 "W4 runtime: %d weight tensors packed int4", n)
+        if getattr(cfg, "cpu_offload_gb", 0) > 0:
+            from .offload import setup_cpu_offload
+
+            setup_cpu_offload(self.model, cfg.cpu_offload_gb)
         self.sampler = Sampler(self.device, cfg.spec.eos_token_id)
         from ..models.lora import LoraBank
 
@@ -347,7 +351,8 @@ class ModelRunner:
                                                    "1") == "1"
                                 and ops.hip_available()))
             if graphs_enabled() and self.eagle is None \
-                    and moe_graph_ok and self.comm.pp_size == 1:
+                    and moe_graph_ok and self.comm.pp_size == 1 \
+                    and self.model.offload is None:
                 self.graph_runner = DecodeGraphRunner(self)
                 self.graph_runner.capture()
         return self.kv

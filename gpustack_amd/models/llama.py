@@ -473,6 +473,7 @@ class LlamaForCausalLM(nn.Module):
             self.final_norm = None
             self.lm_head = None
         self.lm_head_pack: W4Pack | None = None  # W4 runtime (qlinear)
+        self.offload = None  # CpuOffload streamer (engine/offload.py)
         cache = ops.build_cos_sin_cache(
             spec.head_dim, spec.head_dim, cfg.max_model_len,
             base=spec.rope_theta, scaling=spec.rope_scaling,
@@ -494,7 +495,12 @@ class LlamaForCausalLM(nn.Module):
         else:
             x = F.embedding(token_ids, self.embed)
         residual = None
+        off = self.offload
+        if off is not None:
+            off.begin()
         for i, layer in enumerate(self.layers):
+            if off is not None and i >= off.first:
+                off.bind(i)
             x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
         if self.comm.pp_size > 1 and not self.comm.is_last_stage:
             s = (x.float() + residual.float()).to(self.dtype)

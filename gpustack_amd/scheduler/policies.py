@@ -32,6 +32,11 @@ class Candidate:
     # multi-worker TP (reference: subordinate workers,
     # vllm_resource_fit_selector.py:800-867): [(worker_dict, gpu_indexes)]
     subordinates: list = field(default_factory=list)
+    # CPU weight offload (reference: GGUF partial offload / offload_layers
+    # in the claim, schemas/models.py:623-630): host GiB streamed per step
+    ram_claim: int = 2 << 30
+    offload_gb: float = 0.0
+    offload_layers: int = 0
 
 
 def model_spec_for(model: Model | dict) -> ModelSpec | None:
@@ -49,16 +54,45 @@ def model_spec_for(model: Model | dict) -> ModelSpec | None:
         return None
 
 
+def _weight_bytes(model: dict, spec: ModelSpec, tp: int) -> int:
+    """Per-shard resident weight bytes, accounting for runtime W4 packing
+    (backend_parameters quantize_runtime=w4 keeps weights int4: ~0.28x of
+    bf16 incl. scales) and the bf16 default."""
+    bp = model.get("backend_parameters") or {}
+    w = spec.weight_bytes() // tp
+    if bp.get("quantize_runtime") == "w4":
+        w = int(w * 0.28)
+    return w
+
+
 def estimate_vram_claim(model: Model | dict, spec: ModelSpec | None, tp: int) -> int:
-    """Per-GPU VRAM claim (bytes) for one replica shard."""
-    gmu = (model["gpu_memory_utilization"] if isinstance(model, dict)
-           else model.gpu_memory_utilization) or 0.9
+    """Minimum per-GPU VRAM need (bytes) for one replica shard: exact
+    weights + framework/activation overhead + KV-pool floor. The FULL
+    claim recorded on placement additionally includes the KV pool the
+    engine will actually take (gmu x remaining free) — see
+    claim_for_allocatable (reference memory model:
+    policies/utils.py:384-470 + vllm_resource_fit_selector.py:166-206)."""
+    if isinstance(model, Model):
+        model = model.to_dict()
     if spec is None:
         return 16 << 30
-    weights = int(spec.weight_bytes() * WEIGHT_FUDGE) // tp + FRAMEWORK_OVERHEAD
-    # the engine grabs gmu x free for KV; claim the weights + a KV floor and
-    # let the scorer prefer roomier devices
+    weights = int(_weight_bytes(model, spec, tp) * WEIGHT_FUDGE) + FRAMEWORK_OVERHEAD
     return weights + MIN_KV_BYTES
+
+
+def claim_for_allocatable(model: dict, spec: ModelSpec | None, tp: int,
+                          alloc_bytes: int) -> int:
+    """What the engine will actually consume on a GPU with `alloc_bytes`
+    free: weights + overhead + gmu x (free - weights - overhead). Claiming
+    only the floor would let a later placement over-commit the device the
+    moment this engine sizes its KV pool."""
+    need = estimate_vram_claim(model, spec, tp)
+    if spec is None:
+        return need
+    gmu = model.get("gpu_memory_utilization") or 0.9
+    base = need - MIN_KV_BYTES
+    kv = max(MIN_KV_BYTES, int((alloc_bytes - base) * gmu))
+    return min(alloc_bytes, base + kv)
 
 
 # ---- filters (reference: policies/worker_filters/*) -----------------------
@@ -126,8 +160,13 @@ def worker_allocatable(worker: dict, instances: list[dict]) -> dict[int, int]:
 def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -> list[Candidate]:
     spec = model_spec_for(model)
     tp = max(1, model.get("gpus_per_replica") or 1)
-    claim = estimate_vram_claim(model, spec, tp)
+    claim = estimate_vram_claim(model, spec, tp)   # per-GPU floor
     manual = model.get("gpu_selector") or None
+
+    def full_claim(alloc: dict[int, int], picks: list[int]) -> dict[int, int]:
+        return {i: claim_for_allocatable(model, spec, tp, alloc[i])
+                for i in picks}
+
     out: list[Candidate] = []
     for w in workers:
         alloc = worker_allocatable(w, instances)
@@ -139,7 +178,8 @@ def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -
                 if parts[0] in (str(w.get("id")), w.get("name")):
                     picks.append(int(parts[-1]))
             if len(picks) >= tp and all(alloc.get(i, 0) >= claim for i in picks[:tp]):
-                out.append(Candidate(w, picks[:tp], vram_claim={i: claim for i in picks[:tp]}))
+                out.append(Candidate(w, picks[:tp],
+                                     vram_claim=full_claim(alloc, picks[:tp])))
             continue
         fits = sorted(
             (i for i, free in alloc.items() if free >= claim),
@@ -147,12 +187,59 @@ def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -
         )
         if len(fits) >= tp:
             picks = fits[:tp]
-            out.append(Candidate(w, picks, vram_claim={i: claim for i in picks}))
+            out.append(Candidate(w, picks, vram_claim=full_claim(alloc, picks)))
     if not out and model.get("distributed_inference_across_workers"):
         cand = _multi_worker_candidate(tp, claim, workers, instances)
         if cand is not None:
             out.append(cand)
+    if not out and tp == 1:
+        cand = _offload_candidate(model, spec, workers, instances)
+        if cand is not None:
+            out.append(cand)
     return out
+
+
+def _offload_candidate(model: dict, spec: ModelSpec | None,
+                       workers: list[dict],
+                       instances: list[dict]) -> Candidate | None:
+    """Partial CPU offload placement (reference: the GGUF selector's
+    layer-offload path, gguf_resource_fit_selector.py:129-300 +
+    offload_layer_scorer): when no GPU holds the full weights, place on
+    the roomiest GPU and stream the overflow layers' weights from pinned
+    host DRAM (engine/offload.py). Opt-in via backend_parameters
+    cpu_offload=true (matches the reference's explicit offload knobs)."""
+    bp = model.get("backend_parameters") or {}
+    if not bp.get("cpu_offload") or spec is None:
+        return None
+    weights = int(_weight_bytes(model, spec, 1) * WEIGHT_FUDGE)
+    need_floor = FRAMEWORK_OVERHEAD + MIN_KV_BYTES
+    best: tuple[int, dict, int] | None = None
+    for w in workers:
+        alloc = worker_allocatable(w, instances)
+        for i, free in alloc.items():
+            if free > need_floor and (best is None or free > best[0]):
+                best = (free, w, i)
+    if best is None:
+        return None
+    free, w, idx = best
+    resident_budget = int(free * 0.9) - need_floor
+    if resident_budget <= 0 or weights <= resident_budget:
+        return None
+    offload_bytes = weights - resident_budget
+    if offload_bytes > weights * 0.9:
+        return None  # less than 10% resident: refuse (would crawl)
+    per_layer = weights / max(1, spec.num_layers)
+    layers = min(spec.num_layers - 1,
+                 max(1, int(offload_bytes / per_layer + 0.999)))
+    host_ram = (w.get("status") or {}).get("memory", {}).get("total", 0)
+    ram_claim = offload_bytes + (2 << 30)
+    if host_ram and ram_claim > host_ram:
+        return None
+    cand = Candidate(w, [idx], vram_claim={idx: int(free * 0.9)})
+    cand.offload_gb = round(offload_bytes / 2**30 + 0.05, 2)
+    cand.offload_layers = layers
+    cand.ram_claim = ram_claim
+    return cand
 
 
 def _multi_worker_candidate(tp: int, claim: int, workers: list[dict],

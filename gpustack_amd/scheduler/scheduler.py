@@ -104,14 +104,20 @@ class PlacementScheduler:
                 ar_update(s, inst)
                 return False
             tp = max(1, model.gpus_per_replica or 1)
-            claim = estimate_vram_claim(model_d, spec, tp)
+            fallback = estimate_vram_claim(model_d, spec, tp)
             inst.worker_id = cand.worker["id"]
             inst.worker_ip = cand.worker.get("ip", "")
             inst.gpu_indexes = cand.gpu_indexes
             inst.computed_resource_claim = {
-                "vram": {str(i): claim for i in cand.gpu_indexes},
-                "ram": 2 << 30,
+                "vram": {str(i): cand.vram_claim.get(i, fallback)
+                         for i in cand.gpu_indexes},
+                "ram": cand.ram_claim,
             }
+            if cand.offload_gb > 0:
+                # partial CPU offload (reference claim fields:
+                # schemas/models.py:623-630 offload_layers)
+                inst.computed_resource_claim["offload_gb"] = cand.offload_gb
+                inst.computed_resource_claim["offload_layers"] = cand.offload_layers
             if cand.subordinates:
                 # cross-worker TP: rank layout + rendezvous port
                 # (reference: serve_manager.py:1643-1739 port bands +

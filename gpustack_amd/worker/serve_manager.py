@@ -215,6 +215,12 @@ class ServeManager:
         env.update(model.get("env") or {})
 
         bp = dict(model.get("backend_parameters") or {})
+        bp.pop("cpu_offload", None)  # scheduler knob, not an engine field
+        claim = inst.get("computed_resource_claim") or {}
+        if claim.get("offload_gb"):
+            # scheduler placed this instance with partial CPU offload
+            # (engine/offload.py streams the tail layers from host DRAM)
+            bp.setdefault("cpu_offload_gb", claim["offload_gb"])
         if model.get("lora_list"):
             bp.setdefault("lora_dirs", model["lora_list"])
         if model.get("lora_adapters"):
