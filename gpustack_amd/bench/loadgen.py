@@ -154,6 +154,21 @@ class LoadGenerator:
                     await asyncio.wait(tasks, timeout=60)
         return self.summary()
 
+    def raw_records(self, cap: int = 1000) -> list[dict]:
+        """Per-request artifact rows (reference: benchmark artifacts,
+        worker/benchmark/artifacts.py): capped to keep the DB row small."""
+        out = []
+        for r in self.results[:cap]:
+            out.append({
+                "ok": r.ok,
+                "ttft_ms": round(r.ttft * 1000, 2) if r.ttft is not None else None,
+                "latency_ms": round(r.latency * 1000, 2),
+                "output_tokens": r.output_tokens,
+                "prompt_tokens": r.prompt_tokens,
+                "error": r.error or None,
+            })
+        return out
+
     def summary(self) -> dict:
         ok = [r for r in self.results if r.ok]
         dur = self.spec.duration_s
@@ -181,5 +196,10 @@ class LoadGenerator:
         }
 
 
-def run_load(base_url: str, spec: LoadSpec) -> dict:
-    return asyncio.run(LoadGenerator(base_url, spec).run())
+def run_load(base_url: str, spec: LoadSpec,
+             with_artifacts: bool = False) -> dict:
+    gen = LoadGenerator(base_url, spec)
+    res = asyncio.run(gen.run())
+    if with_artifacts:
+        res["artifacts"] = gen.raw_records()
+    return res

@@ -132,10 +132,21 @@ class PlacementScheduler:
                         "rank_base": base,
                     })
                     base += len(gpus)
+                # rendezvous port from a dedicated band, skipping ports any
+                # live instance already holds (reference: distributed port
+                # bands, serve_manager.py:1643-1739 — chosen here because
+                # subordinate workers must agree on it before launch)
+                taken = {
+                    (i.get("distributed_servers") or {}).get("master_port")
+                    for i in others
+                }
+                mp = 45000 + (inst.id * 4) % 1000
+                while mp in taken:
+                    mp = 45000 + (mp - 45000 + 4) % 1000
                 inst.distributed_servers = {
                     "tp": tp,
                     "master_ip": cand.worker.get("ip", ""),
-                    "master_port": 45000 + (inst.id % 1000),
+                    "master_port": mp,
                     "subordinates": ranks,
                 }
             inst.state = ModelInstanceState.SCHEDULED.value

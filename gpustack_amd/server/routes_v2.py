@@ -182,6 +182,53 @@ def register_worker(body: WorkerRegister, request: Request,
         return w.to_dict()
 
 
+class _WorkerStatusBuffer:
+    """Batch worker status writes (reference:
+    gpustack/server/worker_status_buffer.py): at fleet scale per-POST DB
+    writes serialize on the session; posts land in a coalescing buffer
+    (latest status per worker wins) that a flusher drains every interval
+    in ONE transaction. State transitions (NOT_READY -> READY) still
+    write through immediately so reconcilers see them without delay."""
+
+    FLUSH_INTERVAL = 2.0
+
+    def __init__(self):
+        import threading
+
+        self._lock = threading.Lock()
+        self._pending: dict[int, dict] = {}
+        self._timer: threading.Timer | None = None
+
+    def put(self, worker_id: int, status: dict | None) -> None:
+        import threading
+
+        with self._lock:
+            self._pending[worker_id] = {"status": status, "ts": time.time()}
+            if self._timer is None:
+                self._timer = threading.Timer(self.FLUSH_INTERVAL, self.flush)
+                self._timer.daemon = True
+                self._timer.start()
+
+    def flush(self) -> None:
+        with self._lock:
+            batch, self._pending = self._pending, {}
+            self._timer = None
+        if not batch:
+            return
+        with get_session() as s:
+            for wid, entry in batch.items():
+                w = s.get(Worker, wid)
+                if not w:
+                    continue
+                if entry["status"]:
+                    w.status = entry["status"]
+                w.heartbeat_time = entry["ts"]
+            s.commit()
+
+
+_status_buffer = _WorkerStatusBuffer()
+
+
 @router.post("/workers/{worker_id}/status")
 def worker_status(worker_id: int, body: WorkerStatusUpdate,
                   _=Depends(verify_worker_token)):
@@ -189,14 +236,15 @@ def worker_status(worker_id: int, body: WorkerStatusUpdate,
         w = s.get(Worker, worker_id)
         if not w:
             raise HTTPException(404, "worker not found (re-register)")
-        w.status = body.status or w.status
-        w.heartbeat_time = time.time()
         if w.state != WorkerState.READY.value:
+            # state transition: write through + publish for reconcilers
+            w.status = body.status or w.status
+            w.heartbeat_time = time.time()
             w.state = WorkerState.READY.value
             ar_update(s, w)
-        else:
-            s.commit()
-        return {"ok": True}
+            return {"ok": True}
+    _status_buffer.put(worker_id, body.status)
+    return {"ok": True, "buffered": True}
 
 
 @router.post("/workers/{worker_id}/heartbeat")

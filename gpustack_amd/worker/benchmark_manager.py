@@ -84,8 +84,14 @@ class BenchmarkManager:
                     osl=int(cfgd.get("osl", 64)),
                     model=b["model_name"],
                 )
-                point = run_load(f"http://127.0.0.1:{port}", spec)
+                snap_before = self._snapshot(b["model_name"])
+                point = run_load(f"http://127.0.0.1:{port}", spec,
+                                 with_artifacts=bool(cfgd.get("artifacts")))
                 point["value"] = float(v)
+                # per-point worker/GPU/instance snapshot (reference:
+                # schemas/benchmark.py:228-293 snapshot tables)
+                point["snapshot"] = {"before": snap_before,
+                                     "after": self._snapshot(b["model_name"])}
                 profile.append(point)
             results = dict(max(profile, key=lambda r: r.get("output_tps") or 0))
             if len(profile) > 1:
@@ -102,6 +108,31 @@ class BenchmarkManager:
             self._update(bid, state="error", state_message=str(e))
         finally:
             self._running.discard(bid)
+
+    def _snapshot(self, model_name: str) -> dict:
+        """GPU + instance state at a load point (utilization, VRAM,
+        instance pid/port) — the reference snapshots workers/GPUs/instances
+        around every benchmark run."""
+        out: dict = {"ts": time.time()}
+        try:
+            from .detector import detect_gpus
+
+            gpus = detect_gpus(self.cfg.gpu_devices or None)
+            out["gpus"] = [
+                {"index": g.get("index"),
+                 "utilization": (g.get("core") or {}).get("utilization_rate"),
+                 "vram_used": (g.get("memory") or {}).get("used"),
+                 "vram_total": (g.get("memory") or {}).get("total")}
+                for g in gpus
+            ]
+        except Exception:  # noqa: BLE001
+            out["gpus"] = []
+        for ip in self.serve_manager.processes.values():
+            if ip.instance.get("model_name") == model_name:
+                out["instance"] = {"pid": ip.proc.pid, "port": ip.port,
+                                   "healthy": ip.healthy}
+                break
+        return out
 
     def _update(self, bid: int, **fields) -> None:
         try:

@@ -164,6 +164,24 @@ class ServeManager:
 
     # ---- lifecycle -------------------------------------------------------
 
+    LOG_ROTATE_BYTES = 16 << 20
+    LOG_KEEP = 2
+
+    def _rotate_log(self, log_path: Path) -> None:
+        """Size-based rotation (reference: container-log persistence with
+        rotation, serve_manager.py:1131-1518): <name>.log -> .log.1 -> .log.2
+        once the live file passes LOG_ROTATE_BYTES."""
+        try:
+            if not log_path.exists() or log_path.stat().st_size < self.LOG_ROTATE_BYTES:
+                return
+            for i in range(self.LOG_KEEP, 0, -1):
+                src = Path(f"{log_path}.{i - 1}") if i > 1 else log_path
+                dst = Path(f"{log_path}.{i}")
+                if src.exists():
+                    src.replace(dst)
+        except OSError as e:
+            logger.warning("log rotation failed for %s: %s", log_path, e)
+
     def _assign_port(self) -> int:
         lo, hi = self.cfg.engine_port_range()
         with self._lock:
@@ -198,9 +216,23 @@ class ServeManager:
             source, ref = "local_path", local
 
         port = self._assign_port()
+        if inst.get("distributed_servers"):
+            # distributed instances get a fenced port BAND (reference:
+            # serve_manager.py:1643-1739): the rendezvous master port and
+            # follower slots around it must not be handed to the next
+            # instance on this worker
+            ds_ports = inst["distributed_servers"]
+            band = [ds_ports.get("master_port")]
+            with self._lock:
+                for p in band:
+                    if p:
+                        self._used_ports.add(p)
+                for fence in range(port + 1, port + 4):
+                    self._used_ports.add(fence)
         log_dir = Path(self.cfg.data_dir) / "log" / "instances"
         log_dir.mkdir(parents=True, exist_ok=True)
         log_path = log_dir / f"{inst['name']}.log"
+        self._rotate_log(log_path)
 
         env = dict(os.environ)
         ds = inst.get("distributed_servers") or None
