@@ -85,6 +85,11 @@ def test_worker_register_and_status(server):
     r = client.post(f"/v2/workers/{wid}/heartbeat",
                     headers={"Authorization": "Bearer nope"})
     assert r.status_code == 401
+    # status posts are batched (reference worker_status_buffer semantics):
+    # the write lands after the buffer flushes, coalesced per worker
+    from gpustack_amd.server.routes_v2 import _status_buffer
+
+    _status_buffer.flush()
     workers = client.get("/v2/workers").json()["items"]
     assert len(workers) == 1
     assert workers[0]["status"].get("gpu_devices") is None  # status replaced
