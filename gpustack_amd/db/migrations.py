@@ -69,6 +69,10 @@ def _m008_orgs(conn):
     _add_column(conn, "models", "org_id", "INTEGER")
 
 
+def _m009_gpu_type_selector(conn):
+    _add_column(conn, "models", "gpu_type_selector", "JSON")
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -78,6 +82,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (6, "worker_pools table for auto-provisioned capacity", _m006_worker_pools),
     (7, "multi-cluster: clusters table + cluster_id columns", _m007_clusters),
     (8, "orgs table + user/model org scoping", _m008_orgs),
+    (9, "model.gpu_type_selector for device-class placement", _m009_gpu_type_selector),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

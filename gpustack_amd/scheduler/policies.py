@@ -157,11 +157,36 @@ def worker_allocatable(worker: dict, instances: list[dict]) -> dict[int, int]:
 
 # ---- selector -------------------------------------------------------------
 
+def device_matches_type(dev: dict, sel: dict | None) -> bool:
+    """Device-class constraint (reference gpu_type_selector — vGPU slice /
+    MIG partition classes, schemas/models.py:92-175; here the MI355X-native
+    classes: AMD compute-partition mode SPX/DPX/CPX, NPS memory mode,
+    minimum VRAM, device-name substring)."""
+    if not sel:
+        return True
+    part = dev.get("partition") or {}
+    want_c = sel.get("partition_compute")
+    if want_c and (part.get("compute") or "").upper() != want_c.upper():
+        return False
+    want_m = sel.get("partition_memory")
+    if want_m and (part.get("memory") or "").upper() != want_m.upper():
+        return False
+    min_gb = sel.get("min_vram_gb")
+    if min_gb and ((dev.get("memory") or {}).get("total", 0)
+                   < min_gb * (1 << 30)):
+        return False
+    sub = sel.get("name_contains")
+    if sub and sub.lower() not in (dev.get("name") or "").lower():
+        return False
+    return True
+
+
 def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -> list[Candidate]:
     spec = model_spec_for(model)
     tp = max(1, model.get("gpus_per_replica") or 1)
     claim = estimate_vram_claim(model, spec, tp)   # per-GPU floor
     manual = model.get("gpu_selector") or None
+    type_sel = model.get("gpu_type_selector") or None
 
     def full_claim(alloc: dict[int, int], picks: list[int]) -> dict[int, int]:
         return {i: claim_for_allocatable(model, spec, tp, alloc[i])
@@ -170,6 +195,11 @@ def select_candidates(model: dict, workers: list[dict], instances: list[dict]) -
     out: list[Candidate] = []
     for w in workers:
         alloc = worker_allocatable(w, instances)
+        if type_sel:
+            devs = {d.get("index", 0): d
+                    for d in (w.get("status") or {}).get("gpu_devices", [])}
+            alloc = {i: free for i, free in alloc.items()
+                     if device_matches_type(devs.get(i, {}), type_sel)}
         if manual:
             ids = manual.get("gpu_ids", [])
             picks = []

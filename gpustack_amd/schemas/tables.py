@@ -154,6 +154,11 @@ class Model(Base, TimestampMixin, SerializeMixin):
     placement_strategy = Column(String(32), default=PlacementStrategy.BINPACK.value)
     worker_selector = Column(JSON, default=dict)     # label matching
     gpu_selector = Column(JSON, default=None)        # manual worker:gpu ids
+    # device-class constraints (reference gpu_type_selector, vGPU slices /
+    # MIG partitions, schemas/models.py:92-175; MI355X-native: AMD compute
+    # partitioning — SPX/CPX — and NPS memory modes reported by amdsmi):
+    # {"partition_compute": "CPX", "min_vram_gb": 32, "name_contains": "..."}
+    gpu_type_selector = Column(JSON, default=None)
     gpus_per_replica = Column(Integer, default=1)    # TP degree per replica
     backend_parameters = Column(JSON, default=dict)  # engine kwargs overrides
     env = Column(JSON, default=dict)

@@ -301,6 +301,41 @@ async def rerank(request: Request, user: User = Depends(get_current_user)):
     return await _proxy(request, "/v1/rerank", user)
 
 
+# ---- descoped modality surface -------------------------------------------
+# The reference proxies audio (STT/TTS via vox-box), image generation and
+# moderations to external engines (gateway/utils.py:167-194,
+# worker/backends/vox_box.py). This framework's first-party engine serves
+# text model families only; the endpoints exist wire-compatibly and return
+# a structured 501 naming the gap (formal descope — PARITY.md "audio/image
+# modalities") instead of a bare 404, so OpenAI SDK clients fail cleanly.
+_UNSUPPORTED_MODALITIES = [
+    "/v1/audio/transcriptions", "/v1/audio/translations", "/v1/audio/speech",
+    "/v1/images/generations", "/v1/images/edits", "/v1/images/variations",
+    "/v1/moderations",
+]
+
+
+def _register_unsupported(path: str) -> None:
+    @router.post(path, name=f"unsupported{path.replace('/', '_')}")
+    async def _unsupported(user: User = Depends(get_current_user)):
+        raise HTTPException(
+            501,
+            detail={
+                "error": {
+                    "message": f"{path} is not supported: this deployment "
+                               "serves text model families (llm, embedding, "
+                               "reranker) on the first-party MI355X engine; "
+                               "audio/image modalities are descoped",
+                    "type": "unsupported_modality",
+                }
+            },
+        )
+
+
+for _p in _UNSUPPORTED_MODALITIES:
+    _register_unsupported(_p)
+
+
 @router.post("/v1/messages")
 async def anthropic_messages(request: Request, user: User = Depends(get_current_user)):
     """Anthropic-style Messages API, proxied to the placed instance

@@ -39,12 +39,30 @@ def _detect_amdsmi() -> list[dict] | None:
                         amdsmi.AmdSmiTemperatureMetric.CURRENT)
                 except Exception:  # noqa: BLE001
                     temp = 0
-                devs.append(_gpu_entry(
+                entry = _gpu_entry(
                     index=i,
                     name=info.get("market_name", "AMD Instinct"),
                     uuid=str(info.get("asic_serial", f"amd-{i}")),
                     total=mem, used=used, util=act, temperature=temp,
-                ))
+                )
+                # AMD compute/memory partitioning (SPX/DPX/CPX, NPS1/NPS4)
+                # — the MI355X-native device-class the scheduler's
+                # gpu_type_selector matches on (reference: vGPU slice /
+                # MIG-partition classes, schemas/models.py:92-175)
+                part = {}
+                try:
+                    part["compute"] = str(
+                        amdsmi.amdsmi_get_gpu_compute_partition(h))
+                except Exception:  # noqa: BLE001
+                    pass
+                try:
+                    part["memory"] = str(
+                        amdsmi.amdsmi_get_gpu_memory_partition(h))
+                except Exception:  # noqa: BLE001
+                    pass
+                if part:
+                    entry["partition"] = part
+                devs.append(entry)
             return devs
         finally:
             amdsmi.amdsmi_shut_down()

@@ -144,3 +144,34 @@ def test_file_locality_scorer_prefers_holder():
     # without the file record the tie breaks by id order (w1)
     cand = pick_candidate(model, [w1, w2], [], [])
     assert cand.worker["id"] == 1
+
+
+def test_gpu_type_selector_filters_devices():
+    """Device-class placement (reference gpu_type_selector): only devices
+    matching the partition/VRAM class are eligible."""
+    from gpustack_amd.scheduler.policies import (device_matches_type,
+                                                 select_candidates)
+
+    big = {"index": 0, "type": "rocm", "name": "AMD Instinct MI355X",
+           "memory": {"total": 288 << 30},
+           "partition": {"compute": "SPX", "memory": "NPS1"}}
+    slice_ = {"index": 1, "type": "rocm", "name": "AMD Instinct MI355X",
+              "memory": {"total": 36 << 30},
+              "partition": {"compute": "CPX", "memory": "NPS1"}}
+    assert device_matches_type(big, {"partition_compute": "SPX"})
+    assert not device_matches_type(slice_, {"partition_compute": "SPX"})
+    assert device_matches_type(slice_, {"min_vram_gb": 32})
+    assert not device_matches_type(slice_, {"min_vram_gb": 64})
+    assert device_matches_type(big, {"name_contains": "mi355"})
+
+    worker = {"id": 1, "name": "w", "state": "ready",
+              "status": {"gpu_devices": [big, slice_]},
+              "system_reserved": {}}
+    model = {"id": 1, "source": "preset", "model_ref": "tiny",
+             "gpus_per_replica": 1,
+             "gpu_type_selector": {"partition_compute": "CPX"}}
+    cands = select_candidates(model, [worker], [])
+    assert len(cands) == 1 and cands[0].gpu_indexes == [1]
+    model["gpu_type_selector"] = {"partition_compute": "SPX"}
+    cands = select_candidates(model, [worker], [])
+    assert cands and cands[0].gpu_indexes == [0]

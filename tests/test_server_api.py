@@ -460,3 +460,14 @@ def test_org_tenancy(server):
     assert r.status_code == 503
     # admin sees everything; org with members cannot be deleted
     assert client.delete(f"/v2/orgs/{org['id']}").status_code == 409
+
+
+def test_unsupported_modalities_return_501(server):
+    """Audio/image/moderations: wire-compatible endpoints exist and fail
+    with a structured 501 (formal descope, PARITY.md)."""
+    client, app, cfg, reg = server
+    for path in ("/v1/audio/transcriptions", "/v1/images/generations",
+                 "/v1/moderations"):
+        r = client.post(path, json={"model": "x"})
+        assert r.status_code == 501, path
+        assert r.json()["detail"]["error"]["type"] == "unsupported_modality"
