@@ -711,9 +711,15 @@ def delete_benchmark(bench_id: int, _: User = Depends(get_current_user)):
 # ---- tunnel proxy (NAT workers; reference websocket_proxy/) ----------------
 
 @router.get("/tunnel/jobs")
-async def tunnel_jobs(worker_id: int, _=Depends(verify_worker_token)):
+async def tunnel_jobs(worker_id: int, batch: int = 0,
+                      _=Depends(verify_worker_token)):
+    """Long-poll job pickup. `batch=1` returns {"jobs": [...]} with every
+    queued job (first blocks, rest drain) so a burst of concurrent
+    requests does not pay one poll round-trip each."""
     from .tunnel import hub
 
+    if batch:
+        return {"jobs": await hub.next_jobs(worker_id)}
     job = await hub.next_job(worker_id)
     if job is None:
         from fastapi.responses import Response as _Resp

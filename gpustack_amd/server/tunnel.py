@@ -49,6 +49,22 @@ class TunnelHub:
         except asyncio.TimeoutError:
             return None
 
+    async def next_jobs(self, worker_id: int, limit: int = 32) -> list[dict]:
+        """Batch pickup: block for the first job, then drain whatever else
+        is queued — dispatch rate is no longer one job per poll round-trip
+        (the r1 concurrent-streaming-load concern)."""
+        first = await self.next_job(worker_id)
+        if first is None:
+            return []
+        out = [first]
+        q = self._job_q(worker_id)
+        while len(out) < limit:
+            try:
+                out.append(q.get_nowait())
+            except asyncio.QueueEmpty:
+                break
+        return out
+
     async def begin_reply(self, req_id: str, status: int, content_type: str):
         q = self._replies.get(req_id)
         if q is None:

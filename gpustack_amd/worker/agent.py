@@ -101,13 +101,14 @@ class WorkerAgent:
         while not self._stop:
             try:
                 r = self.client._c.get("/v2/tunnel/jobs",
-                                       params={"worker_id": self.worker_id},
+                                       params={"worker_id": self.worker_id,
+                                               "batch": 1},
                                        timeout=_h.Timeout(30.0, read=40.0))
                 if r.status_code != 200:
                     continue
-                job = r.json()
-                threading.Thread(target=self._serve_tunnel_job, args=(job,),
-                                 daemon=True).start()
+                for job in r.json().get("jobs", []):
+                    threading.Thread(target=self._serve_tunnel_job,
+                                     args=(job,), daemon=True).start()
             except Exception as e:  # noqa: BLE001
                 logger.debug("tunnel poll error: %s", e)
                 time.sleep(1)
