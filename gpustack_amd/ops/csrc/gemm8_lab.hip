@@ -250,8 +250,14 @@ __global__ __launch_bounds__(GL_THREADS) void gemm_lab_kernel(
         mfma_burst(p & 1, p & 1);
         if constexpr (MODE == 2) {
           __builtin_amdgcn_s_barrier();
-        } else {
+        } else if constexpr (MODE == 3) {
           if (p & 1) __builtin_amdgcn_s_barrier();
+        } else {  // MODE 5: barriers at EVEN phases — they follow the
+          // p2/p6 vmcnt(4), so each wave's "my tile-T+1 loads landed"
+          // becomes a collective guarantee before any wave's p3/p7 reads
+          // (the odd-phase cadence of MODE 3 lacked this propagation and
+          // raced on the 1002-WG lm_head grid)
+          if ((p & 1) == 0) __builtin_amdgcn_s_barrier();
         }
       }
     }
@@ -304,6 +310,7 @@ void gemm_lab_launch(void* out, const void* x, const void* w, int M, int N,
     *err_unsupported = 1;
     return;
   }
+  if (mode == 5) mode = 4;  // MODE index 4 template instantiation
   const int mtiles = (M + GL_BM - 1) / GL_BM;
   dim3 grid(mtiles * (N / GL_BN));
   dim3 block(GL_THREADS);
@@ -325,6 +332,11 @@ void gemm_lab_launch(void* out, const void* x, const void* w, int M, int N,
       break;
     case 3:
       hipLaunchKernelGGL((gemm_lab_kernel<3>), grid, block, 0, s,
+                         (unsigned short*)out, (const unsigned short*)x,
+                         (const unsigned short*)w, M, N, K);
+      break;
+    case 4:
+      hipLaunchKernelGGL((gemm_lab_kernel<4>), grid, block, 0, s,
                          (unsigned short*)out, (const unsigned short*)x,
                          (const unsigned short*)w, M, N, K);
       break;
