@@ -38,6 +38,10 @@ def parse_args():
     ap.add_argument("--speculative", default=None, choices=["ngram", "eagle", "eagle3"],
                     help="speculative decoding method (BASELINE config 5)")
     ap.add_argument("--draft-tokens", type=int, default=3)
+    ap.add_argument("--draft-dir", default=None,
+                    help="EAGLE/MTP draft checkpoint dir (speculative_config "
+                         "draft_dir; produce one offline with "
+                         "scripts/train_eagle_draft.py)")
     ap.add_argument("--prefix-caching", action="store_true",
                     help="enable automatic prefix caching (shared-prefix workloads)")
     ap.add_argument("--quantize", default=None, choices=["w4"],
@@ -90,7 +94,9 @@ def main():
         quantize_runtime=args.quantize,
         enable_prefix_caching=args.prefix_caching,
         speculative=({"method": args.speculative,
-                      "num_draft_tokens": args.draft_tokens}
+                      "num_draft_tokens": args.draft_tokens,
+                      **({"draft_dir": args.draft_dir}
+                         if args.draft_dir else {})}
                      if args.speculative else None),
         tp_size=tp if comm else 1,
         tp_rank=rank if comm else 0,
