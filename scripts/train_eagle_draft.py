@@ -144,10 +144,11 @@ def main():
     ap.add_argument("--out", default="/tmp/eagle_draft")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--device", default=None)
-    ap.add_argument("--rollouts", type=int, default=48)
-    ap.add_argument("--isl", type=int, default=256)
+    ap.add_argument("--rollouts", type=int, default=256)
+    ap.add_argument("--isl", type=int, default=256, help="max prompt len (prompts vary 32..isl)")
     ap.add_argument("--osl", type=int, default=160)
     ap.add_argument("--steps", type=int, default=400)
+    ap.add_argument("--val", type=int, default=16, help="held-out sequences")
     ap.add_argument("--lr", type=float, default=3e-4)
     ap.add_argument("--selftest", action="store_true")
     args = ap.parse_args()
@@ -167,7 +168,8 @@ def main():
     rng = random.Random(77)
     vocab = cfg.spec.vocab_size
     t0 = time.time()
-    prompts = [[rng.randrange(2, vocab) for _ in range(args.isl)]
+    prompts = [[rng.randrange(2, vocab)
+                for _ in range(rng.randrange(32, args.isl + 1))]
                for _ in range(args.rollouts)]
     outs = eng.generate(prompts, SamplingParams(max_tokens=args.osl,
                                                 ignore_eos=True))
@@ -181,7 +183,9 @@ def main():
     for sq in seqs:
         H = collect_hiddens(runner, sq, dev)
         data.append((torch.tensor(sq, dtype=torch.long, device=dev), H))
-    print(f"hiddens collected ({len(data)} seqs)", flush=True)
+    val = data[:args.val]
+    data = data[args.val:]
+    print(f"hiddens collected ({len(data)} train / {len(val)} val)", flush=True)
 
     draft = DraftTrainer(spec, dev)
     opt = torch.optim.Adam(draft.parameters(), lr=args.lr)
@@ -202,11 +206,19 @@ def main():
         opt.zero_grad()
         loss.backward()
         opt.step()
-        if step % 50 == 0 or step == args.steps - 1:
+        if step % 100 == 0 or step == args.steps - 1:
             with torch.no_grad():
                 acc = (logits.argmax(-1) == toks[2:L]).float().mean().item()
-            print(f"step {step}: loss {loss.item():.3f} "
-                  f"next-token acc {acc:.3f} ({time.time()-t0:.0f}s)",
+                vac, vn = 0.0, 0
+                for vt, vH in val:
+                    Lv = vt.shape[0]
+                    vo = draft(F.embedding(vt[1:Lv - 1], embed), vH[0:Lv - 2],
+                               cos_sin, torch.arange(1, Lv - 1, device=dev))
+                    vl = F.linear(vo, lm_head)
+                    vac += (vl.argmax(-1) == vt[2:Lv]).float().sum().item()
+                    vn += Lv - 2
+            print(f"step {step}: loss {loss.item():.3f} train-acc {acc:.3f} "
+                  f"VAL-acc {vac/max(vn,1):.3f} ({time.time()-t0:.0f}s)",
                   flush=True)
     draft.save(args.out)
     print(f"saved draft checkpoint to {args.out}", flush=True)
