@@ -73,6 +73,12 @@ def _m009_gpu_type_selector(conn):
     _add_column(conn, "models", "gpu_type_selector", "JSON")
 
 
+def _m010_resource_events(conn):
+    from ..schemas.tables import ResourceEvent, ResourceEventArchive
+    ResourceEvent.__table__.create(conn, checkfirst=True)
+    ResourceEventArchive.__table__.create(conn, checkfirst=True)
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -83,6 +89,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (7, "multi-cluster: clusters table + cluster_id columns", _m007_clusters),
     (8, "orgs table + user/model org scoping", _m008_orgs),
     (9, "model.gpu_type_selector for device-class placement", _m009_gpu_type_selector),
+    (10, "resource-event metering pair (hot + archive)", _m010_resource_events),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

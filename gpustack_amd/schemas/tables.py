@@ -293,6 +293,42 @@ class ModelUsageArchive(Base, TimestampMixin, SerializeMixin):
     request_count = Column(Integer, default=0)
 
 
+class ResourceEvent(Base, TimestampMixin, SerializeMixin):
+    """Resource lifecycle events for metering (reference hot table of the
+    resource-event pair, schemas/resource_events.py + the
+    ResourceEventLogger wired at server/server.py:541-595): one row per
+    instance state transition with the resource footprint attached, so
+    billing/capacity pipelines can reconstruct GPU-seconds per model."""
+
+    __tablename__ = "resource_events"
+    id = Column(Integer, primary_key=True)
+    event_type = Column(String(32), index=True)   # scheduled/running/stopped/error
+    instance_id = Column(Integer, index=True)
+    model_id = Column(Integer, index=True)
+    model_name = Column(String(256), index=True)
+    worker_id = Column(Integer, index=True)
+    gpu_indexes = Column(JSON, default=list)
+    vram_bytes = Column(Integer, default=0)       # total claim at event time
+    ram_bytes = Column(Integer, default=0)
+    timestamp = Column(Float, default=time.time, index=True)
+
+
+class ResourceEventArchive(Base, TimestampMixin, SerializeMixin):
+    """Cold tier of the resource-event pair."""
+
+    __tablename__ = "resource_events_archive"
+    id = Column(Integer, primary_key=True)
+    event_type = Column(String(32), index=True)
+    instance_id = Column(Integer, index=True)
+    model_id = Column(Integer, index=True)
+    model_name = Column(String(256), index=True)
+    worker_id = Column(Integer, index=True)
+    gpu_indexes = Column(JSON, default=list)
+    vram_bytes = Column(Integer, default=0)
+    ram_bytes = Column(Integer, default=0)
+    timestamp = Column(Float, index=True)
+
+
 class SystemLoad(Base, SerializeMixin):
     __tablename__ = "system_load"
     id = Column(Integer, primary_key=True)

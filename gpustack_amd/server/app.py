@@ -314,8 +314,9 @@ def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
 def start_background_tasks(cfg: Config, app: FastAPI) -> None:
     from ..scheduler.scheduler import PlacementScheduler
     from .controllers import (
-        ModelController, ScalingScheduler, SystemLoadCollector, UsageArchiver,
-        WorkerMonitor, WorkerPoolController,
+        ModelController, ResourceEventLogger, ScalingScheduler,
+        SystemLoadCollector, UsageArchiver, WorkerMonitor,
+        WorkerPoolController,
     )
     from .coordinator import LeaseCoordinator, LocalCoordinator
 
@@ -336,7 +337,7 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
     sched = PlacementScheduler(cfg)
     tasks = [sched, ModelController(cfg), WorkerMonitor(cfg),
              SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg),
-             WorkerPoolController(cfg)]
+             WorkerPoolController(cfg), ResourceEventLogger(cfg)]
     for t in tasks:
         t.coordinator = coord  # leader-only gating (checked per cycle)
     app.state.scheduler = sched

@@ -250,6 +250,71 @@ class ScalingScheduler(_LeaderGated):
                 time.sleep(1.0)
 
 
+class ResourceEventLogger(_LeaderGated):
+    """Metering event writer (reference: ResourceEventLogger wired at
+    server/server.py:541-595): subscribes to ModelInstance bus events and
+    appends one resource_events row per STATE TRANSITION with the claim
+    footprint attached — the hot half of the hot+archive pair the
+    UsageArchiver drains."""
+
+    METERED_STATES = {"scheduled", "running", "stopped", "error",
+                      "unreachable"}
+
+    def __init__(self, cfg: Config):
+        self.cfg = cfg
+        self._stop = False
+        self._last_state: dict[int, str] = {}
+
+    def stop(self) -> None:
+        self._stop = True
+
+    def _record(self, data: dict) -> None:
+        from ..schemas.tables import ResourceEvent
+
+        iid = data.get("id")
+        state = data.get("state")
+        if iid is None or state not in self.METERED_STATES:
+            return
+        if self._last_state.get(iid) == state:
+            return  # non-transition update (heartbeat field churn)
+        self._last_state[iid] = state
+        claim = data.get("computed_resource_claim") or {}
+        vram = sum((claim.get("vram") or {}).values())
+        with get_session() as s:
+            s.add(ResourceEvent(
+                event_type=state, instance_id=iid,
+                model_id=data.get("model_id"),
+                model_name=data.get("model_name", ""),
+                worker_id=data.get("worker_id"),
+                gpu_indexes=data.get("gpu_indexes") or [],
+                vram_bytes=int(vram), ram_bytes=int(claim.get("ram") or 0),
+            ))
+            s.commit()
+
+    def run(self) -> None:
+        import queue as _q
+
+        from ..db import bus
+
+        q = bus.subscribe("model_instances")
+        try:
+            while not self._stop:
+                try:
+                    ev = q.get(timeout=1.0)
+                except _q.Empty:
+                    continue
+                try:
+                    if self._is_leader() and isinstance(ev.data, dict):
+                        if ev.type.value == "DELETED":
+                            self._last_state.pop(ev.data.get("id"), None)
+                        else:
+                            self._record(ev.data)
+                except Exception:  # noqa: BLE001
+                    logger.exception("resource-event record failed")
+        finally:
+            bus.unsubscribe("model_instances", q)
+
+
 class UsageArchiver(_LeaderGated):
     """Hot -> archive mover for usage rows older than `keep_days`
     (reference: server/usage_archiver.py TableArchiver)."""
@@ -279,6 +344,23 @@ class UsageArchiver(_LeaderGated):
                     prompt_tokens=row.prompt_tokens,
                     completion_tokens=row.completion_tokens,
                     request_count=row.request_count,
+                ))
+                s.delete(row)
+                moved += 1
+            # resource-event pair (same retention window, timestamp-based)
+            import time as _t
+
+            from ..schemas.tables import ResourceEvent, ResourceEventArchive
+
+            ts_cutoff = _t.time() - self.keep_days * 86400
+            for row in s.query(ResourceEvent).filter(
+                    ResourceEvent.timestamp < ts_cutoff).all():
+                s.add(ResourceEventArchive(
+                    event_type=row.event_type, instance_id=row.instance_id,
+                    model_id=row.model_id, model_name=row.model_name,
+                    worker_id=row.worker_id, gpu_indexes=row.gpu_indexes,
+                    vram_bytes=row.vram_bytes, ram_bytes=row.ram_bytes,
+                    timestamp=row.timestamp,
                 ))
                 s.delete(row)
                 moved += 1

@@ -471,3 +471,39 @@ def test_unsupported_modalities_return_501(server):
         r = client.post(path, json={"model": "x"})
         assert r.status_code == 501, path
         assert r.json()["detail"]["error"]["type"] == "unsupported_modality"
+
+
+def test_resource_event_logger_records_transitions(server):
+    """Metering pair (reference resource_events): state transitions append
+    hot rows with the claim footprint; archiver drains old rows."""
+    import time as _time
+
+    from gpustack_amd.db import get_session
+    from gpustack_amd.schemas import ModelInstance
+    from gpustack_amd.schemas.tables import ResourceEvent, ResourceEventArchive
+    from gpustack_amd.server.controllers import (ResourceEventLogger,
+                                                 UsageArchiver)
+
+    client, app, cfg, reg = server
+    lg = ResourceEventLogger(cfg)
+    mid = client.post("/v2/models", json={"name": "re", "model_ref": "tiny"}).json()["id"]
+    data = {"id": 991, "state": "scheduled", "model_id": mid,
+            "model_name": "re", "worker_id": 1, "gpu_indexes": [0],
+            "computed_resource_claim": {"vram": {"0": 1 << 30}, "ram": 2 << 30}}
+    lg._record(data)
+    lg._record(data)  # same state: deduped
+    data2 = dict(data, state="running")
+    lg._record(data2)
+    with get_session() as s:
+        rows = s.query(ResourceEvent).all()
+        assert [r.event_type for r in rows] == ["scheduled", "running"]
+        assert rows[0].vram_bytes == 1 << 30 and rows[0].ram_bytes == 2 << 30
+        # age a row and archive it
+        rows[0].timestamp = _time.time() - 90 * 86400
+        s.commit()
+    arch = UsageArchiver(cfg, keep_days=30)
+    moved = arch.archive_once()
+    assert moved >= 1
+    with get_session() as s:
+        assert s.query(ResourceEventArchive).count() == 1
+        assert s.query(ResourceEvent).count() == 1
