@@ -1,0 +1,316 @@
+#!/usr/bin/env python3
+"""Demonstrate the speculative-decoding SPEEDUP case end-to-end, offline.
+
+profiles/r04 measures that a RANDOM-INIT target cannot yield a speculative
+speedup (its next-token function has no structure a draft can learn:
+held-out acc 0.000). EAGLE's premise requires a TRAINED target — so this
+demo manufactures one: a small Llama-architecture model gradient-trained
+until its greedy behavior follows a deterministic token permutation (a
+Markov-1 "language"), saved as a real HF-named checkpoint, served by the
+engine from disk. The draft is then distilled against the trained target's
+trajectories (scripts/train_eagle_draft.py machinery) and the engine's
+eagle/ngram paths are timed against plain greedy decode — same engine,
+same kernels, same verify-exactness guarantee.
+
+    python scripts/demo_spec_speedup.py                # full GPU demo
+    python scripts/demo_spec_speedup.py --dry          # tiny CPU dry run
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
+import gpustack_amd.engine  # noqa: F401
+import torch
+import torch.nn.functional as F
+
+from train_eagle_draft import DraftTrainer, collect_hiddens, rope_neox
+
+
+class TorchLlama(torch.nn.Module):
+    """Trainable fp32 Llama mirror (same math as the engine's model)."""
+
+    def __init__(self, spec, device):
+        super().__init__()
+        self.spec = spec
+        h, d = spec.hidden_size, spec.head_dim
+        self.eps = spec.rms_norm_eps
+        def P(*shape, std=0.02):
+            return torch.nn.Parameter(torch.randn(*shape, device=device) * std)
+        self.embed = P(spec.vocab_size, h)
+        self.lm_head = P(spec.vocab_size, h)
+        self.final_norm = torch.nn.Parameter(torch.ones(h, device=device))
+        self.layers = torch.nn.ModuleList()
+        for _ in range(spec.num_layers):
+            m = torch.nn.Module()
+            m.q = P(spec.num_heads * d, h)
+            m.k = P(spec.num_kv_heads * d, h)
+            m.v = P(spec.num_kv_heads * d, h)
+            m.o = P(h, spec.num_heads * d)
+            m.gate = P(spec.intermediate_size, h)
+            m.up = P(spec.intermediate_size, h)
+            m.down = P(h, spec.intermediate_size)
+            m.input_norm = torch.nn.Parameter(torch.ones(h, device=device))
+            m.post_norm = torch.nn.Parameter(torch.ones(h, device=device))
+            self.layers.append(m)
+
+    def rms(self, x, w):
+        return x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps) * w
+
+    def forward(self, toks, cos_sin):
+        spec = self.spec
+        d = spec.head_dim
+        T = toks.shape[0]
+        pos = torch.arange(T, device=toks.device)
+        x = F.embedding(toks, self.embed)
+        for m in self.layers:
+            h1 = self.rms(x, m.input_norm)
+            q = rope_neox(F.linear(h1, m.q).view(T, spec.num_heads, d),
+                          cos_sin, pos)
+            k = rope_neox(F.linear(h1, m.k).view(T, spec.num_kv_heads, d),
+                          cos_sin, pos)
+            v = F.linear(h1, m.v).view(T, spec.num_kv_heads, d)
+            gq = spec.num_heads // spec.num_kv_heads
+            k = k.repeat_interleave(gq, dim=1)
+            v = v.repeat_interleave(gq, dim=1)
+            att = F.scaled_dot_product_attention(
+                q.permute(1, 0, 2), k.permute(1, 0, 2), v.permute(1, 0, 2),
+                is_causal=True)
+            x = x + F.linear(att.permute(1, 0, 2).reshape(T, -1), m.o)
+            h2 = self.rms(x, m.post_norm)
+            x = x + F.linear(
+                F.silu(F.linear(h2, m.gate)) * F.linear(h2, m.up), m.down)
+        return F.linear(self.rms(x, self.final_norm), self.lm_head)
+
+    def save_hf(self, out_dir: str) -> None:
+        from safetensors.torch import save_file
+
+        spec = self.spec
+        out = Path(out_dir)
+        out.mkdir(parents=True, exist_ok=True)
+        t = {"model.embed_tokens.weight": self.embed.data,
+             "model.norm.weight": self.final_norm.data,
+             "lm_head.weight": self.lm_head.data}
+        for i, m in enumerate(self.layers):
+            p = f"model.layers.{i}."
+            t[p + "self_attn.q_proj.weight"] = m.q.data
+            t[p + "self_attn.k_proj.weight"] = m.k.data
+            t[p + "self_attn.v_proj.weight"] = m.v.data
+            t[p + "self_attn.o_proj.weight"] = m.o.data
+            t[p + "mlp.gate_proj.weight"] = m.gate.data
+            t[p + "mlp.up_proj.weight"] = m.up.data
+            t[p + "mlp.down_proj.weight"] = m.down.data
+            t[p + "input_layernorm.weight"] = m.input_norm.data
+            t[p + "post_attention_layernorm.weight"] = m.post_norm.data
+        save_file({k: v.to(torch.bfloat16).cpu().contiguous()
+                   for k, v in t.items()}, str(out / "model.safetensors"))
+        cfgj = {
+            "architectures": ["LlamaForCausalLM"],
+            "hidden_size": spec.hidden_size,
+            "intermediate_size": spec.intermediate_size,
+            "num_hidden_layers": spec.num_layers,
+            "num_attention_heads": spec.num_heads,
+            "num_key_value_heads": spec.num_kv_heads,
+            "head_dim": spec.head_dim,
+            "vocab_size": spec.vocab_size,
+            "rope_theta": spec.rope_theta,
+            "rms_norm_eps": spec.rms_norm_eps,
+            "max_position_embeddings": spec.max_position_embeddings,
+            "tie_word_embeddings": False,
+            "eos_token_id": 0,
+            "torch_dtype": "bfloat16",
+        }
+        (out / "config.json").write_text(json.dumps(cfgj))
+
+
+def timed_decode(eng, prompts, osl, warmup=6, steps=30):
+    """Tokens/sec over `steps` engine iterations at fixed concurrency."""
+    from gpustack_amd.engine import SamplingParams
+
+    p = SamplingParams(max_tokens=osl, ignore_eos=True)
+    for pr in prompts:
+        eng.add_request(list(pr), p)
+    for _ in range(warmup):
+        eng.step()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    toks = 0
+    for _ in range(steps):
+        outs = eng.step()
+        toks += len(outs)
+        if not eng.has_unfinished():
+            break
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    while eng.has_unfinished():
+        eng.step()
+    return toks / dt, toks / max(1, steps)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--dry", action="store_true", help="tiny CPU dry run")
+    ap.add_argument("--out", default="/tmp/spec_demo")
+    ap.add_argument("--target-steps", type=int, default=300)
+    ap.add_argument("--draft-steps", type=int, default=300)
+    ap.add_argument("--chain-vocab", type=int, default=4096)
+    args = ap.parse_args()
+
+    import random
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.engine.config import ModelSpec
+
+    dev = "cpu" if args.dry else "cuda:0"
+    if args.dry:
+        spec = ModelSpec(hidden_size=256, intermediate_size=512, num_layers=2,
+                         num_heads=4, num_kv_heads=2, head_dim=64,
+                         vocab_size=512, eos_token_id=0)
+        args.chain_vocab = 128
+        args.target_steps = min(args.target_steps, 120)
+        args.draft_steps = min(args.draft_steps, 60)
+    else:
+        spec = ModelSpec(hidden_size=2048, intermediate_size=8192,
+                         num_layers=6, num_heads=16, num_kv_heads=8,
+                         head_dim=128, vocab_size=32000, eos_token_id=0)
+
+    rng = random.Random(5)
+    lo = 2
+    perm = list(range(lo, lo + args.chain_vocab))
+    rng.shuffle(perm)
+    perm_map = {lo + i: perm[i] for i in range(args.chain_vocab)}
+
+    def chain(start, n):
+        seq = [start]
+        for _ in range(n - 1):
+            seq.append(perm_map[seq[-1]])
+        return seq
+
+    # ---- 1. train the target on the Markov-1 language -------------------
+    from gpustack_amd.ops import build_cos_sin_cache
+
+    target = TorchLlama(spec, dev)
+    cos_sin = build_cos_sin_cache(spec.head_dim, spec.head_dim,
+                                  spec.max_position_embeddings,
+                                  base=spec.rope_theta).to(dev)
+    opt = torch.optim.Adam(target.parameters(), lr=1e-3)
+    t0 = time.time()
+    L = 96 if args.dry else 192
+    for step in range(args.target_steps):
+        start = lo + rng.randrange(args.chain_vocab)
+        toks = torch.tensor(chain(start, L), dtype=torch.long, device=dev)
+        logits = target(toks[:-1], cos_sin)
+        loss = F.cross_entropy(logits, toks[1:])
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        if step % 50 == 0 or step == args.target_steps - 1:
+            acc = (logits.argmax(-1) == toks[1:]).float().mean().item()
+            print(f"target step {step}: loss {loss.item():.3f} "
+                  f"acc {acc:.3f} ({time.time()-t0:.0f}s)", flush=True)
+    ckpt = Path(args.out) / "target"
+    target.save_hf(str(ckpt))
+    del opt
+    if not args.dry:
+        torch.cuda.empty_cache()
+
+    # ---- 2. serve the trained checkpoint; sanity: follows the chain -----
+    eng = LLMEngine(EngineConfig(model=str(ckpt), device=dev,
+                                 max_model_len=1024, max_num_seqs=64,
+                                 gpu_memory_utilization=0.35,
+                                 enforce_random_weights=False,
+                                 kv_cache_blocks=2048 if args.dry else None))
+    starts = [lo + rng.randrange(args.chain_vocab) for _ in range(8)]
+    outs = eng.generate([chain(s, 16) for s in starts],
+                        SamplingParams(max_tokens=32, ignore_eos=True))
+    follow = sum(
+        sum(1 for a, b in zip(o, chain(perm_map[c[-1]], len(o))) if a == b) / len(o)
+        for o, c in zip(outs, [chain(s, 16) for s in starts])) / len(outs)
+    print(f"trained target follows the chain: {follow:.2%}", flush=True)
+
+    # ---- 3. distill the draft from the TRAINED target -------------------
+    prompts = [chain(lo + rng.randrange(args.chain_vocab),
+                     rng.randrange(16, 64)) for _ in range(64)]
+    roll = eng.generate(prompts, SamplingParams(max_tokens=96,
+                                                ignore_eos=True))
+    seqs = [p + o for p, o in zip(prompts, roll)]
+    data = []
+    for sq in seqs:
+        H = collect_hiddens(eng.runner, sq, dev)
+        data.append((torch.tensor(sq, dtype=torch.long, device=dev), H))
+    val, data = data[:8], data[8:]
+    draft = DraftTrainer(spec, dev)
+    dopt = torch.optim.Adam(draft.parameters(), lr=1e-3)
+    embed = eng.runner.model.embed.float()
+    lm_head = eng.runner.model.lm_head.float()
+    mcs = eng.runner.model.cos_sin.float()
+    for step in range(args.draft_steps):
+        toks, H = data[step % len(data)]
+        Ls = toks.shape[0]
+        out = draft(F.embedding(toks[1:Ls - 1], embed), H[0:Ls - 2], mcs,
+                    torch.arange(1, Ls - 1, device=dev))
+        loss = F.cross_entropy(F.linear(out, lm_head), toks[2:Ls])
+        dopt.zero_grad()
+        loss.backward()
+        dopt.step()
+        if step % 50 == 0 or step == args.draft_steps - 1:
+            with torch.no_grad():
+                vac, vn = 0.0, 0
+                for vt, vH in val:
+                    Lv = vt.shape[0]
+                    vo = draft(F.embedding(vt[1:Lv - 1], embed),
+                               vH[0:Lv - 2], mcs,
+                               torch.arange(1, Lv - 1, device=dev))
+                    vac += (F.linear(vo, lm_head).argmax(-1)
+                            == vt[2:Lv]).float().sum().item()
+                    vn += Lv - 2
+            print(f"draft step {step}: loss {loss.item():.3f} "
+                  f"VAL-acc {vac/max(vn,1):.3f}", flush=True)
+    draft_dir = Path(args.out) / "draft"
+    draft.save(str(draft_dir))
+    del eng, dopt
+    if not args.dry:
+        torch.cuda.empty_cache()
+
+    # ---- 4. timed A/B: plain vs eagle vs ngram ---------------------------
+    c = 8
+    bench_prompts = [chain(lo + rng.randrange(args.chain_vocab), 24)
+                     for _ in range(c)]
+    results = {}
+    for name, spec_cfg in [
+        ("plain", None),
+        ("eagle", {"method": "eagle", "num_draft_tokens": 3,
+                   "draft_dir": str(draft_dir)}),
+        ("ngram", {"method": "ngram", "num_draft_tokens": 3}),
+    ]:
+        e = LLMEngine(EngineConfig(model=str(ckpt), device=dev,
+                                   max_model_len=1024, max_num_seqs=c,
+                                   gpu_memory_utilization=0.35,
+                                   enforce_random_weights=False,
+                                   kv_cache_blocks=2048 if args.dry else None,
+                                   speculative=spec_cfg))
+        tps, per_step = timed_decode(e, bench_prompts, osl=256,
+                                     steps=20 if args.dry else 60)
+        results[name] = tps
+        print(f"{name}: {tps:,.0f} tok/s ({per_step:.1f} tokens/step)",
+              flush=True)
+        del e
+        if not args.dry:
+            torch.cuda.empty_cache()
+    print(json.dumps({
+        "target_follows_chain": round(follow, 4),
+        "tok_s": {k: round(v, 1) for k, v in results.items()},
+        "eagle_speedup": round(results["eagle"] / results["plain"], 3),
+        "ngram_speedup": round(results["ngram"] / results["plain"], 3),
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    main()
