@@ -248,7 +248,8 @@ def main():
         data.append((torch.tensor(sq, dtype=torch.long, device=dev), H))
     val, data = data[:8], data[8:]
     draft = DraftTrainer(spec, dev)
-    dopt = torch.optim.Adam(draft.parameters(), lr=1e-3)
+    dopt = torch.optim.Adam(draft.parameters(),
+                            lr=1e-3 if args.dry else 2.5e-4)
     embed = eng.runner.model.embed.float()
     lm_head = eng.runner.model.lm_head.float()
     mcs = eng.runner.model.cos_sin.float()
