@@ -46,15 +46,18 @@ class TorchLlama(torch.nn.Module):
         self.lm_head = P(spec.vocab_size, h)
         self.final_norm = torch.nn.Parameter(torch.ones(h, device=device))
         self.layers = torch.nn.ModuleList()
+        # residual-branch outputs scaled by 1/sqrt(2L): keeps the stream
+        # variance bounded so depth-32 trains from scratch
+        rs = 0.02 / (2 * spec.num_layers) ** 0.5
         for _ in range(spec.num_layers):
             m = torch.nn.Module()
             m.q = P(spec.num_heads * d, h)
             m.k = P(spec.num_kv_heads * d, h)
             m.v = P(spec.num_kv_heads * d, h)
-            m.o = P(h, spec.num_heads * d)
+            m.o = P(h, spec.num_heads * d, std=rs)
             m.gate = P(spec.intermediate_size, h)
             m.up = P(spec.intermediate_size, h)
-            m.down = P(h, spec.intermediate_size)
+            m.down = P(h, spec.intermediate_size, std=rs)
             m.input_norm = torch.nn.Parameter(torch.ones(h, device=device))
             m.post_norm = torch.nn.Parameter(torch.ones(h, device=device))
             self.layers.append(m)
@@ -206,8 +209,13 @@ def main():
     cos_sin = build_cos_sin_cache(spec.head_dim, spec.head_dim,
                                   spec.max_position_embeddings,
                                   base=spec.rope_theta).to(dev)
-    lr = 1e-3 if args.dry else 2.5e-4  # the 6-layer/2048-h target diverges at 1e-3
+    lr = 1e-3 if args.dry else 6e-4
     opt = torch.optim.Adam(target.parameters(), lr=lr)
+
+    def set_lr(step):  # 100-step linear warmup, flat after
+        f = min(1.0, (step + 1) / 100.0) if not args.dry else 1.0
+        for g in opt.param_groups:
+            g["lr"] = lr * f
     t0 = time.time()
     L = 96 if args.dry else 192
     import contextlib
@@ -217,6 +225,7 @@ def main():
     for step in range(args.target_steps):
         start = lo + rng.randrange(args.chain_vocab)
         toks = torch.tensor(chain(start, L), dtype=torch.long, device=dev)
+        set_lr(step)
         with amp:
             logits = target(toks[:-1], cos_sin)
             loss = F.cross_entropy(logits.float(), toks[1:])
