@@ -57,3 +57,31 @@ def test_offload_refused_when_too_little_resident():
              "gpus_per_replica": 1,
              "backend_parameters": {"cpu_offload": True}}
     assert select_candidates(model, [worker], []) == []
+
+
+def test_offload_composes_with_w4_runtime():
+    """W4-packed layers have freed (0-element) parameters; the offload
+    streamer must skip them and still stream any remaining bf16 weights."""
+    p = SamplingParams(max_tokens=6, ignore_eos=True)
+    eng = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                 kv_cache_blocks=64, quantize_runtime="w4",
+                                 cpu_offload_gb=1.0))
+    ref = LLMEngine(EngineConfig(model="tiny", device="cpu",
+                                 kv_cache_blocks=64, quantize_runtime="w4"))
+    out = eng.generate([[3, 4, 5]], p)[0]
+    assert out == ref.generate([[3, 4, 5]], p)[0]
+
+
+def test_engine_server_accepts_w4_and_offload_backend_params():
+    """serve_manager passes quantize_runtime / cpu_offload_gb through
+    backend_parameters into EngineConfig (engine_server filters on
+    dataclass fields)."""
+    from gpustack_amd.engine import EngineConfig
+
+    extra = {"quantize_runtime": "w4", "cpu_offload_gb": 1.5,
+             "bogus_key": 1}
+    kept = {k: v for k, v in extra.items()
+            if k in EngineConfig.__dataclass_fields__}
+    assert kept == {"quantize_runtime": "w4", "cpu_offload_gb": 1.5}
+    cfg = EngineConfig(model="tiny", device="cpu", **kept)
+    assert cfg.quantize_runtime == "w4" and cfg.cpu_offload_gb == 1.5
