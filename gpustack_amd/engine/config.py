@@ -220,6 +220,11 @@ class EngineConfig:
     # weights live pinned in host DRAM and stream to a double-buffered
     # device staging area one layer ahead of use (engine/offload.py)
     cpu_offload_gb: float = 0.0
+    # uneven pipeline-stage layer counts (len == pp_size, sums to
+    # num_layers) — the native analog of the reference's per-GPU GGUF
+    # tensor_split proportions; None = even split (remainder to early
+    # stages so the sampling stage carries less)
+    pp_partition: list[int] | None = None
     # LoRA adapters merged into the weights at load (reference lora_list)
     lora_dirs: list[str] = field(default_factory=list)
     # GGUF checkpoint execution: dequantized to bf16 at load (utils/gguf.py)

@@ -442,8 +442,17 @@ class LlamaForCausalLM(nn.Module):
         # go to the EARLY stages so the last stage (which also runs the
         # lm_head GEMM and sampling) carries less per-step work
         pp, pr = comm.pp_size, comm.pp_rank
-        counts = [spec.num_layers // pp + (1 if i < spec.num_layers % pp else 0)
-                  for i in range(pp)]
+        part = getattr(cfg, "pp_partition", None)
+        if part and len(part) == pp and sum(part) == spec.num_layers:
+            # explicit per-stage layer counts (the native analog of the
+            # reference's per-GPU GGUF tensor_split proportions,
+            # gguf_resource_fit_selector.py:370-441): the scheduler sizes
+            # stages proportionally to each GPU's free VRAM
+            counts = list(part)
+        else:
+            counts = [spec.num_layers // pp
+                      + (1 if i < spec.num_layers % pp else 0)
+                      for i in range(pp)]
         self.layer_offset = sum(counts[:pr])
         self.num_local_layers = counts[pr]
         first, last = comm.is_first_stage, comm.is_last_stage

@@ -71,6 +71,45 @@ class PlacementScheduler:
         for iid in pending + stale:
             self.schedule_one(iid)
 
+    @staticmethod
+    def _pp_partition(model_d, spec, cand, workers, others):
+        """Per-stage layer counts proportional to each stage's free VRAM
+        when backend_parameters requests pipeline parallelism over GPUs
+        with unequal headroom. None for even/trivial cases (the engine's
+        default even split applies)."""
+        bp = model_d.get("backend_parameters") or {}
+        try:
+            pp = int(bp.get("pp_size", 1) or 1)
+        except (TypeError, ValueError):
+            return None
+        gpus = cand.gpu_indexes
+        if (pp <= 1 or spec is None or cand.subordinates
+                or not gpus or len(gpus) % pp):
+            return None
+        from .policies import worker_allocatable
+
+        alloc = worker_allocatable(cand.worker, others)
+        tp = len(gpus) // pp
+        stage_free = [sum(alloc.get(g, 0) for g in gpus[s * tp:(s + 1) * tp])
+                      for s in range(pp)]
+        total = sum(stage_free)
+        if total <= 0:
+            return None
+        n = spec.num_layers
+        part = [max(1, int(n * f / total)) for f in stage_free]
+        # fix rounding: add/remove from the roomiest/most-loaded stages
+        while sum(part) < n:
+            part[stage_free.index(max(stage_free))] += 1
+        while sum(part) > n:
+            i = max(range(pp), key=lambda j: part[j])
+            if part[i] <= 1:
+                return None
+            part[i] -= 1
+        if part == [n // pp + (1 if i < n % pp else 0) for i in range(pp)]:
+            return None  # even: engine default
+        return part
+
+
     def schedule_one(self, instance_id: int) -> bool:
         with get_session() as s:
             inst = s.get(ModelInstance, instance_id)
@@ -118,6 +157,12 @@ class PlacementScheduler:
                 # schemas/models.py:623-630 offload_layers)
                 inst.computed_resource_claim["offload_gb"] = cand.offload_gb
                 inst.computed_resource_claim["offload_layers"] = cand.offload_layers
+            part = self._pp_partition(model_d, spec, cand, workers, others)
+            if part:
+                # uneven pipeline stages sized to each GPU group's free
+                # VRAM — the native analog of the reference's per-GPU GGUF
+                # tensor_split proportions (gguf selector :370-441)
+                inst.computed_resource_claim["pp_partition"] = part
             if cand.subordinates:
                 # cross-worker TP: rank layout + rendezvous port
                 # (reference: serve_manager.py:1643-1739 port bands +

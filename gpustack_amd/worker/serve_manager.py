@@ -253,6 +253,9 @@ class ServeManager:
             # scheduler placed this instance with partial CPU offload
             # (engine/offload.py streams the tail layers from host DRAM)
             bp.setdefault("cpu_offload_gb", claim["offload_gb"])
+        if claim.get("pp_partition"):
+            # uneven pipeline stages (per-GPU tensor_split analog)
+            bp.setdefault("pp_partition", claim["pp_partition"])
         if model.get("lora_list"):
             bp.setdefault("lora_dirs", model["lora_list"])
         if model.get("lora_adapters"):

@@ -386,3 +386,87 @@ def test_pp2_dynamic_lora_matches_single():
 def test_pp2_moe_matches_single_rank():
     """MoE layers partition across pipeline stages like dense ones."""
     assert _run_pp(2, model="tiny-moe") == _single_proc_result("tiny-moe")
+
+
+def _pp_part_rank_main(rank: int, pp: int, port: int, out_path: str,
+                       partition):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch
+
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_parallel
+
+    torch.set_num_threads(1)
+    comm = init_parallel(1, pp, rank, master_port=port, backend="gloo")
+    cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, pp_partition=partition)
+    cfg.spec.num_layers = 4  # tiny has 2; uneven [3,1] needs 4
+    eng = LLMEngine(cfg, comm)
+    if partition:
+        assert eng.runner.model.num_local_layers == partition[rank]
+    results, rids = {}, []
+    if rank == 0:
+        rids = [eng.add_request(p, SamplingParams(max_tokens=6,
+                                                  ignore_eos=True))
+                for p in PROMPTS]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def _run_pp_part(pp: int, partition) -> list[list[int]]:
+    port = _free_port()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pp_part_rank_main,
+                         args=(r, pp, port, out_path, partition))
+             for r in range(pp)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    with open(out_path) as f:
+        return json.load(f)
+
+
+def test_pp_partition_uneven_exact():
+    """Uneven per-stage layer counts (pp_partition, the per-GPU
+    tensor_split analog) produce output identical to the even split."""
+    out_even = _run_pp_part(2, None)
+    out_uneven = _run_pp_part(2, [3, 1])
+    assert out_even == out_uneven
+
+
+def test_scheduler_pp_partition_proportional():
+    from gpustack_amd.scheduler.policies import Candidate
+    from gpustack_amd.scheduler.scheduler import PlacementScheduler
+    from gpustack_amd.engine.config import PRESETS
+
+    spec = PRESETS["llama-3-8b"]  # 32 layers
+    worker = {"id": 1, "status": {"gpu_devices": [
+        {"index": 0, "memory": {"total": 100 << 30}},
+        {"index": 1, "memory": {"total": 300 << 30}},
+    ]}, "system_reserved": {}}
+    cand = Candidate(worker, [0, 1])
+    model_d = {"backend_parameters": {"pp_size": 2}}
+    part = PlacementScheduler._pp_partition(model_d, spec, cand, [worker], [])
+    assert part is not None and sum(part) == 32
+    assert part[1] > part[0]  # roomier stage holds more layers
+    # even headroom -> engine default (None)
+    worker2 = {"id": 1, "status": {"gpu_devices": [
+        {"index": 0, "memory": {"total": 200 << 30}},
+        {"index": 1, "memory": {"total": 200 << 30}},
+    ]}, "system_reserved": {}}
+    assert PlacementScheduler._pp_partition(
+        model_d, spec, Candidate(worker2, [0, 1]), [worker2], []) is None
