@@ -143,6 +143,15 @@ PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=40960, qk_norm=True, eos_token_id=151645,
         num_experts=128, num_experts_per_tok=8, moe_intermediate_size=768,
     ),
+    # Qwen3-235B-A22B (the BASELINE.md 8-GPU MoE row): same Qwen3-MoE
+    # graph as 30B-A3B, bigger dims — serves TP8 over xGMI
+    "qwen3-235b-a22b": ModelSpec(
+        architecture="Qwen3MoeForCausalLM", vocab_size=151936,
+        hidden_size=4096, intermediate_size=12288, num_layers=94,
+        num_heads=64, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
+        max_position_embeddings=40960, qk_norm=True, eos_token_id=151645,
+        num_experts=128, num_experts_per_tok=8, moe_intermediate_size=1536,
+    ),
     # Mixtral 8x7B: 8-expert top-2 MoE on the llama graph
     "mixtral-8x7b": ModelSpec(
         architecture="MixtralForCausalLM", vocab_size=32000,

@@ -35,6 +35,19 @@ def _gen(shape, seed_key: str, base_seed: int, dtype, device, std=0.02):
     return t.to(dtype).to(device)
 
 
+def _kv_slice(full, rank: int, tp: int, n_kv: int, d: int):
+    """KV shard rows for `rank`: plain slice while tp <= num_kv_heads;
+    beyond that, KV heads are REPLICATED — groups of tp/num_kv_heads ranks
+    share one head (standard practice for high-TP GQA, e.g. TP8 over a
+    4-KV-head model like Qwen3-235B-A22B)."""
+    if tp <= n_kv:
+        per = n_kv // tp
+        return full[rank * per * d:(rank + 1) * per * d]
+    repl = tp // n_kv
+    head = rank // repl
+    return full[head * d:(head + 1) * d]
+
+
 def random_init(model, cfg: EngineConfig) -> None:
     spec = model.spec
     tp, rank = cfg.tp_size, cfg.tp_rank
@@ -58,8 +71,8 @@ def random_init(model, cfg: EngineConfig) -> None:
         v_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.v", seed, dtype, device)
         layer.attn.qkv_w.copy_(torch.cat([
             q_full[rank * hq * d:(rank + 1) * hq * d],
-            k_full[rank * hkv * d:(rank + 1) * hkv * d],
-            v_full[rank * hkv * d:(rank + 1) * hkv * d],
+            _kv_slice(k_full, rank, tp, spec.num_kv_heads, d),
+            _kv_slice(v_full, rank, tp, spec.num_kv_heads, d),
         ]))
         if layer.attn.qkv_b is not None:
             qb = _gen((spec.num_heads * d,), f"{li}.qb", seed, dtype, device)
@@ -67,8 +80,10 @@ def random_init(model, cfg: EngineConfig) -> None:
             vb = _gen((spec.num_kv_heads * d,), f"{li}.vb", seed, dtype, device)
             layer.attn.qkv_b.copy_(torch.cat([
                 qb[rank * hq * d:(rank + 1) * hq * d],
-                kb[rank * hkv * d:(rank + 1) * hkv * d],
-                vb[rank * hkv * d:(rank + 1) * hkv * d],
+                _kv_slice(kb.unsqueeze(1), rank, tp, spec.num_kv_heads,
+                          d).squeeze(1),
+                _kv_slice(vb.unsqueeze(1), rank, tp, spec.num_kv_heads,
+                          d).squeeze(1),
             ]))
         o_full = _gen((spec.hidden_size, spec.num_heads * d), f"{li}.o", seed, dtype, device)
         layer.attn.o_w.copy_(o_full[:, rank * hq * d:(rank + 1) * hq * d])
@@ -151,14 +166,18 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
         li = off + local_i
         p = f"{pre}layers.{li}."
         q = row_shard(get(p + "self_attn.q_proj.weight"), hq * d)
-        k = row_shard(get(p + "self_attn.k_proj.weight"), hkv * d)
-        v = row_shard(get(p + "self_attn.v_proj.weight"), hkv * d)
+        k = _kv_slice(get(p + "self_attn.k_proj.weight"), rank, tp,
+                      spec.num_kv_heads, d)
+        v = _kv_slice(get(p + "self_attn.v_proj.weight"), rank, tp,
+                      spec.num_kv_heads, d)
         layer.attn.qkv_w.copy_(torch.cat([q, k, v]))
         if layer.attn.qkv_b is not None:
             layer.attn.qkv_b.copy_(torch.cat([
                 row_shard(get(p + "self_attn.q_proj.bias"), hq * d),
-                row_shard(get(p + "self_attn.k_proj.bias"), hkv * d),
-                row_shard(get(p + "self_attn.v_proj.bias"), hkv * d),
+                _kv_slice(get(p + "self_attn.k_proj.bias").unsqueeze(1),
+                          rank, tp, spec.num_kv_heads, d).squeeze(1),
+                _kv_slice(get(p + "self_attn.v_proj.bias").unsqueeze(1),
+                          rank, tp, spec.num_kv_heads, d).squeeze(1),
             ]))
         o = get(p + "self_attn.o_proj.weight")
         layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])

@@ -98,3 +98,27 @@ def test_tp2_moe_deterministic():
     b = _run_tp2("tiny-moe")
     assert a == b, f"{a} != {b}"
     assert all(len(x) == 6 for x in a)
+
+
+def _run_tpn(model: str, world: int) -> list[list[int]]:
+    port = _free_port()
+    out_path = tempfile.mktemp(suffix=".json")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_tp_rank_main,
+                         args=(r, world, port, out_path, model))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0, f"rank process exited {p.exitcode}"
+    with open(out_path) as f:
+        return json.load(f)
+
+
+@pytest.mark.timeout(600)
+def test_tp4_kv_head_replication_matches_tp1():
+    """tp > num_kv_heads: KV heads replicate across rank groups (the
+    Qwen3-235B TP8-over-4-KV-heads case, scaled down: tiny has 2 KV heads,
+    TP4 => 2 ranks share each head). Output must equal TP1."""
+    assert _run_tpn("tiny", 4) == _single_proc_result("tiny")
