@@ -25,7 +25,11 @@ BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
 
 
 class DecodeGraphRunner:
-    def __init__(self, runner):
+    def __init__(self, runner, want_hidden: bool = False):
+        # want_hidden: also capture the post-final-norm hidden states —
+        # the EAGLE verify forward needs them to condition the next draft
+        # window (runner.execute's return_both path)
+        self.want_hidden = want_hidden
         self.runner = runner
         cfg = runner.cfg
         dev = runner.device
@@ -88,11 +92,13 @@ class DecodeGraphRunner:
         pool = torch.cuda.graph_pool_handle()
         for bs in reversed(self.buckets):  # largest first reserves the pool
             meta = self._meta(bs)
-            model(self.tokens[:bs], meta, kv)  # eager warmup (blas workspaces)
+            model(self.tokens[:bs], meta, kv,
+                  return_both=self.want_hidden)  # eager warmup (blas workspaces)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool):
-                out = model(self.tokens[:bs], meta, kv)
+                out = model(self.tokens[:bs], meta, kv,
+                            return_both=self.want_hidden)
             self.graphs[bs] = g
             self.outs[bs] = out
             self.metas[bs] = meta  # keep every referenced tensor alive
@@ -159,6 +165,9 @@ class DecodeGraphRunner:
                 self.h_bt[dirty_lo:dirty_hi + 1, :max(maxb, 1)], non_blocking=True
             )
         self.graphs[bucket].replay()
+        if self.want_hidden:
+            logits, hidden = self.outs[bucket]
+            return logits[:bs], hidden[:bs]
         return self.outs[bucket][:bs]
 
 

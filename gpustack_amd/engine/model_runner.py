@@ -350,10 +350,15 @@ class ModelRunner:
                                 and os.environ.get("GPUSTACK_AMD_FUSED_MOE",
                                                    "1") == "1"
                                 and ops.hip_available()))
-            if graphs_enabled() and self.eagle is None \
-                    and moe_graph_ok and self.comm.pp_size == 1 \
+            if graphs_enabled() and moe_graph_ok \
+                    and self.comm.pp_size == 1 \
                     and self.model.offload is None:
-                self.graph_runner = DecodeGraphRunner(self)
+                # EAGLE decode graphs since r2: the verify forward (1+k
+                # rows/seq) captures like any decode batch, additionally
+                # returning the hidden states the next draft window needs;
+                # the draft micro-steps stay eager (1-layer head — small)
+                self.graph_runner = DecodeGraphRunner(
+                    self, want_hidden=self.eagle is not None)
                 self.graph_runner.capture()
         return self.kv
 
@@ -590,14 +595,17 @@ class ModelRunner:
     @torch.inference_mode()
     def execute(self, batch: ScheduledBatch) -> list[int]:
         self.last_hidden = None
-        if (self.graph_runner is not None and self.graph_runner.can_run(batch)
-                and not self.batch_uses_lora(batch)):
-            logits = self.graph_runner.run(batch)
-        elif self.eagle is not None:
+        if self.eagle is not None:
             # draft-model speculative: the verify step must also surface the
-            # target hidden states that condition the next draft window
-            tokens, meta = self._meta(batch)
-            if batch.is_prefill:
+            # target hidden states that condition the next draft window.
+            # Decode-shaped verify batches replay the captured hipGraph
+            # (want_hidden=True); prefill seeding stays eager.
+            if (not batch.is_prefill and self.graph_runner is not None
+                    and self.graph_runner.can_run(batch)
+                    and not self.batch_uses_lora(batch)):
+                logits, self.last_hidden = self.graph_runner.run(batch)
+            elif batch.is_prefill:
+                tokens, meta = self._meta(batch)
                 from ..models.llama import qlinear
 
                 hidden_all = self.model(tokens, meta, self.kv, return_hidden=True)
@@ -605,8 +613,12 @@ class ModelRunner:
                                  self.model.lm_head, self.model.lm_head_pack)
                 self.last_hidden = hidden_all
             else:
+                tokens, meta = self._meta(batch)
                 logits, self.last_hidden = self.model(tokens, meta, self.kv,
                                                       return_both=True)
+        elif (self.graph_runner is not None and self.graph_runner.can_run(batch)
+                and not self.batch_uses_lora(batch)):
+            logits = self.graph_runner.run(batch)
         else:
             tokens, meta = self._meta(batch)
             logits = self.model(tokens, meta, self.kv)
