@@ -292,7 +292,11 @@ class EagleProposer:
         return x
 
     def _argmax_token(self, hidden) -> torch.Tensor:
-        return F.linear(hidden, self.model.lm_head).argmax(dim=-1)
+        from ..models.llama import qlinear
+
+        # routes through the W4 runtime pack when lm_head is packed
+        return qlinear(hidden, self.model.lm_head,
+                       self.model.lm_head_pack).argmax(dim=-1)
 
     def _bt_tensor(self, states: list[_DraftState], rows_per: list[int]):
         maxb = max(len(st.block_table) for st in states)

@@ -186,7 +186,9 @@ class PlacementScheduler:
                     for i in others
                 }
                 mp = 45000 + (inst.id * 4) % 1000
-                while mp in taken:
+                for _ in range(250):  # band holds 250 slots of 4
+                    if mp not in taken:
+                        break
                     mp = 45000 + (mp - 45000 + 4) % 1000
                 inst.distributed_servers = {
                     "tp": tp,
