@@ -280,11 +280,16 @@ class MoEMLP(nn.Module):
         #  - decode-shaped torch fallback: ONE padded strided-batched GEMM
         #    pair (pays a per-layer counts.max() host sync)
         #  - prefill-shaped: per-expert loop GEMMs (large per-expert work)
-        if self._fused_ok(x) and flat_exp.numel() <= 32 * self.e:
-            # decode-shaped only: the fused kernels re-stream each expert's
-            # weight panel per 16-row m-tile, which is free when experts
-            # hold <=32 rows but ruinous at prefill occupancy (hundreds of
-            # rows/expert) — prefill keeps the per-expert hipBLASLt loop
+        # decode-shaped batches take the fused kernels (which re-stream
+        # each expert's weight panel per 16-row m-tile — free when experts
+        # hold <=32 rows, ruinous at prefill occupancy, where the
+        # per-expert hipBLASLt loop wins). Under hipGraph CAPTURE the
+        # fused path is forced for ANY size: the torch fallbacks sync with
+        # the host per layer, which capture cannot record (large decode
+        # buckets on many-expert models would otherwise crash at init).
+        capturing = x.is_cuda and torch.cuda.is_current_stream_capturing()
+        if self._fused_ok(x) and (flat_exp.numel() <= 32 * self.e
+                                  or capturing):
             out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w32,
                                        ones_i32)
             return self.comm.all_reduce(out)
