@@ -18,6 +18,11 @@ from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 @pytest.mark.parametrize("seed", [0, 1, 2])
 def test_scheduler_fuzz(seed):
     rng = random.Random(seed)
+    # r2: odd seeds also run packed-W4 weights + CPU weight offload so
+    # the fuzz exercises qlinear dispatch and the layer streamer under
+    # the same allocator pressure
+    extra = ({"quantize_runtime": "w4", "cpu_offload_gb": 0.001}
+             if seed % 2 else {})
     eng = LLMEngine(EngineConfig(
         model="tiny", device="cpu", max_model_len=192,
         kv_cache_blocks=20,                # tiny pool: constant pressure
@@ -25,6 +30,7 @@ def test_scheduler_fuzz(seed):
         enable_chunked_prefill=True, enable_prefix_caching=True,
         kv_offload_gb=0.001,               # a few host blocks
         admission_min_seqs=1, admission_max_wait_s=0.0,
+        **extra,
     ))
     alloc = eng.scheduler.kv.allocator
     shared_prefix = [rng.randrange(2, 500) for _ in range(40)]
