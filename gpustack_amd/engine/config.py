@@ -35,6 +35,18 @@ class ModelSpec:
     num_experts_per_tok: int = 0
     moe_intermediate_size: int = 0
     norm_topk_prob: bool = True
+    # GLM-4.5 / DeepSeek-style MoE extensions (Glm4MoeForCausalLM):
+    # sigmoid router scores + learned correction bias + optional grouped
+    # top-k; a SHARED dense expert added to every token; the first k
+    # layers use a plain dense MLP
+    router_mode: str = "softmax"          # "softmax" | "sigmoid_bias"
+    n_shared_experts: int = 0
+    first_k_dense_replace: int = 0
+    routed_scaling_factor: float = 1.0
+    n_group: int = 1
+    topk_group: int = 1
+    # rotary applied to the first head_dim*partial_rotary_factor dims only
+    partial_rotary_factor: float = 1.0
 
     @property
     def gqa_ratio(self) -> int:
@@ -47,13 +59,21 @@ class ModelSpec:
         h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
         qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
         o = self.num_heads * self.head_dim * h
+        attn_norm = qkv + o + 2 * h
+        dense_mlp = 3 * h * i
         if self.num_experts > 0:
-            mlp = self.num_experts * (3 * h * self.moe_intermediate_size)                 + self.num_experts * h  # router
+            moe_mlp = (self.num_experts * 3 * h * self.moe_intermediate_size
+                       + self.num_experts * h  # router
+                       + 3 * h * self.moe_intermediate_size
+                       * self.n_shared_experts)
+            k_dense = min(self.first_k_dense_replace, self.num_layers)
+            total_layers = (attn_norm * self.num_layers
+                            + dense_mlp * k_dense
+                            + moe_mlp * (self.num_layers - k_dense))
         else:
-            mlp = 3 * h * i
-        per_layer = qkv + o + mlp + 2 * h
+            total_layers = (attn_norm + dense_mlp) * self.num_layers
         emb = v * h * (1 if self.tie_word_embeddings else 2)
-        return (per_layer * self.num_layers + emb + h) * dtype_size
+        return (total_layers + emb + h) * dtype_size
 
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelSpec":
@@ -78,15 +98,28 @@ class ModelSpec:
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=arch.startswith("Qwen2"),
-            qk_norm=arch.startswith("Qwen3"),
+            qk_norm=(arch.startswith("Qwen3")
+                     or bool(cfg.get("use_qk_norm", False))),
             eos_token_id=eos,
             num_experts=cfg.get("num_experts",
-                                cfg.get("num_local_experts", 0)) or 0,
+                                cfg.get("num_local_experts",
+                                        cfg.get("n_routed_experts", 0))) or 0,
             num_experts_per_tok=cfg.get("num_experts_per_tok", 0) or 0,
             moe_intermediate_size=cfg.get("moe_intermediate_size",
                                           cfg.get("intermediate_size", 0))
-            if (cfg.get("num_experts") or cfg.get("num_local_experts")) else 0,
+            if (cfg.get("num_experts") or cfg.get("num_local_experts")
+                or cfg.get("n_routed_experts")) else 0,
             norm_topk_prob=bool(cfg.get("norm_topk_prob", True)),
+            router_mode=("sigmoid_bias" if arch.startswith("Glm4Moe")
+                         else "softmax"),
+            n_shared_experts=cfg.get("n_shared_experts", 0) or 0,
+            first_k_dense_replace=cfg.get("first_k_dense_replace", 0) or 0,
+            routed_scaling_factor=cfg.get("routed_scaling_factor", 1.0) or 1.0,
+            n_group=cfg.get("n_group", 1) or 1,
+            topk_group=cfg.get("topk_group", 1) or 1,
+            partial_rotary_factor=(cfg.get("partial_rotary_factor")
+                                   or (cfg.get("rope_parameters") or {})
+                                   .get("partial_rotary_factor") or 1.0),
         )
 
     @classmethod
@@ -151,6 +184,17 @@ PRESETS: dict[str, ModelSpec] = {
         num_heads=64, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
         max_position_embeddings=40960, qk_norm=True, eos_token_id=151645,
         num_experts=128, num_experts_per_tok=8, moe_intermediate_size=1536,
+    ),
+    # GLM-4.5-Air: Glm4Moe graph — sigmoid+bias
+    # router, 1 shared expert, first layer dense, partial rotary 0.5
+    "glm-4.5-air": ModelSpec(
+        architecture="Glm4MoeForCausalLM", vocab_size=151552,
+        hidden_size=4096, intermediate_size=10944, num_layers=46,
+        num_heads=96, num_kv_heads=8, head_dim=128, rope_theta=1000000.0,
+        max_position_embeddings=131072, qk_norm=True, eos_token_id=151329,
+        num_experts=128, num_experts_per_tok=8, moe_intermediate_size=1408,
+        router_mode="sigmoid_bias", n_shared_experts=1,
+        first_k_dense_replace=1, partial_rotary_factor=0.5,
     ),
     # Mixtral 8x7B: 8-expert top-2 MoE on the llama graph
     "mixtral-8x7b": ModelSpec(

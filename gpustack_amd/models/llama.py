@@ -103,6 +103,8 @@ class Attention(nn.Module):
         self.hq = spec.num_heads // tp_size
         self.hkv = max(1, spec.num_kv_heads // tp_size)
         self.d = spec.head_dim
+        # GLM-style partial rotary: only the first rot_dim dims rotate
+        self.rot_dim = int(spec.head_dim * spec.partial_rotary_factor)
         self.scale = self.d ** -0.5
         h = spec.hidden_size
         qkv_out = (self.hq + 2 * self.hkv) * self.d
@@ -141,7 +143,8 @@ class Attention(nn.Module):
             v = v.contiguous()
             ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
             ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
-        ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d, self.d)
+        ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
+                             self.rot_dim)
         ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
         out = torch.empty(T, self.hq, self.d, dtype=q.dtype, device=q.device)
         if meta.is_prefill:
@@ -224,6 +227,23 @@ class MoEMLP(nn.Module):
             torch.empty(self.e, 2 * self.i, h, dtype=dtype), requires_grad=False)
         self.down_w = nn.Parameter(
             torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
+        # GLM/DeepSeek-style extensions (spec.router_mode "sigmoid_bias"):
+        # learned correction bias on the routing scores and a SHARED dense
+        # expert applied to every token (TP shards its intermediate)
+        if spec.router_mode == "sigmoid_bias":
+            self.router_bias = nn.Parameter(
+                torch.zeros(self.e, dtype=torch.float32), requires_grad=False)
+        else:
+            self.router_bias = None
+        if spec.n_shared_experts > 0:
+            si = spec.moe_intermediate_size * spec.n_shared_experts // tp_size
+            self.shared_i = si
+            self.shared_gate_up_w = nn.Parameter(
+                torch.empty(2 * si, h, dtype=dtype), requires_grad=False)
+            self.shared_down_w = nn.Parameter(
+                torch.empty(h, si, dtype=dtype), requires_grad=False)
+        else:
+            self.shared_gate_up_w = None
 
     def _routing_consts(self, T: int, device):
         """Per-T cached routing constants — rebuilt tensors per layer per
@@ -256,7 +276,27 @@ class MoEMLP(nn.Module):
             self._router_w_f32 = self.router_w.float()
             rw = self._router_w_f32
         logits = F.linear(x.float(), rw)                          # [T, E]
-        if self.spec.norm_topk_prob:
+        if self.spec.router_mode == "sigmoid_bias":
+            # GLM-4.5/DeepSeek routing (HF Glm4MoeTopkRouter): sigmoid
+            # scores; the learned bias only influences the CHOICE, the
+            # routing weight is the raw score; optional grouped top-k
+            scores = torch.sigmoid(logits)
+            choice = scores + self.router_bias
+            ng, tg = self.spec.n_group, self.spec.topk_group
+            if ng > 1:
+                gs = (choice.view(T, ng, self.e // ng)
+                      .topk(2, dim=-1)[0].sum(dim=-1))
+                gi = torch.topk(gs, k=tg, dim=-1, sorted=False)[1]
+                gmask = torch.zeros_like(gs).scatter_(1, gi, 1)
+                smask = gmask.unsqueeze(-1).expand(T, ng, self.e // ng)                     .reshape(T, self.e)
+                choice = choice.masked_fill(~smask.bool(), float("-inf"))
+            experts = torch.topk(choice, k=self.top_k, dim=-1,
+                                 sorted=False)[1]
+            weights = scores.gather(1, experts)
+            if self.spec.norm_topk_prob:
+                weights = weights / (weights.sum(dim=-1, keepdim=True) + 1e-20)
+            weights = weights * self.spec.routed_scaling_factor
+        elif self.spec.norm_topk_prob:
             # softmax over all experts -> top-k -> renormalize == softmax
             # restricted to the top-k logits (Qwen3-MoE default, Mixtral)
             weights, experts = torch.topk(logits, self.top_k, dim=-1)
@@ -292,7 +332,7 @@ class MoEMLP(nn.Module):
                                   or capturing):
             out = self._fused_dispatch(x, flat_exp, flat_tok, flat_w32,
                                        ones_i32)
-            return self.comm.all_reduce(out)
+            return self.comm.all_reduce(self._add_shared(x, out))
         flat_w = flat_w32.to(x.dtype)
         contrib = x.new_zeros(T * self.top_k, x.shape[1])
         # prefill-shaped on GPU with many experts: padded-bmm. The r2
@@ -321,7 +361,18 @@ class MoEMLP(nn.Module):
         else:
             self._loop_dispatch(x, contrib, flat_exp, flat_tok, flat_w)
         out = contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
-        return self.comm.all_reduce(out)
+        return self.comm.all_reduce(self._add_shared(x, out))
+
+    def _add_shared(self, x, out):
+        """Shared dense expert (GLM-4.5: every token, added to the routed
+        mix BEFORE the TP all-reduce so one collective covers both)."""
+        if self.shared_gate_up_w is None:
+            return out
+        gu = F.linear(x, self.shared_gate_up_w)
+        act = torch.empty(x.shape[0], self.shared_i, dtype=x.dtype,
+                          device=x.device)
+        ops.silu_and_mul(act, gu)
+        return out + F.linear(act, self.shared_down_w)
 
     def _fused_ok(self, x) -> bool:
         import os
@@ -405,13 +456,16 @@ class MoEMLP(nn.Module):
 
 
 class DecoderLayer(nn.Module):
-    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator,
+                 dtype, layer_idx: int = 0):
         super().__init__()
         self.spec = spec
         self.attn = Attention(spec, tp_size, comm, dtype)
-        if spec.num_experts > 0:
+        if spec.num_experts > 0 and layer_idx >= spec.first_k_dense_replace:
             self.mlp = MoEMLP(spec, tp_size, comm, dtype)
         else:
+            # dense layer — all layers of a dense model, or the first
+            # first_k_dense_replace layers of a GLM/DeepSeek-style MoE
             self.mlp = MLP(spec, tp_size, comm, dtype)
         h = spec.hidden_size
         self.input_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
@@ -468,8 +522,9 @@ class LlamaForCausalLM(nn.Module):
         else:
             self.embed = None
         self.layers = nn.ModuleList(
-            [DecoderLayer(spec, cfg.tp_size, comm, dtype)
-             for _ in range(self.num_local_layers)]
+            [DecoderLayer(spec, cfg.tp_size, comm, dtype,
+                          layer_idx=self.layer_offset + i)
+             for i in range(self.num_local_layers)]
         )
         for i, layer in enumerate(self.layers):
             layer.attn.layer_idx = self.layer_offset + i
@@ -489,7 +544,9 @@ class LlamaForCausalLM(nn.Module):
         self.lm_head_pack: W4Pack | None = None  # W4 runtime (qlinear)
         self.offload = None  # CpuOffload streamer (engine/offload.py)
         cache = ops.build_cos_sin_cache(
-            spec.head_dim, spec.head_dim, cfg.max_model_len,
+            spec.head_dim,
+            int(spec.head_dim * spec.partial_rotary_factor),
+            cfg.max_model_len,
             base=spec.rope_theta, scaling=spec.rope_scaling,
         )
         self.register_buffer("cos_sin", cache.to(device), persistent=False)

@@ -90,7 +90,7 @@ def random_init(model, cfg: EngineConfig) -> None:
         if spec.qk_norm:
             layer.attn.q_norm.fill_(1.0)
             layer.attn.k_norm.fill_(1.0)
-        if spec.num_experts > 0:
+        if hasattr(layer.mlp, "router_w"):
             _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank)
         else:
             gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
@@ -109,6 +109,19 @@ def _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank):
     mi_loc = spec.moe_intermediate_size // tp
     layer.mlp.router_w.copy_(_gen((spec.num_experts, spec.hidden_size),
                                   f"{li}.router", seed, dtype, device))
+    if layer.mlp.router_bias is not None:
+        layer.mlp.router_bias.copy_(_gen((spec.num_experts,), f"{li}.rbias",
+                                         seed, torch.float32, device,
+                                         std=0.05))
+    if layer.mlp.shared_gate_up_w is not None:
+        si_full = spec.moe_intermediate_size * spec.n_shared_experts
+        si = si_full // tp
+        sg = _gen((si_full, spec.hidden_size), f"{li}.sgate", seed, dtype, device)
+        su = _gen((si_full, spec.hidden_size), f"{li}.sup", seed, dtype, device)
+        layer.mlp.shared_gate_up_w.copy_(torch.cat([
+            sg[rank * si:(rank + 1) * si], su[rank * si:(rank + 1) * si]]))
+        sd = _gen((spec.hidden_size, si_full), f"{li}.sdown", seed, dtype, device)
+        layer.mlp.shared_down_w.copy_(sd[:, rank * si:(rank + 1) * si])
     for e in range(spec.num_experts):
         gate = _gen((spec.moe_intermediate_size, spec.hidden_size),
                     f"{li}.e{e}.gate", seed, dtype, device)
@@ -184,11 +197,24 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
         if spec.qk_norm:
             layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
             layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
-        if spec.num_experts > 0:
+        if hasattr(layer.mlp, "router_w"):
             # MoE (Qwen3-MoE `mlp.gate`, Mixtral `block_sparse_moe.gate`)
             router = (p + "mlp.gate.weight" if p + "mlp.gate.weight" in tensors
                       else p + "block_sparse_moe.gate.weight")
             layer.mlp.router_w.copy_(get(router))
+            if layer.mlp.router_bias is not None                     and p + "mlp.gate.e_score_correction_bias" in tensors:
+                layer.mlp.router_bias.copy_(
+                    tensors[p + "mlp.gate.e_score_correction_bias"].float())
+            if layer.mlp.shared_gate_up_w is not None:
+                si = layer.mlp.shared_i
+                sg = get(p + "mlp.shared_experts.gate_proj.weight")
+                su = get(p + "mlp.shared_experts.up_proj.weight")
+                layer.mlp.shared_gate_up_w.copy_(torch.cat([
+                    sg[rank * si:(rank + 1) * si],
+                    su[rank * si:(rank + 1) * si]]))
+                sd = get(p + "mlp.shared_experts.down_proj.weight")
+                layer.mlp.shared_down_w.copy_(
+                    sd[:, rank * si:(rank + 1) * si])
             mi_loc = spec.moe_intermediate_size // tp
             mi = spec.moe_intermediate_size
             if p + "mlp.experts.gate_up_proj" in tensors:
