@@ -40,7 +40,49 @@ def build_cos_sin_cache(
     """
     half = rot_dim // 2
     inv_freq = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float64) * 2 / rot_dim))
-    if scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
+    attention_factor = 1.0
+    if scaling and scaling.get("rope_type", scaling.get("type")) == "yarn":
+        # YaRN (arXiv 2309.00071; HF _compute_yarn_parameters): blend
+        # interpolated and extrapolated inverse frequencies over a linear
+        # ramp between the beta_fast/beta_slow correction dims, and scale
+        # cos/sin by the attention factor (mscale)
+        import math
+
+        factor = scaling.get("factor")
+        orig = scaling["original_max_position_embeddings"]
+        if factor is None:
+            factor = max_pos / orig
+        beta_fast = scaling.get("beta_fast") or 32
+        beta_slow = scaling.get("beta_slow") or 1
+        truncate = scaling.get("truncate", True)
+
+        def get_mscale(scale, m=1):
+            return 1.0 if scale <= 1 else 0.1 * m * math.log(scale) + 1.0
+
+        attention_factor = scaling.get("attention_factor")
+        if attention_factor is None:
+            ms, msd = scaling.get("mscale"), scaling.get("mscale_all_dim")
+            if ms and msd:
+                attention_factor = float(get_mscale(factor, ms)
+                                         / get_mscale(factor, msd))
+            else:
+                attention_factor = get_mscale(factor)
+
+        def corr_dim(nrot):
+            return (rot_dim * math.log(orig / (nrot * 2 * math.pi))
+                    / (2 * math.log(base)))
+
+        low, high = corr_dim(beta_fast), corr_dim(beta_slow)
+        if truncate:
+            low, high = math.floor(low), math.ceil(high)
+        low, high = max(low, 0), min(high, rot_dim - 1)
+        if low == high:
+            high += 0.001
+        ramp = ((torch.arange(half, dtype=torch.float64) - low)
+                / (high - low)).clamp(0, 1)
+        extrap_f = 1 - ramp
+        inv_freq = (inv_freq / factor) * (1 - extrap_f) + inv_freq * extrap_f
+    elif scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
         factor = scaling["factor"]
         lo = scaling["low_freq_factor"]
         hi = scaling["high_freq_factor"]
@@ -56,6 +98,8 @@ def build_cos_sin_cache(
     t = torch.arange(max_pos, dtype=torch.float64)
     freqs = torch.outer(t, inv_freq)
     cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1).float()
+    if attention_factor != 1.0:
+        cache = cache * attention_factor
     return cache.to(device)
 
 

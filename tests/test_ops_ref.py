@@ -146,3 +146,27 @@ def test_build_prefill_tiles():
     assert ts.tolist() == [0, 0, 70, 134]
     assert tq.tolist() == [0, 64, 0, 0]
     assert tl.tolist() == [70, 70, 64, 10]
+
+
+def test_yarn_rope_matches_hf():
+    """YaRN rope scaling (arXiv 2309.00071) cache == HF transformers'
+    _compute_yarn_parameters for the same config."""
+    import torch
+    from transformers import Qwen2Config
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+    from gpustack_amd.ops.torch_ref import build_cos_sin_cache
+
+    scaling = {"rope_type": "yarn", "factor": 4.0,
+               "original_max_position_embeddings": 32768,
+               "beta_fast": 32, "beta_slow": 1}
+    cfg = Qwen2Config(hidden_size=1024, num_attention_heads=8,
+                      max_position_embeddings=131072,
+                      rope_parameters=dict(scaling, rope_theta=1000000.0))
+    inv, att = ROPE_INIT_FUNCTIONS["yarn"](cfg)
+    cache = build_cos_sin_cache(128, 128, 512, base=1000000.0,
+                                scaling=scaling)
+    t = torch.arange(512, dtype=torch.float32)
+    freqs = torch.outer(t, inv.float())
+    ref = torch.cat([freqs.cos(), freqs.sin()], -1) * att
+    assert (cache - ref).abs().max().item() < 1e-4
