@@ -47,6 +47,19 @@ class ModelSpec:
     topk_group: int = 1
     # rotary applied to the first head_dim*partial_rotary_factor dims only
     partial_rotary_factor: float = 1.0
+    # GPT-OSS extensions (GptOssForCausalLM):
+    # - attention SINKS: per-head learned logit joining the softmax
+    #   denominator only
+    # - alternating sliding-window / full attention layers (layer_types)
+    # - clamped-swiglu expert activation + expert/router biases
+    # - bias on o_proj (attention_bias covers q/k/v)
+    attention_sinks: bool = False
+    sliding_window: int = 0
+    layer_types: tuple | None = None     # ("sliding_attention"|"full_attention", ...)
+    moe_act: str = "silu"                # "silu" | "clamped_swiglu"
+    moe_bias: bool = False               # expert gate_up/down biases
+    router_logit_bias: bool = False      # bias added to router logits
+    o_proj_bias: bool = False
 
     @property
     def gqa_ratio(self) -> int:
@@ -92,8 +105,13 @@ class ModelSpec:
             num_heads=nh,
             num_kv_heads=cfg.get("num_key_value_heads", nh),
             head_dim=hd,
-            rope_theta=cfg.get("rope_theta", 10000.0),
-            rope_scaling=cfg.get("rope_scaling"),
+            rope_theta=cfg.get("rope_theta")
+            or (cfg.get("rope_parameters") or {}).get("rope_theta", 10000.0),
+            rope_scaling=cfg.get("rope_scaling")
+            or ((cfg.get("rope_parameters") or None)
+                if (cfg.get("rope_parameters") or {}).get("rope_type",
+                                                          "default")
+                not in ("default",) else None),
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
@@ -120,6 +138,16 @@ class ModelSpec:
             partial_rotary_factor=(cfg.get("partial_rotary_factor")
                                    or (cfg.get("rope_parameters") or {})
                                    .get("partial_rotary_factor") or 1.0),
+            attention_sinks=arch.startswith("GptOss"),
+            sliding_window=(cfg.get("sliding_window") or 0)
+            if arch.startswith("GptOss") else 0,
+            layer_types=tuple(cfg["layer_types"])
+            if cfg.get("layer_types") else None,
+            moe_act="clamped_swiglu" if arch.startswith("GptOss") else "silu",
+            moe_bias=arch.startswith("GptOss"),
+            router_logit_bias=arch.startswith("GptOss"),
+            o_proj_bias=bool(cfg.get("attention_bias"))
+            if arch.startswith("GptOss") else False,
         )
 
     @classmethod
@@ -195,6 +223,36 @@ PRESETS: dict[str, ModelSpec] = {
         num_experts=128, num_experts_per_tok=8, moe_intermediate_size=1408,
         router_mode="sigmoid_bias", n_shared_experts=1,
         first_k_dense_replace=1, partial_rotary_factor=0.5,
+    ),
+    # GPT-OSS 20B / 120B: 128/32-expert top-4 MoE,
+    # alternating sliding(128)/full attention with sinks, clamped swiglu,
+    # YaRN rope, head_dim 64. CPU-exact; CDNA4 kernels are an r3 item
+    # (the GPU ops fail loudly on D=64/sinks/window).
+    "gpt-oss-20b": ModelSpec(
+        architecture="GptOssForCausalLM", vocab_size=201088,
+        hidden_size=2880, intermediate_size=2880, num_layers=24,
+        num_heads=64, num_kv_heads=8, head_dim=64, rope_theta=150000.0,
+        rope_scaling={"rope_type": "yarn", "factor": 32.0,
+                      "beta_fast": 32.0, "beta_slow": 1.0, "truncate": False,
+                      "original_max_position_embeddings": 4096},
+        max_position_embeddings=131072, eos_token_id=200002,
+        attention_bias=True, o_proj_bias=True, attention_sinks=True,
+        sliding_window=128, num_experts=32, num_experts_per_tok=4,
+        moe_intermediate_size=2880, moe_act="clamped_swiglu", moe_bias=True,
+        router_logit_bias=True, norm_topk_prob=True,
+    ),
+    "gpt-oss-120b": ModelSpec(
+        architecture="GptOssForCausalLM", vocab_size=201088,
+        hidden_size=2880, intermediate_size=2880, num_layers=36,
+        num_heads=64, num_kv_heads=8, head_dim=64, rope_theta=150000.0,
+        rope_scaling={"rope_type": "yarn", "factor": 32.0,
+                      "beta_fast": 32.0, "beta_slow": 1.0, "truncate": False,
+                      "original_max_position_embeddings": 4096},
+        max_position_embeddings=131072, eos_token_id=200002,
+        attention_bias=True, o_proj_bias=True, attention_sinks=True,
+        sliding_window=128, num_experts=128, num_experts_per_tok=4,
+        moe_intermediate_size=2880, moe_act="clamped_swiglu", moe_bias=True,
+        router_logit_bias=True, norm_topk_prob=True,
     ),
     # Mixtral 8x7B: 8-expert top-2 MoE on the llama graph
     "mixtral-8x7b": ModelSpec(

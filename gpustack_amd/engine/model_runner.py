@@ -347,6 +347,8 @@ class ModelRunner:
                             or (cfg.spec.moe_intermediate_size
                                 // max(1, cfg.tp_size) % 128 == 0
                                 and cfg.spec.hidden_size % 128 == 0
+                                and cfg.spec.moe_act == "silu"
+                                and not cfg.spec.moe_bias
                                 and os.environ.get("GPUSTACK_AMD_FUSED_MOE",
                                                    "1") == "1"
                                 and ops.hip_available()))

@@ -96,7 +96,8 @@ def qlinear(x, w, pack: "W4Pack | None", bias=None):
 
 
 class Attention(nn.Module):
-    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator,
+                 dtype, layer_idx: int = 0):
         super().__init__()
         self.spec = spec
         self.comm = comm
@@ -114,6 +115,22 @@ class Attention(nn.Module):
             if spec.attention_bias else None
         )
         self.o_w = nn.Parameter(torch.empty(h, self.hq * self.d, dtype=dtype), requires_grad=False)
+        self.o_b = (nn.Parameter(torch.empty(h, dtype=dtype),
+                                 requires_grad=False)
+                    if spec.o_proj_bias else None)
+        # GPT-OSS: per-head sink logits (TP-sharded with the q heads) and
+        # per-layer sliding window (layer_types or even-layer default)
+        self.sinks = (nn.Parameter(torch.empty(self.hq, dtype=torch.float32),
+                                   requires_grad=False)
+                      if spec.attention_sinks else None)
+        if spec.sliding_window > 0:
+            if spec.layer_types is not None and layer_idx < len(spec.layer_types):
+                sliding = spec.layer_types[layer_idx] == "sliding_attention"
+            else:
+                sliding = layer_idx % 2 == 0  # GPT-OSS default alternation
+            self.window = spec.sliding_window if sliding else 0
+        else:
+            self.window = 0
         self.qkv_pack: W4Pack | None = None   # W4 runtime (qlinear)
         self.o_pack: W4Pack | None = None
         self.layer_idx = 0  # set by LlamaForCausalLM
@@ -147,27 +164,31 @@ class Attention(nn.Module):
                              self.rot_dim)
         ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
         out = torch.empty(T, self.hq, self.d, dtype=q.dtype, device=q.device)
+        sw = {}
+        if self.sinks is not None or self.window:
+            sw = {"sinks": self.sinks, "window": self.window}
         if meta.is_prefill:
             tp = meta.num_prefill_tokens or T
             ops.varlen_prefill_attn(
                 out[:tp], q[:tp], k[:tp], v[:tp], meta.seq_lens_list, self.scale,
-                tiles=(meta.tile_start, meta.tile_q0, meta.tile_len),
+                tiles=(meta.tile_start, meta.tile_q0, meta.tile_len), **sw,
             )
             if tp < T:  # mixed: decode rows ride the same forward
                 ops.paged_attn_decode(
                     out[tp:], q[tp:], k_cache, v_cache,
-                    meta.block_tables, meta.seq_lens, self.scale,
+                    meta.block_tables, meta.seq_lens, self.scale, **sw,
                 )
         elif meta.suffix_meta is not None:
             tiles, starts, hists, news = meta.suffix_meta
             ops.paged_prefill_attn(out, q, k_cache, v_cache,
                                    meta.block_tables, starts, hists, news,
-                                   self.scale, tiles=tiles)
+                                   self.scale, tiles=tiles, **sw)
         else:
             ops.paged_attn_decode(
-                out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, self.scale
+                out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens,
+                self.scale, **sw
             )
-        o = qlinear(out.view(T, -1), self.o_w, self.o_pack)
+        o = qlinear(out.view(T, -1), self.o_w, self.o_pack, self.o_b)
         if meta.lora is not None:
             meta.lora.apply(self.layer_idx, out.view(T, -1), o, self._o_projs)
         return self.comm.all_reduce(o)
@@ -230,11 +251,23 @@ class MoEMLP(nn.Module):
         # GLM/DeepSeek-style extensions (spec.router_mode "sigmoid_bias"):
         # learned correction bias on the routing scores and a SHARED dense
         # expert applied to every token (TP shards its intermediate)
-        if spec.router_mode == "sigmoid_bias":
+        if spec.router_mode == "sigmoid_bias" or spec.router_logit_bias:
             self.router_bias = nn.Parameter(
                 torch.zeros(self.e, dtype=torch.float32), requires_grad=False)
         else:
             self.router_bias = None
+        # GPT-OSS expert biases (TP: gate_up bias shards with the
+        # intermediate; the down bias is added on tp rank 0 only so the
+        # all-reduce sums it exactly once)
+        if spec.moe_bias:
+            self.gate_up_b = nn.Parameter(
+                torch.zeros(self.e, 2 * self.i, dtype=dtype),
+                requires_grad=False)
+            self.down_b = nn.Parameter(
+                torch.zeros(self.e, h, dtype=dtype), requires_grad=False)
+        else:
+            self.gate_up_b = None
+            self.down_b = None
         if spec.n_shared_experts > 0:
             si = spec.moe_intermediate_size * spec.n_shared_experts // tp_size
             self.shared_i = si
@@ -276,6 +309,8 @@ class MoEMLP(nn.Module):
             self._router_w_f32 = self.router_w.float()
             rw = self._router_w_f32
         logits = F.linear(x.float(), rw)                          # [T, E]
+        if self.spec.router_logit_bias and self.router_bias is not None:
+            logits = logits + self.router_bias  # GPT-OSS: bias on logits
         if self.spec.router_mode == "sigmoid_bias":
             # GLM-4.5/DeepSeek routing (HF Glm4MoeTopkRouter): sigmoid
             # scores; the learned bias only influences the CHOICE, the
@@ -379,6 +414,7 @@ class MoEMLP(nn.Module):
 
         return (x.is_cuda and x.dtype == torch.bfloat16
                 and os.environ.get("GPUSTACK_AMD_FUSED_MOE", "1") == "1"
+                and self.spec.moe_act == "silu" and self.gate_up_b is None
                 and self.i % 128 == 0 and x.shape[1] % 128 == 0
                 and ops.hip_available())
 
@@ -404,16 +440,34 @@ class MoEMLP(nn.Module):
                            order.to(torch.int32), flat_w32)
         return contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
 
+    def _act_mul(self, gu):
+        """[T', 2i] -> [T', i]: SiLU-mul (fused kernel/ref) or GPT-OSS
+        clamped swiglu ((up+1) * gate*sigmoid(1.702*gate), clamps +-7).
+        gu is in OUR fused layout [gate; up] (GPT-OSS checkpoints
+        de-interleave at load)."""
+        if self.spec.moe_act == "clamped_swiglu":
+            g = gu[..., :self.i].float().clamp(max=7.0)
+            u = gu[..., self.i:].float().clamp(-7.0, 7.0)
+            return ((u + 1.0) * (g * torch.sigmoid(g * 1.702))).to(gu.dtype)
+        act = torch.empty(gu.shape[0], self.i, dtype=gu.dtype,
+                          device=gu.device)
+        ops.silu_and_mul(act, gu)
+        return act
+
     def _loop_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w):
         hit = torch.bincount(flat_exp, minlength=self.e)
+        tp0 = self.comm.tp_rank == 0
         for e in torch.nonzero(hit, as_tuple=False).flatten().tolist():
             rows = torch.nonzero(flat_exp == e, as_tuple=False).flatten()
             idx = flat_tok[rows]
             xe = x.index_select(0, idx)
-            gu = F.linear(xe, self.gate_up_w[e])
-            act = torch.empty(xe.shape[0], self.i, dtype=x.dtype, device=x.device)
-            ops.silu_and_mul(act, gu)
-            he = F.linear(act, self.down_w[e])
+            gu = F.linear(xe, self.gate_up_w[e],
+                          self.gate_up_b[e] if self.gate_up_b is not None
+                          else None)
+            act = self._act_mul(gu)
+            he = F.linear(act, self.down_w[e],
+                          self.down_b[e] if (self.down_b is not None and tp0)
+                          else None)
             contrib[rows] = he * flat_w[rows].unsqueeze(1)
 
     def _bmm_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w,
@@ -444,12 +498,15 @@ class MoEMLP(nn.Module):
         xpad[s_exp, pos] = x[s_tok]
         _ck("xpad")
         gu = torch.bmm(xpad, self.gate_up_w.transpose(1, 2))   # [E, cap, 2i]
+        if self.gate_up_b is not None:
+            gu = gu + self.gate_up_b.unsqueeze(1)
         _ck("bmm1")
-        act = torch.empty(self.e * cap, self.i, dtype=x.dtype, device=x.device)
-        ops.silu_and_mul(act, gu.reshape(self.e * cap, 2 * self.i))
+        act = self._act_mul(gu.reshape(self.e * cap, 2 * self.i))
         _ck("silu")
         hd = torch.bmm(act.view(self.e, cap, self.i),
                        self.down_w.transpose(1, 2))            # [E, cap, h]
+        if self.down_b is not None and self.comm.tp_rank == 0:
+            hd = hd + self.down_b.unsqueeze(1)
         _ck("bmm2")
         contrib[order] = hd[s_exp, pos] * flat_w[order].unsqueeze(1)
         _ck("combine")
@@ -460,7 +517,7 @@ class DecoderLayer(nn.Module):
                  dtype, layer_idx: int = 0):
         super().__init__()
         self.spec = spec
-        self.attn = Attention(spec, tp_size, comm, dtype)
+        self.attn = Attention(spec, tp_size, comm, dtype, layer_idx=layer_idx)
         if spec.num_experts > 0 and layer_idx >= spec.first_k_dense_replace:
             self.mlp = MoEMLP(spec, tp_size, comm, dtype)
         else:

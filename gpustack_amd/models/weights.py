@@ -87,6 +87,13 @@ def random_init(model, cfg: EngineConfig) -> None:
             ]))
         o_full = _gen((spec.hidden_size, spec.num_heads * d), f"{li}.o", seed, dtype, device)
         layer.attn.o_w.copy_(o_full[:, rank * hq * d:(rank + 1) * hq * d])
+        if layer.attn.o_b is not None:
+            layer.attn.o_b.copy_(_gen((spec.hidden_size,), f"{li}.ob", seed,
+                                      dtype, device))
+        if layer.attn.sinks is not None:
+            sk = _gen((spec.num_heads,), f"{li}.sinks", seed,
+                      torch.float32, device, std=0.5)
+            layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
         if spec.qk_norm:
             layer.attn.q_norm.fill_(1.0)
             layer.attn.k_norm.fill_(1.0)
@@ -113,6 +120,15 @@ def _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank):
         layer.mlp.router_bias.copy_(_gen((spec.num_experts,), f"{li}.rbias",
                                          seed, torch.float32, device,
                                          std=0.05))
+    if getattr(layer.mlp, "gate_up_b", None) is not None:
+        mi = spec.moe_intermediate_size
+        gb = _gen((spec.num_experts, mi), f"{li}.egb", seed, dtype, device)
+        ub = _gen((spec.num_experts, mi), f"{li}.eub", seed, dtype, device)
+        layer.mlp.gate_up_b.copy_(torch.cat([
+            gb[:, rank * mi_loc:(rank + 1) * mi_loc],
+            ub[:, rank * mi_loc:(rank + 1) * mi_loc]], dim=1))
+        layer.mlp.down_b.copy_(_gen((spec.num_experts, spec.hidden_size),
+                                    f"{li}.edb", seed, dtype, device))
     if layer.mlp.shared_gate_up_w is not None:
         si_full = spec.moe_intermediate_size * spec.n_shared_experts
         si = si_full // tp
@@ -194,9 +210,40 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             ]))
         o = get(p + "self_attn.o_proj.weight")
         layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
+        if layer.attn.o_b is not None                 and p + "self_attn.o_proj.bias" in tensors:
+            layer.attn.o_b.copy_(get(p + "self_attn.o_proj.bias"))
+        if layer.attn.sinks is not None and p + "self_attn.sinks" in tensors:
+            sk = tensors[p + "self_attn.sinks"].float()
+            layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
         if spec.qk_norm:
             layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
             layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
+        if hasattr(layer.mlp, "router_w")                 and spec.architecture.startswith("GptOss"):
+            # GPT-OSS: router `mlp.router.{weight,bias}`; experts stored
+            # TRANSPOSED ([E, h, 2i] / [E, i, h]) with INTERLEAVED
+            # gate/up columns ([..., ::2] gate) + per-expert biases —
+            # de-interleave into our fused [gate; up] layout at load
+            layer.mlp.router_w.copy_(get(p + "mlp.router.weight"))
+            layer.mlp.router_bias.copy_(tensors[p + "mlp.router.bias"].float())
+            mi_loc = spec.moe_intermediate_size // tp
+            gu = get(p + "mlp.experts.gate_up_proj")     # [E, h, 2i]
+            gate = gu[:, :, 0::2].transpose(1, 2)        # [E, i, h]
+            up = gu[:, :, 1::2].transpose(1, 2)
+            layer.mlp.gate_up_w.copy_(torch.cat([
+                gate[:, rank * mi_loc:(rank + 1) * mi_loc],
+                up[:, rank * mi_loc:(rank + 1) * mi_loc]], dim=1))
+            gub = get(p + "mlp.experts.gate_up_proj_bias")  # [E, 2i]
+            layer.mlp.gate_up_b.copy_(torch.cat([
+                gub[:, 0::2][:, rank * mi_loc:(rank + 1) * mi_loc],
+                gub[:, 1::2][:, rank * mi_loc:(rank + 1) * mi_loc]], dim=1))
+            dn = get(p + "mlp.experts.down_proj")        # [E, i, h]
+            layer.mlp.down_w.copy_(
+                dn.transpose(1, 2)[:, :, rank * mi_loc:(rank + 1) * mi_loc])
+            layer.mlp.down_b.copy_(get(p + "mlp.experts.down_proj_bias"))
+            layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+            layer.post_attn_norm.copy_(
+                get(p + "post_attention_layernorm.weight"))
+            continue
         if hasattr(layer.mlp, "router_w"):
             # MoE (Qwen3-MoE `mlp.gate`, Mixtral `block_sparse_moe.gate`)
             router = (p + "mlp.gate.weight" if p + "mlp.gate.weight" in tensors

@@ -115,12 +115,21 @@ def greedy_sample(logits) -> torch.Tensor:
     return out
 
 
-def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale: float) -> None:
+def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
+                      scale: float, sinks=None, window: int = 0) -> None:
     hip = _backend(q)
     if hip is not None:
+        if sinks is not None or window:
+            # GPT-OSS sinks/sliding-window: CDNA4 kernel variants are an
+            # r3 item — fail loudly rather than silently mis-attend
+            raise NotImplementedError(
+                "attention sinks / sliding window not yet in the CDNA4 "
+                "decode kernel (gpt-oss GPU serving lands in r3)")
         hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
     else:
-        torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+        torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
+                                    seq_lens, scale, sinks=sinks,
+                                    window=window)
 
 
 _PREFILL_BQ = 64
@@ -140,14 +149,20 @@ def build_prefill_tiles(seq_lens: list[int], device) -> tuple[torch.Tensor, torc
     return mk(starts), mk(q0s), mk(lens)
 
 
-def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float, tiles=None) -> None:
+def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
+                        tiles=None, sinks=None, window: int = 0) -> None:
     hip = _backend(q)
     if hip is not None:
+        if sinks is not None or window:
+            raise NotImplementedError(
+                "attention sinks / sliding window not yet in the CDNA4 "
+                "prefill kernel (gpt-oss GPU serving lands in r3)")
         if tiles is None or tiles[0] is None:
             tiles = build_prefill_tiles(seq_lens, q.device)
         hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale)
     else:
-        torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale)
+        torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale,
+                                      sinks=sinks, window=window)
 
 
 def build_paged_prefill_tiles(seq_starts: list[int], seq_hists: list[int],
@@ -168,12 +183,17 @@ def build_paged_prefill_tiles(seq_starts: list[int], seq_hists: list[int],
 
 def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                        seq_starts: list[int], seq_hists: list[int],
-                       seq_news: list[int], scale: float, tiles=None) -> None:
+                       seq_news: list[int], scale: float, tiles=None,
+                       sinks=None, window: int = 0) -> None:
     """Prefill-with-history: suffix/chunk rows attend to cached paged KV +
     their own freshly written positions through the MFMA flash kernel
     (removes the r1-measured ~3.5x paged-decode-row penalty)."""
     hip = _backend(q)
     if hip is not None:
+        if sinks is not None or window:
+            raise NotImplementedError(
+                "attention sinks / sliding window not yet in the CDNA4 "
+                "paged-prefill kernel (gpt-oss GPU serving lands in r3)")
         if tiles is None:
             tiles = build_paged_prefill_tiles(seq_starts, seq_hists,
                                               seq_news, q.device)
@@ -182,7 +202,8 @@ def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                                 tiles[4], scale)
     else:
         torch_ref.paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
-                                     seq_starts, seq_hists, seq_news, scale)
+                                     seq_starts, seq_hists, seq_news, scale,
+                                     sinks=sinks, window=window)
 
 
 _SG_WORKSPACES: dict = {}

@@ -154,6 +154,27 @@ def greedy_sample(out: torch.Tensor, logits: torch.Tensor) -> None:
     out.copy_(logits.float().argmax(dim=-1))
 
 
+def _sink_softmax(att: torch.Tensor, sinks: torch.Tensor | None):
+    """softmax over the kv axis with optional per-head SINK logits joining
+    the denominator only (GPT-OSS: probability mass the sinks absorb is
+    dropped — no value contribution). att: [Hq, Tq, L]."""
+    if sinks is None:
+        return torch.softmax(att, dim=-1)
+    s = sinks.float().view(-1, 1, 1).expand(att.shape[0], att.shape[1], 1)
+    combined = torch.cat([att, s], dim=-1)
+    probs = torch.softmax(combined, dim=-1)
+    return probs[..., :-1]
+
+
+def _window_mask(qpos: torch.Tensor, kpos: torch.Tensor, window: int):
+    """Causal (+ optional sliding-window) additive mask [Tq, L]."""
+    m = torch.where(kpos.unsqueeze(0) <= qpos.unsqueeze(1), 0.0, float("-inf"))
+    if window:
+        m = torch.where(kpos.unsqueeze(0) > qpos.unsqueeze(1) - window,
+                        m, torch.tensor(float("-inf")))
+    return m
+
+
 def paged_attn_decode(
     out: torch.Tensor,
     q: torch.Tensor,
@@ -162,6 +183,8 @@ def paged_attn_decode(
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: float,
+    sinks: torch.Tensor | None = None,
+    window: int = 0,
 ) -> None:
     """q/out: [N, Hq, D]; caches [B, Hkv, BS, D]."""
     N, Hq, D = q.shape
@@ -177,7 +200,10 @@ def paged_attn_decode(
         keys = keys.repeat_interleave(GQ, dim=0)  # [Hq, L, D]
         vals = vals.repeat_interleave(GQ, dim=0)
         qi = q[i].float().unsqueeze(1)  # [Hq, 1, D]
-        att = torch.softmax((qi @ keys.transpose(1, 2)) * scale, dim=-1)
+        att = (qi @ keys.transpose(1, 2)) * scale
+        att = att + _window_mask(torch.tensor([L - 1], device=q.device),
+                                 torch.arange(L, device=q.device), window)
+        att = _sink_softmax(att, sinks)
         out[i] = (att @ vals).squeeze(1).to(out.dtype)
 
 
@@ -188,6 +214,8 @@ def varlen_prefill_attn(
     v: torch.Tensor,
     seq_lens: list[int],
     scale: float,
+    sinks: torch.Tensor | None = None,
+    window: int = 0,
 ) -> None:
     """q: [T, Hq, D]; k/v: [T, Hkv, D]; causal within each sequence."""
     Hq = q.shape[1]
@@ -199,8 +227,9 @@ def varlen_prefill_attn(
         ks = k[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
         vs = v[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
         att = (qs @ ks.transpose(1, 2)) * scale
-        mask = torch.full((L, L), float("-inf"), device=q.device).triu(1)
-        att = torch.softmax(att + mask, dim=-1)
+        pos = torch.arange(L, device=q.device)
+        att = att + _window_mask(pos, pos, window)
+        att = _sink_softmax(att, sinks)
         out[start : start + L] = (att @ vs).permute(1, 0, 2).to(out.dtype)
         start += L
 
@@ -215,6 +244,8 @@ def paged_prefill_attn(
     seq_hists: list[int],
     seq_news: list[int],
     scale: float,
+    sinks: torch.Tensor | None = None,
+    window: int = 0,
 ) -> None:
     """Prefill-with-history: q/out hold only the NEW (suffix) rows of each
     sequence; K/V for positions [0, hist+new) are gathered from the paged
@@ -241,8 +272,7 @@ def paged_prefill_attn(
         att = (qs @ keys.transpose(1, 2)) * scale           # [Hq, new, L]
         pos = torch.arange(L, device=q.device)
         qpos = hist + torch.arange(new, device=q.device)
-        mask = torch.where(pos.unsqueeze(0) <= qpos.unsqueeze(1), 0.0,
-                           float("-inf"))
-        att = torch.softmax(att + mask, dim=-1)
+        att = att + _window_mask(qpos, pos, window)
+        att = _sink_softmax(att, sinks)
         out[start:start + new] = (att @ vals).permute(1, 0, 2).to(out.dtype)
         row += new
