@@ -337,3 +337,63 @@ def test_deploy_with_lora_adapters(cluster, tmp_path_factory):
     mid = next(m["id"] for m in client.get("/v2/models").json()["items"]
                if m["name"] == "tiny-lora-host")
     client.delete(f"/v2/models/{mid}")
+
+
+def test_deploy_w4_runtime_model(cluster):
+    """Full control-plane path for a W4-runtime model: backend_parameters
+    flow scheduler -> serve manager -> engine_server -> EngineConfig, the
+    engine packs its weights int4 and serves through the gateway."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-w4", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+        "backend_parameters": {"quantize_runtime": "w4"},
+    })
+    assert r.status_code == 201, r.text
+    state = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-w4"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+    r = client.post("/v1/completions", json={
+        "model": "tiny-w4", "prompt": "xyz", "max_tokens": 5,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["usage"]["completion_tokens"] == 5
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-w4'][0]['id']}")
+
+
+def test_deploy_moe_model(cluster):
+    """MoE family end-to-end through the cluster (router + expert bank
+    in the engine subprocess)."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-moe-e2e", "source": "preset", "model_ref": "tiny-moe",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201, r.text
+    state = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-moe-e2e"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-moe-e2e",
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 4, "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-moe-e2e'][0]['id']}")
