@@ -8,6 +8,8 @@ engine/control-plane logic testable without hardware.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import torch_ref
@@ -53,6 +55,21 @@ def _backend(t: torch.Tensor):
     if t.is_cuda:
         return _load_hip()  # raises loudly if the native build is missing
     return None
+
+
+def _oss_kernels_enabled() -> bool:
+    """GPT-OSS CDNA4 kernel variants (head_dim-64 + attention sinks +
+    sliding window in attn_decode/attn_prefill.hip) are written and
+    compile-checked but not yet GPU-validated — opt in explicitly until
+    the r3 numerics pass (tests/test_ops_gpu.py oss tests)."""
+    return os.environ.get("GPUSTACK_AMD_OSS_KERNELS", "0") == "1"
+
+
+def _sinks_f32(sinks):
+    if sinks is not None and (sinks.dtype != torch.float32
+                              or not sinks.is_contiguous()):
+        sinks = sinks.float().contiguous()
+    return sinks
 
 
 # --- op surface -----------------------------------------------------------
@@ -119,13 +136,15 @@ def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
                       scale: float, sinks=None, window: int = 0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if sinks is not None or window:
-            # GPT-OSS sinks/sliding-window: CDNA4 kernel variants are an
-            # r3 item — fail loudly rather than silently mis-attend
+        if (sinks is not None or window) and not _oss_kernels_enabled():
+            # GPT-OSS sinks/sliding-window kernels are written but not yet
+            # GPU-validated — fail loudly rather than silently mis-attend
+            # (opt in with GPUSTACK_AMD_OSS_KERNELS=1; r3 flips the default)
             raise NotImplementedError(
-                "attention sinks / sliding window not yet in the CDNA4 "
-                "decode kernel (gpt-oss GPU serving lands in r3)")
-        hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+                "attention sinks / sliding window CDNA4 decode kernel is "
+                "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+        hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
+                              scale, sinks=_sinks_f32(sinks), window=window)
     else:
         torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                     seq_lens, scale, sinks=sinks,
@@ -153,13 +172,14 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
                         tiles=None, sinks=None, window: int = 0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if sinks is not None or window:
+        if (sinks is not None or window) and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "attention sinks / sliding window not yet in the CDNA4 "
-                "prefill kernel (gpt-oss GPU serving lands in r3)")
+                "attention sinks / sliding window CDNA4 prefill kernel is "
+                "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         if tiles is None or tiles[0] is None:
             tiles = build_prefill_tiles(seq_lens, q.device)
-        hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale)
+        hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale,
+                          sinks=_sinks_f32(sinks), window=window)
     else:
         torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale,
                                       sinks=sinks, window=window)
@@ -190,16 +210,16 @@ def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
     (removes the r1-measured ~3.5x paged-decode-row penalty)."""
     hip = _backend(q)
     if hip is not None:
-        if sinks is not None or window:
+        if (sinks is not None or window) and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "attention sinks / sliding window not yet in the CDNA4 "
-                "paged-prefill kernel (gpt-oss GPU serving lands in r3)")
+                "attention sinks / sliding window CDNA4 paged-prefill kernel "
+                "is unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         if tiles is None:
             tiles = build_paged_prefill_tiles(seq_starts, seq_hists,
                                               seq_news, q.device)
         hip.flash_prefill_paged(out, q, k_cache, v_cache, block_tables,
                                 tiles[0], tiles[1], tiles[2], tiles[3],
-                                tiles[4], scale)
+                                tiles[4], scale, sinks=_sinks_f32(sinks), window=window)
     else:
         torch_ref.paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                                      seq_starts, seq_hists, seq_news, scale,

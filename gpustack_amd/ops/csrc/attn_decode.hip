@@ -32,10 +32,12 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     const void* __restrict__ vc,             // [B, Hkv, BS, D]
     const int* __restrict__ block_tables,    // [N, max_blocks]
     const int* __restrict__ seq_lens,        // [N]
-    int Hkv, int max_blocks, float scale, long q_stride) {
+    int Hkv, int max_blocks, float scale, long q_stride,
+    const float* __restrict__ sinks,         // [Hq] or null (GPT-OSS)
+    int window) {                            // 0 = full attention
   constexpr int LPG = 16;           // lanes per token-group
-  constexpr int DV = D / LPG;       // dims per lane (8 for D=128)
-  static_assert(DV == 8, "decode kernel assumes D = 128");
+  constexpr int DV = D / LPG;       // dims per lane (8 for D=128, 4 for 64)
+  static_assert(DV == 8 || DV == 4, "decode kernel assumes D in {64, 128}");
   // tokens batched per softmax update: trade VALU savings against VGPR
   // pressure (GQ>=4 would spill at TB=4)
   constexpr int TB = (GQ <= 2) ? (BS / 4) : ((GQ <= 5) ? 2 : 1);
@@ -57,17 +59,18 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   // dot-product instruction count (occupancy was the bottleneck: 2-3
   // waves/SIMD at 168-200 VGPRs, profiles/r02_optimization_log.md).
   // fp8 path: f32 q pre-scaled as before (k dequants through f32 anyway).
+  using U16V = std::conditional_t<DV == 8, u16x8, u16x4>;
+  using U8V = std::conditional_t<DV == 8, u8x8, u8x4>;
   float qr[FP8 ? GQ : 1][DV];
-  u16x8 qb[FP8 ? 1 : GQ];
+  U16V qb[FP8 ? 1 : GQ];
 #pragma unroll
   for (int gq = 0; gq < GQ; ++gq) {
     const unsigned short* qp =
         q + (long)seq * q_stride + ((long)h * GQ + gq) * D + sub * DV;
-    u16x8 u = *reinterpret_cast<const u16x8*>(qp);
+    U16V u = *reinterpret_cast<const U16V*>(qp);
     if constexpr (FP8) {
-      bf8_to_f32(u, qr[gq]);
 #pragma unroll
-      for (int j = 0; j < DV; ++j) qr[gq][j] *= scale;
+      for (int j = 0; j < DV; ++j) qr[gq][j] = bf2f(u[j]) * scale;
     } else {
       qb[gq] = u;
     }
@@ -83,8 +86,12 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   }
 
   using KVT = std::conditional_t<FP8, unsigned char, unsigned short>;
+  // sliding window: the single decode query sits at position len-1 and
+  // attends [max(0, len-window), len)
+  const int wstart = (window > 0 && len > window) ? len - window : 0;
+  const int page0 = wstart / BS;
   const int* bt = block_tables + (long)seq * max_blocks;
-  for (int page = wave; page < npages; page += NWAVES) {
+  for (int page = page0 + wave; page < npages; page += NWAVES) {
     const long blk = bt[page];
     const KVT* kbase = (const KVT*)kc + ((blk * Hkv + h) * BS) * D;
     const KVT* vbase = (const KVT*)vc + ((blk * Hkv + h) * BS) * D;
@@ -95,7 +102,7 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
     for (int tb = 0; tb < BS / 4; tb += TB) {
       const int base_tok = page * BS + tb * 4 + g;
-      using VecT = std::conditional_t<FP8, u8x8, u16x8>;
+      using VecT = std::conditional_t<FP8, U8V, U16V>;
       VecT vu[TB];
       VecT ku[TB];              // bf16: K stays packed (dot2 consumes it)
       float kf[FP8 ? TB : 1][DV];  // fp8: K dequanted once, reused per gq
@@ -104,7 +111,10 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         const int tok = (tb + it) * 4 + g;
         ku[it] = *reinterpret_cast<const VecT*>(kbase + tok * D + sub * DV);
         vu[it] = *reinterpret_cast<const VecT*>(vbase + tok * D + sub * DV);
-        if constexpr (FP8) fp8x8_to_f32(ku[it], kf[it]);
+        if constexpr (FP8) {
+#pragma unroll
+          for (int j = 0; j < DV; ++j) kf[it][j] = fp8_to_f32(ku[it][j]);
+        }
       }
 #pragma unroll
       for (int gq = 0; gq < GQ; ++gq) {
@@ -131,7 +141,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         float pmax = -INFINITY;
 #pragma unroll
         for (int it = 0; it < TB; ++it) {
-          const bool valid = base_tok + it * 4 < len;
+          const int tpos = base_tok + it * 4;
+          const bool valid = tpos < len && tpos >= wstart;
           dot[it] = valid ? dot[it] : -INFINITY;
           pmax = fmaxf(pmax, dot[it]);
         }
@@ -209,7 +220,15 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
     for (int w = 0; w < NWAVES; ++w)
       if (lds_s[w][gq] > 0.f) M = fmaxf(M, lds_m[w][gq]);
-    float num = 0.f, den = 0.f;
+    // GPT-OSS attention sink: a per-head learned logit joins the softmax
+    // DENOMINATOR only (no value contribution)
+    float sden = 0.f;
+    if (sinks != nullptr) {
+      const float sk = sinks[h * GQ + gq];
+      M = fmaxf(M, sk);
+      sden = __expf(sk - M);
+    }
+    float num = 0.f, den = sden;
 #pragma unroll
     for (int w = 0; w < NWAVES; ++w) {
       if (lds_s[w][gq] > 0.f) {
@@ -229,16 +248,28 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
                               const void* vc, const int* block_tables,
                               const int* seq_lens, int N, int Hq, int Hkv,
                               int D, int max_blocks, float scale, long q_stride,
-                              int fp8, int* err_unsupported, hipStream_t s) {
+                              int fp8, const float* sinks, int window,
+                              int* err_unsupported, hipStream_t s) {
   const int GQ = Hq / Hkv;
   dim3 grid(N, Hkv);
   dim3 block(THREADS);
   *err_unsupported = 0;
-  if (D != 128) { *err_unsupported = 1; return; }
+  if (D != 128 && D != 64) { *err_unsupported = 1; return; }
 #define LAUNCH_GQ2(G, F)                                                       \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<128, G, F>), grid, block, 0, s, \
-                     (unsigned short*)out, (const unsigned short*)q, kc, vc,   \
-                     block_tables, seq_lens, Hkv, max_blocks, scale, q_stride)
+  do {                                                                         \
+    if (D == 128)                                                              \
+      hipLaunchKernelGGL((paged_attn_decode_kernel<128, G, F>), grid, block,   \
+                         0, s, (unsigned short*)out,                           \
+                         (const unsigned short*)q, kc, vc, block_tables,       \
+                         seq_lens, Hkv, max_blocks, scale, q_stride, sinks,    \
+                         window);                                              \
+    else                                                                       \
+      hipLaunchKernelGGL((paged_attn_decode_kernel<64, G, F>), grid, block,    \
+                         0, s, (unsigned short*)out,                           \
+                         (const unsigned short*)q, kc, vc, block_tables,       \
+                         seq_lens, Hkv, max_blocks, scale, q_stride, sinks,    \
+                         window);                                              \
+  } while (0)
 #define LAUNCH_GQ(G)                                                          \
   do {                                                                        \
     if (fp8) LAUNCH_GQ2(G, true);                                             \

@@ -10,12 +10,16 @@
 //  - swapped QK^T (S^T = K · Q^T) so BOTH operands are row-contiguous
 //    u16x8 reads (K from LDS, Q from registers).
 //  - K tile LDS-staged with the ((row&7)<<4) byte-XOR swizzle (guide §6 G4).
-//  - V staged TRANSPOSED (VT[128][BK+pad]) so PV B-fragments are b128 reads.
+//  - V staged TRANSPOSED (VT[D][BK+pad]) so PV B-fragments are b128 reads.
 //  - v2: BK = 64 kv per tile (twice the MFMA work per barrier pair) and
 //    async-stage split (guide T14): each iteration ISSUES the next tile's
 //    global loads into registers BEFORE the MFMA work, then writes them to
 //    LDS after the barrier — HBM latency hides under compute.
 //  - online softmax in f32 registers; per-q-row stats shared via shuffles.
+//  - templated head_dim D in {64, 128} (GPT-OSS is D=64); optional
+//    per-head attention-SINK logits (join the softmax denominator only)
+//    and sliding WINDOW (kvpos in (qpos-window, qpos]), matching
+//    torch_ref._sink_softmax / _window_mask exactly.
 //
 // Tiling: BQ = 64 q rows per workgroup (4 waves x 16 rows), BK = 64 kv.
 #include "common.h"
@@ -24,14 +28,11 @@ namespace {
 
 constexpr int BQ = 64;
 constexpr int BK = 64;
-constexpr int PF_D = 128;
 constexpr int PF_WAVES = 4;
 constexpr int PF_THREADS = PF_WAVES * WAVE_SIZE;
 constexpr int VT_PAD = 4;  // elements; breaks the bank cycle on VT/P rows
 constexpr int NST = BK / 16;       // S^T stiles per tile (4)
 constexpr int KC2 = BK / 32;       // PV k-chunks per tile (2)
-constexpr int KU = BK * (PF_D / 8) / PF_THREADS;  // K u16x8 units/thread (4)
-constexpr int VU = PF_D * (BK / 8) / PF_THREADS;  // VT units/thread (4)
 
 typedef __attribute__((ext_vector_type(8))) short s16x8;
 
@@ -41,10 +42,12 @@ DEVICE_INLINE f32x4 mfma16x16x32_bf16(u16x8 a, u16x8 b, f32x4 c) {
 }
 
 // byte offset of (kv, dbyte) in the swizzled K tile
+template <int D>
 DEVICE_INLINE int kswz(int kv, int dbyte) {
-  return kv * (PF_D * 2) + (dbyte ^ ((kv & 7) << 4));
+  return kv * (D * 2) + (dbyte ^ ((kv & 7) << 4));
 }
 
+template <int D>
 __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     unsigned short* __restrict__ out,      // [T, Hq, D]
     const unsigned short* __restrict__ q,  // [T, Hq, D]
@@ -54,7 +57,12 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     const int* __restrict__ tile_q0,       // [ntiles] q-tile offset in seq
     const int* __restrict__ tile_len,      // [ntiles] seq length
     int Hq, int Hkv, float scale,
-    long qs, long ks, long vs) {           // row strides (elements)
+    long qs, long ks, long vs,             // row strides (elements)
+    const float* __restrict__ sinks,       // [Hq] or null
+    int window) {                          // 0 = full causal
+  constexpr int KU = BK * (D / 8) / PF_THREADS;  // K u16x8 units/thread
+  constexpr int VU = D * (BK / 8) / PF_THREADS;  // VT units/thread
+  constexpr int QK = D / 32;                     // q k-chunks
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
   const int kvh = qh / (Hq / Hkv);
@@ -68,40 +76,43 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   const int lg = lane >> 4;        // 4-lane group id
   const int tid = threadIdx.x;
 
-  __shared__ unsigned short Kl[BK * PF_D];            // swizzled
-  __shared__ unsigned short VTl[PF_D][BK + VT_PAD];   // transposed V
+  __shared__ unsigned short Kl[BK * D];            // swizzled
+  __shared__ unsigned short VTl[D][BK + VT_PAD];   // transposed V
   __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
 
-  // Hoist this wave's 16 q rows into B-fragments (4 k-chunks of 32).
+  // Hoist this wave's 16 q rows into B-fragments (QK k-chunks of 32).
   const int qrow_local = q0 + wave * 16 + lc;
   const int qrow_clamped = (qrow_local < len) ? qrow_local : (len - 1);
-  u16x8 qfrag[4];
+  u16x8 qfrag[QK];
 #pragma unroll
-  for (int kk = 0; kk < 4; ++kk) {
+  for (int kk = 0; kk < QK; ++kk) {
     const unsigned short* qp = q + (long)(seq0 + qrow_clamped) * qs +
-                               (long)qh * PF_D + kk * 32 + lg * 8;
+                               (long)qh * D + kk * 32 + lg * 8;
     qfrag[kk] = *reinterpret_cast<const u16x8*>(qp);
   }
 
   float mcol = -INFINITY;  // running max for q row `lc` (this wave)
   float lcol = 0.f;        // running denom for q row `lc`
-  f32x4 o[PF_D / 16];      // O[q=(lg*4+r)][d=lc+nt*16]
+  f32x4 o[D / 16];         // O[q=(lg*4+r)][d=lc+nt*16]
 #pragma unroll
-  for (int nt = 0; nt < PF_D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int nt = 0; nt < D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int q_hi = q0 + BQ - 1;
   const int kv_end = min(len, q_hi + 1);           // causal bound
   const int ntiles_kv = (kv_end + BK - 1) / BK;
   const int wave_q_hi = q0 + wave * 16 + 15;       // this wave's causal bound
+  const int wave_q_lo = q0 + wave * 16;            // lowest window start
 
   // staging assignment (fixed per thread):
-  //   K: unit u = tid + r*256 -> (kv = u/16, d0 = (u%16)*8), r in [0,KU)
-  //   VT: unit u -> (d = u%128, kvc = (u/128)*8)
-  const int kst_kv[KU] = {tid / 16, (tid + 256) / 16, (tid + 512) / 16,
-                          (tid + 768) / 16};
-  const int kst_d0 = (tid % 16) * 8;
-  const int vst_d = tid % PF_D;
-  const int vst_kvc0 = (tid / PF_D) * 8;   // + r*16 per round
+  //   K: unit u = tid + r*256 -> (kv = u/(D/8), d0 = (u%(D/8))*8)
+  //   VT: unit u -> (d = u%D, kvc = (u/D)*8)
+  int kst_kv[KU];
+#pragma unroll
+  for (int r = 0; r < KU; ++r) kst_kv[r] = (tid + r * PF_THREADS) / (D / 8);
+  const int kst_d0 = (tid % (D / 8)) * 8;
+  const int vst_d = tid % D;
+  const int vst_kvc0 = (tid / D) * 8;   // + r*(8*PF_THREADS/D) per round
+  constexpr int VST_STEP = 8 * PF_THREADS / D;
 
   u16x8 kstage[KU];
   unsigned short vstage[VU][8];
@@ -112,17 +123,17 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       const int kv = kv0 + kst_kv[r];
       kstage[r] = (kv < len)
           ? *reinterpret_cast<const u16x8*>(
-                k + (long)(seq0 + kv) * ks + (long)kvh * PF_D + kst_d0)
+                k + (long)(seq0 + kv) * ks + (long)kvh * D + kst_d0)
           : u16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
 #pragma unroll
     for (int r = 0; r < VU; ++r) {
-      const int kvc = vst_kvc0 + r * 16;
+      const int kvc = vst_kvc0 + r * VST_STEP;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int kv = kv0 + kvc + j;
         vstage[r][j] = (kv < len)
-            ? v[(long)(seq0 + kv) * vs + (long)kvh * PF_D + vst_d]
+            ? v[(long)(seq0 + kv) * vs + (long)kvh * D + vst_d]
             : (unsigned short)0;
       }
     }
@@ -132,11 +143,11 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 #pragma unroll
     for (int r = 0; r < KU; ++r) {
       *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
-                                kswz(kst_kv[r], kst_d0 * 2)) = kstage[r];
+                                kswz<D>(kst_kv[r], kst_d0 * 2)) = kstage[r];
     }
 #pragma unroll
     for (int r = 0; r < VU; ++r) {
-      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * 16]) =
+      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * VST_STEP]) =
           *reinterpret_cast<u16x8*>(vstage[r]);
     }
   };
@@ -150,18 +161,20 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);  // hide HBM under MFMA
 
     if (kv0 > wave_q_hi) continue;  // fully masked for this wave
+    // sliding window: tile entirely below every row's window -> skip
+    if (window > 0 && kv0 + BK <= wave_q_lo - window + 1) continue;
 
-    // ---- S^T = K · Q^T  (NST stiles x 4 k-chunks) ----
+    // ---- S^T = K · Q^T  (NST stiles x QK k-chunks) ----
     f32x4 st[NST];
 #pragma unroll
     for (int i = 0; i < NST; ++i) st[i] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int stile = 0; stile < NST; ++stile) {
 #pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
+      for (int kk = 0; kk < QK; ++kk) {
         const u16x8 a = *reinterpret_cast<const u16x8*>(
             reinterpret_cast<char*>(Kl) +
-            kswz(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
+            kswz<D>(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
         st[stile] = mfma16x16x32_bf16(a, qfrag[kk], st[stile]);
       }
     }
@@ -176,7 +189,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
-        const bool ok = (kvpos <= qpos) && (kvpos < len) && (qpos < len);
+        bool ok = (kvpos <= qpos) && (kvpos < len) && (qpos < len);
+        if (window > 0) ok = ok && (kvpos > qpos - window);
         x = ok ? x : -INFINITY;
         sv[stile * 4 + r] = x;
         tmax = fmaxf(tmax, x);
@@ -222,7 +236,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       const int orow = lg * 4 + r;
       const float c = __shfl(corr, orow, WAVE_SIZE);
 #pragma unroll
-      for (int nt = 0; nt < PF_D / 16; ++nt) o[nt][r] *= c;
+      for (int nt = 0; nt < D / 16; ++nt) o[nt][r] *= c;
     }
 
     // ---- O += P · V  (A = P from LDS, B = V^T rows from LDS) ----
@@ -231,7 +245,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       const u16x8 pa = *reinterpret_cast<const u16x8*>(
           &Pl[wave][lc][kk2 * 32 + lg * 8]);
 #pragma unroll
-      for (int nt = 0; nt < PF_D / 16; ++nt) {
+      for (int nt = 0; nt < D / 16; ++nt) {
         const u16x8 b = *reinterpret_cast<const u16x8*>(
             &VTl[nt * 16 + lc][kk2 * 32 + lg * 8]);
         o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
@@ -239,17 +253,26 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     }
   }
 
-  // ---- epilogue: normalize rows and write ----
+  // ---- epilogue: normalize rows and write; the SINK logit joins the
+  // denominator only (GPT-OSS: its probability mass drops, no value) ----
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int orow = lg * 4 + r;
-    const float denom = __shfl(lcol, orow, WAVE_SIZE);
+    float denom = __shfl(lcol, orow, WAVE_SIZE);
+    float onum = 1.f;
+    if (sinks != nullptr) {
+      const float m_row = __shfl(mcol, orow, WAVE_SIZE);
+      const float sk = sinks[qh];
+      const float M2 = fmaxf(m_row, sk);
+      onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
+      denom = denom * onum + __expf(sk - M2);
+    }
     const int qrow = q0 + wave * 16 + orow;
     if (qrow >= len || denom <= 0.f) continue;
-    const float inv = 1.f / denom;
+    const float inv = onum / denom;
 #pragma unroll
-    for (int nt = 0; nt < PF_D / 16; ++nt) {
-      out[((long)(seq0 + qrow) * Hq + qh) * PF_D + nt * 16 + lc] =
+    for (int nt = 0; nt < D / 16; ++nt) {
+      out[((long)(seq0 + qrow) * Hq + qh) * D + nt * 16 + lc] =
           f2bf(o[nt][r] * inv);
     }
   }
@@ -271,6 +294,7 @@ namespace {
 
 constexpr int PP_BS = 16;  // pool block size (tokens per KV block)
 
+template <int D>
 __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     unsigned short* __restrict__ out,      // [T, Hq, D] (suffix rows)
     const unsigned short* __restrict__ q,  // [T, Hq, D] (suffix rows)
@@ -282,7 +306,12 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     const int* __restrict__ tile_hist,     // [ntiles] cached tokens (history)
     const int* __restrict__ tile_new,      // [ntiles] suffix length
     const int* __restrict__ tile_seq,      // [ntiles] row into block_tables
-    int Hq, int Hkv, int maxb, float scale, long qs) {
+    int Hq, int Hkv, int maxb, float scale, long qs,
+    const float* __restrict__ sinks,       // [Hq] or null
+    int window) {                          // 0 = full causal
+  constexpr int KU = BK * (D / 8) / PF_THREADS;
+  constexpr int VU = D * (BK / 8) / PF_THREADS;
+  constexpr int QK = D / 32;
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
   const int kvh = qh / (Hq / Hkv);
@@ -299,36 +328,39 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
   const int lg = lane >> 4;
   const int tid = threadIdx.x;
 
-  __shared__ unsigned short Kl[BK * PF_D];
-  __shared__ unsigned short VTl[PF_D][BK + VT_PAD];
+  __shared__ unsigned short Kl[BK * D];
+  __shared__ unsigned short VTl[D][BK + VT_PAD];
   __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
 
   const int qrow_local = q0 + wave * 16 + lc;           // row within suffix
   const int qrow_clamped = (qrow_local < nnew) ? qrow_local : (nnew - 1);
-  u16x8 qfrag[4];
+  u16x8 qfrag[QK];
 #pragma unroll
-  for (int kk = 0; kk < 4; ++kk) {
+  for (int kk = 0; kk < QK; ++kk) {
     const unsigned short* qp = q + (long)(qstart + qrow_clamped) * qs +
-                               (long)qh * PF_D + kk * 32 + lg * 8;
+                               (long)qh * D + kk * 32 + lg * 8;
     qfrag[kk] = *reinterpret_cast<const u16x8*>(qp);
   }
 
   float mcol = -INFINITY;
   float lcol = 0.f;
-  f32x4 o[PF_D / 16];
+  f32x4 o[D / 16];
 #pragma unroll
-  for (int nt = 0; nt < PF_D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int nt = 0; nt < D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int q_hi_abs = hist + q0 + BQ - 1;              // absolute position
   const int kv_end = min(len, q_hi_abs + 1);
   const int ntiles_kv = (kv_end + BK - 1) / BK;
   const int wave_q_hi_abs = hist + q0 + wave * 16 + 15;
+  const int wave_q_lo_abs = hist + q0 + wave * 16;
 
-  const int kst_kv[KU] = {tid / 16, (tid + 256) / 16, (tid + 512) / 16,
-                          (tid + 768) / 16};
-  const int kst_d0 = (tid % 16) * 8;
-  const int vst_d = tid % PF_D;
-  const int vst_kvc0 = (tid / PF_D) * 8;
+  int kst_kv[KU];
+#pragma unroll
+  for (int r = 0; r < KU; ++r) kst_kv[r] = (tid + r * PF_THREADS) / (D / 8);
+  const int kst_d0 = (tid % (D / 8)) * 8;
+  const int vst_d = tid % D;
+  const int vst_kvc0 = (tid / D) * 8;
+  constexpr int VST_STEP = 8 * PF_THREADS / D;
 
   u16x8 kstage[KU];
   unsigned short vstage[VU][8];
@@ -336,7 +368,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
   // pool row base (elements) for token position `kv`
   auto pool_row = [&](int kv) {
     const int blk = bt[kv / PP_BS];
-    return (((long)blk * Hkv + kvh) * PP_BS + kv % PP_BS) * PF_D;
+    return (((long)blk * Hkv + kvh) * PP_BS + kv % PP_BS) * D;
   };
 
   auto issue_loads = [&](int kv0) {
@@ -352,7 +384,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     }
 #pragma unroll
     for (int r = 0; r < VU; ++r) {
-      const int kvc = vst_kvc0 + r * 16;
+      const int kvc = vst_kvc0 + r * VST_STEP;
       // 8 consecutive positions span at most 2 pool blocks; resolving the
       // row base per element keeps the gather exact at block boundaries
 #pragma unroll
@@ -369,11 +401,11 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
 #pragma unroll
     for (int r = 0; r < KU; ++r) {
       *reinterpret_cast<u16x8*>(reinterpret_cast<char*>(Kl) +
-                                kswz(kst_kv[r], kst_d0 * 2)) = kstage[r];
+                                kswz<D>(kst_kv[r], kst_d0 * 2)) = kstage[r];
     }
 #pragma unroll
     for (int r = 0; r < VU; ++r) {
-      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * 16]) =
+      *reinterpret_cast<u16x8*>(&VTl[vst_d][vst_kvc0 + r * VST_STEP]) =
           *reinterpret_cast<u16x8*>(vstage[r]);
     }
   };
@@ -387,6 +419,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);
 
     if (kv0 > wave_q_hi_abs) continue;
+    if (window > 0 && kv0 + BK <= wave_q_lo_abs - window + 1) continue;
 
     f32x4 st[NST];
 #pragma unroll
@@ -394,10 +427,10 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
 #pragma unroll
     for (int stile = 0; stile < NST; ++stile) {
 #pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
+      for (int kk = 0; kk < QK; ++kk) {
         const u16x8 a = *reinterpret_cast<const u16x8*>(
             reinterpret_cast<char*>(Kl) +
-            kswz(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
+            kswz<D>(stile * 16 + lc, (kk * 32 + lg * 8) * 2));
         st[stile] = mfma16x16x32_bf16(a, qfrag[kk], st[stile]);
       }
     }
@@ -412,7 +445,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
-        const bool ok = (kvpos <= qpos) && (kvpos < len) && qvalid;
+        bool ok = (kvpos <= qpos) && (kvpos < len) && qvalid;
+        if (window > 0) ok = ok && (kvpos > qpos - window);
         x = ok ? x : -INFINITY;
         sv[stile * 4 + r] = x;
         tmax = fmaxf(tmax, x);
@@ -456,7 +490,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
       const int orow = lg * 4 + r;
       const float c = __shfl(corr, orow, WAVE_SIZE);
 #pragma unroll
-      for (int nt = 0; nt < PF_D / 16; ++nt) o[nt][r] *= c;
+      for (int nt = 0; nt < D / 16; ++nt) o[nt][r] *= c;
     }
 
 #pragma unroll
@@ -464,7 +498,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
       const u16x8 pa = *reinterpret_cast<const u16x8*>(
           &Pl[wave][lc][kk2 * 32 + lg * 8]);
 #pragma unroll
-      for (int nt = 0; nt < PF_D / 16; ++nt) {
+      for (int nt = 0; nt < D / 16; ++nt) {
         const u16x8 b = *reinterpret_cast<const u16x8*>(
             &VTl[nt * 16 + lc][kk2 * 32 + lg * 8]);
         o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
@@ -475,13 +509,21 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int orow = lg * 4 + r;
-    const float denom = __shfl(lcol, orow, WAVE_SIZE);
+    float denom = __shfl(lcol, orow, WAVE_SIZE);
+    float onum = 1.f;
+    if (sinks != nullptr) {
+      const float m_row = __shfl(mcol, orow, WAVE_SIZE);
+      const float sk = sinks[qh];
+      const float M2 = fmaxf(m_row, sk);
+      onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
+      denom = denom * onum + __expf(sk - M2);
+    }
     const int qrow = q0 + wave * 16 + orow;
     if (qrow >= nnew || denom <= 0.f) continue;
-    const float inv = 1.f / denom;
+    const float inv = onum / denom;
 #pragma unroll
-    for (int nt = 0; nt < PF_D / 16; ++nt) {
-      out[((long)(qstart + qrow) * Hq + qh) * PF_D + nt * 16 + lc] =
+    for (int nt = 0; nt < D / 16; ++nt) {
+      out[((long)(qstart + qrow) * Hq + qh) * D + nt * 16 + lc] =
           f2bf(o[nt][r] * inv);
     }
   }
@@ -494,30 +536,37 @@ void flash_prefill_paged_launch(
     const int* block_tables, const int* tile_qstart, const int* tile_q0,
     const int* tile_hist, const int* tile_new, const int* tile_seq,
     int ntiles, int Hq, int Hkv, int D, int maxb, float scale, long qs,
-    int* err_unsupported, hipStream_t s) {
+    const float* sinks, int window, int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
-  if (D != 128 || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  if ((D != 128 && D != 64) || Hq % Hkv != 0) { *err_unsupported = 1; return; }
   dim3 grid(ntiles, Hq);
-  hipLaunchKernelGGL(flash_prefill_paged_kernel, grid, dim3(PF_THREADS), 0, s,
-                     (unsigned short*)out, (const unsigned short*)q,
-                     (const unsigned short*)kc, (const unsigned short*)vc,
-                     block_tables, tile_qstart, tile_q0, tile_hist, tile_new,
-                     tile_seq, Hq, Hkv, maxb, scale, qs);
+#define PPG_LAUNCH(DD)                                                        \
+  hipLaunchKernelGGL(flash_prefill_paged_kernel<DD>, grid, dim3(PF_THREADS),  \
+                     0, s, (unsigned short*)out, (const unsigned short*)q,    \
+                     (const unsigned short*)kc, (const unsigned short*)vc,    \
+                     block_tables, tile_qstart, tile_q0, tile_hist, tile_new, \
+                     tile_seq, Hq, Hkv, maxb, scale, qs, sinks, window)
+  if (D == 128) PPG_LAUNCH(128); else PPG_LAUNCH(64);
+#undef PPG_LAUNCH
 }
 
 void flash_prefill_launch(void* out, const void* q, const void* k,
                           const void* v, const int* tile_start,
                           const int* tile_q0, const int* tile_len, int ntiles,
                           int Hq, int Hkv, int D, float scale,
-                          long qs, long ks, long vs,
-                          int* err_unsupported, hipStream_t s) {
+                          long qs, long ks, long vs, const float* sinks,
+                          int window, int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
-  if (D != 128 || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  if ((D != 128 && D != 64) || Hq % Hkv != 0) { *err_unsupported = 1; return; }
   dim3 grid(ntiles, Hq);
-  hipLaunchKernelGGL(flash_prefill_kernel, grid, dim3(PF_THREADS), 0, s,
-                     (unsigned short*)out, (const unsigned short*)q,
-                     (const unsigned short*)k, (const unsigned short*)v,
-                     tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks, vs);
+#define PF_LAUNCH(DD)                                                        \
+  hipLaunchKernelGGL(flash_prefill_kernel<DD>, grid, dim3(PF_THREADS), 0, s, \
+                     (unsigned short*)out, (const unsigned short*)q,         \
+                     (const unsigned short*)k, (const unsigned short*)v,     \
+                     tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks,  \
+                     vs, sinks, window)
+  if (D == 128) PF_LAUNCH(128); else PF_LAUNCH(64);
+#undef PF_LAUNCH
 }
 
 // ---------------------------------------------------------------------------

@@ -11,9 +11,9 @@ void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, in
 void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
 void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, int, hipStream_t);
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
-void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, int*, hipStream_t);
-void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, int*, hipStream_t);
-void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, int*, hipStream_t);
+void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, const float*, int, int*, hipStream_t);
+void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, const float*, int, int*, hipStream_t);
+void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, const float*, int, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
@@ -138,9 +138,19 @@ void greedy_sample(at::Tensor out, at::Tensor logits) {
   HIP_CHECK_LAST();
 }
 
+static const float* sink_ptr_checked(const c10::optional<at::Tensor>& sinks,
+                                     int Hq) {
+  if (!sinks.has_value()) return nullptr;
+  TORCH_CHECK(sinks->scalar_type() == at::kFloat && sinks->is_contiguous()
+                  && sinks->numel() == Hq,
+              "sinks must be float32 [Hq]");
+  return sinks->data_ptr<float>();
+}
+
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
-                       at::Tensor seq_lens, double scale) {
+                       at::Tensor seq_lens, double scale,
+                       c10::optional<at::Tensor> sinks, long window) {
   check_bf16(out, "out");
   const long qstride = row_stride_3d(q, "q");
   check_cache(k_cache, "k_cache"); check_cache(v_cache, "v_cache");
@@ -151,12 +161,14 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   const int BS = k_cache.size(2);
   TORCH_CHECK(BS == 16, "decode kernel assumes block_size 16");
   const int max_blocks = block_tables.size(1);
+  const float* sink_ptr = sink_ptr_checked(sinks, Hq);
   int err = 0;
   paged_attn_decode_launch(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                            v_cache.data_ptr(), block_tables.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), N, Hq, Hkv, D, max_blocks,
                            (float)scale, qstride,
-                           is_fp8_cache(k_cache) ? 1 : 0, &err, cur_stream(q));
+                           is_fp8_cache(k_cache) ? 1 : 0, sink_ptr,
+                           (int)window, &err, cur_stream(q));
   TORCH_CHECK(!err, "paged_attn_decode: unsupported head_dim/GQ combination: D=",
               D, " Hq=", Hq, " Hkv=", Hkv);
   HIP_CHECK_LAST();
@@ -164,7 +176,8 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 
 void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                    at::Tensor tile_start, at::Tensor tile_q0,
-                   at::Tensor tile_len, double scale) {
+                   at::Tensor tile_len, double scale,
+                   c10::optional<at::Tensor> sinks, long window) {
   check_bf16(out, "out");
   const long qs = row_stride_3d(q, "q");
   const long ks = row_stride_3d(k, "k");
@@ -176,7 +189,8 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   flash_prefill_launch(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        tile_start.data_ptr<int>(), tile_q0.data_ptr<int>(),
                        tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D,
-                       (float)scale, qs, ks, vs, &err, cur_stream(q));
+                       (float)scale, qs, ks, vs, sink_ptr_checked(sinks, Hq),
+                       (int)window, &err, cur_stream(q));
   TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D);
   HIP_CHECK_LAST();
 }
@@ -185,7 +199,8 @@ void flash_prefill_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                          at::Tensor v_cache, at::Tensor block_tables,
                          at::Tensor tile_qstart, at::Tensor tile_q0,
                          at::Tensor tile_hist, at::Tensor tile_new,
-                         at::Tensor tile_seq, double scale) {
+                         at::Tensor tile_seq, double scale,
+                         c10::optional<at::Tensor> sinks, long window) {
   check_bf16(out, "out");
   const long qs = row_stride_3d(q, "q");
   check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
@@ -204,7 +219,8 @@ void flash_prefill_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       block_tables.data_ptr<int>(), tile_qstart.data_ptr<int>(),
       tile_q0.data_ptr<int>(), tile_hist.data_ptr<int>(),
       tile_new.data_ptr<int>(), tile_seq.data_ptr<int>(), ntiles, Hq, Hkv, D,
-      maxb, (float)scale, qs, &err, cur_stream(q));
+      maxb, (float)scale, qs, sink_ptr_checked(sinks, Hq), (int)window, &err,
+      cur_stream(q));
   TORCH_CHECK(!err, "flash_prefill_paged: unsupported config D=", D);
   HIP_CHECK_LAST();
 }
@@ -347,9 +363,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_and_mul", &silu_and_mul, "silu(x[:d])*x[d:]");
   m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged pool");
   m.def("greedy_sample", &greedy_sample, "argmax over vocab");
-  m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention");
-  m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention");
+  m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention",
+        py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("block_tables"), py::arg("seq_lens"), py::arg("scale"),
+        py::arg("sinks") = py::none(), py::arg("window") = 0);
+  m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention",
+        py::arg("out"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("tile_start"), py::arg("tile_q0"), py::arg("tile_len"),
+        py::arg("scale"), py::arg("sinks") = py::none(),
+        py::arg("window") = 0);
   m.def("flash_prefill_paged", &flash_prefill_paged,
+        py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("block_tables"), py::arg("tile_qstart"), py::arg("tile_q0"),
+        py::arg("tile_hist"), py::arg("tile_new"), py::arg("tile_seq"),
+        py::arg("scale"), py::arg("sinks") = py::none(),
+        py::arg("window") = 0,
         "MFMA prefill attention with paged-KV history (suffix/chunk rows)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");

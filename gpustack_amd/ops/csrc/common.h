@@ -38,6 +38,7 @@ __device__ __forceinline__ unsigned short f2bf(float f) {
 }
 
 typedef __attribute__((ext_vector_type(8))) unsigned char u8x8;
+typedef __attribute__((ext_vector_type(4))) unsigned char u8x4;
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
 
 // OCP fp8 e4m3 (gfx950-native v_cvt_f32_fp8 / v_cvt_pk_fp8_f32)
