@@ -52,6 +52,9 @@ def parse_args():
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree: world becomes ONE replica "
                          "sharded over RCCL/xGMI (default: dp replicas)")
+    ap.add_argument("--cp", type=int, default=1,
+                    help="prefill-context-parallel degree (rows chunked "
+                         "across ranks, replicated decode); world = tp*cp")
     return ap.parse_args()
 
 
@@ -69,17 +72,18 @@ def main():
     dist = None
     comm = None
     tp = max(1, args.tp)
+    cp = max(1, args.cp)
     if world > 1:
         import torch.distributed as tdist
 
         dist = tdist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         dist.init_process_group(backend="nccl" if use_cuda else "gloo")
-        if tp > 1:
-            assert world == tp, "bench --tp requires world_size == tp"
-            from gpustack_amd.parallel import Communicator
+        if tp > 1 or cp > 1:
+            assert world == tp * cp, "bench --tp/--cp require world == tp*cp"
+            from gpustack_amd.parallel import init_parallel
 
-            comm = Communicator(tp, rank)
+            comm = init_parallel(tp, 1, rank, cp_size=cp)
 
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 
@@ -99,11 +103,12 @@ def main():
                          if args.draft_dir else {})}
                      if args.speculative else None),
         tp_size=tp if comm else 1,
-        tp_rank=rank if comm else 0,
+        tp_rank=comm.tp_rank if comm else 0,
     )
     if not use_cuda:  # CPU smoke path: shrink everything
         cfg = EngineConfig(model="tiny", device="cpu", kv_cache_blocks=256, max_model_len=512,
-                           tp_size=tp if comm else 1, tp_rank=rank if comm else 0)
+                           tp_size=tp if comm else 1,
+                           tp_rank=comm.tp_rank if comm else 0)
         args.concurrency = min(args.concurrency, 8)
         args.isl, args.osl = 32, 16
         model = "tiny"

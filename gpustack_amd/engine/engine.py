@@ -24,6 +24,19 @@ class LLMEngine:
         self.cfg = cfg
         self.runner = ModelRunner(cfg, comm)
         self.comm = self.runner.comm
+        if self.comm.cp_size > 1:
+            if cfg.speculative and cfg.speculative.get("method") in (
+                    "eagle", "eagle3", "mtp"):
+                raise ValueError("draft-model speculative decoding is not "
+                                 "supported with context parallelism")
+            if cfg.enable_chunked_prefill:
+                # CP exists to split long prefills ACROSS ranks; chunking
+                # them in time instead would keep every batch under the
+                # budget and CP would never engage (admission stall is
+                # already bounded by the 1/cp prefill speedup)
+                logger.info("disabling chunked prefill under cp=%d",
+                            self.comm.cp_size)
+                cfg.enable_chunked_prefill = False
         kv = self.runner.init_kv_cache()
         self.scheduler = Scheduler(cfg, kv)
         self.seqs: dict[str, Sequence] = {}
@@ -57,11 +70,11 @@ class LLMEngine:
     ) -> str:
         rid = request_id or f"req-{next(self._counter)}"
         params = params or SamplingParams()
-        if (params.guided_json is not None
-                or params.guided_regex is not None) and self.comm.pp_size > 1:
-            raise ValueError("guided_json is not supported with pipeline "
-                             "parallelism (sampling runs on the last stage, "
-                             "which has no token table)")
+        if (params.guided_json is not None or params.guided_regex is not None) \
+                and (self.comm.pp_size > 1 or self.comm.cp_size > 1):
+            raise ValueError("guided decoding is not supported with pipeline "
+                             "or context parallelism (prefill sampling runs "
+                             "on a rank without the token table)")
         if self.comm.world_size > 1:
             assert self.comm.world_rank == 0, "requests enter through rank 0"
             self._pending_ops.append(("add", list(prompt_token_ids),

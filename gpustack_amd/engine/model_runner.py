@@ -211,6 +211,7 @@ class ModelRunner:
     def __init__(self, cfg: EngineConfig, comm: Communicator | None = None):
         self.cfg = cfg
         self.comm = comm or Communicator()
+        self.cp_prefills = 0
         self.device = torch.device(cfg.device)
         import os
 
@@ -402,7 +403,76 @@ class ModelRunner:
         return bool(self.lora_bank.adapters) and any(
             s.lora_slot != 0 for s in batch.seqs)
 
+    def _cp_splittable(self, batch: ScheduledBatch) -> bool:
+        """CP-split applies to pure-prefill batches only; anything else
+        (mixed decode rows, LoRA row groups) falls back to replicated
+        execution — correct by construction, just not accelerated."""
+        if self.comm.cp_size == 1 or not batch.is_prefill:
+            return False
+        n_pre = batch.n_prefill_seqs or len(batch.seqs)
+        if n_pre != len(batch.seqs):
+            return False  # mixed batch: decode rows ride along
+        return not self.batch_uses_lora(batch)
+
+    def _meta_cp(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
+        """Prefill context parallelism (parallel/cp.py): this rank embeds
+        and runs only its contiguous position chunk of every sequence;
+        attention sees the full K/V via the per-layer CP gather + cache
+        write and runs the local rows through the prefill-with-history
+        path (hist = chunk start). Only the tail rank (cp_size-1) produces
+        logits rows; execute() broadcasts its sampled ids world-wide."""
+        from ..parallel import CPMeta, build_cp_prefill
+
+        self.cp_prefills += 1  # observability: tests assert CP engaged
+        dev = self.device
+        comm = self.comm
+        lens = list(batch.seq_lens)
+        local_rows, hists, news, perm, pad_rows, _counts = build_cp_prefill(
+            lens, comm.cp_size, comm.cp_rank)
+        tok_l = [batch.token_ids[i] for i in local_rows]
+        pos_l = [batch.positions[i] for i in local_rows]
+        starts, off = [], 0
+        for n in news:
+            starts.append(off)
+            off += n
+        idx = []
+        if comm.cp_rank == comm.cp_size - 1:
+            # floor-bound partition: the tail chunk is never empty, so this
+            # rank owns every sequence's last row
+            idx = [st + n - 1 for st, n in zip(starts, news)]
+        maxb = max(len(s.block_table) for s in batch.seqs)
+        nseq = len(batch.seqs)
+        dummy = not tok_l
+        bt = torch.zeros(nseq + (1 if dummy else 0), maxb, dtype=torch.int32)
+        for i, s in enumerate(batch.seqs):
+            bt[i, : len(s.block_table)] = torch.tensor(s.block_table,
+                                                       dtype=torch.int32)
+        if dummy:
+            # a rank with zero rows (every seq shorter than cp) still joins
+            # the per-layer CP gathers: run one throwaway row that attends
+            # seq 0's first cached position; perm never selects it and its
+            # output produces no logits
+            tok_l = [batch.token_ids[0]]
+            pos_l = [0]
+            starts, hists, news = [0], [0], [1]
+            bt[nseq] = bt[0]
+        tokens = torch.as_tensor(tok_l, dtype=torch.long).to(dev)
+        tiles = ops.build_paged_prefill_tiles(starts, hists, news, dev)
+        meta = ForwardMeta(
+            is_prefill=False,
+            positions=torch.as_tensor(pos_l, dtype=torch.long).to(dev),
+            slot_mapping=torch.as_tensor(batch.slot_mapping,
+                                         dtype=torch.long).to(dev),
+            logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
+            block_tables=bt.to(dev),
+            suffix_meta=(tiles, starts, hists, news),
+            cp=CPMeta(comm, pad_rows, perm.to(dev)),
+        )
+        return tokens, meta
+
     def _meta(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
+        if self._cp_splittable(batch):
+            return self._meta_cp(batch)
         dev = self.device
         tokens = torch.as_tensor(batch.token_ids, dtype=torch.long).to(dev)
         positions = torch.as_tensor(batch.positions, dtype=torch.long).to(dev)
@@ -634,6 +704,12 @@ class ModelRunner:
             row_seqs = [s for s in batch.seqs for _ in range(batch.rows_per_seq)]
         if self.comm.pp_size > 1:
             return self._pp_finish(batch, logits, row_seqs)
+        if self.comm.cp_size > 1 and self._cp_splittable(batch):
+            # CP prefill: only the tail rank holds logits rows; it samples
+            # and broadcasts (the CP analog of the PP sampling stage)
+            tail = self.comm.cp_rank == self.comm.cp_size - 1
+            return self._pp_finish(batch, logits if tail else None, row_seqs,
+                                   src=self.comm.cp_tail_rank)
         token_ids = self.sampler.sample(logits, row_seqs)
         self.last_logprobs = None
         self.last_top_logprobs = None
@@ -653,26 +729,31 @@ class ModelRunner:
                     list(zip(idx[r].tolist(), vals[r].tolist()))
                     for r in range(lf.shape[0])
                 ]
-        if self.comm.tp_size > 1:
-            # ranks must agree on sampled tokens; rank 0 decides
+        if self.comm.world_size > 1:
+            # ranks must agree on sampled tokens; rank 0 decides (TP
+            # followers and CP replicated-decode ranks receive)
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)
-            self.comm.broadcast(t, src=0)
+            self.comm.broadcast_world(t, src=0)
             token_ids = t.tolist()
         return token_ids
 
-    def _pp_finish(self, batch: ScheduledBatch, logits, row_seqs) -> list[int]:
-        """Pipeline epilogue: the last stage samples (its tp rank 0
-        decides under TPxPP) and every rank receives the token ids so the
-        replicated schedulers stay in lockstep."""
+    def _pp_finish(self, batch: ScheduledBatch, logits, row_seqs,
+                   src: int | None = None) -> list[int]:
+        """Sampling-owner epilogue (PP last stage, or the CP tail rank):
+        the owning rank samples (its tp rank 0 decides) and every rank
+        receives the token ids so the replicated schedulers stay in
+        lockstep."""
         import torch.distributed as dist
 
+        if src is None:
+            src = self.comm.last_stage_rank
         n = len(row_seqs)
         if logits is not None:
             token_ids = self.sampler.sample(logits, row_seqs)
             t = torch.tensor(token_ids, dtype=torch.long, device=self.device)
         else:
             t = torch.empty(n, dtype=torch.long, device=self.device)
-        self.comm.broadcast_world(t, src=self.comm.last_stage_rank)
+        self.comm.broadcast_world(t, src=src)
         token_ids = t.tolist()
         self.last_logprobs = None
         self.last_top_logprobs = None
@@ -696,7 +777,7 @@ class ModelRunner:
                 obj = [((chosen - lse).tolist(), tops)]
             else:
                 obj = [None]
-            dist.broadcast_object_list(obj, src=self.comm.last_stage_rank)
+            dist.broadcast_object_list(obj, src=src)
             if obj[0] is not None:
                 self.last_logprobs, self.last_top_logprobs = obj[0]
         return token_ids

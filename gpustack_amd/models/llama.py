@@ -49,6 +49,12 @@ class ForwardMeta:
     # dynamic multi-LoRA: per-batch adapter row groups (models/lora.py
     # BatchLora); None when no row in the batch uses an adapter
     lora: object | None = None
+    # prefill context parallelism (parallel/cp.py CPMeta): this rank runs a
+    # subset of the batch rows; K/V are all-gathered across the CP group
+    # before the cache write so every rank's paged cache holds the FULL
+    # sequences (slot_mapping is then the full-batch mapping, positions
+    # stay local-row)
+    cp: object | None = None
 
 
 class W4Pack:
@@ -162,7 +168,14 @@ class Attention(nn.Module):
             ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
         ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
                              self.rot_dim)
-        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        if meta.cp is not None:
+            # CP prefill: assemble the full-batch K/V (global position
+            # order) so the cache write below covers every chunk, not just
+            # this rank's rows; q/attention stay local-row
+            kw, vw = meta.cp.gather_kv(k.contiguous(), v.contiguous())
+        else:
+            kw, vw = k, v
+        ops.reshape_and_cache(kw, vw, k_cache, v_cache, meta.slot_mapping)
         out = torch.empty(T, self.hq, self.d, dtype=q.dtype, device=q.device)
         sw = {}
         if self.sinks is not None or self.window:

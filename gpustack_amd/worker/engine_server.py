@@ -1119,6 +1119,11 @@ def main():
     ap.add_argument("--pp", type=int, default=1,
                     help="pipeline stages (layers partitioned across ranks; "
                          "global rank = pp_rank * tp + tp_rank)")
+    ap.add_argument("--cp", type=int, default=1,
+                    help="prefill context parallelism (prompt rows chunked "
+                         "across ranks, per-layer KV all-gather, replicated "
+                         "decode; vLLM --prefill-context-parallel-size "
+                         "analog; mutually exclusive with --pp)")
     ap.add_argument("--tp-rank", type=int, default=None,
                     help="absolute rank of THIS process (set for spawned "
                          "followers; unset = per-worker parent)")
@@ -1140,7 +1145,7 @@ def main():
     device = args.device or ("cuda" if use_cuda else "cpu")
     comm = None
     followers = []
-    world = args.tp * args.pp
+    world = args.tp * args.pp * args.cp
     if args.local_ranks is None:
         args.local_ranks = world if args.tp_rank is None else 1
     if world > 1:
@@ -1163,6 +1168,7 @@ def main():
                 "--gpu-memory-utilization", str(args.gpu_memory_utilization),
                 "--backend-parameters", args.backend_parameters,
                 "--tp", str(args.tp), "--pp", str(args.pp),
+                "--cp", str(args.cp),
                 "--rank-base", str(args.rank_base),
                 "--master-addr", args.master_addr,
             ]
@@ -1181,7 +1187,8 @@ def main():
         comm = init_parallel(args.tp, args.pp, args.tp_rank,
                              master_port=args.master_port,
                              device_id=local_ordinal if use_cuda else None,
-                             master_addr=args.master_addr)
+                             master_addr=args.master_addr,
+                             cp_size=args.cp)
 
     extra = json.loads(args.backend_parameters)
     cfg_kwargs = dict(

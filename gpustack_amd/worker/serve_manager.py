@@ -282,8 +282,12 @@ class ServeManager:
             # backend_parameters pp_size splits layers into pipeline stages
             # instead (tp x pp = scheduled GPUs)
             pp = int(bp.pop("pp_size", 1) or 1)
+            cp = int(bp.pop("cp_size", bp.pop("pcp_size", 1) or 1) or 1)
             if pp > 1 and len(gpus) % pp == 0:
                 args += ["--tp", str(len(gpus) // pp), "--pp", str(pp)]
+            elif cp > 1 and len(gpus) % cp == 0:
+                # prefill context parallelism: tp x cp = scheduled GPUs
+                args += ["--tp", str(len(gpus) // cp), "--cp", str(cp)]
             else:
                 args += ["--tp", str(len(gpus))]
         if bp:
