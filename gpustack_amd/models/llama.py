@@ -425,9 +425,16 @@ class MoEMLP(nn.Module):
     def _fused_ok(self, x) -> bool:
         import os
 
+        # silu/no-bias is the GPU-validated fused configuration; the
+        # GPT-OSS variant (clamped swiglu + expert biases, moe_gemm.hip
+        # act_mode 1) is written + compile-checked but opt-in until the r3
+        # numerics pass (same gate as the attention sinks/window kernels)
+        act_ok = (self.spec.moe_act == "silu" and self.gate_up_b is None) or (
+            self.spec.moe_act == "clamped_swiglu"
+            and os.environ.get("GPUSTACK_AMD_OSS_KERNELS") == "1")
         return (x.is_cuda and x.dtype == torch.bfloat16
                 and os.environ.get("GPUSTACK_AMD_FUSED_MOE", "1") == "1"
-                and self.spec.moe_act == "silu" and self.gate_up_b is None
+                and act_ok
                 and self.i % 128 == 0 and x.shape[1] % 128 == 0
                 and ops.hip_available())
 
@@ -447,10 +454,14 @@ class MoEMLP(nn.Module):
         offs = (counts.cumsum(0, dtype=torch.int32) - counts).to(torch.int32)
         hip = ops._load_hip()
         act = x.new_empty(TK, self.i)
-        hip.moe_gate_up_silu(act, x, self.gate_up_w, s_tok, offs, counts)
+        act_mode = 1 if self.spec.moe_act == "clamped_swiglu" else 0
+        hip.moe_gate_up_silu(act, x, self.gate_up_w, s_tok, offs, counts,
+                             bias=self.gate_up_b, act_mode=act_mode)
         contrib = x.new_empty(TK, x.shape[1])
+        # down bias on tp rank 0 only: the TP all-reduce must sum it once
+        db = self.down_b if self.comm.tp_rank == 0 else None
         hip.moe_down_scale(contrib, act, self.down_w, offs, counts,
-                           order.to(torch.int32), flat_w32)
+                           order.to(torch.int32), flat_w32, bias=db)
         return contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
 
     def _act_mul(self, gu):

@@ -19,8 +19,8 @@ void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, i
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
 void skinny_gemm_v2_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm_lab_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
-void moe_gate_up_silu_launch(void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int*, hipStream_t);
-void moe_down_scale_launch(void*, const void*, const void*, const int*, const int*, const int*, const float*, int, int, int, int*, hipStream_t);
+void moe_gate_up_silu_launch(void*, const void*, const void*, const int*, const int*, const int*, const void*, int, int, int, int, int*, hipStream_t);
+void moe_down_scale_launch(void*, const void*, const void*, const int*, const int*, const int*, const float*, const void*, int, int, int, int*, hipStream_t);
 void w4_gemm_launch(void*, const void*, const void*, const void*, const void*, int, int, int, int*, hipStream_t);
 void w4_dequant_launch(void*, const void*, const void*, const void*, long, int, int*, hipStream_t);
 
@@ -264,7 +264,8 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
 }
 
 void moe_gate_up_silu(at::Tensor act, at::Tensor x, at::Tensor w,
-                      at::Tensor s_tok, at::Tensor offs, at::Tensor counts) {
+                      at::Tensor s_tok, at::Tensor offs, at::Tensor counts,
+                      c10::optional<at::Tensor> bias, long act_mode) {
   check_bf16(act, "act"); check_bf16(x, "x"); check_bf16(w, "w");
   for (auto* t : {&s_tok, &offs, &counts}) {
     TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous());
@@ -274,18 +275,24 @@ void moe_gate_up_silu(at::Tensor act, at::Tensor x, at::Tensor w,
   const int I = (int)(I2 / 2);
   TORCH_CHECK(w.size(2) == H && act.size(1) == I);
   TORCH_CHECK(act.size(0) == s_tok.size(0));
+  const void* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    check_bf16(*bias, "bias");
+    TORCH_CHECK(bias->numel() == (long)E * 2 * I, "gate_up bias must be [E, 2I]");
+    bias_ptr = bias->data_ptr();
+  }
   int err = 0;
   moe_gate_up_silu_launch(act.data_ptr(), x.data_ptr(), w.data_ptr(),
                           s_tok.data_ptr<int>(), offs.data_ptr<int>(),
-                          counts.data_ptr<int>(), E, H, I, &err,
-                          cur_stream(x));
+                          counts.data_ptr<int>(), bias_ptr, (int)act_mode,
+                          E, H, I, &err, cur_stream(x));
   TORCH_CHECK(!err, "moe_gate_up_silu: unsupported dims H=", H, " I=", I);
   HIP_CHECK_LAST();
 }
 
 void moe_down_scale(at::Tensor contrib, at::Tensor act, at::Tensor w,
                     at::Tensor offs, at::Tensor counts, at::Tensor order,
-                    at::Tensor flat_w) {
+                    at::Tensor flat_w, c10::optional<at::Tensor> bias) {
   check_bf16(contrib, "contrib"); check_bf16(act, "act"); check_bf16(w, "w");
   for (auto* t : {&offs, &counts, &order}) {
     TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous());
@@ -294,11 +301,17 @@ void moe_down_scale(at::Tensor contrib, at::Tensor act, at::Tensor w,
   const int E = w.size(0), H = w.size(1), I = w.size(2);
   TORCH_CHECK(act.size(1) == I && contrib.size(1) == H);
   TORCH_CHECK(contrib.size(0) == order.size(0));
+  const void* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    check_bf16(*bias, "bias");
+    TORCH_CHECK(bias->numel() == (long)E * H, "down bias must be [E, H]");
+    bias_ptr = bias->data_ptr();
+  }
   int err = 0;
   moe_down_scale_launch(contrib.data_ptr(), act.data_ptr(), w.data_ptr(),
                         offs.data_ptr<int>(), counts.data_ptr<int>(),
-                        order.data_ptr<int>(), flat_w.data_ptr<float>(), E,
-                        H, I, &err, cur_stream(act));
+                        order.data_ptr<int>(), flat_w.data_ptr<float>(),
+                        bias_ptr, E, H, I, &err, cur_stream(act));
   TORCH_CHECK(!err, "moe_down_scale: unsupported dims H=", H, " I=", I);
   HIP_CHECK_LAST();
 }
@@ -384,8 +397,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm8", &gemm8, "8-phase pipelined 256x256 GEMM (bf16, f32 accum)");
   m.def("gemm_lab", &gemm_lab, "GEMM schedule lab variants (A/B vs hipBLASLt)");
   m.def("moe_gate_up_silu", &moe_gate_up_silu,
+        py::arg("act"), py::arg("x"), py::arg("w"), py::arg("s_tok"),
+        py::arg("offs"), py::arg("counts"), py::arg("bias") = py::none(),
+        py::arg("act_mode") = 0,
         "grouped MoE gate/up GEMM + SiLU (sorted assignments, sync-free)");
   m.def("moe_down_scale", &moe_down_scale,
+        py::arg("contrib"), py::arg("act"), py::arg("w"), py::arg("offs"),
+        py::arg("counts"), py::arg("order"), py::arg("flat_w"),
+        py::arg("bias") = py::none(),
         "grouped MoE down GEMM + routing-weight scale/scatter");
   m.def("w4_gemm", &w4_gemm,
         "W4A16 GEMM: packed-int4 weights dequantized in-register");

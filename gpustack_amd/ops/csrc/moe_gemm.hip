@@ -9,6 +9,12 @@
 //   moe_gate_up_silu: act[j, :] = silu(x[tok_j] @ Wg[e_j]^T) * (x[tok_j] @ Wu[e_j]^T)
 //   moe_down_scale:   contrib[order[j], :] = (act[j] @ Wd[e_j]^T) * w_j
 //
+// GPT-OSS variants (per-expert biases + clamped swiglu, act_mode 1):
+//   gate = min(g + bg, 7); up = clamp(u + bu, -7, 7)
+//   act  = (up + 1) * gate * sigmoid(1.702 * gate)
+//   contrib[order[j], :] = ((act[j] @ Wd[e_j]^T) + bd[e_j]) * w_j
+// (matches models/llama.py MoEMLP._act_mul / the HF GptOss experts)
+//
 // Fixed launch grids (E x N-tiles) independent of the routing outcome:
 // workgroups for empty experts exit immediately. MFMA 16x16x32 tiles,
 // A = 16 gathered token rows staged through LDS in 512-element K chunks,
@@ -55,6 +61,8 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
     const int* __restrict__ s_tok,          // [TK] token index per sorted row
     const int* __restrict__ offs,           // [E]
     const int* __restrict__ counts,         // [E]
+    const unsigned short* __restrict__ bias,  // [E, 2I] bf16 or null
+    int act_mode,                           // 0 = silu, 1 = clamped swiglu
     int H, int I) {
   const int nnt = I / MOE_BN;
   const int e = blockIdx.x / nnt;
@@ -110,13 +118,27 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
         for (int kk = 0; kk < kc; kk += 32) body(kk);
       }
     }
-    // D[row = lg*4 + r][col = lc]; fuse SiLU(gate) * up and write
+    // D[row = lg*4 + r][col = lc]; fuse the activation and write.
+    // bias depends on the column only — one pair of loads per lane.
+    float bg = 0.f, bu = 0.f;
+    if (bias != nullptr) {
+      const int col = nt * MOE_BN + wave * 16 + lc;
+      bg = bf2f(bias[(long)e * 2 * I + col]);
+      bu = bf2f(bias[(long)e * 2 * I + I + col]);
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = lg * 4 + r;
       if (row >= nrows) continue;
-      const float g = ag[r];
-      const float v = g / (1.f + __expf(-g)) * au[r];
+      float v;
+      if (act_mode == 1) {  // GPT-OSS clamped swiglu
+        const float g = fminf(ag[r] + bg, 7.f);
+        const float u = fminf(fmaxf(au[r] + bu, -7.f), 7.f);
+        v = (u + 1.f) * g / (1.f + __expf(-1.702f * g));
+      } else {
+        const float g = ag[r] + bg;
+        v = g / (1.f + __expf(-g)) * (au[r] + bu);
+      }
       act[(long)(base + m0 + row) * I + nt * MOE_BN + wave * 16 + lc] = f2bf(v);
     }
   }
@@ -130,6 +152,7 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
     const int* __restrict__ counts,         // [E]
     const int* __restrict__ order,          // [TK] sorted -> original index
     const float* __restrict__ flat_w,       // [TK] routing weight (original)
+    const unsigned short* __restrict__ bias,  // [E, H] bf16 or null
     int H, int I) {
   const int nnt = H / MOE_BN;
   const int e = blockIdx.x / nnt;
@@ -173,13 +196,15 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
         for (int kk = 0; kk < kc; kk += 32) body(kk);
       }
     }
+    const float bd = (bias != nullptr)
+        ? bf2f(bias[(long)e * H + nt * MOE_BN + wave * 16 + lc]) : 0.f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = lg * 4 + r;
       if (row >= nrows) continue;
       const int oj = order[base + m0 + row];
       contrib[(long)oj * H + nt * MOE_BN + wave * 16 + lc] =
-          f2bf(acc[r] * flat_w[oj]);
+          f2bf((acc[r] + bd) * flat_w[oj]);
     }
   }
 }
@@ -188,19 +213,22 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
 
 void moe_gate_up_silu_launch(void* act, const void* x, const void* w,
                              const int* s_tok, const int* offs,
-                             const int* counts, int E, int H, int I,
+                             const int* counts, const void* bias,
+                             int act_mode, int E, int H, int I,
                              int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
   if (I % MOE_BN != 0 || H % 128 != 0) { *err_unsupported = 1; return; }
   dim3 grid(E * (I / MOE_BN));
   hipLaunchKernelGGL(moe_gate_up_silu_kernel, grid, dim3(MOE_THREADS), 0, s,
                      (unsigned short*)act, (const unsigned short*)x,
-                     (const unsigned short*)w, s_tok, offs, counts, H, I);
+                     (const unsigned short*)w, s_tok, offs, counts,
+                     (const unsigned short*)bias, act_mode, H, I);
 }
 
 void moe_down_scale_launch(void* contrib, const void* act, const void* w,
                            const int* offs, const int* counts,
-                           const int* order, const float* flat_w, int E,
+                           const int* order, const float* flat_w,
+                           const void* bias, int E,
                            int H, int I, int* err_unsupported, hipStream_t s) {
   *err_unsupported = 0;
   if (H % MOE_BN != 0 || I % 128 != 0) { *err_unsupported = 1; return; }
@@ -208,5 +236,5 @@ void moe_down_scale_launch(void* contrib, const void* act, const void* w,
   hipLaunchKernelGGL(moe_down_scale_kernel, grid, dim3(MOE_THREADS), 0, s,
                      (unsigned short*)contrib, (const unsigned short*)act,
                      (const unsigned short*)w, offs, counts, order, flat_w,
-                     H, I);
+                     (const unsigned short*)bias, H, I);
 }

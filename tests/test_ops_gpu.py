@@ -616,3 +616,54 @@ def test_oss_paged_prefill(D, cases, sink, window):
     R.paged_prefill_attn(ref, q, kc, vc, bt, starts, hists, news, scale,
                          sinks=sinks, window=window)
     _close(out, ref)
+
+
+@oss
+@pytest.mark.parametrize("bias", [True, False])
+def test_oss_fused_moe_clamped_swiglu(bias):
+    """moe_gate_up_silu act_mode=1 (+expert biases) + moe_down_scale bias
+    vs plain torch: clamped swiglu per models/llama.py MoEMLP._act_mul."""
+    torch.manual_seed(0)
+    E, H, I, T, K = 8, 256, 128, 33, 2
+    dev = "cuda"
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) / 4
+    w_gu = torch.randn(E, 2 * I, H, dtype=torch.bfloat16, device=dev) / 8
+    w_d = torch.randn(E, H, I, dtype=torch.bfloat16, device=dev) / 8
+    b_gu = (torch.randn(E, 2 * I, dtype=torch.bfloat16, device=dev) / 4
+            if bias else None)
+    b_d = (torch.randn(E, H, dtype=torch.bfloat16, device=dev) / 4
+           if bias else None)
+    flat_exp = torch.randint(0, E, (T * K,), device=dev)
+    flat_tok = torch.arange(T, device=dev).repeat_interleave(K)
+    flat_w32 = torch.rand(T * K, dtype=torch.float32, device=dev)
+
+    order = torch.argsort(flat_exp, stable=True)
+    s_tok = flat_tok[order].to(torch.int32)
+    counts = torch.zeros(E, dtype=torch.int32, device=dev)
+    counts.scatter_add_(0, flat_exp, torch.ones_like(flat_exp, dtype=torch.int32))
+    offs = (counts.cumsum(0, dtype=torch.int32) - counts).to(torch.int32)
+
+    hip = ops._load_hip()
+    act = x.new_empty(T * K, I)
+    hip.moe_gate_up_silu(act, x, w_gu, s_tok, offs, counts, bias=b_gu,
+                         act_mode=1)
+    contrib = x.new_empty(T * K, H)
+    hip.moe_down_scale(contrib, act, w_d, offs, counts,
+                       order.to(torch.int32), flat_w32, bias=b_d)
+
+    # torch reference (fp32)
+    ref = torch.zeros(T * K, H, dtype=torch.float32, device=dev)
+    for j in range(T * K):
+        e = int(flat_exp[j])
+        t = int(flat_tok[j])
+        gu = x[t].float() @ w_gu[e].float().T
+        if b_gu is not None:
+            gu = gu + b_gu[e].float()
+        g = gu[:I].clamp(max=7.0)
+        u = gu[I:].clamp(-7.0, 7.0)
+        a = (u + 1.0) * (g * torch.sigmoid(g * 1.702))
+        d = a @ w_d[e].float().T
+        if b_d is not None:
+            d = d + b_d[e].float()
+        ref[j] = d * flat_w32[j]
+    _close(contrib, ref.to(torch.bfloat16), atol=6e-2, rtol=6e-2)
