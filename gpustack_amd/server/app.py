@@ -252,6 +252,20 @@ def create_app(cfg: Config, start_background: bool = True) -> FastAPI:
     boot = bootstrap_data(cfg)
     if cfg.token is None:
         cfg.token = boot["registration_token"]
+    # persist the registration token where operators (and the all-in-one
+    # container's embedded worker) expect it — reference behavior: the
+    # server writes <data_dir>/token and `gpustack start --server-url`
+    # workers read it
+    from pathlib import Path as _Path
+
+    tok_path = _Path(cfg.data_dir) / "token"
+    try:
+        if (not tok_path.exists()
+                or tok_path.read_text().strip() != cfg.token):
+            tok_path.write_text(cfg.token + "\n")
+            tok_path.chmod(0o600)
+    except OSError as e:  # read-only data dir: log, don't fail startup
+        logger.warning("could not persist registration token: %s", e)
 
     app = FastAPI(title="gpustack_amd", version="0.1.0")
     app.state.config = cfg

@@ -507,3 +507,14 @@ def test_resource_event_logger_records_transitions(server):
     with get_session() as s:
         assert s.query(ResourceEventArchive).count() == 1
         assert s.query(ResourceEvent).count() == 1
+
+
+def test_registration_token_persisted_to_data_dir(server):
+    """The server writes <data_dir>/token at startup (reference behavior;
+    the all-in-one container's embedded worker joins by reading it)."""
+    import pathlib
+
+    _client, _app, cfg, reg_token = server
+    p = pathlib.Path(cfg.data_dir) / "token"
+    assert p.read_text().strip() == reg_token
+    assert (p.stat().st_mode & 0o777) == 0o600
