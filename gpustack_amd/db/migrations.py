@@ -79,6 +79,11 @@ def _m010_resource_events(conn):
     ResourceEventArchive.__table__.create(conn, checkfirst=True)
 
 
+def _m011_gpu_instances(conn):
+    from ..schemas.tables import GPUInstance
+    GPUInstance.__table__.create(conn, checkfirst=True)
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -90,6 +95,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (8, "orgs table + user/model org scoping", _m008_orgs),
     (9, "model.gpu_type_selector for device-class placement", _m009_gpu_type_selector),
     (10, "resource-event metering pair (hot + archive)", _m010_resource_events),
+    (11, "gpu_instances table (operator-analog SSH GPU pods)", _m011_gpu_instances),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

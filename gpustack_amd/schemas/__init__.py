@@ -6,6 +6,7 @@ from pydantic import BaseModel, Field
 from .tables import (  # noqa: F401
     ApiKey, Cluster,
     Benchmark,
+    GPUInstance, GPUInstanceState,
     Model, Org,
     ModelFile,
     ModelInstance,
@@ -173,6 +174,17 @@ class WorkerPoolUpdate(BaseModel):
     replicas: int | None = None
     provider_config: dict | None = None
     labels: dict | None = None
+
+
+class GPUInstanceCreate(BaseModel):
+    name: str
+    flavor: str = "mi355x-1gpu"
+    image: str = "rocm/dev-ubuntu-24.04"
+    ssh_public_key: str = ""
+    provider: str = "k8s"
+    provider_config: dict = Field(default_factory=dict)
+    volumes: list[dict] = Field(default_factory=list)
+    labels: dict = Field(default_factory=dict)
 
 
 class ModelProviderCreate(BaseModel):

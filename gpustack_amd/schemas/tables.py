@@ -253,6 +253,37 @@ class WorkerPool(Base, TimestampMixin, SerializeMixin):
     state_message = Column(Text, default="")
 
 
+class GPUInstanceState(str, enum.Enum):
+    PENDING = "pending"
+    CREATING = "creating"
+    RUNNING = "running"
+    DELETING = "deleting"
+    ERROR = "error"
+
+
+class GPUInstance(Base, TimestampMixin, SerializeMixin):
+    """On-demand SSH-accessible GPU pod (reference: gpustack-operator CRDs
+    + gpustack/gpu_instances/ — templates, flavors, persistent volumes).
+    First-party equivalent: the server's GPUInstanceController drives the
+    pod lifecycle directly through utils/k8s_client.py (or the mock
+    provider for tests/dry-runs) — no Go sidecar process."""
+    __tablename__ = "gpu_instances"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    flavor = Column(String(128), default="mi355x-1gpu")  # gpu count preset
+    image = Column(String(512), default="rocm/dev-ubuntu-24.04")
+    ssh_public_key = Column(Text, default="")
+    provider = Column(String(64), default="k8s")         # k8s | mock
+    provider_config = Column(JSON, default=dict)         # api_server/ns/...
+    volumes = Column(JSON, default=list)  # [{name, size_gb, mount_path}]
+    labels = Column(JSON, default=dict)
+    state = Column(String(32), default=GPUInstanceState.PENDING.value)
+    state_message = Column(Text, default="")
+    external_id = Column(String(256), default="")        # pod name / mock id
+    ssh_host = Column(String(256), default="")
+    ssh_port = Column(Integer, default=0)
+
+
 class Benchmark(Base, TimestampMixin, SerializeMixin):
     """In-product benchmark runs (reference: schemas/benchmark.py)."""
     __tablename__ = "benchmarks"

@@ -348,10 +348,13 @@ def start_background_tasks(cfg: Config, app: FastAPI) -> None:
             coord = LocalCoordinator()
     app.state.coordinator = coord
 
+    from .gpu_instances import GPUInstanceController
+
     sched = PlacementScheduler(cfg)
     tasks = [sched, ModelController(cfg), WorkerMonitor(cfg),
              SystemLoadCollector(cfg), ScalingScheduler(cfg), UsageArchiver(cfg),
-             WorkerPoolController(cfg), ResourceEventLogger(cfg)]
+             WorkerPoolController(cfg), ResourceEventLogger(cfg),
+             GPUInstanceController(cfg)]
     for t in tasks:
         t.coordinator = coord  # leader-only gating (checked per cycle)
     app.state.scheduler = sched

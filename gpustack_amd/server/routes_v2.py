@@ -20,6 +20,7 @@ from ..schemas import (
     ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
     ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
     ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
+    GPUInstanceCreate,
     WorkerPoolCreate, WorkerPoolUpdate, WorkerRegister, WorkerState,
     WorkerStatusUpdate,
 )
@@ -497,6 +498,72 @@ def delete_cluster(cluster_id: int, _: User = Depends(get_admin_user)):
         s.commit()  # tokens must go before the cluster row (FK, no ORM rel)
         ar_delete(s, c)
         return {"deleted": cluster_id}
+
+
+# -- GPU instances (operator-analog SSH GPU pods, server/gpu_instances.py) --
+
+@router.get("/gpu_instances")
+def list_gpu_instances(_: User = Depends(get_current_user)):
+    from ..schemas import GPUInstance
+
+    with get_session() as s:
+        return {"items": [g.to_dict() for g in s.query(GPUInstance).all()]}
+
+
+@router.get("/gpu_instances/{gid}")
+def get_gpu_instance(gid: int, _: User = Depends(get_current_user)):
+    from ..schemas import GPUInstance
+
+    with get_session() as s:
+        g = s.get(GPUInstance, gid)
+        if not g:
+            raise HTTPException(404, "gpu instance not found")
+        return g.to_dict()
+
+
+@router.post("/gpu_instances", status_code=201)
+def create_gpu_instance(body: GPUInstanceCreate,
+                        _: User = Depends(get_admin_user)):
+    from ..schemas import GPUInstance
+    from .gpu_instances import FLAVORS, PROVIDERS
+
+    if body.provider not in PROVIDERS:
+        raise HTTPException(400, f"unknown provider {body.provider!r}")
+    if body.flavor not in FLAVORS:
+        raise HTTPException(400, f"unknown flavor {body.flavor!r}; "
+                                 f"one of {sorted(FLAVORS)}")
+    with get_session() as s:
+        if s.query(GPUInstance).filter_by(name=body.name).first():
+            raise HTTPException(409, "gpu instance exists")
+        g = GPUInstance(name=body.name, flavor=body.flavor, image=body.image,
+                        ssh_public_key=body.ssh_public_key,
+                        provider=body.provider,
+                        provider_config=body.provider_config,
+                        volumes=body.volumes, labels=body.labels)
+        ar_create(s, g)
+        return g.to_dict()
+
+
+@router.delete("/gpu_instances/{gid}")
+def delete_gpu_instance(gid: int, _: User = Depends(get_admin_user)):
+    """Marks DELETING; the controller deprovisions the pod and removes
+    the row (async, like instance teardown elsewhere)."""
+    from ..schemas import GPUInstance, GPUInstanceState
+
+    with get_session() as s:
+        g = s.get(GPUInstance, gid)
+        if not g:
+            raise HTTPException(404, "gpu instance not found")
+        g.state = GPUInstanceState.DELETING.value
+        ar_update(s, g)
+    return {"status": "deleting"}
+
+
+@router.get("/gpu_instance_flavors")
+def list_gpu_instance_flavors(_: User = Depends(get_current_user)):
+    from .gpu_instances import FLAVORS
+
+    return {"items": [{"name": k, **v} for k, v in FLAVORS.items()]}
 
 
 @router.get("/worker_pools")
