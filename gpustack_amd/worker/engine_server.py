@@ -296,6 +296,116 @@ def _apply_lora_routing(sp, body: dict, runner: "EngineRunner") -> None:
         sp.lora_name = name
 
 
+def create_encoder_app(model_dir: str, served_name: str,
+                       device: str = "cpu") -> FastAPI:
+    """Serving app for cross-encoder reranker checkpoints
+    (models/encoder.py): /v1/rerank scores (query, doc) PAIRS through the
+    bidirectional encoder — the real reranker path; generative endpoints
+    404 (the reference's reranker-category instances behave the same)."""
+    import torch as _torch
+
+    from ..models.encoder import (CrossEncoderModel, CrossEncoderRunner,
+                                  EncoderSpec)
+
+    spec = EncoderSpec.from_dir(model_dir)
+    dtype = _torch.bfloat16 if device.startswith("cuda") else _torch.float32
+    model = CrossEncoderModel(spec, device=device, dtype=dtype)
+    from pathlib import Path as _Path
+
+    if list(_Path(model_dir).glob("*.safetensors")):
+        model.load_dir(model_dir)
+    else:
+        model.random_init(seed=0)
+    tok = load_tokenizer(model_dir, spec.vocab_size)
+
+    def _special(name: str, fallback: int) -> int:
+        getter = getattr(tok, "token_to_id", None)
+        if getter is not None:
+            for cand in name.split("|"):
+                tid = getter(cand)
+                if tid is not None:
+                    return tid
+        return fallback
+    if spec.is_roberta:
+        cls_id = _special("<s>|[CLS]", 0)
+        sep_id = _special("</s>|[SEP]", 2)
+    else:
+        cls_id = _special("[CLS]|<s>", min(101, spec.vocab_size - 2))
+        sep_id = _special("[SEP]|</s>", min(102, spec.vocab_size - 1))
+    ce = CrossEncoderRunner(model, cls_id, sep_id, device=device)
+    stats = {"requests": 0, "pairs": 0}
+    app = FastAPI(title=f"gpustack_amd reranker: {served_name}")
+
+    def _enc(text: str) -> list[int]:
+        ids = tok.encode(text)
+        if hasattr(ids, "ids"):
+            ids = ids.ids
+        return list(ids) or [0]
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok", "model": served_name, "mode": "reranker"}
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list", "data": [{
+            "id": served_name, "object": "model", "owned_by": "gpustack_amd",
+            "meta": {"category": "reranker"}}]}
+
+    @app.get("/metrics")
+    async def metrics():
+        return Response("\n".join([
+            "# TYPE gpustack_engine_requests_total counter",
+            f"gpustack_engine_requests_total {stats['requests']}",
+            "# TYPE gpustack_rerank_pairs_total counter",
+            f"gpustack_rerank_pairs_total {stats['pairs']}",
+        ]) + "\n", media_type="text/plain")
+
+    @app.post("/v1/rerank")
+    @app.post("/rerank")
+    async def rerank(request: Request):
+        import math as _math
+
+        body = await request.json()
+        query = body.get("query", "")
+        docs = body.get("documents", [])
+        top_n = int(body.get("top_n", len(docs)) or len(docs))
+        stats["requests"] += 1
+        stats["pairs"] += len(docs)
+        q_ids = _enc(query)
+        doc_ids = [_enc(d) for d in docs]
+        logits = await asyncio.to_thread(ce.score, q_ids, doc_ids)
+        scored = [{"index": i,
+                   "relevance_score": 1.0 / (1.0 + _math.exp(-x)),
+                   "document": {"text": docs[i]}}
+                  for i, x in enumerate(logits)]
+        scored.sort(key=lambda r: -r["relevance_score"])
+        ntok = len(q_ids) * len(docs) + sum(len(d) for d in doc_ids)
+        return {"model": served_name, "results": scored[:top_n],
+                "usage": {"prompt_tokens": ntok, "total_tokens": ntok}}
+
+    @app.post("/v1/score")
+    async def score(request: Request):
+        """Pairwise relevance (sigmoid of the pair logit)."""
+        import math as _math
+
+        body = await request.json()
+        t1 = body.get("text_1", "")
+        t2 = body.get("text_2", [])
+        if isinstance(t2, str):
+            t2 = [t2]
+        stats["requests"] += 1
+        stats["pairs"] += len(t2)
+        logits = await asyncio.to_thread(
+            ce.score, _enc(t1), [_enc(t) for t in t2])
+        return {"object": "list", "model": served_name,
+                "data": [{"index": i,
+                          "score": 1.0 / (1.0 + _math.exp(-x))}
+                         for i, x in enumerate(logits)]}
+
+    return app
+
+
 def create_app(runner: EngineRunner) -> FastAPI:
     app = FastAPI(title="gpustack_amd-engine")
 
@@ -1189,6 +1299,24 @@ def main():
                              device_id=local_ordinal if use_cuda else None,
                              master_addr=args.master_addr,
                              cp_size=args.cp)
+
+    from pathlib import Path as _Path
+
+    _cand = _Path(args.model_ref)
+    if _cand.is_dir() and (_cand / "config.json").exists():
+        import json as _json
+
+        with open(_cand / "config.json") as _f:
+            _arch = (_json.load(_f).get("architectures") or [""])[0]
+        from ..models.encoder import is_encoder_arch
+
+        if is_encoder_arch(_arch):
+            app = create_encoder_app(str(_cand), args.served_name, device)
+            import uvicorn
+
+            uvicorn.run(app, host=args.host, port=args.port,
+                        log_level="warning")
+            return
 
     extra = json.loads(args.backend_parameters)
     cfg_kwargs = dict(
