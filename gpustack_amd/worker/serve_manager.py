@@ -45,6 +45,7 @@ class InstanceProcess:
         self.started_at = time.time()
         self.healthy = False
         self.role = "main"
+        self.fence = None  # CgroupFence when resource limits are requested
 
 
 class ServeManager:
@@ -293,10 +294,20 @@ class ServeManager:
         if bp:
             args += ["--backend-parameters", json.dumps(bp)]
 
+        # resource fencing (reference: gpustack-runtime container limits):
+        # backend_parameters memory_limit_gb / cpu_limit / pids_limit put
+        # the engine in its own cgroup — best-effort, never blocks serving
+        from .isolation import fence_from_backend_parameters
+
+        fence = fence_from_backend_parameters(
+            f"{inst['id']}-{role}", model.get("backend_parameters") or {})
         logf = open(log_path, "ab")
         proc = subprocess.Popen(args, env=env, stdout=logf, stderr=subprocess.STDOUT,
                                 start_new_session=True)
+        if fence is not None and fence.create():
+            fence.attach(proc.pid)
         ip = InstanceProcess(inst, proc, port, log_path)
+        ip.fence = fence
         ip.role = role
         self.processes[iid] = ip
         if role == "main":
@@ -361,6 +372,8 @@ class ServeManager:
                 os.killpg(ip.proc.pid, 9)
             except ProcessLookupError:
                 pass
+        if ip.fence is not None:
+            ip.fence.cleanup()
         with self._lock:
             self._used_ports.discard(ip.port)
 
