@@ -29,14 +29,6 @@ class LLMEngine:
                     "eagle", "eagle3", "mtp"):
                 raise ValueError("draft-model speculative decoding is not "
                                  "supported with context parallelism")
-            if cfg.enable_chunked_prefill:
-                # CP exists to split long prefills ACROSS ranks; chunking
-                # them in time instead would keep every batch under the
-                # budget and CP would never engage (admission stall is
-                # already bounded by the 1/cp prefill speedup)
-                logger.info("disabling chunked prefill under cp=%d",
-                            self.comm.cp_size)
-                cfg.enable_chunked_prefill = False
         kv = self.runner.init_kv_cache()
         self.scheduler = Scheduler(cfg, kv)
         self.seqs: dict[str, Sequence] = {}

@@ -212,6 +212,7 @@ class ModelRunner:
         self.cfg = cfg
         self.comm = comm or Communicator()
         self.cp_prefills = 0
+        self.cp_suffixes = 0
         self.device = torch.device(cfg.device)
         import os
 
@@ -404,15 +405,29 @@ class ModelRunner:
             s.lora_slot != 0 for s in batch.seqs)
 
     def _cp_splittable(self, batch: ScheduledBatch) -> bool:
-        """CP-split applies to pure-prefill batches only; anything else
-        (mixed decode rows, LoRA row groups) falls back to replicated
-        execution — correct by construction, just not accelerated."""
-        if self.comm.cp_size == 1 or not batch.is_prefill:
+        """CP-split applies to pure-prefill batches and to suffix/chunk
+        batches on the paged-prefill path; anything else (mixed decode
+        rows, LoRA row groups, fp8-KV suffix fallback) falls back to
+        replicated execution — correct by construction, just not
+        accelerated."""
+        if self.comm.cp_size == 1:
             return False
-        n_pre = batch.n_prefill_seqs or len(batch.seqs)
-        if n_pre != len(batch.seqs):
-            return False  # mixed batch: decode rows ride along
-        return not self.batch_uses_lora(batch)
+        if self.batch_uses_lora(batch):
+            return False
+        if batch.is_prefill:
+            n_pre = batch.n_prefill_seqs or len(batch.seqs)
+            return n_pre == len(batch.seqs)  # mixed: decode rows ride along
+        if batch.is_suffix:
+            # chunk continuations / prefix-cache suffixes: the CP path runs
+            # the prefill-with-history attention, so it needs the same
+            # conditions as the paged-prefill kernel dispatch
+            if self.kv is None:
+                return False
+            if self.device.type == "cuda":
+                return (self.kv.kv_dtype == torch.bfloat16
+                        and self.cfg.spec.head_dim == 128)
+            return self.kv.kv_dtype != torch.float8_e4m3fn
+        return False
 
     def _meta_cp(self, batch: ScheduledBatch) -> tuple[torch.Tensor, ForwardMeta]:
         """Prefill context parallelism (parallel/cp.py): this rank embeds
@@ -424,11 +439,25 @@ class ModelRunner:
         from ..parallel import CPMeta, build_cp_prefill
 
         self.cp_prefills += 1  # observability: tests assert CP engaged
+        if batch.is_suffix:
+            self.cp_suffixes += 1
         dev = self.device
         comm = self.comm
-        lens = list(batch.seq_lens)
+        if batch.is_suffix:
+            # chunk-continuation / suffix rows: split each seq's NEW rows;
+            # this rank's history = the seq's cached tokens + the chunk
+            # rows owned by earlier CP ranks
+            lens = list(batch.suffix_rows)
+            base_hists = [
+                batch.seq_lens[sum(batch.suffix_rows[:i])] - 1
+                for i in range(len(batch.seqs))
+            ]
+        else:
+            lens = list(batch.seq_lens)
+            base_hists = [0] * len(batch.seqs)
         local_rows, hists, news, perm, pad_rows, _counts = build_cp_prefill(
             lens, comm.cp_size, comm.cp_rank)
+        hists = [b + h for b, h in zip(base_hists, hists)]
         tok_l = [batch.token_ids[i] for i in local_rows]
         pos_l = [batch.positions[i] for i in local_rows]
         starts, off = [], 0
@@ -452,7 +481,7 @@ class ModelRunner:
             # the per-layer CP gathers: run one throwaway row that attends
             # seq 0's first cached position; perm never selects it and its
             # output produces no logits
-            tok_l = [batch.token_ids[0]]
+            tok_l = [int(batch.token_ids[0])]
             pos_l = [0]
             starts, hists, news = [0], [0], [1]
             bt[nseq] = bt[0]

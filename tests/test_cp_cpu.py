@@ -34,18 +34,17 @@ PROMPTS = [[3, 1, 4, 1, 5, 9, 2, 6, 5, 3, 5, 8, 9, 7, 9, 3, 2, 3, 8, 4, 6, 2,
            [7]]
 
 
-def _single_proc_result(model: str = "tiny") -> list[list[int]]:
+def _single_proc_result(model: str = "tiny", **kw) -> list[list[int]]:
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 
     cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
-                       max_model_len=128, seed=0, dtype="float32",
-                       enable_chunked_prefill=False)
+                       max_model_len=128, seed=0, dtype="float32", **kw)
     eng = LLMEngine(cfg)
     return eng.generate(PROMPTS, SamplingParams(max_tokens=6, ignore_eos=True))
 
 
 def _cp_rank_main(rank: int, world: int, port: int, out_path: str,
-                  model: str = "tiny"):
+                  model: str = "tiny", cfg_kw: dict | None = None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
@@ -55,9 +54,9 @@ def _cp_rank_main(rank: int, world: int, port: int, out_path: str,
                          cp_size=world)
     assert comm.cp_rank == rank and comm.world_size == world
     cfg = EngineConfig(model=model, device="cpu", kv_cache_blocks=64,
-                       max_model_len=128, seed=0, dtype="float32")
+                       max_model_len=128, seed=0, dtype="float32",
+                       **(cfg_kw or {}))
     eng = LLMEngine(cfg, comm)
-    assert not eng.cfg.enable_chunked_prefill  # auto-disabled under cp
     results: dict[str, list[int]] = {}
     if rank == 0:
         rids = [eng.add_request(p, SamplingParams(max_tokens=6,
@@ -70,8 +69,11 @@ def _cp_rank_main(rank: int, world: int, port: int, out_path: str,
             for o in outs:
                 results[o.request_id].append(o.token_id)
     # the CP-split path must actually have run (not a silent replicated
-    # fallback) on every rank
+    # fallback) on every rank; with chunked admission the suffix
+    # continuations must ALSO have split
     assert eng.runner.cp_prefills > 0
+    if cfg.enable_chunked_prefill and (cfg_kw or {}).get("max_prefill_tokens"):
+        assert eng.runner.cp_suffixes > 0
     if rank == 0:
         with open(out_path, "w") as f:
             json.dump([results[r] for r in rids], f)
@@ -81,12 +83,12 @@ def _cp_rank_main(rank: int, world: int, port: int, out_path: str,
     dist.destroy_process_group()
 
 
-def _run_cp2(model: str) -> list[list[int]]:
+def _run_cp2(model: str, cfg_kw: dict | None = None) -> list[list[int]]:
     port = _free_port()
     out_path = tempfile.mktemp(suffix=".json")
     ctx = mp.get_context("spawn")
-    procs = [ctx.Process(target=_cp_rank_main, args=(r, 2, port, out_path,
-                                                     model))
+    procs = [ctx.Process(target=_cp_rank_main,
+                         args=(r, 2, port, out_path, model, cfg_kw))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -176,6 +178,14 @@ def test_guided_rejected_under_cp():
 
     with pytest.raises(ValueError, match="context parallelism"):
         eng.add_request([1, 2, 3], SamplingParams(guided_json=True))
+
+
+def test_cp2_chunked_prefill_matches_single_rank():
+    """Chunked admission composes with CP: chunk 0 is a CP-split prefill
+    and every continuation is a CP-split suffix batch (history = cached
+    tokens + earlier ranks' chunk rows) — output must still be exact."""
+    kw = {"enable_chunked_prefill": True, "max_prefill_tokens": 16}
+    assert _run_cp2("tiny", kw) == _single_proc_result("tiny", **kw)
 
 
 def test_cp2_matches_single_rank_moe():
