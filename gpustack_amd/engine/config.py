@@ -60,18 +60,42 @@ class ModelSpec:
     moe_bias: bool = False               # expert gate_up/down biases
     router_logit_bias: bool = False      # bias added to router logits
     o_proj_bias: bool = False
+    # DeepSeek V2/V3/R1 MLA (DeepseekV3ForCausalLM, models/llama.py
+    # MLAAttention): kv_lora_rank > 0 switches attention to the latent
+    # formulation — the paged cache stores [kv_lora_rank + qk_rope]
+    # per token per layer instead of 2 * kv_heads * head_dim
+    q_lora_rank: int = 0
+    kv_lora_rank: int = 0
+    qk_nope_head_dim: int = 0
+    qk_rope_head_dim: int = 0
+    v_head_dim: int = 0
+    rope_interleave: bool = True         # deepseek weights: paired dims
 
     @property
     def gqa_ratio(self) -> int:
         return self.num_heads // self.num_kv_heads
 
     def kv_bytes_per_token(self, dtype_size: int = 2) -> int:
+        if self.kv_lora_rank:
+            # MLA: one compressed latent row per token per layer
+            return (self.num_layers
+                    * (self.kv_lora_rank + self.qk_rope_head_dim)
+                    * dtype_size)
         return 2 * self.num_layers * self.num_kv_heads * self.head_dim * dtype_size
 
     def weight_bytes(self, dtype_size: int = 2) -> int:
         h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
-        qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
-        o = self.num_heads * self.head_dim * h
+        if self.kv_lora_rank:
+            dq = self.qk_nope_head_dim + self.qk_rope_head_dim
+            qp = (self.q_lora_rank * (h + self.num_heads * dq)
+                  if self.q_lora_rank else h * self.num_heads * dq)
+            qkv = (qp + h * (self.kv_lora_rank + self.qk_rope_head_dim)
+                   + self.kv_lora_rank * self.num_heads
+                   * (self.qk_nope_head_dim + self.v_head_dim))
+            o = self.num_heads * self.v_head_dim * h
+        else:
+            qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
+            o = self.num_heads * self.head_dim * h
         attn_norm = qkv + o + 2 * h
         dense_mlp = 3 * h * i
         if self.num_experts > 0:
@@ -128,7 +152,9 @@ class ModelSpec:
             if (cfg.get("num_experts") or cfg.get("num_local_experts")
                 or cfg.get("n_routed_experts")) else 0,
             norm_topk_prob=bool(cfg.get("norm_topk_prob", True)),
-            router_mode=("sigmoid_bias" if arch.startswith("Glm4Moe")
+            router_mode=("sigmoid_bias"
+                         if (arch.startswith("Glm4Moe")
+                             or arch.startswith("Deepseek"))
                          else "softmax"),
             n_shared_experts=cfg.get("n_shared_experts", 0) or 0,
             first_k_dense_replace=cfg.get("first_k_dense_replace", 0) or 0,
@@ -148,6 +174,14 @@ class ModelSpec:
             router_logit_bias=arch.startswith("GptOss"),
             o_proj_bias=bool(cfg.get("attention_bias"))
             if arch.startswith("GptOss") else False,
+            q_lora_rank=(cfg.get("q_lora_rank") or 0)
+            if arch.startswith("Deepseek") else 0,
+            kv_lora_rank=(cfg.get("kv_lora_rank") or 0)
+            if arch.startswith("Deepseek") else 0,
+            qk_nope_head_dim=cfg.get("qk_nope_head_dim", 0) or 0,
+            qk_rope_head_dim=cfg.get("qk_rope_head_dim", 0) or 0,
+            v_head_dim=cfg.get("v_head_dim", 0) or 0,
+            rope_interleave=bool(cfg.get("rope_interleave", True)),
         )
 
     @classmethod
@@ -159,6 +193,26 @@ class ModelSpec:
 # Named presets so the bench / tests / control plane can run with
 # random-init weights and no network (BASELINE.json: synthetic data).
 PRESETS: dict[str, ModelSpec] = {
+    # DeepSeek-V3/R1 671B: MLA latent attention + 256-expert sigmoid+bias
+    # grouped MoE (the R1 checkpoints share this architecture)
+    "deepseek-v3": ModelSpec(
+        architecture="DeepseekV3ForCausalLM", vocab_size=129280,
+        hidden_size=7168, intermediate_size=18432, num_layers=61,
+        num_heads=128, num_kv_heads=128, head_dim=192,
+        rope_theta=10000.0, max_position_embeddings=163840,
+        rope_scaling={"rope_type": "yarn", "factor": 40.0,
+                      "beta_fast": 32.0, "beta_slow": 1.0,
+                      "mscale": 1.0, "mscale_all_dim": 1.0,
+                      "original_max_position_embeddings": 4096},
+        eos_token_id=1, num_experts=256, num_experts_per_tok=8,
+        moe_intermediate_size=2048, router_mode="sigmoid_bias",
+        n_shared_experts=1, first_k_dense_replace=3,
+        routed_scaling_factor=2.5, n_group=8, topk_group=4,
+        norm_topk_prob=True,
+        q_lora_rank=1536, kv_lora_rank=512, qk_nope_head_dim=128,
+        qk_rope_head_dim=64, v_head_dim=128,
+    ),
+
     "llama-3-8b": ModelSpec(),
     "llama-3-70b": ModelSpec(
         hidden_size=8192, intermediate_size=28672, num_layers=80,

@@ -143,7 +143,18 @@ class KVCache:
         self.head_dim = spec.head_dim
         # pipeline stages hold KV only for their own layers
         self.num_layers = num_layers if num_layers is not None else spec.num_layers
-        shape = (num_blocks, kv_heads, cfg.block_size, spec.head_dim)
+        # MLA (DeepSeek): ONE compressed latent row per token — pool is
+        # [nblocks, 1, BS, kv_lora_rank + qk_rope] and there is no V pool
+        # (v_caches stay as zero-size placeholders so the layer-forward
+        # signature is uniform)
+        self.mla = bool(spec.kv_lora_rank)
+        if self.mla:
+            kv_heads = 1
+            self.kv_heads = 1
+            lat = spec.kv_lora_rank + spec.qk_rope_head_dim
+            shape = (num_blocks, 1, cfg.block_size, lat)
+        else:
+            shape = (num_blocks, kv_heads, cfg.block_size, spec.head_dim)
         if cfg.kv_cache_dtype == "fp8":
             dtype = torch.float8_e4m3fn
         else:
@@ -152,9 +163,15 @@ class KVCache:
         self.k_caches = [
             torch.zeros(shape, dtype=dtype, device=device) for _ in range(self.num_layers)
         ]
-        self.v_caches = [
-            torch.zeros(shape, dtype=dtype, device=device) for _ in range(self.num_layers)
-        ]
+        if self.mla:
+            self.v_caches = [
+                torch.zeros(0, dtype=dtype, device=device)
+                for _ in range(self.num_layers)
+            ]
+        else:
+            self.v_caches = [
+                torch.zeros(shape, dtype=dtype, device=device) for _ in range(self.num_layers)
+            ]
         # last block reserved as the graph-padding scratch block (see
         # engine/graph_runner.py): padded rows write/read there, never live KV
         self.pad_block = num_blocks - 1
@@ -171,6 +188,8 @@ class KVCache:
         if host_blocks is None:
             host_blocks = self.compute_host_blocks(cfg, self.num_layers,
                                                    kv_heads)
+        if self.mla:
+            host_blocks = 0  # MLA latent offload tier is an r3 item
         self.host_blocks = host_blocks
         if host_blocks > 0:
             self.host_pool = torch.zeros(
@@ -245,9 +264,14 @@ class KVCache:
         nl = num_layers if num_layers is not None else spec.num_layers
         kv_heads = max(1, spec.num_kv_heads // cfg.tp_size)
         esize = 1 if cfg.kv_cache_dtype == "fp8" else 2
-        per_block = (
-            2 * nl * kv_heads * cfg.block_size * spec.head_dim * esize
-        )
+        if spec.kv_lora_rank:
+            per_block = (nl * cfg.block_size
+                         * (spec.kv_lora_rank + spec.qk_rope_head_dim)
+                         * esize)
+        else:
+            per_block = (
+                2 * nl * kv_heads * cfg.block_size * spec.head_dim * esize
+            )
         return max(1, int(free_bytes * cfg.gpu_memory_utilization) // per_block)
 
     def blocks_needed(self, num_tokens: int) -> int:

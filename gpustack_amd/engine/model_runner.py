@@ -213,6 +213,17 @@ class ModelRunner:
         self.comm = comm or Communicator()
         self.cp_prefills = 0
         self.cp_suffixes = 0
+        if cfg.spec.kv_lora_rank:
+            # MLA (DeepSeek): latent attention serves on the CPU oracle
+            # path today; the CDNA4 absorbed-attention kernels are the r3
+            # item — refuse loudly rather than crash mid-capture
+            if torch.device(cfg.device).type == "cuda":
+                raise NotImplementedError(
+                    "MLA (DeepSeek) GPU serving requires the r3 CDNA4 "
+                    "absorbed-attention kernels; CPU serving is available")
+            if cfg.kv_cache_dtype == "fp8":
+                raise ValueError("fp8 KV is not supported with MLA latent "
+                                 "caches yet")
         self.device = torch.device(cfg.device)
         import os
 
@@ -412,6 +423,8 @@ class ModelRunner:
         accelerated."""
         if self.comm.cp_size == 1:
             return False
+        if self.cfg.spec.kv_lora_rank:
+            return False  # MLA latent gather is an r3 item; replicated CP
         if self.batch_uses_lora(batch):
             return False
         if batch.is_prefill:

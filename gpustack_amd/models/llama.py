@@ -207,7 +207,186 @@ class Attention(nn.Module):
         return self.comm.all_reduce(o)
 
 
+class MLAAttention(nn.Module):
+    """Multi-head Latent Attention (DeepSeek V2/V3/R1 — reference serves
+    these via vLLM; SURVEY.md §2.10 model families). MI355X-first design:
+    the PAGED CACHE stores only the compressed latent per token
+    ([kv_lora_rank + qk_rope_head_dim] = 576 f.ex. — ~57x smaller than
+    expanded MHA KV), and attention runs in the ABSORBED formulation:
+
+        score(t, l) = (q_nope W_uk) · c_l  +  q_rope · k_rope_l
+        out(t)      = (softmax · C) W_uv
+
+    which is algebraically exact vs HF's expand-then-attend and never
+    materializes per-head K/V. The latent projections (kv_a) are
+    REPLICATED across TP ranks (each rank writes the same 576-wide cache
+    — deterministic by construction) while q_b / kv_b / o shard by head.
+    CPU path is the serving oracle (HF-logits-exact,
+    tests/test_deepseek.py); the CDNA4 absorbed-decode kernel is the r3
+    item and this module fails loudly on CUDA until it lands.
+    """
+
+    def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator,
+                 dtype, layer_idx: int = 0):
+        super().__init__()
+        self.spec = spec
+        self.comm = comm
+        self.layer_idx = layer_idx
+        h = spec.hidden_size
+        assert spec.num_heads % tp_size == 0
+        self.nh = spec.num_heads // tp_size
+        self.dn = spec.qk_nope_head_dim
+        self.dr = spec.qk_rope_head_dim
+        self.dq = self.dn + self.dr
+        self.dv = spec.v_head_dim
+        self.r = spec.kv_lora_rank
+        mk = lambda *shape: nn.Parameter(torch.empty(*shape, dtype=dtype),
+                                         requires_grad=False)
+        if spec.q_lora_rank:
+            self.q_a_w = mk(spec.q_lora_rank, h)
+            self.q_a_norm = mk(spec.q_lora_rank)
+            self.q_b_w = mk(self.nh * self.dq, spec.q_lora_rank)
+            self.q_w = None
+        else:
+            self.q_w = mk(self.nh * self.dq, h)
+            self.q_a_w = self.q_a_norm = self.q_b_w = None
+        self.kv_a_w = mk(self.r + self.dr, h)       # replicated
+        self.kv_a_norm = mk(self.r)
+        self.kv_b_w = mk(self.nh * (self.dn + self.dv), self.r)
+        self.o_w = mk(h, self.nh * self.dv)
+        # softmax scale: qk_head_dim^-0.5, yarn-mscale-adjusted exactly as
+        # HF yarn_apply_mscale (modeling_deepseek_v3.py)
+        scale = self.dq ** -0.5
+        rs = spec.rope_scaling or {}
+        if rs.get("rope_type", rs.get("type", "default")) != "default":
+            msd = rs.get("mscale_all_dim", 0)
+            if msd:
+                import math as _math
+
+                f = rs["factor"]
+                m = 1.0 if f <= 1 else 0.1 * msd * _math.log(f) + 1.0
+                scale = scale * m * m
+        self.scale = scale
+
+    def _rope(self, x, cs):
+        """x [T, ..., dr], cs [T, dr] = (cos_half, sin_half). DeepSeek
+        checkpoints store the rotary dims INTERLEAVED (pairs (x0,x1));
+        HF's interleave rope emits the rotated evens then odds — output
+        layout is consistent between q and k so scores match."""
+        half = self.dr // 2
+        shape = [x.shape[0]] + [1] * (x.dim() - 2) + [half]
+        cos = cs[:, :half].view(shape)
+        sin = cs[:, half:].view(shape)
+        if self.spec.rope_interleave:
+            x1, x2 = x[..., 0::2], x[..., 1::2]
+            return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], -1)
+        x1, x2 = x[..., :half], x[..., half:]
+        return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], -1)
+
+    def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
+        if x.is_cuda:
+            raise NotImplementedError(
+                "MLA (DeepSeek) CDNA4 kernels land in r3 — the CPU path is "
+                "the HF-exact oracle those kernels verify against")
+        T = x.shape[0]
+        spec = self.spec
+        if self.q_w is not None:
+            q = F.linear(x, self.q_w)
+        else:
+            qa = F.linear(x, self.q_a_w)
+            qn = torch.empty_like(qa)
+            ops.rms_norm(qn, qa, self.q_a_norm, spec.rms_norm_eps)
+            q = F.linear(qn, self.q_b_w)
+        q = q.view(T, self.nh, self.dq)
+        cs = cos_sin[meta.positions]
+        q_nope, q_rot = q[..., :self.dn], q[..., self.dn:]
+        q_rot = self._rope(q_rot, cs)
+        lat = F.linear(x, self.kv_a_w)              # [T, r + dr]
+        c = torch.empty(T, self.r, dtype=x.dtype, device=x.device)
+        ops.rms_norm(c, lat[:, :self.r].contiguous(), self.kv_a_norm,
+                     spec.rms_norm_eps)
+        k_rot = self._rope(lat[:, self.r:], cs)
+        lat_rows = torch.cat([c, k_rot], dim=-1)    # [T, r + dr]
+        # cache write: one latent row per token ([nblocks, 1, BS, r+dr])
+        slots = meta.slot_mapping
+        valid = slots >= 0
+        flat = k_cache.view(-1, self.r + self.dr)
+        if bool(valid.all()):
+            flat[slots] = lat_rows.to(flat.dtype)
+        elif bool(valid.any()):
+            flat[slots[valid]] = lat_rows[valid].to(flat.dtype)
+        # absorbed projections
+        kvb = self.kv_b_w.view(self.nh, self.dn + self.dv, self.r)
+        uk = kvb[:, :self.dn]                       # [nh, dn, r]
+        uv = kvb[:, self.dn:]                       # [nh, dv, r]
+        q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
+        out = torch.empty(T, self.nh, self.dv, dtype=x.dtype, device=x.device)
+        for rows, ctx, qpos in self._segments(meta, lat_rows, k_cache):
+            C = ctx.float()                         # [L, r+dr]
+            ql = q_lat[rows]                        # [n, nh, r]
+            qr = q_rot[rows].float()                # [n, nh, dr]
+            scores = (torch.einsum("nhr,lr->nhl", ql, C[:, :self.r])
+                      + torch.einsum("nhd,ld->nhl", qr, C[:, self.r:]))
+            scores = scores * self.scale
+            L = C.shape[0]
+            kvpos = torch.arange(L, device=x.device)
+            mask = kvpos.view(1, 1, L) <= qpos.view(-1, 1, 1)
+            scores = scores.masked_fill(~mask, float("-inf"))
+            probs = torch.softmax(scores, dim=-1)
+            ctx_lat = torch.einsum("nhl,lr->nhr", probs, C[:, :self.r])
+            o = torch.einsum("nhr,hdr->nhd", ctx_lat, uv.float())
+            out[rows] = o.to(x.dtype)
+        o = F.linear(out.reshape(T, self.nh * self.dv), self.o_w)
+        return self.comm.all_reduce(o)
+
+    def _segments(self, meta: ForwardMeta, lat_rows, k_cache):
+        """Yield (row_indices, latent_context [L, r+dr], qpos [n]) per
+        sequence for the three batch shapes. Positions are absolute; the
+        latent context is position-ordered from 0."""
+        dev = lat_rows.device
+        D = self.r + self.dr
+        flat = k_cache.view(-1, D)
+        bs = self.spec_block_size(k_cache)
+        if meta.is_prefill:
+            off = 0
+            for L in (meta.seq_lens_list or []):
+                rows = torch.arange(off, off + L, device=dev)
+                yield rows, lat_rows[off:off + L], torch.arange(L, device=dev)
+                off += L
+            tp = meta.num_prefill_tokens or lat_rows.shape[0]
+            if tp < lat_rows.shape[0]:  # mixed: decode rows ride along
+                for j in range(lat_rows.shape[0] - tp):
+                    L = int(meta.seq_lens[j])
+                    bt = meta.block_tables[j]
+                    idx = (bt[torch.arange(L, device=dev) // bs].long() * bs
+                           + torch.arange(L, device=dev) % bs)
+                    yield (torch.tensor([tp + j], device=dev), flat[idx],
+                           torch.tensor([L - 1], device=dev))
+        elif meta.suffix_meta is not None:
+            _tiles, starts, hists, news = meta.suffix_meta
+            for i, (st, hist, n) in enumerate(zip(starts, hists, news)):
+                L = hist + n
+                bt = meta.block_tables[i]
+                idx = (bt[torch.arange(L, device=dev) // bs].long() * bs
+                       + torch.arange(L, device=dev) % bs)
+                yield (torch.arange(st, st + n, device=dev), flat[idx],
+                       torch.arange(hist, L, device=dev))
+        else:
+            for j in range(lat_rows.shape[0]):
+                L = int(meta.seq_lens[j])
+                bt = meta.block_tables[j]
+                idx = (bt[torch.arange(L, device=dev) // bs].long() * bs
+                       + torch.arange(L, device=dev) % bs)
+                yield (torch.tensor([j], device=dev), flat[idx],
+                       torch.tensor([L - 1], device=dev))
+
+    @staticmethod
+    def spec_block_size(k_cache):
+        return k_cache.shape[2]
+
+
 class MLP(nn.Module):
+
     def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator, dtype):
         super().__init__()
         self.comm = comm
@@ -541,7 +720,12 @@ class DecoderLayer(nn.Module):
                  dtype, layer_idx: int = 0):
         super().__init__()
         self.spec = spec
-        self.attn = Attention(spec, tp_size, comm, dtype, layer_idx=layer_idx)
+        if spec.kv_lora_rank:
+            self.attn = MLAAttention(spec, tp_size, comm, dtype,
+                                     layer_idx=layer_idx)
+        else:
+            self.attn = Attention(spec, tp_size, comm, dtype,
+                                  layer_idx=layer_idx)
         if spec.num_experts > 0 and layer_idx >= spec.first_k_dense_replace:
             self.mlp = MoEMLP(spec, tp_size, comm, dtype)
         else:
@@ -624,12 +808,21 @@ class LlamaForCausalLM(nn.Module):
             self.lm_head = None
         self.lm_head_pack: W4Pack | None = None  # W4 runtime (qlinear)
         self.offload = None  # CpuOffload streamer (engine/offload.py)
-        cache = ops.build_cos_sin_cache(
-            spec.head_dim,
-            int(spec.head_dim * spec.partial_rotary_factor),
-            cfg.max_model_len,
-            base=spec.rope_theta, scaling=spec.rope_scaling,
-        )
+        if spec.kv_lora_rank:
+            # MLA: rope acts on the qk_rope dims only (the latent's rope
+            # slice and each head's q tail)
+            cache = ops.build_cos_sin_cache(
+                spec.qk_rope_head_dim, spec.qk_rope_head_dim,
+                cfg.max_model_len,
+                base=spec.rope_theta, scaling=spec.rope_scaling,
+            )
+        else:
+            cache = ops.build_cos_sin_cache(
+                spec.head_dim,
+                int(spec.head_dim * spec.partial_rotary_factor),
+                cfg.max_model_len,
+                base=spec.rope_theta, scaling=spec.rope_scaling,
+            )
         self.register_buffer("cos_sin", cache.to(device), persistent=False)
         self.to(device)
 

@@ -66,6 +66,27 @@ def random_init(model, cfg: EngineConfig) -> None:
 
     for local_i, layer in enumerate(model.layers):
         li = off + local_i
+        if spec.kv_lora_rank:
+            _random_init_mla_attn(layer, spec, li, seed, dtype, device, tp,
+                                  rank)
+            if hasattr(layer.mlp, "router_w"):
+                _random_init_moe_layer(layer, spec, li, seed, dtype, device,
+                                       tp, rank)
+            else:
+                gate = _gen((spec.intermediate_size, spec.hidden_size),
+                            f"{li}.gate", seed, dtype, device)
+                up = _gen((spec.intermediate_size, spec.hidden_size),
+                          f"{li}.up", seed, dtype, device)
+                layer.mlp.gate_up_w.copy_(torch.cat([
+                    gate[rank * i_loc:(rank + 1) * i_loc],
+                    up[rank * i_loc:(rank + 1) * i_loc],
+                ]))
+                down = _gen((spec.hidden_size, spec.intermediate_size),
+                            f"{li}.down", seed, dtype, device)
+                layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
+            layer.input_norm.fill_(1.0)
+            layer.post_attn_norm.fill_(1.0)
+            continue
         q_full = _gen((spec.num_heads * d, spec.hidden_size), f"{li}.q", seed, dtype, device)
         k_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.k", seed, dtype, device)
         v_full = _gen((spec.num_kv_heads * d, spec.hidden_size), f"{li}.v", seed, dtype, device)
@@ -110,6 +131,36 @@ def random_init(model, cfg: EngineConfig) -> None:
             layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.fill_(1.0)
         layer.post_attn_norm.fill_(1.0)
+
+
+def _random_init_mla_attn(layer, spec, li, seed, dtype, device, tp, rank):
+    """DeepSeek MLA attention params (models/llama.py MLAAttention):
+    q_b / kv_b / o shard by head, the latent projections replicate."""
+    a = layer.attn
+    nh_loc = spec.num_heads // tp
+    dq = spec.qk_nope_head_dim + spec.qk_rope_head_dim
+    if spec.q_lora_rank:
+        a.q_a_w.copy_(_gen((spec.q_lora_rank, spec.hidden_size),
+                           f"{li}.qa", seed, dtype, device))
+        a.q_a_norm.fill_(1.0)
+        qb = _gen((spec.num_heads * dq, spec.q_lora_rank), f"{li}.qbw",
+                  seed, dtype, device)
+        a.q_b_w.copy_(qb[rank * nh_loc * dq:(rank + 1) * nh_loc * dq])
+    else:
+        qw = _gen((spec.num_heads * dq, spec.hidden_size), f"{li}.qw",
+                  seed, dtype, device)
+        a.q_w.copy_(qw[rank * nh_loc * dq:(rank + 1) * nh_loc * dq])
+    a.kv_a_w.copy_(_gen((spec.kv_lora_rank + spec.qk_rope_head_dim,
+                         spec.hidden_size), f"{li}.kva", seed, dtype, device))
+    a.kv_a_norm.fill_(1.0)
+    dkv = spec.qk_nope_head_dim + spec.v_head_dim
+    kvb = _gen((spec.num_heads * dkv, spec.kv_lora_rank), f"{li}.kvb",
+               seed, dtype, device)
+    a.kv_b_w.copy_(kvb[rank * nh_loc * dkv:(rank + 1) * nh_loc * dkv])
+    o = _gen((spec.hidden_size, spec.num_heads * spec.v_head_dim),
+             f"{li}.o", seed, dtype, device)
+    dv = spec.v_head_dim
+    a.o_w.copy_(o[:, rank * nh_loc * dv:(rank + 1) * nh_loc * dv])
 
 
 def _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank):
@@ -194,30 +245,54 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
     for local_i, layer in enumerate(model.layers):
         li = off + local_i
         p = f"{pre}layers.{li}."
-        q = row_shard(get(p + "self_attn.q_proj.weight"), hq * d)
-        k = _kv_slice(get(p + "self_attn.k_proj.weight"), rank, tp,
-                      spec.num_kv_heads, d)
-        v = _kv_slice(get(p + "self_attn.v_proj.weight"), rank, tp,
-                      spec.num_kv_heads, d)
-        layer.attn.qkv_w.copy_(torch.cat([q, k, v]))
-        if layer.attn.qkv_b is not None:
-            layer.attn.qkv_b.copy_(torch.cat([
-                row_shard(get(p + "self_attn.q_proj.bias"), hq * d),
-                _kv_slice(get(p + "self_attn.k_proj.bias").unsqueeze(1),
-                          rank, tp, spec.num_kv_heads, d).squeeze(1),
-                _kv_slice(get(p + "self_attn.v_proj.bias").unsqueeze(1),
-                          rank, tp, spec.num_kv_heads, d).squeeze(1),
-            ]))
-        o = get(p + "self_attn.o_proj.weight")
-        layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
-        if layer.attn.o_b is not None                 and p + "self_attn.o_proj.bias" in tensors:
-            layer.attn.o_b.copy_(get(p + "self_attn.o_proj.bias"))
-        if layer.attn.sinks is not None and p + "self_attn.sinks" in tensors:
-            sk = tensors[p + "self_attn.sinks"].float()
-            layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
-        if spec.qk_norm:
-            layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
-            layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
+        if spec.kv_lora_rank:
+            # DeepSeek MLA: latent projections replicate, per-head
+            # projections shard by head (models/llama.py MLAAttention)
+            a = layer.attn
+            nh_loc = spec.num_heads // tp
+            dq = spec.qk_nope_head_dim + spec.qk_rope_head_dim
+            if spec.q_lora_rank:
+                a.q_a_w.copy_(get(p + "self_attn.q_a_proj.weight"))
+                a.q_a_norm.copy_(get(p + "self_attn.q_a_layernorm.weight"))
+                a.q_b_w.copy_(row_shard(
+                    get(p + "self_attn.q_b_proj.weight"), nh_loc * dq))
+            else:
+                a.q_w.copy_(row_shard(
+                    get(p + "self_attn.q_proj.weight"), nh_loc * dq))
+            a.kv_a_w.copy_(get(p + "self_attn.kv_a_proj_with_mqa.weight"))
+            a.kv_a_norm.copy_(get(p + "self_attn.kv_a_layernorm.weight"))
+            dkv = spec.qk_nope_head_dim + spec.v_head_dim
+            a.kv_b_w.copy_(row_shard(
+                get(p + "self_attn.kv_b_proj.weight"), nh_loc * dkv))
+            o = get(p + "self_attn.o_proj.weight")
+            dv = spec.v_head_dim
+            a.o_w.copy_(
+                o[:, rank * nh_loc * dv:(rank + 1) * nh_loc * dv])
+        else:
+            q = row_shard(get(p + "self_attn.q_proj.weight"), hq * d)
+            k = _kv_slice(get(p + "self_attn.k_proj.weight"), rank, tp,
+                          spec.num_kv_heads, d)
+            v = _kv_slice(get(p + "self_attn.v_proj.weight"), rank, tp,
+                          spec.num_kv_heads, d)
+            layer.attn.qkv_w.copy_(torch.cat([q, k, v]))
+            if layer.attn.qkv_b is not None:
+                layer.attn.qkv_b.copy_(torch.cat([
+                    row_shard(get(p + "self_attn.q_proj.bias"), hq * d),
+                    _kv_slice(get(p + "self_attn.k_proj.bias").unsqueeze(1),
+                              rank, tp, spec.num_kv_heads, d).squeeze(1),
+                    _kv_slice(get(p + "self_attn.v_proj.bias").unsqueeze(1),
+                              rank, tp, spec.num_kv_heads, d).squeeze(1),
+                ]))
+            o = get(p + "self_attn.o_proj.weight")
+            layer.attn.o_w.copy_(o[:, rank * hq * d:(rank + 1) * hq * d])
+            if layer.attn.o_b is not None                 and p + "self_attn.o_proj.bias" in tensors:
+                layer.attn.o_b.copy_(get(p + "self_attn.o_proj.bias"))
+            if layer.attn.sinks is not None and p + "self_attn.sinks" in tensors:
+                sk = tensors[p + "self_attn.sinks"].float()
+                layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
+            if spec.qk_norm:
+                layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
+                layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
         if hasattr(layer.mlp, "router_w")                 and spec.architecture.startswith("GptOss"):
             # GPT-OSS: router `mlp.router.{weight,bias}`; experts stored
             # TRANSPOSED ([E, h, 2i] / [E, i, h]) with INTERLEAVED
