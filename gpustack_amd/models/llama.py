@@ -284,10 +284,18 @@ class MLAAttention(nn.Module):
         return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], -1)
 
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
-        if x.is_cuda:
+        import os as _os
+
+        gpu_ok = (x.is_cuda
+                  and _os.environ.get("GPUSTACK_AMD_MLA_KERNEL") == "1"
+                  and not meta.is_prefill and meta.suffix_meta is None
+                  and self.r == 512 and self.dr == 64 and self.nh % 16 == 0)
+        if x.is_cuda and not gpu_ok:
             raise NotImplementedError(
-                "MLA (DeepSeek) CDNA4 kernels land in r3 — the CPU path is "
-                "the HF-exact oracle those kernels verify against")
+                "MLA (DeepSeek) GPU serving: only the absorbed DECODE "
+                "kernel is written (gate GPUSTACK_AMD_MLA_KERNEL=1, "
+                "r3-validated); prefill/suffix CDNA4 paths land in r3 — "
+                "the CPU path is the HF-exact oracle they verify against")
         T = x.shape[0]
         spec = self.spec
         if self.q_w is not None:
@@ -320,6 +328,17 @@ class MLAAttention(nn.Module):
         uk = kvb[:, :self.dn]                       # [nh, dn, r]
         uv = kvb[:, self.dn:]                       # [nh, dv, r]
         q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
+        if x.is_cuda:  # gated decode kernel path (gpu_ok checked above)
+            q_cat = torch.cat([q_lat.to(x.dtype), q_rot.to(x.dtype)],
+                              dim=-1).contiguous()
+            ctx_lat = torch.empty(T, self.nh, self.r, dtype=torch.float32,
+                                  device=x.device)
+            ops.mla_decode(ctx_lat, q_cat, k_cache, meta.block_tables,
+                           meta.seq_lens, self.scale)
+            o = torch.einsum("nhr,hdr->nhd", ctx_lat, uv.float())
+            o = F.linear(o.to(x.dtype).reshape(T, self.nh * self.dv),
+                         self.o_w)
+            return self.comm.all_reduce(o)
         out = torch.empty(T, self.nh, self.dv, dtype=x.dtype, device=x.device)
         for rows, ctx, qpos in self._segments(meta, lat_rows, k_cache):
             C = ctx.float()                         # [L, r+dr]

@@ -151,6 +151,18 @@ def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
                                     window=window)
 
 
+def mla_decode(ctx_out, q_cat, lat_cache, block_tables, seq_lens,
+               scale: float) -> None:
+    """MLA absorbed decode (DeepSeek): ctx_out[N,H,512] f32 = softmax
+    (q_cat · lat) · lat[:, :512] over the paged latent pool. GPU-only —
+    the CPU oracle lives in models/llama.py MLAAttention."""
+    hip = _backend(q_cat)
+    if hip is None:
+        raise NotImplementedError("mla_decode is the CDNA4 kernel entry; "
+                                  "CPU serving uses the torch MLA path")
+    hip.mla_decode(ctx_out, q_cat, lat_cache, block_tables, seq_lens, scale)
+
+
 _PREFILL_BQ = 64
 
 

@@ -18,7 +18,7 @@ CSRC = PKG_DIR / "csrc"
 SO_PATH = PKG_DIR / "_hip_ops.so"
 ARCH = os.environ.get("GPUSTACK_AMD_ARCH", "gfx950")
 
-HIP_SOURCES = ["pointwise.hip", "attn_decode.hip", "attn_prefill.hip", "skinny_gemm.hip", "gemm8.hip", "gemm8_lab.hip", "moe_gemm.hip", "w4_gemm.hip"]
+HIP_SOURCES = ["pointwise.hip", "attn_decode.hip", "attn_prefill.hip", "skinny_gemm.hip", "gemm8.hip", "gemm8_lab.hip", "moe_gemm.hip", "w4_gemm.hip", "mla_decode.hip"]
 CPP_SOURCES = ["bindings.cpp"]
 
 

@@ -11,6 +11,7 @@ void rope_neox_launch(const long*, void*, void*, const float*, int, int, int, in
 void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
 void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, int, hipStream_t);
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
+void mla_decode_launch(float*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, int*, hipStream_t);
 void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, const float*, int, int*, hipStream_t);
 void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, const float*, int, int*, hipStream_t);
 void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, const float*, int, int*, hipStream_t);
@@ -135,6 +136,29 @@ void greedy_sample(at::Tensor out, at::Tensor logits) {
   const int N = logits.size(0), V = logits.size(1);
   greedy_sample_launch(out.data_ptr<long>(), logits.data_ptr(), N, V,
                        cur_stream(logits));
+  HIP_CHECK_LAST();
+}
+
+void mla_decode(at::Tensor ctx_out, at::Tensor q, at::Tensor lat,
+                at::Tensor block_tables, at::Tensor seq_lens, double scale) {
+  TORCH_CHECK(ctx_out.scalar_type() == at::kFloat && ctx_out.is_contiguous(),
+              "ctx_out must be f32 contiguous");
+  check_bf16(q, "q"); check_bf16(lat, "lat");
+  TORCH_CHECK(q.is_contiguous() && lat.is_contiguous());
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt && block_tables.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int N = q.size(0), H = q.size(1), LD = q.size(2);
+  const int BS = lat.size(2);
+  const int R = ctx_out.size(2);
+  TORCH_CHECK(lat.size(3) == LD && ctx_out.size(1) == H);
+  const int max_blocks = block_tables.size(1);
+  int err = 0;
+  mla_decode_launch(ctx_out.data_ptr<float>(), q.data_ptr(), lat.data_ptr(),
+                    block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                    N, H, R, LD - R, BS, max_blocks, (float)scale, &err,
+                    cur_stream(q));
+  TORCH_CHECK(!err, "mla_decode: unsupported config R=", R, " LD=", LD,
+              " BS=", BS, " H=", H);
   HIP_CHECK_LAST();
 }
 
@@ -376,6 +400,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_and_mul", &silu_and_mul, "silu(x[:d])*x[d:]");
   m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged pool");
   m.def("greedy_sample", &greedy_sample, "argmax over vocab");
+  m.def("mla_decode", &mla_decode,
+        "MLA absorbed decode over the paged latent cache (DeepSeek)");
   m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention",
         py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("block_tables"), py::arg("seq_lens"), py::arg("scale"),

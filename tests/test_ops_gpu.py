@@ -667,3 +667,47 @@ def test_oss_fused_moe_clamped_swiglu(bias):
             d = d + b_d[e].float()
         ref[j] = d * flat_w32[j]
     _close(contrib, ref.to(torch.bfloat16), atol=6e-2, rtol=6e-2)
+
+
+mla = pytest.mark.skipif(
+    _os.environ.get("GPUSTACK_AMD_MLA_KERNEL") != "1",
+    reason="set GPUSTACK_AMD_MLA_KERNEL=1 to test the unvalidated MLA "
+           "absorbed-decode kernel")
+
+
+@mla
+@pytest.mark.parametrize("H", [16, 128])
+@pytest.mark.parametrize("lens", [[1], [16], [17, 5, 160, 33], [1200]])
+def test_mla_decode_kernel(H, lens):
+    """mla_decode vs a plain fp32 torch reference of the absorbed
+    formulation over the paged latent pool (R=512, DR=64)."""
+    torch.manual_seed(0)
+    R, DR, BS = 512, 64, 16
+    LD = R + DR
+    N = len(lens)
+    maxb = (max(lens) + BS - 1) // BS
+    nblocks = sum((l + BS - 1) // BS for l in lens) + 2
+    lat = torch.randn(nblocks, 1, BS, LD, dtype=torch.bfloat16,
+                      device="cuda") / 4
+    bt = torch.zeros(N, maxb, dtype=torch.int32, device="cuda")
+    nxt = 0
+    for i, l in enumerate(lens):
+        nb = (l + BS - 1) // BS
+        bt[i, :nb] = torch.arange(nxt, nxt + nb, dtype=torch.int32)
+        nxt += nb
+    q = torch.randn(N, H, LD, dtype=torch.bfloat16, device="cuda") / 4
+    sl = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    scale = 1.0 / (192 ** 0.5)
+    ctx = torch.empty(N, H, R, dtype=torch.float32, device="cuda")
+    ops.mla_decode(ctx, q, lat, bt, sl, scale)
+
+    flat = lat.view(-1, LD).float()
+    for i, L in enumerate(lens):
+        idx = (bt[i][torch.arange(L, device="cuda") // BS].long() * BS
+               + torch.arange(L, device="cuda") % BS)
+        C = flat[idx]                               # [L, LD]
+        scores = (q[i].float() @ C.T) * scale       # [H, L]
+        probs = torch.softmax(scores, dim=-1)
+        want = probs @ C[:, :R]                     # [H, R]
+        assert torch.allclose(ctx[i], want, atol=2e-2, rtol=2e-2), \
+            (ctx[i] - want).abs().max()
