@@ -49,6 +49,30 @@ def model_spec_for(model: Model | dict) -> ModelSpec | None:
             from ..utils.gguf import spec_from_gguf
 
             return spec_from_gguf(ref)
+        import json as _json
+        from pathlib import Path as _Path
+
+        with open(_Path(ref) / "config.json") as f:
+            arch = (_json.load(f).get("architectures") or [""])[0]
+        from ..models.encoder import is_encoder_arch
+
+        if is_encoder_arch(arch):
+            # cross-encoder reranker: synthesize an LLM-shaped spec whose
+            # weight_bytes matches the encoder (placement sizing only —
+            # the engine server builds the real EncoderSpec)
+            from ..models.encoder import EncoderSpec
+
+            e = EncoderSpec.from_dir(ref)
+            return ModelSpec(
+                architecture=e.architecture, vocab_size=e.vocab_size,
+                hidden_size=e.hidden_size,
+                intermediate_size=e.intermediate_size,
+                num_layers=e.num_layers, num_heads=e.num_heads,
+                num_kv_heads=e.num_heads,
+                head_dim=e.hidden_size // e.num_heads,
+                max_position_embeddings=e.max_position_embeddings,
+                tie_word_embeddings=True,
+            )
         return ModelSpec.from_dir(ref)
     except Exception:  # noqa: BLE001
         return None

@@ -290,6 +290,15 @@ def create_model(body: ModelCreate, _: User = Depends(get_current_user)):
         if s.query(Model).filter_by(name=body.name).first():
             raise HTTPException(409, "model name exists")
         m = Model(**body.model_dump())
+        if m.categories == ["llm"]:
+            # auto-categorize from the architecture when the caller left
+            # the default (reference: scheduler model_registry) — rerankers
+            # and embedding checkpoints place and list correctly
+            from ..utils.model_registry import categories_for_model
+
+            cats = categories_for_model(m.source, m.model_ref)
+            if cats:
+                m.categories = cats
         ar_create(s, m)
         return m.to_dict()
 

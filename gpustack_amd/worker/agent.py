@@ -212,12 +212,38 @@ class WorkerAgent:
                     "metadata": meta}
 
         @app.get("/logs/{instance_name}")
-        def logs(instance_name: str, tail: int = 200):
+        def logs(instance_name: str, tail: int = 200, follow: bool = False,
+                 timeout_s: float = 300.0):
+            """Instance log tail; `follow=true` streams appended lines as
+            they land (reference: log_sources follow streaming) until
+            timeout_s or the file disappears."""
             path = Path(self.cfg.data_dir) / "log" / "instances" / f"{instance_name}.log"
             if not path.exists():
                 raise HTTPException(404, "no log for instance")
-            lines = path.read_text(errors="replace").splitlines()
-            return PlainTextResponse("\n".join(lines[-tail:]))
+            if not follow:
+                lines = path.read_text(errors="replace").splitlines()
+                return PlainTextResponse("\n".join(lines[-tail:]))
+
+            def stream():
+                import time as _t
+
+                with open(path, "rb") as f:
+                    # serve the tail first, then poll for appended data
+                    lines = f.read().decode(errors="replace").splitlines(True)
+                    yield "".join(lines[-tail:])
+                    deadline = _t.time() + timeout_s
+                    while _t.time() < deadline:
+                        chunk = f.read()
+                        if chunk:
+                            yield chunk.decode(errors="replace")
+                        elif not path.exists():
+                            return
+                        else:
+                            _t.sleep(0.5)
+
+            from fastapi.responses import StreamingResponse
+
+            return StreamingResponse(stream(), media_type="text/plain")
 
         @app.get("/metrics")
         def metrics():

@@ -248,3 +248,58 @@ def test_encoder_serving_mode(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_model_registry_categories():
+    from gpustack_amd.utils.model_registry import (
+        categories_for_architecture, categories_for_model,
+    )
+
+    assert categories_for_architecture(
+        "XLMRobertaForSequenceClassification") == ["reranker"]
+    assert categories_for_architecture("BertModel") == ["embedding"]
+    assert categories_for_architecture("LlamaForCausalLM") == ["llm"]
+    assert categories_for_architecture(
+        "DeepseekV3ForCausalLM") == ["llm", "moe"]
+    assert categories_for_architecture(
+        "WhisperForConditionalGeneration") == ["speech_to_text"]
+    assert categories_for_model("preset", "deepseek-v3") == ["llm", "moe"]
+    assert categories_for_model("hf", "org/unknown-remote") is None
+
+
+def test_reranker_checkpoint_auto_categorized_and_sized(tmp_path):
+    """A reranker dir: /v2/models create derives categories=[reranker];
+    the scheduler sizing spec resolves (no 'cannot resolve model spec')."""
+    import json as _json
+
+    from gpustack_amd.scheduler.policies import model_spec_for
+
+    (tmp_path / "config.json").write_text(_json.dumps({
+        "architectures": ["XLMRobertaForSequenceClassification"],
+        "vocab_size": 250002, "hidden_size": 1024,
+        "num_hidden_layers": 24, "num_attention_heads": 16,
+        "intermediate_size": 4096, "max_position_embeddings": 8194,
+        "type_vocab_size": 1, "pad_token_id": 1,
+    }))
+    spec = model_spec_for({"source": "local", "model_ref": str(tmp_path)})
+    assert spec is not None and spec.hidden_size == 1024
+    assert 1 << 30 < spec.weight_bytes() < 4 << 30  # ~1.1 GB class model
+
+    import tempfile
+
+    from starlette.testclient import TestClient
+
+    from gpustack_amd.config import Config
+    from gpustack_amd.server.app import create_app
+
+    cfg = Config(data_dir=tempfile.mkdtemp(), bootstrap_password="pw123")
+    app = create_app(cfg, start_background=False)
+    client = TestClient(app)
+    r = client.post("/auth/login", json={"username": "admin",
+                                         "password": "pw123"})
+    client.headers["Authorization"] = f"Bearer {r.json()['token']}"
+    r = client.post("/v2/models", json={
+        "name": "bge-reranker", "source": "local",
+        "model_ref": str(tmp_path)})
+    assert r.status_code == 201, r.text
+    assert r.json()["categories"] == ["reranker"]
