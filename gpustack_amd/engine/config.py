@@ -336,6 +336,19 @@ PRESETS: dict[str, ModelSpec] = {
         num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
         rope_theta=10000.0, eos_token_id=1,
     ),
+    # CPU-test preset: DeepSeek-shaped MLA + MoE (tests/test_deepseek.py,
+    # TP exactness on gloo)
+    "tiny-mla": ModelSpec(
+        architecture="DeepseekV3ForCausalLM", vocab_size=512,
+        hidden_size=128, intermediate_size=256, num_layers=2,
+        num_heads=4, num_kv_heads=4, head_dim=48,
+        max_position_embeddings=256, rope_theta=10000.0, eos_token_id=1,
+        num_experts=8, num_experts_per_tok=2, moe_intermediate_size=64,
+        router_mode="sigmoid_bias", n_shared_experts=1,
+        first_k_dense_replace=1, routed_scaling_factor=1.5, n_group=2,
+        topk_group=1, q_lora_rank=64, kv_lora_rank=96,
+        qk_nope_head_dim=32, qk_rope_head_dim=16, v_head_dim=32,
+    ),
 }
 
 

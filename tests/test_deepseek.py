@@ -248,3 +248,75 @@ def test_deepseek_checkpoint_loader_roundtrip(tmp_path):
     got = _prefill_logits(eng2, prompt)
     assert torch.allclose(got, want, atol=3e-4, rtol=1e-3), \
         (got - want).abs().max()
+
+
+def _mla_tp_rank_main(rank, world, port, out_path):
+    import json
+    import os
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.parallel import init_tp
+
+    prompts = [[3, 1, 4, 1, 5, 9, 2, 6], [11, 22, 33]]
+    comm = init_tp(world, rank, master_port=port, backend="gloo")
+    cfg = EngineConfig(model="tiny-mla", device="cpu", kv_cache_blocks=64,
+                       max_model_len=128, seed=0, dtype="float32",
+                       tp_size=world, tp_rank=rank)
+    eng = LLMEngine(cfg, comm)
+    results = {}
+    if rank == 0:
+        rids = [eng.add_request(p, SamplingParams(max_tokens=6,
+                                                  ignore_eos=True))
+                for p in prompts]
+        results = {r: [] for r in rids}
+    while eng.tp_active():
+        for o in eng.step():
+            if rank == 0 and o.request_id in results:
+                results[o.request_id].append(o.token_id)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump([results[r] for r in rids], f)
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_deepseek_tp2_matches_tp1():
+    """MLA under TP (gloo world 2): latent projections replicate, q_b/
+    kv_b/o shard by head — greedy continuation must be deterministic,
+    run-to-run stable, and equal to the single-rank engine."""
+    import json
+    import multiprocessing as mp
+    import socket
+    import tempfile
+
+    prompts = [[3, 1, 4, 1, 5, 9, 2, 6], [11, 22, 33]]
+
+    def run2():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        out = tempfile.mktemp(suffix=".json")
+        ctx = mp.get_context("spawn")
+        procs = [ctx.Process(target=_mla_tp_rank_main,
+                             args=(r, 2, port, out)) for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+            assert p.exitcode == 0
+        with open(out) as f:
+            return json.load(f)
+
+    a = run2()
+    b = run2()
+    assert a == b and all(len(x) == 6 for x in a)
+    single = LLMEngine(EngineConfig(model="tiny-mla", device="cpu",
+                                    kv_cache_blocks=64, max_model_len=128,
+                                    seed=0, dtype="float32")).generate(
+        prompts, SamplingParams(max_tokens=6, ignore_eos=True))
+    assert a == single
