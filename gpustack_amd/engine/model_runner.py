@@ -534,6 +534,15 @@ class ModelRunner:
             tiles = ops.build_prefill_tiles(pre_lens, dev)
             bt = None
             dec_lens = None
+            if self.cfg.spec.kv_lora_rank and dev.type == "cuda":
+                # MLA gated GPU path: the absorbed kernel reads the latent
+                # through per-seq block tables even for prefill rows
+                maxb_p = max(len(s.block_table) for s in batch.seqs[:n_pre])
+                btp = torch.zeros(n_pre, maxb_p, dtype=torch.int32)
+                for i, sq in enumerate(batch.seqs[:n_pre]):
+                    btp[i, : len(sq.block_table)] = torch.tensor(
+                        sq.block_table, dtype=torch.int32)
+                bt = btp.to(dev)
             if n_dec:
                 maxb = max(len(s.block_table) for s in batch.seqs[n_pre:])
                 btc = torch.zeros(n_dec, maxb, dtype=torch.int32)

@@ -286,17 +286,22 @@ class MLAAttention(nn.Module):
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
         import os as _os
 
+        T = x.shape[0]
+        # mixed prefill batches (decode rows riding along) keep the loud
+        # default: the per-seq tables in meta cover prefill seqs only
+        pure = (not meta.is_prefill
+                or (meta.block_tables is not None
+                    and (meta.num_prefill_tokens or T) == T))
         gpu_ok = (x.is_cuda
                   and _os.environ.get("GPUSTACK_AMD_MLA_KERNEL") == "1"
-                  and not meta.is_prefill and meta.suffix_meta is None
-                  and self.r == 512 and self.dr == 64 and self.nh % 16 == 0)
+                  and self.r == 512 and self.dr == 64 and self.nh % 16 == 0
+                  and pure)
         if x.is_cuda and not gpu_ok:
             raise NotImplementedError(
-                "MLA (DeepSeek) GPU serving: only the absorbed DECODE "
-                "kernel is written (gate GPUSTACK_AMD_MLA_KERNEL=1, "
-                "r3-validated); prefill/suffix CDNA4 paths land in r3 — "
-                "the CPU path is the HF-exact oracle they verify against")
-        T = x.shape[0]
+                "MLA (DeepSeek) GPU serving is gated until the r3 kernel "
+                "validation pass (set GPUSTACK_AMD_MLA_KERNEL=1; requires "
+                "kv_lora_rank 512 / qk_rope 64 / local heads %16==0) — "
+                "the CPU path is the HF-exact oracle it verifies against")
         spec = self.spec
         if self.q_w is not None:
             q = F.linear(x, self.q_w)
@@ -328,13 +333,17 @@ class MLAAttention(nn.Module):
         uk = kvb[:, :self.dn]                       # [nh, dn, r]
         uv = kvb[:, self.dn:]                       # [nh, dv, r]
         q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
-        if x.is_cuda:  # gated decode kernel path (gpu_ok checked above)
+        if x.is_cuda:  # gated kernel path (gpu_ok checked above)
+            # Every batch shape runs the absorbed kernel; prefill/suffix
+            # rows run ROW-WISE (len = abs position + 1 through the seq's
+            # block table) — correct, with O(T*L) latent re-reads; the
+            # tiled absorbed-prefill kernel is the r3 follow-up.
+            bt, lens = self._gpu_row_tables(meta, T, x.device)
             q_cat = torch.cat([q_lat.to(x.dtype), q_rot.to(x.dtype)],
                               dim=-1).contiguous()
             ctx_lat = torch.empty(T, self.nh, self.r, dtype=torch.float32,
                                   device=x.device)
-            ops.mla_decode(ctx_lat, q_cat, k_cache, meta.block_tables,
-                           meta.seq_lens, self.scale)
+            ops.mla_decode(ctx_lat, q_cat, k_cache, bt, lens, self.scale)
             o = torch.einsum("nhr,hdr->nhd", ctx_lat, uv.float())
             o = F.linear(o.to(x.dtype).reshape(T, self.nh * self.dv),
                          self.o_w)
@@ -357,6 +366,29 @@ class MLAAttention(nn.Module):
             out[rows] = o.to(x.dtype)
         o = F.linear(out.reshape(T, self.nh * self.dv), self.o_w)
         return self.comm.all_reduce(o)
+
+    def _gpu_row_tables(self, meta: ForwardMeta, T: int, dev):
+        """Per-ROW (block_table, context_len) tensors for the absorbed
+        kernel across the three batch shapes."""
+        if meta.is_prefill:
+            # per-seq tables in meta.block_tables (rows [0, n_pre));
+            # mixed decode tails reuse their own decode tables — the
+            # prefill _meta packs per-seq tables for ALL seqs here
+            reps, lens = [], []
+            for i, L in enumerate(meta.seq_lens_list or []):
+                reps.extend([i] * L)
+                lens.extend(range(1, L + 1))
+            bt = meta.block_tables[torch.tensor(reps, device=dev)]
+            return bt, torch.tensor(lens, dtype=torch.int32, device=dev)
+        if meta.suffix_meta is not None:
+            _tiles, starts, hists, news = meta.suffix_meta
+            reps, lens = [], []
+            for i, (h, n) in enumerate(zip(hists, news)):
+                reps.extend([i] * n)
+                lens.extend(range(h + 1, h + n + 1))
+            bt = meta.block_tables[torch.tensor(reps, device=dev)]
+            return bt, torch.tensor(lens, dtype=torch.int32, device=dev)
+        return meta.block_tables, meta.seq_lens
 
     def _segments(self, meta: ForwardMeta, lat_rows, k_cache):
         """Yield (row_indices, latent_context [L, r+dr], qpos [n]) per
