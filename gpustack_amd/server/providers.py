@@ -103,7 +103,71 @@ class CommandProvider:
             raise RuntimeError(f"delete_command failed: {out.stderr.strip()}")
 
 
-PROVIDERS = {"mock": MockProvider, "command": CommandProvider}
+class K8sProvider:
+    """Worker pods in a Kubernetes cluster through the first-party kube
+    client (utils/k8s_client.py — the same machinery the GPU-instance
+    operator analog uses). Each pool replica is one ROCm worker pod
+    (kfd/dri devices + `amd.com/gpu` claim) whose command runs the
+    bootstrap script, registering against this server.
+
+    provider_config: api_server, token, namespace, verify, image,
+    gpus_per_worker (default 8)."""
+
+    def __init__(self, config: dict, client=None):
+        self.config = config or {}
+        if client is None:
+            from ..utils.k8s_client import KubeClient
+
+            client = KubeClient(api_server=self.config.get("api_server"),
+                                token=self.config.get("token"),
+                                namespace=self.config.get("namespace",
+                                                          "gpustack"),
+                                verify=self.config.get("verify", True))
+        self.kube = client
+
+    def _manifest(self, name: str, instance_type: str, user_data: str) -> dict:
+        gpus = int(self.config.get("gpus_per_worker", 8))
+        return {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": name, "namespace": self.kube.namespace,
+                         "labels": {"app": "gpustack-amd-pool-worker",
+                                    "gpustack.amd/instance-type":
+                                        instance_type}},
+            "spec": {
+                "restartPolicy": "Always",
+                "containers": [{
+                    "name": "worker",
+                    "image": self.config.get("image", "gpustack-amd:latest"),
+                    "command": ["/bin/sh", "-c", user_data],
+                    "securityContext": {"capabilities": {
+                        "add": ["SYS_PTRACE"]}},
+                    "resources": {"limits": {"amd.com/gpu": str(gpus)}},
+                    "volumeMounts": [
+                        {"name": "kfd", "mountPath": "/dev/kfd"},
+                        {"name": "dri", "mountPath": "/dev/dri"},
+                    ],
+                }],
+                "volumes": [
+                    {"name": "kfd", "hostPath": {"path": "/dev/kfd"}},
+                    {"name": "dri", "hostPath": {"path": "/dev/dri"}},
+                ],
+            },
+        }
+
+    def create(self, name: str, instance_type: str, user_data: str) -> str:
+        pod = self._manifest(name, instance_type, user_data)
+        self.kube.create_pod(pod)
+        return name
+
+    def delete(self, instance_id: str) -> None:
+        self.kube.delete_pod(instance_id)
+
+    def list(self) -> list[str]:  # parity with MockProvider surface
+        return []
+
+
+PROVIDERS = {"mock": MockProvider, "command": CommandProvider,
+             "k8s": K8sProvider}
 
 
 def get_provider(name: str, config: dict | None = None):

@@ -113,3 +113,47 @@ def test_unknown_provider_rejected(server):
     r = client.post("/v2/worker_pools", json={
         "name": "bad", "provider": "droplets", "replicas": 1})
     assert r.status_code == 400
+
+
+def test_k8s_pool_provider_creates_worker_pods():
+    """K8sProvider drives worker pods through the kube client: the pod
+    carries ROCm devices, the amd.com/gpu claim and the bootstrap script
+    as its command."""
+    import json
+
+    import httpx
+
+    from gpustack_amd.server.providers import K8sProvider, bootstrap_script
+    from gpustack_amd.utils.k8s_client import KubeClient
+
+    pods = {}
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        parts = request.url.path.strip("/").split("/")
+        name = parts[5] if len(parts) > 5 else None
+        if request.method == "POST":
+            obj = json.loads(request.content)
+            pods[obj["metadata"]["name"]] = obj
+            return httpx.Response(201, json=obj)
+        if request.method == "DELETE":
+            pods.pop(name, None)
+            return httpx.Response(200, json={})
+        return httpx.Response(404, json={})
+
+    kube = KubeClient(api_server="https://kube.test", token="t",
+                      namespace="gpustack",
+                      transport=httpx.MockTransport(handler))
+    prov = K8sProvider({"image": "gpustack-amd:v1", "gpus_per_worker": 4},
+                       client=kube)
+    ud = bootstrap_script("http://server", "tok_x", {"pool": "a"})
+    iid = prov.create("pool-a-0", "mi355x-4gpu", ud)
+    assert iid == "pool-a-0" and "pool-a-0" in pods
+    c = pods["pool-a-0"]["spec"]["containers"][0]
+    assert c["image"] == "gpustack-amd:v1"
+    assert c["resources"]["limits"]["amd.com/gpu"] == "4"
+    assert "--server-url http://server" in c["command"][2]
+    paths = {v["hostPath"]["path"]
+             for v in pods["pool-a-0"]["spec"]["volumes"]}
+    assert paths == {"/dev/kfd", "/dev/dri"}
+    prov.delete("pool-a-0")
+    assert not pods
