@@ -320,3 +320,20 @@ def test_deepseek_tp2_matches_tp1():
                                     seed=0, dtype="float32")).generate(
         prompts, SamplingParams(max_tokens=6, ignore_eos=True))
     assert a == single
+
+
+def test_deepseek_prefix_cache_hit_matches_cold():
+    """Automatic prefix caching over the LATENT cache: a shared-prefix
+    second prompt recomputes only the suffix (the MLA suffix/absorbed
+    path) and must produce the cold-start continuation exactly."""
+    p = SamplingParams(max_tokens=6, ignore_eos=True)
+    shared = list(range(2, 40))
+    prompts = [shared + [41, 42], shared + [43, 44, 45]]
+    cold = _engine().generate(prompts, p)
+
+    eng = _engine(enable_prefix_caching=True)
+    warm1 = eng.generate([prompts[0]], p)[0]
+    warm2 = eng.generate([prompts[1]], p)[0]  # hits prompts[0]'s prefix
+    assert [warm1, warm2] == cold
+    alloc = eng.scheduler.kv.allocator
+    assert alloc.hits > 0  # the second prompt actually reused blocks

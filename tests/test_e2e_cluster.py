@@ -397,3 +397,77 @@ def test_deploy_moe_model(cluster):
     })
     assert r.status_code == 200, r.text
     client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-moe-e2e'][0]['id']}")
+
+
+@pytest.mark.timeout(240)
+def test_deploy_reranker_model(cluster, tmp_path_factory):
+    """Cross-encoder reranker checkpoint end-to-end: the model is
+    auto-categorized 'reranker' at create, the scheduler sizes it from
+    the encoder config, serve_manager launches engine_server which
+    detects the sequence-classification architecture and serves the
+    pair-scoring /v1/rerank through the gateway."""
+    import json as _json
+
+    import torch
+    from safetensors.torch import save_file
+
+    from gpustack_amd.models.encoder import CrossEncoderModel, EncoderSpec
+
+    client, agent = cluster
+    d = tmp_path_factory.mktemp("reranker")
+    spec = EncoderSpec(
+        architecture="BertForSequenceClassification", vocab_size=256,
+        hidden_size=64, intermediate_size=128, num_layers=2, num_heads=4,
+        max_position_embeddings=96, type_vocab_size=2, num_labels=1,
+        pad_token_id=0)
+    (d / "config.json").write_text(_json.dumps({
+        "architectures": [spec.architecture],
+        "vocab_size": spec.vocab_size, "hidden_size": spec.hidden_size,
+        "num_hidden_layers": spec.num_layers,
+        "num_attention_heads": spec.num_heads,
+        "intermediate_size": spec.intermediate_size,
+        "max_position_embeddings": spec.max_position_embeddings,
+        "type_vocab_size": spec.type_vocab_size,
+        "id2label": {"0": "LABEL_0"}, "pad_token_id": 0,
+    }))
+    from transformers import BertConfig, BertForSequenceClassification
+
+    torch.manual_seed(11)
+    hf = BertForSequenceClassification(BertConfig(
+        vocab_size=spec.vocab_size, hidden_size=spec.hidden_size,
+        num_hidden_layers=spec.num_layers,
+        num_attention_heads=spec.num_heads,
+        intermediate_size=spec.intermediate_size,
+        max_position_embeddings=spec.max_position_embeddings,
+        type_vocab_size=spec.type_vocab_size, num_labels=1,
+        pad_token_id=0)).eval()
+    save_file({k: v.contiguous() for k, v in hf.state_dict().items()
+               if "position_ids" not in k}, str(d / "model.safetensors"))
+
+    r = client.post("/v2/models", json={
+        "name": "bge-e2e", "source": "local", "model_ref": str(d),
+        "replicas": 1,
+    })
+    assert r.status_code == 201, r.text
+    assert r.json()["categories"] == ["reranker"]
+    state = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "bge-e2e"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+    r = client.post("/v1/rerank", json={
+        "model": "bge-e2e", "query": "gpu kernels",
+        "documents": ["mfma tiles", "pasta recipe", "hip streams"],
+        "top_n": 2,
+    })
+    assert r.status_code == 200, r.text
+    res = r.json()["results"]
+    assert len(res) == 2
+    assert res[0]["relevance_score"] >= res[1]["relevance_score"]
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'bge-e2e'][0]['id']}")

@@ -15,16 +15,20 @@ from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
 
 
 @pytest.mark.timeout(600)
-@pytest.mark.parametrize("seed", [0, 1, 2])
+@pytest.mark.parametrize("seed", [0, 1, 2, 3])
 def test_scheduler_fuzz(seed):
     rng = random.Random(seed)
     # r2: odd seeds also run packed-W4 weights + CPU weight offload so
     # the fuzz exercises qlinear dispatch and the layer streamer under
-    # the same allocator pressure
+    # the same allocator pressure; seed 3 runs the MLA latent-cache
+    # family (DeepSeek) through the same chunk/prefix/abort stress
     extra = ({"quantize_runtime": "w4", "cpu_offload_gb": 0.001}
              if seed % 2 else {})
+    model = "tiny"
+    if seed == 3:
+        model, extra = "tiny-mla", {}
     eng = LLMEngine(EngineConfig(
-        model="tiny", device="cpu", max_model_len=192,
+        model=model, device="cpu", max_model_len=192,
         kv_cache_blocks=20,                # tiny pool: constant pressure
         max_num_seqs=6, max_prefill_tokens=48,
         enable_chunked_prefill=True, enable_prefix_caching=True,
