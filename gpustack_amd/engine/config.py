@@ -70,6 +70,17 @@ class ModelSpec:
     qk_rope_head_dim: int = 0
     v_head_dim: int = 0
     rope_interleave: bool = True         # deepseek weights: paired dims
+    # Gemma-2 (Gemma2ForCausalLM): sandwich layer norms (post-attn and
+    # post-ffn norms applied BEFORE the residual adds), sqrt(h)-scaled
+    # embeddings, tanh logit softcapping in attention and on the final
+    # logits, custom attention scale, GeGLU MLP; sliding window
+    # alternates via layer_types (even layers) like GPT-OSS
+    sandwich_norms: bool = False
+    embed_scale: float = 0.0             # 0 = no scaling
+    attn_logit_softcap: float = 0.0      # 0 = off
+    final_logit_softcap: float = 0.0
+    attn_scale: float = 0.0              # 0 = 1/sqrt(head_dim)
+    mlp_act: str = "silu"                # "silu" | "gelu_tanh"
 
     @property
     def gqa_ratio(self) -> int:
@@ -166,7 +177,8 @@ class ModelSpec:
                                    .get("partial_rotary_factor") or 1.0),
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
-            if arch.startswith("GptOss") else 0,
+            if (arch.startswith("GptOss") or arch.startswith("Gemma2"))
+            else 0,
             layer_types=tuple(cfg["layer_types"])
             if cfg.get("layer_types") else None,
             moe_act="clamped_swiglu" if arch.startswith("GptOss") else "silu",
@@ -182,6 +194,17 @@ class ModelSpec:
             qk_rope_head_dim=cfg.get("qk_rope_head_dim", 0) or 0,
             v_head_dim=cfg.get("v_head_dim", 0) or 0,
             rope_interleave=bool(cfg.get("rope_interleave", True)),
+            sandwich_norms=arch.startswith("Gemma2"),
+            embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
+                         if arch.startswith("Gemma") else 0.0),
+            attn_logit_softcap=(cfg.get("attn_logit_softcapping") or 0.0)
+            if arch.startswith("Gemma2") else 0.0,
+            final_logit_softcap=(cfg.get("final_logit_softcapping") or 0.0)
+            if arch.startswith("Gemma2") else 0.0,
+            attn_scale=((cfg.get("query_pre_attn_scalar") or 0) ** -0.5
+                        if (arch.startswith("Gemma2")
+                            and cfg.get("query_pre_attn_scalar")) else 0.0),
+            mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),
         )
 
     @classmethod
@@ -335,6 +358,19 @@ PRESETS: dict[str, ModelSpec] = {
         vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,
         num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
         rope_theta=10000.0, eos_token_id=1,
+    ),
+    # Gemma-2: sandwich norms, GeGLU, softcapping, alternating SWA,
+    # scaled tied embeddings (CPU-oracle family; GPU kernels share the
+    # GPT-OSS r3 window work — head_dim 256 needs a D-template too)
+    "gemma-2-9b": ModelSpec(
+        architecture="Gemma2ForCausalLM", vocab_size=256000,
+        hidden_size=3584, intermediate_size=14336, num_layers=42,
+        num_heads=16, num_kv_heads=8, head_dim=256, rope_theta=10000.0,
+        max_position_embeddings=8192, tie_word_embeddings=True,
+        rms_norm_eps=1e-6, eos_token_id=1, sliding_window=4096,
+        sandwich_norms=True, embed_scale=3584 ** 0.5,
+        attn_logit_softcap=50.0, final_logit_softcap=30.0,
+        attn_scale=224 ** -0.5, mlp_act="gelu_tanh",
     ),
     # CPU-test preset: DeepSeek-shaped MLA + MoE (tests/test_deepseek.py,
     # TP exactness on gloo)

@@ -112,7 +112,8 @@ class Attention(nn.Module):
         self.d = spec.head_dim
         # GLM-style partial rotary: only the first rot_dim dims rotate
         self.rot_dim = int(spec.head_dim * spec.partial_rotary_factor)
-        self.scale = self.d ** -0.5
+        self.scale = spec.attn_scale or self.d ** -0.5
+        self.softcap = spec.attn_logit_softcap
         h = spec.hidden_size
         qkv_out = (self.hq + 2 * self.hkv) * self.d
         self.qkv_w = nn.Parameter(torch.empty(qkv_out, h, dtype=dtype), requires_grad=False)
@@ -180,6 +181,8 @@ class Attention(nn.Module):
         sw = {}
         if self.sinks is not None or self.window:
             sw = {"sinks": self.sinks, "window": self.window}
+        if self.softcap:
+            sw["softcap"] = self.softcap
         if meta.is_prefill:
             tp = meta.num_prefill_tokens or T
             ops.varlen_prefill_attn(
@@ -450,13 +453,22 @@ class MLP(nn.Module):
         self.layer_idx = 0  # set by LlamaForCausalLM
         self._gu_projs = [("gate_proj", 0, self.i), ("up_proj", self.i, self.i)]
         self._down_projs = [("down_proj", 0, h)]
+        self.act = spec.mlp_act
 
     def forward(self, x, meta: ForwardMeta | None = None):
         gu = qlinear(x, self.gate_up_w, self.gate_up_pack)
         if meta is not None and meta.lora is not None:
             meta.lora.apply(self.layer_idx, x, gu, self._gu_projs)
-        act = torch.empty(x.shape[0], self.i, dtype=x.dtype, device=x.device)
-        ops.silu_and_mul(act, gu)
+        if self.act == "gelu_tanh":
+            # Gemma GeGLU (HF gelu_pytorch_tanh); torch pointwise path —
+            # the fused CDNA4 geglu variant is r3 (same slot as the
+            # GPT-OSS clamped-swiglu kernel work)
+            act = (F.gelu(gu[:, :self.i], approximate="tanh")
+                   * gu[:, self.i:])
+        else:
+            act = torch.empty(x.shape[0], self.i, dtype=x.dtype,
+                              device=x.device)
+            ops.silu_and_mul(act, gu)
         down = (qlinear(act, self.down_w, self.down_pack)
                 if self.down_pack is not None
                 else ops.linear_auto(act, self.down_w))
@@ -786,9 +798,21 @@ class DecoderLayer(nn.Module):
         h = spec.hidden_size
         self.input_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
         self.post_attn_norm = nn.Parameter(torch.empty(h, dtype=dtype), requires_grad=False)
+        if spec.sandwich_norms:
+            # Gemma-2: post-attn/post-ffn norms run BEFORE the residual
+            # adds, plus a pre-ffn norm — four norms per layer
+            self.pre_ff_norm = nn.Parameter(torch.empty(h, dtype=dtype),
+                                            requires_grad=False)
+            self.post_ff_norm = nn.Parameter(torch.empty(h, dtype=dtype),
+                                             requires_grad=False)
+        else:
+            self.pre_ff_norm = self.post_ff_norm = None
 
     def forward(self, x, residual, meta, cos_sin, k_cache, v_cache):
         eps = self.spec.rms_norm_eps
+        if self.spec.sandwich_norms:
+            return self._forward_sandwich(x, meta, cos_sin, k_cache,
+                                          v_cache, eps)
         if residual is None:
             residual = x
             h = torch.empty_like(x)
@@ -800,6 +824,21 @@ class DecoderLayer(nn.Module):
         ops.fused_add_rms_norm(a, residual, self.post_attn_norm, eps)
         m = self.mlp(a, meta)
         return m, residual
+
+    def _forward_sandwich(self, x, meta, cos_sin, k_cache, v_cache, eps):
+        """Gemma-2 layer flow (residual carried explicitly — the layer
+        returns the true hidden stream, residual sentinel None):
+            x = x + norm_post_attn(attn(norm_in(x)))
+            x = x + norm_post_ff(mlp(norm_pre_ff(x)))"""
+        h = torch.empty_like(x)
+        ops.rms_norm(h, x, self.input_norm, eps)
+        a = self.attn(h, meta, cos_sin, k_cache, v_cache)
+        ops.rms_norm(a, a, self.post_attn_norm, eps)
+        x = x + a
+        ops.rms_norm(h, x, self.pre_ff_norm, eps)
+        m = self.mlp(h, meta)
+        ops.rms_norm(m, m, self.post_ff_norm, eps)
+        return x + m, None
 
 
 class LlamaForCausalLM(nn.Module):
@@ -890,6 +929,11 @@ class LlamaForCausalLM(nn.Module):
                 self.dtype, self.device)
         else:
             x = F.embedding(token_ids, self.embed)
+            if self.spec.embed_scale:
+                # Gemma: sqrt(hidden) embedding scale, cast like HF
+                # (normalizer materialized in the embed dtype)
+                x = x * torch.tensor(self.spec.embed_scale,
+                                     dtype=x.dtype, device=x.device)
         residual = None
         off = self.offload
         if off is not None:
@@ -899,14 +943,24 @@ class LlamaForCausalLM(nn.Module):
                 off.bind(i)
             x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
         if self.comm.pp_size > 1 and not self.comm.is_last_stage:
-            s = (x.float() + residual.float()).to(self.dtype)
-            self.comm.send_hidden(s)
+            if residual is None:  # sandwich layers carry the true stream
+                self.comm.send_hidden(x.contiguous())
+            else:
+                s = (x.float() + residual.float()).to(self.dtype)
+                self.comm.send_hidden(s)
             return None
-        ops.fused_add_rms_norm(x, residual, self.final_norm, self.spec.rms_norm_eps)
+        if residual is None:
+            ops.rms_norm(x, x, self.final_norm, self.spec.rms_norm_eps)
+        else:
+            ops.fused_add_rms_norm(x, residual, self.final_norm,
+                                   self.spec.rms_norm_eps)
         if return_hidden:
             return x  # all rows, post final norm (embedding serving)
         hidden = x[meta.logits_indices]
         logits = qlinear(hidden, self.lm_head, self.lm_head_pack)
+        if self.spec.final_logit_softcap:
+            cap = self.spec.final_logit_softcap
+            logits = torch.tanh(logits / cap) * cap
         if return_both:  # draft-model speculative needs the features too
             return logits, hidden
         return logits

@@ -131,6 +131,9 @@ def random_init(model, cfg: EngineConfig) -> None:
             layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.fill_(1.0)
         layer.post_attn_norm.fill_(1.0)
+        if layer.pre_ff_norm is not None:  # Gemma-2 sandwich norms
+            layer.pre_ff_norm.fill_(1.0)
+            layer.post_ff_norm.fill_(1.0)
 
 
 def _random_init_mla_attn(layer, spec, li, seed, dtype, device, tp, rank):
@@ -239,7 +242,8 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
     if model.embed is not None:
         model.embed.copy_(get(pre + "embed_tokens.weight"))
     if model.final_norm is not None:
-        model.final_norm.copy_(get(pre + "norm.weight"))
+        fn = get(pre + "norm.weight")
+        model.final_norm.copy_(fn + 1 if spec.sandwich_norms else fn)
     if model.lm_head is not None and not spec.tie_word_embeddings:
         model.lm_head.copy_(get("lm_head.weight"))
     for local_i, layer in enumerate(model.layers):
@@ -373,8 +377,20 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             ]))
             dn = get(p + "mlp.down_proj.weight")
             layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
-        layer.input_norm.copy_(get(p + "input_layernorm.weight"))
-        layer.post_attn_norm.copy_(get(p + "post_attention_layernorm.weight"))
+        if spec.sandwich_norms:
+            # Gemma-2 RMSNorm multiplies by (1 + w); store the EFFECTIVE
+            # weight so the shared rms_norm kernel applies
+            layer.input_norm.copy_(get(p + "input_layernorm.weight") + 1)
+            layer.post_attn_norm.copy_(
+                get(p + "post_attention_layernorm.weight") + 1)
+            layer.pre_ff_norm.copy_(
+                get(p + "pre_feedforward_layernorm.weight") + 1)
+            layer.post_ff_norm.copy_(
+                get(p + "post_feedforward_layernorm.weight") + 1)
+        else:
+            layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+            layer.post_attn_norm.copy_(
+                get(p + "post_attention_layernorm.weight"))
 
 
 def merge_lora(model, cfg: EngineConfig, adapter_dir: str | Path) -> int:

@@ -133,9 +133,14 @@ def greedy_sample(logits) -> torch.Tensor:
 
 
 def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
-                      scale: float, sinks=None, window: int = 0) -> None:
+                      scale: float, sinks=None, window: int = 0,
+                      softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
+        if softcap:
+            raise NotImplementedError(
+                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
+                "decode kernel yet (r3, with the D-256 template)")
         if (sinks is not None or window) and not _oss_kernels_enabled():
             # GPT-OSS sinks/sliding-window kernels are written but not yet
             # GPU-validated — fail loudly rather than silently mis-attend
@@ -148,7 +153,7 @@ def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
     else:
         torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                     seq_lens, scale, sinks=sinks,
-                                    window=window)
+                                    window=window, softcap=softcap)
 
 
 def mla_decode(ctx_out, q_cat, lat_cache, block_tables, seq_lens,
@@ -181,9 +186,14 @@ def build_prefill_tiles(seq_lens: list[int], device) -> tuple[torch.Tensor, torc
 
 
 def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
-                        tiles=None, sinks=None, window: int = 0) -> None:
+                        tiles=None, sinks=None, window: int = 0,
+                        softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
+        if softcap:
+            raise NotImplementedError(
+                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
+                "prefill kernel yet (r3)")
         if (sinks is not None or window) and not _oss_kernels_enabled():
             raise NotImplementedError(
                 "attention sinks / sliding window CDNA4 prefill kernel is "
@@ -194,7 +204,8 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
                           sinks=_sinks_f32(sinks), window=window)
     else:
         torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale,
-                                      sinks=sinks, window=window)
+                                      sinks=sinks, window=window,
+                                      softcap=softcap)
 
 
 def build_paged_prefill_tiles(seq_starts: list[int], seq_hists: list[int],
@@ -216,12 +227,17 @@ def build_paged_prefill_tiles(seq_starts: list[int], seq_hists: list[int],
 def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                        seq_starts: list[int], seq_hists: list[int],
                        seq_news: list[int], scale: float, tiles=None,
-                       sinks=None, window: int = 0) -> None:
+                       sinks=None, window: int = 0,
+                       softcap: float = 0.0) -> None:
     """Prefill-with-history: suffix/chunk rows attend to cached paged KV +
     their own freshly written positions through the MFMA flash kernel
     (removes the r1-measured ~3.5x paged-decode-row penalty)."""
     hip = _backend(q)
     if hip is not None:
+        if softcap:
+            raise NotImplementedError(
+                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
+                "paged-prefill kernel yet (r3)")
         if (sinks is not None or window) and not _oss_kernels_enabled():
             raise NotImplementedError(
                 "attention sinks / sliding window CDNA4 paged-prefill kernel "
@@ -235,7 +251,8 @@ def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
     else:
         torch_ref.paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                                      seq_starts, seq_hists, seq_news, scale,
-                                     sinks=sinks, window=window)
+                                     sinks=sinks, window=window,
+                                     softcap=softcap)
 
 
 _SG_WORKSPACES: dict = {}

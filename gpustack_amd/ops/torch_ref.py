@@ -166,6 +166,14 @@ def _sink_softmax(att: torch.Tensor, sinks: torch.Tensor | None):
     return probs[..., :-1]
 
 
+def _softcap(att: torch.Tensor, cap: float):
+    """Gemma-2 attention logit softcapping: tanh(att/cap)*cap applied to
+    the SCALED scores before the mask (HF eager semantics)."""
+    if cap:
+        return torch.tanh(att / cap) * cap
+    return att
+
+
 def _window_mask(qpos: torch.Tensor, kpos: torch.Tensor, window: int):
     """Causal (+ optional sliding-window) additive mask [Tq, L]."""
     m = torch.where(kpos.unsqueeze(0) <= qpos.unsqueeze(1), 0.0, float("-inf"))
@@ -185,6 +193,7 @@ def paged_attn_decode(
     scale: float,
     sinks: torch.Tensor | None = None,
     window: int = 0,
+    softcap: float = 0.0,
 ) -> None:
     """q/out: [N, Hq, D]; caches [B, Hkv, BS, D]."""
     N, Hq, D = q.shape
@@ -200,7 +209,7 @@ def paged_attn_decode(
         keys = keys.repeat_interleave(GQ, dim=0)  # [Hq, L, D]
         vals = vals.repeat_interleave(GQ, dim=0)
         qi = q[i].float().unsqueeze(1)  # [Hq, 1, D]
-        att = (qi @ keys.transpose(1, 2)) * scale
+        att = _softcap((qi @ keys.transpose(1, 2)) * scale, softcap)
         att = att + _window_mask(torch.tensor([L - 1], device=q.device),
                                  torch.arange(L, device=q.device), window)
         att = _sink_softmax(att, sinks)
@@ -216,6 +225,7 @@ def varlen_prefill_attn(
     scale: float,
     sinks: torch.Tensor | None = None,
     window: int = 0,
+    softcap: float = 0.0,
 ) -> None:
     """q: [T, Hq, D]; k/v: [T, Hkv, D]; causal within each sequence."""
     Hq = q.shape[1]
@@ -226,7 +236,7 @@ def varlen_prefill_attn(
         qs = q[start : start + L].float().permute(1, 0, 2)  # [Hq, L, D]
         ks = k[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
         vs = v[start : start + L].float().permute(1, 0, 2).repeat_interleave(GQ, dim=0)
-        att = (qs @ ks.transpose(1, 2)) * scale
+        att = _softcap((qs @ ks.transpose(1, 2)) * scale, softcap)
         pos = torch.arange(L, device=q.device)
         att = att + _window_mask(pos, pos, window)
         att = _sink_softmax(att, sinks)
@@ -246,6 +256,7 @@ def paged_prefill_attn(
     scale: float,
     sinks: torch.Tensor | None = None,
     window: int = 0,
+    softcap: float = 0.0,
 ) -> None:
     """Prefill-with-history: q/out hold only the NEW (suffix) rows of each
     sequence; K/V for positions [0, hist+new) are gathered from the paged
@@ -269,7 +280,8 @@ def paged_prefill_attn(
         vals = vals.repeat_interleave(GQ, dim=0)
         start = seq_starts[i]
         qs = q[start:start + new].float().permute(1, 0, 2)  # [Hq, new, D]
-        att = (qs @ keys.transpose(1, 2)) * scale           # [Hq, new, L]
+        att = _softcap((qs @ keys.transpose(1, 2)) * scale,
+                       softcap)                             # [Hq, new, L]
         pos = torch.arange(L, device=q.device)
         qpos = hist + torch.arange(new, device=q.device)
         att = att + _window_mask(qpos, pos, window)
