@@ -76,6 +76,10 @@ class ModelSpec:
     # logits, custom attention scale, GeGLU MLP; sliding window
     # alternates via layer_types (even layers) like GPT-OSS
     sandwich_norms: bool = False
+    # Gemma-3: sliding layers rope at a LOCAL base frequency (10k) while
+    # full-attention layers use rope_theta (1M, linearly scaled) — the
+    # model holds two cos/sin caches and each layer picks by window
+    rope_local_theta: float = 0.0
     embed_scale: float = 0.0             # 0 = no scaling
     attn_logit_softcap: float = 0.0      # 0 = off
     final_logit_softcap: float = 0.0
@@ -151,7 +155,7 @@ class ModelSpec:
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=arch.startswith("Qwen2"),
-            qk_norm=(arch.startswith("Qwen3")
+            qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
                      or bool(cfg.get("use_qk_norm", False))),
             eos_token_id=eos,
             num_experts=cfg.get("num_experts",
@@ -177,10 +181,15 @@ class ModelSpec:
                                    .get("partial_rotary_factor") or 1.0),
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
-            if (arch.startswith("GptOss") or arch.startswith("Gemma2"))
+            if (arch.startswith("GptOss") or arch.startswith("Gemma"))
             else 0,
             layer_types=tuple(cfg["layer_types"])
-            if cfg.get("layer_types") else None,
+            if cfg.get("layer_types") else (
+                tuple("sliding_attention"
+                      if (i + 1) % int(cfg.get("sliding_window_pattern", 6))
+                      else "full_attention"
+                      for i in range(cfg.get("num_hidden_layers", 0)))
+                if arch.startswith("Gemma3") else None),
             moe_act="clamped_swiglu" if arch.startswith("GptOss") else "silu",
             moe_bias=arch.startswith("GptOss"),
             router_logit_bias=arch.startswith("GptOss"),
@@ -194,16 +203,19 @@ class ModelSpec:
             qk_rope_head_dim=cfg.get("qk_rope_head_dim", 0) or 0,
             v_head_dim=cfg.get("v_head_dim", 0) or 0,
             rope_interleave=bool(cfg.get("rope_interleave", True)),
-            sandwich_norms=arch.startswith("Gemma2"),
+            sandwich_norms=(arch.startswith("Gemma2")
+                            or arch.startswith("Gemma3")),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
                          if arch.startswith("Gemma") else 0.0),
             attn_logit_softcap=(cfg.get("attn_logit_softcapping") or 0.0)
-            if arch.startswith("Gemma2") else 0.0,
+            if arch.startswith("Gemma") else 0.0,
             final_logit_softcap=(cfg.get("final_logit_softcapping") or 0.0)
-            if arch.startswith("Gemma2") else 0.0,
+            if arch.startswith("Gemma") else 0.0,
             attn_scale=((cfg.get("query_pre_attn_scalar") or 0) ** -0.5
-                        if (arch.startswith("Gemma2")
+                        if (arch.startswith("Gemma")
                             and cfg.get("query_pre_attn_scalar")) else 0.0),
+            rope_local_theta=(cfg.get("rope_local_base_freq") or 0.0)
+            if arch.startswith("Gemma3") else 0.0,
             mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),
         )
 
@@ -371,6 +383,21 @@ PRESETS: dict[str, ModelSpec] = {
         sandwich_norms=True, embed_scale=3584 ** 0.5,
         attn_logit_softcap=50.0, final_logit_softcap=30.0,
         attn_scale=224 ** -0.5, mlp_act="gelu_tanh",
+    ),
+    # Gemma-3: dual rope (local 10k on sliding layers, linear-scaled 1M
+    # global), (1+w) qk-norm, 5:1 SWA pattern, no softcapping
+    "gemma-3-27b": ModelSpec(
+        architecture="Gemma3ForCausalLM", vocab_size=262208,
+        hidden_size=5376, intermediate_size=21504, num_layers=62,
+        num_heads=32, num_kv_heads=16, head_dim=128,
+        rope_theta=1000000.0, rope_local_theta=10000.0,
+        rope_scaling={"rope_type": "linear", "factor": 8.0},
+        max_position_embeddings=131072, tie_word_embeddings=True,
+        rms_norm_eps=1e-6, eos_token_id=1, sliding_window=1024,
+        layer_types=tuple("sliding_attention" if (i + 1) % 6 else
+                          "full_attention" for i in range(62)),
+        sandwich_norms=True, qk_norm=True, embed_scale=5376 ** 0.5,
+        attn_scale=168 ** -0.5, mlp_act="gelu_tanh",
     ),
     # CPU-test preset: DeepSeek-shaped MLA + MoE (tests/test_deepseek.py,
     # TP exactness on gloo)

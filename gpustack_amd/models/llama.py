@@ -914,6 +914,16 @@ class LlamaForCausalLM(nn.Module):
                 base=spec.rope_theta, scaling=spec.rope_scaling,
             )
         self.register_buffer("cos_sin", cache.to(device), persistent=False)
+        if spec.rope_local_theta:
+            # Gemma-3: sliding layers rope at the LOCAL base, unscaled
+            local = ops.build_cos_sin_cache(
+                spec.head_dim, spec.head_dim, cfg.max_model_len,
+                base=spec.rope_local_theta,
+            )
+            self.register_buffer("cos_sin_local", local.to(device),
+                                 persistent=False)
+        else:
+            self.cos_sin_local = None
         self.to(device)
 
     @torch.inference_mode()
@@ -941,7 +951,10 @@ class LlamaForCausalLM(nn.Module):
         for i, layer in enumerate(self.layers):
             if off is not None and i >= off.first:
                 off.bind(i)
-            x, residual = layer(x, residual, meta, self.cos_sin, kv.k_caches[i], kv.v_caches[i])
+            cs = (self.cos_sin_local
+                  if (self.cos_sin_local is not None and layer.attn.window)
+                  else self.cos_sin)
+            x, residual = layer(x, residual, meta, cs, kv.k_caches[i], kv.v_caches[i])
         if self.comm.pp_size > 1 and not self.comm.is_last_stage:
             if residual is None:  # sandwich layers carry the true stream
                 self.comm.send_hidden(x.contiguous())

@@ -295,8 +295,11 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                 sk = tensors[p + "self_attn.sinks"].float()
                 layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
             if spec.qk_norm:
-                layer.attn.q_norm.copy_(get(p + "self_attn.q_norm.weight"))
-                layer.attn.k_norm.copy_(get(p + "self_attn.k_norm.weight"))
+                off = 1 if spec.sandwich_norms else 0  # Gemma: (1+w) norms
+                layer.attn.q_norm.copy_(
+                    get(p + "self_attn.q_norm.weight") + off)
+                layer.attn.k_norm.copy_(
+                    get(p + "self_attn.k_norm.weight") + off)
         if hasattr(layer.mlp, "router_w")                 and spec.architecture.startswith("GptOss"):
             # GPT-OSS: router `mlp.router.{weight,bias}`; experts stored
             # TRANSPOSED ([E, h, 2i] / [E, i, h]) with INTERLEAVED

@@ -82,6 +82,9 @@ def build_cos_sin_cache(
                 / (high - low)).clamp(0, 1)
         extrap_f = 1 - ramp
         inv_freq = (inv_freq / factor) * (1 - extrap_f) + inv_freq * extrap_f
+    elif scaling and scaling.get("rope_type", scaling.get("type")) == "linear":
+        # linear position interpolation (Gemma-3 global layers: factor 8)
+        inv_freq = inv_freq / scaling["factor"]
     elif scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
         factor = scaling["factor"]
         lo = scaling["low_freq_factor"]

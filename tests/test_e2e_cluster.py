@@ -33,24 +33,42 @@ def cluster():
     from gpustack_amd.server.app import create_app
     from gpustack_amd.worker.agent import WorkerAgent
 
-    sport, wport = _free_port(), _free_port()
+    wport = _free_port()
     lo = _free_port()
-    cfg = Config(
-        data_dir=tempfile.mkdtemp(), bootstrap_password="pw",
-        host="127.0.0.1", port=sport,
-    )
-    app = create_app(cfg, start_background=True)
-    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=sport,
-                                           log_level="warning"))
-    st = threading.Thread(target=server.run, daemon=True)
-    st.start()
-    base = f"http://127.0.0.1:{sport}"
-    for _ in range(100):
-        try:
-            httpx.get(base + "/healthz", timeout=1)
+    # retry the server bind: _free_port() closes the probe socket before
+    # uvicorn rebinds it, so a parallel process can steal the port (seen
+    # as a transient fixture error under full-suite load)
+    app = server = None
+    base = ""
+    for _attempt in range(3):
+        sport = _free_port()
+        cfg = Config(
+            data_dir=tempfile.mkdtemp(), bootstrap_password="pw",
+            host="127.0.0.1", port=sport,
+        )
+        app = create_app(cfg, start_background=True)
+        server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1",
+                                               port=sport,
+                                               log_level="warning"))
+        st = threading.Thread(target=server.run, daemon=True)
+        st.start()
+        base = f"http://127.0.0.1:{sport}"
+        up = False
+        for _ in range(100):
+            try:
+                httpx.get(base + "/healthz", timeout=1)
+                up = True
+                break
+            except httpx.HTTPError:
+                time.sleep(0.1)
+        if up:
             break
-        except httpx.HTTPError:
-            time.sleep(0.1)
+        from gpustack_amd.server.app import stop_background_tasks
+
+        stop_background_tasks(app)
+        server.should_exit = True
+    else:
+        raise RuntimeError("e2e server never became healthy")
 
     wcfg = Config(
         data_dir=tempfile.mkdtemp(),
