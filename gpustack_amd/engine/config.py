@@ -127,6 +127,21 @@ class ModelSpec:
         emb = v * h * (1 if self.tie_word_embeddings else 2)
         return (total_layers + emb + h) * dtype_size
 
+    @staticmethod
+    def _rope_scaling_from(cfg: dict) -> dict | None:
+        sc = cfg.get("rope_scaling") or (
+            (cfg.get("rope_parameters") or None)
+            if (cfg.get("rope_parameters") or {}).get("rope_type", "default")
+            not in ("default",) else None)
+        if sc and sc.get("rope_type", sc.get("type")) == "longrope" \
+                and "original_max_position_embeddings" not in sc:
+            # Phi-3/4 keep the pretraining length top-level
+            sc = dict(sc)
+            sc["original_max_position_embeddings"] = cfg.get(
+                "original_max_position_embeddings",
+                cfg.get("max_position_embeddings", 4096))
+        return sc
+
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelSpec":
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
@@ -146,11 +161,7 @@ class ModelSpec:
             head_dim=hd,
             rope_theta=cfg.get("rope_theta")
             or (cfg.get("rope_parameters") or {}).get("rope_theta", 10000.0),
-            rope_scaling=cfg.get("rope_scaling")
-            or ((cfg.get("rope_parameters") or None)
-                if (cfg.get("rope_parameters") or {}).get("rope_type",
-                                                          "default")
-                not in ("default",) else None),
+            rope_scaling=cls._rope_scaling_from(cfg),
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
@@ -370,6 +381,13 @@ PRESETS: dict[str, ModelSpec] = {
         vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,
         num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
         rope_theta=10000.0, eos_token_id=1,
+    ),
+    # Phi-4 14B (Phi3ForCausalLM graph: fused qkv/gate_up, plain rope)
+    "phi-4": ModelSpec(
+        architecture="Phi3ForCausalLM", vocab_size=100352,
+        hidden_size=5120, intermediate_size=17920, num_layers=40,
+        num_heads=40, num_kv_heads=10, head_dim=128, rope_theta=250000.0,
+        max_position_embeddings=16384, eos_token_id=100257,
     ),
     # Gemma-2: sandwich norms, GeGLU, softcapping, alternating SWA,
     # scaled tied embeddings (CPU-oracle family; GPU kernels share the

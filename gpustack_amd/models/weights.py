@@ -249,6 +249,21 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
     for local_i, layer in enumerate(model.layers):
         li = off + local_i
         p = f"{pre}layers.{li}."
+        if p + "self_attn.qkv_proj.weight" in tensors:
+            # Phi-3/4: fused qkv checkpoint tensor [nq + 2*nk, h] — split
+            # then shard exactly like separate projections
+            fused = get(p + "self_attn.qkv_proj.weight")
+            nq_f = spec.num_heads * d
+            nk_f = spec.num_kv_heads * d
+            tensors[p + "self_attn.q_proj.weight"] = fused[:nq_f]
+            tensors[p + "self_attn.k_proj.weight"] = fused[nq_f:nq_f + nk_f]
+            tensors[p + "self_attn.v_proj.weight"] = fused[nq_f + nk_f:]
+        if p + "mlp.gate_up_proj.weight" in tensors \
+                and not hasattr(layer.mlp, "router_w"):
+            fused = get(p + "mlp.gate_up_proj.weight")
+            ii = spec.intermediate_size
+            tensors[p + "mlp.gate_proj.weight"] = fused[:ii]
+            tensors[p + "mlp.up_proj.weight"] = fused[ii:]
         if spec.kv_lora_rank:
             # DeepSeek MLA: latent projections replicate, per-head
             # projections shard by head (models/llama.py MLAAttention)

@@ -82,6 +82,23 @@ def build_cos_sin_cache(
                 / (high - low)).clamp(0, 1)
         extrap_f = 1 - ramp
         inv_freq = (inv_freq / factor) * (1 - extrap_f) + inv_freq * extrap_f
+    elif scaling and scaling.get("rope_type", scaling.get("type")) == "longrope":
+        # LongRoPE (Phi-3/4; HF _compute_longrope_parameters): per-dim
+        # frequency rescale factors, long vs short chosen by the cache's
+        # target length vs the pretraining length — a single static cache
+        # per engine (vLLM-style), not HF's per-forward dynamic switch
+        import math
+
+        orig = scaling["original_max_position_embeddings"]
+        factors = (scaling["long_factor"] if max_pos > orig
+                   else scaling["short_factor"])
+        inv_freq = inv_freq / torch.tensor(factors, dtype=torch.float64)[:half]
+        factor = scaling.get("factor") or (max_pos / orig)
+        attention_factor = scaling.get("attention_factor")
+        if attention_factor is None:
+            attention_factor = (1.0 if factor <= 1.0 else
+                                math.sqrt(1 + math.log(factor)
+                                          / math.log(orig)))
     elif scaling and scaling.get("rope_type", scaling.get("type")) == "linear":
         # linear position interpolation (Gemma-3 global layers: factor 8)
         inv_freq = inv_freq / scaling["factor"]
