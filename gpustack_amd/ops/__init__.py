@@ -137,19 +137,17 @@ def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
                       softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if softcap:
-            raise NotImplementedError(
-                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
-                "decode kernel yet (r3, with the D-256 template)")
-        if (sinks is not None or window) and not _oss_kernels_enabled():
+        if (sinks is not None or window or softcap) \
+                and not _oss_kernels_enabled():
             # GPT-OSS sinks/sliding-window kernels are written but not yet
             # GPU-validated — fail loudly rather than silently mis-attend
             # (opt in with GPUSTACK_AMD_OSS_KERNELS=1; r3 flips the default)
             raise NotImplementedError(
-                "attention sinks / sliding window CDNA4 decode kernel is "
+                "sinks/window/softcap CDNA4 decode kernel variants are "
                 "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
-                              scale, sinks=_sinks_f32(sinks), window=window)
+                              scale, sinks=_sinks_f32(sinks), window=window,
+                              softcap=softcap)
     else:
         torch_ref.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                     seq_lens, scale, sinks=sinks,
@@ -190,18 +188,16 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
                         softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if softcap:
+        if (sinks is not None or window or softcap) \
+                and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
-                "prefill kernel yet (r3)")
-        if (sinks is not None or window) and not _oss_kernels_enabled():
-            raise NotImplementedError(
-                "attention sinks / sliding window CDNA4 prefill kernel is "
+                "sinks/window/softcap CDNA4 prefill kernel variants are "
                 "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         if tiles is None or tiles[0] is None:
             tiles = build_prefill_tiles(seq_lens, q.device)
         hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale,
-                          sinks=_sinks_f32(sinks), window=window)
+                          sinks=_sinks_f32(sinks), window=window,
+                          softcap=softcap)
     else:
         torch_ref.varlen_prefill_attn(out, q, k, v, seq_lens, scale,
                                       sinks=sinks, window=window,
@@ -234,20 +230,18 @@ def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
     (removes the r1-measured ~3.5x paged-decode-row penalty)."""
     hip = _backend(q)
     if hip is not None:
-        if softcap:
+        if (sinks is not None or window or softcap) \
+                and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "attention logit softcapping (Gemma-2) is not in the CDNA4 "
-                "paged-prefill kernel yet (r3)")
-        if (sinks is not None or window) and not _oss_kernels_enabled():
-            raise NotImplementedError(
-                "attention sinks / sliding window CDNA4 paged-prefill kernel "
-                "is unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+                "sinks/window/softcap CDNA4 paged-prefill kernel variants "
+                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         if tiles is None:
             tiles = build_paged_prefill_tiles(seq_starts, seq_hists,
                                               seq_news, q.device)
         hip.flash_prefill_paged(out, q, k_cache, v_cache, block_tables,
                                 tiles[0], tiles[1], tiles[2], tiles[3],
-                                tiles[4], scale, sinks=_sinks_f32(sinks), window=window)
+                                tiles[4], scale, sinks=_sinks_f32(sinks),
+                                window=window, softcap=softcap)
     else:
         torch_ref.paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
                                      seq_starts, seq_hists, seq_news, scale,

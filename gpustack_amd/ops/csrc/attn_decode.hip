@@ -34,10 +34,13 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     const int* __restrict__ seq_lens,        // [N]
     int Hkv, int max_blocks, float scale, long q_stride,
     const float* __restrict__ sinks,         // [Hq] or null (GPT-OSS)
-    int window) {                            // 0 = full attention
+    int window,                              // 0 = full attention
+    float softcap) {                         // 0 = off (Gemma-2 tanh cap)
   constexpr int LPG = 16;           // lanes per token-group
-  constexpr int DV = D / LPG;       // dims per lane (8 for D=128, 4 for 64)
-  static_assert(DV == 8 || DV == 4, "decode kernel assumes D in {64, 128}");
+  constexpr int DV = D / LPG;       // dims per lane (4/8/16 for D 64/128/256)
+  static_assert(DV == 4 || DV == 8 || DV == 16,
+                "decode kernel assumes D in {64, 128, 256}");
+  constexpr int NV = (DV >= 8) ? DV / 8 : 1;  // u16x8 vectors per lane slice
   // tokens batched per softmax update: trade VALU savings against VGPR
   // pressure (GQ>=4 would spill at TB=4)
   constexpr int TB = (GQ <= 2) ? (BS / 4) : ((GQ <= 5) ? 2 : 1);
@@ -59,20 +62,26 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   // dot-product instruction count (occupancy was the bottleneck: 2-3
   // waves/SIMD at 168-200 VGPRs, profiles/r02_optimization_log.md).
   // fp8 path: f32 q pre-scaled as before (k dequants through f32 anyway).
-  using U16V = std::conditional_t<DV == 8, u16x8, u16x4>;
-  using U8V = std::conditional_t<DV == 8, u8x8, u8x4>;
+  // lane slice = DV dims: one u16x4 (DV=4) or NV u16x8 vectors
+  using U16V = std::conditional_t<DV == 4, u16x4, u16x8>;
+  using U8V = std::conditional_t<DV == 4, u8x4, u8x8>;
+  constexpr int VW = (DV == 4) ? 4 : 8;     // elements per vector
   float qr[FP8 ? GQ : 1][DV];
-  U16V qb[FP8 ? 1 : GQ];
+  U16V qb[FP8 ? 1 : GQ][NV];
 #pragma unroll
   for (int gq = 0; gq < GQ; ++gq) {
     const unsigned short* qp =
         q + (long)seq * q_stride + ((long)h * GQ + gq) * D + sub * DV;
-    U16V u = *reinterpret_cast<const U16V*>(qp);
-    if constexpr (FP8) {
 #pragma unroll
-      for (int j = 0; j < DV; ++j) qr[gq][j] = bf2f(u[j]) * scale;
-    } else {
-      qb[gq] = u;
+    for (int vv = 0; vv < NV; ++vv) {
+      U16V u = *reinterpret_cast<const U16V*>(qp + vv * VW);
+      if constexpr (FP8) {
+#pragma unroll
+        for (int j = 0; j < VW; ++j)
+          qr[gq][vv * VW + j] = bf2f(u[j]) * scale;
+      } else {
+        qb[gq][vv] = u;
+      }
     }
   }
 
@@ -103,17 +112,23 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     for (int tb = 0; tb < BS / 4; tb += TB) {
       const int base_tok = page * BS + tb * 4 + g;
       using VecT = std::conditional_t<FP8, U8V, U16V>;
-      VecT vu[TB];
-      VecT ku[TB];              // bf16: K stays packed (dot2 consumes it)
+      VecT vu[TB][NV];
+      VecT ku[TB][NV];          // bf16: K stays packed (dot2 consumes it)
       float kf[FP8 ? TB : 1][DV];  // fp8: K dequanted once, reused per gq
 #pragma unroll
       for (int it = 0; it < TB; ++it) {
         const int tok = (tb + it) * 4 + g;
-        ku[it] = *reinterpret_cast<const VecT*>(kbase + tok * D + sub * DV);
-        vu[it] = *reinterpret_cast<const VecT*>(vbase + tok * D + sub * DV);
-        if constexpr (FP8) {
 #pragma unroll
-          for (int j = 0; j < DV; ++j) kf[it][j] = fp8_to_f32(ku[it][j]);
+        for (int vv = 0; vv < NV; ++vv) {
+          ku[it][vv] = *reinterpret_cast<const VecT*>(
+              kbase + tok * D + sub * DV + vv * VW);
+          vu[it][vv] = *reinterpret_cast<const VecT*>(
+              vbase + tok * D + sub * DV + vv * VW);
+          if constexpr (FP8) {
+#pragma unroll
+            for (int j = 0; j < VW; ++j)
+              kf[it][vv * VW + j] = fp8_to_f32(ku[it][vv][j]);
+          }
         }
       }
 #pragma unroll
@@ -126,8 +141,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
             for (int j = 0; j < DV; ++j) d += qr[gq][j] * kf[it][j];
           } else {
-            const bf16x2* qa = reinterpret_cast<const bf16x2*>(&qb[gq]);
-            const bf16x2* ka = reinterpret_cast<const bf16x2*>(&ku[it]);
+            const bf16x2* qa = reinterpret_cast<const bf16x2*>(&qb[gq][0]);
+            const bf16x2* ka = reinterpret_cast<const bf16x2*>(&ku[it][0]);
 #pragma unroll
             for (int j = 0; j < DV / 2; ++j)
               d = __builtin_amdgcn_fdot2_f32_bf16(qa[j], ka[j], d, false);
@@ -143,7 +158,9 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         for (int it = 0; it < TB; ++it) {
           const int tpos = base_tok + it * 4;
           const bool valid = tpos < len && tpos >= wstart;
-          dot[it] = valid ? dot[it] : -INFINITY;
+          float dv = dot[it];
+          if (softcap > 0.f) dv = tanhf(dv / softcap) * softcap;
+          dot[it] = valid ? dv : -INFINITY;
           pmax = fmaxf(pmax, dot[it]);
         }
         if (pmax == -INFINITY) continue;
@@ -162,7 +179,8 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
           float a = acc[gq][j] * corr;
 #pragma unroll
           for (int it = 0; it < TB; ++it)
-            a += p[it] * (FP8 ? fp8_to_f32(vu[it][j]) : bf2f(vu[it][j]));
+            a += p[it] * (FP8 ? fp8_to_f32(vu[it][j / VW][j % VW])
+                              : bf2f(vu[it][j / VW][j % VW]));
           acc[gq][j] = a;
         }
         m[gq] = nm;
@@ -249,26 +267,24 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
                               const int* seq_lens, int N, int Hq, int Hkv,
                               int D, int max_blocks, float scale, long q_stride,
                               int fp8, const float* sinks, int window,
-                              int* err_unsupported, hipStream_t s) {
+                              float softcap, int* err_unsupported,
+                              hipStream_t s) {
   const int GQ = Hq / Hkv;
   dim3 grid(N, Hkv);
   dim3 block(THREADS);
   *err_unsupported = 0;
-  if (D != 128 && D != 64) { *err_unsupported = 1; return; }
+  if (D != 128 && D != 64 && D != 256) { *err_unsupported = 1; return; }
+  if (D == 256 && GQ > 4) { *err_unsupported = 1; return; }  // VGPR budget
+#define LAUNCH_D(DD, G, F)                                                     \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, G, F>), grid, block, 0, s,  \
+                     (unsigned short*)out, (const unsigned short*)q, kc, vc,   \
+                     block_tables, seq_lens, Hkv, max_blocks, scale, q_stride, \
+                     sinks, window, softcap)
 #define LAUNCH_GQ2(G, F)                                                       \
   do {                                                                         \
-    if (D == 128)                                                              \
-      hipLaunchKernelGGL((paged_attn_decode_kernel<128, G, F>), grid, block,   \
-                         0, s, (unsigned short*)out,                           \
-                         (const unsigned short*)q, kc, vc, block_tables,       \
-                         seq_lens, Hkv, max_blocks, scale, q_stride, sinks,    \
-                         window);                                              \
-    else                                                                       \
-      hipLaunchKernelGGL((paged_attn_decode_kernel<64, G, F>), grid, block,    \
-                         0, s, (unsigned short*)out,                           \
-                         (const unsigned short*)q, kc, vc, block_tables,       \
-                         seq_lens, Hkv, max_blocks, scale, q_stride, sinks,    \
-                         window);                                              \
+    if (D == 128) LAUNCH_D(128, G, F);                                         \
+    else if (D == 64) LAUNCH_D(64, G, F);                                      \
+    else if ((G) <= 4) LAUNCH_D(256, (G) <= 4 ? (G) : 1, F);                   \
   } while (0)
 #define LAUNCH_GQ(G)                                                          \
   do {                                                                        \
@@ -287,4 +303,5 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
   }
 #undef LAUNCH_GQ
 #undef LAUNCH_GQ2
+#undef LAUNCH_D
 }

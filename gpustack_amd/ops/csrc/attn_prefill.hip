@@ -59,7 +59,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     int Hq, int Hkv, float scale,
     long qs, long ks, long vs,             // row strides (elements)
     const float* __restrict__ sinks,       // [Hq] or null
-    int window) {                          // 0 = full causal
+    int window,                            // 0 = full causal
+    float softcap) {                       // 0 = off (Gemma-2 tanh cap)
   constexpr int KU = BK * (D / 8) / PF_THREADS;  // K u16x8 units/thread
   constexpr int VU = D * (BK / 8) / PF_THREADS;  // VT units/thread
   constexpr int QK = D / 32;                     // q k-chunks
@@ -189,6 +190,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
         bool ok = (kvpos <= qpos) && (kvpos < len) && (qpos < len);
         if (window > 0) ok = ok && (kvpos > qpos - window);
         x = ok ? x : -INFINITY;
@@ -308,7 +310,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     const int* __restrict__ tile_seq,      // [ntiles] row into block_tables
     int Hq, int Hkv, int maxb, float scale, long qs,
     const float* __restrict__ sinks,       // [Hq] or null
-    int window) {                          // 0 = full causal
+    int window,                            // 0 = full causal
+    float softcap) {                       // 0 = off (Gemma-2 tanh cap)
   constexpr int KU = BK * (D / 8) / PF_THREADS;
   constexpr int VU = D * (BK / 8) / PF_THREADS;
   constexpr int QK = D / 32;
@@ -445,6 +448,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
         bool ok = (kvpos <= qpos) && (kvpos < len) && qvalid;
         if (window > 0) ok = ok && (kvpos > qpos - window);
         x = ok ? x : -INFINITY;
@@ -536,17 +540,24 @@ void flash_prefill_paged_launch(
     const int* block_tables, const int* tile_qstart, const int* tile_q0,
     const int* tile_hist, const int* tile_new, const int* tile_seq,
     int ntiles, int Hq, int Hkv, int D, int maxb, float scale, long qs,
-    const float* sinks, int window, int* err_unsupported, hipStream_t s) {
+    const float* sinks, int window, float softcap, int* err_unsupported,
+    hipStream_t s) {
   *err_unsupported = 0;
-  if ((D != 128 && D != 64) || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  if ((D != 128 && D != 64 && D != 256) || Hq % Hkv != 0) {
+    *err_unsupported = 1;
+    return;
+  }
   dim3 grid(ntiles, Hq);
 #define PPG_LAUNCH(DD)                                                        \
   hipLaunchKernelGGL(flash_prefill_paged_kernel<DD>, grid, dim3(PF_THREADS),  \
                      0, s, (unsigned short*)out, (const unsigned short*)q,    \
                      (const unsigned short*)kc, (const unsigned short*)vc,    \
                      block_tables, tile_qstart, tile_q0, tile_hist, tile_new, \
-                     tile_seq, Hq, Hkv, maxb, scale, qs, sinks, window)
-  if (D == 128) PPG_LAUNCH(128); else PPG_LAUNCH(64);
+                     tile_seq, Hq, Hkv, maxb, scale, qs, sinks, window,       \
+                     softcap)
+  if (D == 128) PPG_LAUNCH(128);
+  else if (D == 64) PPG_LAUNCH(64);
+  else PPG_LAUNCH(256);
 #undef PPG_LAUNCH
 }
 
@@ -555,17 +566,23 @@ void flash_prefill_launch(void* out, const void* q, const void* k,
                           const int* tile_q0, const int* tile_len, int ntiles,
                           int Hq, int Hkv, int D, float scale,
                           long qs, long ks, long vs, const float* sinks,
-                          int window, int* err_unsupported, hipStream_t s) {
+                          int window, float softcap, int* err_unsupported,
+                          hipStream_t s) {
   *err_unsupported = 0;
-  if ((D != 128 && D != 64) || Hq % Hkv != 0) { *err_unsupported = 1; return; }
+  if ((D != 128 && D != 64 && D != 256) || Hq % Hkv != 0) {
+    *err_unsupported = 1;
+    return;
+  }
   dim3 grid(ntiles, Hq);
 #define PF_LAUNCH(DD)                                                        \
   hipLaunchKernelGGL(flash_prefill_kernel<DD>, grid, dim3(PF_THREADS), 0, s, \
                      (unsigned short*)out, (const unsigned short*)q,         \
                      (const unsigned short*)k, (const unsigned short*)v,     \
                      tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks,  \
-                     vs, sinks, window)
-  if (D == 128) PF_LAUNCH(128); else PF_LAUNCH(64);
+                     vs, sinks, window, softcap)
+  if (D == 128) PF_LAUNCH(128);
+  else if (D == 64) PF_LAUNCH(64);
+  else PF_LAUNCH(256);
 #undef PF_LAUNCH
 }
 

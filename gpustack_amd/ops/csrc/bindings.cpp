@@ -12,9 +12,9 @@ void silu_and_mul_launch(void*, const void*, long, int, hipStream_t);
 void reshape_and_cache_launch(const void*, const void*, void*, void*, const long*, int, int, int, int, long, long, int, hipStream_t);
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
 void mla_decode_launch(float*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, int*, hipStream_t);
-void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, const float*, int, int*, hipStream_t);
-void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, const float*, int, int*, hipStream_t);
-void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, const float*, int, int*, hipStream_t);
+void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, const float*, int, float, int*, hipStream_t);
+void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, const float*, int, float, int*, hipStream_t);
+void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, const float*, int, float, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
 void gemm8_launch(void*, const void*, const void*, int, int, int, int, int*, hipStream_t);
@@ -174,7 +174,8 @@ static const float* sink_ptr_checked(const c10::optional<at::Tensor>& sinks,
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
                        at::Tensor seq_lens, double scale,
-                       c10::optional<at::Tensor> sinks, long window) {
+                       c10::optional<at::Tensor> sinks, long window,
+                       double softcap) {
   check_bf16(out, "out");
   const long qstride = row_stride_3d(q, "q");
   check_cache(k_cache, "k_cache"); check_cache(v_cache, "v_cache");
@@ -192,7 +193,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                            seq_lens.data_ptr<int>(), N, Hq, Hkv, D, max_blocks,
                            (float)scale, qstride,
                            is_fp8_cache(k_cache) ? 1 : 0, sink_ptr,
-                           (int)window, &err, cur_stream(q));
+                           (int)window, (float)softcap, &err, cur_stream(q));
   TORCH_CHECK(!err, "paged_attn_decode: unsupported head_dim/GQ combination: D=",
               D, " Hq=", Hq, " Hkv=", Hkv);
   HIP_CHECK_LAST();
@@ -201,7 +202,8 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                    at::Tensor tile_start, at::Tensor tile_q0,
                    at::Tensor tile_len, double scale,
-                   c10::optional<at::Tensor> sinks, long window) {
+                   c10::optional<at::Tensor> sinks, long window,
+                   double softcap) {
   check_bf16(out, "out");
   const long qs = row_stride_3d(q, "q");
   const long ks = row_stride_3d(k, "k");
@@ -214,7 +216,7 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        tile_start.data_ptr<int>(), tile_q0.data_ptr<int>(),
                        tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D,
                        (float)scale, qs, ks, vs, sink_ptr_checked(sinks, Hq),
-                       (int)window, &err, cur_stream(q));
+                       (int)window, (float)softcap, &err, cur_stream(q));
   TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D);
   HIP_CHECK_LAST();
 }
@@ -224,7 +226,8 @@ void flash_prefill_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                          at::Tensor tile_qstart, at::Tensor tile_q0,
                          at::Tensor tile_hist, at::Tensor tile_new,
                          at::Tensor tile_seq, double scale,
-                         c10::optional<at::Tensor> sinks, long window) {
+                         c10::optional<at::Tensor> sinks, long window,
+                         double softcap) {
   check_bf16(out, "out");
   const long qs = row_stride_3d(q, "q");
   check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
@@ -243,8 +246,8 @@ void flash_prefill_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       block_tables.data_ptr<int>(), tile_qstart.data_ptr<int>(),
       tile_q0.data_ptr<int>(), tile_hist.data_ptr<int>(),
       tile_new.data_ptr<int>(), tile_seq.data_ptr<int>(), ntiles, Hq, Hkv, D,
-      maxb, (float)scale, qs, sink_ptr_checked(sinks, Hq), (int)window, &err,
-      cur_stream(q));
+      maxb, (float)scale, qs, sink_ptr_checked(sinks, Hq), (int)window,
+      (float)softcap, &err, cur_stream(q));
   TORCH_CHECK(!err, "flash_prefill_paged: unsupported config D=", D);
   HIP_CHECK_LAST();
 }
@@ -405,18 +408,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode, "paged GQA decode attention",
         py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("block_tables"), py::arg("seq_lens"), py::arg("scale"),
-        py::arg("sinks") = py::none(), py::arg("window") = 0);
+        py::arg("sinks") = py::none(), py::arg("window") = 0,
+        py::arg("softcap") = 0.0);
   m.def("flash_prefill", &flash_prefill, "varlen causal MFMA prefill attention",
         py::arg("out"), py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("tile_start"), py::arg("tile_q0"), py::arg("tile_len"),
         py::arg("scale"), py::arg("sinks") = py::none(),
-        py::arg("window") = 0);
+        py::arg("window") = 0, py::arg("softcap") = 0.0);
   m.def("flash_prefill_paged", &flash_prefill_paged,
         py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("block_tables"), py::arg("tile_qstart"), py::arg("tile_q0"),
         py::arg("tile_hist"), py::arg("tile_new"), py::arg("tile_seq"),
         py::arg("scale"), py::arg("sinks") = py::none(),
-        py::arg("window") = 0,
+        py::arg("window") = 0, py::arg("softcap") = 0.0,
         "MFMA prefill attention with paged-KV history (suffix/chunk rows)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "split-K skinny GEMM (bf16, f32 accum)");

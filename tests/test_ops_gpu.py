@@ -711,3 +711,45 @@ def test_mla_decode_kernel(H, lens):
         want = probs @ C[:, :R]                     # [H, R]
         assert torch.allclose(ctx[i], want, atol=2e-2, rtol=2e-2), \
             (ctx[i] - want).abs().max()
+
+
+@oss
+@pytest.mark.parametrize("D,Hq,Hkv", [(256, 8, 4), (128, 32, 8)])
+@pytest.mark.parametrize("lens", [[1], [17, 5, 160, 33], [700]])
+@pytest.mark.parametrize("window,softcap", [
+    (0, 50.0), (8, 50.0), (0, 0.0),
+])
+def test_oss_decode_softcap_d256(D, Hq, Hkv, lens, window, softcap):
+    """Gemma-2/3 kernel variants: D=256 decode template + tanh logit
+    softcapping (optionally with a sliding window) vs torch_ref."""
+    kc, vc, bt = _paged_pool(lens, Hkv, D)
+    q = torch.randn(len(lens), Hq, D, dtype=torch.bfloat16, device="cuda")
+    sl = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    ref = torch.empty_like(q)
+    scale = 1 / math.sqrt(D)
+    ops.paged_attn_decode(out, q, kc, vc, bt, sl, scale, window=window,
+                          softcap=softcap)
+    R.paged_attn_decode(ref, q, kc, vc, bt, sl, scale, window=window,
+                        softcap=softcap)
+    _close(out, ref, atol=3e-2, rtol=3e-2)
+
+
+@oss
+@pytest.mark.parametrize("D", [256, 128])
+@pytest.mark.parametrize("lens", [[64], [63, 70, 5], [300]])
+@pytest.mark.parametrize("window,softcap", [(0, 50.0), (8, 50.0)])
+def test_oss_prefill_softcap_d256(D, lens, window, softcap):
+    Hq, Hkv = 8, 4
+    T = sum(lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda") / 4
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda") / 4
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda") / 4
+    out = torch.empty_like(q)
+    ref = torch.empty_like(q)
+    scale = 1 / math.sqrt(D)
+    ops.varlen_prefill_attn(out, q, k, v, lens, scale, window=window,
+                            softcap=softcap)
+    R.varlen_prefill_attn(ref, q, k, v, lens, scale, window=window,
+                          softcap=softcap)
+    _close(out, ref, atol=4e-2, rtol=4e-2)
