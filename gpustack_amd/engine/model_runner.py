@@ -213,6 +213,20 @@ class ModelRunner:
         self.comm = comm or Communicator()
         self.cp_prefills = 0
         self.cp_suffixes = 0
+        if torch.device(cfg.device).type == "cuda" and (
+                cfg.spec.attention_sinks or cfg.spec.sliding_window
+                or cfg.spec.attn_logit_softcap
+                or cfg.spec.head_dim not in (128,)) \
+                and not cfg.spec.kv_lora_rank \
+                and os.environ.get("GPUSTACK_AMD_OSS_KERNELS") != "1":
+            # GPT-OSS / Gemma-class specs need the gated kernel variants;
+            # refuse at init with a clear message instead of crashing
+            # inside hipGraph capture on the first forward
+            raise NotImplementedError(
+                f"{cfg.spec.architecture} GPU serving needs the gated "
+                "CDNA4 kernel variants (sinks/window/softcap/head_dim "
+                f"{cfg.spec.head_dim}) — set GPUSTACK_AMD_OSS_KERNELS=1 "
+                "after the r3 validation pass; CPU serving is available")
         if cfg.spec.kv_lora_rank:
             # MLA (DeepSeek): latent attention serves on the CPU oracle
             # path today; the CDNA4 absorbed-attention kernels are the r3

@@ -137,14 +137,15 @@ def paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
                       softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if (sinks is not None or window or softcap) \
+        if (sinks is not None or window or softcap or q.shape[-1] != 128) \
                 and not _oss_kernels_enabled():
-            # GPT-OSS sinks/sliding-window kernels are written but not yet
-            # GPU-validated — fail loudly rather than silently mis-attend
-            # (opt in with GPUSTACK_AMD_OSS_KERNELS=1; r3 flips the default)
+            # the sinks/window/softcap and D!=128 kernel variants are
+            # written but not yet GPU-validated — fail loudly rather than
+            # silently mis-attend (opt in with GPUSTACK_AMD_OSS_KERNELS=1;
+            # r3 flips the default after the validation pass)
             raise NotImplementedError(
-                "sinks/window/softcap CDNA4 decode kernel variants are "
-                "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+                "sinks/window/softcap/D!=128 CDNA4 decode kernel variants "
+                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         hip.paged_attn_decode(out, q, k_cache, v_cache, block_tables, seq_lens,
                               scale, sinks=_sinks_f32(sinks), window=window,
                               softcap=softcap)
@@ -188,11 +189,11 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
                         softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if (sinks is not None or window or softcap) \
+        if (sinks is not None or window or softcap or q.shape[-1] != 128) \
                 and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "sinks/window/softcap CDNA4 prefill kernel variants are "
-                "unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+                "sinks/window/softcap/D!=128 CDNA4 prefill kernel variants "
+                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
         if tiles is None or tiles[0] is None:
             tiles = build_prefill_tiles(seq_lens, q.device)
         hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale,
@@ -230,11 +231,12 @@ def paged_prefill_attn(out, q, k_cache, v_cache, block_tables,
     (removes the r1-measured ~3.5x paged-decode-row penalty)."""
     hip = _backend(q)
     if hip is not None:
-        if (sinks is not None or window or softcap) \
+        if (sinks is not None or window or softcap or q.shape[-1] != 128) \
                 and not _oss_kernels_enabled():
             raise NotImplementedError(
-                "sinks/window/softcap CDNA4 paged-prefill kernel variants "
-                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+                "sinks/window/softcap/D!=128 CDNA4 paged-prefill kernel "
+                "variants are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 "
+                "to opt in")
         if tiles is None:
             tiles = build_paged_prefill_tiles(seq_starts, seq_hists,
                                               seq_news, q.device)
