@@ -24,7 +24,11 @@ constexpr int BS = 16;       // tokens per KV block (page)
 constexpr int NWAVES = 4;    // waves per workgroup
 constexpr int THREADS = NWAVES * WAVE_SIZE;
 
-template <int D, int GQ, bool FP8>
+// EXT=false compiles the exact validated fast path (no sinks / window /
+// softcap code at all — bit-identical to the r1/r2-measured kernel);
+// EXT=true carries the GPT-OSS/Gemma feature variants and only ever
+// launches when one of those features is requested.
+template <int D, int GQ, bool FP8, bool EXT = false>
 __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,        // [N, Hq, D]
     const unsigned short* __restrict__ q,    // [N, Hq, D]
@@ -95,10 +99,11 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
   }
 
   using KVT = std::conditional_t<FP8, unsigned char, unsigned short>;
-  // sliding window: the single decode query sits at position len-1 and
-  // attends [max(0, len-window), len)
-  const int wstart = (window > 0 && len > window) ? len - window : 0;
-  const int page0 = wstart / BS;
+  // sliding window (EXT): the single decode query sits at position len-1
+  // and attends [max(0, len-window), len)
+  const int wstart =
+      (EXT && window > 0 && len > window) ? len - window : 0;
+  const int page0 = EXT ? wstart / BS : 0;
   const int* bt = block_tables + (long)seq * max_blocks;
   for (int page = page0 + wave; page < npages; page += NWAVES) {
     const long blk = bt[page];
@@ -156,11 +161,16 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
         float pmax = -INFINITY;
 #pragma unroll
         for (int it = 0; it < TB; ++it) {
-          const int tpos = base_tok + it * 4;
-          const bool valid = tpos < len && tpos >= wstart;
-          float dv = dot[it];
-          if (softcap > 0.f) dv = tanhf(dv / softcap) * softcap;
-          dot[it] = valid ? dv : -INFINITY;
+          if constexpr (EXT) {
+            const int tpos = base_tok + it * 4;
+            const bool valid = tpos < len && tpos >= wstart;
+            float dv = dot[it];
+            if (softcap > 0.f) dv = tanhf(dv / softcap) * softcap;
+            dot[it] = valid ? dv : -INFINITY;
+          } else {
+            const bool valid = base_tok + it * 4 < len;
+            dot[it] = valid ? dot[it] : -INFINITY;
+          }
           pmax = fmaxf(pmax, dot[it]);
         }
         if (pmax == -INFINITY) continue;
@@ -238,13 +248,15 @@ __global__ __launch_bounds__(THREADS) void paged_attn_decode_kernel(
 #pragma unroll
     for (int w = 0; w < NWAVES; ++w)
       if (lds_s[w][gq] > 0.f) M = fmaxf(M, lds_m[w][gq]);
-    // GPT-OSS attention sink: a per-head learned logit joins the softmax
-    // DENOMINATOR only (no value contribution)
+    // GPT-OSS attention sink (EXT): a per-head learned logit joins the
+    // softmax DENOMINATOR only (no value contribution)
     float sden = 0.f;
-    if (sinks != nullptr) {
-      const float sk = sinks[h * GQ + gq];
-      M = fmaxf(M, sk);
-      sden = __expf(sk - M);
+    if constexpr (EXT) {
+      if (sinks != nullptr) {
+        const float sk = sinks[h * GQ + gq];
+        M = fmaxf(M, sk);
+        sden = __expf(sk - M);
+      }
     }
     float num = 0.f, den = sden;
 #pragma unroll
@@ -275,16 +287,22 @@ void paged_attn_decode_launch(void* out, const void* q, const void* kc,
   *err_unsupported = 0;
   if (D != 128 && D != 64 && D != 256) { *err_unsupported = 1; return; }
   if (D == 256 && GQ > 4) { *err_unsupported = 1; return; }  // VGPR budget
-#define LAUNCH_D(DD, G, F)                                                     \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, G, F>), grid, block, 0, s,  \
-                     (unsigned short*)out, (const unsigned short*)q, kc, vc,   \
-                     block_tables, seq_lens, Hkv, max_blocks, scale, q_stride, \
-                     sinks, window, softcap)
+  const bool ext = (sinks != nullptr) || window > 0 || softcap > 0.f;
+#define LAUNCH_D(DD, G, F, E)                                                  \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, G, F, E>), grid, block, 0,  \
+                     s, (unsigned short*)out, (const unsigned short*)q, kc,    \
+                     vc, block_tables, seq_lens, Hkv, max_blocks, scale,       \
+                     q_stride, sinks, window, softcap)
 #define LAUNCH_GQ2(G, F)                                                       \
   do {                                                                         \
-    if (D == 128) LAUNCH_D(128, G, F);                                         \
-    else if (D == 64) LAUNCH_D(64, G, F);                                      \
-    else if ((G) <= 4) LAUNCH_D(256, (G) <= 4 ? (G) : 1, F);                   \
+    if (D == 128) {                                                            \
+      if (ext) LAUNCH_D(128, G, F, true);                                      \
+      else LAUNCH_D(128, G, F, false);                                         \
+    } else if (D == 64) {                                                      \
+      LAUNCH_D(64, G, F, true);                                                \
+    } else if ((G) <= 4) {                                                     \
+      LAUNCH_D(256, (G) <= 4 ? (G) : 1, F, true);                              \
+    }                                                                          \
   } while (0)
 #define LAUNCH_GQ(G)                                                          \
   do {                                                                        \

@@ -47,7 +47,9 @@ DEVICE_INLINE int kswz(int kv, int dbyte) {
   return kv * (D * 2) + (dbyte ^ ((kv & 7) << 4));
 }
 
-template <int D>
+// EXT=false is the exact validated fast path (no sinks/window/softcap
+// code); EXT=true carries the GPT-OSS/Gemma variants.
+template <int D, bool EXT = false>
 __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     unsigned short* __restrict__ out,      // [T, Hq, D]
     const unsigned short* __restrict__ q,  // [T, Hq, D]
@@ -162,8 +164,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);  // hide HBM under MFMA
 
     if (kv0 > wave_q_hi) continue;  // fully masked for this wave
-    // sliding window: tile entirely below every row's window -> skip
-    if (window > 0 && kv0 + BK <= wave_q_lo - window + 1) continue;
+    // sliding window (EXT): tile entirely below every row's window
+    if (EXT && window > 0 && kv0 + BK <= wave_q_lo - window + 1) continue;
 
     // ---- S^T = K · Q^T  (NST stiles x QK k-chunks) ----
     f32x4 st[NST];
@@ -190,9 +192,11 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
-        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
         bool ok = (kvpos <= qpos) && (kvpos < len) && (qpos < len);
-        if (window > 0) ok = ok && (kvpos > qpos - window);
+        if constexpr (EXT) {
+          if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+          if (window > 0) ok = ok && (kvpos > qpos - window);
+        }
         x = ok ? x : -INFINITY;
         sv[stile * 4 + r] = x;
         tmax = fmaxf(tmax, x);
@@ -262,12 +266,14 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     const int orow = lg * 4 + r;
     float denom = __shfl(lcol, orow, WAVE_SIZE);
     float onum = 1.f;
-    if (sinks != nullptr) {
-      const float m_row = __shfl(mcol, orow, WAVE_SIZE);
-      const float sk = sinks[qh];
-      const float M2 = fmaxf(m_row, sk);
-      onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
-      denom = denom * onum + __expf(sk - M2);
+    if constexpr (EXT) {
+      if (sinks != nullptr) {
+        const float m_row = __shfl(mcol, orow, WAVE_SIZE);
+        const float sk = sinks[qh];
+        const float M2 = fmaxf(m_row, sk);
+        onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
+        denom = denom * onum + __expf(sk - M2);
+      }
     }
     const int qrow = q0 + wave * 16 + orow;
     if (qrow >= len || denom <= 0.f) continue;
@@ -296,7 +302,7 @@ namespace {
 
 constexpr int PP_BS = 16;  // pool block size (tokens per KV block)
 
-template <int D>
+template <int D, bool EXT = false>
 __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     unsigned short* __restrict__ out,      // [T, Hq, D] (suffix rows)
     const unsigned short* __restrict__ q,  // [T, Hq, D] (suffix rows)
@@ -422,7 +428,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     if (kt + 1 < ntiles_kv) issue_loads(kv0 + BK);
 
     if (kv0 > wave_q_hi_abs) continue;
-    if (window > 0 && kv0 + BK <= wave_q_lo_abs - window + 1) continue;
+    if (EXT && window > 0 && kv0 + BK <= wave_q_lo_abs - window + 1)
+      continue;
 
     f32x4 st[NST];
 #pragma unroll
@@ -448,9 +455,11 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kvpos = kv0 + stile * 16 + lg * 4 + r;
         float x = st[stile][r] * scale;
-        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
         bool ok = (kvpos <= qpos) && (kvpos < len) && qvalid;
-        if (window > 0) ok = ok && (kvpos > qpos - window);
+        if constexpr (EXT) {
+          if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+          if (window > 0) ok = ok && (kvpos > qpos - window);
+        }
         x = ok ? x : -INFINITY;
         sv[stile * 4 + r] = x;
         tmax = fmaxf(tmax, x);
@@ -515,12 +524,14 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_paged_kernel(
     const int orow = lg * 4 + r;
     float denom = __shfl(lcol, orow, WAVE_SIZE);
     float onum = 1.f;
-    if (sinks != nullptr) {
-      const float m_row = __shfl(mcol, orow, WAVE_SIZE);
-      const float sk = sinks[qh];
-      const float M2 = fmaxf(m_row, sk);
-      onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
-      denom = denom * onum + __expf(sk - M2);
+    if constexpr (EXT) {
+      if (sinks != nullptr) {
+        const float m_row = __shfl(mcol, orow, WAVE_SIZE);
+        const float sk = sinks[qh];
+        const float M2 = fmaxf(m_row, sk);
+        onum = (m_row == -INFINITY) ? 0.f : __expf(m_row - M2);
+        denom = denom * onum + __expf(sk - M2);
+      }
     }
     const int qrow = q0 + wave * 16 + orow;
     if (qrow >= nnew || denom <= 0.f) continue;
@@ -548,16 +559,22 @@ void flash_prefill_paged_launch(
     return;
   }
   dim3 grid(ntiles, Hq);
-#define PPG_LAUNCH(DD)                                                        \
-  hipLaunchKernelGGL(flash_prefill_paged_kernel<DD>, grid, dim3(PF_THREADS),  \
-                     0, s, (unsigned short*)out, (const unsigned short*)q,    \
-                     (const unsigned short*)kc, (const unsigned short*)vc,    \
-                     block_tables, tile_qstart, tile_q0, tile_hist, tile_new, \
-                     tile_seq, Hq, Hkv, maxb, scale, qs, sinks, window,       \
-                     softcap)
-  if (D == 128) PPG_LAUNCH(128);
-  else if (D == 64) PPG_LAUNCH(64);
-  else PPG_LAUNCH(256);
+  const bool ext = (sinks != nullptr) || window > 0 || softcap > 0.f;
+#define PPG_LAUNCH(DD, E)                                                     \
+  hipLaunchKernelGGL((flash_prefill_paged_kernel<DD, E>), grid,               \
+                     dim3(PF_THREADS), 0, s, (unsigned short*)out,            \
+                     (const unsigned short*)q, (const unsigned short*)kc,     \
+                     (const unsigned short*)vc, block_tables, tile_qstart,    \
+                     tile_q0, tile_hist, tile_new, tile_seq, Hq, Hkv, maxb,   \
+                     scale, qs, sinks, window, softcap)
+  if (D == 128) {
+    if (ext) PPG_LAUNCH(128, true);
+    else PPG_LAUNCH(128, false);
+  } else if (D == 64) {
+    PPG_LAUNCH(64, true);
+  } else {
+    PPG_LAUNCH(256, true);
+  }
 #undef PPG_LAUNCH
 }
 
@@ -574,15 +591,21 @@ void flash_prefill_launch(void* out, const void* q, const void* k,
     return;
   }
   dim3 grid(ntiles, Hq);
-#define PF_LAUNCH(DD)                                                        \
-  hipLaunchKernelGGL(flash_prefill_kernel<DD>, grid, dim3(PF_THREADS), 0, s, \
-                     (unsigned short*)out, (const unsigned short*)q,         \
+  const bool ext = (sinks != nullptr) || window > 0 || softcap > 0.f;
+#define PF_LAUNCH(DD, E)                                                     \
+  hipLaunchKernelGGL((flash_prefill_kernel<DD, E>), grid, dim3(PF_THREADS),  \
+                     0, s, (unsigned short*)out, (const unsigned short*)q,   \
                      (const unsigned short*)k, (const unsigned short*)v,     \
                      tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks,  \
                      vs, sinks, window, softcap)
-  if (D == 128) PF_LAUNCH(128);
-  else if (D == 64) PF_LAUNCH(64);
-  else PF_LAUNCH(256);
+  if (D == 128) {
+    if (ext) PF_LAUNCH(128, true);
+    else PF_LAUNCH(128, false);
+  } else if (D == 64) {
+    PF_LAUNCH(64, true);
+  } else {
+    PF_LAUNCH(256, true);
+  }
 #undef PF_LAUNCH
 }
 
