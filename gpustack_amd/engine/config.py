@@ -417,6 +417,17 @@ PRESETS: dict[str, ModelSpec] = {
         sandwich_norms=True, qk_norm=True, embed_scale=5376 ** 0.5,
         attn_scale=168 ** -0.5, mlp_act="gelu_tanh",
     ),
+    # CPU-test preset: Gemma-shaped sandwich-norm layers (tests the
+    # residual-None PP handoff + softcap/GeGLU paths on gloo)
+    "tiny-gemma": ModelSpec(
+        architecture="Gemma2ForCausalLM", vocab_size=512, hidden_size=128,
+        intermediate_size=256, num_layers=4, num_heads=4, num_kv_heads=2,
+        head_dim=32, max_position_embeddings=512, rope_theta=10000.0,
+        rms_norm_eps=1e-6, eos_token_id=1, tie_word_embeddings=True,
+        sliding_window=8, sandwich_norms=True, embed_scale=128 ** 0.5,
+        attn_logit_softcap=50.0, final_logit_softcap=30.0,
+        attn_scale=24 ** -0.5, mlp_act="gelu_tanh",
+    ),
     # CPU-test preset: DeepSeek-shaped MLA + MoE (tests/test_deepseek.py,
     # TP exactness on gloo)
     "tiny-mla": ModelSpec(

@@ -255,3 +255,10 @@ def test_cp_with_pp_rejected():
 
     with pytest.raises(ValueError, match="pipeline"):
         init_parallel(1, 2, 0, cp_size=2)
+
+
+def test_cp2_matches_single_rank_sandwich_model():
+    """Gemma-class layers under CP: the per-layer K/V gather + suffix-
+    style attention must compose with sandwich norms, softcap and
+    sliding windows exactly."""
+    assert _run_cp2("tiny-gemma") == _single_proc_result("tiny-gemma")

@@ -470,3 +470,13 @@ def test_scheduler_pp_partition_proportional():
     ]}, "system_reserved": {}}
     assert PlacementScheduler._pp_partition(
         model_d, spec, Candidate(worker2, [0, 1]), [worker2], []) is None
+
+
+@pytest.mark.timeout(300)
+def test_pp2_sandwich_norm_model_matches_single_rank():
+    """Gemma-class sandwich-norm layers carry the TRUE hidden stream
+    (residual sentinel None) — the PP handoff sends x directly instead of
+    summing x+residual; output must equal single-rank exactly."""
+    expected = _single_proc_result("tiny-gemma")
+    got = _run_pp(2, model="tiny-gemma")
+    assert got == expected, f"{got} != {expected}"
