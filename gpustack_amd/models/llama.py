@@ -336,11 +336,29 @@ class MLAAttention(nn.Module):
         uk = kvb[:, :self.dn]                       # [nh, dn, r]
         uv = kvb[:, self.dn:]                       # [nh, dv, r]
         q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
-        if x.is_cuda:  # gated kernel path (gpu_ok checked above)
-            # Every batch shape runs the absorbed kernel; prefill/suffix
-            # rows run ROW-WISE (len = abs position + 1 through the seq's
-            # block table) — correct, with O(T*L) latent re-reads; the
-            # tiled absorbed-prefill kernel is the r3 follow-up.
+        if x.is_cuda and meta.is_prefill:
+            # gated EXPAND-prefill: prefill is compute-bound, so expand
+            # per-head K/V from the latent (two hipBLASLt einsums) and run
+            # the MFMA flash kernel at D_qk=192 over D_v=128 values — full
+            # matrix-core rate, no O(T*L) re-reads
+            k_full = torch.cat([
+                torch.einsum("tr,hdr->thd", c.float(), uk.float()),
+                k_rot.float().unsqueeze(1).expand(T, self.nh, self.dr),
+            ], dim=-1).to(x.dtype).contiguous()
+            v_full = torch.einsum("tr,hdr->thd", c.float(),
+                                  uv.float()).to(x.dtype).contiguous()
+            q192 = torch.cat([q_nope.float(), q_rot.float()],
+                             dim=-1).to(x.dtype).contiguous()
+            out_pf = torch.empty(T, self.nh, self.dv, dtype=x.dtype,
+                                 device=x.device)
+            ops.varlen_prefill_attn(out_pf, q192, k_full, v_full,
+                                    meta.seq_lens_list, self.scale)
+            o = F.linear(out_pf.reshape(T, self.nh * self.dv), self.o_w)
+            return self.comm.all_reduce(o)
+        if x.is_cuda:  # gated absorbed kernel path (decode/suffix rows)
+            # suffix rows run ROW-WISE (len = abs position + 1 through the
+            # seq's block table) — correct, with O(T*L) latent re-reads;
+            # the tiled absorbed-suffix kernel is the r3 follow-up.
             bt, lens = self._gpu_row_tables(meta, T, x.device)
             q_cat = torch.cat([q_lat.to(x.dtype), q_rot.to(x.dtype)],
                               dim=-1).contiguous()

@@ -65,6 +65,10 @@ def _oss_kernels_enabled() -> bool:
     return os.environ.get("GPUSTACK_AMD_OSS_KERNELS", "0") == "1"
 
 
+def _mla_kernel_enabled() -> bool:
+    return os.environ.get("GPUSTACK_AMD_MLA_KERNEL", "0") == "1"
+
+
 def _sinks_f32(sinks):
     if sinks is not None and (sinks.dtype != torch.float32
                               or not sinks.is_contiguous()):
@@ -189,11 +193,15 @@ def varlen_prefill_attn(out, q, k, v, seq_lens: list[int], scale: float,
                         softcap: float = 0.0) -> None:
     hip = _backend(q)
     if hip is not None:
-        if (sinks is not None or window or softcap or q.shape[-1] != 128) \
+        d_ok = (q.shape[-1] == 128
+                or (q.shape[-1] == 192 and _mla_kernel_enabled()))
+        if (sinks is not None or window or softcap or not d_ok) \
                 and not _oss_kernels_enabled():
             raise NotImplementedError(
                 "sinks/window/softcap/D!=128 CDNA4 prefill kernel variants "
-                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 to opt in")
+                "are unvalidated — set GPUSTACK_AMD_OSS_KERNELS=1 (or "
+                "GPUSTACK_AMD_MLA_KERNEL=1 for the 192/128 MLA expand "
+                "shape) to opt in")
         if tiles is None or tiles[0] is None:
             tiles = build_prefill_tiles(seq_lens, q.device)
         hip.flash_prefill(out, q, k, v, tiles[0], tiles[1], tiles[2], scale,

@@ -48,13 +48,15 @@ DEVICE_INLINE int kswz(int kv, int dbyte) {
 }
 
 // EXT=false is the exact validated fast path (no sinks/window/softcap
-// code); EXT=true carries the GPT-OSS/Gemma variants.
-template <int D, bool EXT = false>
+// code); EXT=true carries the GPT-OSS/Gemma variants. DVK is the VALUE
+// head dim (defaults to D; MLA expand-prefill runs D=192 qk over DVK=128
+// values).
+template <int D, bool EXT = false, int DVK = D>
 __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
-    unsigned short* __restrict__ out,      // [T, Hq, D]
+    unsigned short* __restrict__ out,      // [T, Hq, DVK]
     const unsigned short* __restrict__ q,  // [T, Hq, D]
     const unsigned short* __restrict__ k,  // [T, Hkv, D]
-    const unsigned short* __restrict__ v,  // [T, Hkv, D]
+    const unsigned short* __restrict__ v,  // [T, Hkv, DVK]
     const int* __restrict__ tile_start,    // [ntiles] seq start (global row)
     const int* __restrict__ tile_q0,       // [ntiles] q-tile offset in seq
     const int* __restrict__ tile_len,      // [ntiles] seq length
@@ -63,9 +65,9 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     const float* __restrict__ sinks,       // [Hq] or null
     int window,                            // 0 = full causal
     float softcap) {                       // 0 = off (Gemma-2 tanh cap)
-  constexpr int KU = BK * (D / 8) / PF_THREADS;  // K u16x8 units/thread
-  constexpr int VU = D * (BK / 8) / PF_THREADS;  // VT units/thread
-  constexpr int QK = D / 32;                     // q k-chunks
+  constexpr int KU = BK * (D / 8) / PF_THREADS;    // K u16x8 units/thread
+  constexpr int VU = DVK * (BK / 8) / PF_THREADS;  // VT units/thread
+  constexpr int QK = D / 32;                       // q k-chunks
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
   const int kvh = qh / (Hq / Hkv);
@@ -80,7 +82,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
   const int tid = threadIdx.x;
 
   __shared__ unsigned short Kl[BK * D];            // swizzled
-  __shared__ unsigned short VTl[D][BK + VT_PAD];   // transposed V
+  __shared__ unsigned short VTl[DVK][BK + VT_PAD]; // transposed V
   __shared__ unsigned short Pl[PF_WAVES][16][BK + VT_PAD];
 
   // Hoist this wave's 16 q rows into B-fragments (QK k-chunks of 32).
@@ -96,9 +98,9 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 
   float mcol = -INFINITY;  // running max for q row `lc` (this wave)
   float lcol = 0.f;        // running denom for q row `lc`
-  f32x4 o[D / 16];         // O[q=(lg*4+r)][d=lc+nt*16]
+  f32x4 o[DVK / 16];       // O[q=(lg*4+r)][d=lc+nt*16]
 #pragma unroll
-  for (int nt = 0; nt < D / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int nt = 0; nt < DVK / 16; ++nt) o[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int q_hi = q0 + BQ - 1;
   const int kv_end = min(len, q_hi + 1);           // causal bound
@@ -113,9 +115,9 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
 #pragma unroll
   for (int r = 0; r < KU; ++r) kst_kv[r] = (tid + r * PF_THREADS) / (D / 8);
   const int kst_d0 = (tid % (D / 8)) * 8;
-  const int vst_d = tid % D;
-  const int vst_kvc0 = (tid / D) * 8;   // + r*(8*PF_THREADS/D) per round
-  constexpr int VST_STEP = 8 * PF_THREADS / D;
+  const int vst_d = tid % DVK;
+  const int vst_kvc0 = (tid / DVK) * 8;  // + r*(8*PF_THREADS/DVK) per round
+  constexpr int VST_STEP = 8 * PF_THREADS / DVK;
 
   u16x8 kstage[KU];
   unsigned short vstage[VU][8];
@@ -136,7 +138,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       for (int j = 0; j < 8; ++j) {
         const int kv = kv0 + kvc + j;
         vstage[r][j] = (kv < len)
-            ? v[(long)(seq0 + kv) * vs + (long)kvh * D + vst_d]
+            ? v[(long)(seq0 + kv) * vs + (long)kvh * DVK + vst_d]
             : (unsigned short)0;
       }
     }
@@ -242,7 +244,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       const int orow = lg * 4 + r;
       const float c = __shfl(corr, orow, WAVE_SIZE);
 #pragma unroll
-      for (int nt = 0; nt < D / 16; ++nt) o[nt][r] *= c;
+      for (int nt = 0; nt < DVK / 16; ++nt) o[nt][r] *= c;
     }
 
     // ---- O += P · V  (A = P from LDS, B = V^T rows from LDS) ----
@@ -251,7 +253,7 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
       const u16x8 pa = *reinterpret_cast<const u16x8*>(
           &Pl[wave][lc][kk2 * 32 + lg * 8]);
 #pragma unroll
-      for (int nt = 0; nt < D / 16; ++nt) {
+      for (int nt = 0; nt < DVK / 16; ++nt) {
         const u16x8 b = *reinterpret_cast<const u16x8*>(
             &VTl[nt * 16 + lc][kk2 * 32 + lg * 8]);
         o[nt] = mfma16x16x32_bf16(pa, b, o[nt]);
@@ -279,8 +281,8 @@ __global__ __launch_bounds__(PF_THREADS) void flash_prefill_kernel(
     if (qrow >= len || denom <= 0.f) continue;
     const float inv = onum / denom;
 #pragma unroll
-    for (int nt = 0; nt < D / 16; ++nt) {
-      out[((long)(seq0 + qrow) * Hq + qh) * D + nt * 16 + lc] =
+    for (int nt = 0; nt < DVK / 16; ++nt) {
+      out[((long)(seq0 + qrow) * Hq + qh) * DVK + nt * 16 + lc] =
           f2bf(o[nt][r] * inv);
     }
   }
@@ -581,30 +583,36 @@ void flash_prefill_paged_launch(
 void flash_prefill_launch(void* out, const void* q, const void* k,
                           const void* v, const int* tile_start,
                           const int* tile_q0, const int* tile_len, int ntiles,
-                          int Hq, int Hkv, int D, float scale,
+                          int Hq, int Hkv, int D, int DV, float scale,
                           long qs, long ks, long vs, const float* sinks,
                           int window, float softcap, int* err_unsupported,
                           hipStream_t s) {
   *err_unsupported = 0;
-  if ((D != 128 && D != 64 && D != 256) || Hq % Hkv != 0) {
+  const bool mla_dims = (D == 192 && DV == 128);
+  if (((D != 128 && D != 64 && D != 256) && !mla_dims)
+      || (!mla_dims && DV != D) || Hq % Hkv != 0) {
     *err_unsupported = 1;
     return;
   }
   dim3 grid(ntiles, Hq);
   const bool ext = (sinks != nullptr) || window > 0 || softcap > 0.f;
-#define PF_LAUNCH(DD, E)                                                     \
-  hipLaunchKernelGGL((flash_prefill_kernel<DD, E>), grid, dim3(PF_THREADS),  \
-                     0, s, (unsigned short*)out, (const unsigned short*)q,   \
-                     (const unsigned short*)k, (const unsigned short*)v,     \
-                     tile_start, tile_q0, tile_len, Hq, Hkv, scale, qs, ks,  \
-                     vs, sinks, window, softcap)
-  if (D == 128) {
-    if (ext) PF_LAUNCH(128, true);
-    else PF_LAUNCH(128, false);
+#define PF_LAUNCH(DD, E, DVV)                                                \
+  hipLaunchKernelGGL((flash_prefill_kernel<DD, E, DVV>), grid,               \
+                     dim3(PF_THREADS), 0, s, (unsigned short*)out,           \
+                     (const unsigned short*)q, (const unsigned short*)k,     \
+                     (const unsigned short*)v, tile_start, tile_q0,          \
+                     tile_len, Hq, Hkv, scale, qs, ks, vs, sinks, window,    \
+                     softcap)
+  if (mla_dims) {
+    // MLA expand-prefill (DeepSeek): 192-dim qk over 128-dim values
+    PF_LAUNCH(192, false, 128);
+  } else if (D == 128) {
+    if (ext) PF_LAUNCH(128, true, 128);
+    else PF_LAUNCH(128, false, 128);
   } else if (D == 64) {
-    PF_LAUNCH(64, true);
+    PF_LAUNCH(64, true, 64);
   } else {
-    PF_LAUNCH(256, true);
+    PF_LAUNCH(256, true, 256);
   }
 #undef PF_LAUNCH
 }

@@ -13,7 +13,7 @@ void reshape_and_cache_launch(const void*, const void*, void*, void*, const long
 void greedy_sample_launch(long*, const void*, int, int, hipStream_t);
 void mla_decode_launch(float*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, int*, hipStream_t);
 void paged_attn_decode_launch(void*, const void*, const void*, const void*, const int*, const int*, int, int, int, int, int, float, long, int, const float*, int, float, int*, hipStream_t);
-void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, float, long, long, long, const float*, int, float, int*, hipStream_t);
+void flash_prefill_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, int, int, int, int, float, long, long, long, const float*, int, float, int*, hipStream_t);
 void flash_prefill_paged_launch(void*, const void*, const void*, const void*, const int*, const int*, const int*, const int*, const int*, const int*, int, int, int, int, int, float, long, const float*, int, float, int*, hipStream_t);
 void mfma_probe_launch(float*, const void*, const void*, hipStream_t);
 void skinny_gemm_launch(void*, const void*, const void*, void*, int, int, int, int, hipStream_t);
@@ -211,13 +211,15 @@ void flash_prefill(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(tile_start.scalar_type() == at::kInt && tile_start.is_contiguous());
   const int ntiles = tile_start.size(0);
   const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
+  const int DV = out.size(2);  // value dim (== D except MLA 192/128)
+  TORCH_CHECK(v.size(2) == DV, "v/out value dims must match");
   int err = 0;
   flash_prefill_launch(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        tile_start.data_ptr<int>(), tile_q0.data_ptr<int>(),
-                       tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D,
+                       tile_len.data_ptr<int>(), ntiles, Hq, Hkv, D, DV,
                        (float)scale, qs, ks, vs, sink_ptr_checked(sinks, Hq),
                        (int)window, (float)softcap, &err, cur_stream(q));
-  TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D);
+  TORCH_CHECK(!err, "flash_prefill: unsupported config D=", D, " DV=", DV);
   HIP_CHECK_LAST();
 }
 

@@ -753,3 +753,31 @@ def test_oss_prefill_softcap_d256(D, lens, window, softcap):
     R.varlen_prefill_attn(ref, q, k, v, lens, scale, window=window,
                           softcap=softcap)
     _close(out, ref, atol=4e-2, rtol=4e-2)
+
+
+@mla
+@pytest.mark.parametrize("lens", [[64], [63, 70, 5], [300]])
+def test_mla_expand_prefill_shape(lens):
+    """flash_prefill at the MLA expand shape (D_qk=192 over D_v=128) vs
+    torch_ref (which handles asymmetric v dims natively)."""
+    Hq, Hkv, D, DV = 16, 16, 192, 128
+    T = sum(lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda") / 4
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda") / 4
+    v = torch.randn(T, Hkv, DV, dtype=torch.bfloat16, device="cuda") / 4
+    out = torch.empty(T, Hq, DV, dtype=torch.bfloat16, device="cuda")
+    scale = 1 / math.sqrt(D)
+    ops.varlen_prefill_attn(out, q, k, v, lens, scale)
+    # fp32 reference
+    off = 0
+    for L in lens:
+        qs, ks, vs = (t[off:off + L].float() for t in (q, k, v))
+        for h in range(Hq):
+            att = (qs[:, h] @ ks[:, h].T) * scale
+            mask = torch.triu(torch.ones(L, L, device="cuda"), 1).bool()
+            att = att.masked_fill(mask, float("-inf"))
+            want = torch.softmax(att, -1) @ vs[:, h]
+            got = out[off:off + L, h].float()
+            assert torch.allclose(got, want, atol=4e-2, rtol=4e-2), \
+                (got - want).abs().max()
+        off += L
