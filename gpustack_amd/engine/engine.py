@@ -62,7 +62,8 @@ class LLMEngine:
     ) -> str:
         rid = request_id or f"req-{next(self._counter)}"
         params = params or SamplingParams()
-        if (params.guided_json is not None or params.guided_regex is not None) \
+        if (params.guided_json is not None or params.guided_regex is not None
+                or params.guided_grammar is not None) \
                 and (self.comm.pp_size > 1 or self.comm.cp_size > 1):
             raise ValueError("guided decoding is not supported with pipeline "
                              "or context parallelism (prefill sampling runs "

@@ -704,3 +704,265 @@ class GuidedRegexState:
 
     def commit(self, machine) -> None:
         self.machine = machine
+
+
+# ---------------------------------------------------------------------------
+# guided_grammar: constrained decoding against a context-free grammar
+# (reference surface: vLLM guided_grammar, Lark-style EBNF). First-party
+# incremental EARLEY recognizer over characters: prefix-of-language
+# membership is exactly "the next Earley state set is non-empty", which is
+# the same probe/commit contract the JSON PDA and regex NFA use.
+#
+# Supported subset (covers the dialect's common shapes):
+#   root: alt | alt            rule definitions with `:` or `::=`
+#   "literal"                  quoted terminals (\\ \" \n \t escapes)
+#   [a-z0-9_]                  character-class terminals (ranges + chars)
+#   ( ... )  X* X+ X?          groups and postfix quantifiers
+#   lowercase/UPPER rule refs; the start rule is `root` or `start` (or the
+#   first rule defined)
+# ---------------------------------------------------------------------------
+
+class GrammarError(ValueError):
+    pass
+
+
+def _parse_grammar(text: str):
+    """-> (rules: dict[name, list[list[sym]]], start). sym is
+    ("r", name) | ("c", frozenset_of_chars)."""
+    import re as _re
+
+    rules: dict[str, list] = {}
+    aux = [0]
+
+    def fresh() -> str:
+        aux[0] += 1
+        return f"%aux{aux[0]}"
+
+    def parse_class(body: str) -> frozenset:
+        chars: set[str] = set()
+        i = 0
+        while i < len(body):
+            if body[i] == "\\" and i + 1 < len(body):
+                c = {"n": "\n", "t": "\t", "r": "\r"}.get(body[i + 1],
+                                                          body[i + 1])
+                chars.add(c)
+                i += 2
+            elif i + 2 < len(body) and body[i + 1] == "-":
+                lo, hi = body[i], body[i + 2]
+                chars.update(chr(x) for x in range(ord(lo), ord(hi) + 1))
+                i += 3
+            else:
+                chars.add(body[i])
+                i += 1
+        return frozenset(chars)
+
+    TOK = _re.compile(
+        r'\s*(?:(?P<lit>"(?:\\.|[^"\\])*")'
+        r'|(?P<cls>\[(?:\\.|[^\]\\])*\])'
+        r'|(?P<ref>[A-Za-z_][A-Za-z0-9_]*)'
+        r'|(?P<op>[()|*+?])'
+        r')')
+
+    def parse_alts(toks, pos, name):
+        """toks: list of (kind, value); returns (alts, pos)."""
+        alts, seq = [], []
+        while pos < len(toks):
+            kind, val = toks[pos]
+            if kind == "op" and val == "|":
+                alts.append(seq)
+                seq = []
+                pos += 1
+                continue
+            if kind == "op" and val == ")":
+                break
+            if kind == "op" and val == "(":
+                sub, pos = parse_alts(toks, pos + 1, name)
+                if pos >= len(toks) or toks[pos] != ("op", ")"):
+                    raise GrammarError(f"unclosed group in rule {name!r}")
+                pos += 1
+                g = fresh()
+                rules[g] = sub
+                sym = ("r", g)
+            elif kind == "lit":
+                body = val[1:-1]
+                out = []
+                i = 0
+                while i < len(body):
+                    if body[i] == "\\" and i + 1 < len(body):
+                        c = {"n": "\n", "t": "\t", "r": "\r"}.get(
+                            body[i + 1], body[i + 1])
+                        out.append(("c", frozenset((c,))))
+                        i += 2
+                    else:
+                        out.append(("c", frozenset((body[i],))))
+                        i += 1
+                pos += 1
+                # a multi-char literal is a sequence; quantifiers apply to
+                # the whole literal via an aux rule
+                if pos < len(toks) and toks[pos][0] == "op" \
+                        and toks[pos][1] in "*+?":
+                    g = fresh()
+                    rules[g] = [out]
+                    sym = ("r", g)
+                else:
+                    seq.extend(out)
+                    continue
+            elif kind == "cls":
+                sym = ("c", parse_class(val[1:-1]))
+                pos += 1
+            elif kind == "ref":
+                sym = ("r", val)
+                pos += 1
+            else:
+                raise GrammarError(f"unexpected token {val!r} in {name!r}")
+            # postfix quantifier
+            if pos < len(toks) and toks[pos][0] == "op" \
+                    and toks[pos][1] in "*+?":
+                q = toks[pos][1]
+                pos += 1
+                g = fresh()
+                if q == "*":
+                    rules[g] = [[], [sym, ("r", g)]]
+                elif q == "+":
+                    rules[g] = [[sym], [sym, ("r", g)]]
+                else:
+                    rules[g] = [[], [sym]]
+                sym = ("r", g)
+            seq.append(sym)
+        alts.append(seq)
+        return alts, pos
+
+    # split into rule definitions (a line continues until the next
+    # `name:` at line start)
+    lines = [ln for ln in text.splitlines()
+             if ln.strip() and not ln.strip().startswith(("//", "#"))]
+    defs: list[tuple[str, str]] = []
+    head = _re.compile(r'^\s*([A-Za-z_][A-Za-z0-9_]*)\s*(?:::=|:)\s*(.*)$')
+    for ln in lines:
+        m = head.match(ln)
+        if m and not defs or (m and not ln[:1].isspace()):
+            defs.append((m.group(1), m.group(2)))
+        elif defs:
+            defs[-1] = (defs[-1][0], defs[-1][1] + " " + ln.strip())
+        else:
+            raise GrammarError(f"grammar must start with a rule: {ln!r}")
+    if not defs:
+        raise GrammarError("empty grammar")
+    for name, body in defs:
+        toks = []
+        pos = 0
+        while pos < len(body):
+            m = TOK.match(body, pos)
+            if not m:
+                if body[pos:].strip() == "":
+                    break
+                raise GrammarError(
+                    f"bad grammar syntax near {body[pos:pos + 20]!r}")
+            pos = m.end()
+            for kind in ("lit", "cls", "ref", "op"):
+                if m.group(kind) is not None:
+                    toks.append((kind, m.group(kind)))
+                    break
+        alts, end = parse_alts(toks, 0, name)
+        if end != len(toks):
+            raise GrammarError(f"unbalanced ')' in rule {name!r}")
+        rules.setdefault(name, []).extend(alts)
+    start = ("root" if "root" in rules
+             else "start" if "start" in rules else defs[0][0])
+    # validate refs
+    for name, alts in rules.items():
+        for alt in alts:
+            for kind, val in alt:
+                if kind == "r" and val not in rules:
+                    raise GrammarError(f"undefined rule {val!r} "
+                                       f"(referenced from {name!r})")
+    return rules, start
+
+
+class EarleyM:
+    """Incremental Earley recognizer. Items are (rule, alt_idx, dot,
+    origin); `sets[i]` is the closed state set after i characters. Older
+    sets are immutable once built, so clone() is a shallow list copy —
+    the probe/commit pattern stays cheap."""
+
+    def __init__(self, rules, start):
+        self.rules = rules
+        self.start = start
+        s0 = self._closure({(start, a, 0, 0)
+                            for a in range(len(rules[start]))}, 0, [])
+        self.sets = [s0]
+
+    def _closure(self, items: set, idx: int, sets) -> frozenset:
+        rules = self.rules
+        work = list(items)
+        out = set(items)
+        while work:
+            rule, alt, dot, org = work.pop()
+            body = rules[rule][alt]
+            if dot < len(body):
+                kind, val = body[dot]
+                if kind == "r":  # predict
+                    for a in range(len(rules[val])):
+                        it = (val, a, 0, idx)
+                        if it not in out:
+                            out.add(it)
+                            work.append(it)
+            else:  # complete: advance items waiting on `rule` at origin
+                src = out if org == idx else sets[org]
+                for r2, a2, d2, o2 in list(src):
+                    b2 = rules[r2][a2]
+                    if d2 < len(b2) and b2[d2] == ("r", rule):
+                        it = (r2, a2, d2 + 1, o2)
+                        if it not in out:
+                            out.add(it)
+                            work.append(it)
+        return frozenset(out)
+
+    def advance(self, ch: str) -> bool:
+        idx = len(self.sets)
+        scanned = set()
+        for rule, alt, dot, org in self.sets[-1]:
+            body = self.rules[rule][alt]
+            if dot < len(body):
+                kind, val = body[dot]
+                if kind == "c" and ch in val:
+                    scanned.add((rule, alt, dot + 1, org))
+        if not scanned:
+            return False
+        self.sets.append(self._closure(scanned, idx, self.sets))
+        return True
+
+    @property
+    def complete(self) -> bool:
+        return any(r == self.start and o == 0
+                   and d == len(self.rules[r][a])
+                   for r, a, d, o in self.sets[-1])
+
+    def clone(self) -> "EarleyM":
+        m = EarleyM.__new__(EarleyM)
+        m.rules, m.start = self.rules, self.start
+        m.sets = self.sets[:]  # older sets are immutable
+        return m
+
+
+class GuidedGrammarState:
+    """Per-sequence guided_grammar state (same probe/commit API as the
+    JSON PDA and regex NFA states)."""
+
+    def __init__(self, grammar: str):
+        rules, start = _parse_grammar(grammar)
+        self.machine = EarleyM(rules, start)
+
+    @property
+    def complete(self) -> bool:
+        return self.machine.complete
+
+    def try_advance(self, text: str):
+        m = self.machine.clone()
+        for ch in text:
+            if not m.advance(ch):
+                return None
+        return m
+
+    def commit(self, machine) -> None:
+        self.machine = machine

@@ -43,6 +43,10 @@ class Sampler:
                 from .guided import GuidedRegexState
 
                 seq.guided_sm = GuidedRegexState(p.guided_regex)
+            elif p.guided_grammar is not None:
+                from .guided import GuidedGrammarState
+
+                seq.guided_sm = GuidedGrammarState(p.guided_grammar)
             else:
                 schema = p.guided_json if isinstance(p.guided_json, dict) else None
                 seq.guided_sm = GuidedJsonState(schema)
@@ -155,7 +159,8 @@ class Sampler:
         out: list[int] = [0] * len(seqs)
         guided = {i for i, s in enumerate(seqs)
                   if s.params.guided_json is not None
-                  or s.params.guided_regex is not None}
+                  or s.params.guided_regex is not None
+                  or s.params.guided_grammar is not None}
         greedy_idx = [i for i, s in enumerate(seqs)
                       if s.params.greedy and not s.params.needs_logit_processing]
         proc_greedy = [i for i, s in enumerate(seqs)

@@ -30,6 +30,7 @@ class SamplingParams:
     # JSON; a dict = JSON-Schema subset compiled to a template machine
     guided_json: "bool | dict | None" = None
     guided_regex: str | None = None  # regex-subset NFA (engine/guided.py)
+    guided_grammar: str | None = None  # EBNF CFG, Earley (engine/guided.py)
     eos_token_id: int = 0           # used by guided decoding to terminate
     max_tokens: int = 128
     min_tokens: int = 0        # suppress EOS/stop until this many tokens
@@ -48,6 +49,7 @@ class SamplingParams:
         return bool(self.logit_bias) or bool(self.guided_token_seqs) \
             or self.guided_json is not None \
             or self.guided_regex is not None \
+            or self.guided_grammar is not None \
             or self.presence_penalty != 0.0 \
             or self.frequency_penalty != 0.0 \
             or self.repetition_penalty != 1.0 \
@@ -58,7 +60,8 @@ class SamplingParams:
         """Speculative drafts verify greedily row by row; stateful guided
         decoding cannot validate draft rows, so such seqs decode plain."""
         return (self.greedy and self.guided_json is None
-                and self.guided_regex is None)
+                and self.guided_regex is None
+                and self.guided_grammar is None)
 
     @property
     def greedy(self) -> bool:

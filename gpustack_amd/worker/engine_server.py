@@ -275,9 +275,11 @@ def _sampling_params(body: dict, eos_token_id: int, tokenizer=None):
         guided_token_seqs=guided,
         guided_json=gj,
         guided_regex=body.get("guided_regex"),
+        guided_grammar=body.get("guided_grammar"),
         eos_token_id=eos_token_id,
     )
-    if gj is not None or sp.guided_regex is not None:
+    if gj is not None or sp.guided_regex is not None \
+            or sp.guided_grammar is not None:
         sp.ignore_eos = False
     if guided:
         sp.ignore_eos = False
@@ -496,7 +498,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         params = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                   runner.tokenizer)
         _apply_lora_routing(params, body, runner)
-        if params.guided_json is not None or params.guided_regex is not None:
+        if params.guided_json is not None or params.guided_regex is not None \
+                or params.guided_grammar is not None:
             runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         rid, q = runner.submit(prompt_ids, params)
@@ -746,7 +749,8 @@ def create_app(runner: EngineRunner) -> FastAPI:
         base = _sampling_params(body, runner.engine.cfg.spec.eos_token_id,
                                 runner.tokenizer)
         _apply_lora_routing(base, body, runner)
-        if base.guided_json is not None or base.guided_regex is not None:
+        if base.guided_json is not None or base.guided_regex is not None \
+                or base.guided_grammar is not None:
             runner.ensure_token_table()
         stop_strs = _stop_strings(body)
         subs = []
