@@ -888,9 +888,27 @@ class EarleyM:
     def __init__(self, rules, start):
         self.rules = rules
         self.start = start
+        self.nullable = self._nullable(rules)
         s0 = self._closure({(start, a, 0, 0)
                             for a in range(len(rules[start]))}, 0, [])
         self.sets = [s0]
+
+    @staticmethod
+    def _nullable(rules) -> frozenset:
+        """Rules that derive the empty string (fixed point)."""
+        null = set()
+        changed = True
+        while changed:
+            changed = False
+            for name, alts in rules.items():
+                if name in null:
+                    continue
+                for alt in alts:
+                    if all(k == "r" and v in null for k, v in alt):
+                        null.add(name)
+                        changed = True
+                        break
+        return frozenset(null)
 
     def _closure(self, items: set, idx: int, sets) -> frozenset:
         rules = self.rules
@@ -904,6 +922,14 @@ class EarleyM:
                 if kind == "r":  # predict
                     for a in range(len(rules[val])):
                         it = (val, a, 0, idx)
+                        if it not in out:
+                            out.add(it)
+                            work.append(it)
+                    if val in self.nullable:
+                        # Aycock-Horspool: a nullable nonterminal may
+                        # complete within this set regardless of item
+                        # processing order — advance over it eagerly
+                        it = (rule, alt, dot + 1, org)
                         if it not in out:
                             out.add(it)
                             work.append(it)
@@ -940,7 +966,7 @@ class EarleyM:
 
     def clone(self) -> "EarleyM":
         m = EarleyM.__new__(EarleyM)
-        m.rules, m.start = self.rules, self.start
+        m.rules, m.start, m.nullable = self.rules, self.start, self.nullable
         m.sets = self.sets[:]  # older sets are immutable
         return m
 

@@ -505,3 +505,21 @@ def test_guided_grammar_sampled_generation():
     toks = eng.generate([[4, 4]], p)[0]
     text = "".join(table[t] for t in toks if t != 1)
     assert text in ("yes", "no", "maybe"), text
+
+
+def test_grammar_nullable_chains():
+    """Aycock-Horspool nullable handling: nonterminals deriving empty must
+    complete regardless of item processing order."""
+    from gpustack_amd.engine.guided import GuidedGrammarState
+
+    g = GuidedGrammarState('''
+root: a "x"
+a: b b
+b: ""
+''')
+    assert g.try_advance("x").complete
+    g2 = GuidedGrammarState('root: a a "y"\na: "" | "w"')
+    assert g2.try_advance("y").complete
+    assert g2.try_advance("wy").complete
+    assert g2.try_advance("wwy").complete
+    assert g2.try_advance("wwwy") is None
