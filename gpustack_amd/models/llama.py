@@ -335,7 +335,6 @@ class MLAAttention(nn.Module):
         kvb = self.kv_b_w.view(self.nh, self.dn + self.dv, self.r)
         uk = kvb[:, :self.dn]                       # [nh, dn, r]
         uv = kvb[:, self.dn:]                       # [nh, dv, r]
-        q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
         if x.is_cuda and meta.is_prefill:
             # gated EXPAND-prefill: prefill is compute-bound, so expand
             # per-head K/V from the latent (two hipBLASLt einsums) and run
@@ -355,6 +354,9 @@ class MLAAttention(nn.Module):
                                     meta.seq_lens_list, self.scale)
             o = F.linear(out_pf.reshape(T, self.nh * self.dv), self.o_w)
             return self.comm.all_reduce(o)
+        # absorbed projection of q (decode/suffix/CPU paths only — the
+        # expand-prefill branch above never needs it)
+        q_lat = torch.einsum("thd,hdr->thr", q_nope.float(), uk.float())
         if x.is_cuda:  # gated absorbed kernel path (decode/suffix rows)
             # suffix rows run ROW-WISE (len = abs position + 1 through the
             # seq's block table) — correct, with O(T*L) latent re-reads;
