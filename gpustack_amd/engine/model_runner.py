@@ -664,6 +664,45 @@ class ModelRunner:
         return tokens, meta
 
     @torch.inference_mode()
+    def score_tokens(self, prompts: list[list[int]],
+                     token_ids: list[int]) -> list[list[float]]:
+        """Last-token logits restricted to `token_ids`, per prompt — the
+        generative-reranker primitive (Qwen3-Reranker class models score
+        relevance as P("yes") vs P("no") at the judgment position).
+        KV-free prefill like embed()."""
+        dev = self.device
+        lens = [min(len(p), self.cfg.max_model_len) for p in prompts]
+        flat: list[int] = []
+        positions: list[int] = []
+        for p, L in zip(prompts, lens):
+            flat.extend(p[:L])
+            positions.extend(range(L))
+        tokens = torch.tensor(flat, dtype=torch.long, device=dev)
+        pos = torch.tensor(positions, dtype=torch.long, device=dev)
+        slots = torch.full((len(flat),), -1, dtype=torch.long, device=dev)
+        tiles = ops.build_prefill_tiles(lens, dev)
+        idx, off = [], 0
+        for L in lens:
+            idx.append(off + L - 1)
+            off += L
+        meta = ForwardMeta(
+            is_prefill=True, positions=pos, slot_mapping=slots,
+            logits_indices=torch.tensor(idx, dtype=torch.long, device=dev),
+            seq_lens_list=lens,
+            tile_start=tiles[0], tile_q0=tiles[1], tile_len=tiles[2],
+        )
+        logits = self.model(tokens, meta, self.kv)
+        if self.comm.pp_size > 1:
+            if logits is None:
+                logits = torch.empty(len(prompts),
+                                     self.cfg.spec.vocab_size,
+                                     dtype=getattr(torch, self.cfg.dtype),
+                                     device=dev)
+            self.comm.broadcast_world(logits, src=self.comm.last_stage_rank)
+        sel = logits.float()[:, token_ids]
+        return sel.tolist()
+
+    @torch.inference_mode()
     def embed(self, prompts: list[list[int]], pooling: str = "last") -> list[list[float]]:
         """Embedding forward (reference category "embedding",
         schemas/models.py:51): prefill with KV writes disabled

@@ -1083,6 +1083,33 @@ def create_app(runner: EngineRunner) -> FastAPI:
                 ids = ids.ids
             return list(ids) or [0]
 
+        if body.get("mode") == "generative" or "instruction" in body:
+            # Qwen3-Reranker-style scoring: the model judges each
+            # (instruction, query, doc) prompt and relevance is
+            # P(yes | prompt) from the last-token logits
+            import math as _math
+
+            instr = body.get("instruction") or (
+                "Judge whether the Document meets the requirements based "
+                "on the Query. Answer only yes or no.")
+            yes_id = enc(body.get("yes_token", "yes"))[-1]
+            no_id = enc(body.get("no_token", "no"))[-1]
+            prompts = [
+                enc(f"{instr}\nQuery: {query}\nDocument: {d}\nAnswer:")
+                for d in docs
+            ]
+            pairs = await runner.run_aux(
+                runner.engine.runner.score_tokens, prompts, [yes_id, no_id])
+            scored = [{"index": i,
+                       "relevance_score":
+                           1.0 / (1.0 + _math.exp(-(y - n))),
+                       "document": {"text": docs[i]}}
+                      for i, (y, n) in enumerate(pairs)]
+            scored.sort(key=lambda r: -r["relevance_score"])
+            ntok = sum(len(p) for p in prompts)
+            return {"model": runner.served_name, "results": scored[:top_n],
+                    "usage": {"prompt_tokens": ntok, "total_tokens": ntok}}
+
         prompts = [enc(query)] + [enc(d) for d in docs]
         vecs = await runner.run_aux(runner.engine.runner.embed, prompts, "mean")
         import math

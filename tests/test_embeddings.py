@@ -42,3 +42,54 @@ def test_embed_does_not_disturb_generation():
     assert before == after
     # KV pool untouched (slot -1 writes skipped)
     assert eng.scheduler.kv.allocator.num_free == eng.scheduler.kv.allocator.num_blocks
+
+
+def test_generative_rerank_mode():
+    """Qwen3-Reranker-style scoring on a generative engine: relevance =
+    sigmoid(logit_yes - logit_no) at the judgment position; results are
+    sorted, bounded to (0,1), and deterministic."""
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-gr", "--source", "preset",
+        "--model-ref", "tiny", "--port", str(port), "--device", "cpu",
+        "--kv-cache-blocks", "64", "--max-model-len", "256",
+    ])
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 90:
+            if proc.poll() is not None:
+                raise AssertionError(f"server exited {proc.returncode}")
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health",
+                             timeout=2).status_code == 200:
+                    break
+            except httpx.HTTPError:
+                time.sleep(0.5)
+        docs = ["alpha beta", "gamma", "delta epsilon zeta"]
+        r = httpx.post(f"http://127.0.0.1:{port}/v1/rerank", json={
+            "model": "tiny-gr", "query": "alpha", "documents": docs,
+            "mode": "generative"}, timeout=60)
+        assert r.status_code == 200, r.text
+        res = r.json()["results"]
+        assert len(res) == 3
+        assert all(0.0 < x["relevance_score"] < 1.0 for x in res)
+        scores = [x["relevance_score"] for x in res]
+        assert scores == sorted(scores, reverse=True)
+        r2 = httpx.post(f"http://127.0.0.1:{port}/v1/rerank", json={
+            "model": "tiny-gr", "query": "alpha", "documents": docs,
+            "mode": "generative"}, timeout=60)
+        assert r2.json()["results"] == res  # deterministic
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
