@@ -489,3 +489,33 @@ def test_deploy_reranker_model(cluster, tmp_path_factory):
     assert len(res) == 2
     assert res[0]["relevance_score"] >= res[1]["relevance_score"]
     client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'bge-e2e'][0]['id']}")
+
+
+@pytest.mark.timeout(240)
+def test_deploy_gemma_model(cluster):
+    """Gemma-2-class model (sandwich norms, softcapping, GeGLU) served
+    end-to-end through the cluster on the CPU oracle path."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-g2-e2e", "source": "preset", "model_ref": "tiny-gemma",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201, r.text
+    state = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-g2-e2e"]
+        if insts:
+            state = insts[0]["state"]
+            if state == "running":
+                break
+            assert state != "error", insts[0]["state_message"]
+        time.sleep(0.5)
+    assert state == "running", f"instance never ran (last state: {state})"
+    r = client.post("/v1/completions", json={
+        "model": "tiny-g2-e2e", "prompt": "abc", "max_tokens": 5,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["usage"]["completion_tokens"] == 5
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-g2-e2e'][0]['id']}")
