@@ -133,14 +133,39 @@ def create_token(_: User = Depends(get_admin_user)):
 
 # ---- workers -------------------------------------------------------------
 
+def _paginate(rows: list, page: int | None, per_page: int | None,
+              search: str | None = None, search_key: str = "name") -> dict:
+    """Reference-style list envelope: optional `search` substring filter on
+    `search_key` plus `page`/`perPage` windowing with a pagination block.
+    Without params the full item list returns unchanged (existing
+    clients/watchers rely on that)."""
+    if search:
+        needle = search.lower()
+        rows = [r for r in rows
+                if needle in str(r.get(search_key, "")).lower()]
+    total = len(rows)
+    if page is None and per_page is None:
+        return {"items": rows}
+    per_page = max(1, per_page or 100)
+    page = max(1, page or 1)
+    lo = (page - 1) * per_page
+    return {"items": rows[lo:lo + per_page],
+            "pagination": {"page": page, "perPage": per_page,
+                           "total": total,
+                           "totalPage": (total + per_page - 1) // per_page}}
+
+
 @router.get("/workers")
 def list_workers(request: Request, watch: bool = Query(False),
+                 page: int | None = Query(None),
+                 perPage: int | None = Query(None),
+                 search: str | None = Query(None),
                  user: User = Depends(get_current_user)):
     with get_session() as s:
         rows = [w.to_dict() for w in s.query(Worker).all()]
     if watch:
         return _watch_stream("workers", rows, None)
-    return {"items": rows}
+    return _paginate(rows, page, perPage, search)
 
 
 def _token_cluster_id(request: Request) -> int | None:
@@ -273,15 +298,23 @@ def delete_worker(worker_id: int, _: User = Depends(get_admin_user)):
 # ---- models ----------------------------------------------------------------
 
 @router.get("/models")
-def list_models(watch: bool = Query(False), user: User = Depends(get_current_user)):
+def list_models(watch: bool = Query(False),
+                page: int | None = Query(None),
+                perPage: int | None = Query(None),
+                search: str | None = Query(None),
+                categories: str | None = Query(None),
+                user: User = Depends(get_current_user)):
     from .deps import model_allowed_for_user
 
     with get_session() as s:
         rows = [m.to_dict() for m in s.query(Model).all()
                 if model_allowed_for_user(user, m)]
+    if categories:
+        want = set(categories.split(","))
+        rows = [r for r in rows if want & set(r.get("categories") or [])]
     if watch:
         return _watch_stream("models", rows, None)
-    return {"items": rows}
+    return _paginate(rows, page, perPage, search)
 
 
 @router.post("/models", status_code=201)

@@ -518,3 +518,28 @@ def test_registration_token_persisted_to_data_dir(server):
     p = pathlib.Path(cfg.data_dir) / "token"
     assert p.read_text().strip() == reg_token
     assert (p.stat().st_mode & 0o777) == 0o600
+
+
+def test_list_pagination_and_filters(server):
+    """Reference-style list params: page/perPage envelope, substring
+    search, category filter — plain lists stay unchanged for existing
+    clients."""
+    client, app, cfg, _tok = server
+    for i in range(5):
+        r = client.post("/v2/models", json={
+            "name": f"pag-{i}", "source": "preset", "model_ref": "tiny"})
+        assert r.status_code == 201
+    r = client.get("/v2/models")
+    assert "pagination" not in r.json()  # no params: legacy shape
+    r = client.get("/v2/models", params={"page": 2, "perPage": 2,
+                                         "search": "pag-"})
+    body = r.json()
+    assert [m["name"] for m in body["items"]] == ["pag-2", "pag-3"]
+    assert body["pagination"]["total"] == 5
+    assert body["pagination"]["totalPage"] == 3
+    r = client.get("/v2/models", params={"search": "pag-4"})
+    assert [m["name"] for m in r.json()["items"]] == ["pag-4"]
+    r = client.get("/v2/models", params={"categories": "reranker"})
+    assert r.json()["items"] == []
+    r = client.get("/v2/workers", params={"page": 1, "perPage": 10})
+    assert "pagination" in r.json()
