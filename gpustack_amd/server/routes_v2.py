@@ -900,7 +900,64 @@ def version():
     return {"version": __version__}
 
 
-# ---- usage / system load ---------------------------------------------------
+# ---- dashboard / usage / system load ---------------------------------------
+
+@router.get("/dashboard")
+def dashboard(_: User = Depends(get_current_user)):
+    """Aggregate operator view (reference: routes/dashboard.py):
+    resource counts, current + recent system load, and a 7-day model
+    usage summary with per-model totals."""
+    import datetime as _dt
+
+    from ..schemas import GPUInstance, ModelInstanceState
+
+    with get_session() as s:
+        workers = s.query(Worker).all()
+        gpu_count = sum(len(w.gpu_devices or []) for w in workers)
+        vram_total = sum((g.get("memory") or {}).get("total", 0)
+                         for w in workers for g in (w.gpu_devices or []))
+        models = s.query(Model).count()
+        insts = s.query(ModelInstance).all()
+        counts = {
+            "workers": len(workers),
+            "gpus": gpu_count,
+            "vram_total_bytes": vram_total,
+            "models": models,
+            "model_instances": len(insts),
+            "running_instances": sum(
+                1 for i in insts
+                if i.state == ModelInstanceState.RUNNING.value),
+            "gpu_instances": s.query(GPUInstance).count(),
+        }
+        loads = (s.query(SystemLoad)
+                 .order_by(SystemLoad.timestamp.desc()).limit(60).all())
+        current = loads[0].to_dict() if loads else None
+        history = [r.to_dict() for r in reversed(loads)]
+        since = (_dt.date.today()
+                 - _dt.timedelta(days=6)).strftime("%Y-%m-%d")
+        rows = s.query(ModelUsage).filter(ModelUsage.date >= since).all()
+        per_model: dict[str, dict] = {}
+        totals = {"prompt_tokens": 0, "completion_tokens": 0,
+                  "request_count": 0}
+        for u in rows:
+            m = per_model.setdefault(u.model_name, {
+                "model_name": u.model_name, "prompt_tokens": 0,
+                "completion_tokens": 0, "request_count": 0})
+            for k in totals:
+                v = getattr(u, k) or 0
+                m[k] += v
+                totals[k] += v
+        top = sorted(per_model.values(),
+                     key=lambda m: -(m["prompt_tokens"]
+                                     + m["completion_tokens"]))[:10]
+    return {
+        "resource_counts": counts,
+        "system_load": {"current": current, "history": history},
+        "model_usage": {"since": since, "totals": totals,
+                        "top_models": top},
+    }
+
+
 
 @router.get("/usage")
 def usage(user: User = Depends(get_current_user)):

@@ -543,3 +543,28 @@ def test_list_pagination_and_filters(server):
     assert r.json()["items"] == []
     r = client.get("/v2/workers", params={"page": 1, "perPage": 10})
     assert "pagination" in r.json()
+
+
+def test_dashboard(server):
+    """Aggregate dashboard: resource counts, load, 7-day usage summary."""
+    import datetime
+
+    client, app, cfg, _tok = server
+    from gpustack_amd.db import ar_create, get_session
+    from gpustack_amd.schemas import ModelUsage
+
+    today = datetime.date.today().strftime("%Y-%m-%d")
+    with get_session() as s:
+        ar_create(s, ModelUsage(user_id=1, model_id=1, model_name="m1",
+                                date=today, prompt_tokens=100,
+                                completion_tokens=50, request_count=3))
+        ar_create(s, ModelUsage(user_id=1, model_id=2, model_name="m2",
+                                date=today, prompt_tokens=10,
+                                completion_tokens=5, request_count=1))
+    r = client.get("/v2/dashboard")
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["resource_counts"]["models"] >= 0
+    assert body["model_usage"]["totals"]["prompt_tokens"] == 110
+    assert body["model_usage"]["top_models"][0]["model_name"] == "m1"
+    assert "system_load" in body
