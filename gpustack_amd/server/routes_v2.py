@@ -900,6 +900,53 @@ def version():
     return {"version": __version__}
 
 
+@router.get("/gpu_devices")
+def list_gpu_devices(page: int | None = Query(None),
+                     perPage: int | None = Query(None),
+                     search: str | None = Query(None),
+                     _: User = Depends(get_current_user)):
+    """Flattened GPU inventory across workers (reference:
+    routes/gpu_devices.py): device info + per-GPU allocatable VRAM from
+    the same claim accounting the scheduler places against."""
+    from ..scheduler.policies import worker_allocatable
+
+    with get_session() as s:
+        workers = [w.to_dict() for w in s.query(Worker).all()]
+        insts = [i.to_dict() for i in s.query(ModelInstance).all()]
+    rows = []
+    for w in workers:
+        alloc = worker_allocatable(w, insts)
+        gpus = ((w.get("status") or {}).get("gpu_devices")
+                or w.get("gpu_devices") or [])
+        for g in gpus:
+            idx = g.get("index", 0)
+            rows.append({
+                "id": f"{w['name']}:{idx}",
+                "worker_id": w["id"], "worker_name": w["name"],
+                "index": idx, "name": g.get("name", ""),
+                "type": g.get("type", "rocm"),
+                "memory": g.get("memory") or {},
+                "allocatable_vram": alloc.get(idx, 0),
+                "compute_partition": g.get("compute_partition"),
+                "memory_partition": g.get("memory_partition"),
+            })
+    return _paginate(rows, page, perPage, search)
+
+
+@router.get("/config")
+def server_config(request: Request, _: User = Depends(get_admin_user)):
+    """Sanitized effective server config (reference routes/config.py):
+    secrets and tokens are withheld."""
+    cfg = request.app.state.config
+    hide = {"token", "jwt_secret", "bootstrap_password", "database_url"}
+    out = {}
+    for k, v in vars(cfg).items():
+        if k.startswith("_"):
+            continue
+        out[k] = "***" if (k in hide and v) else v
+    return out
+
+
 # ---- dashboard / usage / system load ---------------------------------------
 
 @router.get("/dashboard")

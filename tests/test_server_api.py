@@ -568,3 +568,25 @@ def test_dashboard(server):
     assert body["model_usage"]["totals"]["prompt_tokens"] == 110
     assert body["model_usage"]["top_models"][0]["model_name"] == "m1"
     assert "system_load" in body
+
+
+def test_gpu_devices_and_config(server):
+    client, app, cfg, reg_token = server
+    r = _register_worker(client, reg_token, name="gpuw")
+    assert r.status_code in (200, 201), r.text
+    r = client.get("/v2/gpu_devices")
+    items = r.json()["items"]
+    mine = [g for g in items if g["worker_name"] == "gpuw"]
+    assert len(mine) >= 1
+    assert mine[0]["allocatable_vram"] > 0
+    assert mine[0]["id"] == "gpuw:0"
+    assert mine[0]["memory"].get("total", 0) > 0
+    r = client.get("/v2/gpu_devices", params={"search": "MI355",
+                                              "page": 1, "perPage": 1})
+    assert len(r.json()["items"]) == 1
+
+    r = client.get("/v2/config")
+    assert r.status_code == 200
+    body = r.json()
+    assert body.get("bootstrap_password") == "***"
+    assert "data_dir" in body
