@@ -84,6 +84,12 @@ def _m011_gpu_instances(conn):
     GPUInstance.__table__.create(conn, checkfirst=True)
 
 
+def _m012_gpu_instance_templates(conn):
+    from ..schemas.tables import GPUInstanceTemplate, SSHPublicKey
+    GPUInstanceTemplate.__table__.create(conn, checkfirst=True)
+    SSHPublicKey.__table__.create(conn, checkfirst=True)
+
+
 MIGRATIONS: list[tuple[int, str, object]] = [
     (1, "worker.proxy_mode for tunnel workers", _m001_worker_proxy_mode),
     (2, "model KV/speculative/scaling columns", _m002_model_kv_features),
@@ -96,6 +102,7 @@ MIGRATIONS: list[tuple[int, str, object]] = [
     (9, "model.gpu_type_selector for device-class placement", _m009_gpu_type_selector),
     (10, "resource-event metering pair (hot + archive)", _m010_resource_events),
     (11, "gpu_instances table (operator-analog SSH GPU pods)", _m011_gpu_instances),
+    (12, "gpu-instance templates + ssh public keys", _m012_gpu_instance_templates),
 ]
 
 HEAD = MIGRATIONS[-1][0] if MIGRATIONS else 0

@@ -6,7 +6,7 @@ from pydantic import BaseModel, Field
 from .tables import (  # noqa: F401
     ApiKey, Cluster,
     Benchmark,
-    GPUInstance, GPUInstanceState,
+    GPUInstance, GPUInstanceState, GPUInstanceTemplate, SSHPublicKey,
     Model, Org,
     ModelFile,
     ModelInstance,
@@ -181,10 +181,27 @@ class GPUInstanceCreate(BaseModel):
     flavor: str = "mi355x-1gpu"
     image: str = "rocm/dev-ubuntu-24.04"
     ssh_public_key: str = ""
+    ssh_key_name: str | None = None     # reference a stored SSHPublicKey
+    template: str | None = None         # GPUInstanceTemplate to start from
     provider: str = "k8s"
     provider_config: dict = Field(default_factory=dict)
     volumes: list[dict] = Field(default_factory=list)
     labels: dict = Field(default_factory=dict)
+
+
+class GPUInstanceTemplateCreate(BaseModel):
+    name: str
+    flavor: str = "mi355x-1gpu"
+    image: str = "rocm/dev-ubuntu-24.04"
+    volumes: list[dict] = Field(default_factory=list)
+    labels: dict = Field(default_factory=dict)
+    provider: str = "k8s"
+    provider_config: dict = Field(default_factory=dict)
+
+
+class SSHPublicKeyCreate(BaseModel):
+    name: str
+    public_key: str
 
 
 class ModelProviderCreate(BaseModel):

@@ -284,6 +284,30 @@ class GPUInstance(Base, TimestampMixin, SerializeMixin):
     ssh_port = Column(Integer, default=0)
 
 
+class GPUInstanceTemplate(Base, TimestampMixin, SerializeMixin):
+    """Named GPU-instance preset (reference: gpu_instance_templates):
+    flavor/image/volumes/labels captured once, instantiated many times."""
+    __tablename__ = "gpu_instance_templates"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    flavor = Column(String(128), default="mi355x-1gpu")
+    image = Column(String(512), default="rocm/dev-ubuntu-24.04")
+    volumes = Column(JSON, default=list)
+    labels = Column(JSON, default=dict)
+    provider = Column(String(64), default="k8s")
+    provider_config = Column(JSON, default=dict)
+
+
+class SSHPublicKey(Base, TimestampMixin, SerializeMixin):
+    """Reusable SSH public keys for GPU instances (reference:
+    gpu_instance_ssh_public_keys)."""
+    __tablename__ = "ssh_public_keys"
+    id = Column(Integer, primary_key=True)
+    name = Column(String(256), unique=True, nullable=False, index=True)
+    user_id = Column(Integer, index=True)
+    public_key = Column(Text, nullable=False)
+
+
 class Benchmark(Base, TimestampMixin, SerializeMixin):
     """In-product benchmark runs (reference: schemas/benchmark.py)."""
     __tablename__ = "benchmarks"

@@ -20,7 +20,7 @@ from ..schemas import (
     ModelInstance, ModelInstanceState, ModelInstanceUpdate, ModelProvider,
     ModelProviderCreate, ModelRoute, ModelRouteCreate, ModelUpdate,
     ModelUsage, RegistrationToken, SystemLoad, User, UserCreate, Worker,
-    GPUInstanceCreate,
+    GPUInstanceCreate, GPUInstanceTemplateCreate, SSHPublicKeyCreate,
     WorkerPoolCreate, WorkerPoolUpdate, WorkerRegister, WorkerState,
     WorkerStatusUpdate,
 )
@@ -569,19 +569,42 @@ def create_gpu_instance(body: GPUInstanceCreate,
     from ..schemas import GPUInstance
     from .gpu_instances import FLAVORS, PROVIDERS
 
-    if body.provider not in PROVIDERS:
-        raise HTTPException(400, f"unknown provider {body.provider!r}")
-    if body.flavor not in FLAVORS:
-        raise HTTPException(400, f"unknown flavor {body.flavor!r}; "
-                                 f"one of {sorted(FLAVORS)}")
+    from ..schemas import GPUInstanceTemplate, SSHPublicKey
+
+    fields = {"flavor": body.flavor, "image": body.image,
+              "volumes": body.volumes, "labels": body.labels,
+              "provider": body.provider,
+              "provider_config": body.provider_config}
     with get_session() as s:
+        if body.template:
+            t = (s.query(GPUInstanceTemplate)
+                 .filter_by(name=body.template).first())
+            if not t:
+                raise HTTPException(400,
+                                    f"unknown template {body.template!r}")
+            # caller overrides beat template values; fields left at
+            # their schema defaults inherit from the template
+            set_fields = body.model_fields_set
+            for f in fields:
+                if f not in set_fields:
+                    fields[f] = getattr(t, f)
+        key = body.ssh_public_key
+        if body.ssh_key_name:
+            k = (s.query(SSHPublicKey)
+                 .filter_by(name=body.ssh_key_name).first())
+            if not k:
+                raise HTTPException(400,
+                                    f"unknown ssh key {body.ssh_key_name!r}")
+            key = k.public_key
+        if fields["provider"] not in PROVIDERS:
+            raise HTTPException(400,
+                                f"unknown provider {fields['provider']!r}")
+        if fields["flavor"] not in FLAVORS:
+            raise HTTPException(400, f"unknown flavor {fields['flavor']!r}; "
+                                     f"one of {sorted(FLAVORS)}")
         if s.query(GPUInstance).filter_by(name=body.name).first():
             raise HTTPException(409, "gpu instance exists")
-        g = GPUInstance(name=body.name, flavor=body.flavor, image=body.image,
-                        ssh_public_key=body.ssh_public_key,
-                        provider=body.provider,
-                        provider_config=body.provider_config,
-                        volumes=body.volumes, labels=body.labels)
+        g = GPUInstance(name=body.name, ssh_public_key=key, **fields)
         ar_create(s, g)
         return g.to_dict()
 
@@ -599,6 +622,86 @@ def delete_gpu_instance(gid: int, _: User = Depends(get_admin_user)):
         g.state = GPUInstanceState.DELETING.value
         ar_update(s, g)
     return {"status": "deleting"}
+
+
+# -- templates / ssh keys (reference: gpu_instance_templates,
+# gpu_instance_ssh_public_keys) --
+
+@router.get("/gpu_instance_templates")
+def list_gpu_instance_templates(_: User = Depends(get_current_user)):
+    from ..schemas import GPUInstanceTemplate
+
+    with get_session() as s:
+        return {"items": [t.to_dict()
+                          for t in s.query(GPUInstanceTemplate).all()]}
+
+
+@router.post("/gpu_instance_templates", status_code=201)
+def create_gpu_instance_template(body: GPUInstanceTemplateCreate,
+                                 _: User = Depends(get_admin_user)):
+    from ..schemas import GPUInstanceTemplate
+    from .gpu_instances import FLAVORS
+
+    if body.flavor not in FLAVORS:
+        raise HTTPException(400, f"unknown flavor {body.flavor!r}")
+    with get_session() as s:
+        if s.query(GPUInstanceTemplate).filter_by(name=body.name).first():
+            raise HTTPException(409, "template exists")
+        t = GPUInstanceTemplate(**body.model_dump())
+        ar_create(s, t)
+        return t.to_dict()
+
+
+@router.delete("/gpu_instance_templates/{tid}")
+def delete_gpu_instance_template(tid: int,
+                                 _: User = Depends(get_admin_user)):
+    from ..schemas import GPUInstanceTemplate
+
+    with get_session() as s:
+        t = s.get(GPUInstanceTemplate, tid)
+        if not t:
+            raise HTTPException(404, "template not found")
+        ar_delete(s, t)
+    return {"status": "deleted"}
+
+
+@router.get("/ssh_public_keys")
+def list_ssh_keys(user: User = Depends(get_current_user)):
+    from ..schemas import SSHPublicKey
+
+    with get_session() as s:
+        q = s.query(SSHPublicKey)
+        if not user.is_admin:
+            q = q.filter_by(user_id=user.id)
+        return {"items": [k.to_dict() for k in q.all()]}
+
+
+@router.post("/ssh_public_keys", status_code=201)
+def create_ssh_key(body: SSHPublicKeyCreate,
+                   user: User = Depends(get_current_user)):
+    from ..schemas import SSHPublicKey
+
+    if not body.public_key.strip().startswith(("ssh-", "ecdsa-")):
+        raise HTTPException(400, "not an SSH public key")
+    with get_session() as s:
+        if s.query(SSHPublicKey).filter_by(name=body.name).first():
+            raise HTTPException(409, "key name exists")
+        k = SSHPublicKey(name=body.name, user_id=user.id,
+                         public_key=body.public_key.strip())
+        ar_create(s, k)
+        return k.to_dict()
+
+
+@router.delete("/ssh_public_keys/{kid}")
+def delete_ssh_key(kid: int, user: User = Depends(get_current_user)):
+    from ..schemas import SSHPublicKey
+
+    with get_session() as s:
+        k = s.get(SSHPublicKey, kid)
+        if not k or (not user.is_admin and k.user_id != user.id):
+            raise HTTPException(404, "key not found")
+        ar_delete(s, k)
+    return {"status": "deleted"}
 
 
 @router.get("/gpu_instance_flavors")

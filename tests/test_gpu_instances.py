@@ -194,3 +194,47 @@ def test_migration_creates_gpu_instances_table(tmp_path):
         "SELECT name FROM sqlite_master WHERE type='table'")]
     assert "gpu_instances" in tables
     con.close()
+
+
+def test_templates_and_ssh_keys(server):
+    """Templates preset flavor/image/volumes; instances can start from a
+    template and reference a stored SSH key (reference:
+    gpu_instance_templates + gpu_instance_ssh_public_keys)."""
+    client, cfg = server
+    r = client.post("/v2/gpu_instance_templates", json={
+        "name": "dev-4g", "flavor": "mi355x-4gpu",
+        "image": "rocm/megatron-lm", "provider": "mock",
+        "volumes": [{"size_gb": 100, "mount_path": "/work"}]})
+    assert r.status_code == 201, r.text
+    assert client.post("/v2/gpu_instance_templates", json={
+        "name": "bad", "flavor": "h100"}).status_code == 400
+    r = client.post("/v2/ssh_public_keys", json={
+        "name": "laptop", "public_key": "ssh-ed25519 AAAAC3 u@h"})
+    assert r.status_code == 201
+    assert client.post("/v2/ssh_public_keys", json={
+        "name": "junk", "public_key": "not-a-key"}).status_code == 400
+
+    r = client.post("/v2/gpu_instances", json={
+        "name": "from-tpl", "template": "dev-4g", "ssh_key_name": "laptop"})
+    assert r.status_code == 201, r.text
+    g = r.json()
+    assert g["flavor"] == "mi355x-4gpu"
+    assert g["image"] == "rocm/megatron-lm"
+    assert g["provider"] == "mock"
+    assert g["volumes"][0]["mount_path"] == "/work"
+    assert g["ssh_public_key"] == "ssh-ed25519 AAAAC3 u@h"
+    # caller overrides beat template fields
+    r = client.post("/v2/gpu_instances", json={
+        "name": "from-tpl-2", "template": "dev-4g",
+        "flavor": "mi355x-1gpu", "provider": "mock"})
+    assert r.json()["flavor"] == "mi355x-1gpu"
+    # unknown refs are 400s
+    assert client.post("/v2/gpu_instances", json={
+        "name": "x1", "template": "nope"}).status_code == 400
+    assert client.post("/v2/gpu_instances", json={
+        "name": "x2", "provider": "mock",
+        "ssh_key_name": "nope"}).status_code == 400
+    # listing + delete
+    assert len(client.get("/v2/gpu_instance_templates").json()["items"]) == 1
+    kid = client.get("/v2/ssh_public_keys").json()["items"][0]["id"]
+    assert client.delete(f"/v2/ssh_public_keys/{kid}").status_code == 200
