@@ -77,6 +77,17 @@ class EngineRunner:
         self._wake = threading.Event()
         self._stop = False
         self.stats = {"requests": 0, "generated_tokens": 0, "prompt_tokens": 0}
+        # latency histograms (vLLM-parity observability): TTFT per request
+        # and inter-token gap per emitted token, fixed second buckets
+        self.ttft_buckets = [0.05, 0.1, 0.25, 0.5, 1.0, 2.5, 5.0, 10.0]
+        self.tpot_buckets = [0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1.0]
+        self.ttft_hist = [0] * (len(self.ttft_buckets) + 1)
+        self.ttft_sum = 0.0
+        self.ttft_count = 0
+        self.tpot_hist = [0] * (len(self.tpot_buckets) + 1)
+        self.tpot_sum = 0.0
+        self.tpot_count = 0
+        self._first_seen: dict[str, float] = {}  # rid -> last token time
         self.thread = threading.Thread(target=self._run, name="engine-loop", daemon=True)
 
     def start(self, loop: asyncio.AbstractEventLoop) -> None:
@@ -131,11 +142,25 @@ class EngineRunner:
                         self._push(q, {"error": str(e), "finished": True})
                 time.sleep(1)
                 continue
+            now = time.monotonic()
             for out in outputs:
                 with self._lock:
                     q = self._queues.get(out.request_id)
                 if q is not None:
                     self.stats["generated_tokens"] += 1
+                    prev = self._first_seen.get(out.request_id)
+                    if prev is None:
+                        seq = self.engine.seqs.get(out.request_id)
+                        if seq is not None and seq.first_token_time is not None:
+                            self._observe(
+                                "ttft",
+                                seq.first_token_time - seq.arrival_time)
+                    else:
+                        self._observe("tpot", now - prev)
+                    if out.finished:
+                        self._first_seen.pop(out.request_id, None)
+                    else:
+                        self._first_seen[out.request_id] = now
                     self._push(q, {
                         "token_id": out.token_id,
                         "finished": out.finished,
@@ -143,6 +168,22 @@ class EngineRunner:
                         "logprob": out.logprob,
                         "top_logprobs": out.top_logprobs,
                     })
+
+    def _observe(self, kind: str, v: float) -> None:
+        buckets = self.ttft_buckets if kind == "ttft" else self.tpot_buckets
+        hist = self.ttft_hist if kind == "ttft" else self.tpot_hist
+        for i, b in enumerate(buckets):
+            if v <= b:
+                hist[i] += 1
+                break
+        else:
+            hist[-1] += 1
+        if kind == "ttft":
+            self.ttft_sum += v
+            self.ttft_count += 1
+        else:
+            self.tpot_sum += v
+            self.tpot_count += 1
 
     def _push(self, q: asyncio.Queue, item: dict) -> None:
         assert self.loop is not None
@@ -487,6 +528,22 @@ def create_app(runner: EngineRunner) -> FastAPI:
             "# TYPE gpustack_engine_kv_blocks_total gauge",
             f"gpustack_engine_kv_blocks_total {e.scheduler.kv.allocator.num_blocks}",
         ]
+        for kind, buckets, hist, hsum, hcount in (
+            ("ttft", runner.ttft_buckets, runner.ttft_hist,
+             runner.ttft_sum, runner.ttft_count),
+            ("time_per_output_token", runner.tpot_buckets,
+             runner.tpot_hist, runner.tpot_sum, runner.tpot_count),
+        ):
+            name = f"gpustack_engine_{kind}_seconds"
+            lines.append(f"# TYPE {name} histogram")
+            cum = 0
+            for b, n in zip(buckets, hist):
+                cum += n
+                lines.append(f'{name}_bucket{{le="{b}"}} {cum}')
+            cum += hist[-1]
+            lines.append(f'{name}_bucket{{le="+Inf"}} {cum}')
+            lines.append(f"{name}_sum {hsum}")
+            lines.append(f"{name}_count {hcount}")
         return Response("\n".join(lines) + "\n", media_type="text/plain; version=0.0.4")
 
     async def _generate(request: Request, body: dict, prompt_ids: list[int],

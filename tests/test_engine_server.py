@@ -304,3 +304,42 @@ def test_streaming_logprobs():
             proc.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+def test_engine_metrics_histograms():
+    """vLLM-parity latency observability: TTFT and inter-token-gap
+    histograms appear in the engine /metrics after serving requests."""
+    port = _free_port()
+    proc = subprocess.Popen([
+        sys.executable, "-m", "gpustack_amd.worker.engine_server",
+        "--served-name", "tiny-h", "--source", "preset", "--model-ref",
+        "tiny", "--port", str(port), "--max-model-len", "256",
+        "--device", "cpu", "--kv-cache-blocks", "64",
+    ])
+    try:
+        _wait_health(port, proc)
+        for _ in range(3):
+            r = httpx.post(f"http://127.0.0.1:{port}/v1/completions", json={
+                "model": "tiny-h", "prompt": "hello", "max_tokens": 6,
+                "ignore_eos": True}, timeout=60)
+            assert r.status_code == 200
+        m = httpx.get(f"http://127.0.0.1:{port}/metrics", timeout=10).text
+        assert "gpustack_engine_ttft_seconds_bucket" in m
+        assert "gpustack_engine_time_per_output_token_seconds_count" in m
+        import re as _re
+
+        ttft_count = int(_re.search(
+            r"gpustack_engine_ttft_seconds_count (\d+)", m).group(1))
+        assert ttft_count >= 3
+        tpot_count = int(_re.search(
+            r"gpustack_engine_time_per_output_token_seconds_count (\d+)",
+            m).group(1))
+        assert tpot_count >= 3 * 4  # >=5 gaps per 6-token request, 3 reqs
+        # cumulative histogram sanity: +Inf bucket equals count
+        inf = int(_re.search(
+            r'gpustack_engine_ttft_seconds_bucket\{le="\+Inf"\} (\d+)',
+            m).group(1))
+        assert inf == ttft_count
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
