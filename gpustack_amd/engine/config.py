@@ -76,6 +76,11 @@ class ModelSpec:
     # logits, custom attention scale, GeGLU MLP; sliding window
     # alternates via layer_types (even layers) like GPT-OSS
     sandwich_norms: bool = False
+    # OLMo-2 (Olmo2ForCausalLM): NO input norms — RMSNorm applied to each
+    # sublayer's OUTPUT before the residual add; qk-norm runs over the
+    # FULL projected q/k vectors (num_heads*head_dim), not per head
+    norm_after: bool = False
+    qk_norm_full: bool = False
     # Gemma-3: sliding layers rope at a LOCAL base frequency (10k) while
     # full-attention layers use rope_theta (1M, linearly scaled) — the
     # model holds two cos/sin caches and each layer picks by window
@@ -167,6 +172,7 @@ class ModelSpec:
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=arch.startswith("Qwen2"),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
+                     or arch.startswith("Olmo2")
                      or bool(cfg.get("use_qk_norm", False))),
             eos_token_id=eos,
             num_experts=cfg.get("num_experts",
@@ -216,6 +222,8 @@ class ModelSpec:
             rope_interleave=bool(cfg.get("rope_interleave", True)),
             sandwich_norms=(arch.startswith("Gemma2")
                             or arch.startswith("Gemma3")),
+            norm_after=arch.startswith("Olmo2"),
+            qk_norm_full=arch.startswith("Olmo2"),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
                          if arch.startswith("Gemma") else 0.0),
             attn_logit_softcap=(cfg.get("attn_logit_softcapping") or 0.0)
@@ -381,6 +389,15 @@ PRESETS: dict[str, ModelSpec] = {
         vocab_size=512, hidden_size=128, intermediate_size=256, num_layers=2,
         num_heads=4, num_kv_heads=2, head_dim=32, max_position_embeddings=512,
         rope_theta=10000.0, eos_token_id=1,
+    ),
+    # OLMo-2 13B (norm-after layers + full-projection qk-norm)
+    "olmo-2-13b": ModelSpec(
+        architecture="Olmo2ForCausalLM", vocab_size=100352,
+        hidden_size=5120, intermediate_size=13824, num_layers=40,
+        num_heads=40, num_kv_heads=40, head_dim=128,
+        rope_theta=500000.0, max_position_embeddings=4096,
+        rms_norm_eps=1e-6, eos_token_id=100257, qk_norm=True,
+        norm_after=True, qk_norm_full=True,
     ),
     # Phi-4 14B (Phi3ForCausalLM graph: fused qkv/gate_up, plain rope)
     "phi-4": ModelSpec(

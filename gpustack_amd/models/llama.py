@@ -114,6 +114,9 @@ class Attention(nn.Module):
         self.rot_dim = int(spec.head_dim * spec.partial_rotary_factor)
         self.scale = spec.attn_scale or self.d ** -0.5
         self.softcap = spec.attn_logit_softcap
+        self.hq_full = spec.num_heads
+        self.hkv_full = max(spec.num_kv_heads, tp_size) \
+            if spec.num_kv_heads < tp_size else spec.num_kv_heads
         h = spec.hidden_size
         qkv_out = (self.hq + 2 * self.hkv) * self.d
         self.qkv_w = nn.Parameter(torch.empty(qkv_out, h, dtype=dtype), requires_grad=False)
@@ -146,8 +149,18 @@ class Attention(nn.Module):
                            ("v_proj", nq + nk, nk)]
         self._o_projs = [("o_proj", 0, h)]
         if spec.qk_norm:
-            self.q_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
-            self.k_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
+            if spec.qk_norm_full:
+                # OLMo-2: norm over the FULL projection (sharded weight
+                # slice per rank; the mean reduces across the tp group)
+                self.q_norm = nn.Parameter(
+                    torch.empty(self.hq * self.d, dtype=dtype),
+                    requires_grad=False)
+                self.k_norm = nn.Parameter(
+                    torch.empty(self.hkv * self.d, dtype=dtype),
+                    requires_grad=False)
+            else:
+                self.q_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
+                self.k_norm = nn.Parameter(torch.empty(self.d, dtype=dtype), requires_grad=False)
 
     def forward(self, x, meta: ForwardMeta, cos_sin, k_cache, v_cache):
         T = x.shape[0]
@@ -165,8 +178,11 @@ class Attention(nn.Module):
             q = q.contiguous()
             k = k.contiguous()
             v = v.contiguous()
-            ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
-            ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
+            if self.spec.qk_norm_full:
+                self._full_qk_norm(q, k)
+            else:
+                ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
+                ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
         ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
                              self.rot_dim)
         if meta.cp is not None:
@@ -208,6 +224,25 @@ class Attention(nn.Module):
         if meta.lora is not None:
             meta.lora.apply(self.layer_idx, out.view(T, -1), o, self._o_projs)
         return self.comm.all_reduce(o)
+
+    def _full_qk_norm(self, q, k):
+        """OLMo-2 full-projection RMSNorm: the mean square runs over ALL
+        num_heads*head_dim dims; under TP each rank holds a shard, so the
+        per-row sum-of-squares all-reduces over the tp group before the
+        (sharded) weight scales — bit-compatible with the single-rank
+        math up to summation order."""
+        import torch.distributed as dist
+
+        eps = self.spec.rms_norm_eps
+        for t, w, full in ((q, self.q_norm, self.hq_full * self.d),
+                           (k, self.k_norm, self.hkv_full * self.d)):
+            T = t.shape[0]
+            flat = t.view(T, -1).float()
+            ss = flat.pow(2).sum(-1, keepdim=True)
+            if self.comm.tp_size > 1:
+                dist.all_reduce(ss, group=self.comm.group)
+            inv = torch.rsqrt(ss / full + eps)
+            t.view(T, -1).copy_((flat * inv * w.float()).to(t.dtype))
 
 
 class MLAAttention(nn.Module):
@@ -825,6 +860,11 @@ class DecoderLayer(nn.Module):
                                             requires_grad=False)
             self.post_ff_norm = nn.Parameter(torch.empty(h, dtype=dtype),
                                              requires_grad=False)
+        elif spec.norm_after:
+            # OLMo-2: only output norms (input_norm doubles as the
+            # post-attn norm slot, post_attn_norm as the post-ffn one —
+            # the loader maps HF names accordingly)
+            self.pre_ff_norm = self.post_ff_norm = None
         else:
             self.pre_ff_norm = self.post_ff_norm = None
 
@@ -833,6 +873,9 @@ class DecoderLayer(nn.Module):
         if self.spec.sandwich_norms:
             return self._forward_sandwich(x, meta, cos_sin, k_cache,
                                           v_cache, eps)
+        if self.spec.norm_after:
+            return self._forward_norm_after(x, meta, cos_sin, k_cache,
+                                            v_cache, eps)
         if residual is None:
             residual = x
             h = torch.empty_like(x)
@@ -844,6 +887,17 @@ class DecoderLayer(nn.Module):
         ops.fused_add_rms_norm(a, residual, self.post_attn_norm, eps)
         m = self.mlp(a, meta)
         return m, residual
+
+    def _forward_norm_after(self, x, meta, cos_sin, k_cache, v_cache, eps):
+        """OLMo-2 layer flow (true hidden stream, residual sentinel None):
+            x = x + norm_attn(attn(x)); x = x + norm_ff(mlp(x))
+        (sublayers see the UNNORMED stream; norms scale their outputs)."""
+        a = self.attn(x, meta, cos_sin, k_cache, v_cache)
+        ops.rms_norm(a, a, self.input_norm, eps)
+        x = x + a
+        m = self.mlp(x, meta)
+        ops.rms_norm(m, m, self.post_attn_norm, eps)
+        return x + m, None
 
     def _forward_sandwich(self, x, meta, cos_sin, k_cache, v_cache, eps):
         """Gemma-2 layer flow (residual carried explicitly — the layer

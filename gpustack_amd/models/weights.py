@@ -311,10 +311,14 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                 layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
             if spec.qk_norm:
                 off = 1 if spec.sandwich_norms else 0  # Gemma: (1+w) norms
-                layer.attn.q_norm.copy_(
-                    get(p + "self_attn.q_norm.weight") + off)
-                layer.attn.k_norm.copy_(
-                    get(p + "self_attn.k_norm.weight") + off)
+                qn = get(p + "self_attn.q_norm.weight") + off
+                kn = get(p + "self_attn.k_norm.weight") + off
+                if spec.qk_norm_full:  # OLMo-2: weights shard with heads
+                    qn = qn[rank * hq * d:(rank + 1) * hq * d]
+                    kn = _kv_slice(kn.unsqueeze(1), rank, tp,
+                                   spec.num_kv_heads, d).squeeze(1)
+                layer.attn.q_norm.copy_(qn)
+                layer.attn.k_norm.copy_(kn)
         if hasattr(layer.mlp, "router_w")                 and spec.architecture.startswith("GptOss"):
             # GPT-OSS: router `mlp.router.{weight,bias}`; experts stored
             # TRANSPOSED ([E, h, 2i] / [E, i, h]) with INTERLEAVED
@@ -395,7 +399,14 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             ]))
             dn = get(p + "mlp.down_proj.weight")
             layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
-        if spec.sandwich_norms:
+        if spec.norm_after:
+            # OLMo-2: no input norms — our input_norm slot holds the
+            # post-attention norm, post_attn_norm the post-ffn one
+            layer.input_norm.copy_(
+                get(p + "post_attention_layernorm.weight"))
+            layer.post_attn_norm.copy_(
+                get(p + "post_feedforward_layernorm.weight"))
+        elif spec.sandwich_norms:
             # Gemma-2 RMSNorm multiplies by (1 + w); store the EFFECTIVE
             # weight so the shared rms_norm kernel applies
             layer.input_norm.copy_(get(p + "input_layernorm.weight") + 1)
