@@ -590,3 +590,21 @@ def test_gpu_devices_and_config(server):
     body = r.json()
     assert body.get("bootstrap_password") == "***"
     assert "data_dir" in body
+
+
+def test_catalog_refs_resolve(server):
+    """Every catalog entry's preset ref must exist and its categories
+    match the architecture registry — no dead catalog rows."""
+    client, app, cfg, _ = server
+    from gpustack_amd.engine.config import PRESETS
+    from gpustack_amd.utils.model_registry import categories_for_architecture
+
+    body = client.get("/v2/catalog").json()
+    items = body.get("items") or body.get("models") or []
+    assert len(items) >= 15
+    for it in items:
+        assert it["source"] == "preset"
+        spec = PRESETS.get(it["model_ref"])
+        assert spec is not None, f"dead catalog ref {it['model_ref']}"
+        derived = set(categories_for_architecture(spec.architecture))
+        assert derived & set(it["categories"]), (it["name"], derived)
