@@ -81,6 +81,12 @@ class ModelSpec:
     # FULL projected q/k vectors (num_heads*head_dim), not per head
     norm_after: bool = False
     qk_norm_full: bool = False
+    # Granite (GraniteForCausalLM): llama graph + scalar multipliers —
+    # embeddings (embed_scale above), attention scale (attn_scale above),
+    # sublayer outputs before the residual adds, and a final logits
+    # DIVISOR
+    residual_multiplier: float = 0.0     # 0 = off (1.0)
+    logits_scaling: float = 0.0          # 0 = off (divide logits by this)
     # Gemma-3: sliding layers rope at a LOCAL base frequency (10k) while
     # full-attention layers use rope_theta (1M, linearly scaled) — the
     # model holds two cos/sin caches and each layer picks by window
@@ -225,14 +231,22 @@ class ModelSpec:
             norm_after=arch.startswith("Olmo2"),
             qk_norm_full=arch.startswith("Olmo2"),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
-                         if arch.startswith("Gemma") else 0.0),
+                         if arch.startswith("Gemma")
+                         else (cfg.get("embedding_multiplier") or 0.0)
+                         if arch.startswith("Granite") else 0.0),
             attn_logit_softcap=(cfg.get("attn_logit_softcapping") or 0.0)
             if arch.startswith("Gemma") else 0.0,
             final_logit_softcap=(cfg.get("final_logit_softcapping") or 0.0)
             if arch.startswith("Gemma") else 0.0,
             attn_scale=((cfg.get("query_pre_attn_scalar") or 0) ** -0.5
                         if (arch.startswith("Gemma")
-                            and cfg.get("query_pre_attn_scalar")) else 0.0),
+                            and cfg.get("query_pre_attn_scalar"))
+                        else (cfg.get("attention_multiplier") or 0.0)
+                        if arch.startswith("Granite") else 0.0),
+            residual_multiplier=(cfg.get("residual_multiplier") or 0.0)
+            if arch.startswith("Granite") else 0.0,
+            logits_scaling=(cfg.get("logits_scaling") or 0.0)
+            if arch.startswith("Granite") else 0.0,
             rope_local_theta=(cfg.get("rope_local_base_freq") or 0.0)
             if arch.startswith("Gemma3") else 0.0,
             mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),

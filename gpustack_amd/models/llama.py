@@ -884,8 +884,13 @@ class DecoderLayer(nn.Module):
             h = x
             ops.fused_add_rms_norm(h, residual, self.input_norm, eps)
         a = self.attn(h, meta, cos_sin, k_cache, v_cache)
+        rm = self.spec.residual_multiplier
+        if rm:  # Granite: scale sublayer outputs before the residual add
+            a = a * rm
         ops.fused_add_rms_norm(a, residual, self.post_attn_norm, eps)
         m = self.mlp(a, meta)
+        if rm:
+            m = m * rm
         return m, residual
 
     def _forward_norm_after(self, x, meta, cos_sin, k_cache, v_cache, eps):
@@ -1048,6 +1053,8 @@ class LlamaForCausalLM(nn.Module):
         if self.spec.final_logit_softcap:
             cap = self.spec.final_logit_softcap
             logits = torch.tanh(logits / cap) * cap
+        if self.spec.logits_scaling:
+            logits = logits / self.spec.logits_scaling
         if return_both:  # draft-model speculative needs the features too
             return logits, hidden
         return logits
