@@ -420,6 +420,38 @@ PRESETS: dict[str, ModelSpec] = {
         num_heads=40, num_kv_heads=10, head_dim=128, rope_theta=250000.0,
         max_position_embeddings=16384, eos_token_id=100257,
     ),
+    # DeepSeek-R1: identical architecture/dims to V3 (reasoning ckpt)
+    "deepseek-r1": ModelSpec(
+        architecture="DeepseekV3ForCausalLM", vocab_size=129280,
+        hidden_size=7168, intermediate_size=18432, num_layers=61,
+        num_heads=128, num_kv_heads=128, head_dim=192,
+        rope_theta=10000.0, max_position_embeddings=163840,
+        rope_scaling={"rope_type": "yarn", "factor": 40.0,
+                      "beta_fast": 32.0, "beta_slow": 1.0,
+                      "mscale": 1.0, "mscale_all_dim": 1.0,
+                      "original_max_position_embeddings": 4096},
+        eos_token_id=1, num_experts=256, num_experts_per_tok=8,
+        moe_intermediate_size=2048, router_mode="sigmoid_bias",
+        n_shared_experts=1, first_k_dense_replace=3,
+        routed_scaling_factor=2.5, n_group=8, topk_group=4,
+        norm_topk_prob=True,
+        q_lora_rank=1536, kv_lora_rank=512, qk_nope_head_dim=128,
+        qk_rope_head_dim=64, v_head_dim=128,
+    ),
+    # Kimi-K2 1T: DeepseekV3 architecture at larger expert count
+    "kimi-k2": ModelSpec(
+        architecture="DeepseekV3ForCausalLM", vocab_size=163840,
+        hidden_size=7168, intermediate_size=18432, num_layers=61,
+        num_heads=64, num_kv_heads=64, head_dim=192,
+        rope_theta=50000.0, max_position_embeddings=131072,
+        eos_token_id=163585, num_experts=384, num_experts_per_tok=8,
+        moe_intermediate_size=2048, router_mode="sigmoid_bias",
+        n_shared_experts=1, first_k_dense_replace=1,
+        routed_scaling_factor=2.827, n_group=1, topk_group=1,
+        norm_topk_prob=True,
+        q_lora_rank=1536, kv_lora_rank=512, qk_nope_head_dim=128,
+        qk_rope_head_dim=64, v_head_dim=128,
+    ),
     # Gemma-2: sandwich norms, GeGLU, softcapping, alternating SWA,
     # scaled tied embeddings (CPU-oracle family; GPU kernels share the
     # GPT-OSS r3 window work — head_dim 256 needs a D-template too)
