@@ -43,6 +43,10 @@ def _single_proc_result(model: str = "tiny", **kw) -> list[list[int]]:
     return eng.generate(PROMPTS, SamplingParams(max_tokens=6, ignore_eos=True))
 
 
+NGRAM = {"method": "ngram", "num_draft_tokens": 3, "ngram_max": 3,
+         "ngram_min": 1}
+
+
 def _cp_rank_main(rank: int, world: int, port: int, out_path: str,
                   model: str = "tiny", cfg_kw: dict | None = None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
@@ -178,6 +182,14 @@ def test_guided_rejected_under_cp():
 
     with pytest.raises(ValueError, match="context parallelism"):
         eng.add_request([1, 2, 3], SamplingParams(guided_json=True))
+
+
+def test_cp2_ngram_spec_matches_single_rank():
+    """n-gram speculative decoding under CP: drafts propose from the
+    (identical) output history on every rank and verify rows run the
+    replicated decode path — output must equal single-rank exactly."""
+    kw = {"speculative": dict(NGRAM)}
+    assert _run_cp2("tiny", kw) == _single_proc_result("tiny", **kw)
 
 
 def test_cp2_chunked_prefill_matches_single_rank():
