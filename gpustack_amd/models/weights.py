@@ -625,7 +625,7 @@ def convert_to_w4_runtime(model, cfg: EngineConfig) -> int:
 
     def pack_site(mod, wname: str, pname: str) -> None:
         nonlocal n_packed
-        w = getattr(mod, wname)
+        w = getattr(mod, wname, None)  # MLA attention has no fused qkv
         if w is None or w.numel() == 0:
             return
         pk = quantize(w.data)

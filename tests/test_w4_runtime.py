@@ -53,3 +53,33 @@ def test_engine_w4_runtime_cpu():
                                   kv_cache_blocks=64, quantize_runtime="w4"))
     assert eng2.generate([[1, 2, 3, 4]],
                          SamplingParams(max_tokens=6, ignore_eos=True))[0] == out
+
+
+def test_w4_runtime_on_mla_model():
+    """W4 runtime on a DeepSeek-shaped model: the MLA attention has no
+    fused qkv tensor — conversion must pack what it can (o_proj/lm_head)
+    and serving must stay exact vs the bf16-free fp32 reference for
+    shapes that stay unpacked (tiny dims are W4-ineligible: graceful
+    no-op, not a crash)."""
+    import dataclasses
+
+    import gpustack_amd.engine.config as C
+    from gpustack_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from gpustack_amd.engine.config import PRESETS
+
+    base = PRESETS["tiny-mla"]
+    C.PRESETS["tiny-mla-w4"] = dataclasses.replace(base)
+    try:
+        plain = LLMEngine(EngineConfig(
+            model="tiny-mla-w4", device="cpu", kv_cache_blocks=64,
+            max_model_len=128, seed=0, dtype="float32"))
+        w4 = LLMEngine(EngineConfig(
+            model="tiny-mla-w4", device="cpu", kv_cache_blocks=64,
+            max_model_len=128, seed=0, dtype="float32",
+            quantize_runtime="w4"))
+        p = SamplingParams(max_tokens=5, ignore_eos=True)
+        prompts = [[3, 1, 4, 1, 5]]
+        # tiny dims are ineligible for packing -> identical outputs
+        assert w4.generate(prompts, p) == plain.generate(prompts, p)
+    finally:
+        C.PRESETS.pop("tiny-mla-w4", None)
