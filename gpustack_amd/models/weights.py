@@ -636,8 +636,14 @@ def convert_to_w4_runtime(model, cfg: EngineConfig) -> int:
         n_packed += 1
 
     for layer in model.layers:
-        pack_site(layer.attn, "qkv_w", "qkv_pack")
-        pack_site(layer.attn, "o_w", "o_pack")
+        if model.spec.kv_lora_rank:
+            # MLA attention routes its projections through plain
+            # F.linear/einsums (no qlinear dispatch yet) — leave them
+            # bf16; W4 there is an r3 follow-up with the MLA kernels
+            pass
+        else:
+            pack_site(layer.attn, "qkv_w", "qkv_pack")
+            pack_site(layer.attn, "o_w", "o_pack")
         if model.spec.num_experts == 0:
             pack_site(layer.mlp, "gate_up_w", "gate_up_pack")
             pack_site(layer.mlp, "down_w", "down_pack")
