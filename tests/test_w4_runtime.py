@@ -79,7 +79,15 @@ def test_w4_runtime_on_mla_model():
             quantize_runtime="w4"))
         p = SamplingParams(max_tokens=5, ignore_eos=True)
         prompts = [[3, 1, 4, 1, 5]]
-        # tiny dims are ineligible for packing -> identical outputs
-        assert w4.generate(prompts, p) == plain.generate(prompts, p)
+        out = w4.generate(prompts, p)
+        assert len(out[0]) == 5  # serves (no fused-qkv crash)
+        # MLA attention stays bf16; the (lossy) packing hit lm_head only
+        m = w4.runner.model
+        assert not hasattr(m.layers[0].attn, "qkv_pack") \
+            or m.layers[0].attn.qkv_pack is None
+        assert m.layers[0].attn.o_w.numel() > 0  # unpacked
+        assert m.lm_head_pack is not None        # eligible + packed
+        # determinism holds under the packed head
+        assert w4.generate(prompts, p) == out
     finally:
         C.PRESETS.pop("tiny-mla-w4", None)
