@@ -402,6 +402,10 @@ class ModelRunner:
         return self.kv
 
     def add_lora(self, name: str, adapter_dir: str) -> int:
+        if self.cfg.spec.kv_lora_rank:
+            raise ValueError("dynamic LoRA is not supported on MLA "
+                             "(DeepSeek) models yet — the latent attention "
+                             "projections have no adapter hook")
         return self.lora_bank.add(name, adapter_dir, self.device,
                                   getattr(torch, self.cfg.dtype))
 

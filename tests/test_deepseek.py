@@ -337,3 +337,11 @@ def test_deepseek_prefix_cache_hit_matches_cold():
     assert [warm1, warm2] == cold
     alloc = eng.scheduler.kv.allocator
     assert alloc.hits > 0  # the second prompt actually reused blocks
+
+
+def test_deepseek_dynamic_lora_rejected(tmp_path):
+    """MLA attention has no adapter hook yet — a dynamic LoRA add must
+    refuse loudly instead of silently serving a half-applied adapter."""
+    eng = _engine()
+    with pytest.raises(ValueError, match="MLA"):
+        eng.runner.add_lora("a", str(tmp_path))
