@@ -361,6 +361,13 @@ class ModelRunner:
             if self.comm.pp_size > 1:
                 raise ValueError("draft-model speculative decoding is not "
                                  "supported with pipeline parallelism")
+            if (cfg.spec.kv_lora_rank or cfg.spec.sandwich_norms
+                    or cfg.spec.norm_after or cfg.spec.num_experts):
+                raise ValueError(
+                    "draft-model speculative decoding currently supports "
+                    "dense pre-norm decoder families (llama/qwen/mistral "
+                    "class) — MLA / sandwich-norm / norm-after / MoE "
+                    "targets are not wired to the EAGLE draft layer yet")
             from .eagle import EagleProposer
 
             k = int(spec.get("num_draft_tokens", 3))
