@@ -1003,6 +1003,46 @@ def version():
     return {"version": __version__}
 
 
+@router.get("/model_instances/{instance_id}/logs")
+async def instance_logs(instance_id: int, tail: int = Query(200),
+                        _: User = Depends(get_current_user)):
+    """Instance log tail proxied through the server (reference:
+    routes/model_instances `logs` → worker log API), honoring the
+    worker's proxy mode (direct HTTP or the NAT tunnel)."""
+    from fastapi.responses import PlainTextResponse
+
+    with get_session() as s:
+        inst = s.get(ModelInstance, instance_id)
+        if not inst:
+            raise HTTPException(404, "instance not found")
+        w = s.get(Worker, inst.worker_id) if inst.worker_id else None
+    if w is None:
+        raise HTTPException(409, "instance has no worker yet")
+    path = f"/logs/{inst.name}?tail={int(tail)}"
+    if w.proxy_mode == "tunnel":
+        from .tunnel import hub
+
+        try:
+            status, _ctype, chunks = await hub.request(
+                w.id, w.port, "GET", path, b"")
+        except TimeoutError:
+            raise HTTPException(502, "tunnel reply timeout")
+        buf = b""
+        async for c in chunks:
+            buf += c
+        return PlainTextResponse(buf.decode(errors="replace"),
+                                 status_code=status)
+    import httpx as _httpx
+
+    url = f"http://{w.ip}:{w.port}{path}"
+    try:
+        async with _httpx.AsyncClient(timeout=15) as c:
+            r = await c.get(url)
+    except _httpx.HTTPError:
+        raise HTTPException(502, "worker unreachable")
+    return PlainTextResponse(r.text, status_code=r.status_code)
+
+
 @router.get("/gpu_devices")
 def list_gpu_devices(page: int | None = Query(None),
                      perPage: int | None = Query(None),

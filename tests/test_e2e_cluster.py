@@ -519,3 +519,30 @@ def test_deploy_gemma_model(cluster):
     assert r.status_code == 200, r.text
     assert r.json()["usage"]["completion_tokens"] == 5
     client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-g2-e2e'][0]['id']}")
+
+
+@pytest.mark.timeout(120)
+def test_instance_logs_via_server(cluster):
+    """Instance logs proxied through the server to the worker's log API
+    (reference: model-instance logs route)."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-logs", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201, r.text
+    inst = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-logs"]
+        if insts and insts[0]["state"] == "running":
+            inst = insts[0]
+            break
+        time.sleep(0.5)
+    assert inst is not None
+    r = client.get(f"/v2/model_instances/{inst['id']}/logs",
+                   params={"tail": 50})
+    assert r.status_code == 200, r.text
+    assert "engine" in r.text or len(r.text) > 0  # some log content
+    assert client.get("/v2/model_instances/999999/logs").status_code == 404
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-logs'][0]['id']}")
