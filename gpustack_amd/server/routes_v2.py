@@ -422,6 +422,20 @@ def delete_instance(instance_id: int, _: User = Depends(get_current_user)):
         return {"ok": True}
 
 
+@router.post("/model_instances/{instance_id}/restart")
+def restart_instance(instance_id: int, _: User = Depends(get_admin_user)):
+    """Tear the instance down; the ModelController's replica reconcile
+    recreates it (the reference's restart semantics — state machine runs
+    PENDING→…→RUNNING again on a fresh process)."""
+    with get_session() as s:
+        inst = s.get(ModelInstance, instance_id)
+        if not inst:
+            raise HTTPException(404)
+        model_id = inst.model_id
+        ar_delete(s, inst)
+    return {"status": "restarting", "model_id": model_id}
+
+
 # ---- model routes ----------------------------------------------------------
 
 @router.get("/model_files")

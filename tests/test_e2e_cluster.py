@@ -546,3 +546,42 @@ def test_instance_logs_via_server(cluster):
     assert "engine" in r.text or len(r.text) > 0  # some log content
     assert client.get("/v2/model_instances/999999/logs").status_code == 404
     client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-logs'][0]['id']}")
+
+
+@pytest.mark.timeout(180)
+def test_instance_restart_action(cluster):
+    """POST /restart tears the instance down and the controller brings a
+    fresh one back to RUNNING."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-restart", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201
+    inst = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-restart"]
+        if insts and insts[0]["state"] == "running":
+            inst = insts[0]
+            break
+        time.sleep(0.5)
+    assert inst is not None
+    r = client.post(f"/v2/model_instances/{inst['id']}/restart")
+    assert r.status_code == 200, r.text
+    # a NEW instance (different id) reaches running
+    new_inst = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-restart"]
+        if insts and insts[0]["id"] != inst["id"] \
+                and insts[0]["state"] == "running":
+            new_inst = insts[0]
+            break
+        time.sleep(0.5)
+    assert new_inst is not None, "replacement instance never ran"
+    r = client.post("/v1/completions", json={
+        "model": "tiny-restart", "prompt": "x", "max_tokens": 3,
+        "ignore_eos": True})
+    assert r.status_code == 200
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-restart'][0]['id']}")
