@@ -569,17 +569,22 @@ def test_instance_restart_action(cluster):
     assert inst is not None
     r = client.post(f"/v2/model_instances/{inst['id']}/restart")
     assert r.status_code == 200, r.text
-    # a NEW instance (different id) reaches running
-    new_inst = None
+    # the instance is torn down (row gone or non-running) and a fresh
+    # process comes back to RUNNING (SQLite may reuse the row id)
+    saw_down = False
+    recovered = None
     for _ in range(240):
         insts = [i for i in client.get("/v2/model_instances").json()["items"]
                  if i["model_name"] == "tiny-restart"]
-        if insts and insts[0]["id"] != inst["id"] \
-                and insts[0]["state"] == "running":
-            new_inst = insts[0]
+        if not insts or insts[0]["state"] != "running":
+            saw_down = True
+        elif saw_down and insts[0]["state"] == "running":
+            recovered = insts[0]
             break
-        time.sleep(0.5)
-    assert new_inst is not None, "replacement instance never ran"
+        time.sleep(0.25)
+    assert saw_down, "restart never tore the instance down"
+    assert recovered is not None, "replacement instance never ran"
+    assert recovered.get("pid") != inst.get("pid")  # fresh process
     r = client.post("/v1/completions", json={
         "model": "tiny-restart", "prompt": "x", "max_tokens": 3,
         "ignore_eos": True})
