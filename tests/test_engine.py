@@ -213,3 +213,31 @@ def test_llama31_rope_scaling_preset():
     out = LLMEngine(small).generate(
         [[1, 2, 3]], SamplingParams(max_tokens=4, ignore_eos=True))[0]
     assert len(out) == 4
+
+
+def test_family_presets_sane():
+    """Every serving preset has self-consistent dims and a plausible
+    bf16 weight estimate (catches preset typos before a deploy does)."""
+    from gpustack_amd.engine.config import PRESETS
+
+    expect_gib = {
+        "llama-3-8b": (13, 18), "llama-3-70b": (125, 145),
+        "qwen3-32b": (55, 70), "qwen3-30b-a3b": (50, 62),
+        "deepseek-v3": (1150, 1350), "deepseek-r1": (1150, 1350),
+        "kimi-k2": (1750, 2100), "gemma-2-9b": (15, 20),
+        "gemma-3-27b": (45, 58), "phi-4": (25, 32),
+        "olmo-2-13b": (22, 30), "glm-4.5-air": (180, 250),
+        "gpt-oss-20b": (35, 45), "gpt-oss-120b": (200, 250),
+    }
+    for name, spec in PRESETS.items():
+        if name.startswith("tiny"):
+            continue
+        assert spec.hidden_size % spec.num_heads == 0 or spec.head_dim, name
+        assert spec.num_heads % max(1, spec.num_kv_heads) == 0 \
+            or spec.num_kv_heads % spec.num_heads == 0, name
+        gib = spec.weight_bytes() / 2**30
+        assert 1 < gib < 2200, (name, gib)
+        if name in expect_gib:
+            lo, hi = expect_gib[name]
+            assert lo <= gib <= hi, (name, round(gib))
+        assert spec.kv_bytes_per_token() > 0, name
