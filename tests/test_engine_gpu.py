@@ -245,3 +245,103 @@ def test_guided_json_gpu():
     out = eng.generate([[30, 31]], p)[0]
     doc = _json.loads(tok.decode([t for t in out if t != 1]))
     assert isinstance(doc["ok"], bool)
+
+
+# ---- gated engine-level serving for the r3 kernel variants ----------------
+
+import os as _os
+
+_oss_engine = pytest.mark.skipif(
+    _os.environ.get("GPUSTACK_AMD_OSS_KERNELS") != "1",
+    reason="set GPUSTACK_AMD_OSS_KERNELS=1 after the r3 kernel validation")
+_mla_engine = pytest.mark.skipif(
+    _os.environ.get("GPUSTACK_AMD_MLA_KERNEL") != "1",
+    reason="set GPUSTACK_AMD_MLA_KERNEL=1 after the r3 kernel validation")
+
+
+@_oss_engine
+def test_gpt_oss_shape_engine_gpu():
+    """GPT-OSS-shaped serving end-to-end on the gated kernels: head_dim
+    64, attention sinks, alternating sliding windows, clamped-swiglu
+    biased experts. decode==prefill consistency on-device."""
+    import dataclasses
+
+    cfg = EngineConfig(model="gpt-oss-20b", device="cuda",
+                       max_model_len=512, max_num_seqs=8,
+                       gpu_memory_utilization=0.2)
+    cfg.spec = dataclasses.replace(
+        cfg.spec, num_layers=4, hidden_size=1024, num_heads=16,
+        num_kv_heads=4, vocab_size=2048, intermediate_size=1024,
+        moe_intermediate_size=256, num_experts=8, num_experts_per_tok=2,
+        sliding_window=64, max_position_embeddings=512,
+        rope_scaling=None)
+    eng = LLMEngine(cfg)
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    prompt = list(range(2, 80))  # crosses the 64-token window
+    full = eng.generate([prompt], p)[0]
+    assert len(full) == 8
+    del eng
+    torch.cuda.empty_cache()
+    eng2 = LLMEngine(cfg)
+    cont = eng2.generate([prompt + full[:4]],
+                         SamplingParams(max_tokens=4, ignore_eos=True))[0]
+    assert cont == full[4:]
+
+
+@_oss_engine
+def test_gemma_shape_engine_gpu():
+    """Gemma-2-shaped serving on the gated kernels: head_dim 256,
+    sandwich norms, logit softcapping, GeGLU, alternating windows."""
+    import dataclasses
+
+    cfg = EngineConfig(model="gemma-2-9b", device="cuda",
+                       max_model_len=512, max_num_seqs=8,
+                       gpu_memory_utilization=0.2)
+    cfg.spec = dataclasses.replace(
+        cfg.spec, num_layers=4, hidden_size=1024, num_heads=8,
+        num_kv_heads=4, head_dim=256, vocab_size=2048,
+        intermediate_size=1024, sliding_window=64,
+        max_position_embeddings=512, attn_scale=256 ** -0.5,
+        embed_scale=1024 ** 0.5)
+    eng = LLMEngine(cfg)
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    prompt = list(range(2, 80))
+    full = eng.generate([prompt], p)[0]
+    assert len(full) == 8
+    del eng
+    torch.cuda.empty_cache()
+    eng2 = LLMEngine(cfg)
+    cont = eng2.generate([prompt + full[:4]],
+                         SamplingParams(max_tokens=4, ignore_eos=True))[0]
+    assert cont == full[4:]
+
+
+@_mla_engine
+def test_mla_shape_engine_gpu():
+    """DeepSeek-shaped MLA serving on the gated kernels (expand prefill +
+    absorbed decode over the latent cache): decode==prefill on-device,
+    and the latent pool is the only KV allocation."""
+    import dataclasses
+
+    cfg = EngineConfig(model="deepseek-v3", device="cuda",
+                       max_model_len=512, max_num_seqs=8,
+                       gpu_memory_utilization=0.2)
+    cfg.spec = dataclasses.replace(
+        cfg.spec, num_layers=3, hidden_size=1024, num_heads=16,
+        num_kv_heads=16, vocab_size=2048, intermediate_size=1024,
+        moe_intermediate_size=256, num_experts=8, num_experts_per_tok=2,
+        n_group=2, topk_group=1, first_k_dense_replace=1,
+        q_lora_rank=256, max_position_embeddings=512, rope_scaling=None)
+    eng = LLMEngine(cfg)
+    kv = eng.runner.kv
+    assert kv.k_caches[0].shape[-1] == 576 and kv.v_caches[0].numel() == 0
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    prompt = list(range(2, 50))
+    full = eng.generate([prompt], p)[0]
+    assert len(full) == 8
+    del eng
+    torch.cuda.empty_cache()
+    eng2 = LLMEngine(cfg)
+    cont = eng2.generate([prompt + full[:4]],
+                         SamplingParams(max_tokens=4, ignore_eos=True))[0]
+    assert cont == full[4:]
