@@ -37,6 +37,12 @@ echo "=== 6. kernel micro-bench (commit summaries to profiles/) ==="
 GPUSTACK_AMD_MLA_KERNEL=1 timeout 240 python scripts/bench_mla.py \
     | tee gpurun_out/mla_bench.txt
 
+echo "=== 7. engine-level serving on the gated shapes ==="
+GPUSTACK_AMD_OSS_KERNELS=1 timeout 420 python -m pytest \
+    tests/test_engine_gpu.py -q -m gpu -k "oss_shape or gemma_shape" || exit 1
+GPUSTACK_AMD_MLA_KERNEL=1 timeout 300 python -m pytest \
+    tests/test_engine_gpu.py -q -m gpu -k "mla_shape" || exit 1
+
 echo "ALL GATED KERNEL GROUPS VALIDATED"
 echo "next: flip gate defaults, run full 'pytest -m gpu', then bench"
 echo "  gpt-oss-20b / gemma-2-9b / deepseek CPU-vs-GPU logits spot-checks"
