@@ -558,6 +558,11 @@ class MoEMLP(nn.Module):
             torch.empty(self.e, 2 * self.i, h, dtype=dtype), requires_grad=False)
         self.down_w = nn.Parameter(
             torch.empty(self.e, h, self.i, dtype=dtype), requires_grad=False)
+        # W4 runtime (weights.convert_to_w4_runtime): per-expert packs;
+        # bank-shaped consumers dequant a TRANSIENT bf16 bank per call
+        # (resident capacity is the point: Qwen3-235B W4 fits one GPU)
+        self.gate_up_packs: list | None = None
+        self.down_packs: list | None = None
         # GLM/DeepSeek-style extensions (spec.router_mode "sigmoid_bias"):
         # learned correction bias on the routing scores and a SHARED dense
         # expert applied to every token (TP shards its intermediate)
@@ -708,6 +713,40 @@ class MoEMLP(nn.Module):
         out = contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
         return self.comm.all_reduce(self._add_shared(x, out))
 
+    def _bank(self, which: str):
+        """bf16 expert bank for the fused/bmm paths: the resident tensor,
+        or a transient dequant of the per-expert W4 packs (allocator-
+        cached; hipGraph-capture-safe — the dense W4 decode path already
+        dequants transiently inside captured graphs)."""
+        packs = (self.gate_up_packs if which == "gate_up"
+                 else self.down_packs)
+        w = self.gate_up_w if which == "gate_up" else self.down_w
+        if packs is None:
+            return w
+        p0 = packs[0]
+        dev = p0.qw.device
+        bank = torch.empty((self.e,) + p0.shape, dtype=torch.bfloat16,
+                           device=dev)
+        if dev.type == "cuda":
+            hip = ops._load_hip()
+            for e, pk in enumerate(packs):
+                hip.w4_dequant(bank[e], pk.qw, pk.sc, pk.zs)
+        else:
+            from .quantized import dequant_w4_runtime
+
+            for e, pk in enumerate(packs):
+                bank[e].copy_(dequant_w4_runtime(pk.qw, pk.sc, pk.zs))
+        return bank
+
+    def _expert_w(self, which: str, e: int):
+        """Per-expert weight for the loop path (qlinear handles packs)."""
+        packs = (self.gate_up_packs if which == "gate_up"
+                 else self.down_packs)
+        w = self.gate_up_w if which == "gate_up" else self.down_w
+        if packs is None:
+            return w[e], None
+        return None, packs[e]
+
     def _add_shared(self, x, out):
         """Shared dense expert (GLM-4.5: every token, added to the routed
         mix BEFORE the TP all-reduce so one collective covers both)."""
@@ -752,12 +791,12 @@ class MoEMLP(nn.Module):
         hip = ops._load_hip()
         act = x.new_empty(TK, self.i)
         act_mode = 1 if self.spec.moe_act == "clamped_swiglu" else 0
-        hip.moe_gate_up_silu(act, x, self.gate_up_w, s_tok, offs, counts,
-                             bias=self.gate_up_b, act_mode=act_mode)
+        hip.moe_gate_up_silu(act, x, self._bank("gate_up"), s_tok, offs,
+                             counts, bias=self.gate_up_b, act_mode=act_mode)
         contrib = x.new_empty(TK, x.shape[1])
         # down bias on tp rank 0 only: the TP all-reduce must sum it once
         db = self.down_b if self.comm.tp_rank == 0 else None
-        hip.moe_down_scale(contrib, act, self.down_w, offs, counts,
+        hip.moe_down_scale(contrib, act, self._bank("down"), offs, counts,
                            order.to(torch.int32), flat_w32, bias=db)
         return contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
 
@@ -782,13 +821,15 @@ class MoEMLP(nn.Module):
             rows = torch.nonzero(flat_exp == e, as_tuple=False).flatten()
             idx = flat_tok[rows]
             xe = x.index_select(0, idx)
-            gu = F.linear(xe, self.gate_up_w[e],
-                          self.gate_up_b[e] if self.gate_up_b is not None
-                          else None)
+            gw, gp = self._expert_w("gate_up", e)
+            gu = qlinear(xe, gw, gp,
+                         self.gate_up_b[e] if self.gate_up_b is not None
+                         else None)
             act = self._act_mul(gu)
-            he = F.linear(act, self.down_w[e],
-                          self.down_b[e] if (self.down_b is not None and tp0)
-                          else None)
+            dw, dp = self._expert_w("down", e)
+            he = qlinear(act, dw, dp,
+                         self.down_b[e] if (self.down_b is not None and tp0)
+                         else None)
             contrib[rows] = he * flat_w[rows].unsqueeze(1)
 
     def _bmm_dispatch(self, x, contrib, flat_exp, flat_tok, flat_w,
@@ -818,14 +859,14 @@ class MoEMLP(nn.Module):
         xpad = x.new_zeros(self.e, cap, x.shape[1])
         xpad[s_exp, pos] = x[s_tok]
         _ck("xpad")
-        gu = torch.bmm(xpad, self.gate_up_w.transpose(1, 2))   # [E, cap, 2i]
+        gu = torch.bmm(xpad, self._bank("gate_up").transpose(1, 2))  # [E, cap, 2i]
         if self.gate_up_b is not None:
             gu = gu + self.gate_up_b.unsqueeze(1)
         _ck("bmm1")
         act = self._act_mul(gu.reshape(self.e * cap, 2 * self.i))
         _ck("silu")
         hd = torch.bmm(act.view(self.e, cap, self.i),
-                       self.down_w.transpose(1, 2))            # [E, cap, h]
+                       self._bank("down").transpose(1, 2))     # [E, cap, h]
         if self.down_b is not None and self.comm.tp_rank == 0:
             hd = hd + self.down_b.unsqueeze(1)
         _ck("bmm2")

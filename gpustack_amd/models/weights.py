@@ -647,6 +647,30 @@ def convert_to_w4_runtime(model, cfg: EngineConfig) -> int:
         if model.spec.num_experts == 0:
             pack_site(layer.mlp, "gate_up_w", "gate_up_pack")
             pack_site(layer.mlp, "down_w", "down_pack")
+        elif hasattr(layer.mlp, "router_w"):
+            # MoE: per-expert packing of the 3-D banks — the capacity
+            # bulk of MoE checkpoints lives here (Qwen3-235B: ~97% of
+            # weights). Bank consumers dequant transiently per call;
+            # the loop path dequants per expert via qlinear.
+            for wname, pname in (("gate_up_w", "gate_up_packs"),
+                                 ("down_w", "down_packs")):
+                bank = getattr(layer.mlp, wname)
+                pks = []
+                for e in range(bank.shape[0]):
+                    pk = quantize(bank.data[e])
+                    if pk is None:
+                        pks = None
+                        break
+                    pks.append(W4Pack(*pk))
+                if pks is not None:
+                    setattr(layer.mlp, pname, pks)
+                    bank.data = torch.empty(0, dtype=bank.dtype,
+                                            device=bank.device)
+                    n_packed += len(pks)
+        elif model.spec.num_experts > 0:
+            # dense-first layer of an MoE model: plain MLP, pack normally
+            pack_site(layer.mlp, "gate_up_w", "gate_up_pack")
+            pack_site(layer.mlp, "down_w", "down_pack")
     if model.lm_head is not None and not model.spec.tie_word_embeddings:
         w = model.lm_head
         pk = quantize(w.data)

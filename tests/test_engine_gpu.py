@@ -345,3 +345,29 @@ def test_mla_shape_engine_gpu():
     cont = eng2.generate([prompt + full[:4]],
                          SamplingParams(max_tokens=4, ignore_eos=True))[0]
     assert cont == full[4:]
+
+
+def test_moe_w4_expert_banks_gpu():
+    """MoE with W4-packed expert banks on-device: dispatch dequants a
+    transient bf16 bank (validated w4_dequant + validated MoE kernels)
+    and serving stays deterministic."""
+    import dataclasses
+
+    cfg = EngineConfig(model="qwen3-30b-a3b", device="cuda",
+                       max_model_len=512, max_num_seqs=8,
+                       gpu_memory_utilization=0.2, quantize_runtime="w4")
+    cfg.spec = dataclasses.replace(
+        cfg.spec, num_layers=2, hidden_size=1024, num_heads=8,
+        num_kv_heads=4, head_dim=128, vocab_size=2048,
+        intermediate_size=1024, moe_intermediate_size=128, num_experts=16,
+        num_experts_per_tok=2, max_position_embeddings=512)
+    eng = LLMEngine(cfg)
+    mlp = eng.runner.model.layers[0].mlp
+    assert mlp.gate_up_packs is not None and mlp.gate_up_w.numel() == 0
+    p = SamplingParams(max_tokens=8, ignore_eos=True)
+    out = eng.generate([[1, 2, 3, 4, 5]], p)[0]
+    assert len(out) == 8
+    del eng
+    torch.cuda.empty_cache()
+    eng2 = LLMEngine(cfg)
+    assert eng2.generate([[1, 2, 3, 4, 5]], p)[0] == out

@@ -91,3 +91,69 @@ def test_w4_runtime_on_mla_model():
         assert w4.generate(prompts, p) == out
     finally:
         C.PRESETS.pop("tiny-mla-w4", None)
+
+
+def test_engine_w4_runtime_moe_expert_banks():
+    """MoE W4: the expert banks pack PER EXPERT (the capacity bulk of
+    MoE checkpoints) and dispatch dequants transiently. Output must
+    exactly match a reference engine whose bf16 experts are REPLACED by
+    the dequantized packed values — proving the serving math reads the
+    packed weights and nothing else."""
+    import dataclasses
+
+    import gpustack_amd.engine.config as C
+    from gpustack_amd.engine.config import PRESETS
+    from gpustack_amd.models.quantized import dequant_w4_runtime
+
+    # widen the expert intermediate so BOTH banks are W4-eligible
+    # (gate_up [2i, h]: N%64, K%128; down [h, i]: K=i needs %128)
+    base = PRESETS["tiny-moe"]
+    C.PRESETS["tiny-moe-w4e"] = dataclasses.replace(
+        base, moe_intermediate_size=128)
+    try:
+        p = SamplingParams(max_tokens=6, ignore_eos=True)
+        prompts = [[3, 1, 4, 1, 5, 9]]
+        w4 = LLMEngine(EngineConfig(model="tiny-moe-w4e", device="cpu",
+                                    kv_cache_blocks=64, max_model_len=128,
+                                    seed=0, dtype="float32",
+                                    quantize_runtime="w4"))
+        mlp = w4.runner.model.layers[0].mlp
+        assert mlp.gate_up_packs is not None and mlp.down_packs is not None
+        assert mlp.gate_up_w.numel() == 0 and mlp.down_w.numel() == 0
+
+        ref = LLMEngine(EngineConfig(model="tiny-moe-w4e", device="cpu",
+                                     kv_cache_blocks=64, max_model_len=128,
+                                     seed=0, dtype="float32"))
+        # graft the dequantized packed values into the reference engine
+        for lw4, lref in zip(w4.runner.model.layers,
+                             ref.runner.model.layers):
+            m4, mr = lw4.mlp, lref.mlp
+            if getattr(m4, "gate_up_packs", None) is None:
+                continue
+            for e, pk in enumerate(m4.gate_up_packs):
+                mr.gate_up_w.data[e].copy_(
+                    dequant_w4_runtime(pk.qw, pk.sc, pk.zs))
+            for e, pk in enumerate(m4.down_packs):
+                mr.down_w.data[e].copy_(
+                    dequant_w4_runtime(pk.qw, pk.sc, pk.zs))
+            # attention/lm_head also packed: graft those too
+            if lw4.attn.qkv_pack is not None:
+                from gpustack_amd.models.quantized import \
+                    dequant_w4_runtime as dq
+
+                lref.attn.qkv_w.data.copy_(
+                    dq(lw4.attn.qkv_pack.qw, lw4.attn.qkv_pack.sc,
+                       lw4.attn.qkv_pack.zs))
+            if lw4.attn.o_pack is not None:
+                lref.attn.o_w.data.copy_(
+                    dequant_w4_runtime(lw4.attn.o_pack.qw,
+                                       lw4.attn.o_pack.sc,
+                                       lw4.attn.o_pack.zs))
+        if w4.runner.model.lm_head_pack is not None:
+            ref.runner.model.lm_head.data.copy_(dequant_w4_runtime(
+                w4.runner.model.lm_head_pack.qw,
+                w4.runner.model.lm_head_pack.sc,
+                w4.runner.model.lm_head_pack.zs))
+        assert w4.generate(prompts, p) == ref.generate(prompts, p)
+    finally:
+        C.PRESETS.pop("tiny-moe-w4e", None)
