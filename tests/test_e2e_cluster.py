@@ -590,3 +590,34 @@ def test_instance_restart_action(cluster):
         "ignore_eos": True})
     assert r.status_code == 200
     client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-restart'][0]['id']}")
+
+
+@pytest.mark.timeout(120)
+def test_worker_log_follow_stream(cluster):
+    """Worker /logs?follow=true streams the tail then appended lines
+    until timeout_s (reference: log_sources follow)."""
+    client, agent = cluster
+    r = client.post("/v2/models", json={
+        "name": "tiny-follow", "source": "preset", "model_ref": "tiny",
+        "replicas": 1, "max_model_len": 256,
+    })
+    assert r.status_code == 201
+    inst = None
+    for _ in range(240):
+        insts = [i for i in client.get("/v2/model_instances").json()["items"]
+                 if i["model_name"] == "tiny-follow"]
+        if insts and insts[0]["state"] == "running":
+            inst = insts[0]
+            break
+        time.sleep(0.5)
+    assert inst is not None
+    wport = agent.cfg.worker_port
+    t0 = time.time()
+    r = httpx.get(f"http://127.0.0.1:{wport}/logs/{inst['name']}",
+                  params={"follow": "true", "timeout_s": 1.5, "tail": 20},
+                  timeout=30)
+    took = time.time() - t0
+    assert r.status_code == 200
+    assert len(r.text) > 0          # tail delivered
+    assert 1.0 < took < 10.0        # held open until timeout_s
+    client.delete(f"/v2/models/{[m for m in client.get('/v2/models').json()['items'] if m['name'] == 'tiny-follow'][0]['id']}")
