@@ -200,6 +200,13 @@ def test_cp2_chunked_prefill_matches_single_rank():
     assert _run_cp2("tiny", kw) == _single_proc_result("tiny", **kw)
 
 
+def test_cp2_prefix_cache_matches_single_rank():
+    """Prefix-cache hit suffixes under CP run the CP-split suffix path —
+    shared-prefix prompts must still match single-rank exactly."""
+    kw = {"enable_prefix_caching": True}
+    assert _run_cp2("tiny", kw) == _single_proc_result("tiny", **kw)
+
+
 def test_cp2_matches_single_rank_moe():
     """MoE layers are row-local under CP (router + experts see only this
     rank's rows) — exactness must hold through the grouped-expert path."""
