@@ -247,6 +247,7 @@ class EngineRunner:
 
     def abort(self, rid: str) -> None:
         self.engine.abort_request(rid)
+        self._first_seen.pop(rid, None)
         self.release(rid)
 
 
