@@ -559,11 +559,15 @@ def delete_cluster(cluster_id: int, _: User = Depends(get_admin_user)):
 # -- GPU instances (operator-analog SSH GPU pods, server/gpu_instances.py) --
 
 @router.get("/gpu_instances")
-def list_gpu_instances(_: User = Depends(get_current_user)):
+def list_gpu_instances(watch: bool = Query(False),
+                       _: User = Depends(get_current_user)):
     from ..schemas import GPUInstance
 
     with get_session() as s:
-        return {"items": [g.to_dict() for g in s.query(GPUInstance).all()]}
+        rows = [g.to_dict() for g in s.query(GPUInstance).all()]
+    if watch:
+        return _watch_stream("gpu_instances", rows, None)
+    return {"items": rows}
 
 
 @router.get("/gpu_instances/{gid}")

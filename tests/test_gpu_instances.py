@@ -238,3 +238,37 @@ def test_templates_and_ssh_keys(server):
     assert len(client.get("/v2/gpu_instance_templates").json()["items"]) == 1
     kid = client.get("/v2/ssh_public_keys").json()["items"][0]["id"]
     assert client.delete(f"/v2/ssh_public_keys/{kid}").status_code == 200
+
+
+def test_gpu_instance_watch_stream(server):
+    """watch frames replay the snapshot then relay lifecycle events
+    (generator-level, like the models watch test — TestClient buffers
+    streaming bodies)."""
+    import json as _json
+    import threading
+    import time as _time
+
+    client, cfg = server
+    client.post("/v2/gpu_instances", json={"name": "w1",
+                                           "provider": "mock"})
+    from gpustack_amd.server.routes_v2 import watch_ndjson
+
+    gen = watch_ndjson("gpu_instances", [{"id": 1, "name": "w1"}], None)
+    frames = []
+
+    def reader():
+        for line in gen:
+            frames.append(_json.loads(line))
+            if len(frames) >= 2:
+                break
+
+    t = threading.Thread(target=reader, daemon=True)
+    t.start()
+    _time.sleep(0.2)
+    client.post("/v2/gpu_instances", json={"name": "w2",
+                                           "provider": "mock"})
+    t.join(timeout=10)
+    assert len(frames) >= 2
+    assert frames[0]["data"]["name"] == "w1"
+    assert frames[1]["type"] == "CREATED"
+    assert frames[1]["data"]["name"] == "w2"
