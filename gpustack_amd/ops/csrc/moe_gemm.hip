@@ -54,6 +54,10 @@ DEVICE_INLINE void stage_rows(unsigned short* lds, const unsigned short* src,
   }
 }
 
+// EXTM=false is the exact validated silu/no-bias fast path (identical to
+// the r2-measured kernel); EXTM=true carries the GPT-OSS clamped-swiglu
+// + expert-bias variant and only launches when those are requested.
+template <bool EXTM = false>
 __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
     unsigned short* __restrict__ act,       // [TK, I] bf16 (sorted rows)
     const unsigned short* __restrict__ x,   // [T, H] bf16
@@ -119,31 +123,37 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_gate_up_silu_kernel(
       }
     }
     // D[row = lg*4 + r][col = lc]; fuse the activation and write.
-    // bias depends on the column only — one pair of loads per lane.
+    // EXTM: bias depends on the column only — one pair of loads per lane.
     float bg = 0.f, bu = 0.f;
-    if (bias != nullptr) {
-      const int col = nt * MOE_BN + wave * 16 + lc;
-      bg = bf2f(bias[(long)e * 2 * I + col]);
-      bu = bf2f(bias[(long)e * 2 * I + I + col]);
+    if constexpr (EXTM) {
+      if (bias != nullptr) {
+        const int col = nt * MOE_BN + wave * 16 + lc;
+        bg = bf2f(bias[(long)e * 2 * I + col]);
+        bu = bf2f(bias[(long)e * 2 * I + I + col]);
+      }
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = lg * 4 + r;
       if (row >= nrows) continue;
       float v;
-      if (act_mode == 1) {  // GPT-OSS clamped swiglu
+      if (EXTM && act_mode == 1) {  // GPT-OSS clamped swiglu
         const float g = fminf(ag[r] + bg, 7.f);
         const float u = fminf(fmaxf(au[r] + bu, -7.f), 7.f);
         v = (u + 1.f) * g / (1.f + __expf(-1.702f * g));
-      } else {
+      } else if constexpr (EXTM) {
         const float g = ag[r] + bg;
         v = g / (1.f + __expf(-g)) * (au[r] + bu);
+      } else {
+        const float g = ag[r];
+        v = g / (1.f + __expf(-g)) * au[r];
       }
       act[(long)(base + m0 + row) * I + nt * MOE_BN + wave * 16 + lc] = f2bf(v);
     }
   }
 }
 
+template <bool EXTM = false>
 __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
     unsigned short* __restrict__ contrib,   // [TK, H] bf16 (original order)
     const unsigned short* __restrict__ act, // [TK, I] bf16 (sorted rows)
@@ -196,15 +206,23 @@ __global__ __launch_bounds__(MOE_THREADS) void moe_down_scale_kernel(
         for (int kk = 0; kk < kc; kk += 32) body(kk);
       }
     }
-    const float bd = (bias != nullptr)
-        ? bf2f(bias[(long)e * H + nt * MOE_BN + wave * 16 + lc]) : 0.f;
+    float bd = 0.f;
+    if constexpr (EXTM) {
+      bd = (bias != nullptr)
+          ? bf2f(bias[(long)e * H + nt * MOE_BN + wave * 16 + lc]) : 0.f;
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = lg * 4 + r;
       if (row >= nrows) continue;
       const int oj = order[base + m0 + row];
-      contrib[(long)oj * H + nt * MOE_BN + wave * 16 + lc] =
-          f2bf((acc[r] + bd) * flat_w[oj]);
+      if constexpr (EXTM) {
+        contrib[(long)oj * H + nt * MOE_BN + wave * 16 + lc] =
+            f2bf((acc[r] + bd) * flat_w[oj]);
+      } else {
+        contrib[(long)oj * H + nt * MOE_BN + wave * 16 + lc] =
+            f2bf(acc[r] * flat_w[oj]);
+      }
     }
   }
 }
@@ -219,10 +237,19 @@ void moe_gate_up_silu_launch(void* act, const void* x, const void* w,
   *err_unsupported = 0;
   if (I % MOE_BN != 0 || H % 128 != 0) { *err_unsupported = 1; return; }
   dim3 grid(E * (I / MOE_BN));
-  hipLaunchKernelGGL(moe_gate_up_silu_kernel, grid, dim3(MOE_THREADS), 0, s,
-                     (unsigned short*)act, (const unsigned short*)x,
-                     (const unsigned short*)w, s_tok, offs, counts,
-                     (const unsigned short*)bias, act_mode, H, I);
+  const bool extm = bias != nullptr || act_mode != 0;
+  if (extm)
+    hipLaunchKernelGGL((moe_gate_up_silu_kernel<true>), grid,
+                       dim3(MOE_THREADS), 0, s, (unsigned short*)act,
+                       (const unsigned short*)x, (const unsigned short*)w,
+                       s_tok, offs, counts, (const unsigned short*)bias,
+                       act_mode, H, I);
+  else
+    hipLaunchKernelGGL((moe_gate_up_silu_kernel<false>), grid,
+                       dim3(MOE_THREADS), 0, s, (unsigned short*)act,
+                       (const unsigned short*)x, (const unsigned short*)w,
+                       s_tok, offs, counts, (const unsigned short*)bias,
+                       act_mode, H, I);
 }
 
 void moe_down_scale_launch(void* contrib, const void* act, const void* w,
@@ -233,8 +260,16 @@ void moe_down_scale_launch(void* contrib, const void* act, const void* w,
   *err_unsupported = 0;
   if (H % MOE_BN != 0 || I % 128 != 0) { *err_unsupported = 1; return; }
   dim3 grid(E * (H / MOE_BN));
-  hipLaunchKernelGGL(moe_down_scale_kernel, grid, dim3(MOE_THREADS), 0, s,
-                     (unsigned short*)contrib, (const unsigned short*)act,
-                     (const unsigned short*)w, offs, counts, order, flat_w,
-                     (const unsigned short*)bias, H, I);
+  if (bias != nullptr)
+    hipLaunchKernelGGL((moe_down_scale_kernel<true>), grid,
+                       dim3(MOE_THREADS), 0, s, (unsigned short*)contrib,
+                       (const unsigned short*)act, (const unsigned short*)w,
+                       offs, counts, order, flat_w,
+                       (const unsigned short*)bias, H, I);
+  else
+    hipLaunchKernelGGL((moe_down_scale_kernel<false>), grid,
+                       dim3(MOE_THREADS), 0, s, (unsigned short*)contrib,
+                       (const unsigned short*)act, (const unsigned short*)w,
+                       offs, counts, order, flat_w,
+                       (const unsigned short*)bias, H, I);
 }
