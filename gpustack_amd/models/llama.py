@@ -713,11 +713,16 @@ class MoEMLP(nn.Module):
         out = contrib.view(T, self.top_k, -1).sum(dim=1).to(x.dtype)
         return self.comm.all_reduce(self._add_shared(x, out))
 
+    # ONE scratch per (device, shape) shared by every layer: layers
+    # dequant into it sequentially on the compute stream, so a captured
+    # decode graph records kernels against a STATIC address (no per-
+    # bucket transient replication) and eager mode allocates it once.
+    _bank_scratch: dict = {}
+
     def _bank(self, which: str):
         """bf16 expert bank for the fused/bmm paths: the resident tensor,
-        or a transient dequant of the per-expert W4 packs (allocator-
-        cached; hipGraph-capture-safe — the dense W4 decode path already
-        dequants transiently inside captured graphs)."""
+        or the shared scratch filled by dequanting this layer's per-
+        expert W4 packs."""
         packs = (self.gate_up_packs if which == "gate_up"
                  else self.down_packs)
         w = self.gate_up_w if which == "gate_up" else self.down_w
@@ -725,8 +730,12 @@ class MoEMLP(nn.Module):
             return w
         p0 = packs[0]
         dev = p0.qw.device
-        bank = torch.empty((self.e,) + p0.shape, dtype=torch.bfloat16,
-                           device=dev)
+        key = (str(dev), self.e) + p0.shape
+        bank = MoEMLP._bank_scratch.get(key)
+        if bank is None:
+            bank = torch.empty((self.e,) + p0.shape, dtype=torch.bfloat16,
+                               device=dev)
+            MoEMLP._bank_scratch[key] = bank
         if dev.type == "cuda":
             hip = ops._load_hip()
             for e, pk in enumerate(packs):
