@@ -87,6 +87,9 @@ class ModelSpec:
     # DIVISOR
     residual_multiplier: float = 0.0     # 0 = off (1.0)
     logits_scaling: float = 0.0          # 0 = off (divide logits by this)
+    # SmolLM3 (SmolLM3ForCausalLM): per-layer rope switch — entry 0 means
+    # a NoPE layer (no rotary at all); None = rope everywhere
+    no_rope_layers: tuple | None = None
     # Gemma-3: sliding layers rope at a LOCAL base frequency (10k) while
     # full-attention layers use rope_theta (1M, linearly scaled) — the
     # model holds two cos/sin caches and each layer picks by window
@@ -247,6 +250,12 @@ class ModelSpec:
             if arch.startswith("Granite") else 0.0,
             logits_scaling=(cfg.get("logits_scaling") or 0.0)
             if arch.startswith("Granite") else 0.0,
+            no_rope_layers=(tuple(cfg["no_rope_layers"])
+                            if cfg.get("no_rope_layers") else (
+                                tuple(0 if (i + 1) % 4 == 0 else 1
+                                      for i in range(
+                                          cfg.get("num_hidden_layers", 0)))
+                                if arch.startswith("SmolLM3") else None)),
             rope_local_theta=(cfg.get("rope_local_base_freq") or 0.0)
             if arch.startswith("Gemma3") else 0.0,
             mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),

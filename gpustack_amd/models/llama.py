@@ -115,6 +115,10 @@ class Attention(nn.Module):
         self.scale = spec.attn_scale or self.d ** -0.5
         self.softcap = spec.attn_logit_softcap
         self.hq_full = spec.num_heads
+        # SmolLM3 NoPE layers skip rotary entirely
+        self.use_rope = (spec.no_rope_layers is None
+                         or layer_idx >= len(spec.no_rope_layers)
+                         or bool(spec.no_rope_layers[layer_idx]))
         self.hkv_full = max(spec.num_kv_heads, tp_size) \
             if spec.num_kv_heads < tp_size else spec.num_kv_heads
         h = spec.hidden_size
@@ -183,8 +187,9 @@ class Attention(nn.Module):
             else:
                 ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d), self.q_norm, self.spec.rms_norm_eps)
                 ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
-        ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
-                             self.rot_dim)
+        if self.use_rope:
+            ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
+                                 self.rot_dim)
         if meta.cp is not None:
             # CP prefill: assemble the full-batch K/V (global position
             # order) so the cache write below covers every chunk, not just
