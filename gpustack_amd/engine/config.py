@@ -87,6 +87,10 @@ class ModelSpec:
     # DIVISOR
     residual_multiplier: float = 0.0     # 0 = off (1.0)
     logits_scaling: float = 0.0          # 0 = off (divide logits by this)
+    # rotary application layout: "neox" rotate-half (llama lineage) or
+    # "pairwise" GPT-J-style adjacent-pair rotation (Ernie-4.5; Hunyuan
+    # and MiniMax share it — SURVEY family matrix)
+    rope_mode: str = "neox"
     # SmolLM3 (SmolLM3ForCausalLM): per-layer rope switch — entry 0 means
     # a NoPE layer (no rotary at all); None = rope everywhere
     no_rope_layers: tuple | None = None
@@ -179,7 +183,9 @@ class ModelSpec:
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
-            attention_bias=arch.startswith("Qwen2"),
+            attention_bias=(arch.startswith("Qwen2")
+                            or (arch.startswith("Ernie4_5")
+                                and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
                      or arch.startswith("Olmo2")
                      or bool(cfg.get("use_qk_norm", False))),
@@ -250,6 +256,8 @@ class ModelSpec:
             if arch.startswith("Granite") else 0.0,
             logits_scaling=(cfg.get("logits_scaling") or 0.0)
             if arch.startswith("Granite") else 0.0,
+            rope_mode=("pairwise" if arch.startswith("Ernie4_5")
+                       else "neox"),
             no_rope_layers=(tuple(cfg["no_rope_layers"])
                             if cfg.get("no_rope_layers") else (
                                 tuple(0 if (i + 1) % 4 == 0 else 1

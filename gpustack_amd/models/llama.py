@@ -189,7 +189,7 @@ class Attention(nn.Module):
                 ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d), self.k_norm, self.spec.rms_norm_eps)
         if self.use_rope:
             ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
-                                 self.rot_dim)
+                                 self.rot_dim, mode=self.spec.rope_mode)
         if meta.cp is not None:
             # CP prefill: assemble the full-batch K/V (global position
             # order) so the cache write below covers every chunk, not just

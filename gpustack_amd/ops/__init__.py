@@ -94,12 +94,19 @@ def fused_add_rms_norm(x, residual, weight, eps: float) -> None:
         torch_ref.fused_add_rms_norm(x, residual, weight, eps)
 
 
-def rotary_embedding(positions, q, k, cos_sin, head_dim: int, rot_dim: int) -> None:
+def rotary_embedding(positions, q, k, cos_sin, head_dim: int, rot_dim: int,
+                     mode: str = "neox") -> None:
     hip = _backend(q)
     if hip is not None:
+        if mode != "neox":
+            raise NotImplementedError(
+                "pairwise (GPT-J style) rotary is not in the CDNA4 kernel "
+                "yet (Ernie/Hunyuan/MiniMax GPU serving lands with it in "
+                "r3); CPU serving is available")
         hip.rotary_embedding(positions, q, k, cos_sin, head_dim, rot_dim)
     else:
-        torch_ref.rotary_embedding(positions, q, k, cos_sin, head_dim, rot_dim)
+        torch_ref.rotary_embedding(positions, q, k, cos_sin, head_dim,
+                                   rot_dim, mode=mode)
 
 
 def silu_and_mul(out, x) -> None:

@@ -130,8 +130,12 @@ def rotary_embedding(
     cos_sin: torch.Tensor,
     head_dim: int,
     rot_dim: int,
+    mode: str = "neox",
 ) -> None:
-    """In-place neox-style rotation. q: [T, Hq*D] or [T, Hq, D]; same for k."""
+    """In-place rotation. q: [T, Hq*D] or [T, Hq, D]; same for k.
+    mode "neox" rotates the two halves (llama lineage); "pairwise"
+    rotates adjacent (even, odd) pairs by one frequency each (GPT-J
+    layout; Ernie-4.5 / Hunyuan / MiniMax checkpoints)."""
     half = rot_dim // 2
     cs = cos_sin[positions]  # [T, rot]
     cos = cs[:, :half].unsqueeze(1)  # [T, 1, half]
@@ -139,6 +143,14 @@ def rotary_embedding(
     for x in (q, k):
         T = x.shape[0]
         xs = x if x.dim() == 3 else x.view(T, -1, head_dim)
+        if mode == "pairwise":
+            x1 = xs[..., 0:rot_dim:2].float()
+            x2 = xs[..., 1:rot_dim:2].float()
+            o1 = x1 * cos - x2 * sin
+            o2 = x2 * cos + x1 * sin
+            xs[..., 0:rot_dim:2] = o1.to(x.dtype)
+            xs[..., 1:rot_dim:2] = o2.to(x.dtype)
+            continue
         x1 = xs[..., :half].float()
         x2 = xs[..., half : 2 * half].float()
         o1 = x1 * cos - x2 * sin
