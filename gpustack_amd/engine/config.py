@@ -91,6 +91,9 @@ class ModelSpec:
     # "pairwise" GPT-J-style adjacent-pair rotation (Ernie-4.5; Hunyuan
     # and MiniMax share it — SURVEY family matrix)
     rope_mode: str = "neox"
+    # Hunyuan-dense: per-head qk-norm applied AFTER rope (Qwen3 norms
+    # before rope)
+    qk_norm_after_rope: bool = False
     # SmolLM3 (SmolLM3ForCausalLM): per-layer rope switch — entry 0 means
     # a NoPE layer (no rotary at all); None = rope everywhere
     no_rope_layers: tuple | None = None
@@ -188,7 +191,9 @@ class ModelSpec:
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
                      or arch.startswith("Olmo2")
+                     or arch.startswith("HunYuan")
                      or bool(cfg.get("use_qk_norm", False))),
+            qk_norm_after_rope=arch.startswith("HunYuan"),
             eos_token_id=eos,
             num_experts=cfg.get("num_experts",
                                 cfg.get("num_local_experts",

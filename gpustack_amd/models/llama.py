@@ -178,7 +178,7 @@ class Attention(nn.Module):
         q = qkv[:, :nq].unflatten(1, (self.hq, self.d))
         k = qkv[:, nq:nq + nk].unflatten(1, (self.hkv, self.d))
         v = qkv[:, nq + nk:].unflatten(1, (self.hkv, self.d))
-        if self.spec.qk_norm:
+        if self.spec.qk_norm and not self.spec.qk_norm_after_rope:
             q = q.contiguous()
             k = k.contiguous()
             v = v.contiguous()
@@ -190,6 +190,15 @@ class Attention(nn.Module):
         if self.use_rope:
             ops.rotary_embedding(meta.positions, q, k, cos_sin, self.d,
                                  self.rot_dim, mode=self.spec.rope_mode)
+        if self.spec.qk_norm and self.spec.qk_norm_after_rope:
+            # Hunyuan-dense: per-head norm on the ROTATED q/k
+            q = q.contiguous()
+            k = k.contiguous()
+            v = v.contiguous()
+            ops.rms_norm(q.view(-1, self.d), q.view(-1, self.d),
+                         self.q_norm, self.spec.rms_norm_eps)
+            ops.rms_norm(k.view(-1, self.d), k.view(-1, self.d),
+                         self.k_norm, self.spec.rms_norm_eps)
         if meta.cp is not None:
             # CP prefill: assemble the full-batch K/V (global position
             # order) so the cache write below covers every chunk, not just

@@ -311,8 +311,14 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                 layer.attn.sinks.copy_(sk[rank * hq:(rank + 1) * hq])
             if spec.qk_norm:
                 off = 1 if spec.sandwich_norms else 0  # Gemma: (1+w) norms
-                qn = get(p + "self_attn.q_norm.weight") + off
-                kn = get(p + "self_attn.k_norm.weight") + off
+                qn_name = (p + "self_attn.q_norm.weight"
+                           if p + "self_attn.q_norm.weight" in tensors
+                           else p + "self_attn.query_layernorm.weight")
+                kn_name = (p + "self_attn.k_norm.weight"
+                           if p + "self_attn.k_norm.weight" in tensors
+                           else p + "self_attn.key_layernorm.weight")
+                qn = get(qn_name) + off
+                kn = get(kn_name) + off
                 if spec.qk_norm_full:  # OLMo-2: weights shard with heads
                     qn = qn[rank * hq * d:(rank + 1) * hq * d]
                     kn = _kv_slice(kn.unsqueeze(1), rank, tp,
