@@ -166,6 +166,16 @@ class ModelSpec:
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelSpec":
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        # OLMo-3 keys rope_parameters BY LAYER TYPE: full_attention
+        # defines the main rope (theta + any scaling); sliding layers
+        # rope unscaled at their own theta (the Gemma-3 dual-cache path).
+        slide_theta = 0.0
+        rp = cfg.get("rope_parameters")
+        if isinstance(rp, dict) and "full_attention" in rp:
+            cfg = dict(cfg)
+            cfg["rope_parameters"] = rp.get("full_attention") or {}
+            slide_theta = (rp.get("sliding_attention") or {}).get(
+                "rope_theta", 0.0) or 0.0
         nh = cfg.get("num_attention_heads", 32)
         hd = cfg.get("head_dim") or cfg.get("hidden_size", 4096) // nh
         eos = cfg.get("eos_token_id", 2)
@@ -190,7 +200,7 @@ class ModelSpec:
                             or (arch.startswith("Ernie4_5")
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
-                     or arch.startswith("Olmo2")
+                     or arch.startswith(("Olmo2", "Olmo3"))
                      or arch.startswith("HunYuan")
                      or bool(cfg.get("use_qk_norm", False))),
             qk_norm_after_rope=arch.startswith("HunYuan"),
@@ -218,7 +228,8 @@ class ModelSpec:
                                    .get("partial_rotary_factor") or 1.0),
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
-            if (arch.startswith("GptOss") or arch.startswith("Gemma"))
+            if (arch.startswith("GptOss") or arch.startswith("Gemma")
+                or arch.startswith("Olmo3"))
             else 0,
             layer_types=tuple(cfg["layer_types"])
             if cfg.get("layer_types") else (
@@ -242,8 +253,8 @@ class ModelSpec:
             rope_interleave=bool(cfg.get("rope_interleave", True)),
             sandwich_norms=(arch.startswith("Gemma2")
                             or arch.startswith("Gemma3")),
-            norm_after=arch.startswith("Olmo2"),
-            qk_norm_full=arch.startswith("Olmo2"),
+            norm_after=arch.startswith(("Olmo2", "Olmo3")),
+            qk_norm_full=arch.startswith(("Olmo2", "Olmo3")),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
                          if arch.startswith("Gemma")
                          else (cfg.get("embedding_multiplier") or 0.0)
@@ -270,7 +281,8 @@ class ModelSpec:
                                           cfg.get("num_hidden_layers", 0)))
                                 if arch.startswith("SmolLM3") else None)),
             rope_local_theta=(cfg.get("rope_local_base_freq") or 0.0)
-            if arch.startswith("Gemma3") else 0.0,
+            if arch.startswith("Gemma3")
+            else (slide_theta if arch.startswith("Olmo3") else 0.0),
             mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),
         )
 
