@@ -197,10 +197,11 @@ class ModelSpec:
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=(arch.startswith("Qwen2")
+                            or arch.startswith("SeedOss")
                             or (arch.startswith("Ernie4_5")
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
-                     or arch.startswith(("Olmo2", "Olmo3"))
+                     or arch.startswith(("Olmo2", "Olmo3", "Exaone4"))
                      or arch.startswith("HunYuan")
                      or bool(cfg.get("use_qk_norm", False))),
             qk_norm_after_rope=arch.startswith("HunYuan"),
@@ -229,7 +230,7 @@ class ModelSpec:
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
             if (arch.startswith("GptOss") or arch.startswith("Gemma")
-                or arch.startswith("Olmo3"))
+                or arch.startswith(("Olmo3", "Exaone4")))
             else 0,
             layer_types=tuple(cfg["layer_types"])
             if cfg.get("layer_types") else (
@@ -253,7 +254,7 @@ class ModelSpec:
             rope_interleave=bool(cfg.get("rope_interleave", True)),
             sandwich_norms=(arch.startswith("Gemma2")
                             or arch.startswith("Gemma3")),
-            norm_after=arch.startswith(("Olmo2", "Olmo3")),
+            norm_after=arch.startswith(("Olmo2", "Olmo3", "Exaone4")),
             qk_norm_full=arch.startswith(("Olmo2", "Olmo3")),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
                          if arch.startswith("Gemma")
@@ -276,6 +277,13 @@ class ModelSpec:
                        else "neox"),
             no_rope_layers=(tuple(cfg["no_rope_layers"])
                             if cfg.get("no_rope_layers") else (
+                                # EXAONE-4 hybrid: global-NoPE — rope on
+                                # sliding layers only, full layers unroped
+                                tuple(1 if lt == "sliding_attention" else 0
+                                      for lt in cfg["layer_types"])
+                                if (arch.startswith("Exaone4")
+                                    and cfg.get("sliding_window")
+                                    and cfg.get("layer_types")) else
                                 tuple(0 if (i + 1) % 4 == 0 else 1
                                       for i in range(
                                           cfg.get("num_hidden_layers", 0)))
