@@ -105,7 +105,8 @@ class ModelSpec:
     attn_logit_softcap: float = 0.0      # 0 = off
     final_logit_softcap: float = 0.0
     attn_scale: float = 0.0              # 0 = 1/sqrt(head_dim)
-    mlp_act: str = "silu"                # "silu" | "gelu_tanh"
+    mlp_act: str = "silu"                # "silu" | "gelu_tanh" | "relu2"
+    mlp_no_gate: bool = False            # Arcee: down(act(up(x))), no gate
 
     @property
     def gqa_ratio(self) -> int:
@@ -291,7 +292,10 @@ class ModelSpec:
             rope_local_theta=(cfg.get("rope_local_base_freq") or 0.0)
             if arch.startswith("Gemma3")
             else (slide_theta if arch.startswith("Olmo3") else 0.0),
-            mlp_act=("gelu_tanh" if arch.startswith("Gemma") else "silu"),
+            mlp_act=("gelu_tanh" if arch.startswith("Gemma")
+                     else "relu2" if arch.startswith("Arcee")
+                     else "silu"),
+            mlp_no_gate=arch.startswith("Arcee"),
         )
 
     @classmethod

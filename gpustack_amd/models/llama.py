@@ -515,12 +515,15 @@ class MLP(nn.Module):
         self.comm = comm
         h = spec.hidden_size
         self.i = spec.intermediate_size // tp_size
-        self.gate_up_w = nn.Parameter(torch.empty(2 * self.i, h, dtype=dtype), requires_grad=False)
+        self.no_gate = spec.mlp_no_gate  # Arcee: down(act(up(x)))
+        gu_rows = self.i if self.no_gate else 2 * self.i
+        self.gate_up_w = nn.Parameter(torch.empty(gu_rows, h, dtype=dtype), requires_grad=False)
         self.down_w = nn.Parameter(torch.empty(h, self.i, dtype=dtype), requires_grad=False)
         self.gate_up_pack: W4Pack | None = None  # W4 runtime (qlinear)
         self.down_pack: W4Pack | None = None
         self.layer_idx = 0  # set by LlamaForCausalLM
-        self._gu_projs = [("gate_proj", 0, self.i), ("up_proj", self.i, self.i)]
+        self._gu_projs = ([("up_proj", 0, self.i)] if self.no_gate else
+                          [("gate_proj", 0, self.i), ("up_proj", self.i, self.i)])
         self._down_projs = [("down_proj", 0, h)]
         self.act = spec.mlp_act
 
@@ -528,7 +531,11 @@ class MLP(nn.Module):
         gu = qlinear(x, self.gate_up_w, self.gate_up_pack)
         if meta is not None and meta.lora is not None:
             meta.lora.apply(self.layer_idx, x, gu, self._gu_projs)
-        if self.act == "gelu_tanh":
+        if self.no_gate:
+            # Arcee: single up projection, relu^2 activation
+            act = (torch.square(F.relu(gu)) if self.act == "relu2"
+                   else F.silu(gu))
+        elif self.act == "gelu_tanh":
             # Gemma GeGLU (HF gelu_pytorch_tanh); torch pointwise path —
             # the fused CDNA4 geglu variant is r3 (same slot as the
             # GPT-OSS clamped-swiglu kernel work)

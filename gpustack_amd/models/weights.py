@@ -73,14 +73,18 @@ def random_init(model, cfg: EngineConfig) -> None:
                 _random_init_moe_layer(layer, spec, li, seed, dtype, device,
                                        tp, rank)
             else:
-                gate = _gen((spec.intermediate_size, spec.hidden_size),
-                            f"{li}.gate", seed, dtype, device)
                 up = _gen((spec.intermediate_size, spec.hidden_size),
                           f"{li}.up", seed, dtype, device)
-                layer.mlp.gate_up_w.copy_(torch.cat([
-                    gate[rank * i_loc:(rank + 1) * i_loc],
-                    up[rank * i_loc:(rank + 1) * i_loc],
-                ]))
+                if getattr(layer.mlp, "no_gate", False):
+                    layer.mlp.gate_up_w.copy_(
+                        up[rank * i_loc:(rank + 1) * i_loc])
+                else:
+                    gate = _gen((spec.intermediate_size, spec.hidden_size),
+                                f"{li}.gate", seed, dtype, device)
+                    layer.mlp.gate_up_w.copy_(torch.cat([
+                        gate[rank * i_loc:(rank + 1) * i_loc],
+                        up[rank * i_loc:(rank + 1) * i_loc],
+                    ]))
                 down = _gen((spec.hidden_size, spec.intermediate_size),
                             f"{li}.down", seed, dtype, device)
                 layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
@@ -121,12 +125,15 @@ def random_init(model, cfg: EngineConfig) -> None:
         if hasattr(layer.mlp, "router_w"):
             _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank)
         else:
-            gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
             up = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.up", seed, dtype, device)
-            layer.mlp.gate_up_w.copy_(torch.cat([
-                gate[rank * i_loc:(rank + 1) * i_loc],
-                up[rank * i_loc:(rank + 1) * i_loc],
-            ]))
+            if getattr(layer.mlp, "no_gate", False):
+                layer.mlp.gate_up_w.copy_(up[rank * i_loc:(rank + 1) * i_loc])
+            else:
+                gate = _gen((spec.intermediate_size, spec.hidden_size), f"{li}.gate", seed, dtype, device)
+                layer.mlp.gate_up_w.copy_(torch.cat([
+                    gate[rank * i_loc:(rank + 1) * i_loc],
+                    up[rank * i_loc:(rank + 1) * i_loc],
+                ]))
             down = _gen((spec.hidden_size, spec.intermediate_size), f"{li}.down", seed, dtype, device)
             layer.mlp.down_w.copy_(down[:, rank * i_loc:(rank + 1) * i_loc])
         layer.input_norm.fill_(1.0)
@@ -398,6 +405,12 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                 ]))
                 dn = get(ep + dname + ".weight")
                 layer.mlp.down_w[e].copy_(dn[:, rank * mi_loc:(rank + 1) * mi_loc])
+        elif getattr(layer.mlp, "no_gate", False):
+            # Arcee: up_proj only
+            layer.mlp.gate_up_w.copy_(
+                row_shard(get(p + "mlp.up_proj.weight"), i_loc))
+            dn = get(p + "mlp.down_proj.weight")
+            layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
         else:
             layer.mlp.gate_up_w.copy_(torch.cat([
                 row_shard(get(p + "mlp.gate_proj.weight"), i_loc),
