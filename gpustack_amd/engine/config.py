@@ -106,6 +106,9 @@ class ModelSpec:
     final_logit_softcap: float = 0.0
     attn_scale: float = 0.0              # 0 = 1/sqrt(head_dim)
     mlp_act: str = "silu"                # "silu" | "gelu_tanh" | "relu2"
+    norm_type: str = "rmsnorm"           # "layernorm": Cohere mean-centered
+    parallel_block: bool = False         # Cohere: x + attn(ln(x)) + mlp(ln(x))
+    logits_multiplier: float = 0.0       # Cohere logit_scale (multiplies)
     mlp_no_gate: bool = False            # Arcee: down(act(up(x))), no gate
 
     @property
@@ -167,6 +170,10 @@ class ModelSpec:
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelSpec":
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        if arch.startswith("Cohere") and cfg.get("use_qk_norm"):
+            raise NotImplementedError(
+                "Cohere use_qk_norm (per-head LayerNorm, Command-A) is "
+                "not implemented yet")
         # OLMo-3 keys rope_parameters BY LAYER TYPE: full_attention
         # defines the main rope (theta + any scaling); sliding layers
         # rope unscaled at their own theta (the Gemma-3 dual-cache path).
@@ -194,7 +201,9 @@ class ModelSpec:
             rope_theta=cfg.get("rope_theta")
             or (cfg.get("rope_parameters") or {}).get("rope_theta", 10000.0),
             rope_scaling=cls._rope_scaling_from(cfg),
-            rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
+            rms_norm_eps=(cfg.get("layer_norm_eps", 1e-5)
+                          if arch.startswith("Cohere")
+                          else cfg.get("rms_norm_eps", 1e-6)),
             max_position_embeddings=cfg.get("max_position_embeddings", 4096),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=(arch.startswith("Qwen2")
@@ -231,7 +240,7 @@ class ModelSpec:
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
             if (arch.startswith("GptOss") or arch.startswith("Gemma")
-                or arch.startswith(("Olmo3", "Exaone4")))
+                or arch.startswith(("Olmo3", "Exaone4", "Cohere2")))
             else 0,
             layer_types=tuple(cfg["layer_types"])
             if cfg.get("layer_types") else (
@@ -274,15 +283,21 @@ class ModelSpec:
             if arch.startswith("Granite") else 0.0,
             logits_scaling=(cfg.get("logits_scaling") or 0.0)
             if arch.startswith("Granite") else 0.0,
-            rope_mode=("pairwise" if arch.startswith("Ernie4_5")
+            rope_mode=("pairwise"
+                       if arch.startswith(("Ernie4_5", "Cohere"))
                        else "neox"),
+            norm_type=("layernorm" if arch.startswith("Cohere")
+                       else "rmsnorm"),
+            parallel_block=arch.startswith("Cohere"),
+            logits_multiplier=(cfg.get("logit_scale") or 0.0)
+            if arch.startswith("Cohere") else 0.0,
             no_rope_layers=(tuple(cfg["no_rope_layers"])
                             if cfg.get("no_rope_layers") else (
                                 # EXAONE-4 hybrid: global-NoPE — rope on
                                 # sliding layers only, full layers unroped
                                 tuple(1 if lt == "sliding_attention" else 0
                                       for lt in cfg["layer_types"])
-                                if (arch.startswith("Exaone4")
+                                if (arch.startswith(("Exaone4", "Cohere2"))
                                     and cfg.get("sliding_window")
                                     and cfg.get("layer_types")) else
                                 tuple(0 if (i + 1) % 4 == 0 else 1

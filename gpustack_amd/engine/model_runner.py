@@ -221,6 +221,7 @@ class ModelRunner:
         if torch.device(cfg.device).type == "cuda" and (
                 cfg.spec.attention_sinks or cfg.spec.sliding_window
                 or cfg.spec.attn_logit_softcap
+                or cfg.spec.norm_type != "rmsnorm"
                 or cfg.spec.head_dim not in (128,)) \
                 and not cfg.spec.kv_lora_rank \
                 and os.environ.get("GPUSTACK_AMD_OSS_KERNELS") != "1":
@@ -229,8 +230,9 @@ class ModelRunner:
             # inside hipGraph capture on the first forward
             raise NotImplementedError(
                 f"{cfg.spec.architecture} GPU serving needs the gated "
-                "CDNA4 kernel variants (sinks/window/softcap/head_dim "
-                f"{cfg.spec.head_dim}) — set GPUSTACK_AMD_OSS_KERNELS=1 "
+                "CDNA4 kernel variants (sinks/window/softcap/layernorm/"
+                f"head_dim {cfg.spec.head_dim}) — set "
+                "GPUSTACK_AMD_OSS_KERNELS=1 "
                 "after the r3 validation pass; CPU serving is available")
         if cfg.spec.kv_lora_rank:
             # MLA (DeepSeek): latent attention serves on the CPU oracle
@@ -244,7 +246,6 @@ class ModelRunner:
                 raise ValueError("fp8 KV is not supported with MLA latent "
                                  "caches yet")
         self.device = torch.device(cfg.device)
-        import os
 
         if self.device.type == "cuda" and os.environ.get("GPUSTACK_AMD_BLASLT", "1") == "1":
             try:  # hipBLASLt picks much better skinny-GEMM kernels than rocBLAS
@@ -304,7 +305,6 @@ class ModelRunner:
         by scripts/tune_gemms once on an MI355X) is loaded read-only; set
         GPUSTACK_AMD_TUNE=1 to re-tune and write a fresh table.
         """
-        import os
         from pathlib import Path
 
         try:

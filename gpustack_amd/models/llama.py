@@ -904,6 +904,16 @@ class MoEMLP(nn.Module):
         _ck("combine")
 
 
+
+def _layer_norm(out, x, w, eps):
+    """Cohere mean-centered LayerNorm (weight only, no bias), computed in
+    fp32 like HF CohereLayerNorm (modeling_cohere.py) and cast back."""
+    xf = x.float()
+    mu = xf.mean(-1, keepdim=True)
+    var = (xf - mu).pow(2).mean(-1, keepdim=True)
+    out.copy_(((xf - mu) * torch.rsqrt(var + eps) * w.float()).to(x.dtype))
+
+
 class DecoderLayer(nn.Module):
     def __init__(self, spec: ModelSpec, tp_size: int, comm: Communicator,
                  dtype, layer_idx: int = 0):
@@ -941,6 +951,9 @@ class DecoderLayer(nn.Module):
 
     def forward(self, x, residual, meta, cos_sin, k_cache, v_cache):
         eps = self.spec.rms_norm_eps
+        if self.spec.parallel_block:
+            return self._forward_parallel(x, meta, cos_sin, k_cache,
+                                          v_cache, eps)
         if self.spec.sandwich_norms:
             return self._forward_sandwich(x, meta, cos_sin, k_cache,
                                           v_cache, eps)
@@ -963,6 +976,17 @@ class DecoderLayer(nn.Module):
         if rm:
             m = m * rm
         return m, residual
+
+    def _forward_parallel(self, x, meta, cos_sin, k_cache, v_cache, eps):
+        """Cohere layer flow (true hidden stream, residual sentinel
+        None): one shared input LayerNorm feeds BOTH sublayers and their
+        outputs add to the residual together:
+            x = x + attn(ln(x)) + mlp(ln(x))"""
+        h = torch.empty_like(x)
+        _layer_norm(h, x, self.input_norm, eps)
+        a = self.attn(h, meta, cos_sin, k_cache, v_cache)
+        m = self.mlp(h, meta)
+        return x + a + m, None
 
     def _forward_norm_after(self, x, meta, cos_sin, k_cache, v_cache, eps):
         """OLMo-2 layer flow (true hidden stream, residual sentinel None):
@@ -1112,7 +1136,10 @@ class LlamaForCausalLM(nn.Module):
                 s = (x.float() + residual.float()).to(self.dtype)
                 self.comm.send_hidden(s)
             return None
-        if residual is None:
+        if self.spec.norm_type == "layernorm":
+            assert residual is None  # parallel_block carries true stream
+            _layer_norm(x, x, self.final_norm, self.spec.rms_norm_eps)
+        elif residual is None:
             ops.rms_norm(x, x, self.final_norm, self.spec.rms_norm_eps)
         else:
             ops.fused_add_rms_norm(x, residual, self.final_norm,
@@ -1126,6 +1153,8 @@ class LlamaForCausalLM(nn.Module):
             logits = torch.tanh(logits / cap) * cap
         if self.spec.logits_scaling:
             logits = logits / self.spec.logits_scaling
+        if self.spec.logits_multiplier:
+            logits = logits * self.spec.logits_multiplier
         if return_both:  # draft-model speculative needs the features too
             return logits, hidden
         return logits

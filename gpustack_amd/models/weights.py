@@ -418,7 +418,12 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             ]))
             dn = get(p + "mlp.down_proj.weight")
             layer.mlp.down_w.copy_(dn[:, rank * i_loc:(rank + 1) * i_loc])
-        if spec.norm_after:
+        if spec.parallel_block:
+            # Cohere: ONE shared input LayerNorm per layer; the
+            # post_attn_norm slot is unused (left at identity)
+            layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+            layer.post_attn_norm.fill_(1.0)
+        elif spec.norm_after:
             # OLMo-2: no input norms — our input_norm slot holds the
             # post-attention norm, post_attn_norm the post-ffn one
             layer.input_norm.copy_(
