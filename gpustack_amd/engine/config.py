@@ -208,6 +208,9 @@ class ModelSpec:
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=(arch.startswith("Qwen2")
                             or arch.startswith("SeedOss")
+                            or (arch in ("GlmForCausalLM",
+                                         "Glm4ForCausalLM")
+                                and bool(cfg.get("attention_bias", True)))
                             or (arch.startswith("Ernie4_5")
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
@@ -263,7 +266,10 @@ class ModelSpec:
             v_head_dim=cfg.get("v_head_dim", 0) or 0,
             rope_interleave=bool(cfg.get("rope_interleave", True)),
             sandwich_norms=(arch.startswith("Gemma2")
-                            or arch.startswith("Gemma3")),
+                            or arch.startswith("Gemma3")
+                            # GLM-4 dense: same four-norm layer flow,
+                            # plain RMSNorm weights (no (1+w) offset)
+                            or arch == "Glm4ForCausalLM"),
             norm_after=arch.startswith(("Olmo2", "Olmo3", "Exaone4")),
             qk_norm_full=arch.startswith(("Olmo2", "Olmo3")),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
@@ -284,7 +290,9 @@ class ModelSpec:
             logits_scaling=(cfg.get("logits_scaling") or 0.0)
             if arch.startswith("Granite") else 0.0,
             rope_mode=("pairwise"
-                       if arch.startswith(("Ernie4_5", "Cohere"))
+                       if (arch.startswith(("Ernie4_5", "Cohere"))
+                           or arch in ("GlmForCausalLM",
+                                       "Glm4ForCausalLM"))
                        else "neox"),
             norm_type=("layernorm" if arch.startswith("Cohere")
                        else "rmsnorm"),

@@ -430,6 +430,17 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                 get(p + "post_attention_layernorm.weight"))
             layer.post_attn_norm.copy_(
                 get(p + "post_feedforward_layernorm.weight"))
+        elif spec.sandwich_norms \
+                and p + "post_self_attn_layernorm.weight" in tensors:
+            # GLM-4 dense: same four-norm sandwich flow under different
+            # HF names, plain RMSNorm weights (no offset)
+            layer.input_norm.copy_(get(p + "input_layernorm.weight"))
+            layer.post_attn_norm.copy_(
+                get(p + "post_self_attn_layernorm.weight"))
+            layer.pre_ff_norm.copy_(
+                get(p + "post_attention_layernorm.weight"))
+            layer.post_ff_norm.copy_(
+                get(p + "post_mlp_layernorm.weight"))
         elif spec.sandwich_norms:
             # Gemma-2 RMSNorm multiplies by (1 + w); store the EFFECTIVE
             # weight so the shared rms_norm kernel applies
