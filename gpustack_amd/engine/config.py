@@ -215,6 +215,7 @@ class ModelSpec:
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
                      or arch.startswith(("Olmo2", "Olmo3", "Exaone4"))
+                     or arch.startswith(("Olmoe", "MiniMaxM2"))
                      or arch.startswith("HunYuan")
                      or bool(cfg.get("use_qk_norm", False))),
             qk_norm_after_rope=arch.startswith("HunYuan"),
@@ -230,6 +231,7 @@ class ModelSpec:
             norm_topk_prob=bool(cfg.get("norm_topk_prob", True)),
             router_mode=("sigmoid_bias"
                          if (arch.startswith("Glm4Moe")
+                             or arch.startswith("MiniMaxM2")
                              or arch.startswith("Deepseek"))
                          else "softmax"),
             n_shared_experts=cfg.get("n_shared_experts", 0) or 0,
@@ -239,7 +241,10 @@ class ModelSpec:
             topk_group=cfg.get("topk_group", 1) or 1,
             partial_rotary_factor=(cfg.get("partial_rotary_factor")
                                    or (cfg.get("rope_parameters") or {})
-                                   .get("partial_rotary_factor") or 1.0),
+                                   .get("partial_rotary_factor")
+                                   # MiniMax checkpoints name it rotary_dim
+                                   or (cfg["rotary_dim"] / hd
+                                       if cfg.get("rotary_dim") else 1.0)),
             attention_sinks=arch.startswith("GptOss"),
             sliding_window=(cfg.get("sliding_window") or 0)
             if (arch.startswith("GptOss") or arch.startswith("Gemma")
@@ -271,7 +276,8 @@ class ModelSpec:
                             # plain RMSNorm weights (no (1+w) offset)
                             or arch == "Glm4ForCausalLM"),
             norm_after=arch.startswith(("Olmo2", "Olmo3", "Exaone4")),
-            qk_norm_full=arch.startswith(("Olmo2", "Olmo3")),
+            qk_norm_full=arch.startswith(("Olmo2", "Olmo3", "Olmoe",
+                                          "MiniMaxM2")),
             embed_scale=(cfg.get("hidden_size", 4096) ** 0.5
                          if arch.startswith("Gemma")
                          else (cfg.get("embedding_multiplier") or 0.0)

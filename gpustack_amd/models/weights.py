@@ -363,9 +363,13 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
             router = (p + "mlp.gate.weight" if p + "mlp.gate.weight" in tensors
                       else p + "block_sparse_moe.gate.weight")
             layer.mlp.router_w.copy_(get(router))
-            if layer.mlp.router_bias is not None                     and p + "mlp.gate.e_score_correction_bias" in tensors:
-                layer.mlp.router_bias.copy_(
-                    tensors[p + "mlp.gate.e_score_correction_bias"].float())
+            if layer.mlp.router_bias is not None:
+                # DeepSeek keeps it on the gate; MiniMax-M2 on the block
+                for bn in ("mlp.gate.e_score_correction_bias",
+                           "mlp.e_score_correction_bias"):
+                    if p + bn in tensors:
+                        layer.mlp.router_bias.copy_(tensors[p + bn].float())
+                        break
             if layer.mlp.shared_gate_up_w is not None:
                 si = layer.mlp.shared_i
                 sg = get(p + "mlp.shared_experts.gate_proj.weight")
