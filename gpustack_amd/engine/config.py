@@ -41,6 +41,8 @@ class ModelSpec:
     # layers use a plain dense MLP
     router_mode: str = "softmax"          # "softmax" | "sigmoid_bias"
     n_shared_experts: int = 0
+    shared_expert_gated: bool = False    # Qwen2-MoE: sigmoid(gate(x)) scales
+                                         # the shared expert's output
     first_k_dense_replace: int = 0
     routed_scaling_factor: float = 1.0
     n_group: int = 1
@@ -234,7 +236,13 @@ class ModelSpec:
                              or arch.startswith("MiniMaxM2")
                              or arch.startswith("Deepseek"))
                          else "softmax"),
-            n_shared_experts=cfg.get("n_shared_experts", 0) or 0,
+            n_shared_experts=(cfg.get("n_shared_experts", 0) or 0)
+            if not arch.startswith("Qwen2Moe")
+            # Qwen2-MoE sizes the shared expert directly; express it as
+            # a multiple of the routed intermediate (5632 = 4 x 1408)
+            else (cfg.get("shared_expert_intermediate_size", 0)
+                  // max(cfg.get("moe_intermediate_size", 1), 1)),
+            shared_expert_gated=arch.startswith("Qwen2Moe"),
             first_k_dense_replace=cfg.get("first_k_dense_replace", 0) or 0,
             routed_scaling_factor=cfg.get("routed_scaling_factor", 1.0) or 1.0,
             n_group=cfg.get("n_group", 1) or 1,

@@ -604,6 +604,10 @@ class MoEMLP(nn.Module):
         else:
             self.gate_up_b = None
             self.down_b = None
+        self.shared_gate_w = (
+            nn.Parameter(torch.empty(1, h, dtype=dtype),
+                         requires_grad=False)
+            if spec.shared_expert_gated else None)
         if spec.n_shared_experts > 0:
             si = spec.moe_intermediate_size * spec.n_shared_experts // tp_size
             self.shared_i = si
@@ -786,7 +790,12 @@ class MoEMLP(nn.Module):
         act = torch.empty(x.shape[0], self.shared_i, dtype=x.dtype,
                           device=x.device)
         ops.silu_and_mul(act, gu)
-        return out + F.linear(act, self.shared_down_w)
+        sh = F.linear(act, self.shared_down_w)
+        if self.shared_gate_w is not None:
+            # Qwen2-MoE: token-wise sigmoid gate on the shared expert
+            # (gate is tiny [1, h] — replicated, NOT TP-sharded)
+            sh = sh * torch.sigmoid(F.linear(x, self.shared_gate_w))
+        return out + sh
 
     def _fused_ok(self, x) -> bool:
         import os

@@ -177,6 +177,10 @@ def _random_init_moe_layer(layer, spec, li, seed, dtype, device, tp, rank):
     mi_loc = spec.moe_intermediate_size // tp
     layer.mlp.router_w.copy_(_gen((spec.num_experts, spec.hidden_size),
                                   f"{li}.router", seed, dtype, device))
+    if getattr(layer.mlp, "shared_gate_w", None) is not None:
+        layer.mlp.shared_gate_w.copy_(_gen((1, spec.hidden_size),
+                                           f"{li}.sgate", seed, dtype,
+                                           device))
     if layer.mlp.router_bias is not None:
         layer.mlp.router_bias.copy_(_gen((spec.num_experts,), f"{li}.rbias",
                                          seed, torch.float32, device,
@@ -370,14 +374,22 @@ def load_safetensors(model, cfg: EngineConfig, model_dir: str | Path) -> None:
                     if p + bn in tensors:
                         layer.mlp.router_bias.copy_(tensors[p + bn].float())
                         break
+            if getattr(layer.mlp, "shared_gate_w", None) is not None \
+                    and p + "mlp.shared_expert_gate.weight" in tensors:
+                layer.mlp.shared_gate_w.copy_(
+                    get(p + "mlp.shared_expert_gate.weight"))
             if layer.mlp.shared_gate_up_w is not None:
                 si = layer.mlp.shared_i
-                sg = get(p + "mlp.shared_experts.gate_proj.weight")
-                su = get(p + "mlp.shared_experts.up_proj.weight")
+                # DeepSeek/GLM name it shared_expertS, Qwen2-MoE singular
+                sp = (p + "mlp.shared_experts."
+                      if p + "mlp.shared_experts.gate_proj.weight"
+                      in tensors else p + "mlp.shared_expert.")
+                sg = get(sp + "gate_proj.weight")
+                su = get(sp + "up_proj.weight")
                 layer.mlp.shared_gate_up_w.copy_(torch.cat([
                     sg[rank * si:(rank + 1) * si],
                     su[rank * si:(rank + 1) * si]]))
-                sd = get(p + "mlp.shared_experts.down_proj.weight")
+                sd = get(sp + "down_proj.weight")
                 layer.mlp.shared_down_w.copy_(
                     sd[:, rank * si:(rank + 1) * si])
             mi_loc = spec.moe_intermediate_size // tp
