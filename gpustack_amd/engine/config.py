@@ -217,7 +217,7 @@ class ModelSpec:
                                 and bool(cfg.get("use_bias")))),
             qk_norm=(arch.startswith("Qwen3") or arch.startswith("Gemma3")
                      or arch.startswith(("Olmo2", "Olmo3", "Exaone4"))
-                     or arch.startswith(("Olmoe", "MiniMaxM2"))
+                     or arch.startswith(("Olmoe", "MiniMaxM2", "Dots1"))
                      or arch.startswith("HunYuan")
                      or bool(cfg.get("use_qk_norm", False))),
             qk_norm_after_rope=arch.startswith("HunYuan"),
@@ -233,7 +233,7 @@ class ModelSpec:
             norm_topk_prob=bool(cfg.get("norm_topk_prob", True)),
             router_mode=("sigmoid_bias"
                          if (arch.startswith("Glm4Moe")
-                             or arch.startswith("MiniMaxM2")
+                             or arch.startswith(("MiniMaxM2", "Dots1"))
                              or arch.startswith("Deepseek"))
                          else "softmax"),
             n_shared_experts=(cfg.get("n_shared_experts", 0) or 0)
