@@ -304,7 +304,8 @@ class ModelSpec:
             logits_scaling=(cfg.get("logits_scaling") or 0.0)
             if arch.startswith("Granite") else 0.0,
             rope_mode=("pairwise"
-                       if (arch.startswith(("Ernie4_5", "Cohere"))
+                       if (arch.startswith(("Ernie4_5", "Cohere",
+                                            "Helium"))
                            or arch in ("GlmForCausalLM",
                                        "Glm4ForCausalLM"))
                        else "neox"),
